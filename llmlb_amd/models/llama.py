@@ -1,0 +1,182 @@
+"""Llama-3 family for the MI355X backend (bf16, TP-aware, paged KV).
+
+This is the flagship model family served by the backends behind the
+gateway (BASELINE configs 2-4: Llama-3-8B per-GPU replicas and
+Llama-3-70B TP=4 quads).  GEMMs go through torch.nn.functional.linear
+(hipBLASLt on ROCm); the fused hot ops (rmsnorm+residual, rope, swiglu,
+paged decode attention, kv append) are the hand-written gfx950 kernels
+in llmlb_amd.ops.
+
+Weights are random-initialised (no network for checkpoints; BASELINE
+prescribes synthetic prompts + random-init weights).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..ops import reference as op_ref
+from ..parallel.tp import (ColumnParallelLinear, RowParallelLinear,
+                           TPContext, get_tp_context)
+
+
+@dataclass
+class LlamaConfig:
+    hidden: int = 4096
+    layers: int = 32
+    q_heads: int = 32
+    kv_heads: int = 8
+    head_dim: int = 128
+    inter: int = 14336
+    vocab: int = 128256
+    rope_theta: float = 500000.0
+    max_pos: int = 8192
+    norm_eps: float = 1e-5
+    name: str = "llama-3-8b"
+
+    @classmethod
+    def llama3_8b(cls) -> "LlamaConfig":
+        return cls()
+
+    @classmethod
+    def llama3_70b(cls) -> "LlamaConfig":
+        return cls(hidden=8192, layers=80, q_heads=64, kv_heads=8,
+                   head_dim=128, inter=28672, vocab=128256,
+                   name="llama-3-70b")
+
+    @classmethod
+    def tiny(cls) -> "LlamaConfig":
+        """2-layer Llama-3-shaped model for smoke/tests."""
+        return cls(hidden=256, layers=2, q_heads=8, kv_heads=2, head_dim=32,
+                   inter=512, vocab=1024, max_pos=512, name="llama-tiny")
+
+    @classmethod
+    def small(cls) -> "LlamaConfig":
+        """~0.5B-shaped model for 1-GPU quick benches."""
+        return cls(hidden=1024, layers=8, q_heads=16, kv_heads=4,
+                   head_dim=64, inter=2816, vocab=32000, max_pos=4096,
+                   name="llama-small")
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, tp: TPContext, dtype):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp
+        ws = tp.world_size
+        assert cfg.q_heads % ws == 0 and cfg.kv_heads % ws == 0, \
+            "TP degree must divide q_heads and kv_heads"
+        self.q_local = cfg.q_heads // ws
+        self.kv_local = cfg.kv_heads // ws
+        D = cfg.head_dim
+        qkv_out = (cfg.q_heads + 2 * cfg.kv_heads) * D
+        self.input_norm_w = nn.Parameter(
+            torch.empty(cfg.hidden, dtype=dtype), requires_grad=False)
+        self.post_norm_w = nn.Parameter(
+            torch.empty(cfg.hidden, dtype=dtype), requires_grad=False)
+        self.qkv = ColumnParallelLinear(cfg.hidden, qkv_out, tp, dtype)
+        self.o_proj = RowParallelLinear(cfg.q_heads * D, cfg.hidden, tp,
+                                        dtype)
+        self.gate_up = ColumnParallelLinear(cfg.hidden, 2 * cfg.inter, tp,
+                                            dtype)
+        self.down = RowParallelLinear(cfg.inter, cfg.hidden, tp, dtype)
+        self.scale = D ** -0.5
+
+    def split_qkv(self, qkv: torch.Tensor):
+        D = self.cfg.head_dim
+        T = qkv.shape[0]
+        q, k, v = qkv.split(
+            [self.q_local * D, self.kv_local * D, self.kv_local * D], dim=-1)
+        return (q.view(T, self.q_local, D).contiguous(),
+                k.view(T, self.kv_local, D).contiguous(),
+                v.view(T, self.kv_local, D).contiguous())
+
+
+class LlamaForCausalLM(nn.Module):
+    """Functional-style Llama; attention dispatch is owned by the engine
+    (prefill SDPA vs paged decode), the model exposes per-layer pieces."""
+
+    def __init__(self, cfg: LlamaConfig, tp: TPContext | None = None,
+                 dtype: torch.dtype = torch.bfloat16,
+                 device: str | torch.device = "cpu", seed: int = 1234):
+        super().__init__()
+        self.cfg = cfg
+        self.tp = tp or get_tp_context()
+        self.dtype = dtype
+        self.embed = nn.Parameter(
+            torch.empty(cfg.vocab, cfg.hidden, dtype=dtype),
+            requires_grad=False)
+        self.layers_ = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, self.tp, dtype)
+             for _ in range(cfg.layers)])
+        self.final_norm_w = nn.Parameter(
+            torch.empty(cfg.hidden, dtype=dtype), requires_grad=False)
+        self.lm_head = nn.Parameter(
+            torch.empty(cfg.vocab, cfg.hidden, dtype=dtype),
+            requires_grad=False)
+        self.cos_sin = op_ref.build_cos_sin_table(
+            cfg.max_pos, cfg.head_dim, cfg.rope_theta)
+        self._init_weights(seed)
+        self.to(device)
+        self.cos_sin = self.cos_sin.to(device)
+        self.device = torch.device(device)
+
+    def _init_weights(self, seed: int) -> None:
+        g = torch.Generator().manual_seed(seed)
+        std = 0.02
+
+        def init_(p: torch.Tensor):
+            p.data.copy_(torch.randn(p.shape, generator=g,
+                                     dtype=torch.float32).mul_(std)
+                         .to(p.dtype))
+
+        init_(self.embed)
+        init_(self.lm_head)
+        self.final_norm_w.data.fill_(1.0)
+        for layer in self.layers_:
+            layer.input_norm_w.data.fill_(1.0)
+            layer.post_norm_w.data.fill_(1.0)
+            init_(layer.qkv.weight)
+            init_(layer.o_proj.weight)
+            init_(layer.gate_up.weight)
+            init_(layer.down.weight)
+
+    # -- forward pieces (called by the engine) ---------------------------
+
+    def embed_tokens(self, input_ids: torch.Tensor) -> torch.Tensor:
+        return self.embed[input_ids]
+
+    def layer_attn_prelude(self, layer_idx: int, h: torch.Tensor,
+                           residual: torch.Tensor | None,
+                           positions: torch.Tensor):
+        """norm -> qkv -> rope.  Returns (q, k, v, residual)."""
+        layer = self.layers_[layer_idx]
+        normed, residual = ops.rmsnorm(h, layer.input_norm_w,
+                                       self.cfg.norm_eps, residual)
+        if residual is None:
+            residual = h
+        qkv = layer.qkv(normed)
+        q, k, v = layer.split_qkv(qkv)
+        q, k = ops.rope_apply(q, k, self.cos_sin, positions)
+        return q, k, v, residual
+
+    def layer_attn_epilogue(self, layer_idx: int, attn_out: torch.Tensor,
+                            residual: torch.Tensor):
+        """o_proj -> +residual -> norm -> mlp.  Returns (h, residual)."""
+        layer = self.layers_[layer_idx]
+        T = attn_out.shape[0]
+        o = layer.o_proj(attn_out.reshape(T, -1))
+        normed, residual = ops.rmsnorm(o, layer.post_norm_w,
+                                       self.cfg.norm_eps, residual)
+        gateup = layer.gate_up(normed)
+        act = ops.swiglu(gateup)
+        h = layer.down(act)
+        return h, residual
+
+    def final(self, h: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
+        normed, _ = ops.rmsnorm(h, self.final_norm_w, self.cfg.norm_eps,
+                                residual)
+        return torch.nn.functional.linear(normed, self.lm_head)

@@ -1,0 +1,210 @@
+"""llmlb_amd.ops — MI355X (gfx950) HIP kernels with torch fp32 fallbacks.
+
+The native library (``libllmlb_kernels.so``, built in-tree from
+``csrc/kernels.hip`` by ``__graft_entry__.build()``) is loaded via ctypes
+and called with raw device pointers on the current torch HIP stream.
+
+Policy (required by the harness): on a GPU box the HIP path MUST run —
+if a CUDA/HIP tensor reaches an op and the native library is missing,
+we raise instead of silently falling back to eager PyTorch.  CPU tensors
+always use the fp32 reference implementations (tests, CI without GPU).
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from pathlib import Path
+
+import torch
+
+from . import reference
+
+_LIB_NAME = "libllmlb_kernels.so"
+_LIB_PATH = Path(__file__).resolve().parent / _LIB_NAME
+_lib: ctypes.CDLL | None = None
+_load_error: str | None = None
+
+
+def _try_load() -> None:
+    global _lib, _load_error
+    if _lib is not None:
+        return
+    if not _LIB_PATH.exists():
+        _load_error = f"{_LIB_PATH} not built (run __graft_entry__.build())"
+        return
+    try:
+        _lib = ctypes.CDLL(str(_LIB_PATH))
+    except OSError as e:  # e.g. no ROCm runtime on this host
+        _lib = None
+        _load_error = str(e)
+        return
+    _lib.ll_rmsnorm.restype = ctypes.c_int
+    _lib.ll_rope.restype = ctypes.c_int
+    _lib.ll_swiglu.restype = ctypes.c_int
+    _lib.ll_kv_append.restype = ctypes.c_int
+    _lib.ll_paged_attn_decode.restype = ctypes.c_int
+
+
+_try_load()
+
+
+def native_available() -> bool:
+    return _lib is not None
+
+
+def native_load_error() -> str | None:
+    return _load_error
+
+
+class NativeKernelsMissing(RuntimeError):
+    pass
+
+
+def _require_native() -> ctypes.CDLL:
+    if _lib is None:
+        raise NativeKernelsMissing(
+            "HIP tensor reached llmlb_amd.ops but the native kernel library "
+            f"is not loaded: {_load_error}. Refusing silent eager fallback "
+            "on GPU — build with __graft_entry__.build()."
+        )
+    return _lib
+
+
+def _stream() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _ptr(t: torch.Tensor) -> ctypes.c_void_p:
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _check(rc: int, name: str) -> None:
+    if rc != 0:
+        raise RuntimeError(f"{name} failed with code {rc}")
+
+
+# ---------------------------------------------------------------------------
+# public ops
+# ---------------------------------------------------------------------------
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
+            residual: torch.Tensor | None = None):
+    """out = rmsnorm(x (+ residual)) * weight.
+
+    Returns (out, residual'): when ``residual`` is given it is updated in
+    place to x + residual and returned.  x: [T, H] bf16 (GPU path).
+    """
+    if not x.is_cuda:
+        return reference.rmsnorm(x, weight, eps, residual)
+    lib = _require_native()
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    out = torch.empty_like(x)
+    n_tokens = x.numel() // x.shape[-1]
+    rc = lib.ll_rmsnorm(
+        _ptr(out), _ptr(x), _ptr(weight),
+        _ptr(residual) if residual is not None else None,
+        ctypes.c_int(n_tokens), ctypes.c_int(x.shape[-1]),
+        ctypes.c_float(eps), _stream())
+    _check(rc, "ll_rmsnorm")
+    return out, residual
+
+
+def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
+               positions: torch.Tensor):
+    """In-place rotate-half RoPE on q [T, n_q, D] and k [T, n_kv, D]."""
+    if not q.is_cuda:
+        return reference.rope_apply(q, k, cos_sin, positions)
+    lib = _require_native()
+    assert q.dtype == torch.bfloat16 and q.is_contiguous() and k.is_contiguous()
+    assert cos_sin.dtype == torch.float32
+    pos = positions.to(torch.int32)
+    rc = lib.ll_rope(
+        _ptr(q), _ptr(k), _ptr(cos_sin), _ptr(pos),
+        ctypes.c_int(q.shape[0]), ctypes.c_int(q.shape[1]),
+        ctypes.c_int(k.shape[1]), ctypes.c_int(q.shape[2]), _stream())
+    _check(rc, "ll_rope")
+    return q, k
+
+
+def swiglu(gateup: torch.Tensor) -> torch.Tensor:
+    """gateup [T, 2*inter] -> silu(gate) * up  [T, inter]."""
+    if not gateup.is_cuda:
+        return reference.swiglu(gateup)
+    lib = _require_native()
+    assert gateup.dtype == torch.bfloat16 and gateup.is_contiguous()
+    inter = gateup.shape[-1] // 2
+    n_tokens = gateup.numel() // gateup.shape[-1]
+    out = torch.empty(*gateup.shape[:-1], inter, dtype=gateup.dtype,
+                      device=gateup.device)
+    rc = lib.ll_swiglu(_ptr(out), _ptr(gateup),
+                       ctypes.c_longlong(n_tokens), ctypes.c_int(inter),
+                       _stream())
+    _check(rc, "ll_swiglu")
+    return out
+
+
+def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
+              v: torch.Tensor, slots: torch.Tensor) -> None:
+    """Scatter k/v [T, n_kv, D] into paged caches at flat slot ids."""
+    block_size = kcache.shape[2]
+    if not kcache.is_cuda:
+        reference.kv_append(kcache, vcache, k, v, slots, block_size)
+        return
+    lib = _require_native()
+    assert k.is_contiguous() and v.is_contiguous()
+    slots_i = slots.to(torch.int32)
+    rc = lib.ll_kv_append(
+        _ptr(kcache), _ptr(vcache), _ptr(k), _ptr(v), _ptr(slots_i),
+        ctypes.c_int(k.shape[0]), ctypes.c_int(k.shape[1]),
+        ctypes.c_int(k.shape[2]), ctypes.c_int(block_size), _stream())
+    _check(rc, "ll_kv_append")
+
+
+def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
+                      vcache: torch.Tensor, block_tables: torch.Tensor,
+                      seq_lens: torch.Tensor, scale: float,
+                      n_parts: int | None = None,
+                      workspace: tuple | None = None) -> torch.Tensor:
+    """Decode attention over the paged KV cache.
+
+    q: [S, n_q, D] bf16; caches [num_blocks, n_kv, block_size, D].
+    ``n_parts`` splits long sequences across workgroups (flash-decode);
+    chosen automatically to fill the 256-CU chip when not given.
+    """
+    if not q.is_cuda:
+        return reference.paged_attn_decode(q, kcache, vcache, block_tables,
+                                           seq_lens, scale)
+    lib = _require_native()
+    S, n_q, D = q.shape
+    n_kv = kcache.shape[1]
+    block_size = kcache.shape[2]
+    if n_parts is None:
+        # fill ≫256 workgroups on the 256-CU chip (8 XCDs)
+        base = S * n_kv
+        n_parts = 1
+        while base * n_parts < 512 and n_parts < 16:
+            n_parts *= 2
+    out = torch.empty_like(q)
+    if n_parts > 1:
+        if workspace is not None:
+            ws_acc, ws_ml = workspace
+        else:
+            ws_acc = torch.empty(S, n_q, n_parts, D, dtype=torch.float32,
+                                 device=q.device)
+            ws_ml = torch.empty(S, n_q, n_parts, 2, dtype=torch.float32,
+                                device=q.device)
+        acc_p, ml_p = _ptr(ws_acc), _ptr(ws_ml)
+    else:
+        acc_p = ml_p = None
+    bt = block_tables if block_tables.dtype == torch.int32 \
+        else block_tables.to(torch.int32)
+    sl = seq_lens if seq_lens.dtype == torch.int32 \
+        else seq_lens.to(torch.int32)
+    rc = lib.ll_paged_attn_decode(
+        _ptr(out), acc_p, ml_p, _ptr(q), _ptr(kcache), _ptr(vcache),
+        _ptr(bt), _ptr(sl), ctypes.c_int(S), ctypes.c_int(bt.shape[1]),
+        ctypes.c_int(n_q), ctypes.c_int(n_kv), ctypes.c_int(D),
+        ctypes.c_int(block_size), ctypes.c_int(n_parts),
+        ctypes.c_float(scale), _stream())
+    _check(rc, "ll_paged_attn_decode")
+    return out

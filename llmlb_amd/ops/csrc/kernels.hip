@@ -1,0 +1,555 @@
+// llmlb_amd HIP kernels — hand-written for MI355X (gfx950, CDNA4).
+//
+// These are the inference-backend hot ops for the Llama-3 family served
+// behind the gateway (BASELINE.json configs 2-4).  The reference gateway
+// (akiojin/llmlb) hosts no model kernels; this file is the
+// new MI355X-native backend surface.
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64 lanes everywhere; block sizes are multiples of 64.
+//  * memory-bound ops (rmsnorm/rope/swiglu/kv-append) load bf16 as 16 B
+//    per lane (uint4 = 8 x bf16) — scalar bf16 loads are ~2x slower (G13).
+//  * RoPE uses a host-precomputed cos/sin table (f32) — on-device
+//    sincosf turns a memory-bound op VALU-bound (Appendix B).
+//  * paged decode attention: one workgroup per (seq, kv_head, partition),
+//    4 waves; GQA q-heads of the group are split across waves; K/V pages
+//    are read directly from global memory (the 4 waves share the page
+//    through L1/L2 — LDS staging is not needed for streamed-once data,
+//    common-mistake #7); per-wave online softmax in registers with
+//    4-lanes-per-token shuffle reduction; fp32 accumulation.
+//
+// Compile: hipcc --offload-arch=gfx950 -O3 -std=c++17 -shared -fPIC
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <math.h>
+#include <stdint.h>
+
+#define WAVE 64
+
+typedef __hip_bfloat16 bf16;
+
+__device__ __forceinline__ float bf2f(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = ((unsigned int)u) << 16;
+  return c.f;
+}
+
+// round-to-nearest-even f32 -> bf16 (matches torch's conversion)
+__device__ __forceinline__ unsigned short f2bf(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  unsigned int x = c.i;
+  unsigned int rounding_bias = 0x7FFF + ((x >> 16) & 1);
+  x += rounding_bias;
+  return (unsigned short)(x >> 16);
+}
+
+// unpack a uint4 (8 bf16) into 8 floats
+__device__ __forceinline__ void unpack8(const uint4 v, float* f) {
+  f[0] = bf2f(v.x & 0xffff);        f[1] = bf2f(v.x >> 16);
+  f[2] = bf2f(v.y & 0xffff);        f[3] = bf2f(v.y >> 16);
+  f[4] = bf2f(v.z & 0xffff);        f[5] = bf2f(v.z >> 16);
+  f[6] = bf2f(v.w & 0xffff);        f[7] = bf2f(v.w >> 16);
+}
+
+__device__ __forceinline__ uint4 pack8(const float* f) {
+  uint4 v;
+  v.x = (unsigned int)f2bf(f[0]) | ((unsigned int)f2bf(f[1]) << 16);
+  v.y = (unsigned int)f2bf(f[2]) | ((unsigned int)f2bf(f[3]) << 16);
+  v.z = (unsigned int)f2bf(f[4]) | ((unsigned int)f2bf(f[5]) << 16);
+  v.w = (unsigned int)f2bf(f[6]) | ((unsigned int)f2bf(f[7]) << 16);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm (optionally fused with residual add):
+//   if residual != nullptr:  residual += x;  x' = residual  (updated in place)
+//   out = x' * rsqrt(mean(x'^2) + eps) * weight
+// One 256-thread block per token row; fp32 accumulation; bf16x8 loads.
+// ---------------------------------------------------------------------------
+template <bool FUSE_RESIDUAL>
+__global__ void rmsnorm_kernel(bf16* __restrict__ out,
+                               const bf16* __restrict__ in,
+                               const bf16* __restrict__ weight,
+                               bf16* __restrict__ residual,
+                               int hidden, float eps) {
+  const int row = blockIdx.x;
+  const uint4* inv = reinterpret_cast<const uint4*>(in + (size_t)row * hidden);
+  uint4* resv = FUSE_RESIDUAL
+      ? reinterpret_cast<uint4*>(residual + (size_t)row * hidden) : nullptr;
+  const uint4* wv = reinterpret_cast<const uint4*>(weight);
+  uint4* outv = reinterpret_cast<uint4*>(out + (size_t)row * hidden);
+  const int nvec = hidden >> 3;  // uint4 = 8 bf16
+
+  float ssq = 0.f;
+  // pass 1: (residual add +) sum of squares.  Rows are at most a few K
+  // elements (hidden 4096..16384) so keep the row in registers when it
+  // fits one iteration, else re-read (L1/L2-resident on pass 2).
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    uint4 v = inv[i];
+    float f[8];
+    unpack8(v, f);
+    if (FUSE_RESIDUAL) {
+      float r[8];
+      unpack8(resv[i], r);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] += r[j];
+      resv[i] = pack8(f);  // residual stream carries the sum forward
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ssq += f[j] * f[j];
+  }
+  // wave reduce then cross-wave via LDS
+  for (int off = 32; off > 0; off >>= 1) ssq += __shfl_xor(ssq, off, WAVE);
+  __shared__ float warp_ssq[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_ssq[wid] = ssq;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float total = 0.f;
+#pragma unroll
+  for (int w = 0; w < 4; ++w) total += (w < nw) ? warp_ssq[w] : 0.f;
+  const float rrms = rsqrtf(total / hidden + eps);
+
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    uint4 v = FUSE_RESIDUAL ? resv[i] : inv[i];
+    float f[8], w[8];
+    unpack8(v, f);
+    unpack8(wv[i], w);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = f[j] * rrms * w[j];
+    outv[i] = pack8(f);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (GPT-NeoX rotate-half, Llama-3 style), fused for q and k.
+//   q: [tokens, n_q * D], k: [tokens, n_kv * D], bf16, updated in place.
+//   cos_sin: f32 [max_pos, D]  (cos[0:D/2] then sin[0:D/2] per position)
+// One wave-group slice per (token, head); each thread rotates 2 pairs
+// via 2x bf16 loads; cos/sin from the table (f32).
+// ---------------------------------------------------------------------------
+__global__ void rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
+                            const float* __restrict__ cos_sin,
+                            const int* __restrict__ positions,
+                            int n_tokens, int n_q, int n_kv, int D) {
+  const int half = D / 2;
+  const int heads = n_q + n_kv;
+  // one thread handles 4 rotation pairs (8 bf16 = 16B from each half)
+  const int pairs_per_thread = 4;
+  const int threads_per_head = half / pairs_per_thread;
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long total = (long long)n_tokens * heads * threads_per_head;
+  if (idx >= total) return;
+  const int t = idx / (heads * threads_per_head);
+  const int rem = idx % (heads * threads_per_head);
+  const int h = rem / threads_per_head;
+  const int j0 = (rem % threads_per_head) * pairs_per_thread;
+  const int pos = positions[t];
+
+  bf16* base = (h < n_q) ? (q + ((size_t)t * n_q + h) * D)
+                         : (k + ((size_t)t * n_kv + (h - n_q)) * D);
+  // rotate-half: (x1, x2) at (j, j+half) -> (x1*c - x2*s, x2*c + x1*s)
+  float x1[4], x2[4];
+  uint2 lo = *reinterpret_cast<uint2*>(base + j0);
+  uint2 hi = *reinterpret_cast<uint2*>(base + half + j0);
+  x1[0] = bf2f(lo.x & 0xffff); x1[1] = bf2f(lo.x >> 16);
+  x1[2] = bf2f(lo.y & 0xffff); x1[3] = bf2f(lo.y >> 16);
+  x2[0] = bf2f(hi.x & 0xffff); x2[1] = bf2f(hi.x >> 16);
+  x2[2] = bf2f(hi.y & 0xffff); x2[3] = bf2f(hi.y >> 16);
+  const float* cs = cos_sin + (size_t)pos * D;
+  float4 c = *reinterpret_cast<const float4*>(cs + j0);
+  float4 s = *reinterpret_cast<const float4*>(cs + half + j0);
+  float o1[4], o2[4];
+  o1[0] = x1[0] * c.x - x2[0] * s.x;  o2[0] = x2[0] * c.x + x1[0] * s.x;
+  o1[1] = x1[1] * c.y - x2[1] * s.y;  o2[1] = x2[1] * c.y + x1[1] * s.y;
+  o1[2] = x1[2] * c.z - x2[2] * s.z;  o2[2] = x2[2] * c.z + x1[2] * s.z;
+  o1[3] = x1[3] * c.w - x2[3] * s.w;  o2[3] = x2[3] * c.w + x1[3] * s.w;
+  lo.x = (unsigned)f2bf(o1[0]) | ((unsigned)f2bf(o1[1]) << 16);
+  lo.y = (unsigned)f2bf(o1[2]) | ((unsigned)f2bf(o1[3]) << 16);
+  hi.x = (unsigned)f2bf(o2[0]) | ((unsigned)f2bf(o2[1]) << 16);
+  hi.y = (unsigned)f2bf(o2[2]) | ((unsigned)f2bf(o2[3]) << 16);
+  *reinterpret_cast<uint2*>(base + j0) = lo;
+  *reinterpret_cast<uint2*>(base + half + j0) = hi;
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: out[t, i] = silu(gateup[t, i]) * gateup[t, inter + i]
+// gateup: [tokens, 2*inter] (gate | up), bf16.  Grid-stride, bf16x8.
+// ---------------------------------------------------------------------------
+__global__ void swiglu_kernel(bf16* __restrict__ out,
+                              const bf16* __restrict__ gateup,
+                              long long n_tokens, int inter) {
+  const int nvec = inter >> 3;
+  const long long total = n_tokens * nvec;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (long long)gridDim.x * blockDim.x) {
+    const long long t = idx / nvec;
+    const int i = idx % nvec;
+    const uint4* row = reinterpret_cast<const uint4*>(gateup + t * 2 * inter);
+    uint4 g = row[i];
+    uint4 u = row[nvec + i];
+    float gf[8], uf[8];
+    unpack8(g, gf);
+    unpack8(u, uf);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float x = gf[j];
+      float silu = x / (1.f + __expf(-x));
+      gf[j] = silu * uf[j];
+    }
+    reinterpret_cast<uint4*>(out + t * inter)[i] = pack8(gf);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// KV append (reshape_and_cache): scatter per-token K/V rows into the paged
+// cache.  cache layout: [num_blocks, n_kv, block_size, D] bf16.
+// slots[t] = block_id * block_size + offset   (-1 = skip)
+// One thread per 8 contiguous elements.
+// ---------------------------------------------------------------------------
+__global__ void kv_append_kernel(bf16* __restrict__ kcache,
+                                 bf16* __restrict__ vcache,
+                                 const bf16* __restrict__ k,
+                                 const bf16* __restrict__ v,
+                                 const int* __restrict__ slots,
+                                 int n_tokens, int n_kv, int D,
+                                 int block_size) {
+  const int nvec = D >> 3;
+  const long long total = (long long)n_tokens * n_kv * nvec;
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= total) return;
+  const int t = idx / (n_kv * nvec);
+  const int rem = idx % (n_kv * nvec);
+  const int h = rem / nvec;
+  const int i = rem % nvec;
+  const int slot = slots[t];
+  if (slot < 0) return;
+  const int blk = slot / block_size;
+  const int off = slot % block_size;
+  const size_t src = ((size_t)t * n_kv + h) * D;
+  const size_t dst = (((size_t)blk * n_kv + h) * block_size + off) * D;
+  reinterpret_cast<uint4*>(kcache + dst)[i] =
+      reinterpret_cast<const uint4*>(k + src)[i];
+  reinterpret_cast<uint4*>(vcache + dst)[i] =
+      reinterpret_cast<const uint4*>(v + src)[i];
+}
+
+// ---------------------------------------------------------------------------
+// Paged decode attention (one query token per sequence), GQA, bf16 KV.
+//
+//   q:        [num_seqs, n_q, D] bf16
+//   kcache:   [num_blocks, n_kv, block_size, D] bf16   (block_size = 16)
+//   block_tables: [num_seqs, max_blocks] int32
+//   seq_lens: [num_seqs] int32 (length INCLUDING the current token,
+//             whose K/V is already appended)
+//
+// Grid: (num_seqs, n_kv, n_parts); block = 4 waves (256 threads).
+// Wave w handles q-heads {w, w+4, ...} of the kv-head's GQA group.
+// Within a wave: 4 lanes per token x 16 tokens of a page; each lane
+// computes a 32-element slice of q.k^T, shuffle-reduced over the 4-lane
+// group; online softmax (m shared, l/acc lane-partial, reduced at end).
+//
+// When n_parts > 1 partial (m, l, acc) go to f32 workspace and
+// paged_attn_reduce combines; n_parts == 1 writes bf16 output directly.
+// ---------------------------------------------------------------------------
+#define PA_BLOCK 16        // tokens per KV page
+#define PA_WAVES 4
+
+__global__ __launch_bounds__(PA_WAVES * WAVE)
+void paged_attn_decode_kernel(
+    bf16* __restrict__ out,              // [num_seqs, n_q, D]
+    float* __restrict__ ws_acc,          // [num_seqs, n_q, n_parts, D]
+    float* __restrict__ ws_ml,           // [num_seqs, n_q, n_parts, 2]
+    const bf16* __restrict__ q,
+    const bf16* __restrict__ kcache,
+    const bf16* __restrict__ vcache,
+    const int* __restrict__ block_tables,
+    const int* __restrict__ seq_lens,
+    int n_q, int n_kv, int D, int max_blocks,
+    int n_parts, float scale) {
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int part = blockIdx.z;
+  const int seq_len = seq_lens[seq];
+  const int n_blocks = (seq_len + PA_BLOCK - 1) / PA_BLOCK;
+  const int blocks_per_part = (n_blocks + n_parts - 1) / n_parts;
+  const int blk_begin = part * blocks_per_part;
+  const int blk_end = min(blk_begin + blocks_per_part, n_blocks);
+  const int group = n_q / n_kv;                 // GQA group size
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int t_local = lane >> 2;                // token within page (0..15)
+  const int p = lane & 3;                       // dim-slice (32 elems each)
+  const int DS = D / 4;                         // dim-slice width
+  const int* btab = block_tables + (size_t)seq * max_blocks;
+
+  // each wave serves q-heads wid, wid+PA_WAVES, ... of the group
+  for (int qh_local = wid; qh_local < group; qh_local += PA_WAVES) {
+    const int qh = kvh * group + qh_local;
+    // load this lane's q slice (DS elems, fp32), pre-scaled
+    float qf[32];
+    {
+      const uint4* qv = reinterpret_cast<const uint4*>(
+          q + ((size_t)seq * n_q + qh) * D + p * DS);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        if (i * 8 < DS) {
+          float f[8];
+          unpack8(qv[i], f);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) qf[i * 8 + j] = f[j] * scale;
+        }
+      }
+    }
+    float m = -INFINITY;       // running max (shared across wave)
+    float l = 0.f;             // sumexp, lane-partial (own tokens)
+    float acc[32];             // V-weighted accumulator, lane-partial
+#pragma unroll
+    for (int i = 0; i < 32; ++i) acc[i] = 0.f;
+
+    for (int b = blk_begin; b < blk_end; ++b) {
+      const int blk_id = btab[b];
+      const int tok = b * PA_BLOCK + t_local;
+      const bool valid = tok < seq_len;
+      const bf16* kbase =
+          kcache + (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + t_local) * D +
+          p * DS;
+      // score: dot(q_slice, k_slice) then reduce over the 4-lane group
+      float s = 0.f;
+      {
+        const uint4* kv4 = reinterpret_cast<const uint4*>(kbase);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          if (i * 8 < DS) {
+            float f[8];
+            unpack8(kv4[i], f);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) s += qf[i * 8 + j] * f[j];
+          }
+        }
+      }
+      s += __shfl_xor(s, 1, WAVE);
+      s += __shfl_xor(s, 2, WAVE);
+      if (!valid) s = -INFINITY;
+      // block max over the 16 tokens (lane bits 2..5)
+      float bm = s;
+      bm = fmaxf(bm, __shfl_xor(bm, 4, WAVE));
+      bm = fmaxf(bm, __shfl_xor(bm, 8, WAVE));
+      bm = fmaxf(bm, __shfl_xor(bm, 16, WAVE));
+      bm = fmaxf(bm, __shfl_xor(bm, 32, WAVE));
+      const float m_new = fmaxf(m, bm);
+      if (m_new != -INFINITY) {
+        const float rescale = (m == -INFINITY) ? 0.f : __expf(m - m_new);
+        const float w = valid ? __expf(s - m_new) : 0.f;
+        l = l * rescale + w;
+#pragma unroll
+        for (int i = 0; i < 32; ++i) acc[i] *= rescale;
+        // accumulate w * V[tok][p*DS..]
+        const bf16* vbase =
+            vcache +
+            (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + t_local) * D + p * DS;
+        const uint4* vv4 = reinterpret_cast<const uint4*>(vbase);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          if (i * 8 < DS) {
+            float f[8];
+            unpack8(vv4[i], f);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) acc[i * 8 + j] += w * f[j];
+          }
+        }
+        m = m_new;
+      }
+    }
+    // reduce l and acc across the 16 token groups (butterfly)
+#pragma unroll
+    for (int off = 4; off <= 32; off <<= 1) {
+      l += __shfl_xor(l, off, WAVE);
+#pragma unroll
+      for (int i = 0; i < 32; ++i)
+        if (i < DS) acc[i] += __shfl_xor(acc[i], off, WAVE);
+    }
+    if (n_parts == 1) {
+      // write normalized bf16 output: lanes of token-group 0 hold it all
+      if (t_local == 0) {
+        const float inv_l = (l > 0.f) ? 1.f / l : 0.f;
+        float f[8];
+        uint4* ov = reinterpret_cast<uint4*>(
+            out + ((size_t)seq * n_q + qh) * D + p * DS);
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          if (i * 8 < DS) {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) f[j] = acc[i * 8 + j] * inv_l;
+            ov[i] = pack8(f);
+          }
+        }
+      }
+    } else {
+      if (t_local == 0) {
+        float* wa = ws_acc +
+            (((size_t)seq * n_q + qh) * n_parts + part) * D + p * DS;
+#pragma unroll
+        for (int i = 0; i < 32; ++i)
+          if (i < DS) wa[i] = acc[i];
+        if (p == 0) {
+          float* wm = ws_ml + (((size_t)seq * n_q + qh) * n_parts + part) * 2;
+          wm[0] = m;
+          wm[1] = l;
+        }
+      }
+    }
+  }
+}
+
+// combine partial (m, l, acc) across partitions: one wave per (seq, q-head)
+__global__ __launch_bounds__(WAVE)
+void paged_attn_reduce_kernel(bf16* __restrict__ out,
+                              const float* __restrict__ ws_acc,
+                              const float* __restrict__ ws_ml,
+                              const int* __restrict__ seq_lens,
+                              int n_q, int D, int n_parts) {
+  const int seq = blockIdx.x;
+  const int qh = blockIdx.y;
+  const int lane = threadIdx.x;
+  const float* ml = ws_ml + ((size_t)seq * n_q + qh) * n_parts * 2;
+  const int seq_len = seq_lens[seq];
+  const int n_blocks = (seq_len + PA_BLOCK - 1) / PA_BLOCK;
+  const int blocks_per_part = (n_blocks + n_parts - 1) / n_parts;
+  const int used_parts = min(n_parts, (n_blocks + blocks_per_part - 1) /
+                                          max(blocks_per_part, 1));
+  float m_star = -INFINITY;
+  for (int pp = 0; pp < used_parts; ++pp) m_star = fmaxf(m_star, ml[pp * 2]);
+  float l_tot = 0.f;
+  const int per_lane = (D + WAVE - 1) / WAVE;  // elems per lane (2 for D=128)
+  float o[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int pp = 0; pp < used_parts; ++pp) {
+    const float sc = (ml[pp * 2] == -INFINITY) ? 0.f
+                                               : __expf(ml[pp * 2] - m_star);
+    l_tot += sc * ml[pp * 2 + 1];
+    const float* wa = ws_acc + (((size_t)seq * n_q + qh) * n_parts + pp) * D;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      if (i < per_lane) {
+        const int d = lane * per_lane + i;
+        if (d < D) o[i] += sc * wa[d];
+      }
+    }
+  }
+  const float inv_l = (l_tot > 0.f) ? 1.f / l_tot : 0.f;
+  bf16* ob = out + ((size_t)seq * n_q + qh) * D;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    if (i < per_lane) {
+      const int d = lane * per_lane + i;
+      if (d < D) ob[d] = __float2bfloat16(o[i] * inv_l);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C API (called from Python via ctypes; raw device pointers + hipStream_t)
+// ---------------------------------------------------------------------------
+#define LL_CHECK(expr)                                                       \
+  do {                                                                       \
+    hipError_t _e = (expr);                                                  \
+    if (_e != hipSuccess) return (int)_e;                                    \
+  } while (0)
+
+extern "C" {
+
+int ll_rmsnorm(void* out, const void* in, const void* weight, void* residual,
+               int n_tokens, int hidden, float eps, void* stream) {
+  if (hidden % 8 != 0) return -2;
+  dim3 grid(n_tokens), block(256);
+  hipStream_t s = (hipStream_t)stream;
+  if (residual) {
+    hipLaunchKernelGGL(rmsnorm_kernel<true>, grid, block, 0, s, (bf16*)out,
+                       (const bf16*)in, (const bf16*)weight, (bf16*)residual,
+                       hidden, eps);
+  } else {
+    hipLaunchKernelGGL(rmsnorm_kernel<false>, grid, block, 0, s, (bf16*)out,
+                       (const bf16*)in, (const bf16*)weight, nullptr, hidden,
+                       eps);
+  }
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+int ll_rope(void* q, void* k, const void* cos_sin, const void* positions,
+            int n_tokens, int n_q, int n_kv, int D, void* stream) {
+  if (D % 16 != 0) return -2;
+  const int threads_per_head = (D / 2) / 4;
+  const long long total = (long long)n_tokens * (n_q + n_kv) * threads_per_head;
+  const int block = 256;
+  const long long grid = (total + block - 1) / block;
+  hipLaunchKernelGGL(rope_kernel, dim3((unsigned)grid), dim3(block), 0,
+                     (hipStream_t)stream, (bf16*)q, (bf16*)k,
+                     (const float*)cos_sin, (const int*)positions, n_tokens,
+                     n_q, n_kv, D);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+int ll_swiglu(void* out, const void* gateup, long long n_tokens, int inter,
+              void* stream) {
+  if (inter % 8 != 0) return -2;
+  const long long total = n_tokens * (inter >> 3);
+  const int block = 256;
+  long long grid = (total + block - 1) / block;
+  if (grid > 2048) grid = 2048;  // grid-stride (G11)
+  hipLaunchKernelGGL(swiglu_kernel, dim3((unsigned)grid), dim3(block), 0,
+                     (hipStream_t)stream, (bf16*)out, (const bf16*)gateup,
+                     n_tokens, inter);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+int ll_kv_append(void* kcache, void* vcache, const void* k, const void* v,
+                 const void* slots, int n_tokens, int n_kv, int D,
+                 int block_size, void* stream) {
+  if (D % 8 != 0) return -2;
+  const long long total = (long long)n_tokens * n_kv * (D >> 3);
+  const int block = 256;
+  const long long grid = (total + block - 1) / block;
+  hipLaunchKernelGGL(kv_append_kernel, dim3((unsigned)grid), dim3(block), 0,
+                     (hipStream_t)stream, (bf16*)kcache, (bf16*)vcache,
+                     (const bf16*)k, (const bf16*)v, (const int*)slots,
+                     n_tokens, n_kv, D, block_size);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
+                         const void* kcache, const void* vcache,
+                         const void* block_tables, const void* seq_lens,
+                         int num_seqs, int max_blocks, int n_q, int n_kv,
+                         int D, int block_size, int n_parts, float scale,
+                         void* stream) {
+  if (block_size != PA_BLOCK || D > 128 || D % 32 != 0) return -2;
+  if (n_q % n_kv != 0) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
+  hipLaunchKernelGGL(paged_attn_decode_kernel, grid, block, 0, s, (bf16*)out,
+                     (float*)ws_acc, (float*)ws_ml, (const bf16*)q,
+                     (const bf16*)kcache, (const bf16*)vcache,
+                     (const int*)block_tables, (const int*)seq_lens, n_q,
+                     n_kv, D, max_blocks, n_parts, scale);
+  LL_CHECK(hipGetLastError());
+  if (n_parts > 1) {
+    dim3 rgrid(num_seqs, n_q), rblock(WAVE);
+    hipLaunchKernelGGL(paged_attn_reduce_kernel, rgrid, rblock, 0, s,
+                       (bf16*)out, (const float*)ws_acc, (const float*)ws_ml,
+                       (const int*)seq_lens, n_q, D, n_parts);
+    LL_CHECK(hipGetLastError());
+  }
+  return 0;
+}
+
+int ll_device_sync() {
+  LL_CHECK(hipDeviceSynchronize());
+  return 0;
+}
+
+}  // extern "C"

@@ -1,0 +1,112 @@
+"""Pure-PyTorch fp32 reference implementations of the HIP kernels.
+
+Used (a) as the CPU fallback so the engine/tests run without a GPU, and
+(b) as the numerics reference the GPU tests compare the HIP kernels
+against (per-op, fp32 math).
+"""
+from __future__ import annotations
+
+import torch
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
+            residual: torch.Tensor | None = None):
+    """Returns (out, updated_residual). fp32 accumulation, cast back."""
+    orig_dtype = x.dtype
+    xf = x.float()
+    if residual is not None:
+        xf = xf + residual.float()
+        residual = xf.to(orig_dtype)
+    var = xf.pow(2).mean(dim=-1, keepdim=True)
+    out = (xf * torch.rsqrt(var + eps)) * weight.float()
+    return out.to(orig_dtype), residual
+
+
+def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
+               positions: torch.Tensor):
+    """Rotate-half RoPE in fp32.
+
+    q: [T, n_q, D], k: [T, n_kv, D]; cos_sin: [max_pos, D] fp32 with
+    cos[0:D/2] | sin[0:D/2] per row; positions: [T] int.
+    Returns rotated (q, k) in the input dtype.
+    """
+    D = q.shape[-1]
+    half = D // 2
+    cs = cos_sin[positions]            # [T, D]
+    cos = cs[:, :half].unsqueeze(1)    # [T, 1, half]
+    sin = cs[:, half:].unsqueeze(1)
+
+    def rot(x):
+        xf = x.float()
+        x1, x2 = xf[..., :half], xf[..., half:]
+        o1 = x1 * cos - x2 * sin
+        o2 = x2 * cos + x1 * sin
+        return torch.cat([o1, o2], dim=-1).to(x.dtype)
+
+    return rot(q), rot(k)
+
+
+def build_cos_sin_table(max_pos: int, head_dim: int, rope_theta: float,
+                        device="cpu") -> torch.Tensor:
+    """[max_pos, head_dim] fp32: cos(f_i * p) | sin(f_i * p)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (
+        rope_theta ** (torch.arange(0, half, dtype=torch.float64) / half)
+    )
+    pos = torch.arange(max_pos, dtype=torch.float64)
+    angles = torch.outer(pos, inv_freq)  # [max_pos, half]
+    table = torch.cat([angles.cos(), angles.sin()], dim=-1).float()
+    return table.to(device)
+
+
+def swiglu(gateup: torch.Tensor) -> torch.Tensor:
+    """gateup: [..., 2*inter] (gate | up) -> silu(gate) * up."""
+    inter = gateup.shape[-1] // 2
+    g = gateup[..., :inter].float()
+    u = gateup[..., inter:].float()
+    return (torch.nn.functional.silu(g) * u).to(gateup.dtype)
+
+
+def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
+              v: torch.Tensor, slots: torch.Tensor, block_size: int):
+    """Scatter k/v rows ([T, n_kv, D]) into paged caches
+    ([num_blocks, n_kv, block_size, D]) at flat slot ids."""
+    valid = slots >= 0
+    idx = slots[valid].long()
+    blk = idx // block_size
+    off = idx % block_size
+    # advanced indices (blk, off) broadcast together; slice dims follow
+    kcache[blk, :, off, :] = k[valid]
+    vcache[blk, :, off, :] = v[valid]
+
+
+def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
+                      vcache: torch.Tensor, block_tables: torch.Tensor,
+                      seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+    """Reference decode attention over the paged cache, fp32 math.
+
+    q: [S, n_q, D]; caches: [num_blocks, n_kv, block_size, D];
+    block_tables: [S, max_blocks] int32; seq_lens: [S].
+    Returns [S, n_q, D] in q.dtype.
+    """
+    S, n_q, D = q.shape
+    n_kv = kcache.shape[1]
+    block_size = kcache.shape[2]
+    group = n_q // n_kv
+    out = torch.empty_like(q)
+    for s in range(S):
+        L = int(seq_lens[s])
+        nb = (L + block_size - 1) // block_size
+        blks = block_tables[s, :nb].long()
+        # gather [L, n_kv, D]
+        k = kcache[blks].transpose(1, 2).reshape(nb * block_size, n_kv, D)[:L]
+        v = vcache[blks].transpose(1, 2).reshape(nb * block_size, n_kv, D)[:L]
+        qf = q[s].float()                       # [n_q, D]
+        kf = k.float()                          # [L, n_kv, D]
+        vf = v.float()
+        for h in range(n_q):
+            kvh = h // group
+            scores = (kf[:, kvh] @ qf[h]) * scale          # [L]
+            w = torch.softmax(scores, dim=0)
+            out[s, h] = (w @ vf[:, kvh]).to(q.dtype)
+    return out

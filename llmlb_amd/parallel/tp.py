@@ -1,0 +1,93 @@
+"""Tensor parallelism over torch.distributed.
+
+On ROCm the ``nccl`` backend IS RCCL and collectives ride the 7-link
+xGMI mesh (point-to-point, ≈153 GB/s per link), so a TP group must be
+placed on xGMI-adjacent GPUs — see ``llmlb_amd.parallel.topology`` for
+the quad-placement used by the gateway (BASELINE config 4).  CPU tests
+use ``gloo`` with world_size > 1.
+
+Sharding follows the usual Megatron split: attention qkv and MLP
+gate_up are column-parallel (no comm), o_proj and down_proj are
+row-parallel (one all-reduce each) — two all-reduces per layer.
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+class TPContext:
+    """Process-group wrapper; world_size==1 means no TP (no comm)."""
+
+    def __init__(self, group=None):
+        if group is not None:
+            self.group = group
+            self.world_size = dist.get_world_size(group)
+            self.rank = dist.get_rank(group)
+        elif dist.is_available() and dist.is_initialized():
+            self.group = None
+            self.world_size = dist.get_world_size()
+            self.rank = dist.get_rank()
+        else:
+            self.group = None
+            self.world_size = 1
+            self.rank = 0
+
+    def all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.world_size > 1:
+            dist.all_reduce(t, group=self.group)
+        return t
+
+
+_default_ctx: TPContext | None = None
+
+
+def get_tp_context() -> TPContext:
+    global _default_ctx
+    if _default_ctx is None:
+        _default_ctx = TPContext()
+    return _default_ctx
+
+
+def set_tp_context(ctx: TPContext) -> None:
+    global _default_ctx
+    _default_ctx = ctx
+
+
+class ColumnParallelLinear(nn.Module):
+    """Y = X W^T with W split along the output dim; no communication.
+
+    Output is the local shard [*, out_features / tp]."""
+
+    def __init__(self, in_features: int, out_features: int, tp: TPContext,
+                 dtype=torch.bfloat16):
+        super().__init__()
+        assert out_features % tp.world_size == 0, \
+            f"{out_features} not divisible by tp={tp.world_size}"
+        self.tp = tp
+        self.out_local = out_features // tp.world_size
+        self.weight = nn.Parameter(
+            torch.empty(self.out_local, in_features, dtype=dtype),
+            requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return torch.nn.functional.linear(x, self.weight)
+
+
+class RowParallelLinear(nn.Module):
+    """Y = X W^T with W split along the input dim; all-reduce on output."""
+
+    def __init__(self, in_features: int, out_features: int, tp: TPContext,
+                 dtype=torch.bfloat16):
+        super().__init__()
+        assert in_features % tp.world_size == 0
+        self.tp = tp
+        self.in_local = in_features // tp.world_size
+        self.weight = nn.Parameter(
+            torch.empty(out_features, self.in_local, dtype=dtype),
+            requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        y = torch.nn.functional.linear(x, self.weight)
+        return self.tp.all_reduce(y)

@@ -1,0 +1,130 @@
+"""GPU numerics tests: each gfx950 HIP kernel vs the plain PyTorch fp32
+reference of the same op (tolerances sized for bf16 I/O, fp32 accum)."""
+import pytest
+import torch
+
+import llmlb_amd.ops as ops
+from llmlb_amd.ops import reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def setup_module(module):
+    assert ops.native_available(), (
+        f"native kernels must be loaded on a GPU box: "
+        f"{ops.native_load_error()}")
+
+
+def _mx(a, b):
+    return (a.float() - b.float()).abs().max().item()
+
+
+@pytest.mark.parametrize("tokens,hidden", [(1, 4096), (64, 4096),
+                                           (17, 1024), (256, 8192)])
+def test_rmsnorm(tokens, hidden):
+    torch.manual_seed(0)
+    x = torch.randn(tokens, hidden, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(hidden, dtype=torch.bfloat16, device=DEV)
+    out, _ = ops.rmsnorm(x, w, 1e-5)
+    exp, _ = ref.rmsnorm(x.cpu(), w.cpu(), 1e-5)
+    assert _mx(out.cpu(), exp) < 2e-2
+
+
+def test_rmsnorm_fused_residual():
+    torch.manual_seed(1)
+    x = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    r = torch.randn(32, 4096, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    r_gpu = r.clone()
+    out, r_out = ops.rmsnorm(x, w, 1e-5, residual=r_gpu)
+    exp_out, exp_r = ref.rmsnorm(x.cpu(), w.cpu(), 1e-5, residual=r.cpu())
+    assert _mx(out.cpu(), exp_out) < 2e-2
+    assert _mx(r_out.cpu(), exp_r) < 2e-2
+
+
+@pytest.mark.parametrize("tokens,nq,nkv,D", [(5, 32, 8, 128), (64, 16, 4, 64),
+                                             (3, 8, 8, 128)])
+def test_rope(tokens, nq, nkv, D):
+    torch.manual_seed(2)
+    q = torch.randn(tokens, nq, D, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(tokens, nkv, D, dtype=torch.bfloat16, device=DEV)
+    pos = torch.randint(0, 2048, (tokens,), device=DEV)
+    table = ref.build_cos_sin_table(4096, D, 500000.0, device=DEV)
+    q_cpu, k_cpu = q.cpu(), k.cpu()
+    qo, ko = ops.rope_apply(q.clone(), k.clone(), table, pos)
+    eq, ek = ref.rope_apply(q_cpu, k_cpu, table.cpu(), pos.cpu())
+    assert _mx(qo.cpu(), eq) < 2e-2
+    assert _mx(ko.cpu(), ek) < 2e-2
+
+
+@pytest.mark.parametrize("tokens,inter", [(1, 14336), (64, 14336), (7, 512)])
+def test_swiglu(tokens, inter):
+    torch.manual_seed(3)
+    gu = torch.randn(tokens, 2 * inter, dtype=torch.bfloat16, device=DEV)
+    out = ops.swiglu(gu)
+    exp = ref.swiglu(gu.cpu())
+    assert _mx(out.cpu(), exp) < 2e-2
+
+
+def test_kv_append():
+    torch.manual_seed(4)
+    nb, nkv, bs, D, T = 8, 4, 16, 128, 20
+    kc = torch.zeros(nb, nkv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    k = torch.randn(T, nkv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, nkv, D, dtype=torch.bfloat16, device=DEV)
+    slots = torch.randperm(nb * bs, device=DEV)[:T].to(torch.int32)
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    ops.kv_append(kc, vc, k, v, slots)
+    ref.kv_append(kc_ref, vc_ref, k.cpu(), v.cpu(), slots.cpu(), bs)
+    assert torch.equal(kc.cpu(), kc_ref)
+    assert torch.equal(vc.cpu(), vc_ref)
+
+
+@pytest.mark.parametrize(
+    "S,nq,nkv,D,maxlen,parts",
+    [(4, 32, 8, 128, 100, None),       # llama-3-8b heads
+     (1, 32, 8, 128, 1500, None),      # long seq -> partitioned
+     (64, 16, 4, 64, 300, None),       # many seqs
+     (2, 8, 8, 128, 40, 4),            # forced multi-partition, MHA
+     (3, 16, 2, 32, 17, None)])        # small D
+def test_paged_attn_decode(S, nq, nkv, D, maxlen, parts):
+    torch.manual_seed(5)
+    bs = 16
+    seq_lens = torch.randint(1, maxlen + 1, (S,), device=DEV,
+                             dtype=torch.int32)
+    seq_lens[0] = maxlen
+    max_blocks = (maxlen + bs - 1) // bs
+    nb = S * max_blocks + 1
+    kc = torch.randn(nb, nkv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.randn(nb, nkv, bs, D, dtype=torch.bfloat16, device=DEV)
+    # distinct random block ids per seq
+    perm = torch.randperm(S * max_blocks, device=DEV) + 1
+    bt = perm.reshape(S, max_blocks).to(torch.int32)
+    q = torch.randn(S, nq, D, dtype=torch.bfloat16, device=DEV)
+    scale = D ** -0.5
+    out = ops.paged_attn_decode(q, kc, vc, bt, seq_lens, scale,
+                                n_parts=parts)
+    exp = ref.paged_attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                seq_lens.cpu(), scale)
+    err = _mx(out.cpu(), exp)
+    assert err < 3e-2, f"max err {err}"
+
+
+def test_paged_attn_partition_consistency():
+    """n_parts=1 and n_parts=8 must agree to rounding."""
+    torch.manual_seed(6)
+    S, nq, nkv, D, bs = 2, 32, 8, 128, 16
+    L = 777
+    mb = (L + bs - 1) // bs
+    kc = torch.randn(S * mb + 1, nkv, bs, D, dtype=torch.bfloat16,
+                     device=DEV)
+    vc = torch.randn_like(kc)
+    bt = (torch.arange(S * mb, device=DEV) + 1).reshape(S, mb).to(torch.int32)
+    q = torch.randn(S, nq, D, dtype=torch.bfloat16, device=DEV)
+    sl = torch.tensor([L, L - 5], dtype=torch.int32, device=DEV)
+    o1 = ops.paged_attn_decode(q, kc, vc, bt, sl, D ** -0.5, n_parts=1)
+    o8 = ops.paged_attn_decode(q, kc, vc, bt, sl, D ** -0.5, n_parts=8)
+    assert _mx(o1, o8) < 1e-2
