@@ -17,6 +17,9 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
     if residual is not None:
         xf = xf + residual.float()
         residual = xf.to(orig_dtype)
+        # the residual stream is carried in the storage dtype (bf16), and
+        # the norm reads the stored stream — match the fused kernel
+        xf = residual.float()
     var = xf.pow(2).mean(dim=-1, keepdim=True)
     out = (xf * torch.rsqrt(var + eps)) * weight.float()
     return out.to(orig_dtype), residual
