@@ -1,0 +1,497 @@
+// Local (non-proxied) API handlers: /v1/models, endpoint admin CRUD,
+// api-key management, dashboard summary/stats, request history, GPU
+// telemetry view, audio/images capability routing, drain control.
+#include "gateway.h"
+
+namespace gw {
+
+void start_proxy_direct(Gateway* gw, ClientConn* c, EndpointPtr ep,
+                        ApiKind kind, const std::string& upstream_path,
+                        std::string_view content_type);
+
+// telemetry.cpp
+void telemetry_snapshot_json(std::string* out);
+
+static Json endpoint_json(const EndpointPtr& ep) {
+  JsonObject o;
+  o.emplace_back("id", Json(ep->id));
+  o.emplace_back("name", Json(ep->name));
+  o.emplace_back("base_url", Json(ep->base_url));
+  o.emplace_back("endpoint_type", Json(ep->type));
+  o.emplace_back("status", Json(status_name(ep->status.load())));
+  o.emplace_back("device_index", Json((long long)ep->device_index));
+  o.emplace_back("active_requests", Json((long long)ep->active.load()));
+  o.emplace_back("total_requests",
+                 Json((long long)ep->total_requests.load()));
+  o.emplace_back("total_failures",
+                 Json((long long)ep->total_failures.load()));
+  o.emplace_back("latency_ema_ms", Json(ep->latency_ema_ms.load()));
+  o.emplace_back("health_check_interval_secs",
+                 Json((long long)ep->health_interval_s));
+  o.emplace_back("inference_timeout_secs",
+                 Json((long long)ep->inference_timeout_s));
+  o.emplace_back("created_at", Json(ep->created_at));
+  o.emplace_back("updated_at", Json(ep->updated_at));
+  {
+    std::lock_guard<std::mutex> lk(ep->gpu_mu);
+    if (ep->gpu.updated_ms > 0) {
+      JsonObject g;
+      g.emplace_back("hbm_used_gb", Json(ep->gpu.hbm_used_gb));
+      g.emplace_back("hbm_total_gb", Json(ep->gpu.hbm_total_gb));
+      g.emplace_back("busy_pct", Json(ep->gpu.busy_pct));
+      g.emplace_back("device_index",
+                     Json((long long)ep->gpu.device_index));
+      g.emplace_back("updated_ms", Json((long long)ep->gpu.updated_ms));
+      o.emplace_back("gpu", Json(std::move(g)));
+    }
+  }
+  JsonArray models;
+  for (auto& m : ep->models) {
+    JsonObject mo;
+    mo.emplace_back("id", Json(m.id));
+    mo.emplace_back("canonical_id", Json(m.canonical));
+    mo.emplace_back("capabilities", Json(caps_to_csv(m.caps)));
+    if (m.max_tokens) mo.emplace_back("max_tokens", Json(m.max_tokens));
+    models.emplace_back(Json(std::move(mo)));
+  }
+  o.emplace_back("models", Json(std::move(models)));
+  return Json(std::move(o));
+}
+
+std::string Gateway::render_models_json() {
+  auto table = registry.table();
+  JsonArray data;
+  for (auto& key : table->model_keys_canonical) {
+    auto it = table->by_model.find(key);
+    if (it == table->by_model.end()) continue;
+    const ModelRoute& route = it->second;
+    JsonObject m;
+    m.emplace_back("id", Json(key));
+    m.emplace_back("object", Json("model"));
+    m.emplace_back("created", Json((long long)(wall_ms() / 1000)));
+    m.emplace_back("owned_by", Json("llmlb"));
+    JsonArray eps;
+    for (auto& ep : route.endpoints) eps.emplace_back(Json(ep->id));
+    m.emplace_back("endpoint_ids", Json(std::move(eps)));
+    m.emplace_back("capabilities", Json(caps_to_csv(route.caps)));
+    if (route.max_tokens) m.emplace_back("max_tokens",
+                                        Json(route.max_tokens));
+    data.emplace_back(Json(std::move(m)));
+  }
+  JsonObject o;
+  o.emplace_back("object", Json("list"));
+  o.emplace_back("data", Json(std::move(data)));
+  return Json(std::move(o)).dump();
+}
+
+std::string Gateway::render_summary_json() {
+  // SystemSummary analog (balancer/mod.rs:2470-2588)
+  auto eps = registry.list();
+  JsonArray earr;
+  long long active = 0;
+  for (auto& ep : eps) {
+    active += ep->active.load();
+    earr.emplace_back(endpoint_json(ep));
+  }
+  auto sum60 = history.ring().summarize(60 * 60 * 1000);
+  JsonObject o;
+  o.emplace_back("endpoints", Json(std::move(earr)));
+  o.emplace_back("active_requests", Json(active));
+  o.emplace_back("uptime_ms", Json((long long)(wall_ms() - start_ms)));
+  JsonObject h;
+  h.emplace_back("window_minutes", Json(60));
+  h.emplace_back("requests", Json(sum60.requests));
+  h.emplace_back("failures", Json(sum60.failures));
+  h.emplace_back("prompt_tokens", Json(sum60.prompt_tokens));
+  h.emplace_back("completion_tokens", Json(sum60.completion_tokens));
+  h.emplace_back("avg_duration_ms", Json(sum60.avg_duration_ms));
+  o.emplace_back("last_hour", Json(std::move(h)));
+  return Json(std::move(o)).dump();
+}
+
+std::string Gateway::render_metrics_text() {
+  char buf[2048];
+  int n = snprintf(
+      buf, sizeof buf,
+      "# HELP llmlb_requests_total Total HTTP requests seen\n"
+      "# TYPE llmlb_requests_total counter\n"
+      "llmlb_requests_total %lld\n"
+      "# TYPE llmlb_responses_total counter\n"
+      "llmlb_responses_total{class=\"2xx\"} %lld\n"
+      "llmlb_responses_total{class=\"4xx\"} %lld\n"
+      "llmlb_responses_total{class=\"5xx\"} %lld\n"
+      "# TYPE llmlb_upstream_errors_total counter\n"
+      "llmlb_upstream_errors_total %lld\n"
+      "# TYPE llmlb_streamed_responses_total counter\n"
+      "llmlb_streamed_responses_total %lld\n"
+      "# TYPE llmlb_tokens_streamed_total counter\n"
+      "llmlb_tokens_streamed_total %lld\n"
+      "# TYPE llmlb_cloud_requests_total counter\n"
+      "llmlb_cloud_requests_total %lld\n"
+      "# TYPE llmlb_active_requests gauge\n"
+      "llmlb_active_requests %lld\n",
+      metrics.requests_total.load(), metrics.responses_2xx.load(),
+      metrics.responses_4xx.load(), metrics.responses_5xx.load(),
+      metrics.upstream_errors.load(), metrics.streamed_responses.load(),
+      metrics.tokens_streamed.load(), metrics.cloud_requests.load(),
+      metrics.active_requests.load());
+  return std::string(buf, n);
+}
+
+// capability-routed backend pick: FIRST online endpoint with the
+// capability (reference api/audio.rs:160-190 — deliberately no TPS
+// weighting on the audio/image paths)
+static EndpointPtr pick_by_capability(Registry& reg, uint32_t cap) {
+  for (auto& ep : reg.list()) {
+    if (ep->status.load() != EndpointStatus::kOnline) continue;
+    for (auto& m : ep->models)
+      if (m.caps & cap) return ep;
+  }
+  return nullptr;
+}
+
+void Gateway::handle_local(ClientConn* c, const std::string& path,
+                           std::string_view body) {
+  const HttpRequest& r = c->req;
+  std::string_view method = r.method;
+
+  // ---- /v1/models -------------------------------------------------
+  if (path == "/v1/models" && method == "GET") {
+    c->send_simple(200, "application/json", render_models_json());
+    return;
+  }
+  if (starts_with(path, "/v1/models/") && method == "GET") {
+    std::string id = url_decode(path.substr(strlen("/v1/models/")));
+    auto table = registry.table();
+    auto it = table->by_model.find(id);
+    if (it == table->by_model.end()) {
+      c->send_json_error(404, "model_not_found", "no such model");
+      return;
+    }
+    JsonObject m;
+    m.emplace_back("id", Json(it->second.canonical));
+    m.emplace_back("object", Json("model"));
+    m.emplace_back("owned_by", Json("llmlb"));
+    m.emplace_back("capabilities", Json(caps_to_csv(it->second.caps)));
+    c->send_simple(200, "application/json", Json(std::move(m)).dump());
+    return;
+  }
+
+  // ---- /v1/audio/* + /v1/images/* (capability-routed proxy) --------
+  if (starts_with(path, "/v1/audio/") || starts_with(path, "/v1/images/")) {
+    uint32_t cap;
+    ApiKind kind;
+    if (path == "/v1/audio/transcriptions") {
+      cap = kCapAudioTranscription;
+      kind = ApiKind::kAudio;
+    } else if (path == "/v1/audio/speech") {
+      cap = kCapAudioSpeech;
+      kind = ApiKind::kAudio;
+    } else if (starts_with(path, "/v1/images/")) {
+      cap = kCapImageGen;
+      kind = ApiKind::kImages;
+    } else {
+      c->send_json_error(404, "not_found", "unknown audio route");
+      return;
+    }
+    EndpointPtr ep = pick_by_capability(registry, cap);
+    if (!ep) {
+      c->send_json_error(503, "no_capable_endpoint",
+                         "no online endpoint supports this capability");
+      return;
+    }
+    start_proxy_direct(this, c, ep, kind, path,
+                       r.header("content-type"));
+    return;
+  }
+
+  // ---- admin API (auth: admin token or admin-permission key) ------
+  if (starts_with(path, "/api/")) {
+    std::string_view authz = r.header("authorization");
+    bool admin_ok = false;
+    if (!cfg.admin_token.empty()) {
+      std::string_view key = authz;
+      if (starts_with(key, "Bearer ")) key.remove_prefix(7);
+      if (trim(key) == cfg.admin_token) admin_ok = true;
+    }
+    if (!admin_ok) admin_ok = auth.check_bearer(authz, "admin");
+    if (!admin_ok) {
+      c->send_json_error(401, "unauthorized", "admin credentials required");
+      return;
+    }
+
+    // -- endpoints CRUD (api/endpoints.rs) --
+    if (path == "/api/endpoints" && method == "GET") {
+      JsonArray arr;
+      for (auto& ep : registry.list()) arr.emplace_back(endpoint_json(ep));
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (path == "/api/endpoints" && method == "POST") {
+      Json in;
+      if (!Json::parse(body, in) || !in.is_obj()) {
+        c->send_json_error(400, "bad_json", "invalid JSON body");
+        return;
+      }
+      std::string err;
+      LOG_INFO("add_endpoint: start");
+      EndpointPtr ep = registry.add_endpoint(
+          in["name"].as_str(), in["base_url"].as_str(),
+          in["endpoint_type"].as_str(), in["api_key"].as_str(), &err);
+      if (!ep) {
+        c->send_json_error(422, "invalid_endpoint", err);
+        return;
+      }
+      if (in.find("device_index"))
+        registry.update_endpoint(ep->id, [&](Endpoint& e) {
+          e.device_index = (int)in["device_index"].as_int(-1);
+        });
+      // immediate async probe: health loop will pick it up on its next
+      // tick; optionally detect type + sync now (blocking, admin path)
+      LOG_INFO("add_endpoint: registered, detecting type");
+      std::string dtype = detect_endpoint_type(ep->host, ep->port,
+                                               ep->path_prefix);
+      LOG_INFO("add_endpoint: detected '%s', syncing", dtype.c_str());
+      if (!dtype.empty() && ep->type == "openai_compat")
+        registry.update_endpoint(ep->id, [&](Endpoint& e) {
+          e.type = dtype;
+        });
+      std::string serr;
+      if (sync_endpoint_models(ep, &serr)) {
+        registry.set_status(ep->id, EndpointStatus::kOnline);
+      }
+      LOG_INFO("add_endpoint: done (sync err='%s')", serr.c_str());
+      c->send_simple(201, "application/json", endpoint_json(ep).dump());
+      return;
+    }
+    if (starts_with(path, "/api/endpoints/")) {
+      std::string rest = path.substr(strlen("/api/endpoints/"));
+      size_t slash = rest.find('/');
+      std::string id = url_decode(rest.substr(0, slash));
+      std::string action = slash == std::string::npos
+                               ? ""
+                               : rest.substr(slash + 1);
+      EndpointPtr ep = registry.get(id);
+      if (!ep) {
+        c->send_json_error(404, "not_found", "no such endpoint");
+        return;
+      }
+      if (action.empty() && method == "GET") {
+        c->send_simple(200, "application/json", endpoint_json(ep).dump());
+        return;
+      }
+      if (action.empty() && method == "DELETE") {
+        registry.remove_endpoint(id);
+        c->send_simple(200, "application/json", "{\"deleted\":true}");
+        return;
+      }
+      if (action.empty() && (method == "PUT" || method == "PATCH")) {
+        Json in;
+        if (!Json::parse(body, in)) {
+          c->send_json_error(400, "bad_json", "invalid JSON");
+          return;
+        }
+        registry.update_endpoint(id, [&](Endpoint& e) {
+          if (in.find("name")) e.name = in["name"].as_str();
+          if (in.find("api_key")) e.api_key = in["api_key"].as_str();
+          if (in.find("endpoint_type"))
+            e.type = in["endpoint_type"].as_str();
+          if (in.find("device_index"))
+            e.device_index = (int)in["device_index"].as_int(-1);
+          if (in.find("inference_timeout_secs"))
+            e.inference_timeout_s =
+                (int)in["inference_timeout_secs"].as_int(120);
+        });
+        c->send_simple(200, "application/json",
+                       endpoint_json(registry.get(id)).dump());
+        return;
+      }
+      if (action == "sync" && method == "POST") {
+        std::string err;
+        bool ok = sync_endpoint_models(ep, &err);
+        if (ok) registry.set_status(id, EndpointStatus::kOnline);
+        std::string bodyj = ok ? "{\"synced\":true}"
+                               : "{\"synced\":false,\"error\":\"" + err +
+                                     "\"}";
+        c->send_simple(ok ? 200 : 502, "application/json", bodyj);
+        return;
+      }
+      if (action == "test" && method == "POST") {
+        MiniHttpResult res = mini_http_request(
+            ep->host, ep->port, "GET", ep->path_prefix + "/v1/models", "",
+            cfg.health_timeout_ms);
+        JsonObject o;
+        o.emplace_back("reachable", Json(res.status > 0));
+        o.emplace_back("status", Json((long long)res.status));
+        o.emplace_back("latency_ms", Json(res.elapsed_ms));
+        if (!res.error.empty()) o.emplace_back("error", Json(res.error));
+        c->send_simple(200, "application/json", Json(std::move(o)).dump());
+        return;
+      }
+      c->send_json_error(404, "not_found", "unknown endpoint action");
+      return;
+    }
+
+    // -- api keys --
+    if (path == "/api/api-keys" && method == "POST") {
+      Json in;
+      Json::parse(body, in);
+      std::string name = in["name"].as_str();
+      std::string perms = in["permissions"].as_str();
+      if (perms.empty()) perms = "inference";
+      std::string key = auth.create_key(name.empty() ? "key" : name, perms);
+      JsonObject o;
+      o.emplace_back("api_key", Json(key));
+      o.emplace_back("name", Json(name));
+      o.emplace_back("permissions", Json(perms));
+      c->send_simple(201, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    if (path == "/api/api-keys" && method == "GET") {
+      std::vector<DbRow> rows;
+      db.exec("SELECT id,name,prefix,permissions,created_at,revoked FROM "
+              "api_keys", {}, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("id", Json((long long)row[0].i));
+        o.emplace_back("name", Json(row[1].str()));
+        o.emplace_back("prefix", Json(row[2].str()));
+        o.emplace_back("permissions", Json(row[3].str()));
+        o.emplace_back("created_at", Json(row[4].str()));
+        o.emplace_back("revoked", Json(row[5].i != 0));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (starts_with(path, "/api/api-keys/") && method == "DELETE") {
+      auth.revoke_key(path.substr(strlen("/api/api-keys/")));
+      c->send_simple(200, "application/json", "{\"revoked\":true}");
+      return;
+    }
+
+    // -- dashboard data --
+    if (path == "/api/dashboard/summary" && method == "GET") {
+      c->send_simple(200, "application/json", render_summary_json());
+      return;
+    }
+    if (path == "/api/requests" && method == "GET") {
+      auto recs = history.ring().snapshot(500);
+      JsonArray arr;
+      for (auto& rec : recs) {
+        JsonObject o;
+        o.emplace_back("ts_ms", Json((long long)rec.ts_ms));
+        o.emplace_back("endpoint_id", Json(rec.endpoint_id));
+        o.emplace_back("model", Json(rec.model));
+        o.emplace_back("api_kind", Json(rec.api_kind));
+        o.emplace_back("status", Json((long long)rec.status));
+        o.emplace_back("duration_ms", Json(rec.duration_ms));
+        o.emplace_back("prompt_tokens", Json(rec.prompt_tokens));
+        o.emplace_back("completion_tokens", Json(rec.completion_tokens));
+        o.emplace_back("client_ip", Json(rec.client_ip));
+        if (!rec.error.empty()) o.emplace_back("error", Json(rec.error));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (path == "/api/dashboard/daily-stats" && method == "GET") {
+      std::vector<DbRow> rows;
+      db.exec("SELECT date,endpoint_id,model,api_kind,requests,failures,"
+              "prompt_tokens,completion_tokens,total_duration_ms FROM "
+              "endpoint_daily_stats ORDER BY date DESC LIMIT 500",
+              {}, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("date", Json(row[0].str()));
+        o.emplace_back("endpoint_id", Json(row[1].str()));
+        o.emplace_back("model", Json(row[2].str()));
+        o.emplace_back("api_kind", Json(row[3].str()));
+        o.emplace_back("requests", Json((long long)row[4].i));
+        o.emplace_back("failures", Json((long long)row[5].i));
+        o.emplace_back("prompt_tokens", Json((long long)row[6].i));
+        o.emplace_back("completion_tokens", Json((long long)row[7].i));
+        o.emplace_back("total_duration_ms",
+                       Json(row[8].kind == DbValue::kReal ? row[8].d
+                                                          : (double)row[8].i));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (path == "/api/gpu" && method == "GET") {
+      std::string out;
+      telemetry_snapshot_json(&out);
+      c->send_simple(200, "application/json", out);
+      return;
+    }
+    if (path == "/api/audit/verify" && method == "GET") {
+      std::string err;
+      bool ok = history.verify_audit_chain(&err);
+      JsonObject o;
+      o.emplace_back("valid", Json(ok));
+      if (!ok) o.emplace_back("error", Json(err));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    // -- drain gate control (update/self-update support) --
+    if (path == "/api/system/drain" && method == "POST") {
+      draining.store(true);
+      c->send_simple(200, "application/json", "{\"draining\":true}");
+      return;
+    }
+    if (path == "/api/system/resume" && method == "POST") {
+      draining.store(false);
+      c->send_simple(200, "application/json", "{\"draining\":false}");
+      return;
+    }
+    if (path == "/api/system/info" && method == "GET") {
+      JsonObject o;
+      o.emplace_back("version", Json("0.1.0"));
+      o.emplace_back("workers", Json((long long)workers.size()));
+      o.emplace_back("uptime_ms", Json((long long)(wall_ms() - start_ms)));
+      o.emplace_back("draining", Json(draining.load()));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    // fall through to the Python control plane when configured
+    if (!cfg.control_plane_base.empty()) {
+      start_control_plane_proxy(c);
+      return;
+    }
+    c->send_json_error(404, "not_found", "unknown /api route");
+    return;
+  }
+
+  // dashboard static / ws: control plane
+  if (!cfg.control_plane_base.empty()) {
+    start_control_plane_proxy(c);
+    return;
+  }
+  c->send_json_error(404, "not_found", "no such route");
+}
+
+void Gateway::start_control_plane_proxy(ClientConn* c) {
+  // forward verbatim to the Python control plane (host:port http)
+  std::string base = cfg.control_plane_base;
+  std::string_view s = base;
+  if (starts_with(s, "http://")) s.remove_prefix(7);
+  size_t colon = s.rfind(':');
+  std::string host(s.substr(0, colon));
+  int port = colon == std::string_view::npos
+                 ? 80
+                 : atoi(std::string(s.substr(colon + 1)).c_str());
+  auto ep = std::make_shared<Endpoint>();  // ephemeral descriptor
+  ep->id = "control-plane";
+  ep->host = host;
+  ep->port = port;
+  ep->inference_timeout_s = 30;
+  ep->status = EndpointStatus::kOnline;
+  start_proxy_direct(this, c, ep, ApiKind::kOther,
+                     std::string(c->req.target),
+                     c->req.header("content-type"));
+}
+
+}  // namespace gw

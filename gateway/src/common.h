@@ -5,6 +5,7 @@
 //  with a zero-copy pump.
 #pragma once
 
+#include <algorithm>
 #include <cassert>
 #include <cerrno>
 #include <chrono>
@@ -59,41 +60,56 @@ void log_write(LogLevel lv, const char* fmt, ...)
 class Buffer {
  public:
   const char* data() const { return buf_.data() + off_; }
-  char* wptr() { return buf_.data() + buf_.size(); }
-  size_t size() const { return buf_.size() - off_; }
+  size_t size() const { return len_ - off_; }
   bool empty() const { return size() == 0; }
   std::string_view view() const { return {data(), size()}; }
 
   void append(const void* p, size_t n) {
-    buf_.insert(buf_.end(), (const char*)p, (const char*)p + n);
+    ensure(n);
+    memcpy(buf_.data() + len_, p, n);
+    len_ += n;
   }
   void append(std::string_view s) { append(s.data(), s.size()); }
-  // reserve n writable bytes at the end; returns pointer
+  // reserve n writable bytes at the end; returns pointer (commit() to
+  // make written bytes visible — no zero-fill of the window)
   char* reserve(size_t n) {
-    size_t old = buf_.size();
-    buf_.resize(old + n);
-    buf_.resize(old);
-    return buf_.data() + old;
+    ensure(n);
+    return buf_.data() + len_;
   }
-  void commit(size_t n) { buf_.resize(buf_.size() + n); }
+  void commit(size_t n) {
+    len_ += n;
+    assert(len_ <= buf_.size());
+  }
   void consume(size_t n) {
     off_ += n;
-    assert(off_ <= buf_.size());
-    if (off_ == buf_.size()) clear();
-    // compact when the dead prefix dominates
-    else if (off_ > 65536 && off_ > buf_.size() / 2) {
-      buf_.erase(buf_.begin(), buf_.begin() + off_);
+    assert(off_ <= len_);
+    if (off_ == len_) {
+      off_ = len_ = 0;
+    } else if (off_ > 65536 && off_ > len_ / 2) {
+      memmove(buf_.data(), buf_.data() + off_, len_ - off_);
+      len_ -= off_;
       off_ = 0;
     }
   }
-  void clear() {
-    buf_.clear();
-    off_ = 0;
-  }
+  void clear() { off_ = len_ = 0; }
 
  private:
+  void ensure(size_t n) {
+    if (len_ + n <= buf_.size()) return;
+    // compact the dead prefix first if it buys enough
+    if (off_ > 0 && len_ - off_ + n <= buf_.size() && off_ >= n) {
+      memmove(buf_.data(), buf_.data() + off_, len_ - off_);
+      len_ -= off_;
+      off_ = 0;
+      if (len_ + n <= buf_.size()) return;
+    }
+    size_t want = std::max(buf_.size() * 2, len_ + n);
+    if (want < 4096) want = 4096;
+    buf_.resize(want);
+  }
   std::vector<char> buf_;
-  size_t off_ = 0;
+  size_t off_ = 0;   // consumed prefix
+  size_t len_ = 0;   // valid bytes
 };
 
 // ---------------------------------------------------------------- strings

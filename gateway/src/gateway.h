@@ -1,0 +1,291 @@
+// llmlb-gw server core: worker event loops, client/upstream connections,
+// the zero-copy SSE proxy pump, local admin API.
+#pragma once
+
+#include <atomic>
+#include <memory>
+#include <thread>
+#include <unordered_map>
+#include <unordered_set>
+
+#include "common.h"
+#include "db.h"
+#include "history.h"
+#include "http.h"
+#include "json.h"
+#include "net.h"
+#include "router.h"
+
+namespace gw { class AnthropicStreamEncoder; }
+#include "anthropic.h"
+
+namespace gw {
+
+struct GatewayConfig {
+  std::string host = "0.0.0.0";
+  int port = 32768;
+  int workers = 0;                 // 0 = #cores
+  std::string db_path = "llmlb.sqlite3";
+  size_t body_limit = 20 * 1024 * 1024;   // api/mod.rs:58
+  AdmissionConfig admission;
+  int retention_days = 7;
+  int health_interval_s = 30;
+  int health_timeout_ms = 5000;
+  int auto_sync_interval_s = 900;
+  int consecutive_failures_for_offline = 2;
+  bool require_auth = true;
+  std::string admin_token;          // LLMLB_ADMIN_TOKEN
+  std::string log_level = "info";
+  std::string log_dir;
+  // cloud providers (base_url overridable for tests; http only — this
+  // build environment has no TLS stack and no egress)
+  std::string openai_base = "";     // e.g. http://127.0.0.1:9001/v1
+  std::string google_base = "";
+  std::string anthropic_base = "";
+  std::string openai_key, google_key, anthropic_key;
+  // control plane (Python FastAPI) to which /dashboard + extended /api
+  // routes are proxied; empty = disabled
+  std::string control_plane_base = "";
+};
+
+class Gateway;
+class Worker;
+class ClientConn;
+class UpstreamConn;
+
+// metrics counters (Prometheus text at /api/metrics — cloud_metrics.rs
+// analog, extended with gateway totals)
+struct Metrics {
+  std::atomic<long long> requests_total{0};
+  std::atomic<long long> responses_2xx{0};
+  std::atomic<long long> responses_4xx{0};
+  std::atomic<long long> responses_5xx{0};
+  std::atomic<long long> upstream_errors{0};
+  std::atomic<long long> streamed_responses{0};
+  std::atomic<long long> tokens_streamed{0};
+  std::atomic<long long> cloud_requests{0};
+  std::atomic<long long> active_requests{0};
+};
+
+// auth store: api-key sha256 lookup (auth/middleware.rs analog)
+class AuthStore {
+ public:
+  explicit AuthStore(Db* db) : db_(db) {}
+  void load();
+  // returns true when authorized; perm: "inference" | "admin"
+  bool check_bearer(std::string_view authorization, const char* perm);
+  bool check_key(std::string_view key, const char* perm);
+  std::string create_key(const std::string& name, const std::string& perms);
+  bool revoke_key(const std::string& prefix_or_id);
+  bool any_keys();
+  void set_open_mode(bool open) { open_mode_ = open; }
+  bool open_mode() const { return open_mode_; }
+
+ private:
+  Db* db_;
+  std::mutex mu_;
+  std::unordered_map<std::string, std::string> hash_to_perms_;
+  bool open_mode_ = false;  // no keys configured -> allow (logged)
+};
+
+// ---------------------------------------------------------------------
+// Proxy operation: one in-flight proxied request (client <-> upstream).
+// ---------------------------------------------------------------------
+struct ProxyOp {
+  ClientConn* client = nullptr;       // owner
+  UpstreamConn* upstream = nullptr;
+  RequestLease lease;
+  BodyReader body_reader;
+  SseTap tap;
+  HttpResponse resp;
+  bool resp_head_seen = false;
+  bool streaming = false;
+  bool lease_released_at_head = false;
+  int64_t start_us = 0;
+  uint64_t timer_id = 0;
+  // bookkeeping for history / TPS
+  std::string public_model;
+  std::string upstream_model;
+  std::string endpoint_id;
+  ApiKind kind = ApiKind::kChat;
+  std::string client_ip;
+  long long req_prompt_chars = 0;
+  bool is_cloud = false;
+  std::string cloud_provider;
+  // buffered (non-stream) response body (decoded payload)
+  std::string buffered;
+  // anthropic ingress: re-encode OpenAI SSE -> Anthropic SSE
+  bool anthropic_reencode = false;
+  std::unique_ptr<class AnthropicStreamEncoder> reenc;
+};
+
+// ---------------------------------------------------------------------
+class UpstreamConn : public FdHandler {
+ public:
+  UpstreamConn(Worker* w, EndpointPtr ep, std::string host, int port);
+  ~UpstreamConn() override;
+  void on_event(uint32_t events) override;
+
+  Worker* w;
+  EndpointPtr ep;                 // may be null for cloud
+  std::string pool_key;
+  std::string host;
+  int port;
+  int fd = -1;
+  bool connected = false;
+  bool closed = false;
+  Buffer in, out;
+  ProxyOp* op = nullptr;          // active op (null when idle/pooled)
+  uint32_t events = 0;
+  bool reading_paused = false;
+
+  bool start();                    // begin non-blocking connect
+  void send_buffered();
+  void want(uint32_t ev);
+  void close_now();
+
+ private:
+  void handle_readable();
+  void handle_writable();
+};
+
+// ---------------------------------------------------------------------
+class ClientConn : public FdHandler {
+ public:
+  ClientConn(Worker* w, int fd, std::string peer_ip);
+  ~ClientConn() override;
+  void on_event(uint32_t events) override;
+
+  Worker* w;
+  int fd;
+  std::string peer_ip;
+  Buffer in, out;
+  HttpRequest req;
+  bool have_head = false;
+  bool want_close = false;
+  bool closed = false;
+  std::unique_ptr<ProxyOp> op;
+  uint32_t events = 0;
+  bool response_open = false;  // streaming response in progress
+
+  void flush_out();            // try write; arm EPOLLOUT if partial
+  void close_now();
+  void want(uint32_t ev);
+
+  // response helpers
+  void send_simple(int status, std::string_view content_type,
+                   std::string_view body,
+                   std::string_view extra_headers = {});
+  void send_json_error(int status, std::string_view code,
+                       std::string_view message);
+  void finish_request_cycle();  // reset for next pipelined request
+
+  void process_requests();
+
+ private:
+  void handle_readable();
+  void dispatch();
+};
+
+// ---------------------------------------------------------------------
+class Worker {
+ public:
+  Worker(Gateway* gw, int index);
+  void start();
+  void join();
+
+  Gateway* gw;
+  int index;
+  EventLoop loop;
+  std::thread thread;
+  int listen_fd = -1;
+
+  // keep-alive upstream pool: key -> idle conns
+  std::unordered_map<std::string, std::vector<UpstreamConn*>> pool;
+  UpstreamConn* acquire_upstream(const EndpointPtr& ep,
+                                 const std::string& key,
+                                 const std::string& host, int port);
+  void release_upstream(UpstreamConn* uc);   // pool for reuse
+  void discard_upstream(UpstreamConn* uc);   // close
+
+  struct Acceptor : FdHandler {
+    Worker* w;
+    void on_event(uint32_t events) override;
+  } acceptor;
+};
+
+// ---------------------------------------------------------------------
+class Gateway {
+ public:
+  explicit Gateway(GatewayConfig cfg);
+  ~Gateway();
+  bool init(std::string* err);
+  void run();          // blocks until stop()
+  void stop();
+
+  GatewayConfig cfg;
+  Db db;
+  Registry registry;
+  HistoryWriter history;
+  AuthStore auth;
+  Metrics metrics;
+  std::atomic<bool> draining{false};   // inference gate (self-update)
+  std::atomic<long long> start_ms{0};
+
+  std::vector<std::unique_ptr<Worker>> workers;
+
+  // request dispatch (called on worker threads)
+  void handle_request(ClientConn* c);
+
+  // local (non-proxied) handlers — api_local.cpp
+  void handle_local(ClientConn* c, const std::string& path,
+                    std::string_view body);
+  std::string render_models_json();
+  std::string render_summary_json();
+  std::string render_metrics_text();
+
+  // proxy path — proxy.cpp
+  void start_proxy(ClientConn* c, ApiKind kind, uint32_t required_caps,
+                   const std::string& upstream_path);
+  void start_cloud_proxy(ClientConn* c, ApiKind kind,
+                         const std::string& provider,
+                         const std::string& bare_model,
+                         const std::string& upstream_path,
+                         bool anthropic_native);
+  void start_control_plane_proxy(ClientConn* c);
+  void finalize_op(ClientConn* c, bool success, int status,
+                   const std::string& error);
+
+  // health checking — health.cpp
+  void health_loop();
+  std::thread health_thread;
+  std::atomic<bool> stopping{false};
+
+  // model sync — health.cpp
+  bool sync_endpoint_models(const EndpointPtr& ep, std::string* err);
+  // type detection probe chain — health.cpp
+  std::string detect_endpoint_type(const std::string& host, int port,
+                                   const std::string& prefix);
+};
+
+// blocking mini HTTP client used by the health checker / model sync /
+// detection (not on the hot path)
+struct MiniHttpResult {
+  int status = 0;
+  std::string body;
+  std::string error;
+  std::vector<std::pair<std::string, std::string>> headers;
+  double elapsed_ms = 0;
+  std::string header(std::string_view name) const {
+    for (auto& [k, v] : headers)
+      if (ieq(k, name)) return v;
+    return "";
+  }
+};
+MiniHttpResult mini_http_request(const std::string& host, int port,
+                                 const std::string& method,
+                                 const std::string& path,
+                                 const std::string& body, int timeout_ms,
+                                 const std::vector<std::string>& headers = {});
+
+}  // namespace gw

@@ -1,0 +1,135 @@
+#include "history.h"
+
+namespace gw {
+
+void HistoryWriter::start() {
+  // seed the audit chain tip from the DB (genesis if empty)
+  if (db_ && db_->ok()) {
+    std::vector<DbRow> rows;
+    db_->exec("SELECT hash FROM audit_log ORDER BY id DESC LIMIT 1", {},
+              &rows);
+    last_audit_hash_ = rows.empty() ? sha256_hex("genesis")
+                                    : rows[0][0].str();
+  }
+  stop_ = false;
+  thread_ = std::thread([this] { loop(); });
+}
+
+void HistoryWriter::stop() {
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    if (stop_) return;
+    stop_ = true;
+  }
+  cv_.notify_one();
+  if (thread_.joinable()) thread_.join();
+}
+
+void HistoryWriter::loop() {
+  while (true) {
+    std::vector<RequestRecord> recs;
+    std::vector<AuditRecord> audits;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      cv_.wait_for(lk, std::chrono::milliseconds(500), [this] {
+        return stop_ || queue_.size() > 512 || !audit_queue_.empty();
+      });
+      recs.assign(queue_.begin(), queue_.end());
+      queue_.clear();
+      audits.assign(audit_queue_.begin(), audit_queue_.end());
+      audit_queue_.clear();
+      if (stop_ && recs.empty() && audits.empty()) return;
+    }
+    flush(recs, audits);
+    // hourly retention cleanup (db/request_history.rs:1728)
+    int64_t now = wall_ms();
+    if (db_ && db_->ok() && now - last_cleanup_ms_ > 3600 * 1000) {
+      last_cleanup_ms_ = now;
+      int64_t cutoff = now - (int64_t)retention_days_ * 86400 * 1000;
+      db_->exec("DELETE FROM request_history WHERE ts_ms < ?",
+                {(long long)cutoff});
+      db_->exec("DELETE FROM health_checks WHERE ts_ms < ?",
+                {(long long)cutoff});
+    }
+  }
+}
+
+void HistoryWriter::flush(std::vector<RequestRecord>& recs,
+                          std::vector<AuditRecord>& audits) {
+  if (!db_ || !db_->ok()) return;
+  std::lock_guard<std::mutex> dblk(db_->mutex());
+  db_->exec("BEGIN", {});
+  for (auto& r : recs) {
+    db_->exec(
+        "INSERT INTO request_history (ts_ms,endpoint_id,model,api_kind,"
+        "status,duration_ms,prompt_tokens,completion_tokens,client_ip,"
+        "error) VALUES (?,?,?,?,?,?,?,?,?,?)",
+        {(long long)r.ts_ms, r.endpoint_id, r.model, r.api_kind,
+         (long long)r.status, r.duration_ms, r.prompt_tokens,
+         r.completion_tokens, r.client_ip, r.error});
+    // daily stats upsert (db/endpoint_daily_stats.rs)
+    char date[16];
+    time_t t = (time_t)(r.ts_ms / 1000);
+    struct tm tmv;
+    gmtime_r(&t, &tmv);
+    strftime(date, sizeof date, "%Y-%m-%d", &tmv);
+    bool fail = r.status >= 400 || r.status == 0;
+    db_->exec(
+        "INSERT INTO endpoint_daily_stats (date,endpoint_id,model,api_kind,"
+        "requests,failures,prompt_tokens,completion_tokens,"
+        "total_duration_ms) VALUES (?,?,?,?,1,?,?,?,?) "
+        "ON CONFLICT(date,endpoint_id,model,api_kind) DO UPDATE SET "
+        "requests=requests+1, failures=failures+excluded.failures, "
+        "prompt_tokens=prompt_tokens+excluded.prompt_tokens, "
+        "completion_tokens=completion_tokens+excluded.completion_tokens, "
+        "total_duration_ms=total_duration_ms+excluded.total_duration_ms",
+        {date, r.endpoint_id, r.model, r.api_kind, fail ? 1LL : 0LL,
+         r.prompt_tokens, r.completion_tokens, r.duration_ms});
+  }
+  for (auto& a : audits) {
+    // tamper-evident hash chain: hash = sha256(prev_hash || record)
+    char rec[512];
+    snprintf(rec, sizeof rec, "%lld|%s|%s|%d|%s|%s|%.3f",
+             (long long)a.ts_ms, a.method.c_str(), a.path.c_str(), a.status,
+             a.client_ip.c_str(), a.user.c_str(), a.duration_ms);
+    std::string h = sha256_hex(last_audit_hash_ + "|" + rec);
+    db_->exec(
+        "INSERT INTO audit_log (ts_ms,method,path,status,client_ip,user,"
+        "duration_ms,prev_hash,hash) VALUES (?,?,?,?,?,?,?,?,?)",
+        {(long long)a.ts_ms, a.method, a.path, (long long)a.status,
+         a.client_ip, a.user, a.duration_ms, last_audit_hash_, h});
+    last_audit_hash_ = h;
+  }
+  db_->exec("COMMIT", {});
+}
+
+bool HistoryWriter::verify_audit_chain(std::string* err) {
+  if (!db_ || !db_->ok()) return true;
+  std::vector<DbRow> rows;
+  std::lock_guard<std::mutex> dblk(db_->mutex());
+  if (!db_->exec("SELECT ts_ms,method,path,status,client_ip,user,"
+                 "duration_ms,prev_hash,hash FROM audit_log ORDER BY id",
+                 {}, &rows))
+    return false;
+  std::string prev = sha256_hex("genesis");
+  for (auto& r : rows) {
+    if (r[7].str() != prev) {
+      if (err) *err = "chain broken: prev_hash mismatch";
+      return false;
+    }
+    char rec[512];
+    snprintf(rec, sizeof rec, "%lld|%s|%s|%d|%s|%s|%.3f", r[0].i,
+             r[1].str().c_str(), r[2].str().c_str(), (int)r[3].i,
+             r[4].str().c_str(), r[5].str().c_str(),
+             r[6].kind == DbValue::kReal ? r[6].d : (double)r[6].i);
+    std::string h = sha256_hex(prev + "|" + rec);
+    if (h != r[8].str()) {
+      if (err) *err = "chain broken: record hash mismatch";
+      return false;
+    }
+    prev = h;
+  }
+  return true;
+}
+
+}  // namespace gw
