@@ -2,6 +2,7 @@
 
 #include <arpa/inet.h>
 #include <sys/resource.h>
+#include <unistd.h>
 
 namespace gw {
 
@@ -242,6 +243,7 @@ void ClientConn::want(uint32_t ev) {
 void ClientConn::close_now() {
   if (closed) return;
   closed = true;
+  w->timed_ops.erase(this);
   if (op) {
     // client went away mid-op: tear down upstream, record failure
     if (op->timer_id) w->loop.cancel_timer(op->timer_id);
@@ -336,7 +338,10 @@ void ClientConn::finish_request_cycle() {
   size_t need = req.header_len + (size_t)req.content_length;
   in.consume(need);
   have_head = false;
-  req = HttpRequest{};
+  req.headers.clear();  // keep vector capacity across requests
+  req.header_len = 0;
+  req.content_length = 0;
+  req.chunked = false;
   if (want_close) {
     flush_out();
     if (out.empty()) close_now();
@@ -419,6 +424,42 @@ void Worker::Acceptor::on_event(uint32_t) {
   }
 }
 
+void Worker::arm_record_flush() {
+  loop.add_timer(200, [this] {
+    if (!pending_records.empty()) gw->history.record_batch(pending_records);
+    arm_record_flush();
+  });
+}
+
+void Worker::arm_op_sweep() {
+  loop.add_timer(250, [this] {
+    int64_t now = now_us();
+    // collect first: timeout handling mutates timed_ops
+    std::vector<ClientConn*> expired;
+    for (ClientConn* c : timed_ops)
+      if (c->op && c->op->deadline_us && now > c->op->deadline_us)
+        expired.push_back(c);
+    for (ClientConn* c : expired) {
+      ProxyOp* o = c->op.get();
+      if (!o) continue;
+      if (o->upstream) {
+        UpstreamConn* uc = o->upstream;
+        uc->op = nullptr;
+        o->upstream = nullptr;
+        uc->close_now();
+      }
+      if (!o->resp_head_seen) {
+        c->send_json_error(504, "endpoint_timeout", "upstream timed out");
+        gw->finalize_op(c, false, 504, "timeout");
+      } else {
+        gw->finalize_op(c, false, o->resp.status, "stream_timeout");
+        c->close_now();
+      }
+    }
+    arm_op_sweep();
+  });
+}
+
 void Worker::start() {
   listen_fd = make_listener(gw->cfg.host, gw->cfg.port, true);
   if (listen_fd < 0) {
@@ -427,6 +468,8 @@ void Worker::start() {
     exit(1);
   }
   loop.add(listen_fd, EPOLLIN, &acceptor);
+  arm_record_flush();
+  arm_op_sweep();
   thread = std::thread([this] {
     char name[16];
     snprintf(name, sizeof name, "gw-worker-%d", index);
@@ -512,12 +555,17 @@ void Gateway::run() {
   health_thread = std::thread([this] { health_loop(); });
   LOG_INFO("llmlb-gw listening on %s:%d with %d workers",
            cfg.host.c_str(), cfg.port, (int)workers.size());
-  for (auto& w : workers)
-    if (w->thread.joinable()) w->thread.join();
+  // main thread parks until stop() is requested (signal handlers only
+  // set a flag — never join threads from signal context)
+  while (!stopping.load()) usleep(50 * 1000);
+  shutdown_threads();
 }
 
-void Gateway::stop() {
-  if (stopping.exchange(true)) return;
+void Gateway::stop() { stopping.store(true); }
+
+void Gateway::shutdown_threads() {
+  if (shutdown_done_.exchange(true)) return;
+  stopping.store(true);
   for (auto& w : workers) w->loop.stop();
   if (health_thread.joinable()) health_thread.join();
   for (auto& w : workers) w->join();

@@ -102,7 +102,8 @@ struct ProxyOp {
   bool streaming = false;
   bool lease_released_at_head = false;
   int64_t start_us = 0;
-  uint64_t timer_id = 0;
+  int64_t deadline_us = 0;   // inference timeout (swept by the worker)
+  uint64_t timer_id = 0;     // unused on the fast path
   // bookkeeping for history / TPS
   std::string public_model;
   std::string upstream_model;
@@ -114,6 +115,7 @@ struct ProxyOp {
   std::string cloud_provider;
   // buffered (non-stream) response body (decoded payload)
   std::string buffered;
+  std::shared_ptr<TpsState> tps;   // direct TPS state (no map lookup)
   // anthropic ingress: re-encode OpenAI SSE -> Anthropic SSE
   bool anthropic_reencode = false;
   std::unique_ptr<class AnthropicStreamEncoder> reenc;
@@ -205,6 +207,13 @@ class Worker {
   UpstreamConn* acquire_upstream(const EndpointPtr& ep,
                                  const std::string& key,
                                  const std::string& host, int port);
+  // per-worker record buffer: finalize_op appends lock-free (worker
+  // thread), a 200 ms loop timer flushes to the HistoryWriter
+  std::vector<RequestRecord> pending_records;
+  void arm_record_flush();
+  // ops with an armed inference timeout, swept every 250 ms
+  std::unordered_set<ClientConn*> timed_ops;
+  void arm_op_sweep();
   void release_upstream(UpstreamConn* uc);   // pool for reuse
   void discard_upstream(UpstreamConn* uc);   // close
 
@@ -221,7 +230,8 @@ class Gateway {
   ~Gateway();
   bool init(std::string* err);
   void run();          // blocks until stop()
-  void stop();
+  void stop();         // async-signal-safe: just requests shutdown
+  void shutdown_threads();
 
   GatewayConfig cfg;
   Db db;
@@ -260,6 +270,7 @@ class Gateway {
   void health_loop();
   std::thread health_thread;
   std::atomic<bool> stopping{false};
+  std::atomic<bool> shutdown_done_{false};
 
   // model sync — health.cpp
   bool sync_endpoint_models(const EndpointPtr& ep, std::string* err);

@@ -100,6 +100,18 @@ class HistoryWriter {
     }
     cv_.notify_one();
   }
+  // batched variant used by the workers (one lock for N records)
+  void record_batch(std::vector<RequestRecord>& rs) {
+    if (rs.empty()) return;
+    for (auto& r : rs) ring_.push(r);
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (auto& r : rs) queue_.push_back(std::move(r));
+      while (queue_.size() > 100000) queue_.pop_front();
+    }
+    cv_.notify_one();
+    rs.clear();
+  }
   void audit(AuditRecord r) {
     std::lock_guard<std::mutex> lk(mu_);
     audit_queue_.push_back(std::move(r));

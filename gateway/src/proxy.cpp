@@ -176,25 +176,9 @@ void Gateway::handle_request(ClientConn* c) {
 // Proxy start
 // -------------------------------------------------------------------
 static void arm_timeout(Gateway* gw, ClientConn* c, int timeout_s) {
-  ProxyOp* op = c->op.get();
-  op->timer_id = c->w->loop.add_timer((int64_t)timeout_s * 1000, [gw, c] {
-    ProxyOp* o = c->op.get();
-    if (!o) return;
-    o->timer_id = 0;
-    if (o->upstream) {
-      UpstreamConn* uc = o->upstream;
-      uc->op = nullptr;
-      o->upstream = nullptr;
-      uc->close_now();
-    }
-    if (!o->resp_head_seen) {
-      c->send_json_error(504, "endpoint_timeout", "upstream timed out");
-      gw->finalize_op(c, false, 504, "timeout");
-    } else {
-      gw->finalize_op(c, false, o->resp.status, "stream_timeout");
-      c->close_now();
-    }
-  });
+  (void)gw;
+  c->op->deadline_us = now_us() + (int64_t)timeout_s * 1000000;
+  c->w->timed_ops.insert(c);
 }
 
 void arm_timeout_public(Gateway* gw, ClientConn* c, int secs) {
@@ -263,7 +247,9 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
   op->kind = kind;
   op->client_ip = c->peer_ip;
   op->req_prompt_chars = (long long)body.size();
+  op->tps = sel.tps;
   op->lease = RequestLease(sel.ep, &registry, sel.canonical, kind);
+  op->lease.set_tps(sel.tps);
   c->op = std::move(op);
   uc->op = c->op.get();
 
@@ -628,6 +614,7 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
                           const std::string& error) {
   std::unique_ptr<ProxyOp> op = std::move(c->op);
   if (!op) return;
+  c->w->timed_ops.erase(c);
   if (op->timer_id) c->w->loop.cancel_timer(op->timer_id);
   if (op->upstream) {
     UpstreamConn* uc = op->upstream;
@@ -644,12 +631,19 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
     metrics.tokens_streamed.fetch_add(completion_tokens);
     if (success && op->lease_released_at_head && !op->endpoint_id.empty()) {
       // stream TPS recorded at end (api/proxy.rs:165-183)
-      EndpointPtr ep = registry.get(op->endpoint_id);
+      EndpointPtr ep = op->lease.endpoint();
+      if (!ep) ep = registry.get(op->endpoint_id);
       if (ep) {
         ep->update_latency(dur_ms);
-        if (completion_tokens > 0)
-          registry.record_tps(op->endpoint_id, op->public_model, op->kind,
-                              (double)completion_tokens, dur_ms / 1e3);
+        if (completion_tokens > 0) {
+          if (op->tps)
+            op->tps->update((double)completion_tokens, dur_ms / 1e3, 0.2,
+                            wall_ms());
+          else
+            registry.record_tps(op->endpoint_id, op->public_model,
+                                op->kind, (double)completion_tokens,
+                                dur_ms / 1e3);
+        }
       }
     }
   } else if (!op->buffered.empty()) {
@@ -664,16 +658,18 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
   RequestRecord rec;
   rec.ts_ms = wall_ms();
   rec.endpoint_id = op->is_cloud ? ("cloud:" + op->cloud_provider)
-                                 : op->endpoint_id;
-  rec.model = op->public_model;
+                                 : std::move(op->endpoint_id);
+  rec.model = std::move(op->public_model);
   rec.api_kind = api_kind_name(op->kind);
   rec.status = status;
   rec.duration_ms = dur_ms;
   rec.prompt_tokens = prompt_tokens;
   rec.completion_tokens = completion_tokens;
-  rec.client_ip = op->client_ip;
+  rec.client_ip = std::move(op->client_ip);
   rec.error = error;
-  history.record(std::move(rec));
+  c->w->pending_records.push_back(std::move(rec));
+  if (c->w->pending_records.size() >= 4096)
+    history.record_batch(c->w->pending_records);
 
   bool was_streaming = op->streaming;
   op.reset();

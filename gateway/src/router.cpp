@@ -407,6 +407,7 @@ SelectResult select_endpoint(const RouteTable& table,
   res.ep = route.endpoints[pick.idx];
   res.upstream_model = route.upstream_ids[pick.idx];
   res.canonical = route.canonical;
+  res.tps = route.tps[pick.idx];
   return res;
 }
 

@@ -223,6 +223,7 @@ class RequestLease {
     release_if_needed();
     ep_ = std::move(o.ep_);
     reg_ = o.reg_;
+    tps_ = std::move(o.tps_);
     model_ = std::move(o.model_);
     kind_ = o.kind_;
     start_us_ = o.start_us_;
@@ -237,6 +238,7 @@ class RequestLease {
   EndpointPtr endpoint() const { return ep_; }
   int64_t start_us() const { return start_us_; }
 
+  void set_tps(std::shared_ptr<TpsState> tps) { tps_ = std::move(tps); }
   void complete_success(long long completion_tokens) {
     if (!active()) return;
     done_ = true;
@@ -244,9 +246,13 @@ class RequestLease {
     double dur_s = (now_us() - start_us_) / 1e6;
     ep_->update_latency(dur_s * 1000.0);
     ep_->consecutive_failures.store(0, std::memory_order_relaxed);
-    if (completion_tokens > 0 && reg_)
-      reg_->record_tps(ep_->id, model_, kind_, (double)completion_tokens,
-                       dur_s);
+    if (completion_tokens > 0) {
+      if (tps_)
+        tps_->update((double)completion_tokens, dur_s, 0.2, wall_ms());
+      else if (reg_)
+        reg_->record_tps(ep_->id, model_, kind_, (double)completion_tokens,
+                         dur_s);
+    }
   }
   void complete_failure() {
     if (!active()) return;
@@ -261,6 +267,7 @@ class RequestLease {
   }
   EndpointPtr ep_;
   Registry* reg_ = nullptr;
+  std::shared_ptr<TpsState> tps_;
   std::string model_;
   ApiKind kind_ = ApiKind::kChat;
   int64_t start_us_ = 0;
@@ -279,6 +286,7 @@ struct SelectResult {
   EndpointPtr ep;
   std::string upstream_model;   // model id to send upstream
   std::string canonical;
+  std::shared_ptr<TpsState> tps;  // the (endpoint, model, chat) EMA cell
   int err_status = 0;           // 0 = ok; else 404/429/503
   std::string err_code;
 };
