@@ -1,0 +1,373 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark (driver contract).
+
+Measures BASELINE.json's headline metric: aggregate tokens/sec through
+the MI355X-native gateway with one Llama-3-8B bf16 replica per GPU
+(synthetic prompts, random-init weights), plus the gateway's p50 proxy
+overhead on the CPU-only mock config (config 1).
+
+  python bench.py --gpus N --steps K --warmup W
+
+For N > 1 the driver launches this under torch.distributed.run with one
+rank per GPU; ranks communicate over gloo (control only — serving
+traffic is HTTP through the gateway; RCCL is used inside TP groups,
+not across replicas).  One *step* = one wave of ``conc`` concurrent
+streaming chat requests per GPU (each ``gen_len`` new tokens) completing
+through the gateway.  Rank 0 prints ONE JSON line.
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import socket
+import subprocess
+import sys
+import tempfile
+import threading
+import time
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO))
+
+import torch  # noqa: E402
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def ensure_gateway_binaries() -> Path:
+    bin_dir = REPO / "gateway" / "bin"
+    if not (bin_dir / "llmlb-gw").exists():
+        subprocess.run(["make", "-C", str(REPO / "gateway"), "-j",
+                        str(os.cpu_count() or 4)], check=True,
+                       stdout=subprocess.DEVNULL)
+    return bin_dir
+
+
+# ---------------------------------------------------------------------
+# async load generator (rank 0): drives the gateway with streaming chat
+# ---------------------------------------------------------------------
+async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
+                   gen_len: int, api_key: str) -> tuple[int, int, list]:
+    import aiohttp
+
+    body = {
+        "model": model,
+        "messages": [{"role": "user",
+                      "content": " ".join(f"w{i}" for i in
+                                          range(prompt_words))}],
+        "max_tokens": gen_len,
+        "stream": True,
+        "stream_options": {"include_usage": True},
+    }
+    headers = {"Authorization": f"Bearer {api_key}"}
+    tokens = 0
+    errors = 0
+    latencies: list[float] = []
+
+    async def one(session):
+        nonlocal tokens, errors
+        t0 = time.monotonic()
+        try:
+            async with session.post(f"{base_url}/v1/chat/completions",
+                                    json=body, headers=headers) as resp:
+                if resp.status != 200:
+                    errors += 1
+                    await resp.read()
+                    return
+                usage_tokens = None
+                events = 0
+                async for line in resp.content:
+                    if not line.startswith(b"data:"):
+                        continue
+                    payload = line[5:].strip()
+                    if payload == b"[DONE]":
+                        break
+                    events += 1
+                    if b'"usage"' in payload:
+                        try:
+                            obj = json.loads(payload)
+                            u = obj.get("usage") or {}
+                            if "completion_tokens" in u:
+                                usage_tokens = u["completion_tokens"]
+                        except json.JSONDecodeError:
+                            pass
+                tokens += usage_tokens if usage_tokens is not None \
+                    else max(events - 2, 0)
+        except Exception:
+            errors += 1
+        finally:
+            latencies.append(time.monotonic() - t0)
+
+    connector = aiohttp.TCPConnector(limit=conc + 16)
+    timeout = aiohttp.ClientTimeout(total=600)
+    async with aiohttp.ClientSession(connector=connector,
+                                     timeout=timeout) as session:
+        await asyncio.gather(*[one(session) for _ in range(conc)])
+    return tokens, errors, latencies
+
+
+# ---------------------------------------------------------------------
+def measure_proxy_overhead(bin_dir: Path) -> dict:
+    """Config-1 micro-bench: gateway + mock on loopback, 3 s."""
+    out: dict = {"proxy_rps": None, "p50_proxy_overhead_ms": None}
+    try:
+        env = os.environ.copy()
+        mock_port, gw_port = free_port(), free_port()
+        db = tempfile.mktemp(suffix=".sqlite3")
+        mock = subprocess.Popen(
+            [str(bin_dir / "llmlb-mock"), "--port", str(mock_port),
+             "--workers", "2"], stdout=subprocess.DEVNULL,
+            stderr=subprocess.DEVNULL)
+        env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                    "LLMLB_ADMIN_TOKEN": "bench",
+                    "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+        gw = subprocess.Popen([str(bin_dir / "llmlb-gw"), "--workers",
+                               str(max(2, (os.cpu_count() or 4) // 2))],
+                              env=env, stdout=subprocess.DEVNULL,
+                              stderr=subprocess.DEVNULL)
+        time.sleep(0.6)
+        import urllib.request
+        req = urllib.request.Request(
+            f"http://127.0.0.1:{gw_port}/api/endpoints",
+            data=json.dumps({"name": "mock", "base_url":
+                             f"http://127.0.0.1:{mock_port}"}).encode(),
+            headers={"Authorization": "Bearer bench"})
+        for _ in range(10):
+            try:
+                urllib.request.urlopen(req, timeout=10)
+                break
+            except Exception:
+                time.sleep(0.5)
+        res = subprocess.run(
+            [str(bin_dir / "llmlb-bench"), "--port", str(gw_port),
+             "--threads", "2", "--connections", "50", "--duration", "3",
+             "--json"], capture_output=True, text=True, timeout=60)
+        gw.terminate()
+        mock.terminate()
+        line = res.stdout.strip().splitlines()[-1]
+        data = json.loads(line)
+        if data.get("non2xx", 1) == 0 and data.get("requests", 0) > 0:
+            out["proxy_rps"] = data["rps"]
+            out["p50_proxy_overhead_ms"] = data["p50_ms"]
+    except Exception as e:  # pragma: no cover
+        print(f"proxy-overhead micro-bench skipped: {e}", file=sys.stderr)
+    return out
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--conc", type=int, default=None,
+                    help="concurrent requests per GPU (default 64)")
+    ap.add_argument("--gen-len", type=int, default=128)
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--model", default=None,
+                    help="model preset (default llama-3-8b on GPU, "
+                         "llama-tiny on CPU)")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+    have_gpu = torch.cuda.is_available()
+    model_name = args.model or os.environ.get(
+        "LLMLB_BENCH_MODEL", "llama-3-8b" if have_gpu else "llama-tiny")
+    conc = args.conc or int(os.environ.get("LLMLB_BENCH_CONC", "64"))
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+
+    def sync():
+        if have_gpu:
+            torch.cuda.synchronize()
+
+    # ---- backend replica on every rank ------------------------------
+    from llmlb_amd.backend.launcher import build_engine
+    from llmlb_amd.backend.server import BackendServer, build_asgi_app
+    import uvicorn
+
+    if have_gpu:
+        torch.cuda.set_device(device)
+    kv_blocks = None
+    kv_gb = float(os.environ.get("LLMLB_BENCH_KV_GB", "0"))
+    if kv_gb > 0 and have_gpu:
+        from llmlb_amd.backend.kv_cache import KVCache
+        from llmlb_amd.backend.launcher import PRESETS
+        m = PRESETS[model_name]()
+        per_block = KVCache.bytes_per_block(
+            m.layers, m.kv_heads, m.head_dim)
+        kv_blocks = max(int(kv_gb * (1 << 30)) // per_block, 16)
+    engine = build_engine(model_name, device,
+                          kv_blocks=kv_blocks,
+                          max_batch=max(conc * 2, 128))
+    backend_port = 9300 + rank
+    served = BackendServer(engine, model_name)
+    app = build_asgi_app(served)
+    engine.start()
+    cfg = uvicorn.Config(app, host="127.0.0.1", port=backend_port,
+                         log_level="error", access_log=False)
+    usrv = uvicorn.Server(cfg)
+    threading.Thread(target=usrv.run, daemon=True).start()
+    deadline = time.time() + 60
+    while time.time() < deadline and not usrv.started:
+        time.sleep(0.1)
+
+    barrier()  # all backends up
+
+    gw_proc = None
+    result: dict = {}
+    try:
+        if rank == 0:
+            bin_dir = ensure_gateway_binaries()
+            gw_port = free_port()
+            db = tempfile.mktemp(suffix=".sqlite3")
+            env = os.environ.copy()
+            env.update({"LLMLB_PORT": str(gw_port),
+                        "LLMLB_DATABASE_URL": db,
+                        "LLMLB_ADMIN_TOKEN": "bench",
+                        "LLMLB_HEALTH_CHECK_INTERVAL": "30"})
+            gw_proc = subprocess.Popen(
+                [str(bin_dir / "llmlb-gw"), "--workers",
+                 str(max(2, min(8, (os.cpu_count() or 8) // 2)))],
+                env=env, stdout=subprocess.DEVNULL,
+                stderr=subprocess.DEVNULL)
+            time.sleep(0.6)
+            import urllib.request
+            for r in range(n_gpus):
+                req = urllib.request.Request(
+                    f"http://127.0.0.1:{gw_port}/api/endpoints",
+                    data=json.dumps(
+                        {"name": f"backend-{r}",
+                         "base_url": f"http://127.0.0.1:{9300 + r}",
+                         "device_index": r}).encode(),
+                    headers={"Authorization": "Bearer bench",
+                             "Content-Type": "application/json"})
+                for _ in range(30):
+                    try:
+                        urllib.request.urlopen(req, timeout=10)
+                        break
+                    except Exception:
+                        time.sleep(0.5)
+            # wait until the model routes
+            base = f"http://127.0.0.1:{gw_port}"
+            for _ in range(60):
+                try:
+                    with urllib.request.urlopen(base + "/v1/models",
+                                                timeout=5) as resp:
+                        if model_name in resp.read().decode():
+                            break
+                except Exception:
+                    pass
+                time.sleep(0.5)
+
+            total_conc = conc * n_gpus
+
+            def wave():
+                return asyncio.run(run_wave(base, model_name, total_conc,
+                                            args.prompt_len, args.gen_len,
+                                            "bench"))
+
+            for _ in range(args.warmup):
+                wave()
+        else:
+            pass  # other ranks: engines serve during rank 0's waves
+
+        # warmup complete on rank 0; ranks idle-serve meanwhile
+        barrier()
+        sync()
+        barrier()
+        t0 = time.monotonic()
+        total_tokens = 0
+        total_errors = 0
+        lat_all: list = []
+        if rank == 0:
+            for _ in range(args.steps):
+                tok, err, lats = asyncio.run(
+                    run_wave(f"http://127.0.0.1:{gw_port}", model_name,
+                             conc * n_gpus, args.prompt_len, args.gen_len,
+                             "bench"))
+                total_tokens += tok
+                total_errors += err
+                lat_all.extend(lats)
+        barrier()
+        sync()
+        barrier()
+        elapsed = time.monotonic() - t0
+        # max over ranks
+        if dist is not None:
+            t = torch.tensor([elapsed], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t[0])
+
+        if rank == 0:
+            lat_all.sort()
+            p50_req = lat_all[len(lat_all) // 2] if lat_all else None
+            value = total_tokens / elapsed if elapsed > 0 else 0.0
+            proxy = measure_proxy_overhead(REPO / "gateway" / "bin")
+            vsb = (proxy["proxy_rps"] / 170600.51
+                   if proxy.get("proxy_rps") else None)
+            result = {
+                "metric": "aggregate_tokens_per_sec",
+                "value": round(value, 2),
+                "unit": "tokens/s",
+                "n_gpus": n_gpus,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(elapsed / args.steps * 1000, 2),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": vsb,
+                "dtype": "bfloat16" if have_gpu else "float32",
+                "data": "synthetic",
+                "config": {
+                    "model": model_name,
+                    "global_batch": conc * n_gpus,
+                    "seq_len": args.prompt_len + args.gen_len,
+                    "prompt_len": args.prompt_len,
+                    "gen_len": args.gen_len,
+                    "parallelism": f"dp{n_gpus}",
+                    "route": "/v1/chat/completions (streaming SSE "
+                             "through llmlb-gw)",
+                    "errors": total_errors,
+                    "p50_request_s": round(p50_req, 3) if p50_req else None,
+                    "proxy_rps_config1": proxy.get("proxy_rps"),
+                    "p50_proxy_overhead_ms":
+                        proxy.get("p50_proxy_overhead_ms"),
+                    "baseline_note": "vs_baseline compares config-1 proxy "
+                                     "RPS against the reference's 170600.51"
+                                     " req/s headline (BASELINE.md); no "
+                                     "reference tokens/s exists",
+                },
+            }
+            print(json.dumps(result), flush=True)
+    finally:
+        if gw_proc is not None:
+            gw_proc.terminate()
+        engine.stop()
+        usrv.should_exit = True
+        if dist is not None:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

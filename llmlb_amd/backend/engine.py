@@ -106,11 +106,14 @@ class Engine:
 
     # -- public API ------------------------------------------------------
     def submit(self, prompt_ids: list[int],
-               params: SamplingParams | None = None) -> tuple[int, queue.Queue]:
-        """Queue a request; returns (request_id, token queue).
+               params: SamplingParams | None = None,
+               sink=None) -> tuple[int, queue.Queue]:
+        """Queue a request; returns (request_id, token sink).
 
-        The queue yields (token_id, finish_reason|None); finish_reason is
-        set on the final item."""
+        The sink receives .put((token_id, finish_reason|None)) calls from
+        the engine thread; finish_reason is set on the final item.  Pass
+        a custom ``sink`` (any object with put()) to bridge into asyncio
+        (see server._AsyncSink)."""
         params = params or SamplingParams()
         with self._rid_lock:
             rid = self._next_rid
@@ -119,7 +122,7 @@ class Engine:
         if params.seed is not None:
             gen = torch.Generator(device=self.device)
             gen.manual_seed(params.seed)
-        out_q: queue.Queue = queue.Queue()
+        out_q = sink if sink is not None else queue.Queue()
         seq = _Sequence(rid, list(prompt_ids), params, out_q, gen)
         self.waiting.put(seq)
         return rid, out_q

@@ -1,0 +1,343 @@
+"""OpenAI-compatible HTTP server for the MI355X backend engine.
+
+One server per engine (one engine per GPU; TP replicas run one server
+on rank 0 of the TP group).  Serves the surface the gateway's model
+sync and routing expect:
+
+  GET  /v1/models            — model listing (sync source)
+  POST /v1/chat/completions  — chat, stream + non-stream
+  POST /v1/completions       — legacy completions
+  GET  /api/health           — health with GPU telemetry (xLLM-style)
+  GET  /metrics              — Prometheus counters
+
+Chat messages are rendered with a trivial template and tokenized with a
+deterministic hash tokenizer by default (synthetic serving: BASELINE
+prescribes random-init weights + synthetic prompts; there is no network
+for real tokenizer downloads).  A HF tokenizer can be plugged in when
+available on disk.
+"""
+from __future__ import annotations
+
+import asyncio
+import hashlib
+import json
+import logging
+import time
+import uuid
+
+from .engine import Engine
+from .sampler import SamplingParams
+
+log = logging.getLogger("llmlb.backend")
+
+
+class SyntheticTokenizer:
+    """Deterministic text<->token mapping for synthetic serving."""
+
+    def __init__(self, vocab: int):
+        self.vocab = vocab
+
+    def encode(self, text: str) -> list[int]:
+        toks = []
+        for word in text.split():
+            h = int.from_bytes(
+                hashlib.blake2s(word.encode(), digest_size=4).digest(),
+                "little")
+            toks.append(h % max(self.vocab - 16, 1) + 16)
+        return toks or [1]
+
+    def decode_token(self, tok: int) -> str:
+        return f"tok{tok} "
+
+
+def render_chat(messages: list[dict]) -> str:
+    parts = []
+    for m in messages:
+        content = m.get("content", "")
+        if isinstance(content, list):
+            content = " ".join(
+                b.get("text", "") for b in content
+                if isinstance(b, dict))
+        parts.append(f"<|{m.get('role', 'user')}|> {content}")
+    parts.append("<|assistant|>")
+    return "\n".join(parts)
+
+
+class BackendServer:
+    def __init__(self, engine: Engine, model_name: str,
+                 served_models: list[str] | None = None,
+                 max_tokens_default: int = 128):
+        self.engine = engine
+        self.model_name = model_name
+        self.served = served_models or [model_name]
+        self.tokenizer = SyntheticTokenizer(engine.cfg.model.vocab)
+        self.max_tokens_default = max_tokens_default
+        self.requests_total = 0
+        self.tokens_total = 0
+        self.start_time = time.time()
+
+    # ---------------- request handling (framework-agnostic) ----------
+    def models_payload(self) -> dict:
+        return {
+            "object": "list",
+            "data": [
+                {"id": m, "object": "model", "owned_by": "llmlb_amd",
+                 "max_tokens": self.engine.cfg.model.max_pos}
+                for m in self.served
+            ],
+        }
+
+    def health_payload(self) -> dict:
+        gpu = {}
+        try:
+            import torch
+            if torch.cuda.is_available():
+                free, total = torch.cuda.mem_get_info(self.engine.device)
+                gpu = {
+                    "gpu_memory_total_mb": total // (1 << 20),
+                    "gpu_memory_used_mb": (total - free) // (1 << 20),
+                    "gpu_usage_percent": None,
+                }
+        except Exception:
+            pass
+        return {
+            "status": "ok",
+            "model": self.model_name,
+            "active_requests": self.engine.num_running,
+            "uptime_s": time.time() - self.start_time,
+            **gpu,
+        }
+
+    def parse_sampling(self, body: dict) -> SamplingParams:
+        return SamplingParams(
+            temperature=float(body.get("temperature", 0.0) or 0.0),
+            top_p=float(body.get("top_p", 1.0) or 1.0),
+            max_new_tokens=int(body.get("max_tokens")
+                               or body.get("max_completion_tokens")
+                               or self.max_tokens_default),
+            seed=body.get("seed"),
+        )
+
+    def submit_chat(self, body: dict, sink=None):
+        prompt = render_chat(body.get("messages", []))
+        prompt_ids = self.tokenizer.encode(prompt)
+        params = self.parse_sampling(body)
+        rid, out_q = self.engine.submit(prompt_ids, params, sink=sink)
+        self.requests_total += 1
+        return rid, out_q, prompt_ids
+
+    def submit_completion(self, body: dict, sink=None):
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = " ".join(map(str, prompt))
+        prompt_ids = self.tokenizer.encode(str(prompt))
+        params = self.parse_sampling(body)
+        rid, out_q = self.engine.submit(prompt_ids, params, sink=sink)
+        self.requests_total += 1
+        return rid, out_q, prompt_ids
+
+
+def build_asgi_app(server: BackendServer):
+    """Raw ASGI app (no framework dependency on the hot path)."""
+
+    async def read_body(receive) -> bytes:
+        chunks = []
+        while True:
+            msg = await receive()
+            if msg["type"] == "http.request":
+                chunks.append(msg.get("body", b""))
+                if not msg.get("more_body"):
+                    break
+            elif msg["type"] == "http.disconnect":
+                break
+        return b"".join(chunks)
+
+    async def send_json(send, status: int, obj: dict):
+        data = json.dumps(obj).encode()
+        await send({"type": "http.response.start", "status": status,
+                    "headers": [(b"content-type", b"application/json"),
+                                (b"content-length",
+                                 str(len(data)).encode())]})
+        await send({"type": "http.response.body", "body": data})
+
+    class _AsyncSink:
+        """Engine-thread -> asyncio bridge: put() is called from the
+        engine loop thread; items land in an asyncio.Queue without
+        blocking any executor thread (scales past 64 streams)."""
+
+        def __init__(self):
+            self.loop = asyncio.get_running_loop()
+            self.q: asyncio.Queue = asyncio.Queue()
+
+        def put(self, item):
+            self.loop.call_soon_threadsafe(self.q.put_nowait, item)
+
+    async def drain_queue_async(sink):
+        while True:
+            item = await sink.q.get()
+            yield item
+            if item[1]:
+                return
+
+    async def app(scope, receive, send):
+        if scope["type"] == "lifespan":
+            while True:
+                msg = await receive()
+                if msg["type"] == "lifespan.startup":
+                    await send({"type": "lifespan.startup.complete"})
+                elif msg["type"] == "lifespan.shutdown":
+                    await send({"type": "lifespan.shutdown.complete"})
+                    return
+            return
+        assert scope["type"] == "http"
+        path = scope["path"]
+        method = scope["method"]
+
+        if method == "GET" and path == "/v1/models":
+            await send_json(send, 200, server.models_payload())
+            return
+        if method == "GET" and path in ("/api/health", "/health"):
+            await send_json(send, 200, server.health_payload())
+            return
+        if method == "GET" and path == "/metrics":
+            text = (
+                f"backend_requests_total {server.requests_total}\n"
+                f"backend_tokens_total {server.engine.tokens_out}\n"
+                f"backend_active_requests {server.engine.num_running}\n"
+            ).encode()
+            await send({"type": "http.response.start", "status": 200,
+                        "headers": [(b"content-type", b"text/plain")]})
+            await send({"type": "http.response.body", "body": text})
+            return
+
+        if method == "POST" and path in ("/v1/chat/completions",
+                                         "/v1/completions",
+                                         "/v1/responses"):
+            try:
+                body = json.loads(await read_body(receive) or b"{}")
+            except json.JSONDecodeError:
+                await send_json(send, 400, {"error": {
+                    "message": "invalid JSON",
+                    "type": "invalid_request_error"}})
+                return
+            is_chat = path != "/v1/completions"
+            model = body.get("model", server.model_name)
+            if model not in server.served:
+                await send_json(send, 404, {"error": {
+                    "message": f"model '{model}' not served here",
+                    "type": "invalid_request_error",
+                    "code": "model_not_found"}})
+                return
+            sink = _AsyncSink()
+            try:
+                if is_chat:
+                    rid, out_q, prompt_ids = server.submit_chat(body, sink)
+                else:
+                    rid, out_q, prompt_ids = server.submit_completion(
+                        body, sink)
+            except MemoryError:
+                await send_json(send, 429, {"error": {
+                    "message": "KV cache exhausted",
+                    "type": "server_error",
+                    "code": "capacity_exceeded"}})
+                return
+            created = int(time.time())
+            cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
+            stream = bool(body.get("stream"))
+            include_usage = bool(
+                (body.get("stream_options") or {}).get("include_usage"))
+
+            if not stream:
+                toks, finish = [], "stop"
+                async for tok, fin in drain_queue_async(out_q):
+                    toks.append(tok)
+                    if fin:
+                        finish = fin
+                text = "".join(server.tokenizer.decode_token(t)
+                               for t in toks)
+                usage = {"prompt_tokens": len(prompt_ids),
+                         "completion_tokens": len(toks),
+                         "total_tokens": len(prompt_ids) + len(toks)}
+                if is_chat:
+                    payload = {
+                        "id": cid, "object": "chat.completion",
+                        "created": created, "model": model,
+                        "choices": [{"index": 0,
+                                     "message": {"role": "assistant",
+                                                 "content": text},
+                                     "finish_reason": finish}],
+                        "usage": usage}
+                else:
+                    payload = {
+                        "id": cid, "object": "text_completion",
+                        "created": created, "model": model,
+                        "choices": [{"index": 0, "text": text,
+                                     "finish_reason": finish}],
+                        "usage": usage}
+                await send_json(send, 200, payload)
+                return
+
+            # streaming SSE
+            await send({"type": "http.response.start", "status": 200,
+                        "headers": [(b"content-type",
+                                     b"text/event-stream"),
+                                    (b"cache-control", b"no-cache")]})
+
+            def chunk_payload(delta: dict, finish=None, usage=None):
+                obj = {"id": cid,
+                       "object": ("chat.completion.chunk" if is_chat
+                                  else "text_completion"),
+                       "created": created, "model": model,
+                       "choices": [{"index": 0,
+                                    ("delta" if is_chat else "text"):
+                                        delta if is_chat else
+                                        delta.get("content", ""),
+                                    "finish_reason": finish}]}
+                if usage is not None:
+                    obj["usage"] = usage
+                return ("data: " + json.dumps(obj) + "\n\n").encode()
+
+            if is_chat:
+                await send({"type": "http.response.body",
+                            "body": chunk_payload({"role": "assistant"}),
+                            "more_body": True})
+            ntok = 0
+            finish = "stop"
+            async for tok, fin in drain_queue_async(out_q):
+                ntok += 1
+                if fin:
+                    finish = fin
+                text = server.tokenizer.decode_token(tok)
+                await send({"type": "http.response.body",
+                            "body": chunk_payload({"content": text}),
+                            "more_body": True})
+            usage = ({"prompt_tokens": len(prompt_ids),
+                      "completion_tokens": ntok,
+                      "total_tokens": len(prompt_ids) + ntok}
+                     if include_usage else None)
+            await send({"type": "http.response.body",
+                        "body": chunk_payload({}, finish=finish,
+                                              usage=usage),
+                        "more_body": True})
+            await send({"type": "http.response.body",
+                        "body": b"data: [DONE]\n\n", "more_body": False})
+            return
+
+        await send_json(send, 404, {"error": {"message": "not found",
+                                              "type":
+                                                  "invalid_request_error"}})
+
+    return app
+
+
+def serve(engine: Engine, model_name: str, host: str = "0.0.0.0",
+          port: int = 9200, served_models: list[str] | None = None):
+    """Run the backend server (blocking).  Starts the engine loop."""
+    import uvicorn
+
+    server = BackendServer(engine, model_name, served_models)
+    app = build_asgi_app(server)
+    engine.start()
+    config = uvicorn.Config(app, host=host, port=port, log_level="warning",
+                            access_log=False)
+    uvicorn.Server(config).run()

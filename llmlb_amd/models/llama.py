@@ -119,17 +119,19 @@ class LlamaForCausalLM(nn.Module):
             requires_grad=False)
         self.cos_sin = op_ref.build_cos_sin_table(
             cfg.max_pos, cfg.head_dim, cfg.rope_theta)
-        self._init_weights(seed)
         self.to(device)
         self.cos_sin = self.cos_sin.to(device)
         self.device = torch.device(device)
+        self._init_weights(seed)
 
     def _init_weights(self, seed: int) -> None:
-        g = torch.Generator().manual_seed(seed)
+        # init on the target device (randn of 8B params on CPU is slow)
+        g = torch.Generator(device=self.device).manual_seed(seed)
         std = 0.02
 
         def init_(p: torch.Tensor):
             p.data.copy_(torch.randn(p.shape, generator=g,
+                                     device=self.device,
                                      dtype=torch.float32).mul_(std)
                          .to(p.dtype))
 
