@@ -1,0 +1,333 @@
+"""Integration tests: real llmlb-gw + llmlb-mock binaries over loopback
+(the reference tests the same flows with wiremock backends).
+CPU-only; binaries are built by __graft_entry__.build() / make.
+"""
+import json
+import os
+import socket
+import subprocess
+import tempfile
+import time
+import urllib.error
+import urllib.request
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parents[1]
+BIN = REPO / "gateway" / "bin"
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def _req(method, url, body=None, headers=None, timeout=10):
+    data = json.dumps(body).encode() if isinstance(body, (dict, list)) \
+        else (body.encode() if isinstance(body, str) else body)
+    r = urllib.request.Request(url, data=data, method=method,
+                               headers=headers or {})
+    try:
+        with urllib.request.urlopen(r, timeout=timeout) as resp:
+            return resp.status, resp.read().decode()
+    except urllib.error.HTTPError as e:
+        return e.code, e.read().decode()
+
+
+@pytest.fixture(scope="module")
+def stack():
+    """mock backend + gateway with an admin token and one API key."""
+    if not (BIN / "llmlb-gw").exists():
+        subprocess.run(["make", "-C", str(REPO / "gateway"), "-j", "4"],
+                       check=True, stdout=subprocess.DEVNULL)
+    mock_port, gw_port, cloud_port = _free_port(), _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    mock = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "mock-model,bge-embed"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    cloud = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(cloud_port), "--workers",
+         "1", "--models", "gpt-test"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({
+        "LLMLB_PORT": str(gw_port),
+        "LLMLB_DATABASE_URL": db,
+        "LLMLB_ADMIN_TOKEN": "admintok",
+        "LLMLB_HEALTH_CHECK_INTERVAL": "2",
+        "LLMLB_OPENAI_BASE_URL": f"http://127.0.0.1:{cloud_port}",
+        "OPENAI_API_KEY": "sk-cloudkey",
+    })
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "2"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    admin = {"Authorization": "Bearer admintok"}
+    # wait for the gateway then register the mock
+    for _ in range(50):
+        try:
+            _req("GET", base + "/health", timeout=2)
+            break
+        except Exception:
+            time.sleep(0.1)
+    status, _ = _req("POST", base + "/api/endpoints",
+                     {"name": "mock1",
+                      "base_url": f"http://127.0.0.1:{mock_port}"},
+                     headers=admin)
+    assert status == 201, status
+    # create an inference API key (switches off open mode)
+    status, body = _req("POST", base + "/api/api-keys",
+                        {"name": "t", "permissions": "inference"},
+                        headers=admin)
+    key = json.loads(body)["api_key"]
+    yield {"base": base, "admin": admin,
+           "auth": {"Authorization": f"Bearer {key}",
+                    "Content-Type": "application/json"},
+           "mock_port": mock_port, "gw": gw}
+    gw.terminate()
+    mock.terminate()
+    cloud.terminate()
+    try:
+        gw.wait(timeout=5)
+    except subprocess.TimeoutExpired:
+        gw.kill()
+    for p in Path("/tmp").glob(Path(db).name + "*"):
+        p.unlink(missing_ok=True)
+
+
+def test_models_list(stack):
+    status, body = _req("GET", stack["base"] + "/v1/models",
+                        headers=stack["auth"])
+    assert status == 200
+    ids = [m["id"] for m in json.loads(body)["data"]]
+    assert "mock-model" in ids and "bge-embed" in ids
+
+
+def test_chat_roundtrip(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/chat/completions",
+        {"model": "mock-model",
+         "messages": [{"role": "user", "content": "hi"}]},
+        headers=stack["auth"])
+    assert status == 200, body
+    obj = json.loads(body)
+    assert obj["model"] == "mock-model"
+    assert obj["usage"]["completion_tokens"] == 8
+    assert obj["choices"][0]["message"]["content"]
+
+
+def test_auth_required(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/chat/completions",
+        {"model": "mock-model", "messages": []})
+    assert status == 401
+    assert "invalid_api_key" in body
+
+
+def test_unknown_model_404(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/chat/completions",
+        {"model": "no-such-model", "messages": []},
+        headers=stack["auth"])
+    assert status == 404
+    assert "model_not_found" in body
+
+
+def test_streaming_sse_passthrough(stack):
+    req = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions",
+        data=json.dumps({"model": "mock-model", "stream": True,
+                         "messages": []}).encode(),
+        headers=stack["auth"])
+    with urllib.request.urlopen(req, timeout=10) as resp:
+        assert resp.status == 200
+        assert "text/event-stream" in resp.headers.get("Content-Type", "")
+        raw = resp.read().decode()
+    assert raw.count("data:") >= 9
+    assert "[DONE]" in raw
+    assert '"tok0 "' in raw
+
+
+def test_embeddings(stack):
+    status, body = _req("POST", stack["base"] + "/v1/embeddings",
+                        {"model": "bge-embed", "input": "hello"},
+                        headers=stack["auth"])
+    assert status == 200, body
+    assert json.loads(body)["data"][0]["embedding"]
+
+
+def test_responses_api(stack):
+    status, body = _req("POST", stack["base"] + "/v1/responses",
+                        {"model": "mock-model", "input": "hello"},
+                        headers=stack["auth"])
+    assert status == 200, body
+
+
+def test_anthropic_messages_nonstream(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/messages",
+        {"model": "mock-model", "max_tokens": 32,
+         "messages": [{"role": "user", "content": "hi"}]},
+        headers={**stack["auth"], "anthropic-version": "2023-06-01"})
+    assert status == 200, body
+    obj = json.loads(body)
+    assert obj["type"] == "message"
+    assert obj["role"] == "assistant"
+    assert obj["content"][0]["type"] == "text"
+    assert obj["stop_reason"] == "end_turn"
+    assert obj["usage"]["output_tokens"] == 8
+
+
+def test_anthropic_messages_requires_version_header(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/messages",
+        {"model": "mock-model", "max_tokens": 32, "messages": []},
+        headers=stack["auth"])
+    assert status == 400
+    assert "anthropic-version" in body
+
+
+def test_anthropic_messages_stream_reencoded(stack):
+    req = urllib.request.Request(
+        stack["base"] + "/v1/messages",
+        data=json.dumps({"model": "mock-model", "max_tokens": 32,
+                         "stream": True,
+                         "messages": [{"role": "user",
+                                       "content": "hi"}]}).encode(),
+        headers={**stack["auth"], "anthropic-version": "2023-06-01"})
+    with urllib.request.urlopen(req, timeout=10) as resp:
+        raw = resp.read().decode()
+    # re-encoded Anthropic event stream, in order
+    for ev in ["message_start", "content_block_start",
+               "content_block_delta", "content_block_stop",
+               "message_delta", "message_stop"]:
+        assert f"event: {ev}" in raw, raw[:800]
+    assert raw.index("message_start") < raw.index("content_block_delta") \
+        < raw.index("message_stop")
+    assert '"text_delta"' in raw
+
+
+def test_cloud_prefix_openai(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/chat/completions",
+        {"model": "openai:gpt-test",
+         "messages": [{"role": "user", "content": "hi"}]},
+        headers=stack["auth"])
+    assert status == 200, body
+    # model rewritten back? cloud providers echo the bare name; the
+    # gateway passes the body through with the prefix stripped upstream
+    obj = json.loads(body)
+    assert obj["choices"][0]["message"]["content"]
+
+
+def test_cloud_prefix_unconfigured_provider(stack):
+    status, body = _req(
+        "POST", stack["base"] + "/v1/chat/completions",
+        {"model": "google:gemini-x", "messages": []},
+        headers=stack["auth"])
+    assert status == 502
+    assert "cloud_provider_unconfigured" in body
+
+
+def test_audio_capability_routing_503_when_absent(stack):
+    status, body = _req("POST", stack["base"] + "/v1/audio/speech",
+                        {"model": "tts-1", "input": "hello"},
+                        headers=stack["auth"])
+    # the mock's models don't advertise audio capability
+    assert status == 503
+    assert "no_capable_endpoint" in body
+
+
+def test_request_history_recorded(stack):
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "mock-model", "messages": []}, headers=stack["auth"])
+    time.sleep(0.5)  # batched flush
+    status, body = _req("GET", stack["base"] + "/api/requests",
+                        headers=stack["admin"])
+    assert status == 200
+    recs = json.loads(body)
+    assert any(r["model"] == "mock-model" and r["status"] == 200
+               for r in recs)
+
+
+def test_dashboard_summary(stack):
+    status, body = _req("GET", stack["base"] + "/api/dashboard/summary",
+                        headers=stack["admin"])
+    assert status == 200
+    obj = json.loads(body)
+    assert obj["endpoints"][0]["status"] == "online"
+    assert obj["last_hour"]["requests"] >= 1
+
+
+def test_prometheus_metrics(stack):
+    status, body = _req("GET", stack["base"] + "/api/metrics")
+    assert status == 200
+    assert "llmlb_requests_total" in body
+
+
+def test_audit_chain_verifies(stack):
+    time.sleep(0.6)
+    status, body = _req("GET", stack["base"] + "/api/audit/verify",
+                        headers=stack["admin"])
+    assert status == 200
+    assert json.loads(body)["valid"] is True
+
+
+def test_drain_gate(stack):
+    _req("POST", stack["base"] + "/api/system/drain", {},
+         headers=stack["admin"])
+    status, body = _req("POST", stack["base"] + "/v1/chat/completions",
+                        {"model": "mock-model", "messages": []},
+                        headers=stack["auth"])
+    assert status == 503 and "draining" in body
+    _req("POST", stack["base"] + "/api/system/resume", {},
+         headers=stack["admin"])
+    status, _ = _req("POST", stack["base"] + "/v1/chat/completions",
+                     {"model": "mock-model", "messages": []},
+                     headers=stack["auth"])
+    assert status == 200
+
+
+def test_endpoint_failure_ejected_and_recovers(stack):
+    """Register an endpoint that dies -> health marks it offline and
+    routing stops using it; its model 404s (config 5 behavior)."""
+    port = _free_port()
+    victim = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(port), "--workers", "1",
+         "--models", "ephemeral-model"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    for _ in range(50):  # wait for the victim to accept connections
+        try:
+            socket.create_connection(("127.0.0.1", port), 0.2).close()
+            break
+        except OSError:
+            time.sleep(0.1)
+    status, body = _req("POST", stack["base"] + "/api/endpoints",
+                        {"name": "victim",
+                         "base_url": f"http://127.0.0.1:{port}"},
+                        headers=stack["admin"])
+    assert status == 201
+    ep_id = json.loads(body)["id"]
+    status, _ = _req("POST", stack["base"] + "/v1/chat/completions",
+                     {"model": "ephemeral-model", "messages": []},
+                     headers=stack["auth"])
+    assert status == 200
+    victim.terminate()
+    victim.wait()
+    # 2-fail hysteresis at 2 s interval -> offline within ~8 s
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        status, _ = _req("POST", stack["base"] + "/v1/chat/completions",
+                         {"model": "ephemeral-model", "messages": []},
+                         headers=stack["auth"])
+        if status in (404, 503):
+            break
+        time.sleep(1)
+    assert status in (404, 503), f"endpoint never ejected (last {status})"
+    _req("DELETE", stack["base"] + f"/api/endpoints/{ep_id}",
+         headers=stack["admin"])
