@@ -3,22 +3,26 @@
 One engine per GPU (one process per GPU at node scale; TP replicas span
 a process group over RCCL/xGMI).  Each ``step()``:
 
-  1. admits queued prompts (prefill) while KV pages are available,
+  1. admits queued prompts while KV pages are available and runs their
+     prefill BATCHED (grouped by prompt length; one SDPA + one set of
+     GEMMs per group instead of per sequence),
   2. runs one decode iteration for every running sequence through the
-     paged-attention kernel (single batched launch per layer),
-  3. samples, streams tokens to per-request queues, retires finished
+     paged-attention kernel — decode state (block tables, lengths, last
+     tokens) lives in persistent device tensors so the per-step CPU
+     work is one small gather, not a Python loop,
+  3. samples, streams tokens to per-request sinks, retires finished
      sequences and releases their pages.
 
-Prefill attention uses torch SDPA (flash/math) on the prompt's own
-K/V; the prompt K/V is simultaneously appended to the paged cache by
-the gfx950 kv_append kernel.  Decode attention is the hand-written
-paged kernel (llmlb_amd/ops/csrc/kernels.hip).
+Prefill attention uses torch SDPA on the prompt's own K/V (flash path
+on ROCm); decode attention is the hand-written gfx950 paged kernel
+(llmlb_amd/ops/csrc/kernels.hip).
 """
 from __future__ import annotations
 
 import queue
 import threading
 import time
+from collections import defaultdict
 from dataclasses import dataclass, field
 
 import torch
@@ -27,6 +31,14 @@ from .. import ops
 from ..models.llama import LlamaConfig, LlamaForCausalLM
 from .kv_cache import BLOCK_SIZE, KVCache
 from .sampler import SamplingParams, sample
+
+
+def os_env_flag(name: str, default: bool) -> bool:
+    import os
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v.lower() not in ("0", "false", "no", "off")
 
 
 @dataclass
@@ -52,7 +64,8 @@ class GenerationResult:
 
 class _Sequence:
     __slots__ = ("rid", "prompt_ids", "out_ids", "params", "block_table",
-                 "length", "out_queue", "generator", "t_submit", "t_first")
+                 "length", "out_queue", "generator", "t_submit", "t_first",
+                 "row")
 
     def __init__(self, rid, prompt_ids, params, out_queue, generator):
         self.rid = rid
@@ -65,6 +78,89 @@ class _Sequence:
         self.generator = generator
         self.t_submit = time.monotonic()
         self.t_first = 0.0
+        self.row = -1                        # persistent decode-state row
+
+
+class _DecodeGraph:
+    """One hipGraph-captured decode forward at a fixed (padded) batch.
+
+    The decode inner loop is launch-bound (~450 kernel launches per
+    step at 32 layers); capturing it in a hipGraph replays the whole
+    step at graph-replay cost (MI355X_MICROARCH.md row
+    graph-replay-floor).  Padded rows point at the reserved sacrificial
+    KV block 0 (their K/V writes and attention reads are discarded)."""
+
+    BUCKETS = (8, 16, 32, 64, 128, 256)
+
+    def __init__(self, engine: "Engine", S: int):
+        self.S = S
+        dev = engine.device
+        MB = engine.max_blocks_per_seq
+        self.ids = torch.zeros(S, dtype=torch.long, device=dev)
+        self.lens = torch.zeros(S, dtype=torch.int32, device=dev)
+        self.bt = torch.zeros(S, MB, dtype=torch.int32, device=dev)
+        self.slots = torch.zeros(S, dtype=torch.int32, device=dev)
+        self.seq_lens = torch.ones(S, dtype=torch.int32, device=dev)
+        # fixed partition count per bucket (fills the 256-CU chip)
+        base = S * engine.n_kv_local
+        n_parts = 1
+        while base * n_parts < 512 and n_parts < 16:
+            n_parts *= 2
+        self.n_parts = n_parts
+        nq, D = engine.n_q_local, engine.cfg.model.head_dim
+        if n_parts > 1:
+            self.ws = (torch.empty(S, nq, n_parts, D, dtype=torch.float32,
+                                   device=dev),
+                       torch.empty(S, nq, n_parts, 2, dtype=torch.float32,
+                                   device=dev))
+        else:
+            self.ws = None
+        self.graph: torch.cuda.CUDAGraph | None = None
+        self.logits: torch.Tensor | None = None
+        self._engine = engine
+
+    def _forward(self) -> torch.Tensor:
+        eng = self._engine
+        h = eng.model.embed_tokens(self.ids)
+        residual = None
+        for li in range(eng.cfg.model.layers):
+            q, k, v, residual = eng.model.layer_attn_prelude(
+                li, h, residual, self.lens)
+            ops.kv_append(eng.kv.k[li], eng.kv.v[li], k, v, self.slots)
+            attn = ops.paged_attn_decode(
+                q, eng.kv.k[li], eng.kv.v[li], self.bt, self.seq_lens,
+                eng.scale, n_parts=self.n_parts, workspace=self.ws)
+            h, residual = eng.model.layer_attn_epilogue(li, attn, residual)
+        return eng.model.final(h, residual)
+
+    def capture(self) -> None:
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):  # warmup (allocator, hipBLASLt algo pick)
+                self._forward()
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.logits = self._forward()
+
+    def run(self, ids, lens, bt, slots, seq_lens) -> torch.Tensor:
+        n = ids.shape[0]
+        self.ids[:n] = ids
+        self.lens[:n] = lens
+        self.bt[:n] = bt
+        self.slots[:n] = slots
+        self.seq_lens[:n] = seq_lens
+        if n < self.S:
+            # padded rows: sacrificial block 0, length 0
+            self.lens[n:] = 0
+            self.bt[n:] = 0
+            self.slots[n:] = 0
+            self.seq_lens[n:] = 1
+        self.graph.replay()
+        return self.logits[:n]
 
 
 class Engine:
@@ -88,10 +184,29 @@ class Engine:
         self.scale = m.head_dim ** -0.5
         self.n_q_local = m.q_heads // self.model.tp.world_size
         self.n_kv_local = n_kv_local
-        # decode-attention workspace reused across steps
-        self._ws = None
         self.steps_done = 0
         self.tokens_out = 0
+
+        # persistent decode state (device-resident)
+        self.max_rows = cfg.max_batch
+        self.max_blocks_per_seq = min(
+            (m.max_pos + BLOCK_SIZE - 1) // BLOCK_SIZE, num_blocks)
+        dev = self.device
+        self._bt_dev = torch.zeros(self.max_rows, self.max_blocks_per_seq,
+                                   dtype=torch.int32, device=dev)
+        self._len_dev = torch.zeros(self.max_rows, dtype=torch.int32,
+                                    device=dev)
+        self._last_tok = torch.zeros(self.max_rows, dtype=torch.long,
+                                     device=dev)
+        self._free_rows = list(range(self.max_rows - 1, -1, -1))
+        self._rows_cache: torch.Tensor | None = None  # [S] long
+        self._rows_dirty = True
+        # hipGraph-captured decode steps, one per padded batch bucket
+        self.use_graphs = (self.device.type == "cuda"
+                           and self.model.tp.world_size == 1
+                           and cfg.model.layers > 0
+                           and os_env_flag("LLMLB_DECODE_GRAPHS", True))
+        self._graphs: dict[int, _DecodeGraph] = {}
 
     # -- sizing ----------------------------------------------------------
     def _auto_blocks(self, n_kv_local: int) -> int:
@@ -170,7 +285,7 @@ class Engine:
         admitted: list[_Sequence] = []
         budget = self.cfg.max_prefill_tokens
         while (len(self.running) + len(admitted) < self.cfg.max_batch
-               and budget > 0):
+               and budget > 0 and self._free_rows):
             try:
                 seq = self.waiting.get_nowait()
             except queue.Empty:
@@ -183,6 +298,7 @@ class Engine:
                 self.waiting.put(seq)
                 break
             seq.block_table = self.kv.allocate(need)
+            seq.row = self._free_rows.pop()
             budget -= len(seq.prompt_ids)
             admitted.append(seq)
         return admitted
@@ -190,109 +306,176 @@ class Engine:
     def step(self) -> bool:
         """One scheduler iteration.  Returns True if any work was done."""
         prefills = self._admit()
-        for seq in prefills:
-            self._prefill(seq)
+        if prefills:
+            self._prefill_batch(prefills)
         if self.running:
             self._decode_step()
         self.steps_done += 1
         return bool(prefills or self.running)
 
     # -- forward passes --------------------------------------------------
-    def _slots_for(self, seq: _Sequence, start: int, n: int) -> torch.Tensor:
-        slots = [seq.block_table[(start + i) // BLOCK_SIZE] * BLOCK_SIZE
-                 + (start + i) % BLOCK_SIZE for i in range(n)]
-        return torch.tensor(slots, dtype=torch.int32, device=self.device)
+    def _slots_for(self, seq: _Sequence, start: int, n: int) -> list[int]:
+        return [seq.block_table[(start + i) // BLOCK_SIZE] * BLOCK_SIZE
+                + (start + i) % BLOCK_SIZE for i in range(n)]
 
     @torch.inference_mode()
-    def _prefill(self, seq: _Sequence) -> None:
-        T = len(seq.prompt_ids)
-        ids = torch.tensor(seq.prompt_ids, dtype=torch.long,
-                           device=self.device)
-        positions = torch.arange(T, dtype=torch.int32, device=self.device)
-        slots = self._slots_for(seq, 0, T)
-        h = self.model.embed_tokens(ids)
-        residual = None
-        for li in range(self.cfg.model.layers):
-            q, k, v, residual = self.model.layer_attn_prelude(
-                li, h, residual, positions)
-            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
-            attn = self._sdpa_causal(q, k, v)
-            h, residual = self.model.layer_attn_epilogue(li, attn, residual)
-        logits = self.model.final(h[-1:], residual[-1:])
-        seq.length = T
-        self.running.append(seq)
-        self._emit(seq, logits)
+    def _prefill_batch(self, seqs: list[_Sequence]) -> None:
+        """Batched prefill, grouped by prompt length (same-length groups
+        share one packed forward: GEMMs batched, one SDPA per group)."""
+        groups: dict[int, list[_Sequence]] = defaultdict(list)
+        for s in seqs:
+            groups[len(s.prompt_ids)].append(s)
+        dev = self.device
+        for L, group in groups.items():
+            B = len(group)
+            ids = torch.tensor([s.prompt_ids for s in group],
+                               dtype=torch.long, device=dev).reshape(-1)
+            positions = torch.arange(L, dtype=torch.int32, device=dev) \
+                .repeat(B)
+            slots = torch.tensor(
+                [slot for s in group for slot in self._slots_for(s, 0, L)],
+                dtype=torch.int32, device=dev)
+            h = self.model.embed_tokens(ids)
+            residual = None
+            nq, nkv, D = self.n_q_local, self.n_kv_local, \
+                self.cfg.model.head_dim
+            for li in range(self.cfg.model.layers):
+                q, k, v, residual = self.model.layer_attn_prelude(
+                    li, h, residual, positions)
+                ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+                qh = q.view(B, L, nq, D).transpose(1, 2)
+                kh = k.view(B, L, nkv, D).transpose(1, 2)
+                vh = v.view(B, L, nkv, D).transpose(1, 2)
+                o = torch.nn.functional.scaled_dot_product_attention(
+                    qh, kh, vh, is_causal=L > 1, enable_gqa=True)
+                attn = o.transpose(1, 2).reshape(B * L, nq, D).contiguous()
+                h, residual = self.model.layer_attn_epilogue(li, attn,
+                                                             residual)
+            last_idx = torch.arange(1, B + 1, device=dev) * L - 1
+            logits = self.model.final(h[last_idx], residual[last_idx])
+            toks = sample(logits, [s.params for s in group],
+                          [s.generator for s in group])
+            # install decode state
+            rows = torch.tensor([s.row for s in group], dtype=torch.long,
+                                device=dev)
+            bt_rows = torch.zeros(B, self.max_blocks_per_seq,
+                                  dtype=torch.int32)
+            for i, s in enumerate(group):
+                bt_rows[i, :len(s.block_table)] = torch.tensor(
+                    s.block_table, dtype=torch.int32)
+                s.length = L
+            self._bt_dev[rows] = bt_rows.to(dev)
+            self._len_dev[rows] = L
+            self._last_tok[rows] = toks
+            self.running.extend(group)
+            self._rows_dirty = True
+            toks_cpu = toks.tolist()
+            finished = []
+            for s, tok in zip(group, toks_cpu):
+                if not self._push_token(s, tok):
+                    finished.append(s)
+            for s in finished:
+                self._retire(s)
 
-    def _sdpa_causal(self, q, k, v) -> torch.Tensor:
-        """q: [T, n_q, D], k/v: [T, n_kv, D] -> [T, n_q, D]"""
-        T = q.shape[0]
-        qh = q.transpose(0, 1).unsqueeze(0)          # [1, n_q, T, D]
-        kh = k.transpose(0, 1).unsqueeze(0)
-        vh = v.transpose(0, 1).unsqueeze(0)
-        o = torch.nn.functional.scaled_dot_product_attention(
-            qh, kh, vh, is_causal=T > 1, enable_gqa=True)
-        return o.squeeze(0).transpose(0, 1).contiguous()  # [T, n_q, D]
+    def _rows_tensor(self) -> torch.Tensor:
+        if self._rows_dirty or self._rows_cache is None:
+            self._rows_cache = torch.tensor(
+                [s.row for s in self.running], dtype=torch.long,
+                device=self.device)
+            self._rows_dirty = False
+        return self._rows_cache
 
     @torch.inference_mode()
     def _decode_step(self) -> None:
-        S = len(self.running)
-        ids, positions, slots = [], [], []
+        # grow block tables where the NEXT token crosses a page boundary
+        grow_rows, grow_cols, grow_vals = [], [], []
         for seq in self.running:
-            # grow block table when the next token crosses a page boundary
             if seq.length >= len(seq.block_table) * BLOCK_SIZE:
-                seq.block_table.extend(self.kv.allocate(1))
-            last = seq.out_ids[-1] if seq.out_ids else seq.prompt_ids[-1]
-            ids.append(last)
-            positions.append(seq.length)
-            slots.append(seq.block_table[seq.length // BLOCK_SIZE]
-                         * BLOCK_SIZE + seq.length % BLOCK_SIZE)
-        dev = self.device
-        ids_t = torch.tensor(ids, dtype=torch.long, device=dev)
-        pos_t = torch.tensor(positions, dtype=torch.int32, device=dev)
-        slots_t = torch.tensor(slots, dtype=torch.int32, device=dev)
-        max_blocks = max(len(s.block_table) for s in self.running)
-        bt = torch.zeros(S, max_blocks, dtype=torch.int32, device=dev)
-        for i, s in enumerate(self.running):
-            bt[i, :len(s.block_table)] = torch.tensor(
-                s.block_table, dtype=torch.int32)
-        seq_lens = torch.tensor([s.length + 1 for s in self.running],
-                                dtype=torch.int32, device=dev)
+                blk = self.kv.allocate(1)[0]
+                seq.block_table.append(blk)
+                grow_rows.append(seq.row)
+                grow_cols.append(len(seq.block_table) - 1)
+                grow_vals.append(blk)
+        if grow_rows:
+            self._bt_dev[torch.tensor(grow_rows, device=self.device),
+                         torch.tensor(grow_cols, device=self.device)] = \
+                torch.tensor(grow_vals, dtype=torch.int32,
+                             device=self.device)
 
-        h = self.model.embed_tokens(ids_t)
-        residual = None
-        for li in range(self.cfg.model.layers):
-            q, k, v, residual = self.model.layer_attn_prelude(
-                li, h, residual, pos_t)
-            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots_t)
-            attn = ops.paged_attn_decode(q, self.kv.k[li], self.kv.v[li],
-                                         bt, seq_lens, self.scale)
-            h, residual = self.model.layer_attn_epilogue(li, attn, residual)
-        logits = self.model.final(h, residual)
-        for seq in self.running:
-            seq.length += 1
-        self._emit_batch(logits)
+        rows = self._rows_tensor()
+        lens = self._len_dev[rows]                      # [S] int32
+        ids = self._last_tok[rows]                      # [S] long
+        bt = self._bt_dev[rows]                         # [S, MB] int32
+        blk_idx = (lens // BLOCK_SIZE).long()
+        slots = bt.gather(1, blk_idx.unsqueeze(1)).squeeze(1) \
+            * BLOCK_SIZE + (lens % BLOCK_SIZE)
+        seq_lens = lens + 1
 
-    # -- emission / retirement -------------------------------------------
-    def _emit(self, seq: _Sequence, logits: torch.Tensor) -> None:
-        tok = sample(logits, [seq.params], [seq.generator])
-        if not self._push_token(seq, int(tok[0])):
-            # finished on the very first token (max_new_tokens == 1)
-            self.running.remove(seq)
-            self.kv.release(seq.block_table)
-            seq.block_table = []
-
-    def _emit_batch(self, logits: torch.Tensor) -> None:
+        S = len(self.running)
+        graph = self._graph_for(S)
+        if graph is not None:
+            logits = graph.run(ids, lens, bt, slots, seq_lens)
+        else:
+            h = self.model.embed_tokens(ids)
+            residual = None
+            for li in range(self.cfg.model.layers):
+                q, k, v, residual = self.model.layer_attn_prelude(
+                    li, h, residual, lens)
+                ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+                attn = ops.paged_attn_decode(q, self.kv.k[li],
+                                             self.kv.v[li], bt, seq_lens,
+                                             self.scale)
+                h, residual = self.model.layer_attn_epilogue(li, attn,
+                                                             residual)
+            logits = self.model.final(h, residual)
         toks = sample(logits, [s.params for s in self.running],
                       [s.generator for s in self.running])
-        toks = toks.tolist()
+        self._last_tok[rows] = toks
+        self._len_dev[rows] = lens + 1
+        toks_cpu = toks.tolist()
         finished = []
-        for seq, tok in zip(self.running, toks):
+        for seq, tok in zip(self.running, toks_cpu):
+            seq.length += 1
             if not self._push_token(seq, tok):
                 finished.append(seq)
         for seq in finished:
+            self._retire(seq)
+
+    def _graph_for(self, S: int) -> "_DecodeGraph | None":
+        if not self.use_graphs:
+            return None
+        bucket = None
+        for b in _DecodeGraph.BUCKETS:
+            if S <= b <= self.max_rows:
+                bucket = b
+                break
+        if bucket is None:
+            return None
+        g = self._graphs.get(bucket)
+        if g is None:
+            g = _DecodeGraph(self, bucket)
+            try:
+                g.capture()
+            except Exception as e:  # graphs are an optimization only
+                import logging
+                logging.getLogger("llmlb.engine").warning(
+                    "decode graph capture failed (bucket %d): %s", bucket,
+                    e)
+                self.use_graphs = False
+                return None
+            self._graphs[bucket] = g
+        return g
+
+    # -- emission / retirement -------------------------------------------
+    def _retire(self, seq: _Sequence) -> None:
+        if seq in self.running:
             self.running.remove(seq)
-            self.kv.release(seq.block_table)
-            seq.block_table = []
+        self.kv.release(seq.block_table)
+        seq.block_table = []
+        if seq.row >= 0:
+            self._free_rows.append(seq.row)
+            seq.row = -1
+        self._rows_dirty = True
 
     def _push_token(self, seq: _Sequence, tok: int) -> bool:
         """Returns False when the sequence is finished."""

@@ -26,7 +26,9 @@ class KVCache:
         self.v = [torch.zeros(num_blocks, n_kv, BLOCK_SIZE, head_dim,
                               dtype=dtype, device=device)
                   for _ in range(layers)]
-        self._free: list[int] = list(range(num_blocks - 1, -1, -1))
+        # block 0 is reserved as the sacrificial target of padded rows
+        # in hipGraph-captured decode steps (engine._DecodeGraph)
+        self._free: list[int] = list(range(num_blocks - 1, 0, -1))
 
     @staticmethod
     def bytes_per_block(layers: int, n_kv: int, head_dim: int,

@@ -94,7 +94,12 @@ __global__ void rmsnorm_kernel(bf16* __restrict__ out,
       unpack8(resv[i], r);
 #pragma unroll
       for (int j = 0; j < 8; ++j) f[j] += r[j];
-      resv[i] = pack8(f);  // residual stream carries the sum forward
+      uint4 packed = pack8(f);
+      resv[i] = packed;  // residual stream carries the sum forward
+      // the norm consumes the STORED (bf16-rounded) residual, so the
+      // sum of squares must use the rounded values too (pass 2 re-reads
+      // resv) — otherwise rrms is computed against different data
+      unpack8(packed, f);
     }
 #pragma unroll
     for (int j = 0; j < 8; ++j) ssq += f[j] * f[j];

@@ -38,16 +38,33 @@ def test_gpu_decode_consistency(gpu_engine):
     assert mismatch <= 1, f"{mismatch} mismatches"
 
 
-def test_gpu_matches_cpu_reference_structure(gpu_engine):
-    """Same tiny config on CPU fp32 vs GPU bf16: greedy tokens should
-    broadly agree for a short horizon (random-init model)."""
+def test_gpu_matches_cpu_reference_logits(gpu_engine):
+    """Prefill logits of the GPU bf16 stack (HIP kernels + hipBLASLt)
+    must correlate with the CPU fp32 reference stack on the same
+    random-init weights.  (Token-level argmax agreement is flaky by
+    design on a random-init model with near-tied logits.)"""
     cfg = LlamaConfig.tiny()
     cpu = Engine(EngineConfig(model=cfg, device="cpu", dtype="float32",
                               kv_cache_blocks=64))
-    g = gpu_engine.generate_sync([2, 4, 6], max_new_tokens=3)
-    c = cpu.generate_sync([2, 4, 6], max_new_tokens=3)
-    agree = sum(int(a == b) for a, b in zip(g.token_ids, c.token_ids))
-    assert agree >= 2, (g.token_ids, c.token_ids)
+
+    def prefill_logits(eng, prompt):
+        ids = torch.tensor(prompt, dtype=torch.long, device=eng.device)
+        pos = torch.arange(len(prompt), dtype=torch.int32,
+                           device=eng.device)
+        h = eng.model.embed_tokens(ids)
+        residual = None
+        for li in range(eng.cfg.model.layers):
+            q, k, v, residual = eng.model.layer_attn_prelude(
+                li, h, residual, pos)
+            attn = eng._sdpa_causal(q, k, v)
+            h, residual = eng.model.layer_attn_epilogue(li, attn, residual)
+        return eng.model.final(h[-1:], residual[-1:]).float().cpu()[0]
+
+    prompt = [2, 4, 6, 8, 10]
+    lg = prefill_logits(gpu_engine, prompt)
+    lc = prefill_logits(cpu, prompt)
+    cos = torch.nn.functional.cosine_similarity(lg, lc, dim=0)
+    assert cos > 0.98, f"cosine {cos}"
 
 
 def test_gpu_many_concurrent(gpu_engine):
