@@ -60,7 +60,16 @@ def main():
         sync()
         decode_s = time.monotonic() - t_dec
         total_decoded = sum(q.qsize() for q in qs)
+        timing = {}
+        if eng.timing is not None and eng.timing.get("steps"):
+            n = eng.timing["steps"]
+            timing = {k: round(v / n * 1000, 3)
+                      for k, v in eng.timing.items() if k != "steps"}
+            eng.timing.clear()
         results[f"bs{bs}"] = {
+            "graphs": sorted(eng._graphs.keys()),
+            "use_graphs": eng.use_graphs,
+            "timing_ms_per_step": timing,
             "prefill_tokens_per_s": round(bs * args.prompt_len / prefill_s,
                                           1),
             "prefill_s": round(prefill_s, 3),

@@ -1,0 +1,507 @@
+"""Python control plane for llmlb-gw.
+
+The C++ data plane serves all /v1/* inference routes and the hot admin
+surface; it proxies everything it does not handle (dashboard data
+drill-downs, auth/users, invitations, audit queries, TPS benchmarks,
+catalog, WS events) to this FastAPI app over loopback
+(LLMLB_CONTROL_PLANE_URL).  Shares the gateway's SQLite database (WAL).
+
+Reference coverage: components 20 (dashboard API), 21
+(dashboard WS), 22 (benchmarks), 23 (catalog), 24 (users/api_keys/
+invitations/audit API), 25 (auth: JWT login, password policy).
+bcrypt is not in this image, so passwords use PBKDF2-HMAC-SHA256
+(documented deviation; same role in the schema).
+"""
+from __future__ import annotations
+
+import asyncio
+import base64
+import hashlib
+import hmac
+import json
+import os
+import secrets
+import sqlite3
+import time
+import urllib.request
+from contextlib import contextmanager
+
+from fastapi import (Depends, FastAPI, HTTPException, Query, WebSocket,
+                     WebSocketDisconnect)
+from fastapi.responses import JSONResponse
+from pydantic import BaseModel
+
+# --------------------------------------------------------------- jwt
+def _b64url(data: bytes) -> str:
+    return base64.urlsafe_b64encode(data).rstrip(b"=").decode()
+
+
+def jwt_encode(payload: dict, secret: str) -> str:
+    header = _b64url(json.dumps({"alg": "HS256", "typ": "JWT"}).encode())
+    body = _b64url(json.dumps(payload).encode())
+    msg = f"{header}.{body}".encode()
+    sig = _b64url(hmac.new(secret.encode(), msg, hashlib.sha256).digest())
+    return f"{header}.{body}.{sig}"
+
+
+def jwt_decode(token: str, secret: str) -> dict | None:
+    try:
+        header, body, sig = token.split(".")
+        msg = f"{header}.{body}".encode()
+        expect = _b64url(hmac.new(secret.encode(), msg,
+                                  hashlib.sha256).digest())
+        if not hmac.compare_digest(sig, expect):
+            return None
+        pad = "=" * (-len(body) % 4)
+        payload = json.loads(base64.urlsafe_b64decode(body + pad))
+        if payload.get("exp", 0) < time.time():
+            return None
+        return payload
+    except Exception:
+        return None
+
+
+def hash_password(password: str, salt: bytes | None = None) -> str:
+    salt = salt or secrets.token_bytes(16)
+    dk = hashlib.pbkdf2_hmac("sha256", password.encode(), salt, 100_000)
+    return f"pbkdf2${salt.hex()}${dk.hex()}"
+
+
+def verify_password(password: str, stored: str) -> bool:
+    try:
+        _, salt_hex, dk_hex = stored.split("$")
+        dk = hashlib.pbkdf2_hmac("sha256", password.encode(),
+                                 bytes.fromhex(salt_hex), 100_000)
+        return hmac.compare_digest(dk.hex(), dk_hex)
+    except Exception:
+        return False
+
+
+# --------------------------------------------------------------- app
+class LoginBody(BaseModel):
+    username: str
+    password: str
+
+
+class UserBody(BaseModel):
+    username: str
+    password: str
+    role: str = "user"
+
+
+class ChangePasswordBody(BaseModel):
+    old_password: str
+    new_password: str
+
+
+class InvitationBody(BaseModel):
+    email: str = ""
+    role: str = "user"
+    expires_hours: int = 72
+
+
+class RegisterBody(BaseModel):
+    token: str
+    username: str
+    password: str
+
+
+class TpsBenchmarkBody(BaseModel):
+    # reference api/benchmarks.rs:25-34 defaults and caps
+    model: str
+    requests: int = 20
+    concurrency: int = 4
+    max_tokens: int = 128
+    temperature: float = 0.2
+
+
+def create_app(db_path: str | None = None,
+               gateway_url: str | None = None,
+               jwt_secret: str | None = None,
+               admin_token: str | None = None) -> FastAPI:
+    db_path = db_path or os.environ.get("LLMLB_DATABASE_URL",
+                                        "llmlb.sqlite3")
+    gateway_url = gateway_url or os.environ.get(
+        "LLMLB_GATEWAY_URL", "http://127.0.0.1:32768")
+    jwt_secret = jwt_secret or os.environ.get("LLMLB_JWT_SECRET") \
+        or secrets.token_hex(32)
+    admin_token = admin_token or os.environ.get("LLMLB_ADMIN_TOKEN", "")
+
+    app = FastAPI(title="llmlb control plane", docs_url=None,
+                  redoc_url=None)
+
+    @contextmanager
+    def db():
+        conn = sqlite3.connect(db_path, timeout=10)
+        conn.row_factory = sqlite3.Row
+        try:
+            yield conn
+            conn.commit()
+        finally:
+            conn.close()
+
+    def ensure_admin():
+        admin_user = os.environ.get("LLMLB_ADMIN_USER", "admin")
+        admin_pass = os.environ.get("LLMLB_ADMIN_PASSWORD", "")
+        with db() as conn:
+            row = conn.execute("SELECT COUNT(*) c FROM users").fetchone()
+            if row["c"] == 0:
+                pw = admin_pass or secrets.token_urlsafe(12)
+                conn.execute(
+                    "INSERT INTO users (username,password_hash,role,"
+                    "must_change_password,created_at) VALUES (?,?,?,?,?)",
+                    (admin_user, hash_password(pw), "admin",
+                     0 if admin_pass else 1,
+                     time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())))
+                if not admin_pass:
+                    print(f"[control-plane] created admin user "
+                          f"'{admin_user}' with password: {pw}",
+                          flush=True)
+
+    try:
+        ensure_admin()
+    except sqlite3.OperationalError:
+        pass  # gateway migrates the schema; first boot ordering
+
+    # ---- auth dependency ----
+    from fastapi import Header
+
+    def auth_user(authorization: str | None = Header(None)):
+        if not authorization:
+            raise HTTPException(401, "missing Authorization")
+        token = authorization
+        if token.lower().startswith("bearer "):
+            token = token[7:]
+        if admin_token and token == admin_token:
+            return {"username": "admin-token", "role": "admin"}
+        payload = jwt_decode(token, jwt_secret)
+        if not payload:
+            raise HTTPException(401, "invalid token")
+        return payload
+
+    def require_admin(user=Depends(auth_user)):
+        if user.get("role") != "admin":
+            raise HTTPException(403, "admin role required")
+        return user
+
+    # ---------------- auth ----------------
+    @app.post("/api/auth/login")
+    def login(body: LoginBody):
+        with db() as conn:
+            row = conn.execute(
+                "SELECT id,username,password_hash,role,"
+                "must_change_password FROM users WHERE username=?",
+                (body.username,)).fetchone()
+        if not row or not verify_password(body.password,
+                                          row["password_hash"]):
+            raise HTTPException(401, "invalid credentials")
+        token = jwt_encode(
+            {"sub": row["id"], "username": row["username"],
+             "role": row["role"], "exp": time.time() + 24 * 3600},
+            jwt_secret)
+        return {"token": token, "role": row["role"],
+                "must_change_password": bool(row["must_change_password"])}
+
+    @app.post("/api/auth/change-password")
+    def change_password(body: ChangePasswordBody, user=Depends(auth_user)):
+        with db() as conn:
+            row = conn.execute(
+                "SELECT id,password_hash FROM users WHERE username=?",
+                (user["username"],)).fetchone()
+            if not row or not verify_password(body.old_password,
+                                              row["password_hash"]):
+                raise HTTPException(401, "wrong password")
+            conn.execute(
+                "UPDATE users SET password_hash=?,must_change_password=0 "
+                "WHERE id=?", (hash_password(body.new_password),
+                               row["id"]))
+        return {"changed": True}
+
+    @app.get("/api/auth/me")
+    def me(user=Depends(auth_user)):
+        return {"username": user.get("username"), "role": user.get("role")}
+
+    # ---------------- users ----------------
+    @app.get("/api/users")
+    def list_users(user=Depends(require_admin)):
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT id,username,role,must_change_password,created_at "
+                "FROM users").fetchall()
+        return [dict(r) for r in rows]
+
+    @app.post("/api/users", status_code=201)
+    def create_user(body: UserBody, user=Depends(require_admin)):
+        with db() as conn:
+            try:
+                conn.execute(
+                    "INSERT INTO users (username,password_hash,role,"
+                    "created_at) VALUES (?,?,?,?)",
+                    (body.username, hash_password(body.password),
+                     body.role,
+                     time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())))
+            except sqlite3.IntegrityError:
+                raise HTTPException(409, "username exists")
+        return {"created": body.username}
+
+    @app.delete("/api/users/{username}")
+    def delete_user(username: str, user=Depends(require_admin)):
+        with db() as conn:
+            cur = conn.execute("DELETE FROM users WHERE username=? AND "
+                               "role != 'admin'", (username,))
+            if cur.rowcount == 0:
+                raise HTTPException(404, "no such (non-admin) user")
+        return {"deleted": username}
+
+    # ---------------- invitations ----------------
+    @app.post("/api/invitations", status_code=201)
+    def create_invitation(body: InvitationBody,
+                          user=Depends(require_admin)):
+        token = secrets.token_urlsafe(24)
+        with db() as conn:
+            conn.execute(
+                "INSERT INTO invitations (token,email,role,created_at,"
+                "expires_at) VALUES (?,?,?,?,?)",
+                (token, body.email, body.role,
+                 time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime()),
+                 time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime(
+                     time.time() + body.expires_hours * 3600))))
+        return {"token": token, "role": body.role}
+
+    @app.get("/api/invitations")
+    def list_invitations(user=Depends(require_admin)):
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT id,token,email,role,created_at,expires_at,used_at "
+                "FROM invitations").fetchall()
+        return [dict(r) for r in rows]
+
+    @app.post("/api/auth/register", status_code=201)
+    def register(body: RegisterBody):
+        now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+        with db() as conn:
+            inv = conn.execute(
+                "SELECT id,role,expires_at,used_at FROM invitations "
+                "WHERE token=?", (body.token,)).fetchone()
+            if not inv or inv["used_at"]:
+                raise HTTPException(400, "invalid invitation")
+            if inv["expires_at"] and inv["expires_at"] < now:
+                raise HTTPException(400, "invitation expired")
+            try:
+                conn.execute(
+                    "INSERT INTO users (username,password_hash,role,"
+                    "created_at) VALUES (?,?,?,?)",
+                    (body.username, hash_password(body.password),
+                     inv["role"], now))
+            except sqlite3.IntegrityError:
+                raise HTTPException(409, "username exists")
+            conn.execute("UPDATE invitations SET used_at=? WHERE id=?",
+                         (now, inv["id"]))
+        return {"created": body.username, "role": inv["role"]}
+
+    # ---------------- audit queries ----------------
+    @app.get("/api/audit")
+    def audit_log(limit: int = Query(100, le=1000), offset: int = 0,
+                  q: str = "", user=Depends(require_admin)):
+        sql = ("SELECT id,ts_ms,method,path,status,client_ip,user,"
+               "duration_ms,hash FROM audit_log")
+        params: list = []
+        if q:
+            sql += " WHERE path LIKE ? OR user LIKE ? OR client_ip LIKE ?"
+            params += [f"%{q}%"] * 3
+        sql += " ORDER BY id DESC LIMIT ? OFFSET ?"
+        params += [limit, offset]
+        with db() as conn:
+            rows = conn.execute(sql, params).fetchall()
+        return [dict(r) for r in rows]
+
+    # ---------------- dashboard stats (drill-downs) ----------------
+    @app.get("/api/dashboard/token-stats")
+    def token_stats(days: int = Query(30, le=365),
+                    user=Depends(auth_user)):
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT date, SUM(prompt_tokens) pt, "
+                "SUM(completion_tokens) ct, SUM(requests) reqs, "
+                "SUM(failures) fails FROM endpoint_daily_stats "
+                "GROUP BY date ORDER BY date DESC LIMIT ?",
+                (days,)).fetchall()
+        return [dict(r) for r in rows]
+
+    @app.get("/api/dashboard/clients")
+    def client_analytics(limit: int = Query(20, le=100),
+                         user=Depends(auth_user)):
+        cutoff = int(time.time() * 1000) - 7 * 86400 * 1000
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT client_ip, COUNT(*) requests, "
+                "SUM(completion_tokens) tokens, AVG(duration_ms) avg_ms "
+                "FROM request_history WHERE ts_ms > ? GROUP BY client_ip "
+                "ORDER BY requests DESC LIMIT ?",
+                (cutoff, limit)).fetchall()
+        return [dict(r) for r in rows]
+
+    @app.get("/api/dashboard/timeline")
+    def request_timeline(hours: int = Query(24, le=168),
+                         bucket_minutes: int = 10,
+                         user=Depends(auth_user)):
+        cutoff = int(time.time() * 1000) - hours * 3600 * 1000
+        bucket_ms = bucket_minutes * 60 * 1000
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT (ts_ms / ?) * ? bucket, COUNT(*) requests, "
+                "SUM(CASE WHEN status >= 400 THEN 1 ELSE 0 END) failures, "
+                "SUM(completion_tokens) tokens FROM request_history "
+                "WHERE ts_ms > ? GROUP BY bucket ORDER BY bucket",
+                (bucket_ms, bucket_ms, cutoff)).fetchall()
+        return [dict(r) for r in rows]
+
+    # ---------------- TPS benchmarks (api/benchmarks.rs) ----------------
+    bench_results: dict[str, dict] = {}
+
+    def _gateway_headers():
+        return {"Authorization": f"Bearer {admin_token}",
+                "Content-Type": "application/json"}
+
+    @app.post("/api/benchmarks/tps", status_code=202)
+    async def start_tps_benchmark(body: TpsBenchmarkBody,
+                                  user=Depends(require_admin)):
+        body.requests = min(body.requests, 500)       # caps (ref :25-34)
+        body.concurrency = min(body.concurrency, 64)
+        body.max_tokens = min(body.max_tokens, 4096)
+        bench_id = secrets.token_hex(8)
+        bench_results[bench_id] = {"status": "running",
+                                   "model": body.model}
+
+        async def run():
+            import aiohttp
+            payload = {"model": body.model,
+                       "messages": [{"role": "user",
+                                     "content": "Benchmark prompt."}],
+                       "max_tokens": body.max_tokens,
+                       "temperature": body.temperature}
+            sem = asyncio.Semaphore(body.concurrency)
+            tokens = 0
+            errors = 0
+            t0 = time.monotonic()
+
+            async def one(session):
+                nonlocal tokens, errors
+                async with sem:
+                    try:
+                        async with session.post(
+                                gateway_url + "/v1/chat/completions",
+                                json=payload,
+                                headers=_gateway_headers()) as resp:
+                            data = await resp.json()
+                            if resp.status == 200:
+                                tokens += data.get("usage", {}).get(
+                                    "completion_tokens", 0)
+                            else:
+                                errors += 1
+                    except Exception:
+                        errors += 1
+
+            async with aiohttp.ClientSession() as session:
+                await asyncio.gather(
+                    *[one(session) for _ in range(body.requests)])
+            dur = time.monotonic() - t0
+            bench_results[bench_id] = {
+                "status": "done", "model": body.model,
+                "requests": body.requests, "errors": errors,
+                "duration_s": round(dur, 3),
+                "total_tokens": tokens,
+                "tokens_per_sec": round(tokens / dur, 2) if dur else 0,
+                "source": "benchmark",  # TpsSource::Benchmark
+            }
+
+        asyncio.get_event_loop().create_task(run())
+        return {"benchmark_id": bench_id}
+
+    @app.get("/api/benchmarks/tps/{bench_id}")
+    def get_tps_benchmark(bench_id: str, user=Depends(require_admin)):
+        if bench_id not in bench_results:
+            raise HTTPException(404, "no such benchmark")
+        return bench_results[bench_id]
+
+    # ---------------- catalog (local, no egress) ----------------
+    catalog_path = os.environ.get(
+        "LLMLB_CATALOG_PATH",
+        str(os.path.join(os.path.dirname(__file__), "model_catalog.json")))
+
+    @app.get("/api/catalog/search")
+    def catalog_search(q: str = "", limit: int = Query(20, le=100),
+                       user=Depends(auth_user)):
+        try:
+            with open(catalog_path) as f:
+                catalog = json.load(f)
+        except OSError:
+            catalog = []
+        ql = q.lower()
+        hits = [m for m in catalog
+                if not ql or ql in m.get("id", "").lower()
+                or ql in " ".join(m.get("tags", [])).lower()]
+        return hits[:limit]
+
+    # ---------------- dashboard WebSocket ----------------
+    @app.websocket("/ws/dashboard")
+    async def ws_dashboard(ws: WebSocket):
+        token = ws.query_params.get("token", "")
+        ok = (admin_token and token == admin_token) or \
+            jwt_decode(token, jwt_secret)
+        if not ok:
+            await ws.close(code=4401)
+            return
+        await ws.accept()
+        try:
+            while True:
+                # push a summary snapshot every 2 s (event stream analog)
+                try:
+                    req = urllib.request.Request(
+                        gateway_url + "/api/dashboard/summary",
+                        headers=_gateway_headers())
+                    with urllib.request.urlopen(req, timeout=5) as resp:
+                        summary = json.loads(resp.read())
+                    await ws.send_json({"type": "summary",
+                                        "data": summary,
+                                        "ts": time.time()})
+                except Exception as e:
+                    await ws.send_json({"type": "error", "error": str(e)})
+                await asyncio.sleep(2)
+        except WebSocketDisconnect:
+            return
+
+    @app.get("/api/control/health")
+    def control_health():
+        return {"status": "ok", "component": "control-plane"}
+
+    # minimal dashboard landing page (the reference embeds a React SPA;
+    # our ops surface is the REST + WS API above)
+    @app.get("/dashboard")
+    def dashboard_page():
+        return JSONResponse({
+            "dashboard": "llmlb_amd control plane",
+            "endpoints": ["/api/dashboard/summary (data plane)",
+                          "/api/dashboard/token-stats",
+                          "/api/dashboard/clients",
+                          "/api/dashboard/timeline",
+                          "/ws/dashboard?token=..."],
+        })
+
+    return app
+
+
+def main():
+    import argparse
+
+    import uvicorn
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=32700)
+    args = ap.parse_args()
+    uvicorn.run(create_app(), host=args.host, port=args.port,
+                log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

@@ -207,6 +207,8 @@ class Engine:
                            and cfg.model.layers > 0
                            and os_env_flag("LLMLB_DECODE_GRAPHS", True))
         self._graphs: dict[int, _DecodeGraph] = {}
+        self.timing = defaultdict(float) \
+            if os_env_flag("LLMLB_ENGINE_TIMING", False) else None
 
     # -- sizing ----------------------------------------------------------
     def _auto_blocks(self, n_kv_local: int) -> int:
@@ -387,6 +389,7 @@ class Engine:
 
     @torch.inference_mode()
     def _decode_step(self) -> None:
+        _t0 = time.monotonic()
         # grow block tables where the NEXT token crosses a page boundary
         grow_rows, grow_cols, grow_vals = [], [], []
         for seq in self.running:
@@ -412,6 +415,10 @@ class Engine:
         seq_lens = lens + 1
 
         S = len(self.running)
+        if self.timing is not None:
+            torch.cuda.synchronize() if self.device.type == "cuda" else None
+            self.timing["prep"] += time.monotonic() - _t0
+            _t1 = time.monotonic()
         graph = self._graph_for(S)
         if graph is not None:
             logits = graph.run(ids, lens, bt, slots, seq_lens)
@@ -428,11 +435,18 @@ class Engine:
                 h, residual = self.model.layer_attn_epilogue(li, attn,
                                                              residual)
             logits = self.model.final(h, residual)
+        if self.timing is not None:
+            torch.cuda.synchronize() if self.device.type == "cuda" else None
+            self.timing["forward"] += time.monotonic() - _t1
+            _t2 = time.monotonic()
         toks = sample(logits, [s.params for s in self.running],
                       [s.generator for s in self.running])
         self._last_tok[rows] = toks
         self._len_dev[rows] = lens + 1
         toks_cpu = toks.tolist()
+        if self.timing is not None:
+            self.timing["sample_sync"] += time.monotonic() - _t2
+            _t3 = time.monotonic()
         finished = []
         for seq, tok in zip(self.running, toks_cpu):
             seq.length += 1

@@ -19,6 +19,8 @@ def sample(logits: torch.Tensor, params_list: list[SamplingParams],
            generators: list[torch.Generator | None]) -> torch.Tensor:
     """logits: [S, vocab] -> [S] int64 next tokens."""
     S = logits.shape[0]
+    if all(p.temperature < 1e-5 for p in params_list):
+        return logits.argmax(dim=-1)  # all-greedy fast path (no H2D)
     temps = torch.tensor(
         [max(p.temperature, 0.0) for p in params_list],
         device=logits.device, dtype=torch.float32)

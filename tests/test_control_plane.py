@@ -1,0 +1,142 @@
+"""Control-plane tests (FastAPI TestClient against an in-memory DB)."""
+import sqlite3
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from llmlb_amd.api.control_plane import (create_app, hash_password,
+                                         jwt_decode, jwt_encode,
+                                         verify_password)
+
+
+@pytest.fixture()
+def app_client(tmp_path, monkeypatch):
+    db = str(tmp_path / "cp.sqlite3")
+    # create the schema pieces the control plane reads/writes
+    conn = sqlite3.connect(db)
+    conn.executescript("""
+    CREATE TABLE users (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      username TEXT UNIQUE, password_hash TEXT, role TEXT DEFAULT 'user',
+      must_change_password INTEGER DEFAULT 0, created_at TEXT);
+    CREATE TABLE invitations (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      token TEXT UNIQUE, email TEXT, role TEXT, created_by INTEGER,
+      created_at TEXT, expires_at TEXT, used_at TEXT);
+    CREATE TABLE audit_log (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      ts_ms INTEGER, method TEXT, path TEXT, status INTEGER,
+      client_ip TEXT, user TEXT, duration_ms REAL, prev_hash TEXT,
+      hash TEXT);
+    CREATE TABLE endpoint_daily_stats (date TEXT, endpoint_id TEXT,
+      model TEXT, api_kind TEXT, requests INTEGER, failures INTEGER,
+      prompt_tokens INTEGER, completion_tokens INTEGER,
+      total_duration_ms REAL);
+    CREATE TABLE request_history (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      ts_ms INTEGER, endpoint_id TEXT, model TEXT, api_kind TEXT,
+      status INTEGER, duration_ms REAL, prompt_tokens INTEGER,
+      completion_tokens INTEGER, client_ip TEXT, error TEXT,
+      request_body TEXT, response_body TEXT);
+    """)
+    conn.commit()
+    conn.close()
+    monkeypatch.setenv("LLMLB_ADMIN_USER", "root")
+    monkeypatch.setenv("LLMLB_ADMIN_PASSWORD", "rootpw123")
+    app = create_app(db_path=db, jwt_secret="testsecret",
+                     admin_token="admtok")
+    return TestClient(app)
+
+
+def _login(client, username="root", password="rootpw123"):
+    r = client.post("/api/auth/login",
+                    json={"username": username, "password": password})
+    assert r.status_code == 200, r.text
+    return {"Authorization": f"Bearer {r.json()['token']}"}
+
+
+def test_jwt_roundtrip():
+    tok = jwt_encode({"sub": 1, "role": "admin",
+                      "exp": time.time() + 60}, "s")
+    assert jwt_decode(tok, "s")["role"] == "admin"
+    assert jwt_decode(tok, "wrong") is None
+    expired = jwt_encode({"exp": time.time() - 10}, "s")
+    assert jwt_decode(expired, "s") is None
+
+
+def test_password_hashing():
+    h = hash_password("secret")
+    assert verify_password("secret", h)
+    assert not verify_password("other", h)
+
+
+def test_login_and_me(app_client):
+    hdrs = _login(app_client)
+    r = app_client.get("/api/auth/me", headers=hdrs)
+    assert r.json()["role"] == "admin"
+
+
+def test_login_rejects_bad_password(app_client):
+    r = app_client.post("/api/auth/login",
+                        json={"username": "root", "password": "nope"})
+    assert r.status_code == 401
+
+
+def test_user_crud_requires_admin(app_client):
+    hdrs = _login(app_client)
+    r = app_client.post("/api/users", headers=hdrs,
+                        json={"username": "alice", "password": "pw12345",
+                              "role": "user"})
+    assert r.status_code == 201
+    # alice (non-admin) cannot create users
+    alice = _login(app_client, "alice", "pw12345")
+    r = app_client.post("/api/users", headers=alice,
+                        json={"username": "bob", "password": "x"})
+    assert r.status_code == 403
+    r = app_client.get("/api/users", headers=hdrs)
+    assert {u["username"] for u in r.json()} >= {"root", "alice"}
+
+
+def test_invitation_flow(app_client):
+    hdrs = _login(app_client)
+    r = app_client.post("/api/invitations", headers=hdrs,
+                        json={"role": "user"})
+    token = r.json()["token"]
+    r = app_client.post("/api/auth/register",
+                        json={"token": token, "username": "invitee",
+                              "password": "pw12345"})
+    assert r.status_code == 201
+    # token single-use
+    r = app_client.post("/api/auth/register",
+                        json={"token": token, "username": "x2",
+                              "password": "pw"})
+    assert r.status_code == 400
+    assert _login(app_client, "invitee", "pw12345")
+
+
+def test_change_password(app_client):
+    hdrs = _login(app_client)
+    app_client.post("/api/users", headers=hdrs,
+                    json={"username": "carol", "password": "old12345"})
+    carol = _login(app_client, "carol", "old12345")
+    r = app_client.post("/api/auth/change-password", headers=carol,
+                        json={"old_password": "old12345",
+                              "new_password": "new12345"})
+    assert r.status_code == 200
+    assert _login(app_client, "carol", "new12345")
+
+
+def test_admin_token_accepted(app_client):
+    r = app_client.get("/api/users",
+                       headers={"Authorization": "Bearer admtok"})
+    assert r.status_code == 200
+
+
+def test_catalog_search(app_client):
+    r = app_client.get("/api/catalog/search?q=llama",
+                       headers={"Authorization": "Bearer admtok"})
+    assert r.status_code == 200
+    assert any("llama" in m["id"].lower() for m in r.json())
+
+
+def test_token_stats_empty_ok(app_client):
+    r = app_client.get("/api/dashboard/token-stats",
+                       headers={"Authorization": "Bearer admtok"})
+    assert r.status_code == 200

@@ -53,10 +53,16 @@ def test_gpu_matches_cpu_reference_logits(gpu_engine):
                            device=eng.device)
         h = eng.model.embed_tokens(ids)
         residual = None
+        T = len(prompt)
         for li in range(eng.cfg.model.layers):
             q, k, v, residual = eng.model.layer_attn_prelude(
                 li, h, residual, pos)
-            attn = eng._sdpa_causal(q, k, v)
+            qh = q.transpose(0, 1).unsqueeze(0)
+            kh = k.transpose(0, 1).unsqueeze(0)
+            vh = v.transpose(0, 1).unsqueeze(0)
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qh, kh, vh, is_causal=T > 1, enable_gqa=True)
+            attn = o.squeeze(0).transpose(0, 1).contiguous()
             h, residual = eng.model.layer_attn_epilogue(li, attn, residual)
         return eng.model.final(h[-1:], residual[-1:]).float().cpu()[0]
 
