@@ -294,32 +294,32 @@ void ClientConn::handle_readable() {
 }
 
 void ClientConn::process_requests() {
-  // one request at a time; further pipelined bytes wait in `in`
+  // one request at a time; further pipelined bytes wait in `in`.
+  // NOTE: the parsed head holds string_views INTO `in`; any later read
+  // may reallocate the buffer, so the head is re-parsed on every call
+  // and views never outlive this function (parse cost ~100 ns).
   if (op || response_open) return;
   while (!closed) {
-    if (!have_head) {
-      int rc = parse_request_head(in.view(), req);
-      if (rc == 0) return;
-      if (rc < 0) {
-        send_simple(400, "text/plain", "bad request");
-        want_close = true;
-        flush_out();
-        return;
-      }
-      have_head = true;
-      if (req.chunked) {
-        send_json_error(411, "length_required",
-                        "chunked request bodies are not supported");
-        want_close = true;
-        flush_out();
-        return;
-      }
-      if ((size_t)req.content_length > w->gw->cfg.body_limit) {
-        send_json_error(413, "payload_too_large", "body exceeds limit");
-        want_close = true;
-        flush_out();
-        return;
-      }
+    int rc = parse_request_head(in.view(), req);
+    if (rc == 0) return;
+    if (rc < 0) {
+      send_simple(400, "text/plain", "bad request");
+      want_close = true;
+      flush_out();
+      return;
+    }
+    if (req.chunked) {
+      send_json_error(411, "length_required",
+                      "chunked request bodies are not supported");
+      want_close = true;
+      flush_out();
+      return;
+    }
+    if ((size_t)req.content_length > w->gw->cfg.body_limit) {
+      send_json_error(413, "payload_too_large", "body exceeds limit");
+      want_close = true;
+      flush_out();
+      return;
     }
     size_t need = req.header_len + (size_t)req.content_length;
     if (in.size() < need) return;
@@ -337,7 +337,6 @@ void ClientConn::dispatch() {
 void ClientConn::finish_request_cycle() {
   size_t need = req.header_len + (size_t)req.content_length;
   in.consume(need);
-  have_head = false;
   req.headers.clear();  // keep vector capacity across requests
   req.header_len = 0;
   req.content_length = 0;

@@ -97,7 +97,8 @@ struct ProxyOp {
   RequestLease lease;
   BodyReader body_reader;
   SseTap tap;
-  HttpResponse resp;
+  HttpResponse resp;          // views valid only at head-parse time
+  std::string resp_content_type;  // copied at head time (views dangle)
   bool resp_head_seen = false;
   bool streaming = false;
   bool lease_released_at_head = false;

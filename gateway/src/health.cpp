@@ -120,7 +120,10 @@ MiniHttpResult mini_http_request(const std::string& host, int port,
   }
   close(fd);
   (void)eof;
-  if (!have_head && parse_response_head(raw, resp) != 1) {
+  // re-parse now that `raw` is stable (views taken during the read
+  // loop dangle whenever the string reallocated)
+  resp = HttpResponse{};
+  if (parse_response_head(raw, resp) != 1) {
     res.error = "bad response";
     return res;
   }

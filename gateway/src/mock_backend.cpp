@@ -133,23 +133,20 @@ void MockConn::on_event(uint32_t ev) {
     }
     if ((size_t)n < 32768) break;
   }
-  // serve all complete pipelined requests
+  // serve all complete pipelined requests (head re-parsed per read:
+  // its views must not outlive a buffer reallocation)
   while (!closed) {
-    if (!have_head) {
-      int rc = parse_request_head(in.view(), req);
-      if (rc == 0) return;
-      if (rc < 0) {
-        close_now();
-        return;
-      }
-      have_head = true;
+    int rc = parse_request_head(in.view(), req);
+    if (rc == 0) return;
+    if (rc < 0) {
+      close_now();
+      return;
     }
     size_t need = req.header_len + (size_t)req.content_length;
     if (in.size() < need) return;
     bool keep = req.keep_alive;
     respond();
     in.consume(need);
-    have_head = false;
     if (!keep) {
       // honor Connection: close once the response is flushed
       if (out.empty()) close_now();

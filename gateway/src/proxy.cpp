@@ -169,6 +169,9 @@ void Gateway::handle_request(ClientConn* c) {
     handle_local(c, path, body);
     return;
   }
+  LOG_WARN("unrouted request: method='%.*s' path='%s' (target='%.*s')",
+           (int)r.method.size(), r.method.data(), path.c_str(),
+           (int)r.target.size(), r.target.data());
   c->send_json_error(404, "not_found", "no such route");
 }
 
@@ -472,7 +475,8 @@ void proxy_on_upstream_data(UpstreamConn* uc) {
     }
     op->resp_head_seen = true;
     op->body_reader.init(op->resp);
-    std::string_view ctype = op->resp.header("content-type");
+    op->resp_content_type = std::string(op->resp.header("content-type"));
+    std::string_view ctype = op->resp_content_type;
     bool sse = ctype.find("text/event-stream") != std::string_view::npos;
     op->streaming = op->resp.status / 100 == 2 && sse;
     if (op->streaming) {
@@ -573,7 +577,7 @@ static void respond_buffered(Gateway* gw, ClientConn* c) {
     long long pt = raw_int_field(body, "prompt_tokens");
     if (op->kind == ApiKind::kEmbeddings && ct < 0) ct = 0;
     op->lease.complete_success(ct > 0 ? ct : 0);
-    std::string ctype(op->resp.header("content-type"));
+    std::string ctype = op->resp_content_type;
     if (ctype.empty()) ctype = "application/json";
     std::string extra;
     if (op->anthropic_reencode) {
