@@ -454,6 +454,9 @@ class Engine:
                 finished.append(seq)
         for seq in finished:
             self._retire(seq)
+        if self.timing is not None:
+            self.timing["emit"] += time.monotonic() - _t3
+            self.timing["steps"] += 1
 
     def _graph_for(self, S: int) -> "_DecodeGraph | None":
         if not self.use_graphs:

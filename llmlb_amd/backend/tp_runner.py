@@ -1,0 +1,154 @@
+"""Tensor-parallel backend replica: N ranks, one engine shard each.
+
+Launch (inside an xGMI-adjacent quad chosen by parallel.topology):
+
+  HIP_VISIBLE_DEVICES=0,1,2,3 torchrun --standalone --nnodes=1 \
+      --nproc-per-node 4 --local-addr 127.0.0.1 \
+      -m llmlb_amd.backend.tp_runner --model llama-3-70b --port 9400
+
+Scheduling model: every rank runs an IDENTICAL engine replica of the
+scheduler state; rank 0 owns the HTTP server and broadcasts newly
+submitted requests before each step (control over gloo), so all ranks
+admit the same sequences in the same order and the per-layer tensor
+collectives (RCCL all-reduce in RowParallelLinear) line up.  Sampling
+is deterministic given identical full logits (lm_head is replicated),
+so every rank steps the same token stream without further sync.
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import threading
+import time
+
+import torch
+import torch.distributed as dist
+
+from ..parallel.tp import TPContext
+from .engine import Engine, EngineConfig
+from .launcher import PRESETS
+from .sampler import SamplingParams
+
+
+class TPEngineCoordinator:
+    """Rank-0 facade: queues submissions, broadcasts them each step."""
+
+    def __init__(self, engine: Engine, group_cpu):
+        self.engine = engine
+        self.group_cpu = group_cpu
+        self.rank = dist.get_rank()
+        self._pending: list[tuple] = []
+        self._sinks: dict[int, object] = {}
+        self._lock = threading.Lock()
+        self._next_rid = 0
+        self._stop = False
+
+    # HTTP-facing submit (rank 0 only)
+    def submit(self, prompt_ids, params: SamplingParams | None = None,
+               sink=None):
+        params = params or SamplingParams()
+        with self._lock:
+            rid = self._next_rid
+            self._next_rid += 1
+            self._pending.append((rid, list(prompt_ids),
+                                  params.temperature, params.top_p,
+                                  params.max_new_tokens, params.seed))
+            if sink is not None:
+                self._sinks[rid] = sink
+        return rid, sink
+
+    def _drain_pending(self) -> list[tuple]:
+        with self._lock:
+            out = self._pending
+            self._pending = []
+            return out
+
+    def loop(self):
+        """Lockstep engine loop on every rank."""
+        while not self._stop:
+            if self.rank == 0:
+                batch = self._drain_pending()
+                obj = [batch]
+            else:
+                obj = [None]
+            dist.broadcast_object_list(obj, src=0, group=self.group_cpu)
+            batch = obj[0]
+            for rid, prompt_ids, temp, top_p, max_new, seed in batch:
+                params = SamplingParams(temperature=temp, top_p=top_p,
+                                        max_new_tokens=max_new,
+                                        seed=seed if seed is not None
+                                        else 12345 + rid)
+                sink = self._sinks.get(rid) if self.rank == 0 else None
+                _, q = self.engine.submit(prompt_ids, params, sink=sink)
+            busy = self.engine.step()
+            if not busy and not batch:
+                # idle: cheap sleep, staying in lockstep (all ranks see
+                # the same empty batch + empty running set)
+                time.sleep(0.002)
+
+
+def main(argv=None):
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama-3-70b")
+    ap.add_argument("--served-name", default=None)
+    ap.add_argument("--host", default="0.0.0.0")
+    ap.add_argument("--port", type=int, default=9400)
+    ap.add_argument("--kv-blocks", type=int, default=None)
+    ap.add_argument("--max-batch", type=int, default=128)
+    args = ap.parse_args(argv)
+
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    have_gpu = torch.cuda.is_available()
+    backend = "nccl" if have_gpu else "gloo"  # nccl IS RCCL on ROCm
+    dist.init_process_group(backend)
+    group_cpu = dist.new_group(backend="gloo")  # control-plane casts
+    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    if have_gpu:
+        torch.cuda.set_device(device)
+
+    tp = TPContext()
+    mcfg = PRESETS[args.model]()
+    ecfg = EngineConfig(
+        model=mcfg, device=device,
+        dtype="bfloat16" if have_gpu else "float32",
+        kv_cache_blocks=args.kv_blocks, max_batch=args.max_batch,
+        tp_ctx=tp)
+    engine = Engine(ecfg)
+    coord = TPEngineCoordinator(engine, group_cpu)
+
+    if rank == 0:
+        from .server import BackendServer, build_asgi_app
+        import uvicorn
+
+        served = args.served_name or mcfg.name
+        server = BackendServer(engine, served)
+        # route submissions through the coordinator so they broadcast
+        server.engine = _SubmitProxy(engine, coord)  # type: ignore
+        app = build_asgi_app(server)
+        cfg = uvicorn.Config(app, host=args.host, port=args.port,
+                             log_level="warning", access_log=False)
+        usrv = uvicorn.Server(cfg)
+        threading.Thread(target=usrv.run, daemon=True).start()
+        print(f"llmlb_amd TP backend: model={served} tp={tp.world_size} "
+              f"port={args.port}", flush=True)
+    coord.loop()
+
+
+class _SubmitProxy:
+    """Engine facade handed to the HTTP server on rank 0: submissions
+    go through the coordinator; stats attrs proxy to the engine."""
+
+    def __init__(self, engine: Engine, coord: TPEngineCoordinator):
+        self._engine = engine
+        self._coord = coord
+
+    def submit(self, prompt_ids, params=None, sink=None):
+        return self._coord.submit(prompt_ids, params, sink)
+
+    def __getattr__(self, name):
+        return getattr(self._engine, name)
+
+
+if __name__ == "__main__":
+    main()
