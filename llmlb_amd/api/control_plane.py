@@ -471,6 +471,165 @@ def create_app(db_path: str | None = None,
         except WebSocketDisconnect:
             return
 
+    # ---------------- request-history CSV export (api/mod.rs:248) ----
+    @app.get("/api/requests/export.csv")
+    def export_requests_csv(hours: int = Query(24, le=720),
+                            user=Depends(require_admin)):
+        from fastapi.responses import PlainTextResponse
+        cutoff = int(time.time() * 1000) - hours * 3600 * 1000
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT ts_ms,endpoint_id,model,api_kind,status,"
+                "duration_ms,prompt_tokens,completion_tokens,client_ip,"
+                "error FROM request_history WHERE ts_ms > ? ORDER BY ts_ms",
+                (cutoff,)).fetchall()
+        lines = ["ts_ms,endpoint_id,model,api_kind,status,duration_ms,"
+                 "prompt_tokens,completion_tokens,client_ip,error"]
+        for r in rows:
+            lines.append(",".join(
+                str(r[k]) if r[k] is not None else ""
+                for k in r.keys()))
+        return PlainTextResponse("\n".join(lines) + "\n",
+                                 media_type="text/csv")
+
+    # ---------------- log tail (api 'logs' component) ----------------
+    @app.get("/api/logs")
+    def tail_logs(lines: int = Query(200, le=5000),
+                  user=Depends(require_admin)):
+        log_dir = os.environ.get("LLMLB_LOG_DIR", "logs")
+        path = os.path.join(log_dir, "llmlb-gw.jsonl")
+        out = []
+        try:
+            with open(path, "rb") as f:
+                f.seek(0, 2)
+                size = f.tell()
+                f.seek(max(0, size - 1 << 20))
+                out = f.read().decode(errors="replace").splitlines()
+        except OSError:
+            pass
+        return {"lines": out[-lines:]}
+
+    # ---------------- model registry metadata (registry::models) -----
+    class ModelRegBody(BaseModel):
+        id: str
+        name: str = ""
+        source: str = ""
+        capabilities: str = "chat"
+        size_bytes: int | None = None
+
+    @app.get("/api/models")
+    def list_registered_models(user=Depends(auth_user)):
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT id,name,source,capabilities,size_bytes,created_at"
+                " FROM models").fetchall()
+        return [dict(r) for r in rows]
+
+    @app.post("/api/models", status_code=201)
+    def register_model(body: ModelRegBody, user=Depends(require_admin)):
+        with db() as conn:
+            conn.execute(
+                "INSERT OR REPLACE INTO models (id,name,source,"
+                "capabilities,size_bytes,created_at) VALUES (?,?,?,?,?,?)",
+                (body.id, body.name or body.id, body.source,
+                 body.capabilities, body.size_bytes,
+                 time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())))
+        return {"registered": body.id}
+
+    @app.delete("/api/models/{model_id:path}")
+    def delete_model(model_id: str, user=Depends(require_admin)):
+        with db() as conn:
+            cur = conn.execute("DELETE FROM models WHERE id=?",
+                               (model_id,))
+        if cur.rowcount == 0:
+            raise HTTPException(404, "not registered")
+        return {"deleted": model_id}
+
+    # ---------------- self-update manager (update/ component) --------
+    # No egress in this environment: updates come from a local artifact
+    # path.  Flow mirrors the reference (update/mod.rs): check ->
+    # drain the data plane -> apply (atomic binary replace with .bak
+    # rollback) -> resume.  Restart is left to the process supervisor.
+    update_state: dict = {"state": "idle", "history": []}
+
+    class UpdateBody(BaseModel):
+        artifact_path: str
+        target_path: str = ""
+
+    @app.get("/api/update/status")
+    def update_status(user=Depends(require_admin)):
+        return update_state
+
+    @app.post("/api/update/apply", status_code=202)
+    async def update_apply(body: UpdateBody, user=Depends(require_admin)):
+        import shutil
+        target = body.target_path or os.path.join(
+            os.path.dirname(os.path.dirname(os.path.dirname(
+                os.path.abspath(__file__)))), "gateway", "bin", "llmlb-gw")
+        if not os.path.exists(body.artifact_path):
+            raise HTTPException(400, "artifact not found")
+        update_state.update({"state": "draining"})
+
+        def drain_and_apply():
+            try:
+                req = urllib.request.Request(
+                    gateway_url + "/api/system/drain", data=b"{}",
+                    headers=_gateway_headers())
+                urllib.request.urlopen(req, timeout=10)
+                # wait for in-flight requests to finish (300 s cap,
+                # update/mod.rs:37)
+                deadline = time.time() + 300
+                while time.time() < deadline:
+                    with urllib.request.urlopen(
+                            gateway_url + "/api/metrics",
+                            timeout=5) as resp:
+                        text = resp.read().decode()
+                    for line in text.splitlines():
+                        if line.startswith("llmlb_active_requests"):
+                            if int(float(line.split()[-1])) == 0:
+                                deadline = 0
+                    if deadline == 0:
+                        break
+                    time.sleep(1)
+                update_state.update({"state": "applying"})
+                bak = target + ".bak"
+                if os.path.exists(target):
+                    shutil.copy2(target, bak)
+                shutil.copy2(body.artifact_path, target)
+                os.chmod(target, 0o755)
+                update_state.update({"state": "applied",
+                                     "rollback": bak})
+            except Exception as e:
+                update_state.update({"state": "failed", "error": str(e)})
+            finally:
+                try:
+                    req = urllib.request.Request(
+                        gateway_url + "/api/system/resume", data=b"{}",
+                        headers=_gateway_headers())
+                    urllib.request.urlopen(req, timeout=10)
+                except Exception:
+                    pass
+                update_state["history"] = (
+                    update_state.get("history", [])
+                    + [{"ts": time.time(),
+                        "state": update_state["state"]}])[-20:]
+
+        import threading as _threading
+        _threading.Thread(target=drain_and_apply, daemon=True).start()
+        return {"accepted": True}
+
+    @app.post("/api/update/rollback")
+    def update_rollback(user=Depends(require_admin)):
+        import shutil
+        bak = update_state.get("rollback")
+        if not bak or not os.path.exists(bak):
+            raise HTTPException(400, "no rollback artifact")
+        target = bak[:-4]
+        shutil.copy2(bak, target)
+        os.chmod(target, 0o755)
+        update_state.update({"state": "rolled_back"})
+        return {"rolled_back": True}
+
     @app.get("/api/control/health")
     def control_health():
         return {"status": "ok", "component": "control-plane"}
