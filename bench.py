@@ -32,6 +32,7 @@ from pathlib import Path
 REPO = Path(__file__).resolve().parent
 sys.path.insert(0, str(REPO))
 
+os.environ.setdefault("TORCH_BLAS_PREFER_HIPBLASLT", "1")
 import torch  # noqa: E402
 
 
