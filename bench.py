@@ -73,6 +73,7 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
     tokens = 0
     errors = 0
     latencies: list[float] = []
+    error_samples: list[str] = []
 
     async def one(session):
         nonlocal tokens, errors
@@ -82,7 +83,10 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
                                     json=body, headers=headers) as resp:
                 if resp.status != 200:
                     errors += 1
-                    await resp.read()
+                    detail = (await resp.read())[:200]
+                    if len(error_samples) < 3:
+                        error_samples.append(
+                            f"HTTP {resp.status}: {detail!r}")
                     return
                 usage_tokens = None
                 events = 0
@@ -103,8 +107,10 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
                             pass
                 tokens += usage_tokens if usage_tokens is not None \
                     else max(events - 2, 0)
-        except Exception:
+        except Exception as e:
             errors += 1
+            if len(error_samples) < 3:
+                error_samples.append(f"{type(e).__name__}: {e}")
         finally:
             latencies.append(time.monotonic() - t0)
 
@@ -113,6 +119,9 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
     async with aiohttp.ClientSession(connector=connector,
                                      timeout=timeout) as session:
         await asyncio.gather(*[one(session) for _ in range(conc)])
+    if error_samples:
+        print(f"[bench] request errors: {error_samples}",
+              file=sys.stderr, flush=True)
     return tokens, errors, latencies
 
 
@@ -188,15 +197,23 @@ def main() -> None:
     conc = args.conc or int(os.environ.get("LLMLB_BENCH_CONC", "64"))
     device = f"cuda:{local_rank}" if have_gpu else "cpu"
 
+    def log(msg):
+        print(f"[bench rank{rank}] {msg}", file=sys.stderr, flush=True)
+
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
         dist.init_process_group("gloo", rank=rank, world_size=world)
+        log("process group ready")
 
     def barrier():
+        # async_op + poll: a blocking gloo barrier holds the GIL and
+        # freezes this process's serving threads (uvicorn + engine)
         if dist is not None:
-            dist.barrier()
+            work = dist.barrier(async_op=True)
+            while not work.is_completed():
+                time.sleep(0.005)
 
     def sync():
         if have_gpu:
@@ -218,9 +235,11 @@ def main() -> None:
         per_block = KVCache.bytes_per_block(
             m.layers, m.kv_heads, m.head_dim)
         kv_blocks = max(int(kv_gb * (1 << 30)) // per_block, 16)
+    from llmlb_amd.parallel.tp import TPContext
     engine = build_engine(model_name, device,
                           kv_blocks=kv_blocks,
-                          max_batch=max(conc * 2, 128))
+                          max_batch=max(conc * 2, 128),
+                          tp_ctx=TPContext.solo())
     backend_port = 9300 + rank
     served = BackendServer(engine, model_name)
     app = build_asgi_app(served)
@@ -232,6 +251,7 @@ def main() -> None:
     deadline = time.time() + 60
     while time.time() < deadline and not usrv.started:
         time.sleep(0.1)
+    log(f"backend up on :{backend_port} (started={usrv.started})")
 
     barrier()  # all backends up
 
@@ -247,11 +267,13 @@ def main() -> None:
                         "LLMLB_DATABASE_URL": db,
                         "LLMLB_ADMIN_TOKEN": "bench",
                         "LLMLB_HEALTH_CHECK_INTERVAL": "30"})
+            gw_log = open(os.environ.get("LLMLB_BENCH_GW_LOG",
+                                         os.devnull), "w")
+            log(f"gateway port {gw_port}")
             gw_proc = subprocess.Popen(
                 [str(bin_dir / "llmlb-gw"), "--workers",
                  str(max(2, min(8, (os.cpu_count() or 8) // 2)))],
-                env=env, stdout=subprocess.DEVNULL,
-                stderr=subprocess.DEVNULL)
+                env=env, stdout=gw_log, stderr=gw_log)
             time.sleep(0.6)
             import urllib.request
             for r in range(n_gpus):
@@ -288,8 +310,10 @@ def main() -> None:
                                             args.prompt_len, args.gen_len,
                                             "bench"))
 
+            log("gateway + endpoints ready; warmup")
             for _ in range(args.warmup):
                 wave()
+            log("warmup done")
         else:
             pass  # other ranks: engines serve during rank 0's waves
 
@@ -302,11 +326,12 @@ def main() -> None:
         total_errors = 0
         lat_all: list = []
         if rank == 0:
-            for _ in range(args.steps):
+            for si in range(args.steps):
                 tok, err, lats = asyncio.run(
                     run_wave(f"http://127.0.0.1:{gw_port}", model_name,
                              conc * n_gpus, args.prompt_len, args.gen_len,
                              "bench"))
+                log(f"step {si}: tokens={tok} errors={err}")
                 total_tokens += tok
                 total_errors += err
                 lat_all.extend(lats)
@@ -317,7 +342,9 @@ def main() -> None:
         # max over ranks
         if dist is not None:
             t = torch.tensor([elapsed], dtype=torch.float64)
-            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            work = dist.all_reduce(t, op=dist.ReduceOp.MAX, async_op=True)
+            while not work.is_completed():
+                time.sleep(0.005)
             elapsed = float(t[0])
 
         if rank == 0:
@@ -362,6 +389,13 @@ def main() -> None:
             }
             print(json.dumps(result), flush=True)
     finally:
+        log(f"engine: steps={engine.steps_done} tokens_out="
+            f"{engine.tokens_out} running={engine.num_running} "
+            f"waiting={engine.waiting.qsize()} tp={engine.model.tp.world_size}")
+        hold = float(os.environ.get("LLMLB_BENCH_HOLD", "0"))
+        if hold and rank == 0:
+            log(f"holding stack for {hold}s (LLMLB_BENCH_HOLD)")
+            time.sleep(hold)
         if gw_proc is not None:
             gw_proc.terminate()
         engine.stop()

@@ -29,6 +29,8 @@ PRESETS = {
 def build_engine(model: str, device: str, kv_blocks: int | None = None,
                  max_batch: int = 256, tp_ctx=None,
                  hbm_fraction: float = 0.85) -> Engine:
+    """tp_ctx=None means 'infer from torch.distributed' — pass
+    TPContext.solo() for DP replicas inside a larger world."""
     cfg_fn = PRESETS.get(model)
     if cfg_fn is None:
         raise SystemExit(f"unknown model preset '{model}' "

@@ -20,6 +20,17 @@ import torch.nn as nn
 class TPContext:
     """Process-group wrapper; world_size==1 means no TP (no comm)."""
 
+    @classmethod
+    def solo(cls) -> "TPContext":
+        """A no-comm context for DP replicas living inside a larger
+        torch.distributed world (e.g. bench.py ranks): without it, a
+        replica would mistake the global world for its TP group."""
+        ctx = cls.__new__(cls)
+        ctx.group = None
+        ctx.world_size = 1
+        ctx.rank = 0
+        return ctx
+
     def __init__(self, group=None):
         if group is not None:
             self.group = group
