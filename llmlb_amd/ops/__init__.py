@@ -43,6 +43,7 @@ def _try_load() -> None:
     _lib.ll_swiglu.restype = ctypes.c_int
     _lib.ll_kv_append.restype = ctypes.c_int
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
+    _lib.ll_decode_gemm.restype = ctypes.c_int
 
 
 _try_load()
@@ -158,6 +159,27 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
         ctypes.c_int(k.shape[0]), ctypes.c_int(k.shape[1]),
         ctypes.c_int(k.shape[2]), ctypes.c_int(block_size), _stream())
     _check(rc, "ll_kv_append")
+
+
+def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """out = x @ W^T via the weight-streaming decode GEMM when the
+    shape qualifies (bf16, N<=128, K%256==0, M%16==0); falls back to
+    torch.nn.functional.linear (hipBLASLt) otherwise."""
+    if (not x.is_cuda or _lib is None or x.dtype != torch.bfloat16
+            or x.dim() != 2):
+        return torch.nn.functional.linear(x, weight)
+    N, K = x.shape
+    M = weight.shape[0]
+    if N > 128 or K % 256 != 0 or M % 16 != 0:
+        return torch.nn.functional.linear(x, weight)
+    out = torch.empty(N, M, dtype=x.dtype, device=x.device)
+    rc = _lib.ll_decode_gemm(_ptr(out), _ptr(x), _ptr(weight),
+                             ctypes.c_int(N), ctypes.c_int(M),
+                             ctypes.c_int(K), _stream())
+    if rc == -2:
+        return torch.nn.functional.linear(x, weight)
+    _check(rc, "ll_decode_gemm")
+    return out
 
 
 def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,

@@ -16,6 +16,8 @@ import torch
 import torch.distributed as dist
 import torch.nn as nn
 
+from ..ops import linear_decode
+
 
 class TPContext:
     """Process-group wrapper; world_size==1 means no TP (no comm)."""
@@ -83,6 +85,8 @@ class ColumnParallelLinear(nn.Module):
             requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
+            return linear_decode(x, self.weight)
         return torch.nn.functional.linear(x, self.weight)
 
 
@@ -100,5 +104,8 @@ class RowParallelLinear(nn.Module):
             requires_grad=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = torch.nn.functional.linear(x, self.weight)
+        if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
+            y = linear_decode(x, self.weight)
+        else:
+            y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)
