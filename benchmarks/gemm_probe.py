@@ -1,0 +1,70 @@
+"""Decode-GEMM microbench: custom weight-streaming kernel vs hipBLASLt
+(torch.nn.functional.linear) on the llama-3-8b decode shapes.
+
+Within-probe interleaved A/B (guide §5.4 rule 24): N rounds alternating
+both paths in one process, report median ms and effective weight TB/s.
+"""
+import json
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
+import torch
+
+import llmlb_amd.ops as ops
+
+SHAPES = [  # (N, M, K, label)
+    (64, 6144, 4096, "qkv"),
+    (64, 4096, 4096, "o_proj"),
+    (64, 28672, 4096, "gate_up"),
+    (64, 4096, 14336, "down"),
+    (128, 28672, 4096, "gate_up_n128"),
+    (16, 28672, 4096, "gate_up_n16"),
+]
+ROUNDS = 30
+
+
+def bench_one(fn, x, w, rounds=ROUNDS):
+    for _ in range(5):
+        fn(x, w)
+    torch.cuda.synchronize()
+    times = []
+    for _ in range(rounds):
+        t0 = time.perf_counter()
+        fn(x, w)
+        torch.cuda.synchronize()
+        times.append(time.perf_counter() - t0)
+    times.sort()
+    return times[len(times) // 2]
+
+
+def main():
+    assert torch.cuda.is_available() and ops.native_available(), \
+        ops.native_load_error()
+    dev = "cuda:0"
+    out = {}
+    for N, M, K, label in SHAPES:
+        x = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.3
+        w = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
+        # interleaved A/B
+        t_custom = bench_one(ops.linear_decode, x, w)
+        t_blas = bench_one(torch.nn.functional.linear, x, w)
+        t_custom2 = bench_one(ops.linear_decode, x, w)
+        t_custom = min(t_custom, t_custom2)
+        wbytes = M * K * 2
+        out[label] = {
+            "shape": f"N{N} M{M} K{K}",
+            "custom_us": round(t_custom * 1e6, 1),
+            "hipblaslt_us": round(t_blas * 1e6, 1),
+            "custom_tbps": round(wbytes / t_custom / 1e12, 2),
+            "hipblaslt_tbps": round(wbytes / t_blas / 1e12, 2),
+            "speedup": round(t_blas / t_custom, 2),
+        }
+        print(json.dumps({label: out[label]}), flush=True)
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
