@@ -421,6 +421,25 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }
+    if (path == "/api/events" && method == "GET") {
+      int64_t since = 0;
+      std::string q(r.query);
+      size_t p = q.find("since=");
+      if (p != std::string::npos)
+        since = atoll(q.c_str() + p + 6);
+      JsonArray arr;
+      for (auto& e : events.since(since)) {
+        JsonObject o;
+        o.emplace_back("ts_ms", Json((long long)e.ts_ms));
+        o.emplace_back("type", Json(e.type));
+        Json detail;
+        if (Json::parse(e.detail, detail))
+          o.emplace_back("detail", std::move(detail));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
     if (path == "/api/gpu" && method == "GET") {
       std::string out;
       telemetry_snapshot_json(&out);

@@ -3,6 +3,7 @@
 #pragma once
 
 #include <atomic>
+#include <deque>
 #include <memory>
 #include <thread>
 #include <unordered_map>
@@ -55,6 +56,30 @@ class UpstreamConn;
 
 // metrics counters (Prometheus text at /api/metrics — cloud_metrics.rs
 // analog, extended with gateway totals)
+// broadcast-bus analog (events/mod.rs): bounded ring of recent gateway
+// events, polled by /api/events and pushed onward by the control plane
+struct EventRing {
+  struct Event {
+    int64_t ts_ms;
+    std::string type;     // EndpointStatusChanged | TpsUpdated | ...
+    std::string detail;   // small JSON fragment
+  };
+  std::mutex mu;
+  std::deque<Event> ring;
+  void emit(std::string type, std::string detail) {
+    std::lock_guard<std::mutex> lk(mu);
+    ring.push_back({wall_ms(), std::move(type), std::move(detail)});
+    while (ring.size() > 1024) ring.pop_front();  // cap (events/mod.rs:13)
+  }
+  std::vector<Event> since(int64_t ts_ms) {
+    std::lock_guard<std::mutex> lk(mu);
+    std::vector<Event> out;
+    for (auto& e : ring)
+      if (e.ts_ms > ts_ms) out.push_back(e);
+    return out;
+  }
+};
+
 struct Metrics {
   std::atomic<long long> requests_total{0};
   std::atomic<long long> responses_2xx{0};
@@ -240,6 +265,7 @@ class Gateway {
   HistoryWriter history;
   AuthStore auth;
   Metrics metrics;
+  EventRing events;
   std::atomic<bool> draining{false};   // inference gate (self-update)
   std::atomic<long long> start_ms{0};
 

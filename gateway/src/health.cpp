@@ -174,6 +174,34 @@ std::string Gateway::detect_endpoint_type(const std::string& host, int port,
   return "";  // unreachable / unsupported
 }
 
+// ---------------------------------------------------------------- mapping
+// Static canonical (HF repo id) <-> engine alias table — the analog of
+// the reference's BUILTIN_MAPPINGS (models/mapping.rs:44): any alias a
+// backend reports resolves to the canonical id so every spelling routes.
+struct AliasRow {
+  const char* canonical;
+  const char* alias;
+};
+static const AliasRow kBuiltinMappings[] = {
+    {"meta-llama/Meta-Llama-3-8B-Instruct", "llama3:8b"},
+    {"meta-llama/Meta-Llama-3-8B-Instruct", "llama-3-8b"},
+    {"meta-llama/Meta-Llama-3-8B-Instruct", "Meta-Llama-3-8B-Instruct"},
+    {"meta-llama/Meta-Llama-3-70B-Instruct", "llama3:70b"},
+    {"meta-llama/Meta-Llama-3-70B-Instruct", "llama-3-70b"},
+    {"openai/gpt-oss-20b", "gpt-oss:20b"},
+    {"openai/whisper-large-v3", "whisper-large-v3"},
+    {"openai/whisper-large-v3", "whisper-large-v3-synth"},
+    {"BAAI/bge-large-en-v1.5", "bge-large"},
+    {"mistralai/Mistral-7B-Instruct-v0.3", "mistral:7b"},
+    {"Qwen/Qwen2.5-7B-Instruct", "qwen2.5:7b"},
+};
+
+std::string resolve_canonical_alias(const std::string& model_id) {
+  for (const auto& row : kBuiltinMappings)
+    if (model_id == row.alias) return row.canonical;
+  return "";
+}
+
 // ---------------------------------------------------------------- sync
 // GET /v1/models (OpenAI or Ollama shape) -> model list diff
 // (reference sync/mod.rs:104, sync/parser.rs)
@@ -233,6 +261,8 @@ bool Gateway::sync_endpoint_models(const EndpointPtr& ep, std::string* err) {
       if (mt) em.max_tokens = mt->as_int(0);
       if (const Json* can = m.find("canonical_id"); can)
         em.canonical = can->as_str();
+      if (em.canonical.empty())
+        em.canonical = resolve_canonical_alias(em.id);
       if (em.canonical.empty()) em.canonical = em.id;
       models.push_back(std::move(em));
     }
@@ -275,6 +305,9 @@ void Gateway::health_loop() {
           std::string serr;
           sync_endpoint_models(ep, &serr);
           registry.set_status(ep->id, EndpointStatus::kOnline);
+          events.emit("EndpointStatusChanged",
+                      "{\"endpoint_id\":\"" + ep->id +
+                      "\",\"status\":\"online\"}");
           LOG_INFO("endpoint %s (%s) online", ep->name.c_str(),
                    ep->base_url.c_str());
         } else {
@@ -307,6 +340,9 @@ void Gateway::health_loop() {
                    fails >= cfg.consecutive_failures_for_offline) {
           registry.set_status(ep->id, EndpointStatus::kOffline);
           registry.clear_tps_for_endpoint(ep->id);  // eject + TPS reset
+          events.emit("EndpointStatusChanged",
+                      "{\"endpoint_id\":\"" + ep->id +
+                      "\",\"status\":\"offline\"}");
           LOG_WARN("endpoint %s ejected (offline)", ep->name.c_str());
         }
       }

@@ -671,6 +671,16 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
   rec.completion_tokens = completion_tokens;
   rec.client_ip = std::move(op->client_ip);
   rec.error = error;
+  // TpsUpdated event, throttled to ~1/s (events/mod.rs TpsUpdated)
+  if (success && completion_tokens > 0) {
+    static std::atomic<long long> last_emit{0};
+    long long now = rec.ts_ms, prev = last_emit.load();
+    if (now - prev > 1000 && last_emit.compare_exchange_weak(prev, now)) {
+      events.emit("TpsUpdated",
+                  "{\"endpoint_id\":\"" + rec.endpoint_id +
+                  "\",\"model\":\"" + rec.model + "\"}");
+    }
+  }
   c->w->pending_records.push_back(std::move(rec));
   if (c->w->pending_records.size() >= 4096)
     history.record_batch(c->w->pending_records);
