@@ -13,7 +13,10 @@ sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
 import torch
 
+import os
+os.environ["LLMLB_CUSTOM_DECODE_GEMM"] = "1"
 import llmlb_amd.ops as ops
+ops._USE_CUSTOM_DECODE_GEMM = True
 
 SHAPES = [  # (N, M, K, label)
     (64, 6144, 4096, "qkv"),

@@ -161,12 +161,21 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
     _check(rc, "ll_kv_append")
 
 
+_USE_CUSTOM_DECODE_GEMM = os.environ.get(
+    "LLMLB_CUSTOM_DECODE_GEMM", "0").lower() in ("1", "true", "on")
+
+
 def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
-    """out = x @ W^T via the weight-streaming decode GEMM when the
-    shape qualifies (bf16, N<=128, K%256==0, M%16==0); falls back to
-    torch.nn.functional.linear (hipBLASLt) otherwise."""
-    if (not x.is_cuda or _lib is None or x.dtype != torch.bfloat16
-            or x.dim() != 2):
+    """out = x @ W^T.
+
+    The in-tree weight-streaming kernel (csrc/decode_gemm.hip) is
+    numerically verified but measured SLOWER than hipBLASLt on the
+    llama-3-8b decode shapes (profiles/r01_decode_gemm_analysis.md —
+    the skinny-N regime is per-CU load-path/latency-bound, and
+    hipBLASLt's tiles already sit near that wall), so it is opt-in via
+    LLMLB_CUSTOM_DECODE_GEMM=1 pending the glds+swizzle rewrite."""
+    if (not _USE_CUSTOM_DECODE_GEMM or not x.is_cuda or _lib is None
+            or x.dtype != torch.bfloat16 or x.dim() != 2):
         return torch.nn.functional.linear(x, weight)
     N, K = x.shape
     M = weight.shape[0]
