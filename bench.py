@@ -157,17 +157,20 @@ def measure_proxy_overhead(bin_dir: Path) -> dict:
                 break
             except Exception:
                 time.sleep(0.5)
-        res = subprocess.run(
-            [str(bin_dir / "llmlb-bench"), "--port", str(gw_port),
-             "--threads", "2", "--connections", "50", "--duration", "3",
-             "--json"], capture_output=True, text=True, timeout=60)
-        gw.terminate()
-        mock.terminate()
-        line = res.stdout.strip().splitlines()[-1]
-        data = json.loads(line)
-        if data.get("non2xx", 1) == 0 and data.get("requests", 0) > 0:
-            out["proxy_rps"] = data["rps"]
-            out["p50_proxy_overhead_ms"] = data["p50_ms"]
+        try:
+            res = subprocess.run(
+                [str(bin_dir / "llmlb-bench"), "--port", str(gw_port),
+                 "--threads", "2", "--connections", "50", "--duration",
+                 "3", "--json"], capture_output=True, text=True,
+                timeout=60)
+            line = res.stdout.strip().splitlines()[-1]
+            data = json.loads(line)
+            if data.get("non2xx", 1) == 0 and data.get("requests", 0) > 0:
+                out["proxy_rps"] = data["rps"]
+                out["p50_proxy_overhead_ms"] = data["p50_ms"]
+        finally:
+            gw.terminate()
+            mock.terminate()
     except Exception as e:  # pragma: no cover
         print(f"proxy-overhead micro-bench skipped: {e}", file=sys.stderr)
     return out

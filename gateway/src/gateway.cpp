@@ -373,6 +373,19 @@ void ClientConn::flush_out() {
 void ClientConn::send_simple(int status, std::string_view content_type,
                              std::string_view body,
                              std::string_view extra_headers) {
+  // every-request audit trail (excl. health/metrics/static —
+  // audit/middleware.rs:23-45), buffered per worker
+  if (!req.path.empty() && req.path != "/health" &&
+      req.path != "/api/health" && req.path != "/metrics" &&
+      req.path != "/api/metrics" && !starts_with(req.path, "/dashboard")) {
+    AuditRecord a;
+    a.ts_ms = wall_ms();
+    a.method = std::string(req.method);
+    a.path = std::string(req.path);
+    a.status = status;
+    a.client_ip = peer_ip;
+    w->pending_audits.push_back(std::move(a));
+  }
   append_status_line(out, status, status_reason(status));
   char hdr[256];
   int n = snprintf(hdr, sizeof hdr,
@@ -426,6 +439,7 @@ void Worker::Acceptor::on_event(uint32_t) {
 void Worker::arm_record_flush() {
   loop.add_timer(200, [this] {
     if (!pending_records.empty()) gw->history.record_batch(pending_records);
+    if (!pending_audits.empty()) gw->history.audit_batch(pending_audits);
     arm_record_flush();
   });
 }

@@ -35,6 +35,10 @@ struct GatewayConfig {
   int auto_sync_interval_s = 900;
   int consecutive_failures_for_offline = 2;
   bool require_auth = true;
+  // persist request/response bodies (truncated to 16 KiB) in
+  // request_history — the reference stores sanitized bodies
+  // (api/openai_util.rs:137); off by default for throughput
+  bool capture_bodies = false;
   std::string admin_token;          // LLMLB_ADMIN_TOKEN
   std::string log_level = "info";
   std::string log_dir;
@@ -137,6 +141,7 @@ struct ProxyOp {
   ApiKind kind = ApiKind::kChat;
   std::string client_ip;
   long long req_prompt_chars = 0;
+  std::string captured_request;   // when cfg.capture_bodies
   bool is_cloud = false;
   std::string cloud_provider;
   // buffered (non-stream) response body (decoded payload)
@@ -236,6 +241,7 @@ class Worker {
   // per-worker record buffer: finalize_op appends lock-free (worker
   // thread), a 200 ms loop timer flushes to the HistoryWriter
   std::vector<RequestRecord> pending_records;
+  std::vector<AuditRecord> pending_audits;
   void arm_record_flush();
   // ops with an armed inference timeout, swept every 250 ms
   std::unordered_set<ClientConn*> timed_ops;

@@ -17,6 +17,8 @@ namespace gw {
 
 struct RequestRecord {
   int64_t ts_ms = 0;
+  std::string request_body;    // captured only when LLMLB_CAPTURE_BODIES
+  std::string response_body;
   std::string endpoint_id;
   std::string model;
   std::string api_kind;
@@ -117,6 +119,16 @@ class HistoryWriter {
     audit_queue_.push_back(std::move(r));
     if (audit_queue_.size() > 100000) audit_queue_.pop_front();
     cv_.notify_one();
+  }
+  void audit_batch(std::vector<AuditRecord>& rs) {
+    if (rs.empty()) return;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      for (auto& r : rs) audit_queue_.push_back(std::move(r));
+      while (audit_queue_.size() > 100000) audit_queue_.pop_front();
+    }
+    cv_.notify_one();
+    rs.clear();
   }
 
   HistoryRing& ring() { return ring_; }

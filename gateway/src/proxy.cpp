@@ -671,6 +671,11 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
   rec.completion_tokens = completion_tokens;
   rec.client_ip = std::move(op->client_ip);
   rec.error = error;
+  if (cfg.capture_bodies) {
+    rec.request_body = std::move(op->captured_request);
+    if (!op->streaming)
+      rec.response_body = op->buffered.substr(0, 16384);
+  }
   // TpsUpdated event, throttled to ~1/s (events/mod.rs TpsUpdated)
   if (success && completion_tokens > 0) {
     static std::atomic<long long> last_emit{0};
