@@ -375,7 +375,7 @@ void ClientConn::send_simple(int status, std::string_view content_type,
                              std::string_view extra_headers) {
   // every-request audit trail (excl. health/metrics/static —
   // audit/middleware.rs:23-45), buffered per worker
-  if (!req.path.empty() && req.path != "/health" &&
+  if (getenv("LLMLB_NO_AUDIT") == nullptr && !req.path.empty() && req.path != "/health" &&
       req.path != "/api/health" && req.path != "/metrics" &&
       req.path != "/api/metrics" && !starts_with(req.path, "/dashboard")) {
     AuditRecord a;
