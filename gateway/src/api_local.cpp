@@ -316,6 +316,16 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
         c->send_simple(ok ? 200 : 502, "application/json", bodyj);
         return;
       }
+      if (action == "chat" && method == "POST") {
+        // playground: chat through THIS endpoint, bypassing selection
+        // (reference api/endpoints.rs:1079 proxy_chat_completions)
+        start_proxy_direct(this, c, ep, ApiKind::kChat,
+                           "/v1/chat/completions",
+                           r.header("content-type").empty()
+                               ? std::string_view("application/json")
+                               : r.header("content-type"));
+        return;
+      }
       if (action == "test" && method == "POST") {
         MiniHttpResult res = mini_http_request(
             ep->host, ep->port, "GET", ep->path_prefix + "/v1/models", "",
