@@ -231,6 +231,8 @@ void SseTap::line(std::string_view l) {
     };
     long long pt = grab("\"prompt_tokens\"");
     long long ct = grab("\"completion_tokens\"");
+    if (pt < 0) pt = grab("\"input_tokens\"");    // Responses shape
+    if (ct < 0) ct = grab("\"output_tokens\"");
     if (pt >= 0) usage_prompt_ = pt;
     if (ct >= 0) usage_completion_ = ct;
   }

@@ -575,6 +575,8 @@ static void respond_buffered(Gateway* gw, ClientConn* c) {
     }
     long long ct = raw_int_field(body, "completion_tokens");
     long long pt = raw_int_field(body, "prompt_tokens");
+    if (ct < 0) ct = raw_int_field(body, "output_tokens");
+    if (pt < 0) pt = raw_int_field(body, "input_tokens");
     if (op->kind == ApiKind::kEmbeddings && ct < 0) ct = 0;
     op->lease.complete_success(ct > 0 ? ct : 0);
     std::string ctype = op->resp_content_type;
@@ -651,11 +653,14 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
       }
     }
   } else if (!op->buffered.empty()) {
-    completion_tokens = std::max(0LL,
-                                 raw_int_field(op->buffered,
-                                               "completion_tokens"));
-    prompt_tokens =
-        std::max(0LL, raw_int_field(op->buffered, "prompt_tokens"));
+    completion_tokens = raw_int_field(op->buffered, "completion_tokens");
+    if (completion_tokens < 0)
+      completion_tokens = raw_int_field(op->buffered, "output_tokens");
+    completion_tokens = std::max(0LL, completion_tokens);
+    prompt_tokens = raw_int_field(op->buffered, "prompt_tokens");
+    if (prompt_tokens < 0)
+      prompt_tokens = raw_int_field(op->buffered, "input_tokens");
+    prompt_tokens = std::max(0LL, prompt_tokens);
   }
   if (!success && op->lease.active()) op->lease.complete_failure();
 

@@ -137,6 +137,119 @@ class BackendServer:
         return rid, out_q, prompt_ids
 
 
+async def handle_responses(server: "BackendServer", body: dict, send):
+    """Open Responses API surface (object=response, output_text items;
+    streaming emits response.output_text.delta / response.completed)."""
+    import uuid as _uuid
+    model = body.get("model", server.model_name)
+    if model not in server.served:
+        data = json.dumps({"error": {"message": "model not served",
+                                     "code": "model_not_found"}}).encode()
+        await send({"type": "http.response.start", "status": 404,
+                    "headers": [(b"content-type", b"application/json")]})
+        await send({"type": "http.response.body", "body": data})
+        return
+    inp = body.get("input", "")
+    if isinstance(inp, list):
+        # list of messages or content parts
+        parts = []
+        for item in inp:
+            if isinstance(item, dict):
+                c = item.get("content", "")
+                if isinstance(c, list):
+                    c = " ".join(p.get("text", "") for p in c
+                                 if isinstance(p, dict))
+                parts.append(str(c))
+            else:
+                parts.append(str(item))
+        inp = "\n".join(parts)
+    prompt_ids = server.tokenizer.encode(str(inp) or "hello")
+    params = server.parse_sampling(body)
+    import asyncio as _asyncio
+
+    class _Sink:
+        def __init__(self):
+            self.loop = _asyncio.get_running_loop()
+            self.q: _asyncio.Queue = _asyncio.Queue()
+
+        def put(self, item):
+            self.loop.call_soon_threadsafe(self.q.put_nowait, item)
+
+    sink = _Sink()
+    server.engine.submit(prompt_ids, params, sink=sink)
+    server.requests_total += 1
+    rid = f"resp_{_uuid.uuid4().hex[:24]}"
+    created = int(time.time())
+    stream = bool(body.get("stream"))
+
+    async def collect():
+        toks = []
+        while True:
+            tok, fin = await sink.q.get()
+            toks.append(tok)
+            if fin:
+                return toks, fin
+
+    if not stream:
+        toks, fin = await collect()
+        text = "".join(server.tokenizer.decode_token(t) for t in toks)
+        payload = {
+            "id": rid, "object": "response", "created_at": created,
+            "model": model, "status": "completed",
+            "output": [{"type": "message", "role": "assistant",
+                        "content": [{"type": "output_text",
+                                     "text": text}]}],
+            "usage": {"input_tokens": len(prompt_ids),
+                      "output_tokens": len(toks),
+                      "total_tokens": len(prompt_ids) + len(toks)}}
+        data = json.dumps(payload).encode()
+        await send({"type": "http.response.start", "status": 200,
+                    "headers": [(b"content-type", b"application/json"),
+                                (b"content-length",
+                                 str(len(data)).encode())]})
+        await send({"type": "http.response.body", "body": data})
+        return
+
+    await send({"type": "http.response.start", "status": 200,
+                "headers": [(b"content-type", b"text/event-stream"),
+                            (b"cache-control", b"no-cache")]})
+
+    def ev(obj):
+        return ("data: " + json.dumps(obj) + "\n\n").encode()
+
+    await send({"type": "http.response.body",
+                "body": ev({"type": "response.created",
+                            "response": {"id": rid,
+                                         "object": "response",
+                                         "model": model,
+                                         "status": "in_progress"}}),
+                "more_body": True})
+    ntok = 0
+    while True:
+        tok, fin = await sink.q.get()
+        ntok += 1
+        await send({"type": "http.response.body",
+                    "body": ev({"type": "response.output_text.delta",
+                                "delta":
+                                    server.tokenizer.decode_token(tok)}),
+                    "more_body": True})
+        if fin:
+            break
+    await send({"type": "http.response.body",
+                "body": ev({"type": "response.completed",
+                            "response": {
+                                "id": rid, "object": "response",
+                                "status": "completed", "model": model,
+                                "usage": {
+                                    "input_tokens": len(prompt_ids),
+                                    "output_tokens": ntok,
+                                    "total_tokens":
+                                        len(prompt_ids) + ntok}}}),
+                "more_body": True})
+    await send({"type": "http.response.body",
+                "body": b"data: [DONE]\n\n", "more_body": False})
+
+
 def build_asgi_app(server: BackendServer):
     """Raw ASGI app (no framework dependency on the hot path)."""
 
@@ -219,6 +332,9 @@ def build_asgi_app(server: BackendServer):
                 await send_json(send, 400, {"error": {
                     "message": "invalid JSON",
                     "type": "invalid_request_error"}})
+                return
+            if path == "/v1/responses":
+                await handle_responses(server, body, send)
                 return
             is_chat = path != "/v1/completions"
             model = body.get("model", server.model_name)

@@ -95,10 +95,12 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
     Returns (out, residual'): when ``residual`` is given it is updated in
     place to x + residual and returned.  x: [T, H] bf16 (GPU path).
     """
-    if not x.is_cuda:
+    if not x.is_cuda or x.dtype != torch.bfloat16:
+        # bf16 is the native serving dtype; fp32 GPU tensors (tests,
+        # debugging) use the reference path
         return reference.rmsnorm(x, weight, eps, residual)
     lib = _require_native()
-    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    assert x.is_contiguous()
     out = torch.empty_like(x)
     n_tokens = x.numel() // x.shape[-1]
     rc = lib.ll_rmsnorm(
@@ -113,10 +115,10 @@ def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float,
 def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
                positions: torch.Tensor):
     """In-place rotate-half RoPE on q [T, n_q, D] and k [T, n_kv, D]."""
-    if not q.is_cuda:
+    if not q.is_cuda or q.dtype != torch.bfloat16:
         return reference.rope_apply(q, k, cos_sin, positions)
     lib = _require_native()
-    assert q.dtype == torch.bfloat16 and q.is_contiguous() and k.is_contiguous()
+    assert q.is_contiguous() and k.is_contiguous()
     assert cos_sin.dtype == torch.float32
     pos = positions.to(torch.int32)
     rc = lib.ll_rope(
@@ -129,10 +131,10 @@ def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
 
 def swiglu(gateup: torch.Tensor) -> torch.Tensor:
     """gateup [T, 2*inter] -> silu(gate) * up  [T, inter]."""
-    if not gateup.is_cuda:
+    if not gateup.is_cuda or gateup.dtype != torch.bfloat16:
         return reference.swiglu(gateup)
     lib = _require_native()
-    assert gateup.dtype == torch.bfloat16 and gateup.is_contiguous()
+    assert gateup.is_contiguous()
     inter = gateup.shape[-1] // 2
     n_tokens = gateup.numel() // gateup.shape[-1]
     out = torch.empty(*gateup.shape[:-1], inter, dtype=gateup.dtype,
@@ -148,7 +150,7 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
               v: torch.Tensor, slots: torch.Tensor) -> None:
     """Scatter k/v [T, n_kv, D] into paged caches at flat slot ids."""
     block_size = kcache.shape[2]
-    if not kcache.is_cuda:
+    if not kcache.is_cuda or kcache.dtype != torch.bfloat16:
         reference.kv_append(kcache, vcache, k, v, slots, block_size)
         return
     lib = _require_native()
@@ -202,7 +204,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
     ``n_parts`` splits long sequences across workgroups (flash-decode);
     chosen automatically to fill the 256-CU chip when not given.
     """
-    if not q.is_cuda:
+    if not q.is_cuda or q.dtype != torch.bfloat16:
         return reference.paged_attn_decode(q, kcache, vcache, block_tables,
                                            seq_lens, scale)
     lib = _require_native()

@@ -1,0 +1,109 @@
+"""/v1/responses end-to-end: gateway -> real (tiny) backend serving the
+Open Responses API shape, stream + non-stream, with token accounting."""
+import json
+import socket
+import subprocess
+import sys
+import tempfile
+import time
+import urllib.request
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parents[1]
+BIN = REPO / "gateway" / "bin"
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+@pytest.fixture(scope="module")
+def stack():
+    backend_port, gw_port = _free_port(), _free_port()
+    backend = subprocess.Popen(
+        [sys.executable, "-m", "llmlb_amd.backend.launcher", "--model",
+         "llama-tiny", "--device", "cpu", "--port", str(backend_port)],
+        cwd=str(REPO), stdout=subprocess.DEVNULL,
+        stderr=subprocess.DEVNULL)
+    import os
+    db = tempfile.mktemp(suffix=".sqlite3")
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "2"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        try:
+            socket.create_connection(("127.0.0.1", backend_port),
+                                     0.2).close()
+            break
+        except OSError:
+            time.sleep(0.2)
+    req = urllib.request.Request(
+        base + "/api/endpoints",
+        data=json.dumps({"name": "b", "base_url":
+                         f"http://127.0.0.1:{backend_port}"}).encode(),
+        headers={"Authorization": "Bearer t"})
+    for _ in range(40):
+        try:
+            urllib.request.urlopen(req, timeout=10)
+            break
+        except Exception:
+            time.sleep(0.3)
+    yield {"base": base}
+    gw.terminate()
+    backend.terminate()
+
+
+def test_responses_nonstream(stack):
+    req = urllib.request.Request(
+        stack["base"] + "/v1/responses",
+        data=json.dumps({"model": "llama-tiny",
+                         "input": "say something",
+                         "max_tokens": 6}).encode(),
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as resp:
+        obj = json.loads(resp.read())
+    assert obj["object"] == "response"
+    assert obj["status"] == "completed"
+    text = obj["output"][0]["content"][0]["text"]
+    assert text.strip()
+    assert obj["usage"]["output_tokens"] == 6
+
+
+def test_responses_streaming(stack):
+    req = urllib.request.Request(
+        stack["base"] + "/v1/responses",
+        data=json.dumps({"model": "llama-tiny", "input": "go",
+                         "max_tokens": 5, "stream": True}).encode(),
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as resp:
+        assert "text/event-stream" in resp.headers.get("Content-Type", "")
+        raw = resp.read().decode()
+    assert '"response.created"' in raw
+    assert raw.count('"response.output_text.delta"') == 5
+    assert '"response.completed"' in raw
+    assert '"output_tokens": 5' in raw or '"output_tokens":5' in raw
+    assert "[DONE]" in raw
+
+
+def test_responses_tokens_recorded(stack):
+    time.sleep(0.5)
+    req = urllib.request.Request(stack["base"] + "/api/requests",
+                                 headers={"Authorization": "Bearer t"})
+    with urllib.request.urlopen(req, timeout=10) as resp:
+        recs = json.loads(resp.read())
+    resp_recs = [r for r in recs if r["api_kind"] == "responses"
+                 and r["status"] == 200]
+    assert resp_recs
+    assert any(r["completion_tokens"] > 0 for r in resp_recs)
