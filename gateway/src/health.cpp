@@ -328,6 +328,43 @@ void Gateway::health_loop() {
           if (telemetry_get(ep->device_index, &g))
             registry.update_gpu_info(ep->id, g);
         }
+        // backend-measured tokens/sec (hipEvent-fed) seeds the TPS
+        // tracker until production samples exist — mirrors the
+        // reference's startup TPS seeding (bootstrap.rs:143-159)
+        {
+          MiniHttpResult h = mini_http_request(
+              ep->host, ep->port, "GET", ep->path_prefix + "/api/health",
+              "", cfg.health_timeout_ms);
+          if (h.status == 200) {
+            Json doc;
+            if (Json::parse(h.body, doc)) {
+              double tps = doc["tokens_per_sec"].as_num(0);
+              std::string hm = doc["model"].as_str();
+              if (tps > 0 && !hm.empty()) {
+                auto st = registry.tps_state(ep->id, hm, ApiKind::kChat);
+                std::lock_guard<std::mutex> lk(st->mu);
+                if (st->samples == 0) {   // never override production
+                  st->ema = tps;
+                  st->samples = 1;
+                  st->updated_ms = wall_ms();
+                }
+              }
+              // xLLM-style GPU fields from backends without a local
+              // device binding (endpoint_checker.rs:514-557)
+              if (ep->device_index < 0 &&
+                  doc.find("gpu_memory_total_mb")) {
+                GpuInfo g;
+                g.hbm_total_gb =
+                    doc["gpu_memory_total_mb"].as_num(0) / 1024.0;
+                g.hbm_used_gb =
+                    doc["gpu_memory_used_mb"].as_num(0) / 1024.0;
+                g.device_index = -1;
+                if (g.hbm_total_gb > 0)
+                  registry.update_gpu_info(ep->id, g);
+              }
+            }
+          }
+        }
       } else {
         int fails = ep->consecutive_failures.fetch_add(1) + 1;
         if (cur == EndpointStatus::kPending) {

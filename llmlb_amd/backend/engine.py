@@ -186,6 +186,16 @@ class Engine:
         self.n_kv_local = n_kv_local
         self.steps_done = 0
         self.tokens_out = 0
+        # observability: hipEvent-tagged decode-step GPU time (sampled
+        # every 32 steps) and a tokens/sec EMA — both surfaced through
+        # /api/health so the gateway can seed its TPS estimate
+        # (BASELINE: hipEvent queue-depth tagging + backend-fed TPS)
+        self.gpu_step_ms_ema = 0.0
+        self.tokens_per_sec_ema = 0.0
+        self._tps_window_t0 = time.monotonic()
+        self._tps_window_tokens = 0
+        self._ev_start = None
+        self._ev_end = None
 
         # persistent decode state (device-resident)
         self.max_rows = cfg.max_batch
@@ -405,6 +415,13 @@ class Engine:
                 torch.tensor(grow_vals, dtype=torch.int32,
                              device=self.device)
 
+        sample_gpu_time = (self.device.type == "cuda"
+                           and self.steps_done % 32 == 0)
+        if sample_gpu_time:
+            if self._ev_start is None:
+                self._ev_start = torch.cuda.Event(enable_timing=True)
+                self._ev_end = torch.cuda.Event(enable_timing=True)
+            self._ev_start.record()
         rows = self._rows_tensor()
         lens = self._len_dev[rows]                      # [S] int32
         ids = self._last_tok[rows]                      # [S] long
@@ -443,7 +460,14 @@ class Engine:
                       [s.generator for s in self.running])
         self._last_tok[rows] = toks
         self._len_dev[rows] = lens + 1
+        if sample_gpu_time:
+            self._ev_end.record()
         toks_cpu = toks.tolist()
+        if sample_gpu_time:
+            ms = self._ev_start.elapsed_time(self._ev_end)
+            self.gpu_step_ms_ema = (ms if self.gpu_step_ms_ema == 0
+                                    else 0.2 * ms
+                                    + 0.8 * self.gpu_step_ms_ema)
         if self.timing is not None:
             self.timing["sample_sync"] += time.monotonic() - _t2
             _t3 = time.monotonic()
@@ -500,6 +524,16 @@ class Engine:
             seq.t_first = time.monotonic()
         seq.out_ids.append(tok)
         self.tokens_out += 1
+        self._tps_window_tokens += 1
+        now = time.monotonic()
+        dt = now - self._tps_window_t0
+        if dt >= 2.0:
+            tps = self._tps_window_tokens / dt
+            self.tokens_per_sec_ema = (tps if self.tokens_per_sec_ema == 0
+                                       else 0.2 * tps
+                                       + 0.8 * self.tokens_per_sec_ema)
+            self._tps_window_t0 = now
+            self._tps_window_tokens = 0
         fin = None
         if tok in seq.params.stop_token_ids:
             fin = "stop"

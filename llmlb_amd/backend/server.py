@@ -104,6 +104,9 @@ class BackendServer:
             "status": "ok",
             "model": self.model_name,
             "active_requests": self.engine.num_running,
+            "queued_requests": self.engine.waiting.qsize(),
+            "tokens_per_sec": round(self.engine.tokens_per_sec_ema, 2),
+            "gpu_step_ms": round(self.engine.gpu_step_ms_ema, 3),
             "uptime_s": time.time() - self.start_time,
             **gpu,
         }
