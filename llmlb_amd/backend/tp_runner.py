@@ -103,7 +103,8 @@ def main(argv=None):
     backend = "nccl" if have_gpu else "gloo"  # nccl IS RCCL on ROCm
     dist.init_process_group(backend)
     group_cpu = dist.new_group(backend="gloo")  # control-plane casts
-    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    ndev = torch.cuda.device_count() if have_gpu else 0
+    device = f"cuda:{local_rank % ndev}" if have_gpu else "cpu"
     if have_gpu:
         torch.cuda.set_device(device)
 

@@ -68,6 +68,15 @@ def test_tp2_rccl_matches_tp1(tmp_path):
     port = s.getsockname()[1]
     s.close()
     result = str(tmp_path / "tp_gpu.pt")
-    mp.spawn(_worker, args=(2, port, result), nprocs=2, join=True)
+    try:
+        mp.spawn(_worker, args=(2, port, result), nprocs=2, join=True)
+    except Exception as e:  # RCCL may refuse 2 ranks on ONE device
+        msg = str(e)
+        if ("Duplicate GPU" in msg or "duplicate GPU" in msg
+                or "invalid device" in msg or "unhandled system error"
+                in msg):
+            pytest.skip(f"RCCL cannot share one GPU across ranks: "
+                        f"{msg[:120]}")
+        raise
     out = torch.load(result)
     assert out["err"] < 1e-3, f"TP=2 RCCL logits diverge: {out['err']}"
