@@ -259,6 +259,7 @@ def main() -> None:
     barrier()  # all backends up
 
     gw_proc = None
+    gw_port_holder = [0]
     result: dict = {}
     try:
         if rank == 0:
@@ -306,21 +307,26 @@ def main() -> None:
                     pass
                 time.sleep(0.5)
 
-            total_conc = conc * n_gpus
+            gw_port_holder[0] = gw_port
 
-            def wave():
-                return asyncio.run(run_wave(base, model_name, total_conc,
-                                            args.prompt_len, args.gen_len,
-                                            "bench"))
+        # every rank drives its own share of the load (a single asyncio
+        # driver saturates before the gateway does at 512 streams);
+        # broadcast the gateway port first
+        if dist is not None:
+            obj = [gw_port_holder[0]]
+            work = dist.broadcast_object_list(obj, src=0)
+            gw_port_holder[0] = obj[0]
+        base = f"http://127.0.0.1:{gw_port_holder[0]}"
 
-            log("gateway + endpoints ready; warmup")
-            for _ in range(args.warmup):
-                wave()
-            log("warmup done")
-        else:
-            pass  # other ranks: engines serve during rank 0's waves
+        def wave():
+            return asyncio.run(run_wave(base, model_name, conc,
+                                        args.prompt_len, args.gen_len,
+                                        "bench"))
 
-        # warmup complete on rank 0; ranks idle-serve meanwhile
+        log("warmup")
+        for _ in range(args.warmup):
+            wave()
+        log("warmup done")
         barrier()
         sync()
         barrier()
@@ -328,27 +334,30 @@ def main() -> None:
         total_tokens = 0
         total_errors = 0
         lat_all: list = []
-        if rank == 0:
-            for si in range(args.steps):
-                tok, err, lats = asyncio.run(
-                    run_wave(f"http://127.0.0.1:{gw_port}", model_name,
-                             conc * n_gpus, args.prompt_len, args.gen_len,
-                             "bench"))
-                log(f"step {si}: tokens={tok} errors={err}")
-                total_tokens += tok
-                total_errors += err
-                lat_all.extend(lats)
+        for si in range(args.steps):
+            tok, err, lats = wave()
+            log(f"step {si}: tokens={tok} errors={err}")
+            total_tokens += tok
+            total_errors += err
+            lat_all.extend(lats)
         barrier()
         sync()
         barrier()
         elapsed = time.monotonic() - t0
-        # max over ranks
+        # aggregate across ranks: MAX elapsed, SUM tokens/errors
         if dist is not None:
             t = torch.tensor([elapsed], dtype=torch.float64)
             work = dist.all_reduce(t, op=dist.ReduceOp.MAX, async_op=True)
             while not work.is_completed():
                 time.sleep(0.005)
             elapsed = float(t[0])
+            te = torch.tensor([total_tokens, total_errors],
+                              dtype=torch.float64)
+            work = dist.all_reduce(te, op=dist.ReduceOp.SUM, async_op=True)
+            while not work.is_completed():
+                time.sleep(0.005)
+            total_tokens = int(te[0])
+            total_errors = int(te[1])
 
         if rank == 0:
             lat_all.sort()
