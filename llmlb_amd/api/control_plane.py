@@ -106,6 +106,19 @@ class RegisterBody(BaseModel):
     password: str
 
 
+class ModelRegBody(BaseModel):
+    id: str
+    name: str = ""
+    source: str = ""
+    capabilities: str = "chat"
+    size_bytes: int | None = None
+
+
+class UpdateBody(BaseModel):
+    artifact_path: str
+    target_path: str = ""
+
+
 class TpsBenchmarkBody(BaseModel):
     # reference api/benchmarks.rs:25-34 defaults and caps
     model: str
@@ -453,9 +466,12 @@ def create_app(db_path: str | None = None,
             await ws.close(code=4401)
             return
         await ws.accept()
+        last_event_ms = int(time.time() * 1000)
         try:
             while True:
-                # push a summary snapshot every 2 s (event stream analog)
+                # push a summary snapshot every 2 s plus any new gateway
+                # events (EndpointStatusChanged / TpsUpdated — the
+                # reference's DashboardEvents over /ws/dashboard)
                 try:
                     req = urllib.request.Request(
                         gateway_url + "/api/dashboard/summary",
@@ -465,6 +481,15 @@ def create_app(db_path: str | None = None,
                     await ws.send_json({"type": "summary",
                                         "data": summary,
                                         "ts": time.time()})
+                    ereq = urllib.request.Request(
+                        gateway_url +
+                        f"/api/events?since={last_event_ms}",
+                        headers=_gateway_headers())
+                    with urllib.request.urlopen(ereq, timeout=5) as resp:
+                        events = json.loads(resp.read())
+                    for ev in events:
+                        last_event_ms = max(last_event_ms, ev["ts_ms"])
+                        await ws.send_json({"type": "event", "data": ev})
                 except Exception as e:
                     await ws.send_json({"type": "error", "error": str(e)})
                 await asyncio.sleep(2)
@@ -510,13 +535,6 @@ def create_app(db_path: str | None = None,
         return {"lines": out[-lines:]}
 
     # ---------------- model registry metadata (registry::models) -----
-    class ModelRegBody(BaseModel):
-        id: str
-        name: str = ""
-        source: str = ""
-        capabilities: str = "chat"
-        size_bytes: int | None = None
-
     @app.get("/api/models")
     def list_registered_models(user=Depends(auth_user)):
         with db() as conn:
@@ -551,10 +569,6 @@ def create_app(db_path: str | None = None,
     # drain the data plane -> apply (atomic binary replace with .bak
     # rollback) -> resume.  Restart is left to the process supervisor.
     update_state: dict = {"state": "idle", "history": []}
-
-    class UpdateBody(BaseModel):
-        artifact_path: str
-        target_path: str = ""
 
     @app.get("/api/update/status")
     def update_status(user=Depends(require_admin)):

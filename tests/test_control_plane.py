@@ -140,3 +140,71 @@ def test_token_stats_empty_ok(app_client):
     r = app_client.get("/api/dashboard/token-stats",
                        headers={"Authorization": "Bearer admtok"})
     assert r.status_code == 200
+
+
+def test_update_manager_drain_apply_rollback(app_client, tmp_path,
+                                             monkeypatch):
+    """Self-update flow against a stub gateway (drain -> replace ->
+    resume; rollback restores the .bak)."""
+    import http.server
+    import threading as th
+
+    calls = []
+
+    class StubGW(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):
+            pass
+
+        def do_POST(self):
+            calls.append(self.path)
+            self.send_response(200)
+            self.send_header("Content-Length", "2")
+            self.end_headers()
+            self.wfile.write(b"{}")
+
+        def do_GET(self):
+            body = b"llmlb_active_requests 0\n"
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+    import socket as sk
+    s = sk.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    httpd = http.server.ThreadingHTTPServer(("127.0.0.1", port), StubGW)
+    th.Thread(target=httpd.serve_forever, daemon=True).start()
+
+    from llmlb_amd.api import control_plane as cp
+    app = cp.create_app(db_path=":memory:",
+                        gateway_url=f"http://127.0.0.1:{port}",
+                        jwt_secret="s", admin_token="admtok")
+    from fastapi.testclient import TestClient
+    client = TestClient(app)
+    hdrs = {"Authorization": "Bearer admtok"}
+
+    old = tmp_path / "llmlb-gw"
+    old.write_bytes(b"OLD BINARY")
+    new = tmp_path / "artifact"
+    new.write_bytes(b"NEW BINARY")
+    r = client.post("/api/update/apply", headers=hdrs,
+                    json={"artifact_path": str(new),
+                          "target_path": str(old)})
+    assert r.status_code == 202
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        st = client.get("/api/update/status", headers=hdrs).json()
+        if st["state"] in ("applied", "failed"):
+            break
+        time.sleep(0.2)
+    assert st["state"] == "applied", st
+    assert old.read_bytes() == b"NEW BINARY"
+    assert (tmp_path / "llmlb-gw.bak").read_bytes() == b"OLD BINARY"
+    assert "/api/system/drain" in calls
+    assert "/api/system/resume" in calls
+    r = client.post("/api/update/rollback", headers=hdrs)
+    assert r.status_code == 200
+    assert old.read_bytes() == b"OLD BINARY"
+    httpd.shutdown()
