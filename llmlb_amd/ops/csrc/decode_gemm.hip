@@ -25,6 +25,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 #define WAVE 64
 
@@ -32,8 +33,11 @@ typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-// N_TILES = N/16 rounded up; one kernel instantiation per bucket
-template <int N_TILES, int COLS_W, int SPLITK_W>
+// N_TILES = N/16 rounded up; one kernel instantiation per bucket.
+// UNROLL = 32-wide k-steps kept in flight per iteration: the B stream
+// rides HBM (~900 cyc latency), so enough B loads must be outstanding
+// to cover it (MI355X_MICROARCH per-instruction constants).
+template <int N_TILES, int COLS_W, int SPLITK_W, int UNROLL = 4>
 __global__ __launch_bounds__(512)
 void decode_gemm_kernel(bf16* __restrict__ out,        // [N, M]
                         const bf16* __restrict__ x,    // [N, K]
@@ -64,23 +68,28 @@ void decode_gemm_kernel(bf16* __restrict__ out,        // [N, M]
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // main K loop: 32 per MFMA step, unrolled 4x for load depth
+  // main K loop: 32 per MFMA step, UNROLL-deep for load depth.  B
+  // loads for the whole iteration issue first (the HBM stream), A
+  // (L2-resident x) after, then the MFMAs consume in order.
   const int koff = l_hi * 8;  // this lane's k offset within the step
 #pragma unroll 1
-  for (int k0 = k_begin; k0 < k_begin + Ks; k0 += 128) {
-    bf16x8 bfrag[4];
-    bf16x8 afrag[4][N_TILES];
+  for (int k0 = k_begin; k0 < k_begin + Ks; k0 += 32 * UNROLL) {
+    bf16x8 bfrag[UNROLL];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < UNROLL; ++u)
+      bfrag[u] = *reinterpret_cast<const bf16x8*>(wrow + k0 + u * 32 +
+                                                  koff);
+    bf16x8 afrag[UNROLL][N_TILES];
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
       const int kk = k0 + u * 32 + koff;
-      bfrag[u] = *reinterpret_cast<const bf16x8*>(wrow + kk);
 #pragma unroll
       for (int t = 0; t < N_TILES; ++t)
         afrag[u][t] = *reinterpret_cast<const bf16x8*>(
             x + (size_t)arow[t] * K + kk);
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u)
+    for (int u = 0; u < UNROLL; ++u)
 #pragma unroll
       for (int t = 0; t < N_TILES; ++t)
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -147,6 +156,8 @@ extern "C" {
 int ll_decode_gemm(void* out, const void* x, const void* w, int N, int M,
                    int K, void* stream) {
   if (K % 256 != 0 || M % 16 != 0 || N > 128) return -2;
+  const char* uenv = getenv("LLMLB_DECODE_GEMM_UNROLL");
+  const int unroll = uenv ? atoi(uenv) : 8;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
 
@@ -158,15 +169,22 @@ int ll_decode_gemm(void* out, const void* x, const void* w, int N, int M,
   else cols_w = 1;
   const int splitk = 8 / cols_w;
   if (K % (splitk * 128) != 0) return -2;
+  (void)unroll;
   const int blocks = M / (16 * cols_w);
   const size_t lds = (size_t)cols_w * (splitk - 1) * n_tiles * 256 *
                      sizeof(float);
   if (lds > 160 * 1024) return -2;
 
-#define LAUNCH(NT, CW, SW)                                                  \
-  hipLaunchKernelGGL((decode_gemm_kernel<NT, CW, SW>), dim3(blocks),        \
+#define LAUNCH1(NT, CW, SW, UR)                                             \
+  hipLaunchKernelGGL((decode_gemm_kernel<NT, CW, SW, UR>), dim3(blocks),    \
                      dim3(512), lds, s, (bf16*)out, (const bf16*)x,         \
                      (const bf16*)w, N, M, K)
+#define LAUNCH(NT, CW, SW)                                                  \
+  do {                                                                      \
+    if (unroll >= 16 && (K / SW) % 512 == 0) LAUNCH1(NT, CW, SW, 16);       \
+    else if (unroll >= 8 && (K / SW) % 256 == 0) LAUNCH1(NT, CW, SW, 8);    \
+    else LAUNCH1(NT, CW, SW, 4);                                            \
+  } while (0)
 #define DISPATCH_CW(NT)                                                     \
   do {                                                                      \
     if (cols_w == 8) LAUNCH(NT, 8, 1);                                      \
@@ -186,6 +204,7 @@ int ll_decode_gemm(void* out, const void* x, const void* w, int N, int M,
   }
 #undef DISPATCH_CW
 #undef LAUNCH
+#undef LAUNCH1
   hipError_t e = hipGetLastError();
   return e == hipSuccess ? 0 : (int)e;
 }
