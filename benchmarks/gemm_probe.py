@@ -51,19 +51,33 @@ def main():
     for N, M, K, label in SHAPES:
         x = torch.randn(N, K, dtype=torch.bfloat16, device=dev) * 0.3
         w = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
-        # interleaved A/B
-        t_custom = bench_one(ops.linear_decode, x, w)
+        # interleaved A/B (v1, v2, hipBLASLt)
+        def v1(a, b):
+            ops._DECODE_GEMM_VERSION = 1
+            return ops.linear_decode(a, b)
+
+        def v2(a, b):
+            return ops.decode_gemm_v2(a, b)
+
+        t_v1 = bench_one(v1, x, w)
         t_blas = bench_one(torch.nn.functional.linear, x, w)
-        t_custom2 = bench_one(ops.linear_decode, x, w)
-        t_custom = min(t_custom, t_custom2)
+        try:
+            t_v2 = bench_one(v2, x, w)
+            err = (v2(x, w).float()
+                   - torch.nn.functional.linear(x, w).float()) \
+                .abs().max().item()
+        except ValueError:
+            t_v2, err = float("inf"), -1
         wbytes = M * K * 2
         out[label] = {
             "shape": f"N{N} M{M} K{K}",
-            "custom_us": round(t_custom * 1e6, 1),
+            "v1_us": round(t_v1 * 1e6, 1),
+            "v2_us": round(t_v2 * 1e6, 1),
             "hipblaslt_us": round(t_blas * 1e6, 1),
-            "custom_tbps": round(wbytes / t_custom / 1e12, 2),
+            "v2_tbps": round(wbytes / t_v2 / 1e12, 2),
             "hipblaslt_tbps": round(wbytes / t_blas / 1e12, 2),
-            "speedup": round(t_blas / t_custom, 2),
+            "v2_speedup_vs_blas": round(t_blas / t_v2, 2),
+            "v2_max_err": round(err, 4),
         }
         print(json.dumps({label: out[label]}), flush=True)
     print(json.dumps(out))

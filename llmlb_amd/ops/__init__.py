@@ -44,6 +44,7 @@ def _try_load() -> None:
     _lib.ll_kv_append.restype = ctypes.c_int
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
     _lib.ll_decode_gemm.restype = ctypes.c_int
+    _lib.ll_decode_gemm_v2.restype = ctypes.c_int
 
 
 _try_load()
@@ -165,6 +166,28 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
 
 _USE_CUSTOM_DECODE_GEMM = os.environ.get(
     "LLMLB_CUSTOM_DECODE_GEMM", "0").lower() in ("1", "true", "on")
+_DECODE_GEMM_VERSION = int(os.environ.get("LLMLB_DECODE_GEMM_VERSION",
+                                          "2"))
+
+
+def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
+                   ws: torch.Tensor | None = None) -> torch.Tensor:
+    """Staged-x decode GEMM (v2).  Raises on unsupported shapes."""
+    lib = _require_native()
+    N, K = x.shape
+    M = weight.shape[0]
+    out = torch.empty(N, M, dtype=x.dtype, device=x.device)
+    need_ws = (M // 128) < 224
+    if need_ws and ws is None:
+        ws = torch.empty(N, M, dtype=torch.float32, device=x.device)
+    rc = lib.ll_decode_gemm_v2(
+        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
+        _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
+        _stream())
+    if rc == -2:
+        raise ValueError("unsupported shape for decode_gemm_v2")
+    _check(rc, "ll_decode_gemm_v2")
+    return out
 
 
 def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
@@ -181,6 +204,12 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         return torch.nn.functional.linear(x, weight)
     N, K = x.shape
     M = weight.shape[0]
+    if _DECODE_GEMM_VERSION >= 2 and N <= 128 and K % 256 == 0 \
+            and M % 128 == 0:
+        try:
+            return decode_gemm_v2(x, weight)
+        except ValueError:
+            pass
     if N > 128 or K % 256 != 0 or M % 16 != 0:
         return torch.nn.functional.linear(x, weight)
     out = torch.empty(N, M, dtype=x.dtype, device=x.device)
