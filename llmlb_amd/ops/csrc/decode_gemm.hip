@@ -241,11 +241,8 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
   const int l_lo = lane & 15;
   const int l_hi = lane >> 4;
 
-  // double-buffered x staging, T14 split (guide §6 G15: issue loads
-  // early, ds_write AFTER the barrier, one barrier per chunk)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* const xs0 = smem;
-  char* const xs1 = smem + (size_t)N_TILES * 16 * CHUNK_K * 2;
+  bf16* xs = reinterpret_cast<bf16*>(smem);  // [N_TILES*16][CHUNK_K] swz
 
   const bf16* wrow = w + (size_t)(col_base + l_lo) * K;
   f32x4 acc[N_TILES];
@@ -253,62 +250,30 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
   for (int t = 0; t < N_TILES; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int rows = N_TILES * 16;
-  // staging geometry: each thread copies rows/16 16-byte pieces/chunk
+  // staging geometry: each thread copies (rows*CHUNK_K*2 / 512 / 16)
+  // 16-byte pieces per chunk; source row clamped for padded tails
   const int pieces = rows * CHUNK_K * 2 / (512 * 16);
   const int piece0 = threadIdx.x * pieces;
-  int stage_tok[8], stage_swz[8];
-  const bf16* stage_src[8];
-#pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    if (p < pieces) {
-      const int piece = piece0 + p;
-      const int tok = piece / (CHUNK_K * 2 / 16);
-      const int kb = piece % (CHUNK_K * 2 / 16);
-      const int src_tok = tok < N ? tok : 0;
-      stage_tok[p] = tok;
-      stage_swz[p] = (kb ^ (tok & 15)) * 16 + tok * CHUNK_K * 2;
-      stage_src[p] = x + (size_t)src_tok * K + kb * 8;
-    }
-  }
-  const int nchunks = Ks / CHUNK_K;
-  uint4 v[8];
-  // prologue: load chunk 0, write buf0
-#pragma unroll
-  for (int p = 0; p < 8; ++p)
-    if (p < pieces)
-      v[p] = *reinterpret_cast<const uint4*>(stage_src[p] + k_begin);
-#pragma unroll
-  for (int p = 0; p < 8; ++p)
-    if (p < pieces)
-      *reinterpret_cast<uint4*>(xs0 + stage_swz[p]) = v[p];
-  if (nchunks > 1) {  // preload chunk 1 into registers
-#pragma unroll
-    for (int p = 0; p < 8; ++p)
-      if (p < pieces)
-        v[p] = *reinterpret_cast<const uint4*>(stage_src[p] + k_begin +
-                                               CHUNK_K);
-  }
-  __syncthreads();
 
-  for (int ci = 0; ci < nchunks; ++ci) {
-    const int k0 = k_begin + ci * CHUNK_K;
-    char* cur = (ci & 1) ? xs1 : xs0;
-    // write chunk ci+1 (held in v) into the other buffer, then issue
-    // loads for chunk ci+2 — both overlap this chunk's MFMAs
-    if (ci + 1 < nchunks) {
-      char* nxt = ((ci + 1) & 1) ? xs1 : xs0;
+  for (int k0 = k_begin; k0 < k_begin + Ks; k0 += CHUNK_K) {
+    // ---- stage x[*, k0:k0+CHUNK_K] into LDS (swizzled dest) ----
+    __syncthreads();  // previous chunk fully consumed
 #pragma unroll
-      for (int p = 0; p < 8; ++p)
-        if (p < pieces)
-          *reinterpret_cast<uint4*>(nxt + stage_swz[p]) = v[p];
-      if (ci + 2 < nchunks) {
-#pragma unroll
-        for (int p = 0; p < 8; ++p)
-          if (p < pieces)
-            v[p] = *reinterpret_cast<const uint4*>(
-                stage_src[p] + k0 + 2 * CHUNK_K);
+    for (int p = 0; p < 8; ++p) {
+      if (p < pieces) {
+        const int piece = piece0 + p;
+        const int tok = piece / (CHUNK_K * 2 / 16);   // 16B per piece
+        const int kb = piece % (CHUNK_K * 2 / 16);    // 16B col index
+        const int src_tok = tok < N ? tok : 0;
+        const uint4 v = *reinterpret_cast<const uint4*>(
+            x + (size_t)src_tok * K + k0 + kb * 8);
+        const int swz_kb = kb ^ (tok & 15);
+        *reinterpret_cast<uint4*>(
+            reinterpret_cast<char*>(xs) + (size_t)tok * CHUNK_K * 2 +
+            swz_kb * 16) = v;
       }
     }
+    __syncthreads();
     // ---- consume: STEPS x (B global 16B + N_TILES LDS reads + MFMA) --
 #pragma unroll
     for (int s2 = 0; s2 < STEPS; ++s2) {
@@ -320,12 +285,12 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
         const int tok = t * 16 + l_lo;
         const int swz_kb = kb ^ (tok & 15);
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            cur + tok * CHUNK_K * 2 + swz_kb * 16);
+            reinterpret_cast<char*>(xs) + (size_t)tok * CHUNK_K * 2 +
+            swz_kb * 16);
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag, bfrag, acc[t], 0, 0, 0);
       }
     }
-    __syncthreads();  // next chunk's buffer fully written + consumed
   }
 
   // ---- store ----
@@ -377,7 +342,7 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
   int SK = 1;
   while (mblocks * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
     SK *= 2;
-  const size_t lds = (size_t)n_tiles * 16 * 256 * 2 * 2;  // 2 x-chunks
+  const size_t lds = (size_t)n_tiles * 16 * 256 * 2;  // x chunk
   if (lds > 160 * 1024) return -2;
   if (SK > 1 && ws == nullptr) return -2;
   if (SK > 1)
