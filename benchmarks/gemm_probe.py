@@ -16,7 +16,7 @@ import torch
 import os
 os.environ["LLMLB_CUSTOM_DECODE_GEMM"] = "1"
 import llmlb_amd.ops as ops
-ops._USE_CUSTOM_DECODE_GEMM = True
+ops._USE_CUSTOM_DECODE_GEMM = "1"
 
 SHAPES = [  # (N, M, K, label)
     (64, 6144, 4096, "qkv"),
@@ -53,8 +53,14 @@ def main():
         w = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
         # interleaved A/B (v1, v2, hipBLASLt)
         def v1(a, b):
-            ops._DECODE_GEMM_VERSION = 1
-            return ops.linear_decode(a, b)
+            import ctypes as _ct
+            N2, K2 = a.shape
+            M2 = b.shape[0]
+            o = torch.empty(N2, M2, dtype=a.dtype, device=a.device)
+            ops._lib.ll_decode_gemm(
+                ops._ptr(o), ops._ptr(a), ops._ptr(b), _ct.c_int(N2),
+                _ct.c_int(M2), _ct.c_int(K2), ops._stream())
+            return o
 
         def v2(a, b):
             return ops.decode_gemm_v2(a, b)

@@ -165,7 +165,7 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
 
 
 _USE_CUSTOM_DECODE_GEMM = os.environ.get(
-    "LLMLB_CUSTOM_DECODE_GEMM", "0").lower() in ("1", "true", "on")
+    "LLMLB_CUSTOM_DECODE_GEMM", "auto").lower()
 _DECODE_GEMM_VERSION = int(os.environ.get("LLMLB_DECODE_GEMM_VERSION",
                                           "2"))
 
@@ -193,33 +193,28 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
 def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """out = x @ W^T.
 
-    The in-tree weight-streaming kernel (csrc/decode_gemm.hip) is
-    numerically verified but measured SLOWER than hipBLASLt on the
-    llama-3-8b decode shapes (profiles/r01_decode_gemm_analysis.md —
-    the skinny-N regime is per-CU load-path/latency-bound, and
-    hipBLASLt's tiles already sit near that wall), so it is opt-in via
-    LLMLB_CUSTOM_DECODE_GEMM=1 pending the glds+swizzle rewrite."""
-    if (not _USE_CUSTOM_DECODE_GEMM or not x.is_cuda or _lib is None
+    Dispatch policy (measured, profiles/r01_decode_gemm_v2_ab.md):
+    the staged-x v2 kernel wins on the split-K shapes with large K
+    (down-projection: M/128 < 224 and K >= 8192) and on small batches
+    (N <= 16); hipBLASLt keeps the rest (notably gate_up at N >= 64).
+    LLMLB_CUSTOM_DECODE_GEMM: "auto" (default) = this policy,
+    "1"/"on" = v2 everywhere it is legal, "0"/"off" = hipBLASLt only."""
+    mode = _USE_CUSTOM_DECODE_GEMM
+    if (mode in ("0", "false", "off") or not x.is_cuda or _lib is None
             or x.dtype != torch.bfloat16 or x.dim() != 2):
         return torch.nn.functional.linear(x, weight)
     N, K = x.shape
     M = weight.shape[0]
-    if _DECODE_GEMM_VERSION >= 2 and N <= 128 and K % 256 == 0 \
-            and M % 128 == 0:
+    legal = N <= 128 and K % 256 == 0 and M % 128 == 0
+    if legal and mode == "auto":
+        mblocks = M // 128
+        legal = (mblocks < 224 and K >= 8192) or N <= 16
+    if legal:
         try:
             return decode_gemm_v2(x, weight)
         except ValueError:
             pass
-    if N > 128 or K % 256 != 0 or M % 16 != 0:
-        return torch.nn.functional.linear(x, weight)
-    out = torch.empty(N, M, dtype=x.dtype, device=x.device)
-    rc = _lib.ll_decode_gemm(_ptr(out), _ptr(x), _ptr(weight),
-                             ctypes.c_int(N), ctypes.c_int(M),
-                             ctypes.c_int(K), _stream())
-    if rc == -2:
-        return torch.nn.functional.linear(x, weight)
-    _check(rc, "ll_decode_gemm")
-    return out
+    return torch.nn.functional.linear(x, weight)
 
 
 def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,

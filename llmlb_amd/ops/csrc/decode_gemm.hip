@@ -346,7 +346,7 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
   if (lds > 160 * 1024) return -2;
   if (SK > 1 && ws == nullptr) return -2;
   if (SK > 1)
-    hipMemsetAsync(ws, 0, (size_t)N * M * sizeof(float), s);
+    (void)hipMemsetAsync(ws, 0, (size_t)N * M * sizeof(float), s);
   dim3 grid(mblocks, SK), block(512);
 
 #define LAUNCH_V2(NT)                                                       \

@@ -138,7 +138,7 @@ def test_paged_attn_partition_consistency():
                                    (128, 28672, 4096),    # big bucket
                                    (33, 4096, 4096)])     # padded N
 def test_decode_gemm(N, M, K, monkeypatch):
-    monkeypatch.setattr(ops, "_USE_CUSTOM_DECODE_GEMM", True)
+    monkeypatch.setattr(ops, "_USE_CUSTOM_DECODE_GEMM", "1")
     torch.manual_seed(11)
     x = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.3
     w = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.05
@@ -152,7 +152,7 @@ def test_decode_gemm(N, M, K, monkeypatch):
 
 
 def test_decode_gemm_shape_fallback(monkeypatch):
-    monkeypatch.setattr(ops, "_USE_CUSTOM_DECODE_GEMM", True)
+    monkeypatch.setattr(ops, "_USE_CUSTOM_DECODE_GEMM", "1")
     # K % 1024 != 0 at this split config -> clean fallback to hipBLASLt
     x = torch.randn(16, 512, dtype=torch.bfloat16, device=DEV)
     w = torch.randn(256, 512, dtype=torch.bfloat16, device=DEV)
