@@ -1,0 +1,257 @@
+// llmlb-test: native unit tests for the gateway internals (TPS math,
+// selection, admission, HTTP parsing, chunked scanner, SSE tap, JSON,
+// crypto).  Mirrors the reference's in-file unit-test tier
+// (SURVEY.md §4).  Exit code 0 = all pass.
+#include <cstdio>
+
+#include "http.h"
+#include "json.h"
+#include "router.h"
+#include "sha256.h"
+
+using namespace gw;
+
+static int g_failures = 0;
+#define CHECK(cond)                                                        \
+  do {                                                                     \
+    if (!(cond)) {                                                         \
+      fprintf(stderr, "FAIL %s:%d: %s\n", __FILE__, __LINE__, #cond);      \
+      ++g_failures;                                                        \
+    }                                                                      \
+  } while (0)
+
+static void test_sha256() {
+  // FIPS 180-4 known vectors
+  CHECK(sha256_hex("abc") ==
+        "ba7816bf8f01cfea414140de5dae2223b00361a396177a9cb410ff61f20015ad");
+  CHECK(sha256_hex("") ==
+        "e3b0c44298fc1c149afbf4c8996fb92427ae41e4649b934ca495991b7852b855");
+  // RFC 4231 HMAC-SHA256 test case 2
+  uint8_t mac[32];
+  hmac_sha256("Jefe", "what do ya want for nothing?", mac);
+  CHECK(hex_encode(mac, 32) ==
+        "5bdcc146bf60754e6a042426089575c75a003f089d2739839dec58b964ec3843");
+}
+
+static void test_b64() {
+  CHECK(b64_encode("hello") == "aGVsbG8=");
+  CHECK(b64_decode("aGVsbG8=") == "hello");
+  CHECK(b64_encode("a") == "YQ==");
+  CHECK(b64_decode(b64_encode("\x00\xff\x10binary", false, true)) ==
+        std::string("\x00\xff\x10binary", 9));
+}
+
+static void test_json() {
+  Json v;
+  std::string err;
+  CHECK(Json::parse(R"({"a":[1,2.5,"x\n",true,null],"b":{"c":-3}})", v,
+                    &err));
+  CHECK(v["a"].as_arr().size() == 5);
+  CHECK(v["a"].as_arr()[1].as_num() == 2.5);
+  CHECK(v["a"].as_arr()[2].as_str() == "x\n");
+  CHECK(v["b"]["c"].as_int() == -3);
+  CHECK(!Json::parse("{", v, &err));
+  CHECK(!Json::parse("{\"a\":}", v, &err));
+  // unicode escape
+  CHECK(Json::parse(R"("é😀")", v));
+  CHECK(v.as_str() == "\xc3\xa9\xf0\x9f\x98\x80");
+  // round trip
+  Json obj = JsonObject{{"x", Json("a\"b")}, {"n", Json(42)}};
+  Json back;
+  CHECK(Json::parse(obj.dump(), back));
+  CHECK(back["x"].as_str() == "a\"b" && back["n"].as_int() == 42);
+  // raw field helpers
+  CHECK(raw_string_field(R"({"model":"m1","nested":{"model":"no"}})",
+                         "model") == "m1");
+  CHECK(raw_string_field(R"({"first":"x","model":"m2"})", "model") ==
+        "m2");
+  CHECK(raw_string_field(R"({"a":{"model":"inner"}})", "model") == "");
+}
+
+static void test_http_parse() {
+  HttpRequest req;
+  std::string raw = "POST /v1/chat/completions?x=1 HTTP/1.1\r\n"
+                    "Host: h\r\nContent-Length: 5\r\n\r\nhello";
+  CHECK(parse_request_head(raw, req) == 1);
+  CHECK(req.method == "POST");
+  CHECK(req.path == "/v1/chat/completions");
+  CHECK(req.query == "x=1");
+  CHECK(req.content_length == 5);
+  CHECK(req.keep_alive);
+  HttpRequest req2;
+  CHECK(parse_request_head("GET / HT", req2) == 0);  // incomplete
+  CHECK(parse_request_head("garbage\r\n\r\n", req2) == -1);
+
+  HttpResponse resp;
+  std::string rr = "HTTP/1.1 200 OK\r\nTransfer-Encoding: chunked\r\n"
+                   "Content-Type: text/event-stream\r\n\r\n";
+  CHECK(parse_response_head(rr, resp) == 1);
+  CHECK(resp.status == 200 && resp.chunked);
+}
+
+static void test_chunked_scanner() {
+  // single feed
+  ChunkedScanner cs;
+  std::string body = "5\r\nhello\r\n3\r\nabc\r\n0\r\n\r\n";
+  auto r = cs.scan(body);
+  CHECK(r.complete);
+  std::string payload;
+  for (auto& [off, len] : r.payload) payload += body.substr(off, len);
+  CHECK(payload == "helloabc");
+  // byte-at-a-time feed (split across TCP segments)
+  ChunkedScanner cs2;
+  payload.clear();
+  bool complete = false;
+  for (char c : body) {
+    auto rr = cs2.scan(std::string_view(&c, 1));
+    for (auto& [off, len] : rr.payload) payload += c;
+    if (rr.complete) complete = true;
+  }
+  CHECK(complete && payload == "helloabc");
+  // trailers
+  ChunkedScanner cs3;
+  std::string wt = "2\r\nhi\r\n0\r\nX-T: v\r\n\r\n";
+  auto r3 = cs3.scan(wt);
+  CHECK(r3.complete);
+}
+
+static void test_sse_tap() {
+  SseTap tap;
+  tap.feed("data: {\"choices\":[{\"delta\":{\"content\":\"a\"}}]}\n\n");
+  tap.feed("data: {\"choices\":[{\"delta\":{\"con");  // split line
+  tap.feed("tent\":\"b\"}}]}\n\n");
+  tap.feed("data: {\"usage\":{\"prompt_tokens\":7,"
+           "\"completion_tokens\":42}}\n\ndata: [DONE]\n\n");
+  CHECK(tap.completion_tokens() == 42);
+  CHECK(tap.prompt_tokens() == 7);
+  CHECK(tap.saw_done());
+  // fallback estimate without usage
+  SseTap tap2;
+  for (int i = 0; i < 5; ++i)
+    tap2.feed("data: {\"choices\":[{\"delta\":{\"content\":\"x\"}}]}\n\n");
+  CHECK(tap2.completion_tokens() == 5);
+  // Responses-API usage field names
+  SseTap tap3;
+  tap3.feed("data: {\"type\":\"response.completed\",\"response\":{"
+            "\"usage\":{\"input_tokens\":3,\"output_tokens\":9}}}\n\n");
+  CHECK(tap3.completion_tokens() == 9 && tap3.prompt_tokens() == 3);
+}
+
+static void test_tps_and_selection() {
+  // EMA alpha 0.2 (balancer/types.rs:102-118)
+  TpsState st;
+  st.update(100, 1.0, 0.2, 1000);
+  CHECK(st.score(1000) == 100.0);
+  st.update(200, 1.0, 0.2, 2000);
+  CHECK(st.score(2000) > 119.9 && st.score(2000) < 120.1);
+  CHECK(st.score(2000 + 121000) == 0.0);  // stale after 120 s
+  st.reset();
+  CHECK(st.score(2000) == 0.0);
+
+  // selection: TPS-desc with RR tie-break + admission
+  auto mk = [](const char* id) {
+    auto ep = std::make_shared<Endpoint>();
+    ep->id = id;
+    ep->status = EndpointStatus::kOnline;
+    return ep;
+  };
+  RouteTable table;
+  ModelRoute route;
+  route.model_key = "m";
+  route.canonical = "m";
+  route.rr = std::make_shared<std::atomic<uint64_t>>(0);
+  for (const char* id : {"e1", "e2", "e3"}) {
+    route.endpoints.push_back(mk(id));
+    route.upstream_ids.push_back("m");
+    route.tps.push_back(std::make_shared<TpsState>());
+    route.caps |= kCapChat;
+  }
+  table.by_model["m"] = route;
+  AdmissionConfig adm;
+  // all scores 0 -> pure round robin over all three
+  std::string seen;
+  for (int i = 0; i < 6; ++i)
+    seen += select_endpoint(table, "m", ApiKind::kChat, adm, 5000).ep->id;
+  CHECK(seen.find("e1") != std::string::npos &&
+        seen.find("e2") != std::string::npos &&
+        seen.find("e3") != std::string::npos);
+  // e2 earns the best TPS -> always selected
+  route.tps[1]->update(500, 1.0, 0.2, 5000);
+  table.by_model["m"] = route;
+  for (int i = 0; i < 4; ++i)
+    CHECK(select_endpoint(table, "m", ApiKind::kChat, adm, 5000).ep->id ==
+          "e2");
+  // admission: e2 beyond the 80% reject threshold -> next best
+  route.endpoints[1]->active = 90;  // of max 100
+  CHECK(select_endpoint(table, "m", ApiKind::kChat, adm, 5000).ep->id !=
+        "e2");
+  // HBM gate: e1+e3 over budget and e2 busy -> 429
+  for (auto& ep : route.endpoints) {
+    std::lock_guard<std::mutex> lk(ep->gpu_mu);
+    ep->gpu.hbm_total_gb = 288;
+    ep->gpu.hbm_used_gb = 280;  // 97% > 92% budget
+    ep->gpu.updated_ms = 5000;
+  }
+  auto res = select_endpoint(table, "m", ApiKind::kChat, adm, 5000);
+  CHECK(res.err_status == 429);
+  // unknown model -> 404
+  CHECK(select_endpoint(table, "nope", ApiKind::kChat, adm, 5000)
+            .err_status == 404);
+  // offline endpoints -> 503
+  for (auto& ep : route.endpoints) ep->status = EndpointStatus::kOffline;
+  table.by_model["m"] = route;
+  CHECK(select_endpoint(table, "m", ApiKind::kChat, adm, 5000)
+            .err_status == 503);
+}
+
+static void test_lease() {
+  auto ep = std::make_shared<Endpoint>();
+  ep->status = EndpointStatus::kOnline;
+  {
+    RequestLease lease(ep, nullptr, "m", ApiKind::kChat);
+    CHECK(ep->active.load() == 1);
+    lease.complete_success(10);
+    CHECK(ep->active.load() == 0);
+    CHECK(ep->consecutive_failures.load() == 0);
+  }
+  CHECK(ep->total_failures.load() == 0);
+  {
+    RequestLease lease(ep, nullptr, "m", ApiKind::kChat);
+    // dropped without completion -> counted as failure (lease.rs:71)
+  }
+  CHECK(ep->active.load() == 0);
+  CHECK(ep->total_failures.load() == 1);
+  // move semantics: only the final owner releases
+  {
+    RequestLease a(ep, nullptr, "m", ApiKind::kChat);
+    RequestLease b = std::move(a);
+    CHECK(ep->active.load() == 1);
+    b.complete_success(1);
+  }
+  CHECK(ep->active.load() == 0);
+}
+
+static void test_caps_csv() {
+  CHECK(caps_from_csv("chat,embeddings") == (kCapChat | kCapEmbeddings));
+  CHECK(caps_to_csv(kCapAudioSpeech) == "audio_speech");
+  CHECK(caps_from_csv(caps_to_csv(kCapChat | kCapImageGen)) ==
+        (kCapChat | kCapImageGen));
+}
+
+int main() {
+  test_sha256();
+  test_b64();
+  test_json();
+  test_http_parse();
+  test_chunked_scanner();
+  test_sse_tap();
+  test_tps_and_selection();
+  test_lease();
+  test_caps_csv();
+  if (g_failures == 0) {
+    printf("llmlb-test: all native unit tests passed\n");
+    return 0;
+  }
+  fprintf(stderr, "llmlb-test: %d failure(s)\n", g_failures);
+  return 1;
+}
