@@ -205,7 +205,7 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         return torch.nn.functional.linear(x, weight)
     N, K = x.shape
     M = weight.shape[0]
-    legal = N <= 128 and K % 256 == 0 and M % 128 == 0
+    legal = N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 4096
     if legal and mode == "auto":
         mblocks = M // 128
         legal = (mblocks < 224 and K >= 8192) or N <= 16
