@@ -74,6 +74,7 @@ def cmd_serve(args) -> int:
     admin_token = env.setdefault("LLMLB_ADMIN_TOKEN",
                                  os.urandom(12).hex())
     cp_port = args.control_plane_port
+    env.setdefault("LLMLB_LOG_DIR", "logs")
     env.update({
         "LLMLB_PORT": str(args.port),
         "LLMLB_DATABASE_URL": args.db,
