@@ -53,6 +53,16 @@ def ensure_gateway_binaries() -> Path:
     return bin_dir
 
 
+def ensure_kernels() -> None:
+    lib = REPO / "llmlb_amd" / "ops" / "libllmlb_kernels.so"
+    if not lib.exists():
+        srcs = sorted((REPO / "llmlb_amd/ops/csrc").glob("*.hip"))
+        subprocess.run(["/opt/rocm/bin/hipcc", "--offload-arch=gfx950",
+                        "-O3", "-std=c++17", "-shared", "-fPIC",
+                        *[str(s) for s in srcs], "-o", str(lib)],
+                       check=True)
+
+
 # ---------------------------------------------------------------------
 # async load generator (rank 0): drives the gateway with streaming chat
 # ---------------------------------------------------------------------
@@ -223,6 +233,8 @@ def main() -> None:
             torch.cuda.synchronize()
 
     # ---- backend replica on every rank ------------------------------
+    if have_gpu:
+        ensure_kernels()
     from llmlb_amd.backend.launcher import build_engine
     from llmlb_amd.backend.server import BackendServer, build_asgi_app
     import uvicorn
