@@ -313,47 +313,26 @@ void paged_attn_decode_kernel(
 #pragma unroll
     for (int i = 0; i < 32; ++i) acc[i] = 0.f;
 
-    // K fragments register-double-buffered: the next page's K loads
-    // issue before the current page's V accumulation so the L2/HBM
-    // latency hides under the softmax + V work (per-wave pipelining;
-    // the serial m/l chain otherwise leaves every load exposed).
-    uint4 kreg[4];
-    {
-      const int blk0 = (blk_begin < blk_end) ? btab[blk_begin] : 0;
-      const bf16* kbase0 =
-          kcache + (((size_t)blk0 * n_kv + kvh) * PA_BLOCK + t_local) * D +
-          p * DS;
-      const uint4* kv4 = reinterpret_cast<const uint4*>(kbase0);
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-        if (i * 8 < DS) kreg[i] = kv4[i];
-    }
-
     for (int b = blk_begin; b < blk_end; ++b) {
       const int blk_id = btab[b];
       const int tok = b * PA_BLOCK + t_local;
       const bool valid = tok < seq_len;
+      const bf16* kbase =
+          kcache + (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + t_local) * D +
+          p * DS;
       // score: dot(q_slice, k_slice) then reduce over the 4-lane group
       float s = 0.f;
+      {
+        const uint4* kv4 = reinterpret_cast<const uint4*>(kbase);
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        if (i * 8 < DS) {
-          float f[8];
-          unpack8(kreg[i], f);
+        for (int i = 0; i < 4; ++i) {
+          if (i * 8 < DS) {
+            float f[8];
+            unpack8(kv4[i], f);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) s += qf[i * 8 + j] * f[j];
+            for (int j = 0; j < 8; ++j) s += qf[i * 8 + j] * f[j];
+          }
         }
-      }
-      // prefetch next page's K while the softmax/V phase runs
-      if (b + 1 < blk_end) {
-        const int nblk = btab[b + 1];
-        const bf16* kn =
-            kcache + (((size_t)nblk * n_kv + kvh) * PA_BLOCK + t_local) * D +
-            p * DS;
-        const uint4* kv4n = reinterpret_cast<const uint4*>(kn);
-#pragma unroll
-        for (int i = 0; i < 4; ++i)
-          if (i * 8 < DS) kreg[i] = kv4n[i];
       }
       s += __shfl_xor(s, 1, WAVE);
       s += __shfl_xor(s, 2, WAVE);
