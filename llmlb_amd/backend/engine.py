@@ -104,7 +104,7 @@ class _DecodeGraph:
         # fixed partition count per bucket (fills the 256-CU chip)
         base = S * engine.n_kv_local
         n_parts = 1
-        while base * n_parts < 512 and n_parts < 16:
+        while base * n_parts < 1024 and n_parts < 32:
             n_parts *= 2
         self.n_parts = n_parts
         nq, D = engine.n_q_local, engine.cfg.model.head_dim

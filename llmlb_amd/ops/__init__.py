@@ -239,7 +239,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         # fill ≫256 workgroups on the 256-CU chip (8 XCDs)
         base = S * n_kv
         n_parts = 1
-        while base * n_parts < 512 and n_parts < 16:
+        while base * n_parts < 1024 and n_parts < 32:
             n_parts *= 2
     out = torch.empty_like(q)
     if n_parts > 1:
