@@ -104,14 +104,17 @@ class _DecodeGraph:
         # fixed partition count per bucket (fills the 256-CU chip)
         base = S * engine.n_kv_local
         n_parts = 1
-        while base * n_parts < 1024 and n_parts < 32:
+        while base * n_parts < 512 and n_parts < 16:
             n_parts *= 2
         self.n_parts = n_parts
         nq, D = engine.n_q_local, engine.cfg.model.head_dim
-        if n_parts > 1:
-            self.ws = (torch.empty(S, nq, n_parts, D, dtype=torch.float32,
+        group = nq // engine.n_kv_local
+        use_v2 = ops._PAGED_ATTN_V2 and D == 128 and group <= 4
+        slices = n_parts * 4 if use_v2 else n_parts
+        if slices > 1 or use_v2:
+            self.ws = (torch.empty(S, nq, slices, D, dtype=torch.float32,
                                    device=dev),
-                       torch.empty(S, nq, n_parts, 2, dtype=torch.float32,
+                       torch.empty(S, nq, slices, 2, dtype=torch.float32,
                                    device=dev))
         else:
             self.ws = None

@@ -217,6 +217,10 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, weight)
 
 
+_PAGED_ATTN_V2 = os.environ.get("LLMLB_PAGED_ATTN_V2",
+                                "1").lower() not in ("0", "false", "off")
+
+
 def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
                       vcache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float,
@@ -239,16 +243,20 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         # fill ≫256 workgroups on the 256-CU chip (8 XCDs)
         base = S * n_kv
         n_parts = 1
-        while base * n_parts < 1024 and n_parts < 32:
+        while base * n_parts < 512 and n_parts < 16:
             n_parts *= 2
     out = torch.empty_like(q)
-    if n_parts > 1:
+    group = n_q // n_kv
+    # v2 (GQA-amortized) always reduces through the slice workspace
+    use_v2 = _PAGED_ATTN_V2 and D == 128 and group <= 4
+    slices = n_parts * 4 if use_v2 else n_parts
+    if slices > 1 or use_v2:
         if workspace is not None:
             ws_acc, ws_ml = workspace
         else:
-            ws_acc = torch.empty(S, n_q, n_parts, D, dtype=torch.float32,
+            ws_acc = torch.empty(S, n_q, slices, D, dtype=torch.float32,
                                  device=q.device)
-            ws_ml = torch.empty(S, n_q, n_parts, 2, dtype=torch.float32,
+            ws_ml = torch.empty(S, n_q, slices, 2, dtype=torch.float32,
                                 device=q.device)
         acc_p, ml_p = _ptr(ws_acc), _ptr(ws_ml)
     else:

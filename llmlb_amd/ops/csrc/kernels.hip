@@ -408,6 +408,175 @@ void paged_attn_decode_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Paged decode attention v2 — GQA-amortized (group <= 4, D == 128).
+//
+// v1 gives each wave ONE query head, so the 4 waves of a workgroup
+// each stream the same K/V pages (4x redundant L1/L2 traffic — the
+// measured ceiling at ctx >= 1k).  v2 gives each wave ALL `group`
+// query heads of its kv-head and splits the PAGES across the 4 waves
+// instead: every page is read once per workgroup.  Layout: 8 lanes per
+// token x 16-element slices (D=128 -> p = lane&7, DS=16); per token
+// the lane computes `group` partial dots from one K slice load.  The
+// per-wave online-softmax state (m shared, l/acc lane-partial) is
+// combined across the 4 waves through LDS at the end (flash-decode
+// within the workgroup), then across grid partitions by the shared
+// paged_attn_reduce_kernel.
+// ---------------------------------------------------------------------------
+template <int GROUP>
+__global__ __launch_bounds__(PA_WAVES * WAVE)
+void paged_attn_decode_v2_kernel(
+    bf16* __restrict__ out,              // [num_seqs, n_q, D]
+    float* __restrict__ ws_acc,          // [num_seqs, n_q, n_parts, D]
+    float* __restrict__ ws_ml,           // [num_seqs, n_q, n_parts, 2]
+    const bf16* __restrict__ q,
+    const bf16* __restrict__ kcache,
+    const bf16* __restrict__ vcache,
+    const int* __restrict__ block_tables,
+    const int* __restrict__ seq_lens,
+    int n_q, int n_kv, int max_blocks, int n_parts, float scale) {
+  constexpr int D = 128;
+  constexpr int DS = 16;                 // dim-slice (8 lanes per token)
+  const int seq = blockIdx.x;
+  const int kvh = blockIdx.y;
+  const int part = blockIdx.z;
+  const int seq_len = seq_lens[seq];
+  const int n_blocks = (seq_len + PA_BLOCK - 1) / PA_BLOCK;
+  // page range of this (partition, wave): partition -> waves
+  const int slices = n_parts * PA_WAVES;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int slice = part * PA_WAVES + wid;
+  const int per_slice = (n_blocks + slices - 1) / slices;
+  const int blk_begin = slice * per_slice;
+  const int blk_end = min(blk_begin + per_slice, n_blocks);
+  const int t_local = lane >> 3;         // token within half-page (0..7)
+  const int p = lane & 7;                // dim slice
+  const int* btab = block_tables + (size_t)seq * max_blocks;
+
+  // q fragments for every head of the group, pre-scaled
+  float qf[GROUP][DS];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    const uint4* qv = reinterpret_cast<const uint4*>(
+        q + ((size_t)seq * n_q + kvh * GROUP + g) * D + p * DS);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      float f[8];
+      unpack8(qv[i], f);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qf[g][i * 8 + j] = f[j] * scale;
+    }
+  }
+  float m[GROUP], l[GROUP], acc[GROUP][DS];
+#pragma unroll
+  for (int g = 0; g < GROUP; ++g) {
+    m[g] = -INFINITY;
+    l[g] = 0.f;
+#pragma unroll
+    for (int i = 0; i < DS; ++i) acc[g][i] = 0.f;
+  }
+
+  for (int b = blk_begin; b < blk_end; ++b) {
+    const int blk_id = btab[b];
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int trow = half * 8 + t_local;
+      const int tok = b * PA_BLOCK + trow;
+      const bool valid = tok < seq_len;
+      const bf16* kb =
+          kcache + (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow) * D +
+          p * DS;
+      float kf[DS];
+      {
+        const uint4* kv4 = reinterpret_cast<const uint4*>(kb);
+        float f[8];
+        unpack8(kv4[0], f);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kf[j] = f[j];
+        unpack8(kv4[1], f);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kf[8 + j] = f[j];
+      }
+      float s[GROUP];
+#pragma unroll
+      for (int g = 0; g < GROUP; ++g) {
+        float acc_s = 0.f;
+#pragma unroll
+        for (int j = 0; j < DS; ++j) acc_s += qf[g][j] * kf[j];
+        // reduce over the 8-lane dim group (xor 1,2,4)
+        acc_s += __shfl_xor(acc_s, 1, WAVE);
+        acc_s += __shfl_xor(acc_s, 2, WAVE);
+        acc_s += __shfl_xor(acc_s, 4, WAVE);
+        s[g] = valid ? acc_s : -INFINITY;
+      }
+      // block max over the 8 tokens of this pass (lane bits 3..5)
+      float vf[DS];
+      {
+        const bf16* vb =
+            vcache +
+            (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow) * D + p * DS;
+        const uint4* vv4 = reinterpret_cast<const uint4*>(vb);
+        float f[8];
+        unpack8(vv4[0], f);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vf[j] = f[j];
+        unpack8(vv4[1], f);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vf[8 + j] = f[j];
+      }
+#pragma unroll
+      for (int g = 0; g < GROUP; ++g) {
+        float bm = s[g];
+        bm = fmaxf(bm, __shfl_xor(bm, 8, WAVE));
+        bm = fmaxf(bm, __shfl_xor(bm, 16, WAVE));
+        bm = fmaxf(bm, __shfl_xor(bm, 32, WAVE));
+        const float m_new = fmaxf(m[g], bm);
+        if (m_new != -INFINITY) {
+          const float rescale =
+              (m[g] == -INFINITY) ? 0.f : __expf(m[g] - m_new);
+          const float wgt = valid ? __expf(s[g] - m_new) : 0.f;
+          l[g] = l[g] * rescale + wgt;
+#pragma unroll
+          for (int i = 0; i < DS; ++i)
+            acc[g][i] = acc[g][i] * rescale + wgt * vf[i];
+          m[g] = m_new;
+        }
+      }
+    }
+  }
+  // reduce l and acc across the 8 token groups (butterfly, bits 3..5)
+#pragma unroll
+  for (int off = 8; off <= 32; off <<= 1) {
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      l[g] += __shfl_xor(l[g], off, WAVE);
+#pragma unroll
+      for (int i = 0; i < DS; ++i)
+        acc[g][i] += __shfl_xor(acc[g][i], off, WAVE);
+    }
+  }
+  // each wave is its own flash-decode slice: write (m, l, acc) to the
+  // grid workspace at slice granularity; the reduce kernel combines.
+  if (t_local == 0) {
+#pragma unroll
+    for (int g = 0; g < GROUP; ++g) {
+      const int qh = kvh * GROUP + g;
+      float* wa = ws_acc +
+          (((size_t)seq * n_q + qh) * slices + slice) * D + p * DS;
+#pragma unroll
+      for (int i = 0; i < DS; ++i) wa[i] = acc[g][i];
+      if (p == 0) {
+        float* wm =
+            ws_ml + (((size_t)seq * n_q + qh) * slices + slice) * 2;
+        wm[0] = m[g];
+        wm[1] = l[g];
+      }
+    }
+  }
+}
+
 // combine partial (m, l, acc) across partitions: one wave per (seq, q-head)
 __global__ __launch_bounds__(WAVE)
 void paged_attn_reduce_kernel(bf16* __restrict__ out,
@@ -415,6 +584,8 @@ void paged_attn_reduce_kernel(bf16* __restrict__ out,
                               const float* __restrict__ ws_ml,
                               const int* __restrict__ seq_lens,
                               int n_q, int D, int n_parts) {
+  // n_parts here = total slices in the workspace (grid parts, or
+  // grid parts x PA_WAVES for the v2 kernel)
   const int seq = blockIdx.x;
   const int qh = blockIdx.y;
   const int lane = threadIdx.x;
@@ -535,6 +706,56 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
   if (block_size != PA_BLOCK || D > 128 || D % 32 != 0) return -2;
   if (n_q % n_kv != 0) return -2;
   hipStream_t s = (hipStream_t)stream;
+  const int group = n_q / n_kv;
+  // v2 (GQA-amortized, pages split across waves) needs D=128, group<=4
+  // and the slice workspace (always reduced); host passes ws for it.
+  const bool v2 = (D == 128 && group <= 4 && ws_acc != nullptr);
+  if (v2) {
+    dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
+    const int slices = n_parts * PA_WAVES;
+    switch (group) {
+      case 1:
+        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<1>), grid, block,
+                           0, s, (bf16*)out, (float*)ws_acc,
+                           (float*)ws_ml, (const bf16*)q,
+                           (const bf16*)kcache, (const bf16*)vcache,
+                           (const int*)block_tables,
+                           (const int*)seq_lens, n_q, n_kv, max_blocks,
+                           n_parts, scale);
+        break;
+      case 2:
+        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<2>), grid, block,
+                           0, s, (bf16*)out, (float*)ws_acc,
+                           (float*)ws_ml, (const bf16*)q,
+                           (const bf16*)kcache, (const bf16*)vcache,
+                           (const int*)block_tables,
+                           (const int*)seq_lens, n_q, n_kv, max_blocks,
+                           n_parts, scale);
+        break;
+      case 4:
+        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<4>), grid, block,
+                           0, s, (bf16*)out, (float*)ws_acc,
+                           (float*)ws_ml, (const bf16*)q,
+                           (const bf16*)kcache, (const bf16*)vcache,
+                           (const int*)block_tables,
+                           (const int*)seq_lens, n_q, n_kv, max_blocks,
+                           n_parts, scale);
+        break;
+      default:
+        goto v1_path;
+    }
+    LL_CHECK(hipGetLastError());
+    {
+      dim3 rgrid(num_seqs, n_q), rblock(WAVE);
+      hipLaunchKernelGGL(paged_attn_reduce_kernel, rgrid, rblock, 0, s,
+                         (bf16*)out, (const float*)ws_acc,
+                         (const float*)ws_ml, (const int*)seq_lens, n_q,
+                         D, slices);
+      LL_CHECK(hipGetLastError());
+    }
+    return 0;
+  }
+v1_path:
   dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
   hipLaunchKernelGGL(paged_attn_decode_kernel, grid, block, 0, s, (bf16*)out,
                      (float*)ws_acc, (float*)ws_ml, (const bf16*)q,
