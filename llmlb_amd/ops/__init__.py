@@ -270,6 +270,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         _ptr(bt), _ptr(sl), ctypes.c_int(S), ctypes.c_int(bt.shape[1]),
         ctypes.c_int(n_q), ctypes.c_int(n_kv), ctypes.c_int(D),
         ctypes.c_int(block_size), ctypes.c_int(n_parts),
-        ctypes.c_float(scale), _stream())
+        ctypes.c_float(scale), ctypes.c_int(1 if use_v2 else 0),
+        _stream())
     _check(rc, "ll_paged_attn_decode")
     return out

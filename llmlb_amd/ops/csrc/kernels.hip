@@ -702,14 +702,15 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, int n_q, int n_kv,
                          int D, int block_size, int n_parts, float scale,
-                         void* stream) {
+                         int want_v2, void* stream) {
   if (block_size != PA_BLOCK || D > 128 || D % 32 != 0) return -2;
   if (n_q % n_kv != 0) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int group = n_q / n_kv;
-  // v2 (GQA-amortized, pages split across waves) needs D=128, group<=4
-  // and the slice workspace (always reduced); host passes ws for it.
-  const bool v2 = (D == 128 && group <= 4 && ws_acc != nullptr);
+  // v2 (GQA-amortized, pages split across waves): caller must size the
+  // workspace for n_parts*PA_WAVES slices and pass want_v2 explicitly
+  const bool v2 = (want_v2 && D == 128 && group <= 4 &&
+                   ws_acc != nullptr);
   if (v2) {
     dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
     const int slices = n_parts * PA_WAVES;
