@@ -315,6 +315,13 @@ void ClientConn::process_requests() {
       flush_out();
       return;
     }
+    if (!expect_ack_sent && req.content_length > 0 &&
+        ieq(req.header("expect"), "100-continue") &&
+        in.size() < req.header_len + (size_t)req.content_length) {
+      out.append("HTTP/1.1 100 Continue\r\n\r\n");
+      expect_ack_sent = true;
+      flush_out();
+    }
     if ((size_t)req.content_length > w->gw->cfg.body_limit) {
       send_json_error(413, "payload_too_large", "body exceeds limit");
       want_close = true;
@@ -337,6 +344,7 @@ void ClientConn::dispatch() {
 void ClientConn::finish_request_cycle() {
   size_t need = req.header_len + (size_t)req.content_length;
   in.consume(need);
+  expect_ack_sent = false;
   req.headers.clear();  // keep vector capacity across requests
   req.header_len = 0;
   req.content_length = 0;

@@ -197,6 +197,7 @@ class ClientConn : public FdHandler {
   bool have_head = false;
   bool want_close = false;
   bool closed = false;
+  bool expect_ack_sent = false;
   std::unique_ptr<ProxyOp> op;
   uint32_t events = 0;
   bool response_open = false;  // streaming response in progress
