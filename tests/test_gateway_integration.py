@@ -331,3 +331,33 @@ def test_endpoint_failure_ejected_and_recovers(stack):
     assert status in (404, 503), f"endpoint never ejected (last {status})"
     _req("DELETE", stack["base"] + f"/api/endpoints/{ep_id}",
          headers=stack["admin"])
+
+
+def test_expect_100_continue(stack):
+    """curl sends Expect: 100-continue for large bodies and stalls ~1s
+    unless the server acknowledges with an interim 100 response."""
+    host, port = stack["base"].replace("http://", "").split(":")
+    body = json.dumps({"model": "mock-model",
+                       "messages": [{"role": "user",
+                                     "content": "x" * 2048}]}).encode()
+    s = socket.create_connection((host, int(port)), timeout=10)
+    try:
+        head = (f"POST /v1/chat/completions HTTP/1.1\r\n"
+                f"Host: {host}\r\n"
+                f"Authorization: {stack['auth']['Authorization']}\r\n"
+                f"Content-Type: application/json\r\n"
+                f"Expect: 100-continue\r\n"
+                f"Content-Length: {len(body)}\r\n\r\n").encode()
+        s.sendall(head)  # body deliberately withheld
+        s.settimeout(5)
+        interim = s.recv(4096)
+        assert interim.startswith(b"HTTP/1.1 100 Continue"), interim[:60]
+        s.sendall(body)
+        buf = b""
+        while b"\"chat.completion\"" not in buf:
+            chunk = s.recv(65536)
+            assert chunk, buf[:200]
+            buf += chunk
+        assert b" 200 " in buf.split(b"\r\n", 1)[0]
+    finally:
+        s.close()
