@@ -2,6 +2,7 @@
 // api-key management, dashboard summary/stats, request history, GPU
 // telemetry view, audio/images capability routing, drain control.
 #include "gateway.h"
+#include "dashboard_html.h"
 
 namespace gw {
 
@@ -494,7 +495,20 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
     return;
   }
 
-  // dashboard static / ws: control plane
+  // embedded dashboard SPA (reference serves its React build from the
+  // binary via include_dir, api/mod.rs:56,637-676; ours is a single
+  // self-contained page generated into dashboard_html.h at build time).
+  // The page itself is public; every API it calls requires the token.
+  if (method == "GET" &&
+      (path == "/" || path == "/dashboard" ||
+       starts_with(path, "/dashboard/"))) {
+    c->send_simple(200, "text/html; charset=utf-8",
+                   std::string((const char*)kDashboardHtml,
+                               kDashboardHtml_len));
+    return;
+  }
+
+  // ws / anything else under the ops prefix: control plane
   if (!cfg.control_plane_base.empty()) {
     start_control_plane_proxy(c);
     return;

@@ -361,3 +361,15 @@ def test_expect_100_continue(stack):
         assert b" 200 " in buf.split(b"\r\n", 1)[0]
     finally:
         s.close()
+
+
+def test_dashboard_spa_served(stack):
+    """The gateway serves the embedded single-file dashboard at / and
+    /dashboard (reference: include_dir-embedded React SPA)."""
+    for path in ("/", "/dashboard"):
+        status, body = _req("GET", stack["base"] + path)
+        assert status == 200
+        assert "<html" in body and "llmlb" in body
+    # the APIs the page calls stay token-gated
+    status, _ = _req("GET", stack["base"] + "/api/dashboard/summary")
+    assert status in (401, 403)
