@@ -556,6 +556,9 @@ bool Gateway::init(std::string* err) {
   }
   if (!cfg.db_path.empty() && cfg.db_path != ":memory:none") {
     if (!db.open(cfg.db_path, err)) return false;
+    if (cfg.audit_archive_days > 0)
+      history.set_audit_archive(cfg.db_path + ".audit-archive.sqlite3",
+                                cfg.audit_archive_days);
     if (!db.migrate(err)) return false;
   }
   registry.load_from_db();

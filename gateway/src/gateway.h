@@ -30,6 +30,9 @@ struct GatewayConfig {
   size_t body_limit = 20 * 1024 * 1024;   // api/mod.rs:58
   AdmissionConfig admission;
   int retention_days = 7;
+  // audit rows older than this move to <db>.audit-archive.sqlite3
+  // (reference bootstrap.rs:291-318); 0 disables archiving
+  int audit_archive_days = 90;
   int health_interval_s = 30;
   int health_timeout_ms = 5000;
   int auto_sync_interval_s = 900;

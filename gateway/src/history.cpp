@@ -50,6 +50,8 @@ void HistoryWriter::loop() {
                 {(long long)cutoff});
       db_->exec("DELETE FROM health_checks WHERE ts_ms < ?",
                 {(long long)cutoff});
+      if (!archive_path_.empty())
+        archive_audit_now(now - (int64_t)archive_days_ * 86400 * 1000);
     }
   }
 }
@@ -105,6 +107,52 @@ void HistoryWriter::flush(std::vector<RequestRecord>& recs,
   db_->exec("COMMIT", {});
 }
 
+int HistoryWriter::archive_audit_now(int64_t cutoff_ms,
+                                     std::string* err) {
+  if (!db_ || !db_->ok() || archive_path_.empty()) return 0;
+  std::lock_guard<std::mutex> dblk(db_->mutex());
+  if (!db_->exec("ATTACH ? AS audit_archive", {archive_path_})) {
+    if (err) *err = "attach failed: " + archive_path_;
+    return -1;
+  }
+  int moved = -1;
+  do {
+    if (!db_->exec(
+            "CREATE TABLE IF NOT EXISTS audit_archive.audit_log ("
+            "id INTEGER PRIMARY KEY, ts_ms INTEGER, method TEXT, "
+            "path TEXT, status INTEGER, client_ip TEXT, user TEXT, "
+            "duration_ms REAL, prev_hash TEXT, hash TEXT)", {}))
+      break;
+    db_->exec("BEGIN", {});
+    std::vector<DbRow> rows;
+    db_->exec("SELECT COUNT(*), MAX(id) FROM audit_log WHERE ts_ms < ?",
+              {(long long)cutoff_ms}, &rows);
+    long long n = rows.empty() ? 0 : rows[0][0].i;
+    if (n == 0) {
+      db_->exec("COMMIT", {});
+      moved = 0;
+      break;
+    }
+    long long maxid = rows[0][1].i;
+    rows.clear();
+    db_->exec("SELECT hash FROM audit_log WHERE id = ?", {maxid}, &rows);
+    std::string tip = rows.empty() ? std::string() : rows[0][0].str();
+    db_->exec("INSERT INTO audit_archive.audit_log "
+              "SELECT id,ts_ms,method,path,status,client_ip,user,"
+              "duration_ms,prev_hash,hash FROM main.audit_log "
+              "WHERE ts_ms < ?", {(long long)cutoff_ms});
+    db_->exec("DELETE FROM main.audit_log WHERE ts_ms < ?",
+              {(long long)cutoff_ms});
+    db_->exec("INSERT INTO settings (key,value) VALUES "
+              "('audit_anchor',?) ON CONFLICT(key) DO UPDATE SET "
+              "value=excluded.value", {tip});
+    db_->exec("COMMIT", {});
+    moved = (int)n;
+  } while (false);
+  db_->exec("DETACH audit_archive", {});
+  return moved;
+}
+
 bool HistoryWriter::verify_audit_chain(std::string* err) {
   if (!db_ || !db_->ok()) return true;
   std::vector<DbRow> rows;
@@ -114,6 +162,13 @@ bool HistoryWriter::verify_audit_chain(std::string* err) {
                  {}, &rows))
     return false;
   std::string prev = sha256_hex("genesis");
+  {
+    std::vector<DbRow> anchor;
+    db_->exec("SELECT value FROM settings WHERE key = 'audit_anchor'", {},
+              &anchor);
+    if (!anchor.empty() && !anchor[0][0].str().empty())
+      prev = anchor[0][0].str();
+  }
   for (auto& r : rows) {
     if (r[7].str() != prev) {
       if (err) *err = "chain broken: prev_hash mismatch";

@@ -132,8 +132,21 @@ class HistoryWriter {
   }
 
   HistoryRing& ring() { return ring_; }
-  // verify the audit hash chain (reference audit/hash_chain.rs)
+  // verify the audit hash chain (reference audit/hash_chain.rs); when
+  // old rows have been archived, verification restarts from the
+  // anchored tip hash stored in settings ('audit_anchor')
   bool verify_audit_chain(std::string* err);
+
+  // archive audit rows older than cutoff_ms into a second SQLite file
+  // (reference bootstrap.rs:291-318: 90-day archive to a separate DB).
+  // Returns the number of rows moved, -1 on error.  The hash of the
+  // newest archived row is anchored in settings so the live chain
+  // still verifies; the archive itself verifies from genesis.
+  void set_audit_archive(const std::string& path, int days) {
+    archive_path_ = path;
+    archive_days_ = days;
+  }
+  int archive_audit_now(int64_t cutoff_ms, std::string* err = nullptr);
 
  private:
   void loop();
@@ -151,6 +164,8 @@ class HistoryWriter {
   bool stop_ = false;
   std::string last_audit_hash_;
   int64_t last_cleanup_ms_ = 0;
+  std::string archive_path_;
+  int archive_days_ = 90;
 };
 
 }  // namespace gw
