@@ -29,7 +29,7 @@ import torch
 
 from .. import ops
 from ..models.llama import LlamaConfig, LlamaForCausalLM
-from .kv_cache import BLOCK_SIZE, KVCache
+from .kv_cache import BLOCK_SIZE, KVCache, PrefixCache
 from .sampler import SamplingParams, sample
 
 
@@ -50,6 +50,9 @@ class EngineConfig:
     hbm_fraction: float = 0.85             # of free HBM after weights
     max_batch: int = 256
     max_prefill_tokens: int = 8192
+    # automatic prefix caching: page-aligned prompt KV reuse
+    # (kv_cache.PrefixCache); LLMLB_PREFIX_CACHE=0 disables
+    enable_prefix_cache: bool = True
     tp_ctx: object = None
     seed: int = 1234
 
@@ -65,7 +68,7 @@ class GenerationResult:
 class _Sequence:
     __slots__ = ("rid", "prompt_ids", "out_ids", "params", "block_table",
                  "length", "out_queue", "generator", "t_submit", "t_first",
-                 "row")
+                 "row", "cached_len", "hashes")
 
     def __init__(self, rid, prompt_ids, params, out_queue, generator):
         self.rid = rid
@@ -79,6 +82,8 @@ class _Sequence:
         self.t_submit = time.monotonic()
         self.t_first = 0.0
         self.row = -1                        # persistent decode-state row
+        self.cached_len = 0                  # tokens reused from PrefixCache
+        self.hashes: list[bytes] | None = None
 
 
 class _DecodeGraph:
@@ -178,6 +183,10 @@ class Engine:
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)
         self.kv = KVCache(m.layers, num_blocks, n_kv_local, m.head_dim,
                           self.device, dtype)
+        self.prefix = (PrefixCache(self.kv)
+                       if cfg.enable_prefix_cache
+                       and os_env_flag("LLMLB_PREFIX_CACHE", True)
+                       else None)
         self.running: list[_Sequence] = []
         self.waiting: "queue.Queue[_Sequence]" = queue.Queue()
         self._next_rid = 0
@@ -305,14 +314,30 @@ class Engine:
                 seq = self.waiting.get_nowait()
             except queue.Empty:
                 break
-            need = (len(seq.prompt_ids) + BLOCK_SIZE) // BLOCK_SIZE + 1
-            if need > self.kv.free_blocks():
-                self.waiting.put(seq)  # head-of-line wait for pages
-                break
             if len(seq.prompt_ids) > budget and admitted:
                 self.waiting.put(seq)
                 break
-            seq.block_table = self.kv.allocate(need)
+            need = (len(seq.prompt_ids) + BLOCK_SIZE) // BLOCK_SIZE + 1
+            if self.prefix is not None:
+                if seq.hashes is None:
+                    seq.hashes = PrefixCache.chain_hashes(seq.prompt_ids)
+                # never reuse the whole prompt: at least one token must
+                # prefill to produce the first logits
+                usable = (len(seq.prompt_ids) - 1) // BLOCK_SIZE
+                matched = self.prefix.match(seq.hashes[:usable])
+                seq.cached_len = len(matched) * BLOCK_SIZE
+                need_new = need - len(matched)
+                if not self.prefix.ensure_free(need_new):
+                    self.prefix.release_seq(matched)
+                    seq.cached_len = 0
+                    self.waiting.put(seq)  # head-of-line wait for pages
+                    break
+                seq.block_table = matched + self.kv.allocate(need_new)
+            else:
+                if need > self.kv.free_blocks():
+                    self.waiting.put(seq)  # head-of-line wait for pages
+                    break
+                seq.block_table = self.kv.allocate(need)
             seq.row = self._free_rows.pop()
             budget -= len(seq.prompt_ids)
             admitted.append(seq)
@@ -337,19 +362,28 @@ class Engine:
     def _prefill_batch(self, seqs: list[_Sequence]) -> None:
         """Batched prefill, grouped by prompt length (same-length groups
         share one packed forward: GEMMs batched, one SDPA per group)."""
-        groups: dict[int, list[_Sequence]] = defaultdict(list)
+        groups: dict[tuple[int, int], list[_Sequence]] = defaultdict(list)
         for s in seqs:
-            groups[len(s.prompt_ids)].append(s)
+            groups[(len(s.prompt_ids), s.cached_len)].append(s)
         dev = self.device
-        for L, group in groups.items():
+        for (L, P), group in groups.items():
             B = len(group)
-            ids = torch.tensor([s.prompt_ids for s in group],
+            Ls = L - P                       # suffix actually prefilled
+            ids = torch.tensor([s.prompt_ids[P:] for s in group],
                                dtype=torch.long, device=dev).reshape(-1)
-            positions = torch.arange(L, dtype=torch.int32, device=dev) \
+            positions = torch.arange(P, L, dtype=torch.int32, device=dev) \
                 .repeat(B)
             slots = torch.tensor(
-                [slot for s in group for slot in self._slots_for(s, 0, L)],
+                [slot for s in group for slot in self._slots_for(s, P, Ls)],
                 dtype=torch.int32, device=dev)
+            if P:
+                pslots = torch.tensor(
+                    [sl for s in group for sl in self._slots_for(s, 0, P)],
+                    dtype=torch.long, device=dev)
+                pblks, poffs = pslots // BLOCK_SIZE, pslots % BLOCK_SIZE
+                # rectangular causal mask: suffix query i sees keys 0..P+i
+                mask = (torch.arange(L, device=dev)[None, :]
+                        <= torch.arange(P, L, device=dev)[:, None])
             h = self.model.embed_tokens(ids)
             residual = None
             nq, nkv, D = self.n_q_local, self.n_kv_local, \
@@ -358,15 +392,25 @@ class Engine:
                 q, k, v, residual = self.model.layer_attn_prelude(
                     li, h, residual, positions)
                 ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
-                qh = q.view(B, L, nq, D).transpose(1, 2)
-                kh = k.view(B, L, nkv, D).transpose(1, 2)
-                vh = v.view(B, L, nkv, D).transpose(1, 2)
+                qh = q.view(B, Ls, nq, D).transpose(1, 2)
+                kh = k.view(B, Ls, nkv, D)
+                vh = v.view(B, Ls, nkv, D)
+                if P:
+                    # gather the cached prefix K/V pages for this layer
+                    kpre = self.kv.k[li][pblks, :, poffs].view(B, P, nkv, D)
+                    vpre = self.kv.v[li][pblks, :, poffs].view(B, P, nkv, D)
+                    kh = torch.cat([kpre, kh], dim=1)
+                    vh = torch.cat([vpre, vh], dim=1)
+                kh, vh = kh.transpose(1, 2), vh.transpose(1, 2)
                 o = torch.nn.functional.scaled_dot_product_attention(
-                    qh, kh, vh, is_causal=L > 1, enable_gqa=True)
-                attn = o.transpose(1, 2).reshape(B * L, nq, D).contiguous()
+                    qh, kh, vh,
+                    attn_mask=mask if P else None,
+                    is_causal=Ls > 1 if not P else False,
+                    enable_gqa=True)
+                attn = o.transpose(1, 2).reshape(B * Ls, nq, D).contiguous()
                 h, residual = self.model.layer_attn_epilogue(li, attn,
                                                              residual)
-            last_idx = torch.arange(1, B + 1, device=dev) * L - 1
+            last_idx = torch.arange(1, B + 1, device=dev) * Ls - 1
             logits = self.model.final(h[last_idx], residual[last_idx])
             toks = sample(logits, [s.params for s in group],
                           [s.generator for s in group])
@@ -382,6 +426,15 @@ class Engine:
             self._bt_dev[rows] = bt_rows.to(dev)
             self._len_dev[rows] = L
             self._last_tok[rows] = toks
+            if self.prefix is not None:
+                # publish the freshly-filled full prompt pages (they are
+                # immutable from here on: decode writes start in the next
+                # page slot) and account the hit rate
+                for s in group:
+                    for b in range(P // BLOCK_SIZE, L // BLOCK_SIZE):
+                        self.prefix.register(s.hashes[b], s.block_table[b])
+                    self.prefix.hit_tokens += P
+                    self.prefix.prompt_tokens += L
             self.running.extend(group)
             self._rows_dirty = True
             toks_cpu = toks.tolist()
@@ -514,7 +567,10 @@ class Engine:
     def _retire(self, seq: _Sequence) -> None:
         if seq in self.running:
             self.running.remove(seq)
-        self.kv.release(seq.block_table)
+        if self.prefix is not None:
+            self.prefix.release_seq(seq.block_table)
+        else:
+            self.kv.release(seq.block_table)
         seq.block_table = []
         if seq.row >= 0:
             self._free_rows.append(seq.row)

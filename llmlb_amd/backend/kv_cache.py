@@ -7,8 +7,20 @@ per (block, kv_head) at head_dim=128, read 16 B/lane).
 Sized for 288 GB HBM3E per GPU: the engine asks
 ``blocks_for_memory_fraction`` how many pages fit the configured HBM
 budget after weights.
+
+``PrefixCache`` adds automatic prefix caching on top: full prompt
+pages are content-addressed by a chained BLAKE2 hash of their token
+ids, so a request whose prompt shares a page-aligned prefix with an
+earlier one (shared system prompt, few-shot header, chat history)
+reuses those KV pages and prefills only the suffix.  Retired pages
+with no active users park in an LRU pool and are evicted back to the
+free list on demand — with 288 GB of HBM3E the pool is effectively a
+multi-gigabyte prompt cache.
 """
 from __future__ import annotations
+
+import hashlib
+from collections import OrderedDict
 
 import torch
 
@@ -47,3 +59,89 @@ class KVCache:
 
     def release(self, blocks: list[int]) -> None:
         self._free.extend(blocks)
+
+
+class PrefixCache:
+    """Content-addressed reuse of full prompt KV pages.
+
+    Equivalent in purpose to vLLM's automatic prefix caching; the
+    reference gateway has no engine of its own, so this is part of the
+    new MI355X serving backend (BASELINE.json GPU north star).  All
+    bookkeeping is host-side and O(pages); the GPU never sees it.
+    """
+
+    def __init__(self, kv: KVCache):
+        self.kv = kv
+        self.by_hash: dict[bytes, int] = {}
+        self.block_hash: dict[int, bytes] = {}
+        self.refs: dict[int, int] = {}
+        # zero-ref cached pages, oldest first (eviction order)
+        self.lru: "OrderedDict[int, None]" = OrderedDict()
+        self.hit_tokens = 0
+        self.prompt_tokens = 0
+
+    @staticmethod
+    def chain_hashes(prompt_ids: list[int]) -> list[bytes]:
+        """One digest per FULL page, chained on the previous page so a
+        hash identifies the whole prefix up to and including its page."""
+        out: list[bytes] = []
+        h = b"genesis"
+        for b in range(len(prompt_ids) // BLOCK_SIZE):
+            chunk = prompt_ids[b * BLOCK_SIZE:(b + 1) * BLOCK_SIZE]
+            d = hashlib.blake2b(digest_size=16)
+            d.update(h)
+            d.update(b" ".join(str(t).encode() for t in chunk))
+            h = d.digest()
+            out.append(h)
+        return out
+
+    def match(self, hashes: list[bytes]) -> list[int]:
+        """Longest cached prefix; matched pages gain a reference."""
+        blocks: list[int] = []
+        for h in hashes:
+            blk = self.by_hash.get(h)
+            if blk is None:
+                break
+            blocks.append(blk)
+        for blk in blocks:
+            self.refs[blk] = self.refs.get(blk, 0) + 1
+            self.lru.pop(blk, None)
+        return blocks
+
+    def register(self, h: bytes, blk: int) -> None:
+        """Publish a page the owning sequence just filled.  First
+        writer wins; a duplicate page stays private to its sequence."""
+        if h in self.by_hash:
+            return
+        self.by_hash[h] = blk
+        self.block_hash[blk] = h
+        self.refs[blk] = self.refs.get(blk, 0) + 1
+
+    def release_seq(self, blocks: list[int]) -> None:
+        """Drop a retiring sequence's references.  Cached pages park in
+        the LRU pool; private pages return to the free list."""
+        plain: list[int] = []
+        for blk in blocks:
+            if blk in self.block_hash:
+                n = self.refs.get(blk, 1) - 1
+                if n <= 0:
+                    self.refs.pop(blk, None)
+                    self.lru[blk] = None
+                    self.lru.move_to_end(blk)
+                else:
+                    self.refs[blk] = n
+            else:
+                plain.append(blk)
+        if plain:
+            self.kv.release(plain)
+
+    def ensure_free(self, n: int) -> bool:
+        """Evict parked pages until ``n`` blocks are allocatable."""
+        while self.kv.free_blocks() < n and self.lru:
+            blk, _ = self.lru.popitem(last=False)
+            h = self.block_hash.pop(blk, None)
+            if h is not None:
+                self.by_hash.pop(h, None)
+            self.refs.pop(blk, None)
+            self.kv.release([blk])
+        return self.kv.free_blocks() >= n

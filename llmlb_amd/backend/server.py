@@ -107,6 +107,11 @@ class BackendServer:
             "queued_requests": self.engine.waiting.qsize(),
             "tokens_per_sec": round(self.engine.tokens_per_sec_ema, 2),
             "gpu_step_ms": round(self.engine.gpu_step_ms_ema, 3),
+            "prefix_cache": (
+                {"hit_tokens": self.engine.prefix.hit_tokens,
+                 "prompt_tokens": self.engine.prefix.prompt_tokens,
+                 "parked_pages": len(self.engine.prefix.lru)}
+                if self.engine.prefix is not None else None),
             "uptime_s": time.time() - self.start_time,
             **gpu,
         }
