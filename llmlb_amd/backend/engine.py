@@ -53,6 +53,9 @@ class EngineConfig:
     # automatic prefix caching: page-aligned prompt KV reuse
     # (kv_cache.PrefixCache); LLMLB_PREFIX_CACHE=0 disables
     enable_prefix_cache: bool = True
+    # prompts longer than this prefill in chunks (activation-memory
+    # bound for long prompts; keeps decode latency smooth)
+    prefill_chunk: int = 4096
     tp_ctx: object = None
     seed: int = 1234
 
@@ -360,56 +363,28 @@ class Engine:
 
     @torch.inference_mode()
     def _prefill_batch(self, seqs: list[_Sequence]) -> None:
-        """Batched prefill, grouped by prompt length (same-length groups
-        share one packed forward: GEMMs batched, one SDPA per group)."""
+        """Batched prefill, grouped by (prompt length, cached prefix):
+        same-shape groups share one packed forward (GEMMs batched, one
+        SDPA per group).  A group longer than ``prefill_chunk`` runs in
+        chunks — every chunk after the first behaves exactly like a
+        cached-prefix suffix (its keys are already in the paged cache),
+        which bounds activation memory for long prompts and keeps
+        prefill waves from monopolising the GPU."""
         groups: dict[tuple[int, int], list[_Sequence]] = defaultdict(list)
         for s in seqs:
             groups[(len(s.prompt_ids), s.cached_len)].append(s)
         dev = self.device
+        C = max(self.cfg.prefill_chunk, BLOCK_SIZE)
         for (L, P), group in groups.items():
             B = len(group)
-            Ls = L - P                       # suffix actually prefilled
-            ids = torch.tensor([s.prompt_ids[P:] for s in group],
-                               dtype=torch.long, device=dev).reshape(-1)
-            positions = torch.arange(P, L, dtype=torch.int32, device=dev) \
-                .repeat(B)
-            slots = torch.tensor(
-                [slot for s in group for slot in self._slots_for(s, P, Ls)],
-                dtype=torch.int32, device=dev)
-            if P:
-                pslots = torch.tensor(
-                    [sl for s in group for sl in self._slots_for(s, 0, P)],
-                    dtype=torch.long, device=dev)
-                pblks, poffs = pslots // BLOCK_SIZE, pslots % BLOCK_SIZE
-                # rectangular causal mask: suffix query i sees keys 0..P+i
-                mask = (torch.arange(L, device=dev)[None, :]
-                        <= torch.arange(P, L, device=dev)[:, None])
-            h = self.model.embed_tokens(ids)
-            residual = None
-            nq, nkv, D = self.n_q_local, self.n_kv_local, \
-                self.cfg.model.head_dim
-            for li in range(self.cfg.model.layers):
-                q, k, v, residual = self.model.layer_attn_prelude(
-                    li, h, residual, positions)
-                ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
-                qh = q.view(B, Ls, nq, D).transpose(1, 2)
-                kh = k.view(B, Ls, nkv, D)
-                vh = v.view(B, Ls, nkv, D)
-                if P:
-                    # gather the cached prefix K/V pages for this layer
-                    kpre = self.kv.k[li][pblks, :, poffs].view(B, P, nkv, D)
-                    vpre = self.kv.v[li][pblks, :, poffs].view(B, P, nkv, D)
-                    kh = torch.cat([kpre, kh], dim=1)
-                    vh = torch.cat([vpre, vh], dim=1)
-                kh, vh = kh.transpose(1, 2), vh.transpose(1, 2)
-                o = torch.nn.functional.scaled_dot_product_attention(
-                    qh, kh, vh,
-                    attn_mask=mask if P else None,
-                    is_causal=Ls > 1 if not P else False,
-                    enable_gqa=True)
-                attn = o.transpose(1, 2).reshape(B * Ls, nq, D).contiguous()
-                h, residual = self.model.layer_attn_epilogue(li, attn,
-                                                             residual)
+            c0 = P
+            while True:
+                c1 = L if L - c0 <= C else c0 + C
+                h, residual = self._prefill_chunk(group, B, c0, c1)
+                if c1 == L:
+                    break
+                c0 = c1
+            Ls = L - c0                      # tokens in the final chunk
             last_idx = torch.arange(1, B + 1, device=dev) * Ls - 1
             logits = self.model.final(h[last_idx], residual[last_idx])
             toks = sample(logits, [s.params for s in group],
@@ -444,6 +419,57 @@ class Engine:
                     finished.append(s)
             for s in finished:
                 self._retire(s)
+
+    def _prefill_chunk(self, group: list[_Sequence], B: int, P: int,
+                       L: int) -> tuple[torch.Tensor, torch.Tensor]:
+        """One prefill pass over prompt positions [P, L) for a
+        same-shape group whose positions [0, P) already have KV in the
+        paged cache (cached prefix pages and/or earlier chunks).
+        Returns the final (h, residual) of the chunk's tokens."""
+        dev = self.device
+        Ls = L - P
+        ids = torch.tensor([s.prompt_ids[P:L] for s in group],
+                           dtype=torch.long, device=dev).reshape(-1)
+        positions = torch.arange(P, L, dtype=torch.int32, device=dev) \
+            .repeat(B)
+        slots = torch.tensor(
+            [slot for s in group for slot in self._slots_for(s, P, Ls)],
+            dtype=torch.int32, device=dev)
+        if P:
+            pslots = torch.tensor(
+                [sl for s in group for sl in self._slots_for(s, 0, P)],
+                dtype=torch.long, device=dev)
+            pblks, poffs = pslots // BLOCK_SIZE, pslots % BLOCK_SIZE
+            # rectangular causal mask: chunk query i sees keys 0..P+i
+            mask = (torch.arange(L, device=dev)[None, :]
+                    <= torch.arange(P, L, device=dev)[:, None])
+        h = self.model.embed_tokens(ids)
+        residual = None
+        nq, nkv, D = self.n_q_local, self.n_kv_local, \
+            self.cfg.model.head_dim
+        for li in range(self.cfg.model.layers):
+            q, k, v, residual = self.model.layer_attn_prelude(
+                li, h, residual, positions)
+            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+            qh = q.view(B, Ls, nq, D).transpose(1, 2)
+            kh = k.view(B, Ls, nkv, D)
+            vh = v.view(B, Ls, nkv, D)
+            if P:
+                # gather this layer's cached K/V for positions [0, P)
+                kpre = self.kv.k[li][pblks, :, poffs].view(B, P, nkv, D)
+                vpre = self.kv.v[li][pblks, :, poffs].view(B, P, nkv, D)
+                kh = torch.cat([kpre, kh], dim=1)
+                vh = torch.cat([vpre, vh], dim=1)
+            kh, vh = kh.transpose(1, 2), vh.transpose(1, 2)
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qh, kh, vh,
+                attn_mask=mask if P else None,
+                is_causal=Ls > 1 if not P else False,
+                enable_gqa=True)
+            attn = o.transpose(1, 2).reshape(B * Ls, nq, D).contiguous()
+            h, residual = self.model.layer_attn_epilogue(li, attn,
+                                                         residual)
+        return h, residual
 
     def _rows_tensor(self) -> torch.Tensor:
         if self._rows_dirty or self._rows_cache is None:

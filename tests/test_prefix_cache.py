@@ -69,3 +69,16 @@ def test_chain_hash_is_prefix_sensitive():
     assert a[0] == b[0]
     assert a[1] != b[1]
     assert len(PrefixCache.chain_hashes(list(range(15)))) == 0
+
+
+def test_chunked_prefill_matches_unchunked():
+    prompt = list(range(100))
+    ref = _greedy(_engine(prefix=False), prompt, n=6)
+    cfg = EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                       dtype="float32", kv_cache_blocks=64,
+                       enable_prefix_cache=True, prefill_chunk=32, seed=7)
+    eng = Engine(cfg)
+    assert _greedy(eng, prompt, n=6) == ref
+    # chunking composes with a cached prefix (second request hits)
+    assert _greedy(eng, prompt, n=6) == ref
+    assert eng.prefix.hit_tokens > 0
