@@ -1,4 +1,5 @@
-"""Synthetic audio backends: ASR (transcription) and TTS (speech).
+"""Synthetic media backends: ASR (transcription), TTS (speech) and
+image generation.
 
 BASELINE config 5 runs a heterogeneous fleet (8B text + TTS + ASR
 backends).  There is no egress for real Whisper/TTS checkpoints, so
@@ -14,8 +15,13 @@ Both advertise their models via GET /v1/models with names the gateway's
 capability inference maps to audio_transcription / audio_speech
 (health.cpp sync_endpoint_models).
 
+  * image `/v1/images/generations` — returns a deterministic
+    procedurally-generated PNG (b64_json) whose pattern derives from
+    the prompt hash (reference api/images.rs proxies the same surface).
+
   python -m llmlb_amd.backend.audio_server --kind asr --port 9500
   python -m llmlb_amd.backend.audio_server --kind tts --port 9501
+  python -m llmlb_amd.backend.audio_server --kind image --port 9502
 """
 from __future__ import annotations
 
@@ -27,6 +33,7 @@ import math
 import struct
 import time
 import wave
+import zlib
 
 WORDS = ("the quick brown fox jumps over a lazy dog near river stone "
          "light sound model audio gateway stream token").split()
@@ -62,6 +69,27 @@ def synth_wav(text: str) -> bytes:
             frames += struct.pack("<h", s)
         w.writeframes(bytes(frames))
     return buf.getvalue()
+
+
+def synth_png(prompt: str, size: int = 64) -> bytes:
+    """Minimal deterministic PNG (RGB, no deps beyond zlib)."""
+    seed = hashlib.blake2b(prompt.encode(), digest_size=32).digest()
+    raw = bytearray()
+    for y in range(size):
+        raw.append(0)  # filter: none
+        for x in range(size):
+            raw.append((x * 4 + seed[y % 32]) & 0xFF)
+            raw.append((y * 4 + seed[x % 32]) & 0xFF)
+            raw.append((seed[(x ^ y) % 32]) & 0xFF)
+
+    def chunk(tag: bytes, data: bytes) -> bytes:
+        return (struct.pack(">I", len(data)) + tag + data
+                + struct.pack(">I", zlib.crc32(tag + data) & 0xFFFFFFFF))
+
+    ihdr = struct.pack(">IIBBBBB", size, size, 8, 2, 0, 0, 0)
+    return (b"\x89PNG\r\n\x1a\n" + chunk(b"IHDR", ihdr)
+            + chunk(b"IDAT", zlib.compress(bytes(raw)))
+            + chunk(b"IEND", b""))
 
 
 def build_app(kind: str, model_name: str):
@@ -116,6 +144,22 @@ def build_app(kind: str, model_name: str):
             await send_bytes(send, 200, "application/json",
                              json.dumps({"text": text}).encode())
             return
+        if kind == "image" and path == "/v1/images/generations":
+            import base64
+            payload = await read_body(receive)
+            try:
+                req = json.loads(payload or b"{}")
+            except json.JSONDecodeError:
+                req = {}
+            n = int(req.get("n", 1) or 1)
+            prompt = str(req.get("prompt", ""))
+            data = [{"b64_json": base64.b64encode(
+                        synth_png(f"{prompt}#{i}")).decode()}
+                    for i in range(min(n, 4))]
+            body = json.dumps({"created": int(time.time()),
+                               "data": data}).encode()
+            await send_bytes(send, 200, "application/json", body)
+            return
         if kind == "tts" and path == "/v1/audio/speech":
             payload = await read_body(receive)
             try:
@@ -133,13 +177,16 @@ def build_app(kind: str, model_name: str):
 def main(argv=None):
     import uvicorn
     ap = argparse.ArgumentParser()
-    ap.add_argument("--kind", choices=("asr", "tts"), required=True)
+    ap.add_argument("--kind", choices=("asr", "tts", "image"),
+                    required=True)
     ap.add_argument("--model", default=None)
     ap.add_argument("--host", default="0.0.0.0")
     ap.add_argument("--port", type=int, default=9500)
     args = ap.parse_args(argv)
-    model = args.model or ("whisper-large-v3-synth" if args.kind == "asr"
-                           else "kokoro-tts-synth")
+    model = args.model or {
+        "asr": "whisper-large-v3-synth",
+        "tts": "kokoro-tts-synth",
+        "image": "stable-diffusion-xl-synth"}[args.kind]
     uvicorn.run(build_app(args.kind, model), host=args.host,
                 port=args.port, log_level="warning", access_log=False)
 

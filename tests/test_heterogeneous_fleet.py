@@ -59,7 +59,7 @@ def fleet():
         subprocess.run(["make", "-C", str(REPO / "gateway"), "-j", "4"],
                        check=True, stdout=subprocess.DEVNULL)
     ports = {k: _free_port() for k in
-             ("text_ok", "text_bad", "asr", "tts", "gw")}
+             ("text_ok", "text_bad", "asr", "tts", "image", "gw")}
     procs = [
         subprocess.Popen([str(BIN / "llmlb-mock"), "--port",
                           str(ports["text_ok"]), "--workers", "1",
@@ -81,6 +81,11 @@ def fleet():
                           "tts", "--port", str(ports["tts"])],
                          cwd=str(REPO), stdout=subprocess.DEVNULL,
                          stderr=subprocess.DEVNULL),
+        subprocess.Popen([sys.executable, "-m",
+                          "llmlb_amd.backend.audio_server", "--kind",
+                          "image", "--port", str(ports["image"])],
+                         cwd=str(REPO), stdout=subprocess.DEVNULL,
+                         stderr=subprocess.DEVNULL),
     ]
     db = tempfile.mktemp(suffix=".sqlite3")
     import os
@@ -95,10 +100,10 @@ def fleet():
     procs.append(gw)
     base = f"http://127.0.0.1:{ports['gw']}"
     admin = {"Authorization": "Bearer admintok"}
-    for p in ("text_ok", "text_bad", "asr", "tts"):
+    for p in ("text_ok", "text_bad", "asr", "tts", "image"):
         assert _wait_port(ports[p]), f"{p} backend failed to start"
     assert _wait_port(ports["gw"])
-    for name in ("text_ok", "text_bad", "asr", "tts"):
+    for name in ("text_ok", "text_bad", "asr", "tts", "image"):
         status, body = _req("POST", base + "/api/endpoints",
                             {"name": name,
                              "base_url":
@@ -145,6 +150,22 @@ def test_tts_routing(fleet):
     assert status == 200
     assert body[:4] == b"RIFF"  # WAV container
     assert len(body) > 4000
+
+
+def test_image_generation_routing(fleet):
+    """/v1/images/generations routes by capability to the synthetic
+    image backend and returns a decodable PNG (reference api/images.rs
+    proxies the same surface)."""
+    import base64
+    status, body = _req("POST",
+                        fleet["base"] + "/v1/images/generations",
+                        {"model": "stable-diffusion-xl-synth",
+                         "prompt": "a red square", "n": 2})
+    assert status == 200, body
+    data = json.loads(body)["data"]
+    assert len(data) == 2
+    png = base64.b64decode(data[0]["b64_json"])
+    assert png.startswith(b"\x89PNG")
 
 
 def test_failure_injection_tps_steering(fleet):
