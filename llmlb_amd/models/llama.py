@@ -90,9 +90,13 @@ class LlamaDecoderLayer(nn.Module):
         T = qkv.shape[0]
         q, k, v = qkv.split(
             [self.q_local * D, self.kv_local * D, self.kv_local * D], dim=-1)
-        return (q.view(T, self.q_local, D).contiguous(),
-                k.view(T, self.kv_local, D).contiguous(),
-                v.view(T, self.kv_local, D).contiguous())
+        # strided views into the fused qkv output — rope / kv_append /
+        # paged attention consume them with an explicit row stride, so
+        # no .contiguous() copies (3 kernels/layer saved on the decode
+        # hot path; profiles/r01_decode_gemm_analysis.md "elementwise")
+        return (q.view(T, self.q_local, D),
+                k.view(T, self.kv_local, D),
+                v.view(T, self.kv_local, D))
 
 
 class LlamaForCausalLM(nn.Module):

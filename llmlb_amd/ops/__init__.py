@@ -119,13 +119,17 @@ def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
     if not q.is_cuda or q.dtype != torch.bfloat16:
         return reference.rope_apply(q, k, cos_sin, positions)
     lib = _require_native()
-    assert q.is_contiguous() and k.is_contiguous()
+    # q/k may be strided views into the fused qkv projection (head dim
+    # contiguous, head stride == D): no .contiguous() copies needed
+    assert q.stride(-1) == 1 and q.stride(1) == q.shape[2]
+    assert k.stride(-1) == 1 and k.stride(1) == k.shape[2]
     assert cos_sin.dtype == torch.float32
     pos = positions.to(torch.int32)
     rc = lib.ll_rope(
         _ptr(q), _ptr(k), _ptr(cos_sin), _ptr(pos),
         ctypes.c_int(q.shape[0]), ctypes.c_int(q.shape[1]),
-        ctypes.c_int(k.shape[1]), ctypes.c_int(q.shape[2]), _stream())
+        ctypes.c_int(k.shape[1]), ctypes.c_int(q.shape[2]),
+        ctypes.c_int(q.stride(0)), ctypes.c_int(k.stride(0)), _stream())
     _check(rc, "ll_rope")
     return q, k
 
@@ -155,12 +159,14 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
         reference.kv_append(kcache, vcache, k, v, slots, block_size)
         return
     lib = _require_native()
-    assert k.is_contiguous() and v.is_contiguous()
+    assert k.stride(-1) == 1 and k.stride(1) == k.shape[2]
+    assert v.stride(-1) == 1 and v.stride(1) == v.shape[2]
     slots_i = slots.to(torch.int32)
     rc = lib.ll_kv_append(
         _ptr(kcache), _ptr(vcache), _ptr(k), _ptr(v), _ptr(slots_i),
         ctypes.c_int(k.shape[0]), ctypes.c_int(k.shape[1]),
-        ctypes.c_int(k.shape[2]), ctypes.c_int(block_size), _stream())
+        ctypes.c_int(k.shape[2]), ctypes.c_int(block_size),
+        ctypes.c_int(k.stride(0)), ctypes.c_int(v.stride(0)), _stream())
     _check(rc, "ll_kv_append")
 
 
@@ -245,7 +251,8 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         n_parts = 1
         while base * n_parts < 512 and n_parts < 16:
             n_parts *= 2
-    out = torch.empty_like(q)
+    assert q.stride(-1) == 1 and q.stride(1) == D
+    out = torch.empty(S, n_q, D, dtype=q.dtype, device=q.device)
     group = n_q // n_kv
     # v2 (GQA-amortized) always reduces through the slice workspace
     use_v2 = _PAGED_ATTN_V2 and D == 128 and group <= 4
@@ -271,6 +278,6 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         ctypes.c_int(n_q), ctypes.c_int(n_kv), ctypes.c_int(D),
         ctypes.c_int(block_size), ctypes.c_int(n_parts),
         ctypes.c_float(scale), ctypes.c_int(1 if use_v2 else 0),
-        _stream())
+        ctypes.c_int(q.stride(0)), _stream())
     _check(rc, "ll_paged_attn_decode")
     return out

@@ -137,7 +137,8 @@ __global__ void rmsnorm_kernel(bf16* __restrict__ out,
 __global__ void rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
                             const float* __restrict__ cos_sin,
                             const int* __restrict__ positions,
-                            int n_tokens, int n_q, int n_kv, int D) {
+                            int n_tokens, int n_q, int n_kv, int D,
+                            int q_rstride, int k_rstride) {
   const int half = D / 2;
   const int heads = n_q + n_kv;
   // one thread handles 4 rotation pairs (8 bf16 = 16B from each half)
@@ -152,8 +153,9 @@ __global__ void rope_kernel(bf16* __restrict__ q, bf16* __restrict__ k,
   const int j0 = (rem % threads_per_head) * pairs_per_thread;
   const int pos = positions[t];
 
-  bf16* base = (h < n_q) ? (q + ((size_t)t * n_q + h) * D)
-                         : (k + ((size_t)t * n_kv + (h - n_q)) * D);
+  bf16* base = (h < n_q)
+      ? (q + (size_t)t * q_rstride + (size_t)h * D)
+      : (k + (size_t)t * k_rstride + (size_t)(h - n_q) * D);
   // rotate-half: (x1, x2) at (j, j+half) -> (x1*c - x2*s, x2*c + x1*s)
   float x1[4], x2[4];
   uint2 lo = *reinterpret_cast<uint2*>(base + j0);
@@ -219,7 +221,8 @@ __global__ void kv_append_kernel(bf16* __restrict__ kcache,
                                  const bf16* __restrict__ v,
                                  const int* __restrict__ slots,
                                  int n_tokens, int n_kv, int D,
-                                 int block_size) {
+                                 int block_size, int k_rstride,
+                                 int v_rstride) {
   const int nvec = D >> 3;
   const long long total = (long long)n_tokens * n_kv * nvec;
   const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -232,12 +235,13 @@ __global__ void kv_append_kernel(bf16* __restrict__ kcache,
   if (slot < 0) return;
   const int blk = slot / block_size;
   const int off = slot % block_size;
-  const size_t src = ((size_t)t * n_kv + h) * D;
   const size_t dst = (((size_t)blk * n_kv + h) * block_size + off) * D;
   reinterpret_cast<uint4*>(kcache + dst)[i] =
-      reinterpret_cast<const uint4*>(k + src)[i];
+      reinterpret_cast<const uint4*>(
+          k + (size_t)t * k_rstride + (size_t)h * D)[i];
   reinterpret_cast<uint4*>(vcache + dst)[i] =
-      reinterpret_cast<const uint4*>(v + src)[i];
+      reinterpret_cast<const uint4*>(
+          v + (size_t)t * v_rstride + (size_t)h * D)[i];
 }
 
 // ---------------------------------------------------------------------------
@@ -272,7 +276,7 @@ void paged_attn_decode_kernel(
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     int n_q, int n_kv, int D, int max_blocks,
-    int n_parts, float scale) {
+    int n_parts, float scale, int q_rstride) {
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
   const int part = blockIdx.z;
@@ -296,7 +300,7 @@ void paged_attn_decode_kernel(
     float qf[32];
     {
       const uint4* qv = reinterpret_cast<const uint4*>(
-          q + ((size_t)seq * n_q + qh) * D + p * DS);
+          q + (size_t)seq * q_rstride + (size_t)qh * D + p * DS);
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         if (i * 8 < DS) {
@@ -435,7 +439,8 @@ void paged_attn_decode_v2_kernel(
     const bf16* __restrict__ vcache,
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
-    int n_q, int n_kv, int max_blocks, int n_parts, float scale) {
+    int n_q, int n_kv, int max_blocks, int n_parts, float scale,
+    int q_rstride) {
   constexpr int D = 128;
   constexpr int DS = 16;                 // dim-slice (8 lanes per token)
   const int seq = blockIdx.x;
@@ -460,7 +465,8 @@ void paged_attn_decode_v2_kernel(
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
     const uint4* qv = reinterpret_cast<const uint4*>(
-        q + ((size_t)seq * n_q + kvh * GROUP + g) * D + p * DS);
+        q + (size_t)seq * q_rstride + (size_t)(kvh * GROUP + g) * D
+        + p * DS);
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       float f[8];
@@ -654,7 +660,8 @@ int ll_rmsnorm(void* out, const void* in, const void* weight, void* residual,
 }
 
 int ll_rope(void* q, void* k, const void* cos_sin, const void* positions,
-            int n_tokens, int n_q, int n_kv, int D, void* stream) {
+            int n_tokens, int n_q, int n_kv, int D, int q_rstride,
+            int k_rstride, void* stream) {
   if (D % 16 != 0) return -2;
   const int threads_per_head = (D / 2) / 4;
   const long long total = (long long)n_tokens * (n_q + n_kv) * threads_per_head;
@@ -663,7 +670,7 @@ int ll_rope(void* q, void* k, const void* cos_sin, const void* positions,
   hipLaunchKernelGGL(rope_kernel, dim3((unsigned)grid), dim3(block), 0,
                      (hipStream_t)stream, (bf16*)q, (bf16*)k,
                      (const float*)cos_sin, (const int*)positions, n_tokens,
-                     n_q, n_kv, D);
+                     n_q, n_kv, D, q_rstride, k_rstride);
   LL_CHECK(hipGetLastError());
   return 0;
 }
@@ -684,7 +691,8 @@ int ll_swiglu(void* out, const void* gateup, long long n_tokens, int inter,
 
 int ll_kv_append(void* kcache, void* vcache, const void* k, const void* v,
                  const void* slots, int n_tokens, int n_kv, int D,
-                 int block_size, void* stream) {
+                 int block_size, int k_rstride, int v_rstride,
+                 void* stream) {
   if (D % 8 != 0) return -2;
   const long long total = (long long)n_tokens * n_kv * (D >> 3);
   const int block = 256;
@@ -692,7 +700,7 @@ int ll_kv_append(void* kcache, void* vcache, const void* k, const void* v,
   hipLaunchKernelGGL(kv_append_kernel, dim3((unsigned)grid), dim3(block), 0,
                      (hipStream_t)stream, (bf16*)kcache, (bf16*)vcache,
                      (const bf16*)k, (const bf16*)v, (const int*)slots,
-                     n_tokens, n_kv, D, block_size);
+                     n_tokens, n_kv, D, block_size, k_rstride, v_rstride);
   LL_CHECK(hipGetLastError());
   return 0;
 }
@@ -702,7 +710,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* block_tables, const void* seq_lens,
                          int num_seqs, int max_blocks, int n_q, int n_kv,
                          int D, int block_size, int n_parts, float scale,
-                         int want_v2, void* stream) {
+                         int want_v2, int q_rstride, void* stream) {
   if (block_size != PA_BLOCK || D > 128 || D % 32 != 0) return -2;
   if (n_q % n_kv != 0) return -2;
   hipStream_t s = (hipStream_t)stream;
@@ -722,7 +730,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            (const bf16*)kcache, (const bf16*)vcache,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale);
+                           n_parts, scale, q_rstride);
         break;
       case 2:
         hipLaunchKernelGGL((paged_attn_decode_v2_kernel<2>), grid, block,
@@ -731,7 +739,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            (const bf16*)kcache, (const bf16*)vcache,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale);
+                           n_parts, scale, q_rstride);
         break;
       case 4:
         hipLaunchKernelGGL((paged_attn_decode_v2_kernel<4>), grid, block,
@@ -740,7 +748,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            (const bf16*)kcache, (const bf16*)vcache,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale);
+                           n_parts, scale, q_rstride);
         break;
       default:
         goto v1_path;
@@ -762,7 +770,7 @@ v1_path:
                      (float*)ws_acc, (float*)ws_ml, (const bf16*)q,
                      (const bf16*)kcache, (const bf16*)vcache,
                      (const int*)block_tables, (const int*)seq_lens, n_q,
-                     n_kv, D, max_blocks, n_parts, scale);
+                     n_kv, D, max_blocks, n_parts, scale, q_rstride);
   LL_CHECK(hipGetLastError());
   if (n_parts > 1) {
     dim3 rgrid(num_seqs, n_q), rblock(WAVE);

@@ -158,3 +158,52 @@ def test_decode_gemm_shape_fallback(monkeypatch):
     w = torch.randn(256, 512, dtype=torch.bfloat16, device=DEV)
     out = ops.linear_decode(x, w)
     assert out.shape == (16, 256)
+
+
+@pytest.mark.gpu
+def test_strided_qkv_views_match_contiguous():
+    """rope / kv_append / paged attention consume strided views into the
+    fused qkv projection (no .contiguous() copies on the decode hot
+    path); results must match running the same ops on dense copies."""
+    torch.manual_seed(11)
+    dev = "cuda:0"
+    T, nq, nkv, D, BS = 24, 8, 2, 128, 16
+    width = (nq + 2 * nkv) * D
+    qkv = torch.randn(T, width, dtype=torch.bfloat16, device=dev)
+    qs, ks, vs = qkv.split([nq * D, nkv * D, nkv * D], dim=-1)
+    qs = qs.view(T, nq, D)
+    ks = ks.view(T, nkv, D)
+    vs = vs.view(T, nkv, D)
+    qc, kc, vc = (x.contiguous().clone() for x in (qs, ks, vs))
+
+    from llmlb_amd.ops.reference import build_cos_sin_table
+    cs = build_cos_sin_table(256, D, 1e4, device=dev)
+    pos = torch.arange(T, dtype=torch.int32, device=dev)
+    ops.rope_apply(qs, ks, cs, pos)   # in place on the views
+    ops.rope_apply(qc, kc, cs, pos)
+    assert torch.equal(qs.contiguous(), qc)
+    assert torch.equal(ks.contiguous(), kc)
+
+    nb = 8
+    kcache_s = torch.zeros(nb, nkv, BS, D, dtype=torch.bfloat16,
+                           device=dev)
+    vcache_s = torch.zeros_like(kcache_s)
+    kcache_c = torch.zeros_like(kcache_s)
+    vcache_c = torch.zeros_like(kcache_s)
+    slots = torch.arange(16, 16 + T, dtype=torch.int32, device=dev)
+    ops.kv_append(kcache_s, vcache_s, ks, vs, slots)
+    ops.kv_append(kcache_c, vcache_c, kc, vc, slots)
+    assert torch.equal(kcache_s, kcache_c)
+    assert torch.equal(vcache_s, vcache_c)
+
+    # one decode token per "sequence": reuse rows of qs as queries
+    S = 4
+    bt = torch.tensor([[1, 2], [2, 3], [3, 4], [4, 5]],
+                      dtype=torch.int32, device=dev)
+    sl = torch.tensor([20, 24, 18, 30], dtype=torch.int32, device=dev)
+    sl = torch.clamp(sl, max=2 * BS)
+    out_s = ops.paged_attn_decode(qs[:S], kcache_s, vcache_s, bt, sl,
+                                  D ** -0.5)
+    out_c = ops.paged_attn_decode(qc[:S], kcache_c, vcache_c, bt, sl,
+                                  D ** -0.5)
+    assert torch.equal(out_s, out_c)
