@@ -14,6 +14,8 @@ from __future__ import annotations
 
 from dataclasses import dataclass
 
+import os
+
 import torch
 import torch.nn as nn
 
@@ -21,6 +23,10 @@ from .. import ops
 from ..ops import reference as op_ref
 from ..parallel.tp import (ColumnParallelLinear, RowParallelLinear,
                            TPContext, get_tp_context)
+
+
+_QKV_VIEWS = os.environ.get("LLMLB_QKV_VIEWS", "1").lower() not in (
+    "0", "false", "off")
 
 
 @dataclass
@@ -93,10 +99,15 @@ class LlamaDecoderLayer(nn.Module):
         # strided views into the fused qkv output — rope / kv_append /
         # paged attention consume them with an explicit row stride, so
         # no .contiguous() copies (3 kernels/layer saved on the decode
-        # hot path; profiles/r01_decode_gemm_analysis.md "elementwise")
-        return (q.view(T, self.q_local, D),
-                k.view(T, self.kv_local, D),
-                v.view(T, self.kv_local, D))
+        # hot path; profiles/r01_decode_gemm_analysis.md "elementwise").
+        # LLMLB_QKV_VIEWS=0 restores the copies (A/B baseline).
+        if _QKV_VIEWS:
+            return (q.view(T, self.q_local, D),
+                    k.view(T, self.kv_local, D),
+                    v.view(T, self.kv_local, D))
+        return (q.view(T, self.q_local, D).contiguous(),
+                k.view(T, self.kv_local, D).contiguous(),
+                v.view(T, self.kv_local, D).contiguous())
 
 
 class LlamaForCausalLM(nn.Module):
