@@ -189,8 +189,7 @@ class LlamaForCausalLM(nn.Module):
         normed, residual = ops.rmsnorm(o, layer.post_norm_w,
                                        self.cfg.norm_eps, residual)
         gateup = layer.gate_up(normed)
-        act = ops.swiglu(gateup)
-        h = layer.down(act)
+        h = layer.down.forward_swiglu(gateup)
         return h, residual
 
     def final(self, h: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:

@@ -189,7 +189,7 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     rc = lib.ll_decode_gemm_v2(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
         _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
-        _stream())
+        ctypes.c_int(0), _stream())
     if rc == -2:
         raise ValueError("unsupported shape for decode_gemm_v2")
     _check(rc, "ll_decode_gemm_v2")
@@ -221,6 +221,39 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         except ValueError:
             pass
     return torch.nn.functional.linear(x, weight)
+
+
+def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
+    """Fused silu(gate)*up @ W^T: the SwiGLU epilogue is applied while
+    staging x chunks into LDS inside the v2 decode GEMM, removing the
+    standalone swiglu kernel and the activation round trip.  Returns
+    None when the fused path does not apply (caller falls back)."""
+    if (_USE_CUSTOM_DECODE_GEMM in ("0", "false", "off")
+            or not gateup.is_cuda or _lib is None
+            or gateup.dtype != torch.bfloat16 or gateup.dim() != 2
+            or not gateup.is_contiguous()):
+        return None
+    N, K2 = gateup.shape
+    K = K2 // 2
+    M = weight.shape[0]
+    if weight.shape[1] != K:
+        return None
+    # same measured policy as linear_decode's down-projection arm
+    if not (N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 8192
+            and (M // 128) < 224):
+        return None
+    lib = _require_native()
+    out = torch.empty(N, M, dtype=gateup.dtype, device=gateup.device)
+    ws = torch.empty(N, M, dtype=torch.float32, device=gateup.device) \
+        if (M // 128) < 224 else None
+    rc = lib.ll_decode_gemm_v2(
+        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(gateup),
+        _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
+        ctypes.c_int(1), _stream())
+    if rc == -2:
+        return None
+    _check(rc, "ll_decode_gemm_v2(swiglu)")
+    return out
 
 
 _PAGED_ATTN_V2 = os.environ.get("LLMLB_PAGED_ATTN_V2",

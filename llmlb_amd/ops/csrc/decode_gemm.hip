@@ -33,6 +33,32 @@ typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+__device__ __forceinline__ float dg_bf2f(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = ((unsigned int)u) << 16;
+  return c.f;
+}
+__device__ __forceinline__ unsigned short dg_f2bf(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  unsigned int x = c.i + (0x7FFF + ((c.i >> 16) & 1));
+  return (unsigned short)(x >> 16);
+}
+__device__ __forceinline__ void dg_unpack8(const uint4 v, float* f) {
+  f[0] = dg_bf2f(v.x & 0xffff);  f[1] = dg_bf2f(v.x >> 16);
+  f[2] = dg_bf2f(v.y & 0xffff);  f[3] = dg_bf2f(v.y >> 16);
+  f[4] = dg_bf2f(v.z & 0xffff);  f[5] = dg_bf2f(v.z >> 16);
+  f[6] = dg_bf2f(v.w & 0xffff);  f[7] = dg_bf2f(v.w >> 16);
+}
+__device__ __forceinline__ uint4 dg_pack8(const float* f) {
+  uint4 v;
+  v.x = (unsigned)dg_f2bf(f[0]) | ((unsigned)dg_f2bf(f[1]) << 16);
+  v.y = (unsigned)dg_f2bf(f[2]) | ((unsigned)dg_f2bf(f[3]) << 16);
+  v.z = (unsigned)dg_f2bf(f[4]) | ((unsigned)dg_f2bf(f[5]) << 16);
+  v.w = (unsigned)dg_f2bf(f[6]) | ((unsigned)dg_f2bf(f[7]) << 16);
+  return v;
+}
+
 // N_TILES = N/16 rounded up; one kernel instantiation per bucket.
 // UNROLL = 32-wide k-steps kept in flight per iteration: the B stream
 // rides HBM (~900 cyc latency), so enough B loads must be outstanding
@@ -222,7 +248,7 @@ int ll_decode_gemm(void* out, const void* x, const void* w, int N, int M,
 // a k-offset and differ in token row) land on 16 distinct bank slots.
 // ===========================================================================
 
-template <int N_TILES, bool ATOMIC>
+template <int N_TILES, bool ATOMIC, bool SWIGLU = false>
 __global__ __launch_bounds__(512)
 void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
                            float* __restrict__ out_ws,  // [N, M] (SK>1)
@@ -265,8 +291,26 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
         const int tok = piece / (CHUNK_K * 2 / 16);   // 16B per piece
         const int kb = piece % (CHUNK_K * 2 / 16);    // 16B col index
         const int src_tok = tok < N ? tok : 0;
-        const uint4 v = *reinterpret_cast<const uint4*>(
-            x + (size_t)src_tok * K + k0 + kb * 8);
+        uint4 v;
+        if (SWIGLU) {
+          // x is gateup [N, 2K]: stage silu(gate)*up on the fly
+          // (bit-identical to the standalone swiglu kernel)
+          const bf16* row = x + (size_t)src_tok * 2 * K;
+          const uint4 gv = *reinterpret_cast<const uint4*>(
+              row + k0 + kb * 8);
+          const uint4 uv = *reinterpret_cast<const uint4*>(
+              row + K + k0 + kb * 8);
+          float gf[8], uf[8];
+          dg_unpack8(gv, gf);
+          dg_unpack8(uv, uf);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            gf[j] = gf[j] / (1.f + __expf(-gf[j])) * uf[j];
+          v = dg_pack8(gf);
+        } else {
+          v = *reinterpret_cast<const uint4*>(
+              x + (size_t)src_tok * K + k0 + kb * 8);
+        }
         const int swz_kb = kb ^ (tok & 15);
         *reinterpret_cast<uint4*>(
             reinterpret_cast<char*>(xs) + (size_t)tok * CHUNK_K * 2 +
@@ -334,7 +378,7 @@ extern "C" {
 // v2 entry; ws = fp32 workspace [N, M] (required when SK would be > 1,
 // may be null when M/128 >= 224).  Returns -2 on unsupported shape.
 int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
-                      int N, int M, int K, void* stream) {
+                      int N, int M, int K, int swiglu_x, void* stream) {
   if (K % 256 != 0 || M % 128 != 0 || N > 128) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
@@ -349,16 +393,19 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
     (void)hipMemsetAsync(ws, 0, (size_t)N * M * sizeof(float), s);
   dim3 grid(mblocks, SK), block(512);
 
+#define LAUNCH_V2_(NT, AT, SW)                                              \
+  hipLaunchKernelGGL((decode_gemm_v2_kernel<NT, AT, SW>), grid, block, lds, \
+                     s, (bf16*)out, (float*)ws, (const bf16*)x,             \
+                     (const bf16*)w, N, M, K, SK)
 #define LAUNCH_V2(NT)                                                       \
   do {                                                                      \
-    if (SK > 1)                                                             \
-      hipLaunchKernelGGL((decode_gemm_v2_kernel<NT, true>), grid, block,    \
-                         lds, s, (bf16*)out, (float*)ws, (const bf16*)x,    \
-                         (const bf16*)w, N, M, K, SK);                      \
-    else                                                                    \
-      hipLaunchKernelGGL((decode_gemm_v2_kernel<NT, false>), grid, block,   \
-                         lds, s, (bf16*)out, (float*)ws, (const bf16*)x,    \
-                         (const bf16*)w, N, M, K, SK);                      \
+    if (SK > 1) {                                                           \
+      if (swiglu_x) LAUNCH_V2_(NT, true, true);                             \
+      else LAUNCH_V2_(NT, true, false);                                     \
+    } else {                                                                \
+      if (swiglu_x) LAUNCH_V2_(NT, false, true);                            \
+      else LAUNCH_V2_(NT, false, false);                                    \
+    }                                                                       \
   } while (0)
   switch (n_tiles) {
     case 1: LAUNCH_V2(1); break;

@@ -109,3 +109,15 @@ class RowParallelLinear(nn.Module):
         else:
             y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)
+
+    def forward_swiglu(self, gateup: torch.Tensor) -> torch.Tensor:
+        """down(silu(gate)*up) with the SwiGLU fused into the decode
+        GEMM's x staging when the measured policy applies (each rank
+        holds its own [gate_local | up_local] halves, so the fusion is
+        TP-transparent)."""
+        from .. import ops as _ops
+        if gateup.dim() == 2 and gateup.shape[0] <= 128 and gateup.is_cuda:
+            y = _ops.swiglu_linear(gateup, self.weight)
+            if y is not None:
+                return self.tp.all_reduce(y)
+        return self.forward(_ops.swiglu(gateup))

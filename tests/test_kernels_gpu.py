@@ -5,6 +5,7 @@ import torch
 
 import llmlb_amd.ops as ops
 from llmlb_amd.ops import reference as ref
+from llmlb_amd.ops import reference as ref_mod
 
 pytestmark = pytest.mark.gpu
 
@@ -207,3 +208,25 @@ def test_strided_qkv_views_match_contiguous():
     out_c = ops.paged_attn_decode(qc[:S], kcache_c, vcache_c, bt, sl,
                                   D ** -0.5)
     assert torch.equal(out_s, out_c)
+
+
+@pytest.mark.gpu
+def test_swiglu_linear_fused():
+    """Fused silu(gate)*up @ W^T (swiglu applied while staging x in the
+    v2 decode GEMM) must be bit-identical to the unfused pipeline and
+    close to the fp32 reference."""
+    torch.manual_seed(3)
+    dev = "cuda:0"
+    N, inter, M = 64, 14336, 4096
+    gateup = (torch.randn(N, 2 * inter, dtype=torch.float32, device=dev)
+              .to(torch.bfloat16))
+    w = (torch.randn(M, inter, dtype=torch.float32, device=dev)
+         * 0.02).to(torch.bfloat16)
+    fused = ops.swiglu_linear(gateup, w)
+    assert fused is not None
+    act = ops.swiglu(gateup)
+    unfused = ops.decode_gemm_v2(act, w)
+    assert torch.equal(fused, unfused)
+    ref = ref_mod.swiglu(gateup.float()) @ w.float().T
+    rel = (fused.float() - ref).abs().max() / ref.abs().max()
+    assert rel < 0.02, rel.item()
