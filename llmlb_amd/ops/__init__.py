@@ -223,12 +223,17 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, weight)
 
 
+_FUSED_SWIGLU = os.environ.get("LLMLB_FUSED_SWIGLU",
+                               "1").lower() not in ("0", "false", "off")
+
+
 def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
     """Fused silu(gate)*up @ W^T: the SwiGLU epilogue is applied while
     staging x chunks into LDS inside the v2 decode GEMM, removing the
     standalone swiglu kernel and the activation round trip.  Returns
     None when the fused path does not apply (caller falls back)."""
-    if (_USE_CUSTOM_DECODE_GEMM in ("0", "false", "off")
+    if (not _FUSED_SWIGLU
+            or _USE_CUSTOM_DECODE_GEMM in ("0", "false", "off")
             or not gateup.is_cuda or _lib is None
             or gateup.dtype != torch.bfloat16 or gateup.dim() != 2
             or not gateup.is_contiguous()):
