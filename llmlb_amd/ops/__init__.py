@@ -223,8 +223,11 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.linear(x, weight)
 
 
+# fusing swiglu into the staging measured ~4% SLOWER end-to-end (the
+# per-M-block silu recompute sits on the single-buffered staging
+# critical path; profiles/r01_fused_swiglu_ab.md) — default OFF
 _FUSED_SWIGLU = os.environ.get("LLMLB_FUSED_SWIGLU",
-                               "1").lower() not in ("0", "false", "off")
+                               "0").lower() in ("1", "true", "on")
 
 
 def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):

@@ -222,11 +222,20 @@ def test_swiglu_linear_fused():
               .to(torch.bfloat16))
     w = (torch.randn(M, inter, dtype=torch.float32, device=dev)
          * 0.02).to(torch.bfloat16)
-    fused = ops.swiglu_linear(gateup, w)
+    import llmlb_amd.ops as ops_mod
+    old = ops_mod._FUSED_SWIGLU
+    ops_mod._FUSED_SWIGLU = True   # measured slower -> default off
+    try:
+        fused = ops.swiglu_linear(gateup, w)
+    finally:
+        ops_mod._FUSED_SWIGLU = old
     assert fused is not None
     act = ops.swiglu(gateup)
     unfused = ops.decode_gemm_v2(act, w)
-    assert torch.equal(fused, unfused)
+    # split-K accumulates through fp32 atomics, so run-to-run ordering
+    # differs by a few bf16 ULP; compare with matching tolerance
+    assert torch.allclose(fused.float(), unfused.float(), atol=0.07,
+                          rtol=0.02)
     ref = ref_mod.swiglu(gateup.float()) @ w.float().T
     rel = (fused.float() - ref).abs().max() / ref.abs().max()
     assert rel < 0.02, rel.item()
