@@ -388,7 +388,9 @@ class Engine:
             last_idx = torch.arange(1, B + 1, device=dev) * Ls - 1
             logits = self.model.final(h[last_idx], residual[last_idx])
             toks = sample(logits, [s.params for s in group],
-                          [s.generator for s in group])
+                          [s.generator for s in group],
+                          histories=[(s.prompt_ids, s.out_ids)
+                                     for s in group])
             # install decode state
             rows = torch.tensor([s.row for s in group], dtype=torch.long,
                                 device=dev)
@@ -539,7 +541,9 @@ class Engine:
             self.timing["forward"] += time.monotonic() - _t1
             _t2 = time.monotonic()
         toks = sample(logits, [s.params for s in self.running],
-                      [s.generator for s in self.running])
+                      [s.generator for s in self.running],
+                      histories=[(s.prompt_ids, s.out_ids)
+                                 for s in self.running])
         self._last_tok[rows] = toks
         self._len_dev[rows] = lens + 1
         if sample_gpu_time:

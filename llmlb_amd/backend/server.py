@@ -120,6 +120,14 @@ class BackendServer:
         return SamplingParams(
             temperature=float(body.get("temperature", 0.0) or 0.0),
             top_p=float(body.get("top_p", 1.0) or 1.0),
+            top_k=int(body.get("top_k", 0) or 0),
+            min_p=float(body.get("min_p", 0.0) or 0.0),
+            frequency_penalty=float(body.get("frequency_penalty", 0.0)
+                                    or 0.0),
+            presence_penalty=float(body.get("presence_penalty", 0.0)
+                                   or 0.0),
+            repetition_penalty=float(body.get("repetition_penalty", 1.0)
+                                     or 1.0),
             max_new_tokens=int(body.get("max_tokens")
                                or body.get("max_completion_tokens")
                                or self.max_tokens_default),

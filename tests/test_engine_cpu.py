@@ -95,3 +95,34 @@ def test_sampling_temperature_seeded(tiny_engine):
     p2 = SamplingParams(temperature=0.8, top_p=0.9, max_new_tokens=6, seed=42)
     r2 = tiny_engine.generate_sync([5, 6], max_new_tokens=6, params=p2)
     assert r1.token_ids == r2.token_ids
+
+
+def test_sampling_filters_and_penalties():
+    """top_k / min_p / penalties: shape-level behavior on a controlled
+    logits row (unit test of sampler.sample, no engine needed)."""
+    import torch
+    from llmlb_amd.backend.sampler import SamplingParams, sample
+    logits = torch.tensor([[5.0, 4.0, 3.0, 2.0, 1.0]])
+    # top_k=1 must always pick the argmax even at high temperature
+    g = torch.Generator().manual_seed(0)
+    p = SamplingParams(temperature=5.0, top_k=1)
+    assert sample(logits, [p], [g]).item() == 0
+    # presence penalty pushes a repeated token below the runner-up
+    p2 = SamplingParams(temperature=0.0, presence_penalty=2.0)
+    tok = sample(logits, [p2], [None],
+                 histories=[([1, 2], [0])]).item()
+    assert tok == 1
+    # frequency penalty scales with count
+    p3 = SamplingParams(temperature=0.0, frequency_penalty=0.6)
+    tok = sample(logits, [p3], [None],
+                 histories=[([], [0, 0])]).item()
+    assert tok == 1
+    # repetition penalty suppresses prompt tokens too
+    p4 = SamplingParams(temperature=0.0, repetition_penalty=2.0)
+    tok = sample(logits, [p4], [None], histories=[([0], [])]).item()
+    assert tok == 1
+    # min_p keeps only near-max tokens
+    p5 = SamplingParams(temperature=1.0, min_p=0.9, seed=1)
+    g5 = torch.Generator().manual_seed(1)
+    for _ in range(5):
+        assert sample(logits, [p5], [g5]).item() == 0
