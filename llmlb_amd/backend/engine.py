@@ -252,8 +252,10 @@ class Engine:
                sink=None) -> tuple[int, queue.Queue]:
         """Queue a request; returns (request_id, token sink).
 
-        The sink receives .put((token_id, finish_reason|None)) calls from
-        the engine thread; finish_reason is set on the final item.  Pass
+        The sink receives .put((token_id, finish_reason|None, extra))
+        calls from the engine thread; finish_reason is set on the final
+        item and ``extra`` carries per-token logprob info when
+        params.logprobs > 0 (None otherwise).  Pass
         a custom ``sink`` (any object with put()) to bridge into asyncio
         (see server._AsyncSink)."""
         params = params or SamplingParams()
@@ -282,7 +284,8 @@ class Engine:
             if not running_loop:
                 self.step()
             try:
-                tok, fin = out_q.get(block=running_loop, timeout=60.0)
+                tok, fin, _extra = out_q.get(block=running_loop,
+                                             timeout=60.0)
             except queue.Empty:
                 continue
             res.token_ids.append(tok)
@@ -414,10 +417,11 @@ class Engine:
                     self.prefix.prompt_tokens += L
             self.running.extend(group)
             self._rows_dirty = True
+            extras = self._logprob_extras(logits, toks, group)
             toks_cpu = toks.tolist()
             finished = []
-            for s, tok in zip(group, toks_cpu):
-                if not self._push_token(s, tok):
+            for s, tok, ex in zip(group, toks_cpu, extras):
+                if not self._push_token(s, tok, ex):
                     finished.append(s)
             for s in finished:
                 self._retire(s)
@@ -557,10 +561,11 @@ class Engine:
         if self.timing is not None:
             self.timing["sample_sync"] += time.monotonic() - _t2
             _t3 = time.monotonic()
+        extras = self._logprob_extras(logits, toks, self.running)
         finished = []
-        for seq, tok in zip(self.running, toks_cpu):
+        for seq, tok, ex in zip(self.running, toks_cpu, extras):
             seq.length += 1
-            if not self._push_token(seq, tok):
+            if not self._push_token(seq, tok, ex):
                 finished.append(seq)
         for seq in finished:
             self._retire(seq)
@@ -607,7 +612,29 @@ class Engine:
             seq.row = -1
         self._rows_dirty = True
 
-    def _push_token(self, seq: _Sequence, tok: int) -> bool:
+    def _logprob_extras(self, logits: torch.Tensor, toks: torch.Tensor,
+                        seqs: list[_Sequence]) -> list[dict | None]:
+        """Per-token logprob payloads for sequences that asked for them
+        (params.logprobs > 0).  Slow path: full-vocab log_softmax over
+        just the requesting rows."""
+        idxs = [i for i, s in enumerate(seqs) if s.params.logprobs > 0]
+        extras: list[dict | None] = [None] * len(seqs)
+        if not idxs:
+            return extras
+        sub = logits[idxs].float().log_softmax(dim=-1)
+        toks_cpu = toks.tolist()
+        for j, i in enumerate(idxs):
+            k = min(seqs[i].params.logprobs, 20)
+            row = sub[j]
+            topv, topi = row.topk(k)
+            extras[i] = {
+                "logprob": row[toks_cpu[i]].item(),
+                "top": list(zip(topi.tolist(), topv.tolist())),
+            }
+        return extras
+
+    def _push_token(self, seq: _Sequence, tok: int,
+                    extra: dict | None = None) -> bool:
         """Returns False when the sequence is finished."""
         if not seq.out_ids:
             seq.t_first = time.monotonic()
@@ -628,7 +655,7 @@ class Engine:
             fin = "stop"
         elif len(seq.out_ids) >= seq.params.max_new_tokens:
             fin = "length"
-        seq.out_queue.put((tok, fin))
+        seq.out_queue.put((tok, fin, extra))
         return fin is None
 
     @property

@@ -22,6 +22,7 @@ class SamplingParams:
     max_new_tokens: int = 128
     stop_token_ids: tuple = ()
     seed: int | None = None
+    logprobs: int = 0            # 0 = off, else top-k alternatives
 
     def needs_history(self) -> bool:
         return (self.frequency_penalty != 0.0

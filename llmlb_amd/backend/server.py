@@ -132,7 +132,17 @@ class BackendServer:
                                or body.get("max_completion_tokens")
                                or self.max_tokens_default),
             seed=body.get("seed"),
+            logprobs=self._parse_logprobs(body),
         )
+
+    @staticmethod
+    def _parse_logprobs(body: dict) -> int:
+        lp = body.get("logprobs")
+        if lp is True:                    # chat form
+            return int(body.get("top_logprobs", 0) or 0) or 1
+        if isinstance(lp, int) and not isinstance(lp, bool):
+            return min(lp, 20)            # completions form
+        return 0
 
     def submit_chat(self, body: dict, sink=None):
         prompt = render_chat(body.get("messages", []))
@@ -201,7 +211,7 @@ async def handle_responses(server: "BackendServer", body: dict, send):
     async def collect():
         toks = []
         while True:
-            tok, fin = await sink.q.get()
+            tok, fin, _extra = await sink.q.get()
             toks.append(tok)
             if fin:
                 return toks, fin
@@ -242,7 +252,7 @@ async def handle_responses(server: "BackendServer", body: dict, send):
                 "more_body": True})
     ntok = 0
     while True:
-        tok, fin = await sink.q.get()
+        tok, fin, _extra = await sink.q.get()
         ntok += 1
         await send({"type": "http.response.body",
                     "body": ev({"type": "response.output_text.delta",
@@ -379,10 +389,24 @@ def build_asgi_app(server: BackendServer):
             include_usage = bool(
                 (body.get("stream_options") or {}).get("include_usage"))
 
+            want_lp = server._parse_logprobs(body) > 0
+
+            def lp_entry(tok, extra):
+                return {
+                    "token": server.tokenizer.decode_token(tok).strip(),
+                    "logprob": extra["logprob"], "bytes": None,
+                    "top_logprobs": [
+                        {"token":
+                             server.tokenizer.decode_token(t).strip(),
+                         "logprob": v, "bytes": None}
+                        for t, v in extra["top"]]}
+
             if not stream:
-                toks, finish = [], "stop"
-                async for tok, fin in drain_queue_async(out_q):
+                toks, finish, lp_content = [], "stop", []
+                async for tok, fin, extra in drain_queue_async(out_q):
                     toks.append(tok)
+                    if extra is not None:
+                        lp_content.append(lp_entry(tok, extra))
                     if fin:
                         finish = fin
                 text = "".join(server.tokenizer.decode_token(t)
@@ -397,6 +421,8 @@ def build_asgi_app(server: BackendServer):
                         "choices": [{"index": 0,
                                      "message": {"role": "assistant",
                                                  "content": text},
+                                     "logprobs": ({"content": lp_content}
+                                                  if want_lp else None),
                                      "finish_reason": finish}],
                         "usage": usage}
                 else:
@@ -435,13 +461,20 @@ def build_asgi_app(server: BackendServer):
                             "more_body": True})
             ntok = 0
             finish = "stop"
-            async for tok, fin in drain_queue_async(out_q):
+            async for tok, fin, extra in drain_queue_async(out_q):
                 ntok += 1
                 if fin:
                     finish = fin
                 text = server.tokenizer.decode_token(tok)
+                payload = chunk_payload({"content": text})
+                if extra is not None and is_chat:
+                    obj = json.loads(payload[len(b"data: "):])
+                    obj["choices"][0]["logprobs"] = {
+                        "content": [lp_entry(tok, extra)]}
+                    payload = ("data: " + json.dumps(obj)
+                               + "\n\n").encode()
                 await send({"type": "http.response.body",
-                            "body": chunk_payload({"content": text}),
+                            "body": payload,
                             "more_body": True})
             usage = ({"prompt_tokens": len(prompt_ids),
                       "completion_tokens": ntok,

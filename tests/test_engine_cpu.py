@@ -80,7 +80,7 @@ def _drain(q: queue.Queue) -> list:
     toks = []
     while True:
         try:
-            tok, fin = q.get_nowait()
+            tok, fin, _ex = q.get_nowait()
         except queue.Empty:
             break
         toks.append(tok)
@@ -126,3 +126,26 @@ def test_sampling_filters_and_penalties():
     g5 = torch.Generator().manual_seed(1)
     for _ in range(5):
         assert sample(logits, [p5], [g5]).item() == 0
+
+
+def test_logprobs_chat(tiny_engine):
+    """params.logprobs=k yields per-token chosen logprob + top-k
+    alternatives; chosen logprob must be <= 0 and appear in top-k for
+    greedy decoding."""
+    from llmlb_amd.backend.sampler import SamplingParams
+    eng = tiny_engine
+    p = SamplingParams(temperature=0.0, max_new_tokens=4, logprobs=3)
+    rid, q = eng.submit([5, 6, 7], p)
+    for _ in range(30):
+        if not eng.step():
+            break
+    items = []
+    while not q.empty():
+        items.append(q.get_nowait())
+    assert len(items) == 4
+    for tok, fin, extra in items:
+        assert extra is not None
+        assert extra["logprob"] <= 0.0
+        assert len(extra["top"]) == 3
+        # greedy: the chosen token is the top-1 alternative
+        assert extra["top"][0][0] == tok

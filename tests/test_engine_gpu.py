@@ -92,7 +92,7 @@ def test_gpu_many_concurrent(gpu_engine):
     for q in qs:
         toks = []
         while True:
-            tok, fin = q.get_nowait()
+            tok, fin, _ex = q.get_nowait()
             toks.append(tok)
             if fin:
                 break

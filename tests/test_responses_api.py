@@ -107,3 +107,22 @@ def test_responses_tokens_recorded(stack):
                  and r["status"] == 200]
     assert resp_recs
     assert any(r["completion_tokens"] > 0 for r in resp_recs)
+
+
+def test_chat_logprobs_through_gateway(stack):
+    """OpenAI chat logprobs: logprobs=true + top_logprobs=k returns a
+    content array with per-token logprob and k alternatives."""
+    body = json.dumps({"model": "llama-tiny",
+                       "messages": [{"role": "user", "content": "hi"}],
+                       "max_tokens": 3, "logprobs": True,
+                       "top_logprobs": 2}).encode()
+    req = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions", data=body,
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        out = json.loads(r.read())
+    lp = out["choices"][0]["logprobs"]
+    assert lp and len(lp["content"]) == 3
+    first = lp["content"][0]
+    assert first["logprob"] <= 0.0
+    assert len(first["top_logprobs"]) == 2
