@@ -110,6 +110,7 @@ bool anthropic_request_to_openai(const Json& in, Json* out,
   if (in.find("temperature")) o.emplace_back("temperature",
                                              in["temperature"]);
   if (in.find("top_p")) o.emplace_back("top_p", in["top_p"]);
+  if (in.find("top_k")) o.emplace_back("top_k", in["top_k"]);
   if (in.find("stream")) o.emplace_back("stream", in["stream"]);
   const Json* stops = in.find("stop_sequences");
   if (stops && stops->is_arr()) o.emplace_back("stop", *stops);
