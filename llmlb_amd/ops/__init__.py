@@ -176,6 +176,22 @@ _DECODE_GEMM_VERSION = int(os.environ.get("LLMLB_DECODE_GEMM_VERSION",
                                           "2"))
 
 
+# split-K fp32 workspaces, reused per shape: allocated zeroed once;
+# every kernel invocation's convert pass re-zeroes after reading, so
+# replays (and hipGraph captures) carry no per-call memset.  Calls are
+# stream-ordered on the engine's stream, so reuse is race-free.
+_WS_CACHE: dict = {}
+
+
+def _split_k_ws(N: int, M: int, device) -> torch.Tensor:
+    key = (N, M, str(device))
+    t = _WS_CACHE.get(key)
+    if t is None:
+        t = torch.zeros(N, M, dtype=torch.float32, device=device)
+        _WS_CACHE[key] = t
+    return t
+
+
 def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
                    ws: torch.Tensor | None = None) -> torch.Tensor:
     """Staged-x decode GEMM (v2).  Raises on unsupported shapes."""
@@ -185,7 +201,7 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     out = torch.empty(N, M, dtype=x.dtype, device=x.device)
     need_ws = (M // 128) < 224
     if need_ws and ws is None:
-        ws = torch.empty(N, M, dtype=torch.float32, device=x.device)
+        ws = _split_k_ws(N, M, x.device)
     rc = lib.ll_decode_gemm_v2(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
         _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
@@ -252,8 +268,7 @@ def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
         return None
     lib = _require_native()
     out = torch.empty(N, M, dtype=gateup.dtype, device=gateup.device)
-    ws = torch.empty(N, M, dtype=torch.float32, device=gateup.device) \
-        if (M // 128) < 224 else None
+    ws = _split_k_ws(N, M, gateup.device) if (M // 128) < 224 else None
     rc = lib.ll_decode_gemm_v2(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(gateup),
         _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),

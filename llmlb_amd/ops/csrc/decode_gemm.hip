@@ -360,13 +360,18 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
   }
 }
 
+// converts the fp32 split-K accumulator to bf16 AND writes zeros back,
+// so the workspace is ready for the next call without a separate
+// hipMemset pass (the memset cost ~5 us per GEMM in-graph; the caller
+// must hand over a zeroed workspace the first time)
 __global__ void f32_to_bf16_kernel(bf16* __restrict__ out,
-                                   const float* __restrict__ in,
+                                   float* __restrict__ in,
                                    long long n) {
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   union { float f; unsigned int u; } c2;
   c2.f = in[i];
+  in[i] = 0.f;
   unsigned int bias = 0x7FFF + ((c2.u >> 16) & 1);
   union { unsigned short u; __hip_bfloat16 b; } cv;
   cv.u = (unsigned short)((c2.u + bias) >> 16);
@@ -389,8 +394,8 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
   const size_t lds = (size_t)n_tiles * 16 * 256 * 2;  // x chunk
   if (lds > 160 * 1024) return -2;
   if (SK > 1 && ws == nullptr) return -2;
-  if (SK > 1)
-    (void)hipMemsetAsync(ws, 0, (size_t)N * M * sizeof(float), s);
+  // no memset: the convert pass re-zeroes the workspace after reading
+  // it, and the Python side allocates it zeroed
   dim3 grid(mblocks, SK), block(512);
 
 #define LAUNCH_V2_(NT, AT, SW)                                              \
@@ -422,7 +427,7 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
     int thr = 256;
     long long blocks2 = (n + thr - 1) / thr;
     hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((unsigned)blocks2),
-                       dim3(thr), 0, s, (bf16*)out, (const float*)ws, n);
+                       dim3(thr), 0, s, (bf16*)out, (float*)ws, n);
   }
   hipError_t e = hipGetLastError();
   return e == hipSuccess ? 0 : (int)e;
