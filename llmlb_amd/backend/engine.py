@@ -119,7 +119,7 @@ class _DecodeGraph:
         group = nq // engine.n_kv_local
         use_v2 = ops._PAGED_ATTN_V2 and D == 128 and group <= 4
         slices = n_parts * 4 if use_v2 else n_parts
-        if slices > 1 or use_v2:
+        if n_parts > 1:
             self.ws = (torch.empty(S, nq, slices, D, dtype=torch.float32,
                                    device=dev),
                        torch.empty(S, nq, slices, 2, dtype=torch.float32,

@@ -313,7 +313,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
     # v2 (GQA-amortized) always reduces through the slice workspace
     use_v2 = _PAGED_ATTN_V2 and D == 128 and group <= 4
     slices = n_parts * 4 if use_v2 else n_parts
-    if slices > 1 or use_v2:
+    if n_parts > 1:
         if workspace is not None:
             ws_acc, ws_ml = workspace
         else:

@@ -563,8 +563,59 @@ void paged_attn_decode_v2_kernel(
         acc[g][i] += __shfl_xor(acc[g][i], off, WAVE);
     }
   }
-  // each wave is its own flash-decode slice: write (m, l, acc) to the
-  // grid workspace at slice granularity; the reduce kernel combines.
+  if (gridDim.z == 1) {
+    // single grid partition: the 4 wave-slices live in THIS workgroup,
+    // so combine through LDS and write bf16 directly — no global f32
+    // slice round trip, no reduce kernel (saves ~8 MB/layer of ws
+    // traffic + one launch at the common short-context sizes)
+    __shared__ float s_acc[PA_WAVES][GROUP][D];
+    __shared__ float s_ml[PA_WAVES][GROUP][2];
+    if (t_local == 0) {
+#pragma unroll
+      for (int g = 0; g < GROUP; ++g) {
+#pragma unroll
+        for (int i = 0; i < DS; ++i) s_acc[wid][g][p * DS + i] = acc[g][i];
+        if (p == 0) {
+          s_ml[wid][g][0] = m[g];
+          s_ml[wid][g][1] = l[g];
+        }
+      }
+    }
+    __syncthreads();
+    if (wid == 0 && t_local == 0) {
+#pragma unroll
+      for (int g = 0; g < GROUP; ++g) {
+        float m_tot = -INFINITY;
+#pragma unroll
+        for (int w2 = 0; w2 < PA_WAVES; ++w2)
+          m_tot = fmaxf(m_tot, s_ml[w2][g][0]);
+        float l_tot = 0.f, o[DS];
+#pragma unroll
+        for (int i = 0; i < DS; ++i) o[i] = 0.f;
+#pragma unroll
+        for (int w2 = 0; w2 < PA_WAVES; ++w2) {
+          const float mw = s_ml[w2][g][0];
+          const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - m_tot);
+          l_tot += s_ml[w2][g][1] * sc;
+#pragma unroll
+          for (int i = 0; i < DS; ++i)
+            o[i] += s_acc[w2][g][p * DS + i] * sc;
+        }
+        const float inv = l_tot > 0.f ? 1.f / l_tot : 0.f;
+        const int qh = kvh * GROUP + g;
+        bf16* ob = out + ((size_t)seq * n_q + qh) * D + p * DS;
+#pragma unroll
+        for (int i = 0; i < DS; ++i) {
+          union { unsigned short u; __hip_bfloat16 b; } cv;
+          cv.u = f2bf(o[i] * inv);
+          ob[i] = cv.b;
+        }
+      }
+    }
+    return;
+  }
+  // multi-partition: each wave is its own flash-decode slice — write
+  // (m, l, acc) to the grid workspace; the reduce kernel combines.
   if (t_local == 0) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
@@ -718,7 +769,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
   // v2 (GQA-amortized, pages split across waves): caller must size the
   // workspace for n_parts*PA_WAVES slices and pass want_v2 explicitly
   const bool v2 = (want_v2 && D == 128 && group <= 4 &&
-                   ws_acc != nullptr);
+                   (n_parts == 1 || ws_acc != nullptr));
   if (v2) {
     dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
     const int slices = n_parts * PA_WAVES;
@@ -754,7 +805,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
         goto v1_path;
     }
     LL_CHECK(hipGetLastError());
-    {
+    if (n_parts > 1) {   // single partition combined in-workgroup
       dim3 rgrid(num_seqs, n_q), rblock(WAVE);
       hipLaunchKernelGGL(paged_attn_reduce_kernel, rgrid, rblock, 0, s,
                          (bf16*)out, (const float*)ws_acc,
