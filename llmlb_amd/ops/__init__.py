@@ -215,10 +215,11 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
 def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """out = x @ W^T.
 
-    Dispatch policy (measured, profiles/r01_decode_gemm_v2_ab.md):
-    the staged-x v2 kernel wins on the split-K shapes with large K
-    (down-projection: M/128 < 224 and K >= 8192) and on small batches
-    (N <= 16); hipBLASLt keeps the rest (notably gate_up at N >= 64).
+    Dispatch policy (measured IN-ENGINE, profiles/r01_attn_lds_and_
+    gemm_policy.md): isolated microbenches had v2 winning the
+    down-projection, but inside the captured decode step hipBLASLt is
+    ~3% faster end-to-end at bs64 (cache interactions differ), so auto
+    keeps v2 only for tiny batches (N <= 16) where its advantage held.
     LLMLB_CUSTOM_DECODE_GEMM: "auto" (default) = this policy,
     "1"/"on" = v2 everywhere it is legal, "0"/"off" = hipBLASLt only."""
     mode = _USE_CUSTOM_DECODE_GEMM
@@ -229,8 +230,7 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     M = weight.shape[0]
     legal = N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 4096
     if legal and mode == "auto":
-        mblocks = M // 128
-        legal = (mblocks < 224 and K >= 8192) or N <= 16
+        legal = N <= 16
     if legal:
         try:
             return decode_gemm_v2(x, weight)
