@@ -56,6 +56,9 @@ class EngineConfig:
     # prompts longer than this prefill in chunks (activation-memory
     # bound for long prompts; keeps decode latency smooth)
     prefill_chunk: int = 4096
+    # HF-format safetensors checkpoint (file or shard dir); None =
+    # deterministic random init (synthetic serving)
+    weights_path: str | None = None
     tp_ctx: object = None
     seed: int = 1234
 
@@ -180,7 +183,10 @@ class Engine:
         self.device = torch.device(cfg.device)
         dtype = getattr(torch, cfg.dtype)
         self.model = LlamaForCausalLM(cfg.model, tp=cfg.tp_ctx, dtype=dtype,
-                                      device=cfg.device, seed=cfg.seed)
+                                      device=cfg.device, seed=cfg.seed,
+                                      init_weights=not cfg.weights_path)
+        if cfg.weights_path:
+            self.model.load_safetensors(cfg.weights_path)
         m = cfg.model
         n_kv_local = m.kv_heads // self.model.tp.world_size
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)

@@ -28,7 +28,8 @@ PRESETS = {
 
 def build_engine(model: str, device: str, kv_blocks: int | None = None,
                  max_batch: int = 256, tp_ctx=None,
-                 hbm_fraction: float = 0.85) -> Engine:
+                 hbm_fraction: float = 0.85,
+                 weights: str | None = None) -> Engine:
     """tp_ctx=None means 'infer from torch.distributed' — pass
     TPContext.solo() for DP replicas inside a larger world."""
     cfg_fn = PRESETS.get(model)
@@ -39,7 +40,8 @@ def build_engine(model: str, device: str, kv_blocks: int | None = None,
     dtype = "bfloat16" if device.startswith("cuda") else "float32"
     ecfg = EngineConfig(model=mcfg, device=device, dtype=dtype,
                         kv_cache_blocks=kv_blocks, max_batch=max_batch,
-                        tp_ctx=tp_ctx, hbm_fraction=hbm_fraction)
+                        tp_ctx=tp_ctx, hbm_fraction=hbm_fraction,
+                        weights_path=weights)
     return Engine(ecfg)
 
 
@@ -52,6 +54,8 @@ def main(argv=None) -> None:
     ap.add_argument("--host", default="0.0.0.0")
     ap.add_argument("--port", type=int, default=9200)
     ap.add_argument("--kv-blocks", type=int, default=None)
+    ap.add_argument("--weights", default=None,
+                    help="HF safetensors checkpoint (file or dir)")
     ap.add_argument("--max-batch", type=int, default=256)
     args = ap.parse_args(argv)
 
@@ -61,7 +65,7 @@ def main(argv=None) -> None:
     if device.startswith("cuda"):
         torch.cuda.set_device(device)
     engine = build_engine(args.model, device, args.kv_blocks,
-                          args.max_batch)
+                          args.max_batch, weights=args.weights)
     served = args.served_name or engine.cfg.model.name
     from .server import serve
     print(f"llmlb_amd backend: model={served} device={device} "

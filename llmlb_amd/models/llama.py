@@ -116,7 +116,8 @@ class LlamaForCausalLM(nn.Module):
 
     def __init__(self, cfg: LlamaConfig, tp: TPContext | None = None,
                  dtype: torch.dtype = torch.bfloat16,
-                 device: str | torch.device = "cpu", seed: int = 1234):
+                 device: str | torch.device = "cpu", seed: int = 1234,
+                 init_weights: bool = True):
         super().__init__()
         self.cfg = cfg
         self.tp = tp or get_tp_context()
@@ -137,7 +138,8 @@ class LlamaForCausalLM(nn.Module):
         self.to(device)
         self.cos_sin = self.cos_sin.to(device)
         self.device = torch.device(device)
-        self._init_weights(seed)
+        if init_weights:          # skipped when a checkpoint follows
+            self._init_weights(seed)
 
     def _init_weights(self, seed: int) -> None:
         # init on the target device (randn of 8B params on CPU is slow)
@@ -160,6 +162,107 @@ class LlamaForCausalLM(nn.Module):
             init_(layer.o_proj.weight)
             init_(layer.gate_up.weight)
             init_(layer.down.weight)
+
+    # -- checkpoint I/O ---------------------------------------------------
+
+    @staticmethod
+    def _shard(t: torch.Tensor, dim: int, ws: int, rank: int
+               ) -> torch.Tensor:
+        if ws == 1:
+            return t
+        size = t.shape[dim] // ws
+        return t.narrow(dim, rank * size, size)
+
+    def load_safetensors(self, path) -> int:
+        """Load HuggingFace-format Llama weights (*.safetensors file,
+        or a directory of shards).  Maps HF names onto the fused
+        qkv / gate_up layouts and applies this rank's TP shard (column
+        split for qkv/gate_up, input split for o_proj/down — Megatron
+        convention, parallel/tp.py).  Returns the number of tensors
+        consumed.  HF q/k use the same rotate-half RoPE layout as
+        ops.rope_apply, so no permutation is needed."""
+        from pathlib import Path as _P
+
+        from safetensors import safe_open
+
+        p = _P(path)
+        files = sorted(p.glob("*.safetensors")) if p.is_dir() else [p]
+        if not files:
+            raise FileNotFoundError(f"no .safetensors under {path}")
+        ws, rank = self.tp.world_size, self.tp.rank
+        D = self.cfg.head_dim
+        tensors: dict = {}
+        n = 0
+        for f in files:
+            with safe_open(str(f), framework="pt") as sf:
+                for name in sf.keys():
+                    tensors[name] = sf.get_tensor(name)
+                    n += 1
+
+        def take(name):
+            t = tensors[name]
+            return t.to(self.dtype)
+
+        def assign(param, t):
+            assert param.shape == t.shape, (param.shape, t.shape)
+            param.data.copy_(t.to(self.device))
+
+        assign(self.embed, take("model.embed_tokens.weight"))
+        assign(self.final_norm_w, take("model.norm.weight"))
+        lm = ("lm_head.weight" if "lm_head.weight" in tensors
+              else "model.embed_tokens.weight")   # tied embeddings
+        assign(self.lm_head, take(lm))
+        for i, layer in enumerate(self.layers_):
+            pre = f"model.layers.{i}."
+            assign(layer.input_norm_w, take(pre + "input_layernorm.weight"))
+            assign(layer.post_norm_w,
+                   take(pre + "post_attention_layernorm.weight"))
+            q = self._shard(take(pre + "self_attn.q_proj.weight"), 0,
+                            ws, rank)
+            k = self._shard(take(pre + "self_attn.k_proj.weight"), 0,
+                            ws, rank)
+            v = self._shard(take(pre + "self_attn.v_proj.weight"), 0,
+                            ws, rank)
+            assign(layer.qkv.weight, torch.cat([q, k, v], dim=0))
+            assign(layer.o_proj.weight,
+                   self._shard(take(pre + "self_attn.o_proj.weight"), 1,
+                               ws, rank))
+            gate = self._shard(take(pre + "mlp.gate_proj.weight"), 0,
+                               ws, rank)
+            up = self._shard(take(pre + "mlp.up_proj.weight"), 0,
+                             ws, rank)
+            assign(layer.gate_up.weight, torch.cat([gate, up], dim=0))
+            assign(layer.down.weight,
+                   self._shard(take(pre + "mlp.down_proj.weight"), 1,
+                               ws, rank))
+        return n
+
+    def save_safetensors(self, path) -> None:
+        """Write HF-format weights (inverse of load; world_size 1 only
+        — TP ranks hold shards, gather before saving)."""
+        from safetensors.torch import save_file
+
+        assert self.tp.world_size == 1, "save requires an unsharded model"
+        D = self.cfg.head_dim
+        out = {"model.embed_tokens.weight": self.embed.data,
+               "model.norm.weight": self.final_norm_w.data,
+               "lm_head.weight": self.lm_head.data}
+        for i, layer in enumerate(self.layers_):
+            pre = f"model.layers.{i}."
+            qn, kn = self.cfg.q_heads * D, self.cfg.kv_heads * D
+            qkv = layer.qkv.weight.data
+            out[pre + "self_attn.q_proj.weight"] = qkv[:qn]
+            out[pre + "self_attn.k_proj.weight"] = qkv[qn:qn + kn]
+            out[pre + "self_attn.v_proj.weight"] = qkv[qn + kn:]
+            out[pre + "self_attn.o_proj.weight"] = layer.o_proj.weight.data
+            gu = layer.gate_up.weight.data
+            out[pre + "mlp.gate_proj.weight"] = gu[:self.cfg.inter]
+            out[pre + "mlp.up_proj.weight"] = gu[self.cfg.inter:]
+            out[pre + "mlp.down_proj.weight"] = layer.down.weight.data
+            out[pre + "input_layernorm.weight"] = layer.input_norm_w.data
+            out[pre + "post_attention_layernorm.weight"] =                 layer.post_norm_w.data
+        save_file({k: v.contiguous().cpu() for k, v in out.items()},
+                  str(path))
 
     # -- forward pieces (called by the engine) ---------------------------
 
