@@ -56,6 +56,8 @@ def main(argv=None) -> None:
     ap.add_argument("--kv-blocks", type=int, default=None)
     ap.add_argument("--weights", default=None,
                     help="HF safetensors checkpoint (file or dir)")
+    ap.add_argument("--tokenizer", default=None,
+                    help="local tokenizer.json (tokenizers format)")
     ap.add_argument("--max-batch", type=int, default=256)
     args = ap.parse_args(argv)
 
@@ -70,7 +72,8 @@ def main(argv=None) -> None:
     from .server import serve
     print(f"llmlb_amd backend: model={served} device={device} "
           f"port={args.port} kv_blocks={engine.kv.num_blocks}", flush=True)
-    serve(engine, served, host=args.host, port=args.port)
+    serve(engine, served, host=args.host, port=args.port,
+          tokenizer_path=args.tokenizer)
 
 
 if __name__ == "__main__":

@@ -31,6 +31,25 @@ from .sampler import SamplingParams
 log = logging.getLogger("llmlb.backend")
 
 
+class HFTokenizer:
+    """Wrap a local HuggingFace tokenizer.json (``tokenizers`` package;
+    no network).  Used when the operator passes --tokenizer so real
+    checkpoints serve real text; falls back to SyntheticTokenizer
+    otherwise."""
+
+    def __init__(self, path: str, vocab: int):
+        from tokenizers import Tokenizer
+        self.tok = Tokenizer.from_file(path)
+        self.vocab = vocab
+
+    def encode(self, text: str) -> list[int]:
+        ids = [t % self.vocab for t in self.tok.encode(text).ids]
+        return ids or [1]
+
+    def decode_token(self, tok: int) -> str:
+        return self.tok.decode([tok])
+
+
 class SyntheticTokenizer:
     """Deterministic text<->token mapping for synthetic serving."""
 
@@ -66,11 +85,15 @@ def render_chat(messages: list[dict]) -> str:
 class BackendServer:
     def __init__(self, engine: Engine, model_name: str,
                  served_models: list[str] | None = None,
-                 max_tokens_default: int = 128):
+                 max_tokens_default: int = 128,
+                 tokenizer_path: str | None = None):
         self.engine = engine
         self.model_name = model_name
         self.served = served_models or [model_name]
-        self.tokenizer = SyntheticTokenizer(engine.cfg.model.vocab)
+        vocab = engine.cfg.model.vocab
+        self.tokenizer = (HFTokenizer(tokenizer_path, vocab)
+                          if tokenizer_path
+                          else SyntheticTokenizer(vocab))
         self.max_tokens_default = max_tokens_default
         self.requests_total = 0
         self.tokens_total = 0
@@ -496,11 +519,13 @@ def build_asgi_app(server: BackendServer):
 
 
 def serve(engine: Engine, model_name: str, host: str = "0.0.0.0",
-          port: int = 9200, served_models: list[str] | None = None):
+          port: int = 9200, served_models: list[str] | None = None,
+          tokenizer_path: str | None = None):
     """Run the backend server (blocking).  Starts the engine loop."""
     import uvicorn
 
-    server = BackendServer(engine, model_name, served_models)
+    server = BackendServer(engine, model_name, served_models,
+                           tokenizer_path=tokenizer_path)
     app = build_asgi_app(server)
     engine.start()
     config = uvicorn.Config(app, host=host, port=port, log_level="warning",

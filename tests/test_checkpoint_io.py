@@ -49,3 +49,23 @@ def test_tp_shard_slicing():
     assert torch.equal(torch.cat([s0, s1], 0), t)
     c1 = LlamaForCausalLM._shard(t, 1, 2, 1)
     assert torch.equal(c1, t[:, 2:])
+
+
+def test_hf_tokenizer_wrapper(tmp_path):
+    """BackendServer accepts a local tokenizer.json (tokenizers pkg)."""
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    vocab = {"<unk>": 0, "hello": 1, "world": 2, "hi": 3}
+    tok = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    p = tmp_path / "tokenizer.json"
+    tok.save(str(p))
+
+    from llmlb_amd.backend.server import BackendServer
+    eng = Engine(EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                              dtype="float32", kv_cache_blocks=16))
+    srv = BackendServer(eng, "m", tokenizer_path=str(p))
+    assert srv.tokenizer.encode("hello world") == [1, 2]
+    assert "hello" in srv.tokenizer.decode_token(1)
