@@ -59,6 +59,11 @@ class EngineConfig:
     # HF-format safetensors checkpoint (file or shard dir); None =
     # deterministic random init (synthetic serving)
     weights_path: str | None = None
+    # prompt-lookup speculative decoding: draft up to this many tokens
+    # per step from the sequence's own n-gram history and verify them
+    # in one batched multi-token pass (greedy sequences only).
+    # 0 = off.  Env LLMLB_SPEC_NGRAM overrides.
+    spec_ngram: int = 0
     tp_ctx: object = None
     seed: int = 1234
 
@@ -74,7 +79,7 @@ class GenerationResult:
 class _Sequence:
     __slots__ = ("rid", "prompt_ids", "out_ids", "params", "block_table",
                  "length", "out_queue", "generator", "t_submit", "t_first",
-                 "row", "cached_len", "hashes")
+                 "row", "cached_len", "hashes", "bigram_pos")
 
     def __init__(self, rid, prompt_ids, params, out_queue, generator):
         self.rid = rid
@@ -90,6 +95,7 @@ class _Sequence:
         self.row = -1                        # persistent decode-state row
         self.cached_len = 0                  # tokens reused from PrefixCache
         self.hashes: list[bytes] | None = None
+        self.bigram_pos: dict | None = None  # spec: bigram -> last pos
 
 
 class _DecodeGraph:
@@ -192,6 +198,11 @@ class Engine:
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)
         self.kv = KVCache(m.layers, num_blocks, n_kv_local, m.head_dim,
                           self.device, dtype)
+        import os as _os
+        self.spec_ngram = int(_os.environ.get("LLMLB_SPEC_NGRAM",
+                                              str(cfg.spec_ngram)) or 0)
+        self.spec_proposed = 0
+        self.spec_accepted = 0
         self.prefix = (PrefixCache(self.kv)
                        if cfg.enable_prefix_cache
                        and os_env_flag("LLMLB_PREFIX_CACHE", True)
@@ -361,7 +372,13 @@ class Engine:
         if prefills:
             self._prefill_batch(prefills)
         if self.running:
-            self._decode_step()
+            if (self.spec_ngram > 0
+                    and all(s.params.temperature < 1e-5
+                            and not s.params.needs_history()
+                            for s in self.running)):
+                self._spec_step()
+            else:
+                self._decode_step()
         self.steps_done += 1
         return bool(prefills or self.running)
 
@@ -578,6 +595,140 @@ class Engine:
         if self.timing is not None:
             self.timing["emit"] += time.monotonic() - _t3
             self.timing["steps"] += 1
+
+    # -- prompt-lookup speculative decoding ------------------------------
+    def _draft_for(self, seq: _Sequence) -> list[int]:
+        """Up to spec_ngram draft tokens: continuation after the last
+        earlier occurrence of the current trailing bigram (prompt-lookup
+        decoding).  The bigram index is maintained incrementally."""
+        ctx = seq.prompt_ids + seq.out_ids
+        if len(ctx) < 3:
+            return []
+        if seq.bigram_pos is None:
+            seq.bigram_pos = {}
+            for i in range(len(ctx) - 2):      # exclude trailing bigram
+                seq.bigram_pos[(ctx[i], ctx[i + 1])] = i
+        pos = seq.bigram_pos.get((ctx[-2], ctx[-1]))
+        if pos is None:
+            return []
+        start = pos + 2
+        return ctx[start:start + self.spec_ngram]
+
+    def _index_token(self, seq: _Sequence) -> None:
+        """Advance the bigram index by the token just emitted."""
+        if seq.bigram_pos is None:
+            return
+        ctx = seq.prompt_ids + seq.out_ids
+        if len(ctx) >= 3:
+            # the bigram ENDING at the previous token becomes eligible
+            seq.bigram_pos[(ctx[-3], ctx[-2])] = len(ctx) - 3
+
+    @torch.inference_mode()
+    def _spec_step(self) -> None:
+        """One speculative step for ALL running (greedy) sequences:
+        feed [last_token, draft...] padded to a uniform length, verify
+        with one batched multi-token forward through the paged cache
+        (rectangular causal masks), accept the longest matching prefix,
+        emit accepted+1 tokens.  Rejected positions leave stale KV that
+        the very next write at that slot overwrites (seq.length is the
+        source of truth)."""
+        dev = self.device
+        seqs = self.running
+        B = len(seqs)
+        drafts = [self._draft_for(s) for s in seqs]
+        F = 1 + max((len(d) for d in drafts), default=0)
+        if F == 1:
+            self._decode_step()
+            return
+        last = self._last_tok[self._rows_tensor()].tolist()
+        feed_rows, pos_rows, slot_rows = [], [], []
+        for s, d, t0 in zip(seqs, drafts, last):
+            # grow the block table to cover length+F new slots
+            need_blocks = (s.length + F + BLOCK_SIZE - 1) // BLOCK_SIZE
+            while len(s.block_table) < need_blocks:
+                if self.prefix is not None:
+                    self.prefix.ensure_free(1)
+                s.block_table.append(self.kv.allocate(1)[0])
+            feed = [t0] + d + [t0] * (F - 1 - len(d))
+            feed_rows.append(feed)
+            pos_rows.append(list(range(s.length, s.length + F)))
+            slot_rows.append(self._slots_for(s, s.length, F))
+        ids = torch.tensor(feed_rows, dtype=torch.long,
+                           device=dev).reshape(-1)
+        positions = torch.tensor(pos_rows, dtype=torch.int32,
+                                 device=dev).reshape(-1)
+        slots = torch.tensor(slot_rows, dtype=torch.int32,
+                             device=dev).reshape(-1)
+        Lmax = max(s.length for s in seqs) + F
+        # gather slots for key positions [0, Lmax) per seq (pad: slot 0,
+        # masked out below)
+        gslots = torch.zeros(B, Lmax, dtype=torch.long)
+        for i, s in enumerate(seqs):
+            sl = self._slots_for(s, 0, s.length)
+            gslots[i, :s.length] = torch.tensor(sl, dtype=torch.long)
+            gslots[i, s.length:s.length + F] = torch.tensor(
+                slot_rows[i], dtype=torch.long)
+        gslots = gslots.to(dev)
+        gblks, goffs = gslots // BLOCK_SIZE, gslots % BLOCK_SIZE
+        lens = torch.tensor([s.length for s in seqs], device=dev)
+        qpos = (lens[:, None] + torch.arange(F, device=dev)[None, :])
+        mask = (torch.arange(Lmax, device=dev)[None, None, :]
+                <= qpos[:, :, None])              # [B, F, Lmax]
+        mask = mask[:, None, :, :]                # broadcast over heads
+
+        nq, nkv, D = self.n_q_local, self.n_kv_local, self.cfg.model.head_dim
+        h = self.model.embed_tokens(ids)
+        residual = None
+        for li in range(self.cfg.model.layers):
+            q, k, v, residual = self.model.layer_attn_prelude(
+                li, h, residual, positions)
+            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+            qh = q.view(B, F, nq, D).transpose(1, 2)
+            kh = self.kv.k[li][gblks, :, goffs].view(B, Lmax, nkv, D)                 .transpose(1, 2)
+            vh = self.kv.v[li][gblks, :, goffs].view(B, Lmax, nkv, D)                 .transpose(1, 2)
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qh, kh, vh, attn_mask=mask, enable_gqa=True)
+            attn = o.transpose(1, 2).reshape(B * F, nq, D).contiguous()
+            h, residual = self.model.layer_attn_epilogue(li, attn,
+                                                         residual)
+        logits = self.model.final(h, residual).view(B, F, -1)
+        preds = logits.argmax(dim=-1).tolist()     # [B][F]
+
+        rows_idx, new_last, new_len = [], [], []
+        finished = []
+        for i, (s, d) in enumerate(zip(seqs, drafts)):
+            m = 0
+            while m < len(d) and preds[i][m] == d[m]:
+                m += 1
+            emit = preds[i][:m + 1]
+            self.spec_proposed += len(d)
+            self.spec_accepted += m
+            s.length += 1 + m
+            rows_idx.append(s.row)
+            new_last.append(emit[-1])
+            new_len.append(s.length)
+            done = False
+            for tok in emit:
+                if done:
+                    break
+                if not self._push_token(s, tok):
+                    done = True
+                self._index_token(s)
+            if done:
+                finished.append(s)
+        rows_t = torch.tensor(rows_idx, dtype=torch.long, device=dev)
+        self._last_tok[rows_t] = torch.tensor(new_last, dtype=torch.long,
+                                              device=dev)
+        self._len_dev[rows_t] = torch.tensor(new_len, dtype=torch.int32,
+                                             device=dev)
+        bt_rows = torch.zeros(len(seqs), self.max_blocks_per_seq,
+                              dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            bt_rows[i, :len(s.block_table)] = torch.tensor(
+                s.block_table, dtype=torch.int32)
+        self._bt_dev[rows_t] = bt_rows.to(dev)
+        for s in finished:
+            self._retire(s)
 
     def _graph_for(self, S: int) -> "_DecodeGraph | None":
         if not self.use_graphs:
