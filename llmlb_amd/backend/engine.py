@@ -59,6 +59,11 @@ class EngineConfig:
     # HF-format safetensors checkpoint (file or shard dir); None =
     # deterministic random init (synthetic serving)
     weights_path: str | None = None
+    # KV cache storage: "bfloat16" (default) or "fp8_e4m3" (OCP e4m3
+    # bytes + per-token-head scales: half the attention bandwidth,
+    # double the resident context; needs D=128 and GQA group <= 4 on
+    # GPU).  Env LLMLB_KV_DTYPE overrides.
+    kv_dtype: str = "bfloat16"
     # prompt-lookup speculative decoding: draft up to this many tokens
     # per step from the sequence's own n-gram history and verify them
     # in one batched multi-token pass (greedy sequences only).
@@ -146,10 +151,12 @@ class _DecodeGraph:
         for li in range(eng.cfg.model.layers):
             q, k, v, residual = eng.model.layer_attn_prelude(
                 li, h, residual, self.lens)
-            ops.kv_append(eng.kv.k[li], eng.kv.v[li], k, v, self.slots)
+            eng._kv_append(li, k, v, self.slots)
             attn = ops.paged_attn_decode(
                 q, eng.kv.k[li], eng.kv.v[li], self.bt, self.seq_lens,
-                eng.scale, n_parts=self.n_parts, workspace=self.ws)
+                eng.scale, n_parts=self.n_parts, workspace=self.ws,
+                k_scale=(eng.kv.k_scale[li] if eng.kv.fp8 else None),
+                v_scale=(eng.kv.v_scale[li] if eng.kv.fp8 else None))
             h, residual = eng.model.layer_attn_epilogue(li, attn, residual)
         return eng.model.final(h, residual)
 
@@ -195,9 +202,12 @@ class Engine:
             self.model.load_safetensors(cfg.weights_path)
         m = cfg.model
         n_kv_local = m.kv_heads // self.model.tp.world_size
+        import os as _os0
+        self.kv_dtype = _os0.environ.get("LLMLB_KV_DTYPE",
+                                         cfg.kv_dtype) or "bfloat16"
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)
         self.kv = KVCache(m.layers, num_blocks, n_kv_local, m.head_dim,
-                          self.device, dtype)
+                          self.device, dtype, kv_dtype=self.kv_dtype)
         import os as _os
         self.spec_ngram = int(_os.environ.get("LLMLB_SPEC_NGRAM",
                                               str(cfg.spec_ngram)) or 0)
@@ -255,7 +265,9 @@ class Engine:
     # -- sizing ----------------------------------------------------------
     def _auto_blocks(self, n_kv_local: int) -> int:
         m = self.cfg.model
-        per_block = KVCache.bytes_per_block(m.layers, n_kv_local, m.head_dim)
+        per_block = KVCache.bytes_per_block(m.layers, n_kv_local,
+                                            m.head_dim,
+                                            kv_dtype=self.kv_dtype)
         if self.device.type == "cuda":
             free, _total = torch.cuda.mem_get_info(self.device)
             budget = int(free * self.cfg.hbm_fraction)
@@ -383,6 +395,13 @@ class Engine:
         return bool(prefills or self.running)
 
     # -- forward passes --------------------------------------------------
+    def _kv_append(self, li: int, k, v, slots) -> None:
+        ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots,
+                      k_scale=(self.kv.k_scale[li] if self.kv.fp8
+                               else None),
+                      v_scale=(self.kv.v_scale[li] if self.kv.fp8
+                               else None))
+
     def _slots_for(self, seq: _Sequence, start: int, n: int) -> list[int]:
         return [seq.block_table[(start + i) // BLOCK_SIZE] * BLOCK_SIZE
                 + (start + i) % BLOCK_SIZE for i in range(n)]
@@ -479,16 +498,17 @@ class Engine:
         for li in range(self.cfg.model.layers):
             q, k, v, residual = self.model.layer_attn_prelude(
                 li, h, residual, positions)
-            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+            self._kv_append(li, k, v, slots)
             qh = q.view(B, Ls, nq, D).transpose(1, 2)
             kh = k.view(B, Ls, nkv, D)
             vh = v.view(B, Ls, nkv, D)
             if P:
                 # gather this layer's cached K/V for positions [0, P)
-                kpre = self.kv.k[li][pblks, :, poffs].view(B, P, nkv, D)
-                vpre = self.kv.v[li][pblks, :, poffs].view(B, P, nkv, D)
-                kh = torch.cat([kpre, kh], dim=1)
-                vh = torch.cat([vpre, vh], dim=1)
+                kpre, vpre = self.kv.gather(li, pblks, poffs)
+                kh = torch.cat([kpre.view(B, P, nkv, D).to(kh.dtype), kh],
+                               dim=1)
+                vh = torch.cat([vpre.view(B, P, nkv, D).to(vh.dtype), vh],
+                               dim=1)
             kh, vh = kh.transpose(1, 2), vh.transpose(1, 2)
             o = torch.nn.functional.scaled_dot_product_attention(
                 qh, kh, vh,
@@ -556,10 +576,14 @@ class Engine:
             for li in range(self.cfg.model.layers):
                 q, k, v, residual = self.model.layer_attn_prelude(
                     li, h, residual, lens)
-                ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
-                attn = ops.paged_attn_decode(q, self.kv.k[li],
-                                             self.kv.v[li], bt, seq_lens,
-                                             self.scale)
+                self._kv_append(li, k, v, slots)
+                attn = ops.paged_attn_decode(
+                    q, self.kv.k[li], self.kv.v[li], bt, seq_lens,
+                    self.scale,
+                    k_scale=(self.kv.k_scale[li] if self.kv.fp8
+                             else None),
+                    v_scale=(self.kv.v_scale[li] if self.kv.fp8
+                             else None))
                 h, residual = self.model.layer_attn_epilogue(li, attn,
                                                              residual)
             logits = self.model.final(h, residual)
@@ -682,10 +706,11 @@ class Engine:
         for li in range(self.cfg.model.layers):
             q, k, v, residual = self.model.layer_attn_prelude(
                 li, h, residual, positions)
-            ops.kv_append(self.kv.k[li], self.kv.v[li], k, v, slots)
+            self._kv_append(li, k, v, slots)
+            kg, vg = self.kv.gather(li, gblks, goffs)
             qh = q.view(B, F, nq, D).transpose(1, 2)
-            kh = self.kv.k[li][gblks, :, goffs].view(B, Lmax, nkv, D)                 .transpose(1, 2)
-            vh = self.kv.v[li][gblks, :, goffs].view(B, Lmax, nkv, D)                 .transpose(1, 2)
+            kh = kg.view(B, Lmax, nkv, D).to(q.dtype).transpose(1, 2)
+            vh = vg.view(B, Lmax, nkv, D).to(q.dtype).transpose(1, 2)
             o = torch.nn.functional.scaled_dot_product_attention(
                 qh, kh, vh, attn_mask=mask, enable_gqa=True)
             attn = o.transpose(1, 2).reshape(B * F, nq, D).contiguous()

@@ -43,6 +43,8 @@ def _try_load() -> None:
     _lib.ll_swiglu.restype = ctypes.c_int
     _lib.ll_kv_append.restype = ctypes.c_int
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
+    _lib.ll_paged_attn_decode_fp8.restype = ctypes.c_int
+    _lib.ll_kv_append_fp8.restype = ctypes.c_int
     _lib.ll_decode_gemm.restype = ctypes.c_int
     _lib.ll_decode_gemm_v2.restype = ctypes.c_int
 
@@ -152,9 +154,29 @@ def swiglu(gateup: torch.Tensor) -> torch.Tensor:
 
 
 def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
-              v: torch.Tensor, slots: torch.Tensor) -> None:
-    """Scatter k/v [T, n_kv, D] into paged caches at flat slot ids."""
+              v: torch.Tensor, slots: torch.Tensor,
+              k_scale: torch.Tensor | None = None,
+              v_scale: torch.Tensor | None = None) -> None:
+    """Scatter k/v [T, n_kv, D] into paged caches at flat slot ids.
+    uint8 caches are the FP8 (OCP e4m3) path: values are quantized with
+    one scale per (page, head, token) written to k_scale/v_scale."""
     block_size = kcache.shape[2]
+    if kcache.dtype == torch.uint8:
+        if not kcache.is_cuda:
+            reference.kv_append_fp8(kcache, vcache, k_scale, v_scale,
+                                    k, v, slots, block_size)
+            return
+        lib = _require_native()
+        assert k.stride(-1) == 1 and k.stride(1) == k.shape[2]
+        slots_i = slots.to(torch.int32)
+        rc = lib.ll_kv_append_fp8(
+            _ptr(kcache), _ptr(vcache), _ptr(k_scale), _ptr(v_scale),
+            _ptr(k), _ptr(v), _ptr(slots_i), ctypes.c_int(k.shape[0]),
+            ctypes.c_int(k.shape[1]), ctypes.c_int(k.shape[2]),
+            ctypes.c_int(block_size), ctypes.c_int(k.stride(0)),
+            ctypes.c_int(v.stride(0)), _stream())
+        _check(rc, "ll_kv_append_fp8")
+        return
     if not kcache.is_cuda or kcache.dtype != torch.bfloat16:
         reference.kv_append(kcache, vcache, k, v, slots, block_size)
         return
@@ -287,7 +309,10 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
                       vcache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float,
                       n_parts: int | None = None,
-                      workspace: tuple | None = None) -> torch.Tensor:
+                      workspace: tuple | None = None,
+                      k_scale: torch.Tensor | None = None,
+                      v_scale: torch.Tensor | None = None
+                      ) -> torch.Tensor:
     """Decode attention over the paged KV cache.
 
     q: [S, n_q, D] bf16; caches [num_blocks, n_kv, block_size, D].
@@ -295,6 +320,11 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
     chosen automatically to fill the 256-CU chip when not given.
     """
     if not q.is_cuda or q.dtype != torch.bfloat16:
+        if kcache.dtype == torch.uint8:
+            kd, vd = reference.dequant_fp8_cache(kcache, k_scale), \
+                reference.dequant_fp8_cache(vcache, v_scale)
+            return reference.paged_attn_decode(q, kd, vd, block_tables,
+                                               seq_lens, scale)
         return reference.paged_attn_decode(q, kcache, vcache, block_tables,
                                            seq_lens, scale)
     lib = _require_native()
@@ -328,6 +358,18 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         else block_tables.to(torch.int32)
     sl = seq_lens if seq_lens.dtype == torch.int32 \
         else seq_lens.to(torch.int32)
+    if kcache.dtype == torch.uint8:
+        assert use_v2, "FP8 KV requires the v2 attention shape " \
+            "(D=128, GQA group <= 4)"
+        rc = lib.ll_paged_attn_decode_fp8(
+            _ptr(out), acc_p, ml_p, _ptr(q), _ptr(kcache), _ptr(vcache),
+            _ptr(k_scale), _ptr(v_scale), _ptr(bt), _ptr(sl),
+            ctypes.c_int(S), ctypes.c_int(bt.shape[1]),
+            ctypes.c_int(n_q), ctypes.c_int(n_kv), ctypes.c_int(D),
+            ctypes.c_int(block_size), ctypes.c_int(n_parts),
+            ctypes.c_float(scale), ctypes.c_int(q.stride(0)), _stream())
+        _check(rc, "ll_paged_attn_decode_fp8")
+        return out
     rc = lib.ll_paged_attn_decode(
         _ptr(out), acc_p, ml_p, _ptr(q), _ptr(kcache), _ptr(vcache),
         _ptr(bt), _ptr(sl), ctypes.c_int(S), ctypes.c_int(bt.shape[1]),

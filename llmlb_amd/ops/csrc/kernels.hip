@@ -21,6 +21,7 @@
 // Compile: hipcc --offload-arch=gfx950 -O3 -std=c++17 -shared -fPIC
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 #include <math.h>
 #include <stdint.h>
 
@@ -245,6 +246,85 @@ __global__ void kv_append_kernel(bf16* __restrict__ kcache,
 }
 
 // ---------------------------------------------------------------------------
+// FP8 (OCP e4m3) KV path: K/V stored as e4m3 bytes with one fp32
+// scale per (page, head, token) — half the attention bandwidth and
+// double the resident context.  gfx950's FP8 is OCP e4m3fn (the MI300X
+// fnuz variant is gone), so bytes round-trip exactly with torch's
+// float8_e4m3fn.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float e4m3_to_f32(int b) {
+  const int e = (b >> 3) & 0xF, m = b & 7;
+  float v;
+  if (e == 0)
+    v = ldexpf((float)m / 8.f, -6);
+  else if (e == 15 && m == 7)
+    v = __int_as_float(0x7fc00000);           // e4m3fn: only NaN, no inf
+  else
+    v = ldexpf(1.f + (float)m / 8.f, e - 7);
+  return (b & 0x80) ? -v : v;
+}
+
+// round-to-nearest-even f32 -> e4m3 byte (input pre-scaled to +-448)
+__device__ __forceinline__ unsigned char f32_to_e4m3(float f) {
+  __hip_fp8_e4m3 q(f);
+  return q.__x;
+}
+
+// one wave per (token, head): amax over D, scale = amax/448, quantize
+__global__ __launch_bounds__(256)
+void kv_append_fp8_kernel(unsigned char* __restrict__ kcache,
+                          unsigned char* __restrict__ vcache,
+                          float* __restrict__ k_scale,
+                          float* __restrict__ v_scale,
+                          const bf16* __restrict__ k,
+                          const bf16* __restrict__ v,
+                          const int* __restrict__ slots,
+                          int n_tokens, int n_kv, int D, int block_size,
+                          int k_rstride, int v_rstride) {
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  if (wave_id >= n_tokens * n_kv) return;
+  const int t = wave_id / n_kv;
+  const int h = wave_id % n_kv;
+  const int slot = slots[t];
+  if (slot < 0) return;
+  const int blk = slot / block_size;
+  const int off = slot % block_size;
+  const int per_lane = D / WAVE;               // 2 at D=128
+  const size_t dst = (((size_t)blk * n_kv + h) * block_size + off) * D;
+  const size_t sidx = ((size_t)blk * n_kv + h) * block_size + off;
+#pragma unroll
+  for (int which = 0; which < 2; ++which) {
+    const bf16* src = which ? v : k;
+    const int rstride = which ? v_rstride : k_rstride;
+    unsigned char* cache = which ? vcache : kcache;
+    float* scale_out = which ? v_scale : k_scale;
+    float x[4];
+    float amax = 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      if (i < per_lane) {
+        x[i] = bf2f(__builtin_bit_cast(
+            unsigned short,
+            src[(size_t)t * rstride + (size_t)h * D + lane * per_lane +
+                i]));
+        amax = fmaxf(amax, fabsf(x[i]));
+      }
+    }
+#pragma unroll
+    for (int o = 32; o > 0; o >>= 1)
+      amax = fmaxf(amax, __shfl_xor(amax, o, WAVE));
+    const float scale = amax > 0.f ? amax / 448.f : 1.f;
+    const float inv = 1.f / scale;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      if (i < per_lane)
+        cache[dst + lane * per_lane + i] = f32_to_e4m3(x[i] * inv);
+    if (lane == 0) scale_out[sidx] = scale;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Paged decode attention (one query token per sequence), GQA, bf16 KV.
 //
 //   q:        [num_seqs, n_q, D] bf16
@@ -428,20 +508,28 @@ void paged_attn_decode_kernel(
 // within the workgroup), then across grid partitions by the shared
 // paged_attn_reduce_kernel.
 // ---------------------------------------------------------------------------
-template <int GROUP>
+template <int GROUP, bool KV8 = false>
 __global__ __launch_bounds__(PA_WAVES * WAVE)
 void paged_attn_decode_v2_kernel(
     bf16* __restrict__ out,              // [num_seqs, n_q, D]
     float* __restrict__ ws_acc,          // [num_seqs, n_q, n_parts, D]
     float* __restrict__ ws_ml,           // [num_seqs, n_q, n_parts, 2]
     const bf16* __restrict__ q,
-    const bf16* __restrict__ kcache,
+    const bf16* __restrict__ kcache,     // e4m3 bytes when KV8
     const bf16* __restrict__ vcache,
+    const float* __restrict__ k_scale,   // per (page, head, token); KV8
+    const float* __restrict__ v_scale,
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     int n_q, int n_kv, int max_blocks, int n_parts, float scale,
     int q_rstride) {
   constexpr int D = 128;
+  __shared__ float s_lut[256];           // e4m3 byte -> float
+  if (KV8) {
+    for (int i = threadIdx.x; i < 256; i += blockDim.x)
+      s_lut[i] = e4m3_to_f32(i);
+    __syncthreads();
+  }
   constexpr int DS = 16;                 // dim-slice (8 lanes per token)
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
@@ -491,12 +579,20 @@ void paged_attn_decode_v2_kernel(
       const int trow = half * 8 + t_local;
       const int tok = b * PA_BLOCK + trow;
       const bool valid = tok < seq_len;
-      const bf16* kb =
-          kcache + (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow) * D +
-          p * DS;
+      const size_t row = ((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow;
       float kf[DS];
-      {
-        const uint4* kv4 = reinterpret_cast<const uint4*>(kb);
+      if (KV8) {
+        const uint4 raw = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const unsigned char*>(kcache) + row * D +
+            p * DS);
+        const unsigned char* by =
+            reinterpret_cast<const unsigned char*>(&raw);
+        const float sc = k_scale[row];
+#pragma unroll
+        for (int j = 0; j < DS; ++j) kf[j] = s_lut[by[j]] * sc;
+      } else {
+        const uint4* kv4 = reinterpret_cast<const uint4*>(
+            kcache + row * D + p * DS);
         float f[8];
         unpack8(kv4[0], f);
 #pragma unroll
@@ -519,11 +615,18 @@ void paged_attn_decode_v2_kernel(
       }
       // block max over the 8 tokens of this pass (lane bits 3..5)
       float vf[DS];
-      {
-        const bf16* vb =
-            vcache +
-            (((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow) * D + p * DS;
-        const uint4* vv4 = reinterpret_cast<const uint4*>(vb);
+      if (KV8) {
+        const uint4 raw = *reinterpret_cast<const uint4*>(
+            reinterpret_cast<const unsigned char*>(vcache) + row * D +
+            p * DS);
+        const unsigned char* by =
+            reinterpret_cast<const unsigned char*>(&raw);
+        const float sc = v_scale[row];
+#pragma unroll
+        for (int j = 0; j < DS; ++j) vf[j] = s_lut[by[j]] * sc;
+      } else {
+        const uint4* vv4 = reinterpret_cast<const uint4*>(
+            vcache + row * D + p * DS);
         float f[8];
         unpack8(vv4[0], f);
 #pragma unroll
@@ -756,6 +859,25 @@ int ll_kv_append(void* kcache, void* vcache, const void* k, const void* v,
   return 0;
 }
 
+int ll_kv_append_fp8(void* kcache, void* vcache, void* k_scale,
+                     void* v_scale, const void* k, const void* v,
+                     const void* slots, int n_tokens, int n_kv, int D,
+                     int block_size, int k_rstride, int v_rstride,
+                     void* stream) {
+  if (D != 128) return -2;
+  const long long waves = (long long)n_tokens * n_kv;
+  const int block = 256;
+  const long long grid = (waves * WAVE + block - 1) / block;
+  hipLaunchKernelGGL(kv_append_fp8_kernel, dim3((unsigned)grid),
+                     dim3(block), 0, (hipStream_t)stream,
+                     (unsigned char*)kcache, (unsigned char*)vcache,
+                     (float*)k_scale, (float*)v_scale, (const bf16*)k,
+                     (const bf16*)v, (const int*)slots, n_tokens, n_kv,
+                     D, block_size, k_rstride, v_rstride);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
 int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                          const void* kcache, const void* vcache,
                          const void* block_tables, const void* seq_lens,
@@ -779,6 +901,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            0, s, (bf16*)out, (float*)ws_acc,
                            (float*)ws_ml, (const bf16*)q,
                            (const bf16*)kcache, (const bf16*)vcache,
+                           nullptr, nullptr,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
                            n_parts, scale, q_rstride);
@@ -788,6 +911,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            0, s, (bf16*)out, (float*)ws_acc,
                            (float*)ws_ml, (const bf16*)q,
                            (const bf16*)kcache, (const bf16*)vcache,
+                           nullptr, nullptr,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
                            n_parts, scale, q_rstride);
@@ -797,6 +921,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                            0, s, (bf16*)out, (float*)ws_acc,
                            (float*)ws_ml, (const bf16*)q,
                            (const bf16*)kcache, (const bf16*)vcache,
+                           nullptr, nullptr,
                            (const int*)block_tables,
                            (const int*)seq_lens, n_q, n_kv, max_blocks,
                            n_parts, scale, q_rstride);
@@ -828,6 +953,52 @@ v1_path:
     hipLaunchKernelGGL(paged_attn_reduce_kernel, rgrid, rblock, 0, s,
                        (bf16*)out, (const float*)ws_acc, (const float*)ws_ml,
                        (const int*)seq_lens, n_q, D, n_parts);
+    LL_CHECK(hipGetLastError());
+  }
+  return 0;
+}
+
+// FP8-KV decode attention: v2 kernel only (D=128, group<=4); same
+// workspace/reduce and single-partition LDS-combine semantics as bf16
+int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
+                             const void* q, const void* kcache,
+                             const void* vcache, const void* k_scale,
+                             const void* v_scale,
+                             const void* block_tables,
+                             const void* seq_lens, int num_seqs,
+                             int max_blocks, int n_q, int n_kv, int D,
+                             int block_size, int n_parts, float scale,
+                             int q_rstride, void* stream) {
+  if (block_size != PA_BLOCK || D != 128) return -2;
+  if (n_q % n_kv != 0) return -2;
+  const int group = n_q / n_kv;
+  if (group > 4 || (group == 3)) return -2;
+  if (n_parts > 1 && ws_acc == nullptr) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
+  const int slices = n_parts * PA_WAVES;
+#define LAUNCH_V2_FP8(G)                                                    \
+  hipLaunchKernelGGL((paged_attn_decode_v2_kernel<G, true>), grid, block,   \
+                     0, s, (bf16*)out, (float*)ws_acc, (float*)ws_ml,       \
+                     (const bf16*)q, (const bf16*)kcache,                   \
+                     (const bf16*)vcache, (const float*)k_scale,            \
+                     (const float*)v_scale, (const int*)block_tables,       \
+                     (const int*)seq_lens, n_q, n_kv, max_blocks,           \
+                     n_parts, scale, q_rstride)
+  switch (group) {
+    case 1: LAUNCH_V2_FP8(1); break;
+    case 2: LAUNCH_V2_FP8(2); break;
+    case 4: LAUNCH_V2_FP8(4); break;
+    default: return -2;
+  }
+#undef LAUNCH_V2_FP8
+  LL_CHECK(hipGetLastError());
+  if (n_parts > 1) {
+    dim3 rgrid(num_seqs, n_q), rblock(WAVE);
+    hipLaunchKernelGGL(paged_attn_reduce_kernel, rgrid, rblock, 0, s,
+                       (bf16*)out, (const float*)ws_acc,
+                       (const float*)ws_ml, (const int*)seq_lens, n_q, D,
+                       slices);
     LL_CHECK(hipGetLastError());
   }
   return 0;

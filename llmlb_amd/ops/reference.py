@@ -83,6 +83,37 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
     vcache[blk, :, off, :] = v[valid]
 
 
+def quantize_fp8(x: torch.Tensor):
+    """Per-(row=last-2-dims-leading) e4m3 quantization used by the FP8
+    KV cache: one scale per [..., D] vector, scale = amax/448."""
+    amax = x.float().abs().amax(dim=-1, keepdim=True)
+    scale = torch.where(amax > 0, amax / 448.0, torch.ones_like(amax))
+    q = (x.float() / scale).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale.squeeze(-1).float()
+
+
+def dequant_fp8_cache(cache_u8: torch.Tensor,
+                      scale: torch.Tensor) -> torch.Tensor:
+    """[nb, n_kv, bs, D] uint8 + [nb, n_kv, bs] scale -> bf16."""
+    f = cache_u8.view(torch.float8_e4m3fn).to(torch.float32)
+    return (f * scale.unsqueeze(-1)).to(torch.bfloat16) \
+        if cache_u8.is_cuda else (f * scale.unsqueeze(-1))
+
+
+def kv_append_fp8(kcache, vcache, k_scale, v_scale, k, v, slots,
+                  block_size):
+    valid = slots >= 0
+    idx = slots[valid].long()
+    blk = idx // block_size
+    off = idx % block_size
+    kq, ks = quantize_fp8(k[valid])
+    vq, vs = quantize_fp8(v[valid])
+    kcache[blk, :, off, :] = kq
+    vcache[blk, :, off, :] = vq
+    k_scale[blk, :, off] = ks
+    v_scale[blk, :, off] = vs
+
+
 def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
                       vcache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float) -> torch.Tensor:

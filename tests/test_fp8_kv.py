@@ -1,0 +1,105 @@
+"""FP8 (OCP e4m3) KV cache: quantization round trip, CPU engine path
+through the reference ops, and GPU kernel numerics (marked gpu)."""
+import pytest
+import torch
+
+import llmlb_amd.ops as ops
+from llmlb_amd.ops import reference
+from llmlb_amd.backend.engine import Engine, EngineConfig
+from llmlb_amd.backend.sampler import SamplingParams
+from llmlb_amd.models.llama import LlamaConfig
+
+
+def test_quantize_round_trip_cpu():
+    x = torch.randn(8, 4, 128) * 3
+    q, scale = reference.quantize_fp8(x)
+    deq = q.view(torch.float8_e4m3fn).float() * scale.unsqueeze(-1)
+    rel = (deq - x).abs().max() / x.abs().max()
+    assert rel < 0.07, rel.item()     # e4m3 step ~6% of amax
+
+
+def test_fp8_engine_cpu():
+    """Engine with fp8 KV on CPU (reference ops): prefill is exact (it
+    attends dense K/V), decode reads the dequantized cache."""
+    cfg = EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                       dtype="float32", kv_cache_blocks=64,
+                       kv_dtype="fp8_e4m3", seed=5)
+    eng = Engine(cfg)
+    assert eng.kv.fp8
+    p = SamplingParams(temperature=0.0)
+    out = eng.generate_sync([2, 7, 1, 8, 2, 8], 12, p).token_ids
+    assert len(out) == 12
+    ref_eng = Engine(EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                                  dtype="float32", kv_cache_blocks=64,
+                                  seed=5))
+    ref = ref_eng.generate_sync([2, 7, 1, 8, 2, 8], 12, p).token_ids
+    # the first token depends only on the exact dense prefill
+    assert out[0] == ref[0]
+    # and the stream stays plausible (same vocab, mostly agrees for a
+    # tiny model whose margins are wide)
+    agree = sum(a == b for a, b in zip(out, ref))
+    assert agree >= 8, (out, ref)
+
+
+@pytest.mark.gpu
+def test_fp8_append_and_attention_gpu():
+    torch.manual_seed(2)
+    dev = "cuda:0"
+    T, nkv, nq, D, BS, nb = 48, 2, 8, 128, 16, 8
+    k = torch.randn(T, nkv, D, dtype=torch.bfloat16, device=dev)
+    v = torch.randn(T, nkv, D, dtype=torch.bfloat16, device=dev)
+    kc = torch.zeros(nb, nkv, BS, D, dtype=torch.uint8, device=dev)
+    vc = torch.zeros_like(kc)
+    ks = torch.zeros(nb, nkv, BS, dtype=torch.float32, device=dev)
+    vs = torch.zeros_like(ks)
+    slots = torch.arange(16, 16 + T, dtype=torch.int32, device=dev)
+    ops.kv_append(kc, vc, k, v, slots, k_scale=ks, v_scale=vs)
+    # quantization matches the torch e4m3 reference within 1 ULP
+    kc_ref = torch.zeros_like(kc).cpu()
+    vc_ref = torch.zeros_like(vc).cpu()
+    ks_ref = torch.zeros_like(ks).cpu()
+    vs_ref = torch.zeros_like(vs).cpu()
+    reference.kv_append_fp8(kc_ref, vc_ref, ks_ref, vs_ref, k.cpu(),
+                            v.cpu(), slots.cpu(), BS)
+    assert torch.allclose(ks.cpu(), ks_ref, rtol=1e-6)
+    deq_gpu = reference.dequant_fp8_cache(kc, ks).cpu().float()
+    deq_ref = reference.dequant_fp8_cache(kc_ref, ks_ref).float()
+    assert (deq_gpu - deq_ref).abs().max() < 0.05
+
+    # attention over the fp8 cache vs SDPA on the dequantized values
+    S = 3
+    bt = torch.tensor([[1, 2], [2, 3], [3, 4]], dtype=torch.int32,
+                      device=dev)
+    sl = torch.tensor([20, 30, 17], dtype=torch.int32, device=dev)
+    q = torch.randn(S, nq, D, dtype=torch.bfloat16, device=dev)
+    out = ops.paged_attn_decode(q, kc, vc, bt, sl, D ** -0.5,
+                                k_scale=ks, v_scale=vs)
+    kd = reference.dequant_fp8_cache(kc, ks)
+    vd = reference.dequant_fp8_cache(vc, vs)
+    ref = reference.paged_attn_decode(q.float().cpu(),
+                                      kd.cpu().float(),
+                                      vd.cpu().float(),
+                                      bt.cpu(), sl.cpu(), D ** -0.5)
+    rel = (out.cpu().float() - ref).abs().max() / ref.abs().max()
+    assert rel < 0.02, rel.item()
+
+
+@pytest.mark.gpu
+def test_fp8_engine_gpu_decode():
+    """End-to-end fp8-KV engine on GPU: greedy decode runs through the
+    fp8 kernels (graphs on) and stays close to the bf16-KV engine."""
+    cfg = LlamaConfig.small()
+    p = SamplingParams(temperature=0.0)
+    outs = {}
+    for kvd in ("bfloat16", "fp8_e4m3"):
+        eng = Engine(EngineConfig(model=cfg, device="cuda:0",
+                                  dtype="bfloat16", kv_cache_blocks=256,
+                                  kv_dtype=kvd, seed=3))
+        outs[kvd] = eng.generate_sync(list(range(24)), 16, p).token_ids
+        del eng
+        torch.cuda.empty_cache()
+    assert len(outs["fp8_e4m3"]) == 16
+    assert outs["fp8_e4m3"][0] == outs["bfloat16"][0]
+    agree = sum(a == b for a, b in zip(outs["fp8_e4m3"],
+                                       outs["bfloat16"]))
+    assert agree >= 10, outs
