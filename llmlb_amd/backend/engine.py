@@ -205,6 +205,13 @@ class Engine:
         import os as _os0
         self.kv_dtype = _os0.environ.get("LLMLB_KV_DTYPE",
                                          cfg.kv_dtype) or "bfloat16"
+        if (self.kv_dtype == "fp8_e4m3" and self.device.type == "cuda"
+                and (m.head_dim != 128
+                     or m.q_heads // m.kv_heads > 4)):
+            raise ValueError(
+                "fp8_e4m3 KV on GPU requires head_dim=128 and GQA "
+                f"group <= 4 (got D={m.head_dim}, "
+                f"group={m.q_heads // m.kv_heads})")
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)
         self.kv = KVCache(m.layers, num_blocks, n_kv_local, m.head_dim,
                           self.device, dtype, kv_dtype=self.kv_dtype)

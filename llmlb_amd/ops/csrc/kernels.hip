@@ -252,6 +252,15 @@ __global__ void kv_append_kernel(bf16* __restrict__ kcache,
 // fnuz variant is gone), so bytes round-trip exactly with torch's
 // float8_e4m3fn.
 // ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(2))) float f32x2_t;
+
+// one dword of packed e4m3 -> 4 floats (gfx950 native converts)
+__device__ __forceinline__ void fp8x4_to_f32(unsigned int d, float* o) {
+  f32x2_t lo = __builtin_amdgcn_cvt_pk_f32_fp8(d, false);
+  f32x2_t hi = __builtin_amdgcn_cvt_pk_f32_fp8(d, true);
+  o[0] = lo.x; o[1] = lo.y; o[2] = hi.x; o[3] = hi.y;
+}
+
 __device__ __forceinline__ float e4m3_to_f32(int b) {
   const int e = (b >> 3) & 0xF, m = b & 7;
   float v;
@@ -524,12 +533,6 @@ void paged_attn_decode_v2_kernel(
     int n_q, int n_kv, int max_blocks, int n_parts, float scale,
     int q_rstride) {
   constexpr int D = 128;
-  __shared__ float s_lut[256];           // e4m3 byte -> float
-  if (KV8) {
-    for (int i = threadIdx.x; i < 256; i += blockDim.x)
-      s_lut[i] = e4m3_to_f32(i);
-    __syncthreads();
-  }
   constexpr int DS = 16;                 // dim-slice (8 lanes per token)
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
@@ -585,11 +588,13 @@ void paged_attn_decode_v2_kernel(
         const uint4 raw = *reinterpret_cast<const uint4*>(
             reinterpret_cast<const unsigned char*>(kcache) + row * D +
             p * DS);
-        const unsigned char* by =
-            reinterpret_cast<const unsigned char*>(&raw);
         const float sc = k_scale[row];
+        fp8x4_to_f32(raw.x, kf + 0);
+        fp8x4_to_f32(raw.y, kf + 4);
+        fp8x4_to_f32(raw.z, kf + 8);
+        fp8x4_to_f32(raw.w, kf + 12);
 #pragma unroll
-        for (int j = 0; j < DS; ++j) kf[j] = s_lut[by[j]] * sc;
+        for (int j = 0; j < DS; ++j) kf[j] *= sc;
       } else {
         const uint4* kv4 = reinterpret_cast<const uint4*>(
             kcache + row * D + p * DS);
@@ -619,11 +624,13 @@ void paged_attn_decode_v2_kernel(
         const uint4 raw = *reinterpret_cast<const uint4*>(
             reinterpret_cast<const unsigned char*>(vcache) + row * D +
             p * DS);
-        const unsigned char* by =
-            reinterpret_cast<const unsigned char*>(&raw);
         const float sc = v_scale[row];
+        fp8x4_to_f32(raw.x, vf + 0);
+        fp8x4_to_f32(raw.y, vf + 4);
+        fp8x4_to_f32(raw.z, vf + 8);
+        fp8x4_to_f32(raw.w, vf + 12);
 #pragma unroll
-        for (int j = 0; j < DS; ++j) vf[j] = s_lut[by[j]] * sc;
+        for (int j = 0; j < DS; ++j) vf[j] *= sc;
       } else {
         const uint4* vv4 = reinterpret_cast<const uint4*>(
             vcache + row * D + p * DS);

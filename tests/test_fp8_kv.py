@@ -88,7 +88,10 @@ def test_fp8_append_and_attention_gpu():
 def test_fp8_engine_gpu_decode():
     """End-to-end fp8-KV engine on GPU: greedy decode runs through the
     fp8 kernels (graphs on) and stays close to the bf16-KV engine."""
-    cfg = LlamaConfig.small()
+    # fp8 KV needs D=128 / group<=4 on GPU (the v2 attention shape)
+    cfg = LlamaConfig(hidden=512, layers=4, q_heads=4, kv_heads=2,
+                      head_dim=128, inter=1024, vocab=4096, max_pos=1024,
+                      name="fp8-test")
     p = SamplingParams(temperature=0.0)
     outs = {}
     for kvd in ("bfloat16", "fp8_e4m3"):
