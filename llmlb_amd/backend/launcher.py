@@ -29,7 +29,8 @@ PRESETS = {
 def build_engine(model: str, device: str, kv_blocks: int | None = None,
                  max_batch: int = 256, tp_ctx=None,
                  hbm_fraction: float = 0.85,
-                 weights: str | None = None) -> Engine:
+                 weights: str | None = None,
+                 max_prefill_tokens: int | None = None) -> Engine:
     """tp_ctx=None means 'infer from torch.distributed' — pass
     TPContext.solo() for DP replicas inside a larger world."""
     cfg_fn = PRESETS.get(model)
@@ -42,6 +43,8 @@ def build_engine(model: str, device: str, kv_blocks: int | None = None,
                         kv_cache_blocks=kv_blocks, max_batch=max_batch,
                         tp_ctx=tp_ctx, hbm_fraction=hbm_fraction,
                         weights_path=weights)
+    if max_prefill_tokens:
+        ecfg.max_prefill_tokens = max_prefill_tokens
     return Engine(ecfg)
 
 
