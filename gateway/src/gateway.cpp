@@ -337,6 +337,17 @@ void ClientConn::process_requests() {
 }
 
 void ClientConn::dispatch() {
+  // request tracing: honour a client-supplied X-Request-Id, else mint
+  // one; echoed on the response and forwarded upstream
+  std::string_view rid = req.header("x-request-id");
+  if (!rid.empty() && rid.size() <= 64) {
+    request_id.assign(rid);
+  } else {
+    char buf[32];
+    snprintf(buf, sizeof buf, "req_%08x%06llx", (unsigned)rand_r(&w->rng),
+             (unsigned long long)(w->request_seq++ & 0xffffff));
+    request_id = buf;
+  }
   w->gw->handle_request(this);
   if (!op && !response_open) finish_request_cycle();
 }
@@ -395,12 +406,17 @@ void ClientConn::send_simple(int status, std::string_view content_type,
     w->pending_audits.push_back(std::move(a));
   }
   append_status_line(out, status, status_reason(status));
-  char hdr[256];
+  char hdr[320];
   int n = snprintf(hdr, sizeof hdr,
                    "Content-Type: %.*s\r\nContent-Length: %zu\r\n",
                    (int)content_type.size(), content_type.data(),
                    body.size());
   out.append(hdr, n);
+  if (!request_id.empty()) {
+    out.append("X-Request-Id: ");
+    out.append(request_id);
+    out.append("\r\n");
+  }
   if (!extra_headers.empty()) out.append(extra_headers);
   if (want_close) out.append("Connection: close\r\n");
   out.append("\r\n");
@@ -429,6 +445,7 @@ void ClientConn::send_json_error(int status, std::string_view code,
 // ======================================================== Worker
 Worker::Worker(Gateway* gw_, int index_) : gw(gw_), index(index_) {
   acceptor.w = this;
+  rng = (unsigned)(wall_ms() ^ (0x9e3779b9u * (unsigned)(index_ + 1)));
 }
 
 void Worker::Acceptor::on_event(uint32_t) {

@@ -195,6 +195,7 @@ class ClientConn : public FdHandler {
   Worker* w;
   int fd;
   std::string peer_ip;
+  std::string request_id;   // X-Request-Id: client-supplied or generated
   Buffer in, out;
   HttpRequest req;
   bool have_head = false;
@@ -247,6 +248,9 @@ class Worker {
   std::vector<RequestRecord> pending_records;
   std::vector<AuditRecord> pending_audits;
   void arm_record_flush();
+  // request-id minting state (tracing)
+  unsigned rng = 0;
+  unsigned long long request_seq = 0;
   // ops with an armed inference timeout, swept every 250 ms
   std::unordered_set<ClientConn*> timed_ops;
   void arm_op_sweep();

@@ -322,6 +322,11 @@ void start_proxy_direct(Gateway* gw, ClientConn* c, EndpointPtr ep,
                    ep->path_prefix.c_str(), upstream_path.c_str(),
                    ep->host.c_str(), ep->port, body.size());
   uc->out.append(head, n);
+  if (!c->request_id.empty()) {
+    uc->out.append("X-Request-Id: ");
+    uc->out.append(c->request_id);
+    uc->out.append("\r\n");
+  }
   if (!content_type.empty()) {
     uc->out.append("Content-Type: ");
     uc->out.append(content_type);
@@ -486,16 +491,28 @@ void proxy_on_upstream_data(UpstreamConn* uc) {
         append_status_line(c->out, 200, "OK");
         c->out.append(
             "Content-Type: text/event-stream\r\nCache-Control: no-cache\r\n"
-            "Transfer-Encoding: chunked\r\nConnection: keep-alive\r\n\r\n");
+            "Transfer-Encoding: chunked\r\nConnection: keep-alive\r\n");
+        if (!c->request_id.empty()) {
+          c->out.append("X-Request-Id: ");
+          c->out.append(c->request_id);
+          c->out.append("\r\n");
+        }
+        c->out.append("\r\n");
       } else {
         // forward upstream head verbatim minus hop-by-hop Connection
         append_status_line(c->out, op->resp.status, op->resp.reason);
         for (auto& h : op->resp.headers) {
-          if (ieq(h.name, "connection") || ieq(h.name, "keep-alive"))
+          if (ieq(h.name, "connection") || ieq(h.name, "keep-alive") ||
+              ieq(h.name, "x-request-id"))
             continue;
           c->out.append(h.name);
           c->out.append(": ");
           c->out.append(h.value);
+          c->out.append("\r\n");
+        }
+        if (!c->request_id.empty()) {
+          c->out.append("X-Request-Id: ");
+          c->out.append(c->request_id);
           c->out.append("\r\n");
         }
         c->out.append("\r\n");

@@ -130,6 +130,10 @@ class BackendServer:
             "queued_requests": self.engine.waiting.qsize(),
             "tokens_per_sec": round(self.engine.tokens_per_sec_ema, 2),
             "gpu_step_ms": round(self.engine.gpu_step_ms_ema, 3),
+            "spec_decode": (
+                {"proposed": self.engine.spec_proposed,
+                 "accepted": self.engine.spec_accepted}
+                if self.engine.spec_ngram > 0 else None),
             "prefix_cache": (
                 {"hit_tokens": self.engine.prefix.hit_tokens,
                  "prompt_tokens": self.engine.prefix.prompt_tokens,

@@ -373,3 +373,29 @@ def test_dashboard_spa_served(stack):
     # the APIs the page calls stay token-gated
     status, _ = _req("GET", stack["base"] + "/api/dashboard/summary")
     assert status in (401, 403)
+
+
+def test_request_id_tracing(stack):
+    """Every response carries X-Request-Id: echoed when the client
+    supplies one, minted otherwise; streamed responses included."""
+    body = {"model": "mock-model",
+            "messages": [{"role": "user", "content": "hi"}]}
+    r = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions",
+        data=json.dumps(body).encode(),
+        headers={**stack["auth"], "X-Request-Id": "trace-me-42"})
+    with urllib.request.urlopen(r, timeout=10) as resp:
+        assert resp.headers["X-Request-Id"] == "trace-me-42"
+    r2 = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions",
+        data=json.dumps({**body, "stream": True}).encode(),
+        headers=stack["auth"])
+    with urllib.request.urlopen(r2, timeout=10) as resp:
+        rid = resp.headers["X-Request-Id"]
+        assert rid and rid.startswith("req_")
+        resp.read()
+    # local API responses carry one too
+    r3 = urllib.request.Request(stack["base"] + "/v1/models",
+                                headers=stack["auth"])
+    with urllib.request.urlopen(r3, timeout=10) as resp:
+        assert resp.headers["X-Request-Id"]
