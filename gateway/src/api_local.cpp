@@ -401,6 +401,8 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
         o.emplace_back("prompt_tokens", Json(rec.prompt_tokens));
         o.emplace_back("completion_tokens", Json(rec.completion_tokens));
         o.emplace_back("client_ip", Json(rec.client_ip));
+        if (!rec.request_id.empty())
+          o.emplace_back("request_id", Json(rec.request_id));
         if (!rec.error.empty()) o.emplace_back("error", Json(rec.error));
         arr.emplace_back(Json(std::move(o)));
       }

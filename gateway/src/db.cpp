@@ -224,6 +224,10 @@ bool Db::migrate(std::string* err) {
     if (em) sqlite3_free(em);
     return false;
   }
+  // additive migrations (idempotent: ignore duplicate-column errors)
+  sqlite3_exec(db_,
+               "ALTER TABLE request_history ADD COLUMN request_id TEXT",
+               nullptr, nullptr, nullptr);
   return true;
 }
 

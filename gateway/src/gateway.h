@@ -143,6 +143,7 @@ struct ProxyOp {
   std::string endpoint_id;
   ApiKind kind = ApiKind::kChat;
   std::string client_ip;
+  std::string request_id;     // tracing (copied from the client conn)
   long long req_prompt_chars = 0;
   std::string captured_request;   // when cfg.capture_bodies
   bool is_cloud = false;

@@ -65,12 +65,12 @@ void HistoryWriter::flush(std::vector<RequestRecord>& recs,
     db_->exec(
         "INSERT INTO request_history (ts_ms,endpoint_id,model,api_kind,"
         "status,duration_ms,prompt_tokens,completion_tokens,client_ip,"
-        "error,request_body,response_body) VALUES (?,?,?,?,?,?,?,?,?,?,"
-        "?,?)",
+        "error,request_body,response_body,request_id) VALUES "
+        "(?,?,?,?,?,?,?,?,?,?,?,?,?)",
         {(long long)r.ts_ms, r.endpoint_id, r.model, r.api_kind,
          (long long)r.status, r.duration_ms, r.prompt_tokens,
          r.completion_tokens, r.client_ip, r.error, r.request_body,
-         r.response_body});
+         r.response_body, r.request_id});
     // daily stats upsert (db/endpoint_daily_stats.rs)
     char date[16];
     time_t t = (time_t)(r.ts_ms / 1000);

@@ -17,6 +17,7 @@ namespace gw {
 
 struct RequestRecord {
   int64_t ts_ms = 0;
+  std::string request_id;
   std::string request_body;    // captured only when LLMLB_CAPTURE_BODIES
   std::string response_body;
   std::string endpoint_id;

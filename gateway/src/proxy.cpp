@@ -249,6 +249,7 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
   op->endpoint_id = sel.ep->id;
   op->kind = kind;
   op->client_ip = c->peer_ip;
+  op->request_id = c->request_id;
   op->req_prompt_chars = (long long)body.size();
   op->tps = sel.tps;
   op->lease = RequestLease(sel.ep, &registry, sel.canonical, kind);
@@ -308,6 +309,7 @@ void start_proxy_direct(Gateway* gw, ClientConn* c, EndpointPtr ep,
   op->endpoint_id = ep->id;
   op->kind = kind;
   op->client_ip = c->peer_ip;
+  op->request_id = c->request_id;
   op->public_model = raw_string_field(body, "model");
   op->upstream_model = op->public_model;
   op->lease = RequestLease(ep, &gw->registry, op->public_model, kind);
@@ -426,6 +428,7 @@ void Gateway::start_cloud_proxy(ClientConn* c, ApiKind kind,
   op->is_cloud = true;
   op->cloud_provider = provider;
   op->client_ip = c->peer_ip;
+  op->request_id = c->request_id;
   c->op = std::move(op);
   uc->op = c->op.get();
   metrics.cloud_requests.fetch_add(1);
@@ -692,6 +695,7 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
   rec.prompt_tokens = prompt_tokens;
   rec.completion_tokens = completion_tokens;
   rec.client_ip = std::move(op->client_ip);
+  rec.request_id = std::move(op->request_id);
   rec.error = error;
   if (cfg.capture_bodies) {
     rec.request_body = std::move(op->captured_request);

@@ -243,6 +243,26 @@ def test_audio_capability_routing_503_when_absent(stack):
     assert "no_capable_endpoint" in body
 
 
+def test_request_id_in_history(stack):
+    """The minted/echoed request id lands in /api/requests records."""
+    body = {"model": "mock-model",
+            "messages": [{"role": "user", "content": "hi"}]}
+    r = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions",
+        data=json.dumps(body).encode(),
+        headers={**stack["auth"], "X-Request-Id": "hist-trace-7"})
+    urllib.request.urlopen(r, timeout=10).read()
+    deadline = time.time() + 5
+    while time.time() < deadline:
+        _, recs = _req("GET", stack["base"] + "/api/requests",
+                       headers=stack["admin"])
+        if any(rec.get("request_id") == "hist-trace-7"
+               for rec in json.loads(recs)):
+            return
+        time.sleep(0.2)
+    raise AssertionError("request id not recorded")
+
+
 def test_request_history_recorded(stack):
     _req("POST", stack["base"] + "/v1/chat/completions",
          {"model": "mock-model", "messages": []}, headers=stack["auth"])
