@@ -208,7 +208,16 @@ def main() -> None:
     model_name = args.model or os.environ.get(
         "LLMLB_BENCH_MODEL", "llama-3-8b" if have_gpu else "llama-tiny")
     conc = args.conc or int(os.environ.get("LLMLB_BENCH_CONC", "64"))
-    device = f"cuda:{local_rank}" if have_gpu else "cpu"
+    if have_gpu:
+        # one rank per GPU; wrap when oversubscribed (e.g. validating
+        # the multi-rank path on a single-GPU box)
+        dev_idx = local_rank % torch.cuda.device_count()
+        if dev_idx != local_rank:
+            print(f"[bench rank{rank}] oversubscribed: rank "
+                  f"{local_rank} -> cuda:{dev_idx}", flush=True)
+        device = f"cuda:{dev_idx}"
+    else:
+        device = "cpu"
 
     def log(msg):
         print(f"[bench rank{rank}] {msg}", file=sys.stderr, flush=True)
