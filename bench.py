@@ -208,11 +208,14 @@ def main() -> None:
     model_name = args.model or os.environ.get(
         "LLMLB_BENCH_MODEL", "llama-3-8b" if have_gpu else "llama-tiny")
     conc = args.conc or int(os.environ.get("LLMLB_BENCH_CONC", "64"))
+    oversubscribed = False
     if have_gpu:
         # one rank per GPU; wrap when oversubscribed (e.g. validating
         # the multi-rank path on a single-GPU box)
-        dev_idx = local_rank % torch.cuda.device_count()
-        if dev_idx != local_rank:
+        n_dev = torch.cuda.device_count()
+        dev_idx = local_rank % n_dev
+        oversubscribed = world > n_dev
+        if oversubscribed:
             print(f"[bench rank{rank}] oversubscribed: rank "
                   f"{local_rank} -> cuda:{dev_idx}", flush=True)
         device = f"cuda:{dev_idx}"
@@ -252,6 +255,8 @@ def main() -> None:
         torch.cuda.set_device(device)
     kv_blocks = None
     kv_gb = float(os.environ.get("LLMLB_BENCH_KV_GB", "0"))
+    if kv_gb == 0 and oversubscribed:
+        kv_gb = 8.0   # fixed cap: ranks share one GPU in validation
     if kv_gb > 0 and have_gpu:
         from llmlb_amd.backend.kv_cache import KVCache
         from llmlb_amd.backend.launcher import PRESETS
