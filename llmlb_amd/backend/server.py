@@ -397,19 +397,28 @@ def build_asgi_app(server: BackendServer):
                     "type": "invalid_request_error",
                     "code": "model_not_found"}})
                 return
-            sink = _AsyncSink()
+            n_choices = max(1, min(int(body.get("n", 1) or 1), 8))
+            sinks = [_AsyncSink() for _ in range(n_choices)]
             try:
-                if is_chat:
-                    rid, out_q, prompt_ids = server.submit_chat(body, sink)
-                else:
-                    rid, out_q, prompt_ids = server.submit_completion(
-                        body, sink)
+                for ci, sink in enumerate(sinks):
+                    # distinct seeds per choice so sampled choices
+                    # differ (greedy choices are identical by design)
+                    cbody = dict(body)
+                    if body.get("seed") is not None and ci:
+                        cbody["seed"] = int(body["seed"]) + ci
+                    if is_chat:
+                        rid, out_q, prompt_ids = server.submit_chat(
+                            cbody, sink)
+                    else:
+                        rid, out_q, prompt_ids = server.submit_completion(
+                            cbody, sink)
             except MemoryError:
                 await send_json(send, 429, {"error": {
                     "message": "KV cache exhausted",
                     "type": "server_error",
                     "code": "capacity_exceeded"}})
                 return
+            out_q = sinks[0]
             created = int(time.time())
             cid = f"chatcmpl-{uuid.uuid4().hex[:24]}"
             stream = bool(body.get("stream"))
@@ -429,36 +438,40 @@ def build_asgi_app(server: BackendServer):
                         for t, v in extra["top"]]}
 
             if not stream:
-                toks, finish, lp_content = [], "stop", []
-                async for tok, fin, extra in drain_queue_async(out_q):
-                    toks.append(tok)
-                    if extra is not None:
-                        lp_content.append(lp_entry(tok, extra))
-                    if fin:
-                        finish = fin
-                text = "".join(server.tokenizer.decode_token(t)
-                               for t in toks)
+                choices = []
+                total_out = 0
+                for ci, sink_i in enumerate(sinks):
+                    toks, finish, lp_content = [], "stop", []
+                    async for tok, fin, extra in drain_queue_async(
+                            sink_i):
+                        toks.append(tok)
+                        if extra is not None:
+                            lp_content.append(lp_entry(tok, extra))
+                        if fin:
+                            finish = fin
+                    text = "".join(server.tokenizer.decode_token(t)
+                                   for t in toks)
+                    total_out += len(toks)
+                    if is_chat:
+                        choices.append(
+                            {"index": ci,
+                             "message": {"role": "assistant",
+                                         "content": text},
+                             "logprobs": ({"content": lp_content}
+                                          if want_lp else None),
+                             "finish_reason": finish})
+                    else:
+                        choices.append({"index": ci, "text": text,
+                                        "finish_reason": finish})
                 usage = {"prompt_tokens": len(prompt_ids),
-                         "completion_tokens": len(toks),
-                         "total_tokens": len(prompt_ids) + len(toks)}
-                if is_chat:
-                    payload = {
-                        "id": cid, "object": "chat.completion",
-                        "created": created, "model": model,
-                        "choices": [{"index": 0,
-                                     "message": {"role": "assistant",
-                                                 "content": text},
-                                     "logprobs": ({"content": lp_content}
-                                                  if want_lp else None),
-                                     "finish_reason": finish}],
-                        "usage": usage}
-                else:
-                    payload = {
-                        "id": cid, "object": "text_completion",
-                        "created": created, "model": model,
-                        "choices": [{"index": 0, "text": text,
-                                     "finish_reason": finish}],
-                        "usage": usage}
+                         "completion_tokens": total_out,
+                         "total_tokens": len(prompt_ids) + total_out}
+                payload = {
+                    "id": cid,
+                    "object": ("chat.completion" if is_chat
+                               else "text_completion"),
+                    "created": created, "model": model,
+                    "choices": choices, "usage": usage}
                 await send_json(send, 200, payload)
                 return
 
@@ -468,12 +481,13 @@ def build_asgi_app(server: BackendServer):
                                      b"text/event-stream"),
                                     (b"cache-control", b"no-cache")]})
 
-            def chunk_payload(delta: dict, finish=None, usage=None):
+            def chunk_payload(delta: dict, finish=None, usage=None,
+                              index=0):
                 obj = {"id": cid,
                        "object": ("chat.completion.chunk" if is_chat
                                   else "text_completion"),
                        "created": created, "model": model,
-                       "choices": [{"index": 0,
+                       "choices": [{"index": index,
                                     ("delta" if is_chat else "text"):
                                         delta if is_chat else
                                         delta.get("content", ""),
@@ -487,29 +501,36 @@ def build_asgi_app(server: BackendServer):
                             "body": chunk_payload({"role": "assistant"}),
                             "more_body": True})
             ntok = 0
-            finish = "stop"
-            async for tok, fin, extra in drain_queue_async(out_q):
-                ntok += 1
-                if fin:
-                    finish = fin
-                text = server.tokenizer.decode_token(tok)
-                payload = chunk_payload({"content": text})
-                if extra is not None and is_chat:
-                    obj = json.loads(payload[len(b"data: "):])
-                    obj["choices"][0]["logprobs"] = {
-                        "content": [lp_entry(tok, extra)]}
-                    payload = ("data: " + json.dumps(obj)
-                               + "\n\n").encode()
-                await send({"type": "http.response.body",
-                            "body": payload,
-                            "more_body": True})
+            for ci, sink_i in enumerate(sinks):
+                finish = "stop"
+                async for tok, fin, extra in drain_queue_async(sink_i):
+                    ntok += 1
+                    if fin:
+                        finish = fin
+                    text = server.tokenizer.decode_token(tok)
+                    payload = chunk_payload({"content": text}, index=ci)
+                    if extra is not None and is_chat:
+                        obj = json.loads(payload[len(b"data: "):])
+                        obj["choices"][0]["logprobs"] = {
+                            "content": [lp_entry(tok, extra)]}
+                        payload = ("data: " + json.dumps(obj)
+                                   + "\n\n").encode()
+                    await send({"type": "http.response.body",
+                                "body": payload,
+                                "more_body": True})
+                if ci + 1 < len(sinks):
+                    await send({"type": "http.response.body",
+                                "body": chunk_payload({}, finish=finish,
+                                                      index=ci),
+                                "more_body": True})
             usage = ({"prompt_tokens": len(prompt_ids),
                       "completion_tokens": ntok,
                       "total_tokens": len(prompt_ids) + ntok}
                      if include_usage else None)
             await send({"type": "http.response.body",
                         "body": chunk_payload({}, finish=finish,
-                                              usage=usage),
+                                              usage=usage,
+                                              index=len(sinks) - 1),
                         "more_body": True})
             await send({"type": "http.response.body",
                         "body": b"data: [DONE]\n\n", "more_body": False})

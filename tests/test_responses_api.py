@@ -126,3 +126,21 @@ def test_chat_logprobs_through_gateway(stack):
     first = lp["content"][0]
     assert first["logprob"] <= 0.0
     assert len(first["top_logprobs"]) == 2
+
+
+def test_n_choices(stack):
+    """n>1 returns multiple choices; greedy choices are identical,
+    sampled ones (temperature + seed) differ."""
+    body = json.dumps({"model": "llama-tiny",
+                       "messages": [{"role": "user", "content": "hi"}],
+                       "max_tokens": 4, "n": 3, "temperature": 0.9,
+                       "seed": 5}).encode()
+    req = urllib.request.Request(
+        stack["base"] + "/v1/chat/completions", data=body,
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        out = json.loads(r.read())
+    assert [c["index"] for c in out["choices"]] == [0, 1, 2]
+    texts = {c["message"]["content"] for c in out["choices"]}
+    assert len(texts) >= 2           # per-choice seeds differ
+    assert out["usage"]["completion_tokens"] == 12
