@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import argparse
 import asyncio
+import itertools
 import json
 import os
 import socket
@@ -66,19 +67,28 @@ def ensure_kernels() -> None:
 # ---------------------------------------------------------------------
 # async load generator (rank 0): drives the gateway with streaming chat
 # ---------------------------------------------------------------------
+_REQ_UID = itertools.count()
+
+
 async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
                    gen_len: int, api_key: str) -> tuple[int, int, list]:
     import aiohttp
 
-    body = {
-        "model": model,
-        "messages": [{"role": "user",
-                      "content": " ".join(f"w{i}" for i in
-                                          range(prompt_words))}],
-        "max_tokens": gen_len,
-        "stream": True,
-        "stream_options": {"include_usage": True},
-    }
+    def fresh_body():
+        # every request gets a UNIQUE prompt: identical prompts would
+        # let the engine's prefix cache skip prefill work inside the
+        # timed region, which would not be an honest measurement
+        uid = next(_REQ_UID)
+        return {
+            "model": model,
+            "messages": [{"role": "user",
+                          "content": " ".join(f"u{uid}w{i}" for i in
+                                              range(prompt_words))}],
+            "max_tokens": gen_len,
+            "stream": True,
+            "stream_options": {"include_usage": True},
+        }
+
     headers = {"Authorization": f"Bearer {api_key}"}
     tokens = 0
     errors = 0
@@ -90,7 +100,8 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
         t0 = time.monotonic()
         try:
             async with session.post(f"{base_url}/v1/chat/completions",
-                                    json=body, headers=headers) as resp:
+                                    json=fresh_body(),
+                                    headers=headers) as resp:
                 if resp.status != 200:
                     errors += 1
                     detail = (await resp.read())[:200]
@@ -245,14 +256,11 @@ def main() -> None:
             torch.cuda.synchronize()
 
     # ---- backend replica on every rank ------------------------------
+    # The backend (engine + HTTP server) runs in its OWN process: the
+    # load-generation client below is Python too, and sharing a GIL
+    # with the serving side depresses the measured throughput.
     if have_gpu:
         ensure_kernels()
-    from llmlb_amd.backend.launcher import build_engine
-    from llmlb_amd.backend.server import BackendServer, build_asgi_app
-    import uvicorn
-
-    if have_gpu:
-        torch.cuda.set_device(device)
     kv_blocks = None
     kv_gb = float(os.environ.get("LLMLB_BENCH_KV_GB", "0"))
     if kv_gb == 0 and oversubscribed:
@@ -264,23 +272,39 @@ def main() -> None:
         per_block = KVCache.bytes_per_block(
             m.layers, m.kv_heads, m.head_dim)
         kv_blocks = max(int(kv_gb * (1 << 30)) // per_block, 16)
-    from llmlb_amd.parallel.tp import TPContext
-    engine = build_engine(model_name, device,
-                          kv_blocks=kv_blocks,
-                          max_batch=max(conc * 2, 128),
-                          tp_ctx=TPContext.solo())
     backend_port = 9300 + rank
-    served = BackendServer(engine, model_name)
-    app = build_asgi_app(served)
-    engine.start()
-    cfg = uvicorn.Config(app, host="127.0.0.1", port=backend_port,
-                         log_level="error", access_log=False)
-    usrv = uvicorn.Server(cfg)
-    threading.Thread(target=usrv.run, daemon=True).start()
-    deadline = time.time() + 60
-    while time.time() < deadline and not usrv.started:
-        time.sleep(0.1)
-    log(f"backend up on :{backend_port} (started={usrv.started})")
+    be_cmd = [sys.executable, "-m", "llmlb_amd.backend.launcher",
+              "--model", model_name, "--device", device,
+              "--port", str(backend_port),
+              "--max-batch", str(max(conc * 2, 128))]
+    if kv_blocks:
+        be_cmd += ["--kv-blocks", str(kv_blocks)]
+    be_env = os.environ.copy()
+    # the launcher process must not join this torchrun world
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+              "MASTER_PORT", "GROUP_RANK", "LOCAL_WORLD_SIZE",
+              "TORCHELASTIC_USE_AGENT_STORE"):
+        be_env.pop(k, None)
+    be_log = open(f"/tmp/bench_backend_r{rank}.log", "w")
+    be_proc = subprocess.Popen(be_cmd, env=be_env, cwd=str(REPO),
+                               stdout=be_log, stderr=be_log)
+    deadline = time.time() + 300
+    be_up = False
+    while time.time() < deadline:
+        if be_proc.poll() is not None:
+            raise SystemExit(
+                f"backend died; see /tmp/bench_backend_r{rank}.log")
+        try:
+            import urllib.request as _ur
+            with _ur.urlopen("http://127.0.0.1:"
+                             f"{backend_port}/api/health",
+                             timeout=2) as r:
+                r.read()
+            be_up = True
+            break
+        except Exception:
+            time.sleep(0.5)
+    log(f"backend up on :{backend_port} (started={be_up})")
 
     barrier()  # all backends up
 
@@ -427,17 +451,25 @@ def main() -> None:
             }
             print(json.dumps(result), flush=True)
     finally:
-        log(f"engine: steps={engine.steps_done} tokens_out="
-            f"{engine.tokens_out} running={engine.num_running} "
-            f"waiting={engine.waiting.qsize()} tp={engine.model.tp.world_size}")
+        try:
+            import urllib.request as _ur
+            with _ur.urlopen("http://127.0.0.1:"
+                             f"{backend_port}/api/health",
+                             timeout=5) as r:
+                log(f"backend health: {r.read().decode()[:220]}")
+        except Exception as e:
+            log(f"backend health unavailable: {e}")
         hold = float(os.environ.get("LLMLB_BENCH_HOLD", "0"))
         if hold and rank == 0:
             log(f"holding stack for {hold}s (LLMLB_BENCH_HOLD)")
             time.sleep(hold)
         if gw_proc is not None:
             gw_proc.terminate()
-        engine.stop()
-        usrv.should_exit = True
+        be_proc.terminate()
+        try:
+            be_proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            be_proc.kill()
         if dist is not None:
             dist.destroy_process_group()
 
