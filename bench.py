@@ -111,21 +111,29 @@ async def run_wave(base_url: str, model: str, conc: int, prompt_words: int,
                     return
                 usage_tokens = None
                 events = 0
-                async for line in resp.content:
-                    if not line.startswith(b"data:"):
-                        continue
-                    payload = line[5:].strip()
-                    if payload == b"[DONE]":
+                pending = b""
+                done = False
+                async for chunk in resp.content.iter_any():
+                    lines = (pending + chunk).split(b"\n")
+                    pending = lines.pop()
+                    for line in lines:
+                        if not line.startswith(b"data:"):
+                            continue
+                        payload = line[5:].strip()
+                        if payload == b"[DONE]":
+                            done = True
+                            break
+                        events += 1
+                        if b'"usage"' in payload:
+                            try:
+                                obj = json.loads(payload)
+                                u = obj.get("usage") or {}
+                                if "completion_tokens" in u:
+                                    usage_tokens = u["completion_tokens"]
+                            except json.JSONDecodeError:
+                                pass
+                    if done:
                         break
-                    events += 1
-                    if b'"usage"' in payload:
-                        try:
-                            obj = json.loads(payload)
-                            u = obj.get("usage") or {}
-                            if "completion_tokens" in u:
-                                usage_tokens = u["completion_tokens"]
-                        except json.JSONDecodeError:
-                            pass
                 tokens += usage_tokens if usage_tokens is not None \
                     else max(events - 2, 0)
         except Exception as e:
