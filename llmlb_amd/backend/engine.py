@@ -359,7 +359,12 @@ class Engine:
             if len(seq.prompt_ids) > budget and admitted:
                 self.waiting.put(seq)
                 break
-            need = (len(seq.prompt_ids) + BLOCK_SIZE) // BLOCK_SIZE + 1
+            # reserve worst-case pages (prompt + full generation) so a
+            # mid-decode growth can never exhaust the pool and kill the
+            # engine thread; with 288 GB of HBM the utilisation cost is
+            # noise, and parked prefix pages remain evictable
+            need = (len(seq.prompt_ids) + seq.params.max_new_tokens
+                    + BLOCK_SIZE) // BLOCK_SIZE + 1
             if self.prefix is not None:
                 if seq.hashes is None:
                     seq.hashes = PrefixCache.chain_hashes(seq.prompt_ids)
@@ -542,6 +547,8 @@ class Engine:
         grow_rows, grow_cols, grow_vals = [], [], []
         for seq in self.running:
             if seq.length >= len(seq.block_table) * BLOCK_SIZE:
+                if self.prefix is not None:
+                    self.prefix.ensure_free(1)
                 blk = self.kv.allocate(1)[0]
                 seq.block_table.append(blk)
                 grow_rows.append(seq.row)

@@ -149,3 +149,26 @@ def test_logprobs_chat(tiny_engine):
         assert len(extra["top"]) == 3
         # greedy: the chosen token is the top-1 alternative
         assert extra["top"][0][0] == tok
+
+
+def test_capacity_never_kills_engine():
+    """Worst-case page reservation at admission: a flood of requests
+    against a tiny KV pool queues (head-of-line) instead of crashing
+    the engine thread mid-decode, and everything eventually finishes."""
+    from llmlb_amd.models.llama import LlamaConfig
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    eng = Engine(EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                              dtype="float32", kv_cache_blocks=24,
+                              seed=3))
+    p = SamplingParams(temperature=0.0, max_new_tokens=40)
+    sinks = [eng.submit([i, i + 1, i + 2], p)[1] for i in range(12)]
+    for _ in range(3000):
+        busy = eng.step()
+        if not busy and eng.waiting.empty() and not eng.running:
+            break
+    for i, q in enumerate(sinks):
+        toks = []
+        while not q.empty():
+            toks.append(q.get_nowait())
+        assert len(toks) == 40, (i, len(toks))
+    assert eng.kv.free_blocks() > 0
