@@ -256,6 +256,11 @@ bool Gateway::sync_endpoint_models(const EndpointPtr& ep, std::string* err) {
           lower.find("stable-diffusion") != std::string::npos ||
           lower.find("image") != std::string::npos)
         caps = kCapImageGen;
+      // a backend that declares its own capabilities wins over the
+      // name heuristic (our MI355X backend self-reports)
+      if (const Json* cj = m.find("capabilities");
+          cj && cj->is_str() && !cj->as_str().empty())
+        caps = caps_from_csv(cj->as_str());
       em.caps = caps;
       const Json* mt = m.find("max_tokens");
       if (mt) em.max_tokens = mt->as_int(0);

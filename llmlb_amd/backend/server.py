@@ -98,6 +98,8 @@ class BackendServer:
         self.requests_total = 0
         self.tokens_total = 0
         self.start_time = time.time()
+        import threading
+        self.embed_lock = threading.Lock()
 
     # ---------------- request handling (framework-agnostic) ----------
     def models_payload(self) -> dict:
@@ -105,7 +107,9 @@ class BackendServer:
             "object": "list",
             "data": [
                 {"id": m, "object": "model", "owned_by": "llmlb_amd",
-                 "max_tokens": self.engine.cfg.model.max_pos}
+                 "max_tokens": self.engine.cfg.model.max_pos,
+                 "capabilities":
+                     "chat,completions,responses,embeddings"}
                 for m in self.served
             ],
         }
@@ -374,6 +378,42 @@ def build_asgi_app(server: BackendServer):
             await send({"type": "http.response.start", "status": 200,
                         "headers": [(b"content-type", b"text/plain")]})
             await send({"type": "http.response.body", "body": text})
+            return
+
+        if method == "POST" and path == "/v1/embeddings":
+            try:
+                body = json.loads(await read_body(receive) or b"{}")
+            except json.JSONDecodeError:
+                await send_json(send, 400, {"error": {
+                    "message": "invalid JSON",
+                    "type": "invalid_request_error"}})
+                return
+            inputs = body.get("input", "")
+            if isinstance(inputs, str):
+                inputs = [inputs]
+            inputs = [str(x) for x in inputs][:64]
+            import torch as _t
+
+            def _embed():
+                data, ntok = [], 0
+                with server.embed_lock:
+                    for i, text in enumerate(inputs):
+                        ids = server.tokenizer.encode(text)
+                        ntok += len(ids)
+                        t = _t.tensor([ids], dtype=_t.long,
+                                      device=server.engine.device)
+                        h = server.engine.model.forward_dense(t)
+                        vec = h[0, -1].float()
+                        vec = vec / (vec.norm() + 1e-8)
+                        data.append({"object": "embedding", "index": i,
+                                     "embedding": vec.cpu().tolist()})
+                return data, ntok
+
+            data, ntok = await asyncio.to_thread(_embed)
+            await send_json(send, 200, {
+                "object": "list", "data": data,
+                "model": body.get("model", server.model_name),
+                "usage": {"prompt_tokens": ntok, "total_tokens": ntok}})
             return
 
         if method == "POST" and path in ("/v1/chat/completions",

@@ -163,6 +163,35 @@ class LlamaForCausalLM(nn.Module):
             init_(layer.gate_up.weight)
             init_(layer.down.weight)
 
+    @torch.inference_mode()
+    def forward_dense(self, input_ids: torch.Tensor) -> torch.Tensor:
+        """Plain causal forward over a [B, L] batch with NO KV cache —
+        used for embeddings (backend /v1/embeddings): returns the
+        final-norm hidden states [B, L, hidden]."""
+        B, L = input_ids.shape
+        dev = input_ids.device
+        ids = input_ids.reshape(-1)
+        positions = torch.arange(L, dtype=torch.int32,
+                                 device=dev).repeat(B)
+        nq, nkv, D = (self.cfg.q_heads // self.tp.world_size,
+                      self.cfg.kv_heads // self.tp.world_size,
+                      self.cfg.head_dim)
+        h = self.embed_tokens(ids)
+        residual = None
+        for li in range(self.cfg.layers):
+            q, k, v, residual = self.layer_attn_prelude(
+                li, h, residual, positions)
+            qh = q.view(B, L, nq, D).transpose(1, 2)
+            kh = k.view(B, L, nkv, D).transpose(1, 2)
+            vh = v.view(B, L, nkv, D).transpose(1, 2)
+            o = torch.nn.functional.scaled_dot_product_attention(
+                qh, kh, vh, is_causal=L > 1, enable_gqa=True)
+            attn = o.transpose(1, 2).reshape(B * L, nq, D).contiguous()
+            h, residual = self.layer_attn_epilogue(li, attn, residual)
+        final = ops.rmsnorm(h + residual if residual is not None else h,
+                            self.final_norm_w, self.cfg.norm_eps, None)[0]
+        return final.view(B, L, -1)
+
     # -- checkpoint I/O ---------------------------------------------------
 
     @staticmethod

@@ -144,3 +144,27 @@ def test_n_choices(stack):
     texts = {c["message"]["content"] for c in out["choices"]}
     assert len(texts) >= 2           # per-choice seeds differ
     assert out["usage"]["completion_tokens"] == 12
+
+
+def test_embeddings_from_engine(stack):
+    """/v1/embeddings served by the REAL backend (dense forward,
+    unit-normalized last-token hidden state), routed by the gateway."""
+    body = json.dumps({"model": "llama-tiny",
+                       "input": ["hello world", "another text"]}).encode()
+    req = urllib.request.Request(
+        stack["base"] + "/v1/embeddings", data=body,
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        out = json.loads(r.read())
+    assert len(out["data"]) == 2
+    v0 = out["data"][0]["embedding"]
+    assert len(v0) == 256                # tiny hidden size
+    import math
+    assert abs(math.sqrt(sum(x * x for x in v0)) - 1.0) < 1e-3
+    # deterministic: same input -> same vector
+    with urllib.request.urlopen(urllib.request.Request(
+            stack["base"] + "/v1/embeddings", data=body,
+            headers={"Content-Type": "application/json"}),
+            timeout=30) as r:
+        out2 = json.loads(r.read())
+    assert out2["data"][0]["embedding"] == v0
