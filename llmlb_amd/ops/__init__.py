@@ -215,8 +215,12 @@ def _split_k_ws(N: int, M: int, device) -> torch.Tensor:
 
 
 def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
-                   ws: torch.Tensor | None = None) -> torch.Tensor:
-    """Staged-x decode GEMM (v2).  Raises on unsupported shapes."""
+                   ws: torch.Tensor | None = None,
+                   w_scale: torch.Tensor | None = None) -> torch.Tensor:
+    """Staged-x decode GEMM (v2).  Raises on unsupported shapes.
+    ``w_scale`` non-None = fp8-weight arm: ``weight`` holds e4m3 bytes
+    and per-row scales fold into the store (half the weight traffic on
+    the bandwidth-bound decode shapes)."""
     lib = _require_native()
     N, K = x.shape
     M = weight.shape[0]
@@ -226,7 +230,9 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
         ws = _split_k_ws(N, M, x.device)
     rc = lib.ll_decode_gemm_v2(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
-        _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
+        _ptr(weight),
+        _ptr(w_scale) if w_scale is not None else None,
+        ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
         ctypes.c_int(0), _stream())
     if rc == -2:
         raise ValueError("unsupported shape for decode_gemm_v2")
@@ -234,7 +240,21 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     return out
 
 
-def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+_FP8_WEIGHTS = os.environ.get("LLMLB_FP8_WEIGHTS",
+                              "0").lower() in ("1", "true", "on")
+
+
+def quantize_fp8_weight(w: torch.Tensor):
+    """Per-output-row e4m3 quantization of a [M, K] weight: returns
+    (bytes [M, K] u8, scale [M] f32) with scale = row amax / 448."""
+    amax = w.float().abs().amax(dim=1, keepdim=True)
+    scale = torch.where(amax > 0, amax / 448.0, torch.ones_like(amax))
+    q = (w.float() / scale).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), scale.squeeze(1).contiguous()
+
+
+def linear_decode(x: torch.Tensor, weight: torch.Tensor,
+                  w8: tuple | None = None) -> torch.Tensor:
     """out = x @ W^T.
 
     Dispatch policy (measured IN-ENGINE, profiles/r01_attn_lds_and_
@@ -250,9 +270,14 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         return torch.nn.functional.linear(x, weight)
     N, K = x.shape
     M = weight.shape[0]
-    legal = N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 4096
-    if legal and mode == "auto":
-        legal = N <= 16
+    shape_ok = N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 4096
+    if w8 is not None and shape_ok:
+        # fp8 weights halve the dominant traffic: always take them
+        try:
+            return decode_gemm_v2(x, w8[0], w_scale=w8[1])
+        except ValueError:
+            pass
+    legal = shape_ok and (mode != "auto" or N <= 16)
     if legal:
         try:
             return decode_gemm_v2(x, weight)
@@ -293,7 +318,8 @@ def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
     ws = _split_k_ws(N, M, gateup.device) if (M // 128) < 224 else None
     rc = lib.ll_decode_gemm_v2(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(gateup),
-        _ptr(weight), ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
+        _ptr(weight), None,
+        ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),
         ctypes.c_int(1), _stream())
     if rc == -2:
         return None

@@ -50,6 +50,16 @@ __device__ __forceinline__ void dg_unpack8(const uint4 v, float* f) {
   f[4] = dg_bf2f(v.z & 0xffff);  f[5] = dg_bf2f(v.z >> 16);
   f[6] = dg_bf2f(v.w & 0xffff);  f[7] = dg_bf2f(v.w >> 16);
 }
+typedef __attribute__((ext_vector_type(2))) float dg_f32x2;
+
+// one dword of packed e4m3 -> 4 floats (gfx950 native converts)
+__device__ __forceinline__ void dg_fp8x4_to_f32(unsigned int d,
+                                                float* o) {
+  dg_f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(d, false);
+  dg_f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(d, true);
+  o[0] = lo.x; o[1] = lo.y; o[2] = hi.x; o[3] = hi.y;
+}
+
 __device__ __forceinline__ uint4 dg_pack8(const float* f) {
   uint4 v;
   v.x = (unsigned)dg_f2bf(f[0]) | ((unsigned)dg_f2bf(f[1]) << 16);
@@ -248,12 +258,14 @@ int ll_decode_gemm(void* out, const void* x, const void* w, int N, int M,
 // a k-offset and differ in token row) land on 16 distinct bank slots.
 // ===========================================================================
 
-template <int N_TILES, bool ATOMIC, bool SWIGLU = false>
+template <int N_TILES, bool ATOMIC, bool SWIGLU = false, bool W8 = false>
 __global__ __launch_bounds__(512)
 void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
                            float* __restrict__ out_ws,  // [N, M] (SK>1)
                            const bf16* __restrict__ x,  // [N, K]
-                           const bf16* __restrict__ w,  // [M, K]
+                           const bf16* __restrict__ w,  // [M, K] (e4m3
+                                                        // bytes when W8)
+                           const float* __restrict__ w_scale,  // [M]; W8
                            int N, int M, int K, int SK) {
   constexpr int CHUNK_K = 256;              // k per staged chunk
   constexpr int STEPS = CHUNK_K / 32;       // MFMA steps per chunk
@@ -270,7 +282,13 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* xs = reinterpret_cast<bf16*>(smem);  // [N_TILES*16][CHUNK_K] swz
 
-  const bf16* wrow = w + (size_t)(col_base + l_lo) * K;
+  const bf16* wrow = W8 ? nullptr
+                        : w + (size_t)(col_base + l_lo) * K;
+  const unsigned char* wrow8 =
+      W8 ? reinterpret_cast<const unsigned char*>(w) +
+               (size_t)(col_base + l_lo) * K
+         : nullptr;
+  const float wsc = W8 ? w_scale[col_base + l_lo] : 1.f;
   f32x4 acc[N_TILES];
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
@@ -322,7 +340,20 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
 #pragma unroll
     for (int s2 = 0; s2 < STEPS; ++s2) {
       const int kk = k0 + s2 * 32 + l_hi * 8;
-      bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(wrow + kk);
+      bf16x8 bfrag;
+      if (W8) {
+        // 8 e4m3 bytes -> bf16 fragment (native converts); the per-row
+        // scale folds into the store, not the inner loop
+        const uint2 braw =
+            *reinterpret_cast<const uint2*>(wrow8 + kk);
+        float f[8];
+        dg_fp8x4_to_f32(braw.x, f);
+        dg_fp8x4_to_f32(braw.y, f + 4);
+        const uint4 packed = dg_pack8(f);
+        bfrag = *reinterpret_cast<const bf16x8*>(&packed);
+      } else {
+        bfrag = *reinterpret_cast<const bf16x8*>(wrow + kk);
+      }
       const int kb = (s2 * 32 + l_hi * 8) * 2 / 16;  // 16B col in chunk
 #pragma unroll
       for (int t = 0; t < N_TILES; ++t) {
@@ -346,10 +377,10 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
       if (row < N) {
         if (ATOMIC) {
           atomicAdd(out_ws + (size_t)row * M + col_base + l_lo,
-                    acc[t][r]);
+                    W8 ? acc[t][r] * wsc : acc[t][r]);
         } else {
           union { float f; unsigned int i; } c2;
-          c2.f = acc[t][r];
+          c2.f = W8 ? acc[t][r] * wsc : acc[t][r];
           unsigned int bias = 0x7FFF + ((c2.i >> 16) & 1);
           union { unsigned short u; __hip_bfloat16 b; } cv;
           cv.u = (unsigned short)((c2.i + bias) >> 16);
@@ -383,7 +414,8 @@ extern "C" {
 // v2 entry; ws = fp32 workspace [N, M] (required when SK would be > 1,
 // may be null when M/128 >= 224).  Returns -2 on unsupported shape.
 int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
-                      int N, int M, int K, int swiglu_x, void* stream) {
+                      const void* w_scale, int N, int M, int K,
+                      int swiglu_x, void* stream) {
   if (K % 256 != 0 || M % 128 != 0 || N > 128) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
@@ -398,18 +430,22 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
   // it, and the Python side allocates it zeroed
   dim3 grid(mblocks, SK), block(512);
 
-#define LAUNCH_V2_(NT, AT, SW)                                              \
-  hipLaunchKernelGGL((decode_gemm_v2_kernel<NT, AT, SW>), grid, block, lds, \
-                     s, (bf16*)out, (float*)ws, (const bf16*)x,             \
-                     (const bf16*)w, N, M, K, SK)
+  const bool w8 = w_scale != nullptr;
+#define LAUNCH_V2_(NT, AT, SW, W8F)                                         \
+  hipLaunchKernelGGL((decode_gemm_v2_kernel<NT, AT, SW, W8F>), grid,        \
+                     block, lds, s, (bf16*)out, (float*)ws,                 \
+                     (const bf16*)x, (const bf16*)w,                        \
+                     (const float*)w_scale, N, M, K, SK)
 #define LAUNCH_V2(NT)                                                       \
   do {                                                                      \
     if (SK > 1) {                                                           \
-      if (swiglu_x) LAUNCH_V2_(NT, true, true);                             \
-      else LAUNCH_V2_(NT, true, false);                                     \
+      if (swiglu_x) LAUNCH_V2_(NT, true, true, false);                      \
+      else if (w8) LAUNCH_V2_(NT, true, false, true);                       \
+      else LAUNCH_V2_(NT, true, false, false);                              \
     } else {                                                                \
-      if (swiglu_x) LAUNCH_V2_(NT, false, true);                            \
-      else LAUNCH_V2_(NT, false, false);                                    \
+      if (swiglu_x) LAUNCH_V2_(NT, false, true, false);                     \
+      else if (w8) LAUNCH_V2_(NT, false, false, true);                      \
+      else LAUNCH_V2_(NT, false, false, false);                             \
     }                                                                       \
   } while (0)
   switch (n_tiles) {

@@ -68,7 +68,25 @@ def set_tp_context(ctx: TPContext) -> None:
     _default_ctx = ctx
 
 
-class ColumnParallelLinear(nn.Module):
+class _Fp8WeightMixin:
+    """Opt-in fp8 (e4m3) weight storage for the DECODE path
+    (LLMLB_FP8_WEIGHTS=1): per-row-quantized bytes + scales feed the
+    W8 arm of the v2 decode GEMM (half the weight traffic); prefill
+    and large batches keep the bf16 weights + hipBLASLt."""
+
+    _w8 = None
+
+    def _w8_pair(self):
+        from .. import ops as _ops
+        if not _ops._FP8_WEIGHTS or not self.weight.is_cuda \
+                or self.weight.dtype != torch.bfloat16:
+            return None
+        if self._w8 is None:
+            self._w8 = _ops.quantize_fp8_weight(self.weight.data)
+        return self._w8
+
+
+class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
     """Y = X W^T with W split along the output dim; no communication.
 
     Output is the local shard [*, out_features / tp]."""
@@ -86,11 +104,11 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
-            return linear_decode(x, self.weight)
+            return linear_decode(x, self.weight, w8=self._w8_pair())
         return torch.nn.functional.linear(x, self.weight)
 
 
-class RowParallelLinear(nn.Module):
+class RowParallelLinear(nn.Module, _Fp8WeightMixin):
     """Y = X W^T with W split along the input dim; all-reduce on output."""
 
     def __init__(self, in_features: int, out_features: int, tp: TPContext,
@@ -105,7 +123,7 @@ class RowParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
-            y = linear_decode(x, self.weight)
+            y = linear_decode(x, self.weight, w8=self._w8_pair())
         else:
             y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)
