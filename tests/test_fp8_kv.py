@@ -136,9 +136,11 @@ def test_fp8_weights_engine_gpu():
     from llmlb_amd.backend.engine import Engine, EngineConfig
     from llmlb_amd.backend.sampler import SamplingParams
     from llmlb_amd.models.llama import LlamaConfig
-    cfg = LlamaConfig(hidden=512, layers=4, q_heads=4, kv_heads=2,
-                      head_dim=128, inter=1024, vocab=4096, max_pos=1024,
-                      name="w8-test")
+    # real 8B-shaped layers so every projection passes the fp8
+    # dispatch legality (K >= 4096, M % 128 == 0)
+    cfg = LlamaConfig(hidden=4096, layers=2, q_heads=32, kv_heads=8,
+                      head_dim=128, inter=14336, vocab=8192,
+                      max_pos=1024, name="w8-test")
     p = SamplingParams(temperature=0.0)
     outs = {}
     for arm in (False, True):
