@@ -240,6 +240,11 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     return out
 
 
+# EXPERIMENTAL: measured SLOWER than bf16+hipBLASLt on the current v2
+# kernel structure (the e4m3->bf16 converts sit on the load-latency
+# critical path and outweigh the halved weight traffic —
+# profiles/r01_fp8_weights_ab.md).  Kept for the round-2 pipelined
+# staging kernel; default OFF.
 _FP8_WEIGHTS = os.environ.get("LLMLB_FP8_WEIGHTS",
                               "0").lower() in ("1", "true", "on")
 
