@@ -336,35 +336,23 @@ void decode_gemm_v2_kernel(bf16* __restrict__ out,      // [N, M] (bf16)
       }
     }
     __syncthreads();
-    // ---- consume ----------------------------------------------------
-    // Phase 1: issue ALL of this chunk's B loads up front so the HBM
-    // stream overlaps the converts/LDS reads/MFMAs instead of
-    // serialising per step (the measured critical path for the W8 and
-    // fused-epilogue arms, profiles/r01_fp8_weights_ab.md).
-    uint2 braw8[STEPS];
-    bf16x8 bfrags[STEPS];
-    if (W8) {
-#pragma unroll
-      for (int s2 = 0; s2 < STEPS; ++s2)
-        braw8[s2] = *reinterpret_cast<const uint2*>(
-            wrow8 + k0 + s2 * 32 + l_hi * 8);
-    } else {
-#pragma unroll
-      for (int s2 = 0; s2 < STEPS; ++s2)
-        bfrags[s2] = *reinterpret_cast<const bf16x8*>(
-            wrow + k0 + s2 * 32 + l_hi * 8);
-    }
+    // ---- consume: STEPS x (B global 16B + N_TILES LDS reads + MFMA) --
 #pragma unroll
     for (int s2 = 0; s2 < STEPS; ++s2) {
+      const int kk = k0 + s2 * 32 + l_hi * 8;
       bf16x8 bfrag;
       if (W8) {
+        // 8 e4m3 bytes -> bf16 fragment (native converts); the per-row
+        // scale folds into the store, not the inner loop
+        const uint2 braw =
+            *reinterpret_cast<const uint2*>(wrow8 + kk);
         float f[8];
-        dg_fp8x4_to_f32(braw8[s2].x, f);
-        dg_fp8x4_to_f32(braw8[s2].y, f + 4);
+        dg_fp8x4_to_f32(braw.x, f);
+        dg_fp8x4_to_f32(braw.y, f + 4);
         const uint4 packed = dg_pack8(f);
         bfrag = *reinterpret_cast<const bf16x8*>(&packed);
       } else {
-        bfrag = bfrags[s2];
+        bfrag = *reinterpret_cast<const bf16x8*>(wrow + kk);
       }
       const int kb = (s2 * 32 + l_hi * 8) * 2 / 16;  // 16B col in chunk
 #pragma unroll
