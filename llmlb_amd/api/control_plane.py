@@ -132,12 +132,12 @@ def create_app(db_path: str | None = None,
                gateway_url: str | None = None,
                jwt_secret: str | None = None,
                admin_token: str | None = None) -> FastAPI:
-    db_path = db_path or os.environ.get("LLMLB_DATABASE_URL",
-                                        "llmlb.sqlite3")
+    from ..config import get_config
+    cfg = get_config(refresh=True)
+    db_path = db_path or cfg.database_url
     gateway_url = gateway_url or os.environ.get(
         "LLMLB_GATEWAY_URL", "http://127.0.0.1:32768")
-    jwt_secret = jwt_secret or os.environ.get("LLMLB_JWT_SECRET") \
-        or secrets.token_hex(32)
+    jwt_secret = jwt_secret or cfg.jwt_secret or secrets.token_hex(32)
     admin_token = admin_token or os.environ.get("LLMLB_ADMIN_TOKEN", "")
 
     app = FastAPI(title="llmlb control plane", docs_url=None,
@@ -154,8 +154,8 @@ def create_app(db_path: str | None = None,
             conn.close()
 
     def ensure_admin():
-        admin_user = os.environ.get("LLMLB_ADMIN_USER", "admin")
-        admin_pass = os.environ.get("LLMLB_ADMIN_PASSWORD", "")
+        admin_user = cfg.admin_user
+        admin_pass = cfg.admin_password
         with db() as conn:
             row = conn.execute("SELECT COUNT(*) c FROM users").fetchone()
             if row["c"] == 0:
@@ -521,7 +521,7 @@ def create_app(db_path: str | None = None,
     @app.get("/api/logs")
     def tail_logs(lines: int = Query(200, le=5000),
                   user=Depends(require_admin)):
-        log_dir = os.environ.get("LLMLB_LOG_DIR", "logs")
+        log_dir = cfg.log_dir
         path = os.path.join(log_dir, "llmlb-gw.jsonl")
         out = []
         try:
