@@ -194,8 +194,6 @@ def kv_append(kcache: torch.Tensor, vcache: torch.Tensor, k: torch.Tensor,
 
 _USE_CUSTOM_DECODE_GEMM = os.environ.get(
     "LLMLB_CUSTOM_DECODE_GEMM", "auto").lower()
-_DECODE_GEMM_VERSION = int(os.environ.get("LLMLB_DECODE_GEMM_VERSION",
-                                          "2"))
 
 
 # split-K fp32 workspaces, reused per shape: allocated zeroed once;

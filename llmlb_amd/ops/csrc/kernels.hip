@@ -261,17 +261,6 @@ __device__ __forceinline__ void fp8x4_to_f32(unsigned int d, float* o) {
   o[0] = lo.x; o[1] = lo.y; o[2] = hi.x; o[3] = hi.y;
 }
 
-__device__ __forceinline__ float e4m3_to_f32(int b) {
-  const int e = (b >> 3) & 0xF, m = b & 7;
-  float v;
-  if (e == 0)
-    v = ldexpf((float)m / 8.f, -6);
-  else if (e == 15 && m == 7)
-    v = __int_as_float(0x7fc00000);           // e4m3fn: only NaN, no inf
-  else
-    v = ldexpf(1.f + (float)m / 8.f, e - 7);
-  return (b & 0x80) ? -v : v;
-}
 
 // round-to-nearest-even f32 -> e4m3 byte (input pre-scaled to +-448)
 __device__ __forceinline__ unsigned char f32_to_e4m3(float f) {
