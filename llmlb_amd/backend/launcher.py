@@ -21,6 +21,8 @@ from .engine import Engine, EngineConfig
 PRESETS = {
     "llama-3-8b": LlamaConfig.llama3_8b,
     "llama-3-70b": LlamaConfig.llama3_70b,
+    "qwen2-7b": LlamaConfig.qwen2_7b,
+    "qwen2-0.5b": LlamaConfig.qwen2_0_5b,
     "llama-tiny": LlamaConfig.tiny,
     "llama-small": LlamaConfig.small,
 }

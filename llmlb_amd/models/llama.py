@@ -41,6 +41,10 @@ class LlamaConfig:
     rope_theta: float = 500000.0
     max_pos: int = 8192
     norm_eps: float = 1e-5
+    # Qwen-2-style variants: bias on the fused qkv projection and
+    # tied input/output embeddings
+    qkv_bias: bool = False
+    tie_embeddings: bool = False
     name: str = "llama-3-8b"
 
     @classmethod
@@ -58,6 +62,20 @@ class LlamaConfig:
         """2-layer Llama-3-shaped model for smoke/tests."""
         return cls(hidden=256, layers=2, q_heads=8, kv_heads=2, head_dim=32,
                    inter=512, vocab=1024, max_pos=512, name="llama-tiny")
+
+    @classmethod
+    def qwen2_7b(cls) -> "LlamaConfig":
+        return cls(hidden=3584, layers=28, q_heads=28, kv_heads=4,
+                   head_dim=128, inter=18944, vocab=152064,
+                   rope_theta=1e6, max_pos=32768, qkv_bias=True,
+                   tie_embeddings=False, name="qwen2-7b")
+
+    @classmethod
+    def qwen2_0_5b(cls) -> "LlamaConfig":
+        return cls(hidden=896, layers=24, q_heads=14, kv_heads=2,
+                   head_dim=64, inter=4864, vocab=151936,
+                   rope_theta=1e6, max_pos=4096, qkv_bias=True,
+                   tie_embeddings=True, name="qwen2-0.5b")
 
     @classmethod
     def small(cls) -> "LlamaConfig":
@@ -83,7 +101,8 @@ class LlamaDecoderLayer(nn.Module):
             torch.empty(cfg.hidden, dtype=dtype), requires_grad=False)
         self.post_norm_w = nn.Parameter(
             torch.empty(cfg.hidden, dtype=dtype), requires_grad=False)
-        self.qkv = ColumnParallelLinear(cfg.hidden, qkv_out, tp, dtype)
+        self.qkv = ColumnParallelLinear(cfg.hidden, qkv_out, tp, dtype,
+                                        bias=cfg.qkv_bias)
         self.o_proj = RowParallelLinear(cfg.q_heads * D, cfg.hidden, tp,
                                         dtype)
         self.gate_up = ColumnParallelLinear(cfg.hidden, 2 * cfg.inter, tp,
@@ -153,12 +172,17 @@ class LlamaForCausalLM(nn.Module):
                          .to(p.dtype))
 
         init_(self.embed)
-        init_(self.lm_head)
+        if self.cfg.tie_embeddings:
+            self.lm_head.data.copy_(self.embed.data)
+        else:
+            init_(self.lm_head)
         self.final_norm_w.data.fill_(1.0)
         for layer in self.layers_:
             layer.input_norm_w.data.fill_(1.0)
             layer.post_norm_w.data.fill_(1.0)
             init_(layer.qkv.weight)
+            if layer.qkv.bias is not None:
+                layer.qkv.bias.data.normal_(0.0, 0.02, generator=g)
             init_(layer.o_proj.weight)
             init_(layer.gate_up.weight)
             init_(layer.down.weight)
@@ -253,6 +277,14 @@ class LlamaForCausalLM(nn.Module):
             v = self._shard(take(pre + "self_attn.v_proj.weight"), 0,
                             ws, rank)
             assign(layer.qkv.weight, torch.cat([q, k, v], dim=0))
+            if layer.qkv.bias is not None:
+                qb = self._shard(take(pre + "self_attn.q_proj.bias"), 0,
+                                 ws, rank)
+                kb = self._shard(take(pre + "self_attn.k_proj.bias"), 0,
+                                 ws, rank)
+                vb = self._shard(take(pre + "self_attn.v_proj.bias"), 0,
+                                 ws, rank)
+                assign(layer.qkv.bias, torch.cat([qb, kb, vb], dim=0))
             assign(layer.o_proj.weight,
                    self._shard(take(pre + "self_attn.o_proj.weight"), 1,
                                ws, rank))
@@ -283,6 +315,11 @@ class LlamaForCausalLM(nn.Module):
             out[pre + "self_attn.q_proj.weight"] = qkv[:qn]
             out[pre + "self_attn.k_proj.weight"] = qkv[qn:qn + kn]
             out[pre + "self_attn.v_proj.weight"] = qkv[qn + kn:]
+            if layer.qkv.bias is not None:
+                qkvb = layer.qkv.bias.data
+                out[pre + "self_attn.q_proj.bias"] = qkvb[:qn]
+                out[pre + "self_attn.k_proj.bias"] = qkvb[qn:qn + kn]
+                out[pre + "self_attn.v_proj.bias"] = qkvb[qn + kn:]
             out[pre + "self_attn.o_proj.weight"] = layer.o_proj.weight.data
             gu = layer.gate_up.weight.data
             out[pre + "mlp.gate_proj.weight"] = gu[:self.cfg.inter]

@@ -92,7 +92,7 @@ class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
     Output is the local shard [*, out_features / tp]."""
 
     def __init__(self, in_features: int, out_features: int, tp: TPContext,
-                 dtype=torch.bfloat16):
+                 dtype=torch.bfloat16, bias: bool = False):
         super().__init__()
         assert out_features % tp.world_size == 0, \
             f"{out_features} not divisible by tp={tp.world_size}"
@@ -101,11 +101,15 @@ class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
         self.weight = nn.Parameter(
             torch.empty(self.out_local, in_features, dtype=dtype),
             requires_grad=False)
+        self.bias = nn.Parameter(
+            torch.zeros(self.out_local, dtype=dtype),
+            requires_grad=False) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
-            return linear_decode(x, self.weight, w8=self._w8_pair())
-        return torch.nn.functional.linear(x, self.weight)
+            y = linear_decode(x, self.weight, w8=self._w8_pair())
+            return y + self.bias if self.bias is not None else y
+        return torch.nn.functional.linear(x, self.weight, self.bias)
 
 
 class RowParallelLinear(nn.Module, _Fp8WeightMixin):

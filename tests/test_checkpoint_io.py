@@ -69,3 +69,27 @@ def test_hf_tokenizer_wrapper(tmp_path):
     srv = BackendServer(eng, "m", tokenizer_path=str(p))
     assert srv.tokenizer.encode("hello world") == [1, 2]
     assert "hello" in srv.tokenizer.decode_token(1)
+
+
+def test_qwen2_style_round_trip():
+    """qkv-bias + tied-embedding configs (Qwen-2 family) generate,
+    checkpoint and reload exactly."""
+    cfg = LlamaConfig(hidden=128, layers=2, q_heads=4, kv_heads=2,
+                      head_dim=32, inter=256, vocab=512, max_pos=256,
+                      qkv_bias=True, tie_embeddings=True,
+                      name="qwen-test")
+    a = Engine(EngineConfig(model=cfg, device="cpu", dtype="float32",
+                            kv_cache_blocks=32, seed=4))
+    assert torch.equal(a.model.embed, a.model.lm_head)  # tied
+    assert a.model.layers_[0].qkv.bias is not None
+    import tempfile
+    from pathlib import Path
+    with tempfile.TemporaryDirectory() as d:
+        p = Path(d) / "m.safetensors"
+        a.model.save_safetensors(p)
+        b = Engine(EngineConfig(model=cfg, device="cpu",
+                                dtype="float32", kv_cache_blocks=32,
+                                seed=99, weights_path=str(p)))
+    prm = SamplingParams(temperature=0.0)
+    assert a.generate_sync([5, 6, 7], 6, prm).token_ids == \
+        b.generate_sync([5, 6, 7], 6, prm).token_ids
