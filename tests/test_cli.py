@@ -91,3 +91,49 @@ def test_assistant_blocks_injection():
         capture_output=True, text=True, timeout=15)
     assert res.returncode == 2
     assert "blocked" in res.stderr
+
+
+def test_serve_with_backend_chat_e2e(tmp_path):
+    """`llmlb serve --backends llama-tiny:x1` brings up the full node
+    (gateway + control plane + backend replica) and serves a chat
+    round trip through the gateway."""
+    port, cp, be = _free_port(), _free_port(), _free_port()
+    env = os.environ.copy()
+    env.update({"LLMLB_RUN_DIR": str(tmp_path / "run"),
+                "LLMLB_DATABASE_URL": str(tmp_path / "db.sqlite3"),
+                "LLMLB_ADMIN_TOKEN": "clitok",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "2"})
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "llmlb_amd.cli", "serve", "--port",
+         str(port), "--db", str(tmp_path / "db.sqlite3"),
+         "--control-plane-port", str(cp),
+         "--backends", "llama-tiny:x1",
+         "--backend-base-port", str(be)],
+        cwd=str(REPO), env=env, stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT)
+    try:
+        body = json.dumps({
+            "model": "llama-tiny", "max_tokens": 4,
+            "messages": [{"role": "user", "content": "hi"}]}).encode()
+        deadline = time.time() + 90
+        ok = False
+        while time.time() < deadline:
+            try:
+                req = urllib.request.Request(
+                    f"http://127.0.0.1:{port}/v1/chat/completions",
+                    data=body,
+                    headers={"Content-Type": "application/json"})
+                with urllib.request.urlopen(req, timeout=10) as r:
+                    out = json.loads(r.read())
+                    if out["choices"][0]["message"]["content"]:
+                        ok = True
+                        break
+            except Exception:
+                time.sleep(1.0)
+        assert ok, "chat through the CLI-supervised node never succeeded"
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
