@@ -228,6 +228,7 @@ class Engine:
         self.waiting: "queue.Queue[_Sequence]" = queue.Queue()
         self._next_rid = 0
         self._rid_lock = threading.Lock()
+        self._abort_rids: set[int] = set()
         self._stop = threading.Event()
         self._thread: threading.Thread | None = None
         self.scale = m.head_dim ** -0.5
@@ -390,8 +391,35 @@ class Engine:
             admitted.append(seq)
         return admitted
 
+    def abort(self, rid: int) -> None:
+        """Cancel a request (client disconnect): the sequence is
+        retired at the next step boundary and its pages are freed."""
+        with self._rid_lock:
+            self._abort_rids.add(rid)
+
+    def _apply_aborts(self) -> None:
+        with self._rid_lock:
+            if not self._abort_rids:
+                return
+            rids = self._abort_rids
+            self._abort_rids = set()
+        for seq in [s for s in self.running if s.rid in rids]:
+            self._retire(seq)
+        # drop queued ones too (rebuild the waiting queue)
+        pending = []
+        while True:
+            try:
+                s = self.waiting.get_nowait()
+            except queue.Empty:
+                break
+            if s.rid not in rids:
+                pending.append(s)
+        for s in pending:
+            self.waiting.put(s)
+
     def step(self) -> bool:
         """One scheduler iteration.  Returns True if any work was done."""
+        self._apply_aborts()
         prefills = self._admit()
         if prefills:
             self._prefill_batch(prefills)

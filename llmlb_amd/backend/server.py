@@ -439,6 +439,7 @@ def build_asgi_app(server: BackendServer):
                 return
             n_choices = max(1, min(int(body.get("n", 1) or 1), 8))
             sinks = [_AsyncSink() for _ in range(n_choices)]
+            rids = []
             try:
                 for ci, sink in enumerate(sinks):
                     # distinct seeds per choice so sampled choices
@@ -452,6 +453,7 @@ def build_asgi_app(server: BackendServer):
                     else:
                         rid, out_q, prompt_ids = server.submit_completion(
                             cbody, sink)
+                    rids.append(rid)
             except MemoryError:
                 await send_json(send, 429, {"error": {
                     "message": "KV cache exhausted",
@@ -540,8 +542,22 @@ def build_asgi_app(server: BackendServer):
                 await send({"type": "http.response.body",
                             "body": chunk_payload({"role": "assistant"}),
                             "more_body": True})
+            # watch for client disconnect so an abandoned stream stops
+            # burning decode steps (engine.abort frees its pages)
+            disconnected = asyncio.Event()
+
+            async def _watch_disconnect():
+                while True:
+                    msg = await receive()
+                    if msg["type"] == "http.disconnect":
+                        disconnected.set()
+                        return
+
+            watcher = asyncio.create_task(_watch_disconnect())
             ntok = 0
             for ci, sink_i in enumerate(sinks):
+                if disconnected.is_set():
+                    break
                 finish = "stop"
                 async for tok, fin, extra in drain_queue_async(sink_i):
                     ntok += 1
@@ -558,6 +574,10 @@ def build_asgi_app(server: BackendServer):
                     await send({"type": "http.response.body",
                                 "body": payload,
                                 "more_body": True})
+                    if disconnected.is_set():
+                        break
+                if disconnected.is_set():
+                    break
                 if ci + 1 < len(sinks):
                     await send({"type": "http.response.body",
                                 "body": chunk_payload({}, finish=finish,
@@ -572,6 +592,11 @@ def build_asgi_app(server: BackendServer):
                                               usage=usage,
                                               index=len(sinks) - 1),
                         "more_body": True})
+            watcher.cancel()
+            if disconnected.is_set():
+                for r in rids:
+                    server.engine.abort(r)
+                return
             await send({"type": "http.response.body",
                         "body": b"data: [DONE]\n\n", "more_body": False})
             return

@@ -172,3 +172,25 @@ def test_capacity_never_kills_engine():
             toks.append(q.get_nowait())
         assert len(toks) == 40, (i, len(toks))
     assert eng.kv.free_blocks() > 0
+
+
+def test_abort_frees_pages(tiny_engine):
+    """abort(rid) retires a running sequence at the next step and
+    releases everything (pages, row, queue slot)."""
+    eng = tiny_engine
+    free0 = eng.kv.free_blocks()
+    p = SamplingParams(temperature=0.0, max_new_tokens=200)
+    rid, q = eng.submit([1, 2, 3, 4], p)
+    for _ in range(5):
+        eng.step()
+    assert eng.num_running == 1
+    eng.abort(rid)
+    eng.step()
+    assert eng.num_running == 0
+    parked = len(eng.prefix.lru) if eng.prefix else 0
+    assert eng.kv.free_blocks() + parked == free0
+    # aborting a queued (not yet admitted) request also works
+    rid2, _ = eng.submit([5] * 8, p)
+    eng.abort(rid2)
+    eng.step()
+    assert eng.num_running == 0 and eng.waiting.empty()

@@ -60,7 +60,7 @@ def stack():
             break
         except Exception:
             time.sleep(0.3)
-    yield {"base": base}
+    yield {"base": base, "backend": f"http://127.0.0.1:{backend_port}"}
     gw.terminate()
     backend.terminate()
 
@@ -168,3 +168,29 @@ def test_embeddings_from_engine(stack):
             timeout=30) as r:
         out2 = json.loads(r.read())
     assert out2["data"][0]["embedding"] == v0
+
+
+def test_disconnect_aborts_generation(stack):
+    """Closing a streaming connection mid-generation aborts the
+    request on the engine (pages freed, active count drops)."""
+    import socket as _s
+    host, port = stack["backend"].replace("http://", "").split(":")
+    body = json.dumps({"model": "llama-tiny", "max_tokens": 500,
+                       "stream": True,
+                       "messages": [{"role": "user",
+                                     "content": "hi"}]}).encode()
+    c = _s.create_connection((host, int(port)), timeout=10)
+    c.sendall((f"POST /v1/chat/completions HTTP/1.1\r\n"
+               f"Host: {host}\r\nContent-Type: application/json\r\n"
+               f"Content-Length: {len(body)}\r\n\r\n").encode() + body)
+    c.recv(4096)          # first bytes of the stream
+    c.close()             # abandon mid-generation
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        with urllib.request.urlopen(stack["backend"] + "/api/health",
+                                    timeout=5) as r:
+            h = json.loads(r.read())
+        if h["active_requests"] == 0 and h["queued_requests"] == 0:
+            return
+        time.sleep(0.3)
+    raise AssertionError(f"generation not aborted: {h}")
