@@ -480,9 +480,21 @@ def build_asgi_app(server: BackendServer):
                         for t, v in extra["top"]]}
 
             if not stream:
+                disconnected = asyncio.Event()
+
+                async def _watch_disconnect_ns():
+                    while True:
+                        msg = await receive()
+                        if msg["type"] == "http.disconnect":
+                            disconnected.set()
+                            return
+
+                watcher = asyncio.create_task(_watch_disconnect_ns())
                 choices = []
                 total_out = 0
                 for ci, sink_i in enumerate(sinks):
+                    if disconnected.is_set():
+                        break
                     toks, finish, lp_content = [], "stop", []
                     async for tok, fin, extra in drain_queue_async(
                             sink_i):
@@ -491,6 +503,10 @@ def build_asgi_app(server: BackendServer):
                             lp_content.append(lp_entry(tok, extra))
                         if fin:
                             finish = fin
+                        if disconnected.is_set():
+                            break
+                    if disconnected.is_set():
+                        break
                     text = "".join(server.tokenizer.decode_token(t)
                                    for t in toks)
                     total_out += len(toks)
@@ -505,6 +521,11 @@ def build_asgi_app(server: BackendServer):
                     else:
                         choices.append({"index": ci, "text": text,
                                         "finish_reason": finish})
+                watcher.cancel()
+                if disconnected.is_set():
+                    for r in rids:
+                        server.engine.abort(r)
+                    return
                 usage = {"prompt_tokens": len(prompt_ids),
                          "completion_tokens": total_out,
                          "total_tokens": len(prompt_ids) + total_out}
