@@ -38,7 +38,9 @@ class TPEngineCoordinator:
         self.group_cpu = group_cpu
         self.rank = dist.get_rank()
         self._pending: list[tuple] = []
+        self._pending_aborts: list[int] = []
         self._sinks: dict[int, object] = {}
+        self._rid_map: dict[int, int] = {}   # coord rid -> engine rid
         self._lock = threading.Lock()
         self._next_rid = 0
         self._stop = False
@@ -46,16 +48,23 @@ class TPEngineCoordinator:
     # HTTP-facing submit (rank 0 only)
     def submit(self, prompt_ids, params: SamplingParams | None = None,
                sink=None):
+        from dataclasses import asdict
         params = params or SamplingParams()
         with self._lock:
             rid = self._next_rid
             self._next_rid += 1
-            self._pending.append((rid, list(prompt_ids),
-                                  params.temperature, params.top_p,
-                                  params.max_new_tokens, params.seed))
+            # broadcast the FULL params dict so every sampler feature
+            # (top_k/min_p/penalties/logprobs/stops) works under TP
+            self._pending.append((rid, list(prompt_ids), asdict(params)))
             if sink is not None:
                 self._sinks[rid] = sink
         return rid, sink
+
+    def abort(self, rid: int) -> None:
+        """Cancellation must reach every rank in the same step, or the
+        TP group's collectives desynchronise."""
+        with self._lock:
+            self._pending_aborts.append(rid)
 
     def _drain_pending(self) -> list[tuple]:
         with self._lock:
@@ -67,19 +76,28 @@ class TPEngineCoordinator:
         """Lockstep engine loop on every rank."""
         while not self._stop:
             if self.rank == 0:
+                with self._lock:
+                    aborts = self._pending_aborts
+                    self._pending_aborts = []
                 batch = self._drain_pending()
-                obj = [batch]
+                obj = [(batch, aborts)]
             else:
                 obj = [None]
             dist.broadcast_object_list(obj, src=0, group=self.group_cpu)
-            batch = obj[0]
-            for rid, prompt_ids, temp, top_p, max_new, seed in batch:
-                params = SamplingParams(temperature=temp, top_p=top_p,
-                                        max_new_tokens=max_new,
-                                        seed=seed if seed is not None
-                                        else 12345 + rid)
+            batch, aborts = obj[0]
+            for rid, prompt_ids, pdict in batch:
+                params = SamplingParams(**pdict)
+                if params.seed is None:
+                    params.seed = 12345 + rid
+                params.stop_token_ids = tuple(params.stop_token_ids)
                 sink = self._sinks.get(rid) if self.rank == 0 else None
-                _, q = self.engine.submit(prompt_ids, params, sink=sink)
+                erid, q = self.engine.submit(prompt_ids, params,
+                                             sink=sink)
+                self._rid_map[rid] = erid
+            for rid in aborts:
+                erid = self._rid_map.get(rid)
+                if erid is not None:
+                    self.engine.abort(erid)
             busy = self.engine.step()
             if not busy and not batch:
                 # idle: cheap sleep, staying in lockstep (all ranks see
@@ -146,6 +164,9 @@ class _SubmitProxy:
 
     def submit(self, prompt_ids, params=None, sink=None):
         return self._coord.submit(prompt_ids, params, sink)
+
+    def abort(self, rid):
+        self._coord.abort(rid)
 
     def __getattr__(self, name):
         return getattr(self._engine, name)
