@@ -84,7 +84,7 @@ class GenerationResult:
 class _Sequence:
     __slots__ = ("rid", "prompt_ids", "out_ids", "params", "block_table",
                  "length", "out_queue", "generator", "t_submit", "t_first",
-                 "row", "cached_len", "hashes", "bigram_pos")
+                 "row", "cached_len", "hashes", "bigram_pos", "max_new")
 
     def __init__(self, rid, prompt_ids, params, out_queue, generator):
         self.rid = rid
@@ -101,6 +101,7 @@ class _Sequence:
         self.cached_len = 0                  # tokens reused from PrefixCache
         self.hashes: list[bytes] | None = None
         self.bigram_pos: dict | None = None  # spec: bigram -> last pos
+        self.max_new = params.max_new_tokens # clamped to max_pos at submit
 
 
 class _DecodeGraph:
@@ -305,6 +306,11 @@ class Engine:
             gen.manual_seed(params.seed)
         out_q = sink if sink is not None else queue.Queue()
         seq = _Sequence(rid, list(prompt_ids), params, out_q, gen)
+        # a generation can never pass the model's context window: clamp
+        # so page reservation stays within max_blocks_per_seq
+        seq.max_new = max(1, min(params.max_new_tokens,
+                                 self.cfg.model.max_pos
+                                 - len(seq.prompt_ids) - 1))
         self.waiting.put(seq)
         return rid, out_q
 
@@ -364,8 +370,9 @@ class Engine:
             # mid-decode growth can never exhaust the pool and kill the
             # engine thread; with 288 GB of HBM the utilisation cost is
             # noise, and parked prefix pages remain evictable
-            need = (len(seq.prompt_ids) + seq.params.max_new_tokens
-                    + BLOCK_SIZE) // BLOCK_SIZE + 1
+            need = min((len(seq.prompt_ids) + seq.max_new
+                        + BLOCK_SIZE) // BLOCK_SIZE + 1,
+                       self.max_blocks_per_seq)
             if self.prefix is not None:
                 if seq.hashes is None:
                     seq.hashes = PrefixCache.chain_hashes(seq.prompt_ids)
@@ -877,7 +884,7 @@ class Engine:
         fin = None
         if tok in seq.params.stop_token_ids:
             fin = "stop"
-        elif len(seq.out_ids) >= seq.params.max_new_tokens:
+        elif len(seq.out_ids) >= seq.max_new:
             fin = "length"
         seq.out_queue.put((tok, fin, extra))
         return fin is None

@@ -66,6 +66,12 @@ class TPEngineCoordinator:
         with self._lock:
             self._pending_aborts.append(rid)
 
+    def stop(self) -> None:
+        """Rank 0: broadcast a shutdown sentinel so EVERY rank leaves
+        the lockstep loop together."""
+        with self._lock:
+            self._pending_aborts.append(-1)
+
     def _drain_pending(self) -> list[tuple]:
         with self._lock:
             out = self._pending
@@ -95,6 +101,9 @@ class TPEngineCoordinator:
                                              sink=sink)
                 self._rid_map[rid] = erid
             for rid in aborts:
+                if rid == -1:          # group shutdown sentinel
+                    self._stop = True
+                    continue
                 erid = self._rid_map.get(rid)
                 if erid is not None:
                     self.engine.abort(erid)

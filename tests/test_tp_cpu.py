@@ -12,7 +12,11 @@ import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
+from pathlib import Path
+
 from llmlb_amd.models.llama import LlamaConfig
+
+REPO = Path(__file__).resolve().parents[1]
 
 
 def _master_weights(cfg, seed=7):
@@ -143,3 +147,18 @@ def test_tp2_matches_tp1(tmp_path):
     mp.spawn(_worker, args=(2, port, result), nprocs=2, join=True)
     out = torch.load(result)
     assert out["err"] < 1e-4, f"TP=2 logits diverge: {out['err']}"
+
+
+def test_tp_coordinator_lockstep_2rank():
+    """The lockstep coordinator (broadcast submissions + aborts +
+    shutdown) over gloo world_size=2 matches a solo engine."""
+    import subprocess
+    import sys
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29761",
+         "tests/helpers/tp_coord_check.py"],
+        cwd=str(REPO), capture_output=True, text=True, timeout=240)
+    assert res.returncode == 0, res.stdout[-1500:] + res.stderr[-1500:]
+    assert "tp-coord-check OK" in res.stdout
