@@ -437,6 +437,18 @@ def build_asgi_app(server: BackendServer):
                     "type": "invalid_request_error",
                     "code": "model_not_found"}})
                 return
+            # context-window guard (OpenAI: context_length_exceeded)
+            probe_ids = server.tokenizer.encode(
+                render_chat(body.get("messages", []))
+                if is_chat else str(body.get("prompt", "")))
+            if len(probe_ids) >= server.engine.cfg.model.max_pos:
+                await send_json(send, 400, {"error": {
+                    "message": f"prompt has {len(probe_ids)} tokens; "
+                               f"model context is "
+                               f"{server.engine.cfg.model.max_pos}",
+                    "type": "invalid_request_error",
+                    "code": "context_length_exceeded"}})
+                return
             n_choices = max(1, min(int(body.get("n", 1) or 1), 8))
             sinks = [_AsyncSink() for _ in range(n_choices)]
             rids = []

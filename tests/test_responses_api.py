@@ -194,3 +194,20 @@ def test_disconnect_aborts_generation(stack):
             return
         time.sleep(0.3)
     raise AssertionError(f"generation not aborted: {h}")
+
+
+def test_context_length_exceeded(stack):
+    body = json.dumps({"model": "llama-tiny", "max_tokens": 4,
+                       "messages": [{"role": "user",
+                                     "content": " ".join(
+                                         f"w{i}" for i in range(600))}]
+                       }).encode()
+    req = urllib.request.Request(
+        stack["backend"] + "/v1/chat/completions", data=body,
+        headers={"Content-Type": "application/json"})
+    try:
+        urllib.request.urlopen(req, timeout=15)
+        raise AssertionError("expected 400")
+    except urllib.error.HTTPError as e:
+        assert e.code == 400
+        assert b"context_length_exceeded" in e.read()
