@@ -24,6 +24,7 @@ struct MockConfig {
   int workers = 2;
   std::vector<std::string> models = {"mock-model"};
   int delay_ms = 0;          // simulated inference latency
+  bool delay_inference_only = false;
   int stream_chunks = 8;     // SSE deltas per streaming response
   double fail_rate = 0.0;    // inject failures (config 5 testing)
   int completion_tokens = 8;
@@ -295,7 +296,13 @@ void MockConn::respond() {
     }
     flush();
   };
-  if (g_cfg.delay_ms > 0) {
+  // --delay-inference-only: control/health/models respond instantly
+  // so registration+sync work while inference stalls (timeout tests)
+  const bool inference = starts_with(path, "/v1/chat") ||
+                         starts_with(path, "/v1/completions") ||
+                         starts_with(path, "/v1/embeddings") ||
+                         starts_with(path, "/v1/responses");
+  if (g_cfg.delay_ms > 0 && (!g_cfg.delay_inference_only || inference)) {
     w->loop.add_timer(g_cfg.delay_ms, do_respond);
   } else {
     do_respond();
@@ -312,6 +319,8 @@ int main(int argc, char** argv) {
     else if (a == "--workers") g_cfg.workers = atoi(next());
     else if (a == "--models") g_cfg.models = split(next(), ',');
     else if (a == "--delay-ms") g_cfg.delay_ms = atoi(next());
+    else if (a == "--delay-inference-only")
+      g_cfg.delay_inference_only = true;
     else if (a == "--stream-chunks") g_cfg.stream_chunks = atoi(next());
     else if (a == "--fail-rate") g_cfg.fail_rate = atof(next());
   }

@@ -419,3 +419,65 @@ def test_request_id_tracing(stack):
                                 headers=stack["auth"])
     with urllib.request.urlopen(r3, timeout=10) as resp:
         assert resp.headers["X-Request-Id"]
+
+
+def test_inference_timeout_sweep():
+    """An endpoint whose backend stalls past inference_timeout_secs
+    gets its request killed by the deadline sweep (504) instead of
+    hanging the client."""
+    mock_port, gw_port = _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    slow = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "slow-model", "--delay-ms", "30000",
+         "--delay-inference-only"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    admin = {"Authorization": "Bearer t"}
+    try:
+        for _ in range(50):
+            try:
+                _req("GET", base + "/health", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        status, body = _req("POST", base + "/api/endpoints",
+                            {"name": "slow",
+                             "base_url": f"http://127.0.0.1:{mock_port}"},
+                            headers=admin)
+        assert status == 201, body
+        ep_id = json.loads(body)["id"]
+        status, _ = _req("PUT", base + f"/api/endpoints/{ep_id}",
+                         {"inference_timeout_secs": 1}, headers=admin)
+        assert status == 200
+        # wait for model sync (the slow mock delays /v1/models too)
+        deadline = time.time() + 30
+        while time.time() < deadline:
+            _, models = _req("GET", base + "/v1/models")
+            if "slow-model" in models:
+                break
+            time.sleep(0.5)
+        t0 = time.time()
+        status, body = _req("POST", base + "/v1/chat/completions",
+                            {"model": "slow-model",
+                             "messages": [{"role": "user",
+                                           "content": "hi"}]},
+                            timeout=20)
+        dt = time.time() - t0
+        assert status == 504, (status, body)
+        assert dt < 8, dt
+    finally:
+        gw.terminate()
+        slow.terminate()
+        for p in (gw, slow):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
