@@ -481,3 +481,60 @@ def test_inference_timeout_sweep():
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def test_admission_429_when_saturated():
+    """LLMLB_QUEUE_MAX=2: with both slots held by stalled requests, a
+    third gets a fast 429 (capacity) instead of queueing forever."""
+    mock_port, gw_port = _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    slow = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "busy-model", "--delay-ms", "8000",
+         "--delay-inference-only"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t", "LLMLB_QUEUE_MAX": "2",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        for _ in range(50):
+            try:
+                _req("GET", base + "/health", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        status, _ = _req("POST", base + "/api/endpoints",
+                         {"name": "busy",
+                          "base_url": f"http://127.0.0.1:{mock_port}"},
+                         headers={"Authorization": "Bearer t"})
+        assert status == 201
+        import threading
+        body = {"model": "busy-model",
+                "messages": [{"role": "user", "content": "hi"}]}
+        holders = [threading.Thread(
+            target=lambda: _req("POST", base + "/v1/chat/completions",
+                                body, timeout=30)) for _ in range(2)]
+        for h in holders:
+            h.start()
+        time.sleep(1.0)   # both slots occupied (0.8 * 2 = admission cap)
+        t0 = time.time()
+        status, resp = _req("POST", base + "/v1/chat/completions", body,
+                            timeout=10)
+        dt = time.time() - t0
+        assert status == 429, (status, resp)
+        assert dt < 3, dt
+        for h in holders:
+            h.join(timeout=30)
+    finally:
+        gw.terminate()
+        slow.terminate()
+        for p in (gw, slow):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
