@@ -208,3 +208,18 @@ def test_update_manager_drain_apply_rollback(app_client, tmp_path,
     assert r.status_code == 200
     assert old.read_bytes() == b"OLD BINARY"
     httpd.shutdown()
+
+
+def test_ws_dashboard_auth_and_stream(app_client):
+    """/ws/dashboard rejects bad tokens and streams typed events for a
+    valid admin token (gateway absent -> error-typed events still show
+    the loop is alive)."""
+    import pytest as _pytest
+    from starlette.websockets import WebSocketDisconnect
+    with _pytest.raises(WebSocketDisconnect):
+        with app_client.websocket_connect("/ws/dashboard?token=bad"):
+            pass
+    with app_client.websocket_connect(
+            "/ws/dashboard?token=admtok") as ws:
+        msg = ws.receive_json()
+        assert msg["type"] in ("summary", "event", "error")
