@@ -1,6 +1,7 @@
 #include "common.h"
 
 #include <fcntl.h>
+#include <dirent.h>
 #include <sys/stat.h>
 #include <sys/types.h>
 #include <unistd.h>
@@ -14,6 +15,9 @@ namespace gw {
 LogLevel g_log_level = LogLevel::kInfo;
 static FILE* g_log_file = nullptr;
 static std::mutex g_log_mu;
+static std::string g_log_dir;
+static int g_log_retention_days = 7;
+static int g_log_day = -1;   // yyyymmdd of the open file
 
 std::string iso8601_now() {
   time_t t = time(nullptr);
@@ -24,15 +28,60 @@ std::string iso8601_now() {
   return buf;
 }
 
-void log_init(const std::string& level, const std::string& jsonl_dir) {
+static int today_yyyymmdd() {
+  time_t t = time(nullptr);
+  struct tm tmv;
+  gmtime_r(&t, &tmv);
+  return (tmv.tm_year + 1900) * 10000 + (tmv.tm_mon + 1) * 100 +
+         tmv.tm_mday;
+}
+
+// delete llmlb-gw-*.jsonl older than the retention window
+// (reference logging.rs retention; mtime-based, best effort)
+static void purge_old_logs() {
+  if (g_log_dir.empty()) return;
+  DIR* d = opendir(g_log_dir.c_str());
+  if (!d) return;
+  const time_t cutoff =
+      time(nullptr) - (time_t)g_log_retention_days * 86400;
+  struct dirent* e;
+  while ((e = readdir(d)) != nullptr) {
+    std::string name = e->d_name;
+    if (name.rfind("llmlb-gw-", 0) != 0 ||
+        name.find(".jsonl") == std::string::npos)
+      continue;
+    std::string path = g_log_dir + "/" + name;
+    struct stat st;
+    if (stat(path.c_str(), &st) == 0 && st.st_mtime < cutoff)
+      unlink(path.c_str());
+  }
+  closedir(d);
+}
+
+// open (or roll to) today's dated file; caller holds g_log_mu
+static void open_log_for_today() {
+  const int day = today_yyyymmdd();
+  if (day == g_log_day && g_log_file) return;
+  if (g_log_file) fclose(g_log_file);
+  char name[64];
+  snprintf(name, sizeof name, "/llmlb-gw-%08d.jsonl", day);
+  g_log_file = fopen((g_log_dir + name).c_str(), "a");
+  g_log_day = day;
+  purge_old_logs();
+}
+
+void log_init(const std::string& level, const std::string& jsonl_dir,
+              int retention_days) {
   if (level == "debug") g_log_level = LogLevel::kDebug;
   else if (level == "info") g_log_level = LogLevel::kInfo;
   else if (level == "warn" || level == "warning") g_log_level = LogLevel::kWarn;
   else if (level == "error") g_log_level = LogLevel::kError;
   if (!jsonl_dir.empty()) {
     ::mkdir(jsonl_dir.c_str(), 0755);
-    std::string path = jsonl_dir + "/llmlb-gw.jsonl";
-    g_log_file = fopen(path.c_str(), "a");
+    g_log_dir = jsonl_dir;
+    g_log_retention_days = retention_days > 0 ? retention_days : 7;
+    std::lock_guard<std::mutex> lk(g_log_mu);
+    open_log_for_today();
   }
 }
 
@@ -47,6 +96,7 @@ void log_write(LogLevel lv, const char* fmt, ...) {
   std::string ts = iso8601_now();
   std::lock_guard<std::mutex> lk(g_log_mu);
   fprintf(stderr, "%s %-5s %s\n", ts.c_str(), names[(int)lv], msg);
+  if (!g_log_dir.empty()) open_log_for_today();   // daily rotation
   if (g_log_file) {
     // minimal JSONL sink (stdout + file, reference logging.rs:18)
     fprintf(g_log_file, "{\"ts\":\"%s\",\"level\":\"%s\",\"msg\":\"", ts.c_str(),

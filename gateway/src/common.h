@@ -40,7 +40,8 @@ std::string iso8601_now();
 // ---------------------------------------------------------------- logging
 enum class LogLevel { kDebug = 0, kInfo, kWarn, kError };
 extern LogLevel g_log_level;
-void log_init(const std::string& level, const std::string& jsonl_dir);
+void log_init(const std::string& level, const std::string& jsonl_dir,
+              int retention_days = 7);
 void log_write(LogLevel lv, const char* fmt, ...)
     __attribute__((format(printf, 2, 3)));
 

@@ -564,7 +564,7 @@ Gateway::Gateway(GatewayConfig cfg_)
 Gateway::~Gateway() { stop(); }
 
 bool Gateway::init(std::string* err) {
-  log_init(cfg.log_level, cfg.log_dir);
+  log_init(cfg.log_level, cfg.log_dir, cfg.log_retention_days);
   // raise RLIMIT_NOFILE (bootstrap.rs:374-422)
   rlimit rl{};
   if (getrlimit(RLIMIT_NOFILE, &rl) == 0 && rl.rlim_cur < 65536) {

@@ -45,6 +45,7 @@ struct GatewayConfig {
   std::string admin_token;          // LLMLB_ADMIN_TOKEN
   std::string log_level = "info";
   std::string log_dir;
+  int log_retention_days = 7;
   // cloud providers (base_url overridable for tests; http only — this
   // build environment has no TLS stack and no egress)
   std::string openai_base = "";     // e.g. http://127.0.0.1:9001/v1

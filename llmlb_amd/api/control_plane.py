@@ -522,7 +522,13 @@ def create_app(db_path: str | None = None,
     def tail_logs(lines: int = Query(200, le=5000),
                   user=Depends(require_admin)):
         log_dir = cfg.log_dir
-        path = os.path.join(log_dir, "llmlb-gw.jsonl")
+        # newest dated file (daily rotation: llmlb-gw-YYYYMMDD.jsonl),
+        # falling back to the legacy fixed name
+        import glob as _glob
+        dated = sorted(_glob.glob(
+            os.path.join(log_dir, "llmlb-gw-*.jsonl")))
+        path = dated[-1] if dated else os.path.join(log_dir,
+                                                    "llmlb-gw.jsonl")
         out = []
         try:
             with open(path, "rb") as f:
