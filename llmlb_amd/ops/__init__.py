@@ -370,7 +370,8 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
     out = torch.empty(S, n_q, D, dtype=q.dtype, device=q.device)
     group = n_q // n_kv
     # v2 (GQA-amortized) always reduces through the slice workspace
-    use_v2 = _PAGED_ATTN_V2 and D == 128 and group <= 4
+    use_v2 = (_PAGED_ATTN_V2 and D == 128
+              and group in (1, 2, 4, 8))
     slices = n_parts * 4 if use_v2 else n_parts
     if n_parts > 1:
         if workspace is not None:
@@ -389,7 +390,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         else seq_lens.to(torch.int32)
     if kcache.dtype == torch.uint8:
         assert use_v2, "FP8 KV requires the v2 attention shape " \
-            "(D=128, GQA group <= 4)"
+            "(D=128, GQA group in 1/2/4/8)"
         rc = lib.ll_paged_attn_decode_fp8(
             _ptr(out), acc_p, ml_p, _ptr(q), _ptr(kcache), _ptr(vcache),
             _ptr(k_scale), _ptr(v_scale), _ptr(bt), _ptr(sl),

@@ -520,7 +520,7 @@ void paged_attn_decode_v2_kernel(
     const int* __restrict__ block_tables,
     const int* __restrict__ seq_lens,
     int n_q, int n_kv, int max_blocks, int n_parts, float scale,
-    int q_rstride) {
+    int q_rstride, int group_stride, int g_base) {
   constexpr int D = 128;
   constexpr int DS = 16;                 // dim-slice (8 lanes per token)
   const int seq = blockIdx.x;
@@ -545,8 +545,8 @@ void paged_attn_decode_v2_kernel(
 #pragma unroll
   for (int g = 0; g < GROUP; ++g) {
     const uint4* qv = reinterpret_cast<const uint4*>(
-        q + (size_t)seq * q_rstride + (size_t)(kvh * GROUP + g) * D
-        + p * DS);
+        q + (size_t)seq * q_rstride
+        + (size_t)(kvh * group_stride + g_base + g) * D + p * DS);
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
       float f[8];
@@ -701,7 +701,7 @@ void paged_attn_decode_v2_kernel(
             o[i] += s_acc[w2][g][p * DS + i] * sc;
         }
         const float inv = l_tot > 0.f ? 1.f / l_tot : 0.f;
-        const int qh = kvh * GROUP + g;
+        const int qh = kvh * group_stride + g_base + g;
         bf16* ob = out + ((size_t)seq * n_q + qh) * D + p * DS;
 #pragma unroll
         for (int i = 0; i < DS; ++i) {
@@ -718,7 +718,7 @@ void paged_attn_decode_v2_kernel(
   if (t_local == 0) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
-      const int qh = kvh * GROUP + g;
+      const int qh = kvh * group_stride + g_base + g;
       float* wa = ws_acc +
           (((size_t)seq * n_q + qh) * slices + slice) * D + p * DS;
 #pragma unroll
@@ -886,45 +886,32 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
   const int group = n_q / n_kv;
   // v2 (GQA-amortized, pages split across waves): caller must size the
   // workspace for n_parts*PA_WAVES slices and pass want_v2 explicitly
-  const bool v2 = (want_v2 && D == 128 && group <= 4 &&
+  const bool v2 = (want_v2 && D == 128 && (group <= 4 || group == 8) &&
                    (n_parts == 1 || ws_acc != nullptr));
   if (v2) {
     dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
     const int slices = n_parts * PA_WAVES;
+#define LAUNCH_PA_V2(G, GS, GB)                                             \
+  hipLaunchKernelGGL((paged_attn_decode_v2_kernel<G>), grid, block, 0, s,   \
+                     (bf16*)out, (float*)ws_acc, (float*)ws_ml,             \
+                     (const bf16*)q, (const bf16*)kcache,                   \
+                     (const bf16*)vcache, nullptr, nullptr,                 \
+                     (const int*)block_tables, (const int*)seq_lens, n_q,   \
+                     n_kv, max_blocks, n_parts, scale, q_rstride, GS, GB)
     switch (group) {
-      case 1:
-        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<1>), grid, block,
-                           0, s, (bf16*)out, (float*)ws_acc,
-                           (float*)ws_ml, (const bf16*)q,
-                           (const bf16*)kcache, (const bf16*)vcache,
-                           nullptr, nullptr,
-                           (const int*)block_tables,
-                           (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale, q_rstride);
-        break;
-      case 2:
-        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<2>), grid, block,
-                           0, s, (bf16*)out, (float*)ws_acc,
-                           (float*)ws_ml, (const bf16*)q,
-                           (const bf16*)kcache, (const bf16*)vcache,
-                           nullptr, nullptr,
-                           (const int*)block_tables,
-                           (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale, q_rstride);
-        break;
-      case 4:
-        hipLaunchKernelGGL((paged_attn_decode_v2_kernel<4>), grid, block,
-                           0, s, (bf16*)out, (float*)ws_acc,
-                           (float*)ws_ml, (const bf16*)q,
-                           (const bf16*)kcache, (const bf16*)vcache,
-                           nullptr, nullptr,
-                           (const int*)block_tables,
-                           (const int*)seq_lens, n_q, n_kv, max_blocks,
-                           n_parts, scale, q_rstride);
+      case 1: LAUNCH_PA_V2(1, 1, 0); break;
+      case 2: LAUNCH_PA_V2(2, 2, 0); break;
+      case 4: LAUNCH_PA_V2(4, 4, 0); break;
+      case 8:
+        // 70B-class GQA group: two GROUP=4 passes (2 page reads per
+        // workgroup vs 8 under the v1 head-per-wave layout)
+        LAUNCH_PA_V2(4, 8, 0);
+        LAUNCH_PA_V2(4, 8, 4);
         break;
       default:
         goto v1_path;
     }
+#undef LAUNCH_PA_V2
     LL_CHECK(hipGetLastError());
     if (n_parts > 1) {   // single partition combined in-workgroup
       dim3 rgrid(num_seqs, n_q), rblock(WAVE);
@@ -968,23 +955,27 @@ int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
   if (block_size != PA_BLOCK || D != 128) return -2;
   if (n_q % n_kv != 0) return -2;
   const int group = n_q / n_kv;
-  if (group > 4 || (group == 3)) return -2;
+  if ((group > 4 && group != 8) || group == 3) return -2;
   if (n_parts > 1 && ws_acc == nullptr) return -2;
   hipStream_t s = (hipStream_t)stream;
   dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
   const int slices = n_parts * PA_WAVES;
-#define LAUNCH_V2_FP8(G)                                                    \
+#define LAUNCH_V2_FP8(G, GS, GB)                                            \
   hipLaunchKernelGGL((paged_attn_decode_v2_kernel<G, true>), grid, block,   \
                      0, s, (bf16*)out, (float*)ws_acc, (float*)ws_ml,       \
                      (const bf16*)q, (const bf16*)kcache,                   \
                      (const bf16*)vcache, (const float*)k_scale,            \
                      (const float*)v_scale, (const int*)block_tables,       \
                      (const int*)seq_lens, n_q, n_kv, max_blocks,           \
-                     n_parts, scale, q_rstride)
+                     n_parts, scale, q_rstride, GS, GB)
   switch (group) {
-    case 1: LAUNCH_V2_FP8(1); break;
-    case 2: LAUNCH_V2_FP8(2); break;
-    case 4: LAUNCH_V2_FP8(4); break;
+    case 1: LAUNCH_V2_FP8(1, 1, 0); break;
+    case 2: LAUNCH_V2_FP8(2, 2, 0); break;
+    case 4: LAUNCH_V2_FP8(4, 4, 0); break;
+    case 8:
+      LAUNCH_V2_FP8(4, 8, 0);
+      LAUNCH_V2_FP8(4, 8, 4);
+      break;
     default: return -2;
   }
 #undef LAUNCH_V2_FP8
