@@ -132,8 +132,7 @@ class _DecodeGraph:
         self.n_parts = n_parts
         nq, D = engine.n_q_local, engine.cfg.model.head_dim
         group = nq // engine.n_kv_local
-        use_v2 = (ops._PAGED_ATTN_V2 and D == 128
-                  and group in (1, 2, 4, 8))
+        use_v2 = ops._PAGED_ATTN_V2 and D == 128 and group <= 16
         slices = n_parts * 4 if use_v2 else n_parts
         if n_parts > 1:
             self.ws = (torch.empty(S, nq, slices, D, dtype=torch.float32,
@@ -209,10 +208,10 @@ class Engine:
                                          cfg.kv_dtype) or "bfloat16"
         if (self.kv_dtype == "fp8_e4m3" and self.device.type == "cuda"
                 and (m.head_dim != 128
-                     or m.q_heads // m.kv_heads not in (1, 2, 4, 8))):
+                     or m.q_heads // m.kv_heads > 16)):
             raise ValueError(
                 "fp8_e4m3 KV on GPU requires head_dim=128 and GQA "
-                f"group in 1/2/4/8 (got D={m.head_dim}, "
+                f"group <= 16 (got D={m.head_dim}, "
                 f"group={m.q_heads // m.kv_heads})")
         num_blocks = cfg.kv_cache_blocks or self._auto_blocks(n_kv_local)
         self.kv = KVCache(m.layers, num_blocks, n_kv_local, m.head_dim,

@@ -370,8 +370,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
     out = torch.empty(S, n_q, D, dtype=q.dtype, device=q.device)
     group = n_q // n_kv
     # v2 (GQA-amortized) always reduces through the slice workspace
-    use_v2 = (_PAGED_ATTN_V2 and D == 128
-              and group in (1, 2, 4, 8))
+    use_v2 = _PAGED_ATTN_V2 and D == 128 and group <= 16
     slices = n_parts * 4 if use_v2 else n_parts
     if n_parts > 1:
         if workspace is not None:

@@ -886,7 +886,7 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
   const int group = n_q / n_kv;
   // v2 (GQA-amortized, pages split across waves): caller must size the
   // workspace for n_parts*PA_WAVES slices and pass want_v2 explicitly
-  const bool v2 = (want_v2 && D == 128 && (group <= 4 || group == 8) &&
+  const bool v2 = (want_v2 && D == 128 && group <= 16 &&
                    (n_parts == 1 || ws_acc != nullptr));
   if (v2) {
     dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
@@ -898,18 +898,19 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                      (const bf16*)vcache, nullptr, nullptr,                 \
                      (const int*)block_tables, (const int*)seq_lens, n_q,   \
                      n_kv, max_blocks, n_parts, scale, q_rstride, GS, GB)
-    switch (group) {
-      case 1: LAUNCH_PA_V2(1, 1, 0); break;
-      case 2: LAUNCH_PA_V2(2, 2, 0); break;
-      case 4: LAUNCH_PA_V2(4, 4, 0); break;
-      case 8:
-        // 70B-class GQA group: two GROUP=4 passes (2 page reads per
-        // workgroup vs 8 under the v1 head-per-wave layout)
-        LAUNCH_PA_V2(4, 8, 0);
-        LAUNCH_PA_V2(4, 8, 4);
-        break;
-      default:
-        goto v1_path;
+    // any GQA group: greedy 4/2/1 passes at the same group_stride —
+    // ceil(group/4) page reads per workgroup instead of `group` under
+    // the v1 head-per-wave layout (e.g. 70B group 8 -> 2 reads,
+    // Qwen2-7B group 7 -> 3)
+    for (int gb = 0, rem = group; rem > 0;) {
+      const int g = rem >= 4 ? 4 : (rem >= 2 ? 2 : 1);
+      switch (g) {
+        case 4: LAUNCH_PA_V2(4, group, gb); break;
+        case 2: LAUNCH_PA_V2(2, group, gb); break;
+        default: LAUNCH_PA_V2(1, group, gb); break;
+      }
+      gb += g;
+      rem -= g;
     }
 #undef LAUNCH_PA_V2
     LL_CHECK(hipGetLastError());
@@ -955,7 +956,7 @@ int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
   if (block_size != PA_BLOCK || D != 128) return -2;
   if (n_q % n_kv != 0) return -2;
   const int group = n_q / n_kv;
-  if ((group > 4 && group != 8) || group == 3) return -2;
+  if (group > 16) return -2;
   if (n_parts > 1 && ws_acc == nullptr) return -2;
   hipStream_t s = (hipStream_t)stream;
   dim3 grid(num_seqs, n_kv, n_parts), block(PA_WAVES * WAVE);
@@ -968,15 +969,15 @@ int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
                      (const float*)v_scale, (const int*)block_tables,       \
                      (const int*)seq_lens, n_q, n_kv, max_blocks,           \
                      n_parts, scale, q_rstride, GS, GB)
-  switch (group) {
-    case 1: LAUNCH_V2_FP8(1, 1, 0); break;
-    case 2: LAUNCH_V2_FP8(2, 2, 0); break;
-    case 4: LAUNCH_V2_FP8(4, 4, 0); break;
-    case 8:
-      LAUNCH_V2_FP8(4, 8, 0);
-      LAUNCH_V2_FP8(4, 8, 4);
-      break;
-    default: return -2;
+  for (int gb = 0, rem = group; rem > 0;) {
+    const int g = rem >= 4 ? 4 : (rem >= 2 ? 2 : 1);
+    switch (g) {
+      case 4: LAUNCH_V2_FP8(4, group, gb); break;
+      case 2: LAUNCH_V2_FP8(2, group, gb); break;
+      default: LAUNCH_V2_FP8(1, group, gb); break;
+    }
+    gb += g;
+    rem -= g;
   }
 #undef LAUNCH_V2_FP8
   LL_CHECK(hipGetLastError());

@@ -267,3 +267,28 @@ def test_paged_attn_v2_group8():
                                  n_parts=2)
     rel2 = (out2.float().cpu() - ref).abs().max() / ref.abs().max()
     assert rel2 < 5e-2, rel2.item()
+
+
+@pytest.mark.gpu
+def test_paged_attn_v2_odd_groups():
+    """Greedy 4/2/1 decomposition covers any GQA group (3, 5, 7...)."""
+    torch.manual_seed(8)
+    dev = "cuda:0"
+    import llmlb_amd.ops.reference as R
+    for nq, nkv in ((6, 2), (14, 2), (5, 1)):   # groups 3, 7, 5
+        S, D, BS, nb = 3, 128, 16, 10
+        kcache = torch.randn(nb, nkv, BS, D, dtype=torch.bfloat16,
+                             device=dev)
+        vcache = torch.randn_like(kcache)
+        q = torch.randn(S, nq, D, dtype=torch.bfloat16, device=dev)
+        bt = torch.arange(1, 1 + S * 2, dtype=torch.int32,
+                          device=dev).reshape(S, 2)
+        sl = torch.tensor([19, 32, 8], dtype=torch.int32, device=dev)
+        out = ops.paged_attn_decode(q, kcache, vcache, bt, sl,
+                                    D ** -0.5)
+        ref = R.paged_attn_decode(q.float().cpu(),
+                                  kcache.float().cpu(),
+                                  vcache.float().cpu(), bt.cpu(),
+                                  sl.cpu(), D ** -0.5)
+        rel = (out.float().cpu() - ref).abs().max() / ref.abs().max()
+        assert rel < 5e-2, (nq, nkv, rel.item())
