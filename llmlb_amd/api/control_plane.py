@@ -534,7 +534,7 @@ def create_app(db_path: str | None = None,
             with open(path, "rb") as f:
                 f.seek(0, 2)
                 size = f.tell()
-                f.seek(max(0, size - 1 << 20))
+                f.seek(max(0, size - (1 << 20)))
                 out = f.read().decode(errors="replace").splitlines()
         except OSError:
             pass

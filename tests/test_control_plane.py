@@ -223,3 +223,25 @@ def test_ws_dashboard_auth_and_stream(app_client):
             "/ws/dashboard?token=admtok") as ws:
         msg = ws.receive_json()
         assert msg["type"] in ("summary", "event", "error")
+
+
+def test_logs_tail_reads_rotated_file(app_client, tmp_path,
+                                      monkeypatch):
+    """/api/logs tails the newest dated llmlb-gw-YYYYMMDD.jsonl."""
+    import llmlb_amd.api.control_plane as cp
+    logd = tmp_path / "logs"
+    logd.mkdir()
+    (logd / "llmlb-gw-20260101.jsonl").write_text('{"old":1}\n')
+    (logd / "llmlb-gw-20260914.jsonl").write_text(
+        '{"msg":"a"}\n{"msg":"b"}\n')
+    monkeypatch.setenv("LLMLB_LOG_DIR", str(logd))
+    from llmlb_amd.config import get_config
+    get_config(refresh=True)
+    app = cp.create_app(db_path=":memory:", jwt_secret="s",
+                        admin_token="tok2")
+    from fastapi.testclient import TestClient
+    c = TestClient(app)
+    r = c.get("/api/logs", headers={"Authorization": "Bearer tok2"})
+    assert r.status_code == 200
+    lines = r.json()["lines"]
+    assert '{"msg":"b"}' in lines and '{"old":1}' not in lines
