@@ -538,3 +538,73 @@ def test_admission_429_when_saturated():
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def test_endpoint_persistence_across_restart():
+    """Registered endpoints survive a gateway restart (SQLite-backed
+    registry reload + health re-check brings them back online)."""
+    mock_port, gw_port = _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    mock = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "persist-model"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "2"})
+
+    def start_gw():
+        return subprocess.Popen(
+            [str(BIN / "llmlb-gw"), "--workers", "1"], env=env,
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+
+    def wait_health(timeout=15):
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            try:
+                _req("GET", f"http://127.0.0.1:{gw_port}/health",
+                     timeout=2)
+                return True
+            except Exception:
+                time.sleep(0.1)
+        return False
+
+    gw = start_gw()
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        assert wait_health()
+        status, _ = _req("POST", base + "/api/endpoints",
+                         {"name": "p1",
+                          "base_url": f"http://127.0.0.1:{mock_port}"},
+                         headers={"Authorization": "Bearer t"})
+        assert status == 201
+        status, _ = _req("POST", base + "/v1/chat/completions",
+                         {"model": "persist-model",
+                          "messages": [{"role": "user",
+                                        "content": "x"}]})
+        assert status == 200
+        gw.terminate()
+        gw.wait(timeout=10)
+        gw = start_gw()
+        assert wait_health()
+        deadline = time.time() + 20
+        ok = False
+        while time.time() < deadline:
+            status, _ = _req("POST", base + "/v1/chat/completions",
+                             {"model": "persist-model",
+                              "messages": [{"role": "user",
+                                            "content": "x"}]})
+            if status == 200:
+                ok = True
+                break
+            time.sleep(0.5)
+        assert ok, f"model not routable after restart (last {status})"
+    finally:
+        gw.terminate()
+        mock.terminate()
+        for p in (gw, mock):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
