@@ -75,6 +75,7 @@ struct Histo {
 
 struct ThreadStats {
   Histo histo;
+  Histo ttfb;       // time to first response byte (streaming TTFT proxy)
   uint64_t non2xx = 0;
   uint64_t socket_errors = 0;
 };
@@ -90,6 +91,7 @@ struct BenchConn : FdHandler {
   size_t out_off = 0;
   bool connected = false;
   int64_t sent_us = 0;
+  bool ttfb_recorded = false;
   HttpResponse resp;
   bool have_head = false;
   BodyReader body;
@@ -124,6 +126,7 @@ struct BenchConn : FdHandler {
   }
   void send_request() {
     sent_us = now_us();
+    ttfb_recorded = false;
     out_off = 0;
     have_head = false;
     resp = HttpResponse{};
@@ -176,6 +179,10 @@ struct BenchConn : FdHandler {
       char* p = in.reserve(32768);
       ssize_t n = ::recv(fd, p, 32768, 0);
       if (n > 0) {
+        if (!ttfb_recorded) {
+          stats->ttfb.add(now_us() - sent_us);
+          ttfb_recorded = true;
+        }
         in.commit((size_t)n);
       } else if (n == 0) {
         ++stats->socket_errors;
@@ -294,10 +301,11 @@ int main(int argc, char** argv) {
   for (auto& th : threads) th.join();
   double elapsed = (now_us() - t0) / 1e6;
 
-  Histo total;
+  Histo total, ttfb;
   uint64_t non2xx = 0, socket_errors = 0;
   for (auto& s : stats) {
     total.merge(s.histo);
+    ttfb.merge(s.ttfb);
     non2xx += s.non2xx;
     socket_errors += s.socket_errors;
   }
@@ -305,21 +313,27 @@ int main(int argc, char** argv) {
   if (g_cfg.json_out) {
     printf("{\"label\":\"%s\",\"rps\":%.2f,\"p50_ms\":%.3f,\"p75_ms\":%.3f,"
            "\"p90_ms\":%.3f,\"p95_ms\":%.3f,\"p99_ms\":%.3f,"
-           "\"mean_ms\":%.3f,\"non2xx\":%llu,\"socket_errors\":%llu,"
+           "\"mean_ms\":%.3f,\"ttfb_p50_ms\":%.3f,"
+           "\"ttfb_p99_ms\":%.3f,\"non2xx\":%llu,"
+           "\"socket_errors\":%llu,"
            "\"requests\":%llu,\"duration_s\":%.2f}\n",
            g_cfg.label.c_str(), rps, total.percentile(0.50),
            total.percentile(0.75), total.percentile(0.90),
            total.percentile(0.95), total.percentile(0.99),
            total.count ? (double)total.sum_us / total.count / 1000.0 : 0.0,
+           ttfb.percentile(0.50), ttfb.percentile(0.99),
            (unsigned long long)non2xx, (unsigned long long)socket_errors,
            (unsigned long long)total.count, elapsed);
   } else {
-    printf("label,rps,p50_ms,p75_ms,p90_ms,p95_ms,p99_ms,non2xx,"
+    printf("label,rps,p50_ms,p75_ms,p90_ms,p95_ms,p99_ms,"
+           "ttfb_p50_ms,ttfb_p99_ms,non2xx,"
            "socket_errors,requests,duration_s\n");
-    printf("%s,%.2f,%.3f,%.3f,%.3f,%.3f,%.3f,%llu,%llu,%llu,%.2f\n",
+    printf("%s,%.2f,%.3f,%.3f,%.3f,%.3f,%.3f,%.3f,%.3f,%llu,%llu,%llu,"
+           "%.2f\n",
            g_cfg.label.c_str(), rps, total.percentile(0.50),
            total.percentile(0.75), total.percentile(0.90),
            total.percentile(0.95), total.percentile(0.99),
+           ttfb.percentile(0.50), ttfb.percentile(0.99),
            (unsigned long long)non2xx, (unsigned long long)socket_errors,
            (unsigned long long)total.count, elapsed);
   }
