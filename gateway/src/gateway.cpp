@@ -519,6 +519,19 @@ void Worker::start() {
 void Worker::join() {
   loop.stop();
   if (thread.joinable()) thread.join();
+  // free idle pooled upstream connections: their deferred-delete path
+  // runs on the (now stopped) event loop, so reclaim directly.  Only
+  // still-open conns are deleted here — closed ones are owned by the
+  // loop's deferred queue.
+  for (auto& [key, conns] : pool) {
+    for (UpstreamConn* uc : conns) {
+      if (!uc->closed) {
+        if (uc->fd >= 0) ::close(uc->fd);
+        delete uc;
+      }
+    }
+  }
+  pool.clear();
 }
 
 UpstreamConn* Worker::acquire_upstream(const EndpointPtr& ep,
