@@ -69,9 +69,9 @@ void EventLoop::wakeup() {
 }
 
 void EventLoop::run() {
-  running_ = true;
+  running_.store(true, std::memory_order_relaxed);
   epoll_event events[256];
-  while (running_) {
+  while (running_.load(std::memory_order_relaxed)) {
     int timeout = -1;
     int64_t now = now_us();
     while (!timers_.empty()) {

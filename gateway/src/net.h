@@ -9,6 +9,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <functional>
 #include <queue>
 #include <vector>
@@ -42,7 +43,7 @@ class EventLoop {
   void post(std::function<void()> fn);
 
   void run();
-  void stop() { running_ = false; wakeup(); }
+  void stop() { running_.store(false, std::memory_order_relaxed); wakeup(); }
   void wakeup();
 
   // deferred deletion (safe while iterating epoll events)
@@ -58,7 +59,7 @@ class EventLoop {
   };
   int epfd_ = -1;
   int wake_fd_ = -1;
-  bool running_ = false;
+  std::atomic<bool> running_{false};
   std::priority_queue<Timer, std::vector<Timer>, std::greater<Timer>> timers_;
   std::unordered_map<uint64_t, TimerFn> timer_fns_;
   uint64_t next_timer_id_ = 1;
