@@ -4,7 +4,7 @@
 # concurrency.  Emits the loadgen CSV + gateway and backend health.
 set -e
 cd "$(dirname "$0")/.."
-DUR=${DUR:-45} CONNS=${CONNS:-192} GEN=${GEN:-16}
+DUR=${DUR:-45} CONNS=${CONNS:-192} GEN=${GEN:-16} PROMPT_WORDS=${PROMPT_WORDS:-8}
 GW=19610 BE=19611
 export LLMLB_NO_AUDIT=   # audit ON (realistic)
 python -m llmlb_amd.backend.launcher --model llama-3-8b --device cuda:0 \
@@ -23,7 +23,8 @@ curl -sf -X POST 127.0.0.1:$GW/api/endpoints \
   -H 'Authorization: Bearer admintok' \
   -d "{\"name\":\"soak\",\"base_url\":\"http://127.0.0.1:$BE\",\"device_index\":0}" >/dev/null
 sleep 3   # let sync discover the model
-BODY="{\"model\":\"llama-3-8b\",\"max_tokens\":$GEN,\"messages\":[{\"role\":\"user\",\"content\":\"soak test prompt with some words\"}]}"
+PROMPT=$(python3 -c "import sys; print(' '.join(f'w{i}' for i in range(int(sys.argv[1]))))" $PROMPT_WORDS)
+BODY="{\"model\":\"llama-3-8b\",\"max_tokens\":$GEN,\"stream\":true,\"messages\":[{\"role\":\"user\",\"content\":\"$PROMPT\"}]}"
 gateway/bin/llmlb-bench --host 127.0.0.1 --port $GW \
   --path /v1/chat/completions --method POST --body "$BODY" \
   --threads 8 --connections $CONNS --duration $DUR --label gpu_soak

@@ -56,6 +56,11 @@ class EngineConfig:
     # prompts longer than this prefill in chunks (activation-memory
     # bound for long prompts; keeps decode latency smooth)
     prefill_chunk: int = 4096
+    # cap on prompt tokens ADMITTED per step (fairness: a decode step
+    # runs between admission waves, bounding time-between-tokens for
+    # running streams).  None = max_prefill_tokens.  Env
+    # LLMLB_PREFILL_STEP_BUDGET overrides.
+    prefill_step_budget: int | None = None
     # HF-format safetensors checkpoint (file or shard dir); None =
     # deterministic random init (synthetic serving)
     weights_path: str | None = None
@@ -270,6 +275,11 @@ class Engine:
         self._graphs: dict[int, _DecodeGraph] = {}
         self.timing = defaultdict(float) \
             if os_env_flag("LLMLB_ENGINE_TIMING", False) else None
+        import os as _os2
+        _sb = _os2.environ.get("LLMLB_PREFILL_STEP_BUDGET")
+        self._step_budget = (int(_sb) if _sb
+                             else (cfg.prefill_step_budget
+                                   or cfg.max_prefill_tokens))
 
     # -- sizing ----------------------------------------------------------
     def _auto_blocks(self, n_kv_local: int) -> int:
@@ -356,7 +366,7 @@ class Engine:
     # -- scheduling ------------------------------------------------------
     def _admit(self) -> list[_Sequence]:
         admitted: list[_Sequence] = []
-        budget = self.cfg.max_prefill_tokens
+        budget = self._step_budget
         while (len(self.running) + len(admitted) < self.cfg.max_batch
                and budget > 0 and self._free_rows):
             try:
