@@ -5,7 +5,7 @@
 set -e
 cd "$(dirname "$0")/.."
 DUR=${DUR:-45} CONNS=${CONNS:-192} GEN=${GEN:-16} PROMPT_WORDS=${PROMPT_WORDS:-8}
-GW=19610 BE=19611
+GW=${GW_PORT:-19610} BE=${BE_PORT:-19611}
 export LLMLB_NO_AUDIT=   # audit ON (realistic)
 python -m llmlb_amd.backend.launcher --model llama-3-8b --device cuda:0 \
   --port $BE --kv-blocks 16384 >/tmp/soak_be.log 2>&1 &
