@@ -136,7 +136,29 @@ std::string Gateway::render_metrics_text() {
       metrics.upstream_errors.load(), metrics.streamed_responses.load(),
       metrics.tokens_streamed.load(), metrics.cloud_requests.load(),
       metrics.active_requests.load());
-  return std::string(buf, n);
+  std::string out(buf, n);
+  // per-endpoint gauges (ops dashboards scrape these to see the
+  // balancer's view of each replica)
+  out += "# TYPE llmlb_endpoint_active gauge\n"
+         "# TYPE llmlb_endpoint_requests_total counter\n"
+         "# TYPE llmlb_endpoint_failures_total counter\n"
+         "# TYPE llmlb_endpoint_online gauge\n";
+  for (auto& ep : registry.list()) {
+    char line[512];
+    int m = snprintf(
+        line, sizeof line,
+        "llmlb_endpoint_active{endpoint=\"%s\"} %lld\n"
+        "llmlb_endpoint_requests_total{endpoint=\"%s\"} %lld\n"
+        "llmlb_endpoint_failures_total{endpoint=\"%s\"} %lld\n"
+        "llmlb_endpoint_online{endpoint=\"%s\"} %d\n",
+        ep->name.c_str(), (long long)ep->active.load(),
+        ep->name.c_str(), (long long)ep->total_requests.load(),
+        ep->name.c_str(), (long long)ep->total_failures.load(),
+        ep->name.c_str(),
+        ep->status.load() == EndpointStatus::kOnline ? 1 : 0);
+    out.append(line, m);
+  }
+  return out;
 }
 
 // capability-routed backend pick: FIRST online endpoint with the

@@ -284,6 +284,12 @@ def test_dashboard_summary(stack):
     assert obj["last_hour"]["requests"] >= 1
 
 
+def test_prometheus_per_endpoint_series(stack):
+    _, text = _req("GET", stack["base"] + "/metrics")
+    assert 'llmlb_endpoint_online{endpoint="mock1"} 1' in text
+    assert 'llmlb_endpoint_requests_total{endpoint="mock1"}' in text
+
+
 def test_prometheus_metrics(stack):
     status, body = _req("GET", stack["base"] + "/api/metrics")
     assert status == 200
