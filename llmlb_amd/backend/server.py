@@ -370,10 +370,24 @@ def build_asgi_app(server: BackendServer):
             await send_json(send, 200, server.health_payload())
             return
         if method == "GET" and path == "/metrics":
+            eng = server.engine
+            extra = ""
+            if eng.prefix is not None:
+                extra += (f"backend_prefix_hit_tokens "
+                          f"{eng.prefix.hit_tokens}\n"
+                          f"backend_prefix_prompt_tokens "
+                          f"{eng.prefix.prompt_tokens}\n")
+            if eng.spec_ngram > 0:
+                extra += (f"backend_spec_proposed {eng.spec_proposed}\n"
+                          f"backend_spec_accepted {eng.spec_accepted}\n")
             text = (
                 f"backend_requests_total {server.requests_total}\n"
-                f"backend_tokens_total {server.engine.tokens_out}\n"
-                f"backend_active_requests {server.engine.num_running}\n"
+                f"backend_tokens_total {eng.tokens_out}\n"
+                f"backend_active_requests {eng.num_running}\n"
+                f"backend_tokens_per_sec {eng.tokens_per_sec_ema:.2f}\n"
+                f"backend_gpu_step_ms {eng.gpu_step_ms_ema:.3f}\n"
+                f"backend_kv_free_pages {eng.kv.free_blocks()}\n"
+                f"{extra}"
             ).encode()
             await send({"type": "http.response.start", "status": 200,
                         "headers": [(b"content-type", b"text/plain")]})
