@@ -614,3 +614,16 @@ def test_endpoint_persistence_across_restart():
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def test_cloud_prefix_typo_alias(stack):
+    """`ahtnorpic:` is the reference's documented typo-alias for
+    anthropic (api/openai.rs:637-655); unconfigured here -> 502-class
+    error naming the anthropic provider, not a 404."""
+    status, body = _req("POST", stack["base"] + "/v1/chat/completions",
+                        {"model": "ahtnorpic:claude-3-5-haiku",
+                         "messages": [{"role": "user",
+                                       "content": "hi"}]},
+                        headers=stack["auth"])
+    assert status in (400, 502, 503), (status, body)
+    assert "anthropic" in body.lower()
