@@ -333,14 +333,35 @@ def build_asgi_app(server: BackendServer):
     class _AsyncSink:
         """Engine-thread -> asyncio bridge: put() is called from the
         engine loop thread; items land in an asyncio.Queue without
-        blocking any executor thread (scales past 64 streams)."""
+        blocking any executor thread (scales past 64 streams).
+
+        Writes COALESCE: a decode step emits one token for every
+        running stream back to back, so put() buffers under a lock
+        and schedules at most one loop wakeup until drained — ~64x
+        fewer call_soon_threadsafe round trips per step."""
 
         def __init__(self):
+            import threading as _th
             self.loop = asyncio.get_running_loop()
             self.q: asyncio.Queue = asyncio.Queue()
+            self._buf: list = []
+            self._mu = _th.Lock()
+            self._scheduled = False
 
         def put(self, item):
-            self.loop.call_soon_threadsafe(self.q.put_nowait, item)
+            with self._mu:
+                self._buf.append(item)
+                if self._scheduled:
+                    return
+                self._scheduled = True
+            self.loop.call_soon_threadsafe(self._drain)
+
+        def _drain(self):
+            with self._mu:
+                items, self._buf = self._buf, []
+                self._scheduled = False
+            for it in items:
+                self.q.put_nowait(it)
 
     async def drain_queue_async(sink):
         while True:
