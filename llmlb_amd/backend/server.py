@@ -515,6 +515,19 @@ def build_asgi_app(server: BackendServer):
                 (body.get("stream_options") or {}).get("include_usage"))
 
             want_lp = server._parse_logprobs(body) > 0
+            stops = body.get("stop") or []
+            if isinstance(stops, str):
+                stops = [stops]
+            stops = [s for s in map(str, stops)][:4]
+
+            def find_stop(text: str):
+                """Earliest stop-sequence hit -> cut position."""
+                best = -1
+                for s in stops:
+                    i = text.find(s)
+                    if i >= 0 and (best < 0 or i < best):
+                        best = i
+                return best
 
             def lp_entry(tok, extra):
                 return {
@@ -543,19 +556,26 @@ def build_asgi_app(server: BackendServer):
                     if disconnected.is_set():
                         break
                     toks, finish, lp_content = [], "stop", []
+                    text = ""
                     async for tok, fin, extra in drain_queue_async(
                             sink_i):
                         toks.append(tok)
+                        text += server.tokenizer.decode_token(tok)
                         if extra is not None:
                             lp_content.append(lp_entry(tok, extra))
                         if fin:
                             finish = fin
+                        if stops and find_stop(text) >= 0:
+                            finish = "stop"
+                            server.engine.abort(rids[ci])
+                            break
                         if disconnected.is_set():
                             break
                     if disconnected.is_set():
                         break
-                    text = "".join(server.tokenizer.decode_token(t)
-                                   for t in toks)
+                    cut = find_stop(text) if stops else -1
+                    if cut >= 0:
+                        text = text[:cut]
                     total_out += len(toks)
                     if is_chat:
                         choices.append(
@@ -627,11 +647,27 @@ def build_asgi_app(server: BackendServer):
                 if disconnected.is_set():
                     break
                 finish = "stop"
+                acc = ""
                 async for tok, fin, extra in drain_queue_async(sink_i):
                     ntok += 1
                     if fin:
                         finish = fin
                     text = server.tokenizer.decode_token(tok)
+                    if stops:
+                        cut = find_stop(acc + text)
+                        if cut >= 0:
+                            # emit only up to the stop sequence
+                            text = (acc + text)[:cut][len(acc):]
+                            finish = "stop"
+                            server.engine.abort(rids[ci])
+                            if text:
+                                await send({
+                                    "type": "http.response.body",
+                                    "body": chunk_payload(
+                                        {"content": text}, index=ci),
+                                    "more_body": True})
+                            break
+                        acc += text
                     payload = chunk_payload({"content": text}, index=ci)
                     if extra is not None and is_chat:
                         obj = json.loads(payload[len(b"data: "):])

@@ -211,3 +211,31 @@ def test_context_length_exceeded(stack):
     except urllib.error.HTTPError as e:
         assert e.code == 400
         assert b"context_length_exceeded" in e.read()
+
+
+def test_stop_strings(stack):
+    """OpenAI `stop`: generation truncates at the stop sequence and
+    reports finish_reason=stop (synthetic tokenizer emits 'tokN '
+    pieces, so stopping on a piece we know will appear works)."""
+    # discover the first generated piece for this prompt
+    base_req = {"model": "llama-tiny", "max_tokens": 8,
+                "messages": [{"role": "user", "content": "stop test"}]}
+    r = urllib.request.Request(
+        stack["backend"] + "/v1/chat/completions",
+        data=json.dumps(base_req).encode(),
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(r, timeout=30) as resp:
+        full = json.loads(resp.read())["choices"][0]["message"]["content"]
+    pieces = full.split()
+    assert len(pieces) >= 3
+    stop_piece = pieces[2]
+    r2 = urllib.request.Request(
+        stack["backend"] + "/v1/chat/completions",
+        data=json.dumps({**base_req, "stop": stop_piece}).encode(),
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(r2, timeout=30) as resp:
+        out = json.loads(resp.read())
+    ch = out["choices"][0]
+    assert ch["finish_reason"] == "stop"
+    assert stop_piece not in ch["message"]["content"]
+    assert ch["message"]["content"].split() == pieces[:2]
