@@ -238,4 +238,6 @@ def test_stop_strings(stack):
     ch = out["choices"][0]
     assert ch["finish_reason"] == "stop"
     assert stop_piece not in ch["message"]["content"]
-    assert ch["message"]["content"].split() == pieces[:2]
+    # truncation happens at the FIRST occurrence of the stop string
+    # anywhere in the decoded text (OpenAI substring semantics)
+    assert ch["message"]["content"] == full[:full.index(stop_piece)]
