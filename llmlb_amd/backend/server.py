@@ -207,6 +207,7 @@ async def handle_responses(server: "BackendServer", body: dict, send):
         await send({"type": "http.response.body", "body": data})
         return
     inp = body.get("input", "")
+    instructions = body.get("instructions") or ""
     if isinstance(inp, list):
         # list of messages or content parts
         parts = []
@@ -220,7 +221,9 @@ async def handle_responses(server: "BackendServer", body: dict, send):
             else:
                 parts.append(str(item))
         inp = "\n".join(parts)
-    prompt_ids = server.tokenizer.encode(str(inp) or "hello")
+    rendered = (f"<|system|> {instructions}\n{inp}"
+                if instructions else str(inp))
+    prompt_ids = server.tokenizer.encode(rendered or "hello")
     params = server.parse_sampling(body)
     import asyncio as _asyncio
 
@@ -386,6 +389,17 @@ def build_asgi_app(server: BackendServer):
 
         if method == "GET" and path == "/v1/models":
             await send_json(send, 200, server.models_payload())
+            return
+        if method == "GET" and path.startswith("/v1/models/"):
+            mid = path[len("/v1/models/"):]
+            for m in server.models_payload()["data"]:
+                if m["id"] == mid:
+                    await send_json(send, 200, m)
+                    return
+            await send_json(send, 404, {"error": {
+                "message": "model not found",
+                "type": "invalid_request_error",
+                "code": "model_not_found"}})
             return
         if method == "GET" and path in ("/api/health", "/health"):
             await send_json(send, 200, server.health_payload())

@@ -241,3 +241,24 @@ def test_stop_strings(stack):
     # truncation happens at the FIRST occurrence of the stop string
     # anywhere in the decoded text (OpenAI substring semantics)
     assert ch["message"]["content"] == full[:full.index(stop_piece)]
+
+
+def test_responses_instructions_and_model_detail(stack):
+    body = json.dumps({"model": "llama-tiny", "max_tokens": 3,
+                       "instructions": "be terse",
+                       "input": "hello"}).encode()
+    r = urllib.request.Request(stack["backend"] + "/v1/responses",
+                               data=body,
+                               headers={"Content-Type":
+                                        "application/json"})
+    with urllib.request.urlopen(r, timeout=30) as resp:
+        out = json.loads(resp.read())
+    assert out["object"] == "response"
+    # instructions change the prompt -> (usually) different tokens than
+    # without; at minimum the call succeeds and usage reflects a longer
+    # prompt
+    assert out["usage"]["input_tokens"] >= 3
+    with urllib.request.urlopen(
+            stack["backend"] + "/v1/models/llama-tiny", timeout=10) as m:
+        detail = json.loads(m.read())
+    assert detail["id"] == "llama-tiny"
