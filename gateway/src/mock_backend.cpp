@@ -25,6 +25,7 @@ struct MockConfig {
   std::vector<std::string> models = {"mock-model"};
   int delay_ms = 0;          // simulated inference latency
   bool delay_inference_only = false;
+  double health_tps = 0;     // advertise tokens_per_sec in /api/health
   int stream_chunks = 8;     // SSE deltas per streaming response
   double fail_rate = 0.0;    // inject failures (config 5 testing)
   int completion_tokens = 8;
@@ -287,6 +288,21 @@ void MockConn::respond() {
                        b.size());
       out.append(h, n);
       out.append(b);
+    } else if (path == "/api/health" && g_cfg.health_tps > 0) {
+      char b[256];
+      int bn = snprintf(b, sizeof b,
+                        "{\"status\":\"ok\",\"model\":\"%s\","
+                        "\"tokens_per_sec\":%.1f}",
+                        g_cfg.models.empty() ? "m"
+                                             : g_cfg.models[0].c_str(),
+                        g_cfg.health_tps);
+      char h[128];
+      int hn = snprintf(h, sizeof h,
+                        "Content-Type: application/json\r\n"
+                        "Content-Length: %d\r\n\r\n", bn);
+      append_status_line(out, 200, "OK");
+      out.append(h, hn);
+      out.append(b, bn);
     } else if (path == "/health") {
       append_status_line(out, 200, "OK");
       out.append("Content-Length: 2\r\n\r\nok");
@@ -321,6 +337,7 @@ int main(int argc, char** argv) {
     else if (a == "--delay-ms") g_cfg.delay_ms = atoi(next());
     else if (a == "--delay-inference-only")
       g_cfg.delay_inference_only = true;
+    else if (a == "--health-tps") g_cfg.health_tps = atof(next());
     else if (a == "--stream-chunks") g_cfg.stream_chunks = atoi(next());
     else if (a == "--fail-rate") g_cfg.fail_rate = atof(next());
   }

@@ -205,3 +205,66 @@ def test_upstream_failures_normalized_502(fleet):
     # TPS steering may avoid the bad endpoint entirely — both outcomes
     # are correct; the shape assertion only applies when a 502 occurs
     assert True or seen502
+
+
+def test_backend_tps_seeds_routing():
+    """A backend advertising tokens_per_sec in /api/health seeds the
+    router's TPS estimate (bootstrap.rs:143-159 analog): with no
+    production samples yet, traffic prefers the faster-advertising
+    replica."""
+    ports = {k: _free_port() for k in ("fast", "slow", "gw")}
+    procs = [
+        subprocess.Popen([str(BIN / "llmlb-mock"), "--port",
+                          str(ports["fast"]), "--workers", "1",
+                          "--models", "seeded", "--health-tps", "900"],
+                         stdout=subprocess.DEVNULL,
+                         stderr=subprocess.DEVNULL),
+        subprocess.Popen([str(BIN / "llmlb-mock"), "--port",
+                          str(ports["slow"]), "--workers", "1",
+                          "--models", "seeded", "--health-tps", "3"],
+                         stdout=subprocess.DEVNULL,
+                         stderr=subprocess.DEVNULL),
+    ]
+    import os
+    db = tempfile.mktemp(suffix=".sqlite3")
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(ports["gw"]),
+                "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "1"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    procs.append(gw)
+    base = f"http://127.0.0.1:{ports['gw']}"
+    admin = {"Authorization": "Bearer t"}
+    try:
+        for p in ("fast", "slow"):
+            assert _wait_port(ports[p])
+        assert _wait_port(ports["gw"])
+        for name in ("fast", "slow"):
+            status, _ = _req("POST", base + "/api/endpoints",
+                             {"name": name,
+                              "base_url":
+                                  f"http://127.0.0.1:{ports[name]}"},
+                             headers=admin)
+            assert status == 201
+        # wait until the health loop has seeded TPS for both
+        time.sleep(2.5)
+        for _ in range(10):
+            _req("POST", base + "/v1/chat/completions",
+                 {"model": "seeded",
+                  "messages": [{"role": "user", "content": "x"}]})
+        _, body = _req("GET", base + "/api/dashboard/summary",
+                       headers=admin)
+        eps = {e["name"]: e for e in json.loads(body)["endpoints"]}
+        assert eps["fast"]["total_requests"] > \
+            eps["slow"]["total_requests"], eps
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
