@@ -627,3 +627,43 @@ def test_cloud_prefix_typo_alias(stack):
                         headers=stack["auth"])
     assert status in (400, 502, 503), (status, body)
     assert "anthropic" in body.lower()
+
+
+def test_garbage_storm_survival(stack):
+    """Random garbage, truncated heads, oversized lines and abrupt
+    closes must not crash or wedge the gateway."""
+    import random
+    host, port = stack["base"].replace("http://", "").split(":")
+    rng = random.Random(7)
+    payloads = [
+        b"\x00\xff\xfe\x01" * 64,
+        b"GET / HTTP/9.9\r\n\r\n",
+        b"POST /v1/chat/completions HTTP/1.1\r\nContent-Length: "
+        b"99999999999999999999\r\n\r\n",
+        b"A" * 70000,
+        b"GET " + b"/x" * 8000 + b" HTTP/1.1\r\n\r\n",
+        b"POST /v1/chat/completions HTTP/1.1\r\nTransfer-Encoding: "
+        b"chunked\r\n\r\nzzz\r\n",
+    ]
+    for _ in range(40):
+        p = rng.choice(payloads)
+        try:
+            s = socket.create_connection((host, int(port)), timeout=5)
+            s.sendall(p[:rng.randrange(1, len(p) + 1)])
+            if rng.random() < 0.5:
+                s.close()
+            else:
+                s.settimeout(0.5)
+                try:
+                    s.recv(4096)
+                except OSError:
+                    pass
+                s.close()
+        except OSError:
+            pass
+    # gateway still serves normal traffic afterwards
+    status, _ = _req("POST", stack["base"] + "/v1/chat/completions",
+                     {"model": "mock-model",
+                      "messages": [{"role": "user", "content": "ok"}]},
+                     headers=stack["auth"])
+    assert status == 200
