@@ -645,7 +645,7 @@ def test_garbage_storm_survival(stack):
         b"POST /v1/chat/completions HTTP/1.1\r\nTransfer-Encoding: "
         b"chunked\r\n\r\nzzz\r\n",
     ]
-    for _ in range(40):
+    for _ in range(16):
         p = rng.choice(payloads)
         try:
             s = socket.create_connection((host, int(port)), timeout=5)
@@ -653,7 +653,7 @@ def test_garbage_storm_survival(stack):
             if rng.random() < 0.5:
                 s.close()
             else:
-                s.settimeout(0.5)
+                s.settimeout(0.2)
                 try:
                     s.recv(4096)
                 except OSError:
