@@ -245,3 +245,39 @@ def test_logs_tail_reads_rotated_file(app_client, tmp_path,
     assert r.status_code == 200
     lines = r.json()["lines"]
     assert '{"msg":"b"}' in lines and '{"old":1}' not in lines
+
+
+def test_requests_csv_export(app_client, tmp_path, monkeypatch):
+    import llmlb_amd.api.control_plane as cp
+    db = str(tmp_path / "csv.sqlite3")
+    conn = sqlite3.connect(db)
+    conn.executescript("""
+    CREATE TABLE users (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      username TEXT UNIQUE, password_hash TEXT, role TEXT DEFAULT 'user',
+      must_change_password INTEGER DEFAULT 0, created_at TEXT);
+    CREATE TABLE request_history (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      ts_ms INTEGER, endpoint_id TEXT, model TEXT, api_kind TEXT,
+      status INTEGER, duration_ms REAL, prompt_tokens INTEGER,
+      completion_tokens INTEGER, client_ip TEXT, error TEXT,
+      request_body TEXT, response_body TEXT, request_id TEXT);
+    """)
+    import time as _t
+    now = int(_t.time() * 1000)
+    conn.execute(
+        "INSERT INTO request_history (ts_ms,endpoint_id,model,api_kind,"
+        "status,duration_ms,prompt_tokens,completion_tokens,client_ip,"
+        "error) VALUES (?,?,?,?,?,?,?,?,?,?)",
+        (now, "e1", "m1", "chat_completions", 200, 12.5, 10, 20,
+         "1.2.3.4", ""))
+    conn.commit()
+    conn.close()
+    app = cp.create_app(db_path=db, jwt_secret="s", admin_token="tok3")
+    from fastapi.testclient import TestClient
+    c = TestClient(app)
+    r = c.get("/api/requests/export.csv",
+              headers={"Authorization": "Bearer tok3"})
+    assert r.status_code == 200
+    lines = r.text.strip().splitlines()
+    assert lines[0].startswith("ts_ms,endpoint_id,model")
+    assert any("m1" in ln and "chat_completions" in ln
+               for ln in lines[1:])

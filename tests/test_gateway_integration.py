@@ -667,3 +667,63 @@ def test_garbage_storm_survival(stack):
                       "messages": [{"role": "user", "content": "ok"}]},
                      headers=stack["auth"])
     assert status == 200
+
+
+def test_capture_bodies_flag():
+    """LLMLB_CAPTURE_BODIES=1 persists truncated request/response
+    bodies into request_history (off by default for throughput)."""
+    mock_port, gw_port = _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    mock = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "cap-model"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t", "LLMLB_CAPTURE_BODIES": "1",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        for _ in range(50):
+            try:
+                _req("GET", base + "/health", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        _req("POST", base + "/api/endpoints",
+             {"name": "c", "base_url": f"http://127.0.0.1:{mock_port}"},
+             headers={"Authorization": "Bearer t"})
+        _req("POST", base + "/v1/chat/completions",
+             {"model": "cap-model",
+              "messages": [{"role": "user",
+                            "content": "capture-me-marker"}]})
+        import sqlite3 as sq
+        deadline = time.time() + 10
+        found = False
+        while time.time() < deadline and not found:
+            try:
+                conn = sq.connect(db)
+                row = conn.execute(
+                    "SELECT request_body, response_body FROM "
+                    "request_history ORDER BY id DESC LIMIT 1"
+                ).fetchone()
+                conn.close()
+                if row and row[0] and "capture-me-marker" in row[0] \
+                        and row[1]:
+                    found = True
+                    break
+            except Exception:
+                pass
+            time.sleep(0.5)
+        assert found
+    finally:
+        gw.terminate()
+        mock.terminate()
+        for p in (gw, mock):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
