@@ -250,6 +250,8 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
   op->kind = kind;
   op->client_ip = c->peer_ip;
   op->request_id = c->request_id;
+  if (cfg.capture_bodies)
+    op->captured_request = std::string(body.substr(0, 16384));
   op->req_prompt_chars = (long long)body.size();
   op->tps = sel.tps;
   op->lease = RequestLease(sel.ep, &registry, sel.canonical, kind);
@@ -310,6 +312,8 @@ void start_proxy_direct(Gateway* gw, ClientConn* c, EndpointPtr ep,
   op->kind = kind;
   op->client_ip = c->peer_ip;
   op->request_id = c->request_id;
+  if (gw->cfg.capture_bodies)
+    op->captured_request = std::string(body.substr(0, 16384));
   op->public_model = raw_string_field(body, "model");
   op->upstream_model = op->public_model;
   op->lease = RequestLease(ep, &gw->registry, op->public_model, kind);
@@ -429,6 +433,8 @@ void Gateway::start_cloud_proxy(ClientConn* c, ApiKind kind,
   op->cloud_provider = provider;
   op->client_ip = c->peer_ip;
   op->request_id = c->request_id;
+  if (cfg.capture_bodies)
+    op->captured_request = std::string(body.substr(0, 16384));
   c->op = std::move(op);
   uc->op = c->op.get();
   metrics.cloud_requests.fetch_add(1);
