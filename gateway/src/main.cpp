@@ -47,6 +47,8 @@ int main(int argc, char** argv) {
   cfg.retention_days = envi("LLMLB_REQUEST_HISTORY_RETENTION_DAYS", 7);
   cfg.audit_archive_days = envi("LLMLB_AUDIT_RETENTION_DAYS", 90);
   cfg.log_retention_days = envi("LLMLB_LOG_RETENTION_DAYS", 7);
+  cfg.admission.round_robin =
+      envs("LLMLB_LOAD_BALANCER_MODE", "tps") == std::string("round_robin");
   cfg.capture_bodies = envi("LLMLB_CAPTURE_BODIES", 0) != 0;
   cfg.openai_base = envs("LLMLB_OPENAI_BASE_URL", "");
   cfg.google_base = envs("LLMLB_GOOGLE_BASE_URL", "");

@@ -386,7 +386,7 @@ SelectResult select_endpoint(const RouteTable& table,
         continue;
     }
     all_rejecting = false;
-    double score = route.tps[i]->score(now_ms);
+    double score = adm.round_robin ? 0.0 : route.tps[i]->score(now_ms);
     cands.push_back({i, score});
   }
   if (cands.empty()) {

@@ -280,6 +280,9 @@ struct AdmissionConfig {
   double delay_fraction = 0.5;
   double reject_fraction = 0.8;
   double hbm_admission_fraction = 0.92;    // of 288 GB HBM3E
+  // LLMLB_LOAD_BALANCER_MODE=round_robin ignores TPS scores (pure RR
+  // over admitting endpoints; reference config.rs load_balancer mode)
+  bool round_robin = false;
 };
 
 struct SelectResult {

@@ -268,3 +268,63 @@ def test_backend_tps_seeds_routing():
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def test_round_robin_mode_ignores_tps():
+    """LLMLB_LOAD_BALANCER_MODE=round_robin spreads traffic evenly even
+    when one backend advertises a huge TPS."""
+    ports = {k: _free_port() for k in ("a", "b", "gw")}
+    procs = [
+        subprocess.Popen([str(BIN / "llmlb-mock"), "--port",
+                          str(ports["a"]), "--workers", "1", "--models",
+                          "rrm", "--health-tps", "900"],
+                         stdout=subprocess.DEVNULL,
+                         stderr=subprocess.DEVNULL),
+        subprocess.Popen([str(BIN / "llmlb-mock"), "--port",
+                          str(ports["b"]), "--workers", "1", "--models",
+                          "rrm", "--health-tps", "2"],
+                         stdout=subprocess.DEVNULL,
+                         stderr=subprocess.DEVNULL),
+    ]
+    import os
+    db = tempfile.mktemp(suffix=".sqlite3")
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(ports["gw"]),
+                "LLMLB_DATABASE_URL": db, "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_LOAD_BALANCER_MODE": "round_robin",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "1"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    procs.append(gw)
+    base = f"http://127.0.0.1:{ports['gw']}"
+    admin = {"Authorization": "Bearer t"}
+    try:
+        for p in ("a", "b"):
+            assert _wait_port(ports[p])
+        assert _wait_port(ports["gw"])
+        for name in ("a", "b"):
+            status, _ = _req("POST", base + "/api/endpoints",
+                             {"name": name,
+                              "base_url":
+                                  f"http://127.0.0.1:{ports[name]}"},
+                             headers=admin)
+            assert status == 201
+        time.sleep(2.5)
+        for _ in range(12):
+            _req("POST", base + "/v1/chat/completions",
+                 {"model": "rrm",
+                  "messages": [{"role": "user", "content": "x"}]})
+        _, body = _req("GET", base + "/api/dashboard/summary",
+                       headers=admin)
+        eps = {e["name"]: e["total_requests"]
+               for e in json.loads(body)["endpoints"]}
+        assert abs(eps["a"] - eps["b"]) <= 2, eps
+    finally:
+        for p in procs:
+            p.terminate()
+        for p in procs:
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
