@@ -34,6 +34,8 @@ struct GatewayConfig {
   // (reference bootstrap.rs:291-318); 0 disables archiving
   int audit_archive_days = 90;
   int health_interval_s = 30;
+  double tps_ema_alpha = 0.2;
+  double amdsmi_poll_interval_s = 2.0;
   int health_timeout_ms = 5000;
   int auto_sync_interval_s = 900;
   int consecutive_failures_for_offline = 2;

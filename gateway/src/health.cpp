@@ -284,7 +284,8 @@ void Gateway::health_loop() {
   while (!stopping.load()) {
     int64_t loop_start = now_us();
     // amdsmi poll (MI355X-native GPU telemetry -> admission control)
-    if (now_us() - last_telemetry_us > 2 * 1000 * 1000) {
+    if (now_us() - last_telemetry_us >
+        (int64_t)(cfg.amdsmi_poll_interval_s * 1e6)) {
       telemetry_poll_once();
       last_telemetry_us = now_us();
     }

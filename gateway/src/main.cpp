@@ -49,6 +49,8 @@ int main(int argc, char** argv) {
   cfg.log_retention_days = envi("LLMLB_LOG_RETENTION_DAYS", 7);
   cfg.admission.round_robin =
       envs("LLMLB_LOAD_BALANCER_MODE", "tps") == std::string("round_robin");
+  cfg.tps_ema_alpha = envf("LLMLB_TPS_EMA_ALPHA", 0.2);
+  cfg.amdsmi_poll_interval_s = envf("LLMLB_AMDSMI_POLL_INTERVAL_SECS", 2.0);
   cfg.capture_bodies = envi("LLMLB_CAPTURE_BODIES", 0) != 0;
   cfg.openai_base = envs("LLMLB_OPENAI_BASE_URL", "");
   cfg.google_base = envs("LLMLB_GOOGLE_BASE_URL", "");

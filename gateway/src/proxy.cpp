@@ -669,7 +669,8 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
         ep->update_latency(dur_ms);
         if (completion_tokens > 0) {
           if (op->tps)
-            op->tps->update((double)completion_tokens, dur_ms / 1e3, 0.2,
+            op->tps->update((double)completion_tokens, dur_ms / 1e3,
+                            cfg.tps_ema_alpha,
                             wall_ms());
           else
             registry.record_tps(op->endpoint_id, op->public_model,
