@@ -281,3 +281,67 @@ def test_requests_csv_export(app_client, tmp_path, monkeypatch):
     assert lines[0].startswith("ts_ms,endpoint_id,model")
     assert any("m1" in ln and "chat_completions" in ln
                for ln in lines[1:])
+
+
+def test_dashboard_analytics_and_registry(tmp_path):
+    """clients / timeline / audit query / model-registry CRUD /
+    control health on a seeded DB."""
+    import llmlb_amd.api.control_plane as cp
+    import time as _t
+    db = str(tmp_path / "an.sqlite3")
+    conn = sqlite3.connect(db)
+    conn.executescript("""
+    CREATE TABLE users (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      username TEXT UNIQUE, password_hash TEXT, role TEXT DEFAULT 'user',
+      must_change_password INTEGER DEFAULT 0, created_at TEXT);
+    CREATE TABLE request_history (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      ts_ms INTEGER, endpoint_id TEXT, model TEXT, api_kind TEXT,
+      status INTEGER, duration_ms REAL, prompt_tokens INTEGER,
+      completion_tokens INTEGER, client_ip TEXT, error TEXT,
+      request_body TEXT, response_body TEXT, request_id TEXT);
+    CREATE TABLE audit_log (id INTEGER PRIMARY KEY AUTOINCREMENT,
+      ts_ms INTEGER, method TEXT, path TEXT, status INTEGER,
+      client_ip TEXT, user TEXT, duration_ms REAL,
+      prev_hash TEXT, hash TEXT);
+    CREATE TABLE models (id TEXT PRIMARY KEY, name TEXT NOT NULL,
+      source TEXT NOT NULL DEFAULT '', capabilities TEXT NOT NULL
+      DEFAULT 'chat', size_bytes INTEGER, created_at TEXT NOT NULL);
+    """)
+    now = int(_t.time() * 1000)
+    for i in range(5):
+        conn.execute(
+            "INSERT INTO request_history (ts_ms,endpoint_id,model,"
+            "api_kind,status,duration_ms,prompt_tokens,"
+            "completion_tokens,client_ip,error) VALUES "
+            "(?,?,?,?,?,?,?,?,?,?)",
+            (now - i * 60000, "e1", "m1", "chat_completions",
+             200 if i else 500, 10.0, 5, 7, f"10.0.0.{i % 2}", ""))
+    conn.execute("INSERT INTO audit_log (ts_ms,method,path,status,"
+                 "client_ip,user,duration_ms,prev_hash,hash) VALUES "
+                 "(?,?,?,?,?,?,?,?,?)",
+                 (now, "GET", "/x", 200, "10.0.0.1", "", 1.0, "p", "h"))
+    conn.commit()
+    conn.close()
+    app = cp.create_app(db_path=db, jwt_secret="s", admin_token="tk")
+    from fastapi.testclient import TestClient
+    c = TestClient(app)
+    hdr = {"Authorization": "Bearer tk"}
+    r = c.get("/api/dashboard/clients", headers=hdr)
+    assert r.status_code == 200
+    assert any(cl["client_ip"].startswith("10.0.0.")
+               for cl in r.json())
+    r = c.get("/api/dashboard/timeline?hours=2", headers=hdr)
+    assert r.status_code == 200 and r.json()
+    r = c.get("/api/audit?limit=10", headers=hdr)
+    assert r.status_code == 200 and len(r.json()) == 1
+    # model-registry CRUD (schema shared with the data plane's db.cpp)
+    r = c.post("/api/models", headers=hdr,
+               json={"id": "meta-llama/Meta-Llama-3-8B",
+                     "capabilities": "chat"})
+    assert r.status_code in (200, 201), r.text
+    r = c.get("/api/models", headers=hdr)
+    assert any(m["id"] == "meta-llama/Meta-Llama-3-8B"
+               for m in r.json())
+    r = c.delete("/api/models/meta-llama/Meta-Llama-3-8B", headers=hdr)
+    assert r.status_code == 200, r.text
+    assert c.get("/api/control/health").status_code == 200
