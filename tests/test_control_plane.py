@@ -345,3 +345,89 @@ def test_dashboard_analytics_and_registry(tmp_path):
     r = c.delete("/api/models/meta-llama/Meta-Llama-3-8B", headers=hdr)
     assert r.status_code == 200, r.text
     assert c.get("/api/control/health").status_code == 200
+
+
+def test_tps_benchmark_runner(tmp_path):
+    """POST /api/benchmarks/tps drives real requests through a live
+    gateway and records tokens/sec; GET returns the finished result."""
+    import socket
+    import subprocess
+    import sys
+    import tempfile
+    import llmlb_amd.api.control_plane as cp
+    from pathlib import Path
+    REPO = Path(__file__).resolve().parents[1]
+    BIN = REPO / "gateway" / "bin"
+
+    def free():
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        p = s.getsockname()[1]
+        s.close()
+        return p
+
+    import os
+    mock_port, gw_port = free(), free()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    mock = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "bm"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "bt",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    import urllib.request
+    import json as _json
+    import time as _t
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        deadline = _t.time() + 15
+        while _t.time() < deadline:
+            try:
+                urllib.request.urlopen(base + "/health", timeout=2)
+                break
+            except Exception:
+                _t.sleep(0.1)
+        req = urllib.request.Request(
+            base + "/api/endpoints",
+            data=_json.dumps({"name": "b",
+                              "base_url":
+                                  f"http://127.0.0.1:{mock_port}"
+                              }).encode(),
+            headers={"Authorization": "Bearer bt"})
+        urllib.request.urlopen(req, timeout=10)
+
+        os.environ["LLMLB_ADMIN_TOKEN"] = "bt"
+        app = cp.create_app(db_path=":memory:", jwt_secret="s",
+                            admin_token="bt", gateway_url=base)
+        from fastapi.testclient import TestClient
+        c = TestClient(app)
+        hdr = {"Authorization": "Bearer bt"}
+        r = c.post("/api/benchmarks/tps", headers=hdr,
+                   json={"model": "bm", "requests": 6,
+                         "concurrency": 2, "max_tokens": 16})
+        assert r.status_code == 202, r.text
+        bid = r.json()["benchmark_id"]
+        deadline = _t.time() + 30
+        res = None
+        while _t.time() < deadline:
+            res = c.get(f"/api/benchmarks/tps/{bid}",
+                        headers=hdr).json()
+            if res.get("status") == "done":
+                break
+            _t.sleep(0.3)
+        assert res and res["status"] == "done", res
+        assert res["errors"] == 0 and res["tokens_per_sec"] > 0
+        assert res["requests"] == 6 and res["source"] == "benchmark"
+    finally:
+        gw.terminate()
+        mock.terminate()
+        for p in (gw, mock):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
