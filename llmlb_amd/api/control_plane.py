@@ -387,6 +387,7 @@ def create_app(db_path: str | None = None,
                                    "model": body.model}
 
         async def run():
+          try:
             import aiohttp
             payload = {"model": body.model,
                        "messages": [{"role": "user",
@@ -427,6 +428,10 @@ def create_app(db_path: str | None = None,
                 "tokens_per_sec": round(tokens / dur, 2) if dur else 0,
                 "source": "benchmark",  # TpsSource::Benchmark
             }
+          except Exception as e:  # a crashed run must not stay
+            bench_results[bench_id] = {  # "running" forever
+                "status": "failed", "model": body.model,
+                "error": repr(e)}
 
         asyncio.get_event_loop().create_task(run())
         return {"benchmark_id": bench_id}

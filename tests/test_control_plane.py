@@ -401,28 +401,59 @@ def test_tps_benchmark_runner(tmp_path):
             headers={"Authorization": "Bearer bt"})
         urllib.request.urlopen(req, timeout=10)
 
-        os.environ["LLMLB_ADMIN_TOKEN"] = "bt"
-        app = cp.create_app(db_path=":memory:", jwt_secret="s",
-                            admin_token="bt", gateway_url=base)
-        from fastapi.testclient import TestClient
-        c = TestClient(app)
-        hdr = {"Authorization": "Bearer bt"}
-        r = c.post("/api/benchmarks/tps", headers=hdr,
-                   json={"model": "bm", "requests": 6,
-                         "concurrency": 2, "max_tokens": 16})
-        assert r.status_code == 202, r.text
-        bid = r.json()["benchmark_id"]
-        deadline = _t.time() + 30
-        res = None
-        while _t.time() < deadline:
-            res = c.get(f"/api/benchmarks/tps/{bid}",
-                        headers=hdr).json()
-            if res.get("status") == "done":
-                break
-            _t.sleep(0.3)
-        assert res and res["status"] == "done", res
-        assert res["errors"] == 0 and res["tokens_per_sec"] > 0
-        assert res["requests"] == 6 and res["source"] == "benchmark"
+        # real control-plane process: background benchmark tasks need
+        # a continuously-running event loop (TestClient suspends it)
+        cp_port = free()
+        cp_env = os.environ.copy()
+        cp_env.update({"LLMLB_DATABASE_URL": tempfile.mktemp(),
+                       "LLMLB_ADMIN_TOKEN": "bt",
+                       "LLMLB_JWT_SECRET": "s",
+                       "LLMLB_GATEWAY_URL": base})
+        cpp = subprocess.Popen(
+            [sys.executable, "-m", "llmlb_amd.api.control_plane",
+             "--port", str(cp_port)], env=cp_env, cwd=str(REPO),
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+        try:
+            cp_base = f"http://127.0.0.1:{cp_port}"
+            deadline = _t.time() + 30
+            while _t.time() < deadline:
+                try:
+                    urllib.request.urlopen(cp_base
+                                           + "/api/control/health",
+                                           timeout=2)
+                    break
+                except Exception:
+                    _t.sleep(0.3)
+            req = urllib.request.Request(
+                cp_base + "/api/benchmarks/tps",
+                data=_json.dumps({"model": "bm", "requests": 6,
+                                  "concurrency": 2,
+                                  "max_tokens": 16}).encode(),
+                headers={"Authorization": "Bearer bt",
+                         "Content-Type": "application/json"})
+            with urllib.request.urlopen(req, timeout=10) as r:
+                bid = _json.loads(r.read())["benchmark_id"]
+            deadline = _t.time() + 30
+            res = None
+            while _t.time() < deadline:
+                q = urllib.request.Request(
+                    cp_base + f"/api/benchmarks/tps/{bid}",
+                    headers={"Authorization": "Bearer bt"})
+                with urllib.request.urlopen(q, timeout=10) as r:
+                    res = _json.loads(r.read())
+                if res.get("status") != "running":
+                    break
+                _t.sleep(0.3)
+            assert res and res["status"] == "done", res
+            assert res["errors"] == 0 and res["tokens_per_sec"] > 0
+            assert res["requests"] == 6
+            assert res["source"] == "benchmark"
+        finally:
+            cpp.terminate()
+            try:
+                cpp.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                cpp.kill()
     finally:
         gw.terminate()
         mock.terminate()
