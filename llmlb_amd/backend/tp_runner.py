@@ -160,6 +160,12 @@ def main(argv=None):
         threading.Thread(target=usrv.run, daemon=True).start()
         print(f"llmlb_amd TP backend: model={served} tp={tp.world_size} "
               f"port={args.port}", flush=True)
+    if rank == 0:
+        # SIGTERM drains the whole group via the lockstep sentinel so
+        # every rank leaves its collective loop together
+        import signal as _sig
+        _sig.signal(_sig.SIGTERM, lambda *_: coord.stop())
+        _sig.signal(_sig.SIGINT, lambda *_: coord.stop())
     coord.loop()
 
 
