@@ -161,3 +161,20 @@ def test_fp8_weights_engine_gpu():
     # numerics are covered by test_fp8_weight_gemm_gpu.  Check sanity.
     assert all(0 <= t < cfg.vocab for t in outs[True])
     assert len(outs[False]) == 16
+
+
+def test_fp8_with_prefix_and_spec_cpu():
+    """Feature interplay on CPU: fp8 KV + prefix caching + speculative
+    decoding together stay self-consistent (repeatable greedy)."""
+    cfg = EngineConfig(model=LlamaConfig.tiny(), device="cpu",
+                       dtype="float32", kv_cache_blocks=96,
+                       kv_dtype="fp8_e4m3", spec_ngram=4, seed=9)
+    p = SamplingParams(temperature=0.0)
+    prompt = [3, 1, 3, 1, 3, 1] + list(range(20))
+    eng = Engine(cfg)
+    a = eng.generate_sync(prompt, 12, p).token_ids
+    b = eng.generate_sync(prompt, 12, p).token_ids   # prefix-cache hit
+    assert a == b
+    assert len(a) == 12
+    if eng.prefix:
+        assert eng.prefix.hit_tokens > 0
