@@ -178,3 +178,24 @@ def test_fp8_with_prefix_and_spec_cpu():
     assert len(a) == 12
     if eng.prefix:
         assert eng.prefix.hit_tokens > 0
+
+
+@pytest.mark.gpu
+def test_fp8_group8_engine_gpu():
+    """fp8 KV with GQA group 8 (the 70B shape): exercises the two-pass
+    v2 fp8 attention end to end."""
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.sampler import SamplingParams
+    from llmlb_amd.models.llama import LlamaConfig
+    cfg = LlamaConfig(hidden=2048, layers=2, q_heads=16, kv_heads=2,
+                      head_dim=128, inter=4096, vocab=8192,
+                      max_pos=1024, name="fp8-g8-test")
+    eng = Engine(EngineConfig(model=cfg, device="cuda:0",
+                              dtype="bfloat16", kv_cache_blocks=256,
+                              kv_dtype="fp8_e4m3", seed=6))
+    p = SamplingParams(temperature=0.0)
+    out = eng.generate_sync(list(range(40)), 12, p).token_ids
+    assert len(out) == 12
+    assert all(0 <= t < cfg.vocab for t in out)
+    out2 = eng.generate_sync(list(range(40)), 12, p).token_ids
+    assert out == out2
