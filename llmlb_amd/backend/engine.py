@@ -443,7 +443,7 @@ class Engine:
         if self.running:
             if (self.spec_ngram > 0
                     and all(s.params.temperature < 1e-5
-                            and not s.params.needs_history()
+                            and not s.params.needs_slow_path()
                             for s in self.running)):
                 self._spec_step()
             else:

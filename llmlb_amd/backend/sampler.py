@@ -23,11 +23,15 @@ class SamplingParams:
     stop_token_ids: tuple = ()
     seed: int | None = None
     logprobs: int = 0            # 0 = off, else top-k alternatives
+    logit_bias: tuple = ()       # ((token_id, bias), ...) -100..100
 
     def needs_history(self) -> bool:
         return (self.frequency_penalty != 0.0
                 or self.presence_penalty != 0.0
                 or self.repetition_penalty != 1.0)
+
+    def needs_slow_path(self) -> bool:
+        return self.needs_history() or bool(self.logit_bias)
 
 
 def _apply_penalties(lg: torch.Tensor, p: SamplingParams,
@@ -57,13 +61,13 @@ def sample(logits: torch.Tensor, params_list: list[SamplingParams],
     ``histories[i]`` = (prompt_ids, out_ids) — only consulted for
     sequences whose params request a penalty."""
     S = logits.shape[0]
-    if all(p.temperature < 1e-5 and not p.needs_history()
+    if all(p.temperature < 1e-5 and not p.needs_slow_path()
            for p in params_list):
         return logits.argmax(dim=-1)  # all-greedy fast path (no H2D)
     out = torch.empty(S, dtype=torch.long, device=logits.device)
     for i in range(S):
         p = params_list[i]
-        if p.temperature < 1e-5 and not p.needs_history():
+        if p.temperature < 1e-5 and not p.needs_slow_path():
             out[i] = logits[i].argmax(dim=-1)
             continue
         # clone: .float() aliases when logits are already fp32 and the
@@ -72,6 +76,9 @@ def sample(logits: torch.Tensor, params_list: list[SamplingParams],
         if p.needs_history() and histories is not None:
             prompt_ids, out_ids = histories[i]
             _apply_penalties(lg, p, out_ids, prompt_ids)
+        for tok_id, bias in p.logit_bias:
+            if 0 <= tok_id < lg.shape[-1]:
+                lg[tok_id] += bias
         if p.temperature < 1e-5:
             out[i] = lg.argmax(dim=-1)
             continue

@@ -164,6 +164,9 @@ class BackendServer:
                                or self.max_tokens_default),
             seed=body.get("seed"),
             logprobs=self._parse_logprobs(body),
+            logit_bias=tuple(
+                (int(k), max(-100.0, min(100.0, float(vv))))
+                for k, vv in (body.get("logit_bias") or {}).items()),
         )
 
     @staticmethod

@@ -218,3 +218,18 @@ def test_engine_reproducibility_across_instances():
     p2 = SamplingParams(temperature=0.8, seed=5)
     b = eng.generate_sync([4, 4, 4], 6, p2).token_ids
     assert a == b
+
+
+def test_logit_bias(tiny_engine):
+    """A +100 logit_bias forces the token; -100 suppresses it."""
+    eng = tiny_engine
+    p0 = SamplingParams(temperature=0.0, max_new_tokens=3)
+    base = eng.generate_sync([2, 3, 4], 3, p0).token_ids
+    p1 = SamplingParams(temperature=0.0, max_new_tokens=3,
+                        logit_bias=((7, 100.0),))
+    forced = eng.generate_sync([2, 3, 4], 3, p1).token_ids
+    assert forced == [7, 7, 7]
+    p2 = SamplingParams(temperature=0.0, max_new_tokens=1,
+                        logit_bias=((base[0], -100.0),))
+    avoided = eng.generate_sync([2, 3, 4], 1, p2).token_ids
+    assert avoided[0] != base[0]
