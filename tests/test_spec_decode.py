@@ -1,6 +1,7 @@
 """Prompt-lookup speculative decoding: outputs must be IDENTICAL to
 plain greedy decoding (verification is exact for greedy), and the
 n-gram drafter must actually accept tokens on repetitive sequences."""
+import pytest
 import torch
 
 from llmlb_amd.backend.engine import Engine, EngineConfig
@@ -77,3 +78,24 @@ def test_spec_batched_mixed_drafts():
     assert len(outs["a"]) == 12 and len(outs["b"]) == 12
     assert outs["a"] == _greedy(_engine(spec=0), [7, 3, 7, 3, 7, 3], 12)
     assert outs["b"] == _greedy(_engine(spec=0), list(range(30)), 12)
+
+
+@pytest.mark.gpu
+def test_spec_matches_plain_greedy_gpu():
+    """Speculative verify on GPU (gathered K/V + masked SDPA + real
+    kernels) is output-identical to plain greedy decoding."""
+    import torch
+    cfg = LlamaConfig(hidden=1024, layers=2, q_heads=8, kv_heads=2,
+                      head_dim=128, inter=2048, vocab=4096,
+                      max_pos=512, name="spec-gpu")
+    outs = {}
+    for spec in (0, 4):
+        eng = Engine(EngineConfig(model=cfg, device="cuda:0",
+                                  dtype="bfloat16", kv_cache_blocks=128,
+                                  spec_ngram=spec, seed=13))
+        p = SamplingParams(temperature=0.0)
+        outs[spec] = eng.generate_sync([5, 6, 5, 6, 5, 6, 5, 6], 20,
+                                       p).token_ids
+        del eng
+        torch.cuda.empty_cache()
+    assert outs[4] == outs[0]
