@@ -16,6 +16,14 @@ a process group over RCCL/xGMI).  Each ``step()``:
 Prefill attention uses torch SDPA on the prompt's own K/V (flash path
 on ROCm); decode attention is the hand-written gfx950 paged kernel
 (llmlb_amd/ops/csrc/kernels.hip).
+
+Optional serving features (all measured in profiles/):
+  * automatic prefix caching (kv_cache.PrefixCache) — default on,
+  * chunked prefill for long prompts (prefill_chunk),
+  * prompt-lookup speculative decoding (spec_ngram / LLMLB_SPEC_NGRAM),
+  * FP8 e4m3 KV cache (kv_dtype="fp8_e4m3"),
+  * HF-safetensors checkpoint loading (weights_path),
+  * request cancellation (abort()).
 """
 from __future__ import annotations
 
