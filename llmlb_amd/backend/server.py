@@ -603,6 +603,8 @@ def build_asgi_app(server: BackendServer):
                                           if want_lp else None),
                              "finish_reason": finish})
                     else:
+                        if body.get("echo"):
+                            text = str(body.get("prompt", "")) + text
                         choices.append({"index": ci, "text": text,
                                         "finish_reason": finish})
                 watcher.cancel()

@@ -262,3 +262,14 @@ def test_responses_instructions_and_model_detail(stack):
             stack["backend"] + "/v1/models/llama-tiny", timeout=10) as m:
         detail = json.loads(m.read())
     assert detail["id"] == "llama-tiny"
+
+
+def test_completions_echo(stack):
+    body = json.dumps({"model": "llama-tiny", "prompt": "abc def",
+                       "max_tokens": 3, "echo": True}).encode()
+    r = urllib.request.Request(
+        stack["backend"] + "/v1/completions", data=body,
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(r, timeout=30) as resp:
+        out = json.loads(resp.read())
+    assert out["choices"][0]["text"].startswith("abc def")
