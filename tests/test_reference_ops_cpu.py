@@ -1,0 +1,73 @@
+"""The fp32 reference ops themselves (llmlb_amd/ops/reference.py) vs
+straightforward formula implementations — these are the ground truth
+the GPU kernels are tested against, so they get their own checks."""
+import math
+
+import torch
+
+from llmlb_amd.ops import reference as R
+
+
+def test_rmsnorm_formula():
+    x = torch.randn(4, 64)
+    w = torch.randn(64)
+    out, _ = R.rmsnorm(x, w, 1e-5, None)
+    manual = x / torch.sqrt((x * x).mean(-1, keepdim=True) + 1e-5) * w
+    assert torch.allclose(out, manual, atol=1e-5)
+
+
+def test_rmsnorm_fused_residual_semantics():
+    x = torch.randn(3, 32)
+    res = torch.randn(3, 32)
+    w = torch.ones(32)
+    out, new_res = R.rmsnorm(x, w, 1e-6, res)
+    assert torch.allclose(new_res, x + res, atol=1e-6)
+    manual = new_res / torch.sqrt(
+        (new_res * new_res).mean(-1, keepdim=True) + 1e-6)
+    assert torch.allclose(out, manual, atol=1e-5)
+
+
+def test_swiglu_formula():
+    gu = torch.randn(5, 16)
+    out = R.swiglu(gu)
+    g, u = gu[:, :8], gu[:, 8:]
+    manual = g * torch.sigmoid(g) * u
+    assert torch.allclose(out, manual, atol=1e-5)
+
+
+def test_rope_is_norm_preserving_rotation():
+    q = torch.randn(6, 2, 32)
+    k = torch.randn(6, 1, 32)
+    cs = R.build_cos_sin_table(64, 32, 1e4)
+    pos = torch.arange(6)
+    q2, k2 = R.rope_apply(q.clone(), k.clone(), cs, pos)
+    # a rotation preserves pairwise norms
+    assert torch.allclose(q2.norm(dim=-1), q.norm(dim=-1), atol=1e-4)
+    assert torch.allclose(k2.norm(dim=-1), k.norm(dim=-1), atol=1e-4)
+    # position 0 is the identity
+    q0, _ = R.rope_apply(q[:1].clone(), k[:1].clone(), cs,
+                         torch.zeros(1, dtype=torch.long))
+    assert torch.allclose(q0, q[:1], atol=1e-5)
+
+
+def test_paged_attn_reference_matches_sdpa():
+    torch.manual_seed(0)
+    S, nq, nkv, D, BS, nb = 2, 4, 2, 32, 16, 6
+    kc = torch.randn(nb, nkv, BS, D)
+    vc = torch.randn(nb, nkv, BS, D)
+    q = torch.randn(S, nq, D)
+    bt = torch.tensor([[1, 2], [3, 4]], dtype=torch.int32)
+    sl = torch.tensor([20, 30], dtype=torch.int32)
+    out = R.paged_attn_decode(q, kc, vc, bt, sl, D ** -0.5)
+    for s in range(S):
+        L = int(sl[s])
+        ks = torch.cat([kc[bt[s, i]].transpose(0, 1)
+                        for i in range((L + BS - 1) // BS)], 0)[:L]
+        vs = torch.cat([vc[bt[s, i]].transpose(0, 1)
+                        for i in range((L + BS - 1) // BS)], 0)[:L]
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q[s].unsqueeze(0).transpose(0, 1),
+            ks.transpose(0, 1).repeat_interleave(nq // nkv, 0),
+            vs.transpose(0, 1).repeat_interleave(nq // nkv, 0),
+        ).transpose(0, 1).squeeze(0)
+        assert torch.allclose(out[s], ref, atol=1e-4), s
