@@ -742,3 +742,22 @@ def test_cloud_prefix_google(stack):
     assert status == 502
     assert "google" in body.lower()
     assert "LLMLB_GOOGLE_BASE_URL".lower() in body.lower()
+
+
+def test_anthropic_unicode_roundtrip(stack):
+    """Multibyte content survives request conversion + response
+    re-encode (the mock echoes a fixed reply; the conversion path must
+    not mangle or reject the UTF-8 request)."""
+    body = {"model": "mock-model", "max_tokens": 16,
+            "messages": [{"role": "user",
+                          "content": "héllo 世界 🚀 \"quoted\" \\ backslash"}]}
+    r = urllib.request.Request(
+        stack["base"] + "/v1/messages",
+        data=json.dumps(body, ensure_ascii=False).encode("utf-8"),
+        headers={**stack["auth"],
+                 "anthropic-version": "2023-06-01"})
+    with urllib.request.urlopen(r, timeout=10) as resp:
+        out = json.loads(resp.read())
+    assert out["type"] == "message"
+    assert out["content"][0]["type"] == "text"
+    assert out["usage"]["input_tokens"] > 0
