@@ -128,8 +128,11 @@ def build_app(kind: str, model_name: str):
             return
         path = scope["path"]
         if path == "/v1/models":
+            caps = {"asr": "audio_transcription", "tts": "audio_speech",
+                    "image": "image_generation"}[kind]
             body = json.dumps({"object": "list", "data": [
-                {"id": m, "object": "model", "owned_by": "llmlb_amd"}
+                {"id": m, "object": "model", "owned_by": "llmlb_amd",
+                 "capabilities": caps}
                 for m in served]}).encode()
             await send_bytes(send, 200, "application/json", body)
             return
