@@ -13,7 +13,8 @@ void start_proxy_direct(Gateway* gw, ClientConn* c, EndpointPtr ep,
 // telemetry.cpp
 void telemetry_snapshot_json(std::string* out);
 
-static Json endpoint_json(const EndpointPtr& ep) {
+static Json endpoint_json(const EndpointPtr& ep,
+                          Registry* reg = nullptr) {
   JsonObject o;
   o.emplace_back("id", Json(ep->id));
   o.emplace_back("name", Json(ep->name));
@@ -53,6 +54,12 @@ static Json endpoint_json(const EndpointPtr& ep) {
     mo.emplace_back("canonical_id", Json(m.canonical));
     mo.emplace_back("capabilities", Json(caps_to_csv(m.caps)));
     if (m.max_tokens) mo.emplace_back("max_tokens", Json(m.max_tokens));
+    if (reg) {
+      auto st = reg->tps_state(ep->id, m.canonical, ApiKind::kChat);
+      std::lock_guard<std::mutex> lk(st->mu);
+      if (st->samples > 0)
+        mo.emplace_back("tokens_per_sec", Json(st->ema));
+    }
     models.emplace_back(Json(std::move(mo)));
   }
   o.emplace_back("models", Json(std::move(models)));
@@ -92,7 +99,7 @@ std::string Gateway::render_summary_json() {
   long long active = 0;
   for (auto& ep : eps) {
     active += ep->active.load();
-    earr.emplace_back(endpoint_json(ep));
+    earr.emplace_back(endpoint_json(ep, &registry));
   }
   auto sum60 = history.ring().summarize(60 * 60 * 1000);
   JsonObject o;
@@ -246,7 +253,8 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
     // -- endpoints CRUD (api/endpoints.rs) --
     if (path == "/api/endpoints" && method == "GET") {
       JsonArray arr;
-      for (auto& ep : registry.list()) arr.emplace_back(endpoint_json(ep));
+      for (auto& ep : registry.list())
+        arr.emplace_back(endpoint_json(ep, &registry));
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }

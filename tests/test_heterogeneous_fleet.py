@@ -260,6 +260,9 @@ def test_backend_tps_seeds_routing():
         eps = {e["name"]: e for e in json.loads(body)["endpoints"]}
         assert eps["fast"]["total_requests"] > \
             eps["slow"]["total_requests"], eps
+        # per-model TPS (seeded) is surfaced in the endpoint JSON
+        assert any(m.get("tokens_per_sec", 0) > 0
+                   for m in eps["fast"]["models"]), eps["fast"]
     finally:
         for p in procs:
             p.terminate()
