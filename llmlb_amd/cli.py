@@ -213,6 +213,8 @@ def cmd_bench(args) -> int:
 
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser(prog="llmlb")
+    ap.add_argument("--version", action="version",
+                    version="llmlb-amd 0.1.0 (MI355X-native)")
     sub = ap.add_subparsers(dest="cmd", required=True)
 
     sp = sub.add_parser("serve", help="start gateway + control plane")
