@@ -131,6 +131,13 @@ def test_serve_with_backend_chat_e2e(tmp_path):
             except Exception:
                 time.sleep(1.0)
         assert ok, "chat through the CLI-supervised node never succeeded"
+        # assistant helper: allowed local URL round-trips
+        res = subprocess.run(
+            [sys.executable, "-m", "llmlb_amd.cli", "assistant",
+             f"http://127.0.0.1:{port}/health"], cwd=str(REPO), env=env,
+            capture_output=True, text=True, timeout=20)
+        assert res.returncode == 0, res.stderr
+        assert "ok" in res.stdout or "status" in res.stdout
     finally:
         proc.send_signal(signal.SIGTERM)
         try:
