@@ -308,7 +308,8 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
         return;
       }
       if (action.empty() && method == "GET") {
-        c->send_simple(200, "application/json", endpoint_json(ep).dump());
+        c->send_simple(200, "application/json",
+                       endpoint_json(ep, &registry).dump());
         return;
       }
       if (action.empty() && method == "DELETE") {
