@@ -727,6 +727,13 @@ class Engine:
         seqs = self.running
         B = len(seqs)
         drafts = [self._draft_for(s) for s in seqs]
+        # never let the fed window (last token + drafts) write KV past
+        # the context window / per-seq page capacity
+        cap = self.max_blocks_per_seq * BLOCK_SIZE
+        for i, s in enumerate(seqs):
+            room = min(self.cfg.model.max_pos, cap) - s.length - 1
+            if len(drafts[i]) > max(room, 0):
+                drafts[i] = drafts[i][:max(room, 0)]
         F = 1 + max((len(d) for d in drafts), default=0)
         if F == 1:
             self._decode_step()

@@ -99,3 +99,20 @@ def test_spec_matches_plain_greedy_gpu():
         del eng
         torch.cuda.empty_cache()
     assert outs[4] == outs[0]
+
+
+def test_spec_near_context_window_end():
+    """Drafts are clamped so speculative feeds never write KV past the
+    context window (was an overrun at max_pos - spec_ngram)."""
+    cfg = LlamaConfig(hidden=128, layers=2, q_heads=4, kv_heads=2,
+                      head_dim=32, inter=256, vocab=256, max_pos=64,
+                      name="spec-edge")
+    eng = Engine(EngineConfig(model=cfg, device="cpu", dtype="float32",
+                              kv_cache_blocks=32, spec_ngram=6, seed=2))
+    # repetitive prompt near the window: generation runs to the clamp
+    prompt = [5, 6] * 20                      # 40 tokens, window 64
+    p = SamplingParams(temperature=0.0, max_new_tokens=200)
+    res = eng.generate_sync(prompt, 200, p)
+    assert res.finish_reason in ("length", "stop")
+    assert len(res.token_ids) <= 64 - 40
+    assert eng.kv.free_blocks() >= 0
