@@ -8,6 +8,7 @@
 #include "http.h"
 #include "json.h"
 #include "router.h"
+#include "anthropic.h"
 #include "db.h"
 #include "history.h"
 #include "sha256.h"
@@ -295,6 +296,42 @@ static void test_audit_archive() {
   remove(arch.c_str());
 }
 
+static void test_anthropic_stream_encoder_fragmented() {
+  // OpenAI SSE -> Anthropic SSE re-encode must produce the same event
+  // sequence whether fed whole or byte-at-a-time
+  const std::string in =
+      "data: {\"choices\":[{\"delta\":{\"role\":\"assistant\"}}]}\n\n"
+      "data: {\"choices\":[{\"delta\":{\"content\":\"Hel\"}}]}\n\n"
+      "data: {\"choices\":[{\"delta\":{\"content\":\"lo\"}}]}\n\n"
+      "data: {\"choices\":[{\"delta\":{},\"finish_reason\":\"stop\"}]"
+      ",\"usage\":{\"prompt_tokens\":3,\"completion_tokens\":2}}\n\n"
+      "data: [DONE]\n\n";
+  std::string whole, frag;
+  {
+    AnthropicStreamEncoder enc("m");
+    enc.feed(in, whole);
+    enc.finish(whole);
+  }
+  {
+    AnthropicStreamEncoder enc("m");
+    for (char ch : in) enc.feed(std::string_view(&ch, 1), frag);
+    enc.finish(frag);
+  }
+  // normalise the per-message random id before comparing
+  auto strip_id = [](std::string s) {
+    size_t p;
+    while ((p = s.find("msg_")) != std::string::npos)
+      s.replace(p, 4 + 36, "msg_X");
+    return s;
+  };
+  CHECK(strip_id(whole) == strip_id(frag));
+  CHECK(whole.find("message_start") != std::string::npos);
+  CHECK(whole.find("content_block_delta") != std::string::npos);
+  CHECK(whole.find("Hel") != std::string::npos);
+  CHECK(whole.find("message_stop") != std::string::npos);
+  CHECK(whole.find("end_turn") != std::string::npos);
+}
+
 static void test_caps_csv() {
   CHECK(caps_from_csv("chat,embeddings") == (kCapChat | kCapEmbeddings));
   CHECK(caps_to_csv(kCapAudioSpeech) == "audio_speech");
@@ -312,6 +349,7 @@ int main() {
   test_tps_and_selection();
   test_lease();
   test_audit_archive();
+  test_anthropic_stream_encoder_fragmented();
   test_caps_csv();
   if (g_failures == 0) {
     printf("llmlb-test: all native unit tests passed\n");
