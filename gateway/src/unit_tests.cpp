@@ -319,9 +319,11 @@ static void test_anthropic_stream_encoder_fragmented() {
   }
   // normalise the per-message random id before comparing
   auto strip_id = [](std::string s) {
-    size_t p;
-    while ((p = s.find("msg_")) != std::string::npos)
+    size_t p = 0;
+    while ((p = s.find("msg_", p)) != std::string::npos) {
       s.replace(p, 4 + 36, "msg_X");
+      p += 5;
+    }
     return s;
   };
   CHECK(strip_id(whole) == strip_id(frag));
