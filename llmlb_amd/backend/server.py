@@ -236,7 +236,10 @@ async def handle_responses(server: "BackendServer", body: dict, send):
             self.q: _asyncio.Queue = _asyncio.Queue()
 
         def put(self, item):
-            self.loop.call_soon_threadsafe(self.q.put_nowait, item)
+            try:
+                self.loop.call_soon_threadsafe(self.q.put_nowait, item)
+            except RuntimeError:
+                pass  # loop closed during shutdown
 
     sink = _Sink()
     server.engine.submit(prompt_ids, params, sink=sink)
@@ -360,7 +363,12 @@ def build_asgi_app(server: BackendServer):
                 if self._scheduled:
                     return
                 self._scheduled = True
-            self.loop.call_soon_threadsafe(self._drain)
+            try:
+                self.loop.call_soon_threadsafe(self._drain)
+            except RuntimeError:
+                # event loop closed (server shutting down): the engine
+                # thread must not die over an orphaned stream
+                pass
 
         def _drain(self):
             with self._mu:
