@@ -235,8 +235,27 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
     return;
   }
 
-  // ---- admin API (auth: admin token or admin-permission key) ------
+  // ---- admin API: admin token, or a key holding the route's scope
+  // (reference per-group middleware matrix, api/mod.rs:70-635) --------
   if (starts_with(path, "/api/")) {
+    // route -> required scope (read vs manage split per method)
+    const bool is_read = method == "GET" || method == "HEAD";
+    const char* scope = "admin";
+    if (starts_with(path, "/api/endpoints"))
+      scope = is_read ? "endpoints.read" : "endpoints.manage";
+    else if (starts_with(path, "/api/api-keys"))
+      scope = "api_keys.manage";
+    else if (starts_with(path, "/api/models/registry"))
+      scope = "registry.read";
+    else if (starts_with(path, "/api/models"))
+      scope = is_read ? "registry.read" : "models.manage";
+    else if (starts_with(path, "/api/dashboard") ||
+             path == "/api/requests" || path == "/api/events" ||
+             path == "/api/gpu")
+      scope = "metrics.read";
+    else if (starts_with(path, "/api/audit") ||
+             starts_with(path, "/api/logs"))
+      scope = "logs.read";
     std::string_view authz = r.header("authorization");
     bool admin_ok = false;
     if (!cfg.admin_token.empty()) {
@@ -244,7 +263,9 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       if (starts_with(key, "Bearer ")) key.remove_prefix(7);
       if (trim(key) == cfg.admin_token) admin_ok = true;
     }
-    if (!admin_ok) admin_ok = auth.check_bearer(authz, "admin");
+    if (!admin_ok)
+      admin_ok = auth.check_bearer(authz, scope,
+                                   is_loopback_ip(c->peer_ip));
     if (!admin_ok) {
       c->send_json_error(401, "unauthorized", "admin credentials required");
       return;
@@ -382,6 +403,11 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       std::string perms = in["permissions"].as_str();
       if (perms.empty()) perms = "inference";
       std::string key = auth.create_key(name.empty() ? "key" : name, perms);
+      if (key.empty()) {
+        c->send_json_error(500, "entropy_unavailable",
+                           "could not generate key material");
+        return;
+      }
       JsonObject o;
       o.emplace_back("api_key", Json(key));
       o.emplace_back("name", Json(name));

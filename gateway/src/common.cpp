@@ -1,8 +1,10 @@
 #include "common.h"
 
+#include <errno.h>
 #include <fcntl.h>
 #include <dirent.h>
 #include <sys/stat.h>
+#include <sys/syscall.h>
 #include <sys/types.h>
 #include <unistd.h>
 
@@ -152,9 +154,49 @@ std::string url_decode(std::string_view s) {
   return out;
 }
 
+// OS entropy (getrandom(2), /dev/urandom fallback).  API keys and the
+// uuid4 seed come from here: the reference generates token material
+// with a CSPRNG (rand::rng ThreadRng), and mt19937 output alone is
+// predictable/recoverable.
+bool secure_random_bytes(void* buf, size_t n) {
+#ifdef SYS_getrandom
+  size_t off = 0;
+  while (off < n) {
+    long r = syscall(SYS_getrandom, (char*)buf + off, n - off, 0);
+    if (r > 0) { off += (size_t)r; continue; }
+    if (r < 0 && errno == EINTR) continue;
+    break;
+  }
+  if (off == n) return true;
+#endif
+  int fd = open("/dev/urandom", O_RDONLY | O_CLOEXEC);
+  if (fd < 0) return false;
+  size_t off2 = 0;
+  while (off2 < n) {
+    ssize_t r = read(fd, (char*)buf + off2, n - off2);
+    if (r <= 0) { if (r < 0 && errno == EINTR) continue; break; }
+    off2 += (size_t)r;
+  }
+  close(fd);
+  return off2 == n;
+}
+
+bool is_loopback_ip(std::string_view ip) {
+  if (ip == "::1" || ip == "localhost") return true;
+  if (starts_with(ip, "::ffff:")) ip.remove_prefix(7);
+  return starts_with(ip, "127.");
+}
+
 std::string uuid4() {
-  static thread_local std::mt19937_64 rng(
-      std::random_device{}() ^ (uint64_t)(uintptr_t)&rng);
+  // ids only (request/message ids) — key material uses
+  // secure_random_bytes directly.  Seeded from OS entropy so even id
+  // streams are not recoverable from a weak seed.
+  static thread_local std::mt19937_64 rng([] {
+    uint64_t seed[2] = {0, 0};
+    if (!secure_random_bytes(seed, sizeof seed))
+      seed[0] = std::random_device{}() ^ (uint64_t)(uintptr_t)&seed;
+    return seed[0] ^ (seed[1] << 1);
+  }());
   uint64_t a = rng(), b = rng();
   a = (a & 0xffffffffffff0fffULL) | 0x4000ULL;         // version 4
   b = (b & 0x3fffffffffffffffULL) | 0x8000000000000000ULL;  // variant

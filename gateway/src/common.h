@@ -138,6 +138,11 @@ std::string url_decode(std::string_view s);
 
 // uuid v4 (for endpoint / request ids)
 std::string uuid4();
+// fills buf with n bytes of OS entropy (getrandom/urandom); false on
+// failure.  Use for all key/token material.
+bool secure_random_bytes(void* buf, size_t n);
+// true for 127.0.0.0/8, ::1 and their IPv4-mapped forms
+bool is_loopback_ip(std::string_view ip);
 
 // base64 / base64url
 std::string b64_encode(std::string_view in, bool url_safe = false,

@@ -21,35 +21,66 @@ void AuthStore::load() {
     hash_to_perms_[sha256_hex(env_key)] = "inference,admin";
   open_mode_ = hash_to_perms_.empty();
   if (open_mode_)
-    LOG_WARN("no API keys configured: /v1/* runs in open mode "
-             "(create one via POST /api/api-keys)");
+    LOG_WARN("no API keys configured: /v1/* runs in open mode for "
+             "LOOPBACK clients only (create a key via POST /api/api-keys "
+             "or set LLMLB_API_KEY to serve remote clients)");
 }
 
-bool AuthStore::check_key(std::string_view key, const char* perm) {
-  if (open_mode_) return true;
+// exact comma-separated token matching; "admin" grants everything, the
+// legacy "inference" token covers the two openai.* read/infer scopes.
+bool AuthStore::perms_grant(std::string_view perm_list, const char* perm) {
+  std::string_view want(perm);
+  size_t pos = 0;
+  while (pos <= perm_list.size()) {
+    size_t comma = perm_list.find(',', pos);
+    std::string_view tok = perm_list.substr(
+        pos, comma == std::string_view::npos ? std::string_view::npos
+                                             : comma - pos);
+    tok = trim(tok);
+    if (!tok.empty()) {
+      if (tok == want || tok == "admin") return true;
+      if (tok == "inference" &&
+          (want == "openai.inference" || want == "openai.models.read"))
+        return true;
+    }
+    if (comma == std::string_view::npos) break;
+    pos = comma + 1;
+  }
+  return false;
+}
+
+bool AuthStore::check_key(std::string_view key, const char* perm,
+                          bool client_is_local) {
+  if (open_mode_) return client_is_local;
   std::string h = sha256_hex(key);
   std::lock_guard<std::mutex> lk(mu_);
   auto it = hash_to_perms_.find(h);
   if (it == hash_to_perms_.end()) return false;
-  return it->second.find(perm) != std::string::npos ||
-         it->second.find("admin") != std::string::npos;
+  return perms_grant(it->second, perm);
 }
 
 bool AuthStore::check_bearer(std::string_view authorization,
-                             const char* perm) {
-  if (open_mode_) return true;
+                             const char* perm, bool client_is_local) {
+  if (open_mode_) return client_is_local;
   std::string_view key = authorization;
   if (starts_with(key, "Bearer ") || starts_with(key, "bearer "))
     key.remove_prefix(7);
   key = trim(key);
   if (key.empty()) return false;
-  return check_key(key, perm);
+  return check_key(key, perm, client_is_local);
 }
 
 std::string AuthStore::create_key(const std::string& name,
                                   const std::string& perms) {
-  std::string key = "sk-" + b64_encode(uuid4() + uuid4(), true, false)
-                                .substr(0, 43);
+  unsigned char raw[32];
+  if (!secure_random_bytes(raw, sizeof raw)) {
+    LOG_ERROR("secure_random_bytes failed; refusing to mint an API key");
+    return "";
+  }
+  std::string key =
+      "sk-" + b64_encode(std::string_view((char*)raw, sizeof raw), true,
+                         false)
+                  .substr(0, 43);
   std::string h = sha256_hex(key);
   {
     std::lock_guard<std::mutex> lk(mu_);

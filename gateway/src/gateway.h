@@ -103,24 +103,38 @@ struct Metrics {
 };
 
 // auth store: api-key sha256 lookup (auth/middleware.rs analog)
+// Per-key permission matrix mirroring the reference's 11 scopes
+// (common/auth.rs:60-95: openai.inference, openai.models.read,
+// endpoints.read, endpoints.manage, api_keys.manage, users.manage,
+// invitations.manage, models.manage, registry.read, logs.read,
+// metrics.read) plus "admin" (grants all) and the legacy "inference"
+// (grants openai.inference + openai.models.read).  Stored perms are a
+// comma-separated token list compared EXACTLY per token — substring
+// grants ("badminton" ⊃ "admin") are rejected.
 class AuthStore {
  public:
   explicit AuthStore(Db* db) : db_(db) {}
   void load();
-  // returns true when authorized; perm: "inference" | "admin"
-  bool check_bearer(std::string_view authorization, const char* perm);
-  bool check_key(std::string_view key, const char* perm);
+  // returns true when authorized; perm = one scope name above.
+  // client_is_local gates open mode (no keys configured): an exposed
+  // gateway without its first key only serves loopback clients.
+  bool check_bearer(std::string_view authorization, const char* perm,
+                    bool client_is_local = false);
+  bool check_key(std::string_view key, const char* perm,
+                 bool client_is_local = false);
   std::string create_key(const std::string& name, const std::string& perms);
   bool revoke_key(const std::string& prefix_or_id);
   bool any_keys();
   void set_open_mode(bool open) { open_mode_ = open; }
   bool open_mode() const { return open_mode_; }
+  // exact-token permission-set test (public for unit tests)
+  static bool perms_grant(std::string_view perm_list, const char* perm);
 
  private:
   Db* db_;
   std::mutex mu_;
   std::unordered_map<std::string, std::string> hash_to_perms_;
-  bool open_mode_ = false;  // no keys configured -> allow (logged)
+  bool open_mode_ = false;  // no keys configured -> loopback-only allow
 };
 
 // ---------------------------------------------------------------------

@@ -127,12 +127,19 @@ void Gateway::handle_request(ClientConn* c) {
       return;
     }
     // auth: Bearer key (x-api-key also accepted on /v1/messages,
-    // mirroring the Anthropic-native surface)
+    // mirroring the Anthropic-native surface).  Scope per route:
+    // /v1/models needs openai.models.read, everything else
+    // openai.inference (reference permission matrix auth.rs:60-95).
+    const bool local_client = is_loopback_ip(c->peer_ip);
+    const char* scope = (path == "/v1/models" ||
+                         starts_with(path, "/v1/models/"))
+                            ? "openai.models.read"
+                            : "openai.inference";
     std::string_view authz = r.header("authorization");
-    bool authed = auth.check_bearer(authz, "inference");
+    bool authed = auth.check_bearer(authz, scope, local_client);
     if (!authed && path == "/v1/messages") {
       std::string_view xkey = r.header("x-api-key");
-      authed = !xkey.empty() && auth.check_key(xkey, "inference");
+      authed = !xkey.empty() && auth.check_key(xkey, scope, local_client);
     }
     if (!authed) {
       c->send_json_error(401, "invalid_api_key",

@@ -7,6 +7,7 @@
 
 #include "http.h"
 #include "json.h"
+#include "gateway.h"
 #include "router.h"
 #include "anthropic.h"
 #include "db.h"
@@ -334,6 +335,46 @@ static void test_anthropic_stream_encoder_fragmented() {
   CHECK(whole.find("end_turn") != std::string::npos);
 }
 
+static void test_permission_matrix() {
+  // exact token matching: no substring grants
+  CHECK(!AuthStore::perms_grant("badminton", "admin"));
+  CHECK(!AuthStore::perms_grant("inference-disabled", "openai.inference"));
+  CHECK(!AuthStore::perms_grant("inferenc", "openai.inference"));
+  CHECK(AuthStore::perms_grant("admin", "users.manage"));
+  CHECK(AuthStore::perms_grant("a,admin,b", "logs.read"));
+  // legacy "inference" covers the two openai scopes only
+  CHECK(AuthStore::perms_grant("inference", "openai.inference"));
+  CHECK(AuthStore::perms_grant("inference", "openai.models.read"));
+  CHECK(!AuthStore::perms_grant("inference", "endpoints.manage"));
+  // scoped keys: exact scope + no cross-grant (the 11-scope matrix)
+  const char* scopes[] = {"openai.inference",  "openai.models.read",
+                          "endpoints.read",    "endpoints.manage",
+                          "api_keys.manage",   "users.manage",
+                          "invitations.manage","models.manage",
+                          "registry.read",     "logs.read",
+                          "metrics.read"};
+  for (const char* a : scopes)
+    for (const char* b : scopes)
+      CHECK(AuthStore::perms_grant(a, b) == (strcmp(a, b) == 0));
+  // whitespace + multi-token lists
+  CHECK(AuthStore::perms_grant(" endpoints.read , metrics.read ",
+                               "metrics.read"));
+  CHECK(!AuthStore::perms_grant("", "openai.inference"));
+  // loopback helper
+  CHECK(is_loopback_ip("127.0.0.1"));
+  CHECK(is_loopback_ip("::1"));
+  CHECK(is_loopback_ip("::ffff:127.0.0.1"));
+  CHECK(!is_loopback_ip("10.0.0.5"));
+  CHECK(!is_loopback_ip("1270.0.0.1"));
+}
+
+static void test_secure_random() {
+  unsigned char a[32] = {0}, b[32] = {0};
+  CHECK(secure_random_bytes(a, sizeof a));
+  CHECK(secure_random_bytes(b, sizeof b));
+  CHECK(memcmp(a, b, sizeof a) != 0);  // astronomically unlikely equal
+}
+
 static void test_caps_csv() {
   CHECK(caps_from_csv("chat,embeddings") == (kCapChat | kCapEmbeddings));
   CHECK(caps_to_csv(kCapAudioSpeech) == "audio_speech");
@@ -352,6 +393,8 @@ int main() {
   test_lease();
   test_audit_archive();
   test_anthropic_stream_encoder_fragmented();
+  test_permission_matrix();
+  test_secure_random();
   test_caps_csv();
   if (g_failures == 0) {
     printf("llmlb-test: all native unit tests passed\n");

@@ -27,11 +27,14 @@ Optional serving features (all measured in profiles/):
 """
 from __future__ import annotations
 
+import logging
 import queue
 import threading
 import time
 from collections import defaultdict
 from dataclasses import dataclass, field
+
+log = logging.getLogger("llmlb.engine")
 
 import torch
 
@@ -74,7 +77,7 @@ class EngineConfig:
     weights_path: str | None = None
     # KV cache storage: "bfloat16" (default) or "fp8_e4m3" (OCP e4m3
     # bytes + per-token-head scales: half the attention bandwidth,
-    # double the resident context; needs D=128 and GQA group <= 4 on
+    # double the resident context; needs D=128, GQA group <= 16 on
     # GPU).  Env LLMLB_KV_DTYPE overrides.
     kv_dtype: str = "bfloat16"
     # prompt-lookup speculative decoding: draft up to this many tokens
@@ -367,9 +370,35 @@ class Engine:
 
     def _loop(self) -> None:
         while not self._stop.is_set():
-            busy = self.step()
+            try:
+                busy = self.step()
+            except Exception:  # noqa: BLE001 — engine thread must survive
+                # A step() failure (OOM, kernel error, scheduler bug)
+                # must not kill the daemon thread and hang every
+                # in-flight and queued request: fail them all with an
+                # "error" finish and keep serving.
+                log.exception("engine step failed; failing in-flight work")
+                self._fail_all_inflight("engine step error")
+                busy = False
             if not busy:
                 time.sleep(0.002)
+
+    def _fail_all_inflight(self, msg: str) -> None:
+        for seq in list(self.running):
+            try:
+                seq.out_queue.put((0, "error", {"error": msg}))
+            except Exception:  # noqa: BLE001
+                pass
+            self._retire(seq)
+        while True:
+            try:
+                seq = self.waiting.get_nowait()
+            except queue.Empty:
+                break
+            try:
+                seq.out_queue.put((0, "error", {"error": msg}))
+            except Exception:  # noqa: BLE001
+                pass
 
     # -- scheduling ------------------------------------------------------
     def _admit(self) -> list[_Sequence]:
@@ -596,17 +625,29 @@ class Engine:
     @torch.inference_mode()
     def _decode_step(self) -> None:
         _t0 = time.monotonic()
-        # grow block tables where the NEXT token crosses a page boundary
+        # grow block tables where the NEXT token crosses a page boundary.
+        # Admission reserves worst-case pages, so exhaustion here means a
+        # logic bug or prefix-cache pressure — fail just the starved
+        # sequence rather than raising out of the engine thread.
         grow_rows, grow_cols, grow_vals = [], [], []
+        starved = []
         for seq in self.running:
             if seq.length >= len(seq.block_table) * BLOCK_SIZE:
                 if self.prefix is not None:
                     self.prefix.ensure_free(1)
+                if self.kv.free_blocks() < 1:
+                    starved.append(seq)
+                    continue
                 blk = self.kv.allocate(1)[0]
                 seq.block_table.append(blk)
                 grow_rows.append(seq.row)
                 grow_cols.append(len(seq.block_table) - 1)
                 grow_vals.append(blk)
+        for seq in starved:
+            log.error("KV pool exhausted for rid=%s; failing sequence",
+                      seq.rid)
+            seq.out_queue.put((0, "error", {"error": "kv pages exhausted"}))
+            self._retire(seq)
         if grow_rows:
             self._bt_dev[torch.tensor(grow_rows, device=self.device),
                          torch.tensor(grow_cols, device=self.device)] = \
@@ -732,10 +773,30 @@ class Engine:
         cap = self.max_blocks_per_seq * BLOCK_SIZE
         for i, s in enumerate(seqs):
             room = min(self.cfg.model.max_pos, cap) - s.length - 1
+            # also bound by what the KV pool can actually grow to right
+            # now: pages beyond the current table must be allocatable,
+            # else truncate the draft instead of raising mid-step
+            # 
+            free_room = (len(s.block_table) * BLOCK_SIZE
+                         + self.kv.free_blocks() * BLOCK_SIZE
+                         - s.length - 1)
+            room = min(room, free_room)
             if len(drafts[i]) > max(room, 0):
                 drafts[i] = drafts[i][:max(room, 0)]
         F = 1 + max((len(d) for d in drafts), default=0)
         if F == 1:
+            self._decode_step()
+            return
+        # whole-batch page pre-check: per-seq free_room above cannot see
+        # the OTHER sequences' concurrent growth, so verify the total
+        # demand fits before touching the pool; plain decode needs only
+        # the admission-time reservation and always proceeds
+        total_need = sum(
+            max(0, (s.length + F + BLOCK_SIZE - 1) // BLOCK_SIZE
+                - len(s.block_table)) for s in seqs)
+        if self.prefix is not None and total_need:
+            self.prefix.ensure_free(total_need)
+        if total_need > self.kv.free_blocks():
             self._decode_step()
             return
         last = self._last_tok[self._rows_tensor()].tolist()
@@ -744,8 +805,6 @@ class Engine:
             # grow the block table to cover length+F new slots
             need_blocks = (s.length + F + BLOCK_SIZE - 1) // BLOCK_SIZE
             while len(s.block_table) < need_blocks:
-                if self.prefix is not None:
-                    self.prefix.ensure_free(1)
                 s.block_table.append(self.kv.allocate(1)[0])
             feed = [t0] + d + [t0] * (F - 1 - len(d))
             feed_rows.append(feed)

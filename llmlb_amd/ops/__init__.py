@@ -389,7 +389,7 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
         else seq_lens.to(torch.int32)
     if kcache.dtype == torch.uint8:
         assert use_v2, "FP8 KV requires the v2 attention shape " \
-            "(D=128, GQA group in 1/2/4/8)"
+            "(D=128, GQA group <= 16)"
         rc = lib.ll_paged_attn_decode_fp8(
             _ptr(out), acc_p, ml_p, _ptr(q), _ptr(kcache), _ptr(vcache),
             _ptr(k_scale), _ptr(v_scale), _ptr(bt), _ptr(sl),

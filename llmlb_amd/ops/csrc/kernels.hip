@@ -492,7 +492,8 @@ void paged_attn_decode_kernel(
 
 
 // ---------------------------------------------------------------------------
-// Paged decode attention v2 — GQA-amortized (group <= 4, D == 128).
+// Paged decode attention v2 — GQA-amortized (any group <= 16 via
+// greedy 4/2/1 sub-group passes, D == 128).
 //
 // v1 gives each wave ONE query head, so the 4 waves of a workgroup
 // each stream the same K/V pages (4x redundant L1/L2 traffic — the
@@ -942,7 +943,7 @@ v1_path:
   return 0;
 }
 
-// FP8-KV decode attention: v2 kernel only (D=128, group<=4); same
+// FP8-KV decode attention: v2 kernel only (D=128, group<=16); same
 // workspace/reduce and single-partition LDS-combine semantics as bf16
 int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
                              const void* q, const void* kcache,

@@ -110,3 +110,49 @@ def test_permission_matrix(gw):
     assert _req("POST", gw + "/v1/chat/completions", chat,
                 headers={"Authorization":
                          f"Bearer {keep}"})[0] == 200
+
+
+def test_scoped_permission_matrix(gw):
+    """Reference 11-scope matrix (common/auth.rs:60-95): scoped keys
+    reach exactly their routes; substring tokens grant nothing."""
+    admin = {"Authorization": "Bearer adm"}
+
+    def mint(perms):
+        status, body = _req("POST", gw + "/api/api-keys",
+                            {"name": perms, "permissions": perms},
+                            headers=admin)
+        assert status in (200, 201), body
+        return {"Authorization": f"Bearer {json.loads(body)['api_key']}"}
+
+    chat = {"model": "pm", "messages": [{"role": "user", "content": "x"}]}
+
+    ep_read = mint("endpoints.read")
+    assert _req("GET", gw + "/api/endpoints", headers=ep_read)[0] == 200
+    assert _req("POST", gw + "/api/endpoints", {"name": "x"},
+                headers=ep_read)[0] in (401, 403)          # no manage
+    assert _req("POST", gw + "/v1/chat/completions", chat,
+                headers=ep_read)[0] in (401, 403)          # no inference
+
+    models_read = mint("openai.models.read")
+    assert _req("GET", gw + "/v1/models", headers=models_read)[0] == 200
+    assert _req("POST", gw + "/v1/chat/completions", chat,
+                headers=models_read)[0] in (401, 403)
+
+    metrics = mint("metrics.read")
+    assert _req("GET", gw + "/api/dashboard/summary",
+                headers=metrics)[0] == 200
+    assert _req("GET", gw + "/api/endpoints",
+                headers=metrics)[0] in (401, 403)
+
+    # legacy "inference" implies models.read but not admin reads
+    inf = mint("inference")
+    assert _req("GET", gw + "/v1/models", headers=inf)[0] == 200
+    assert _req("GET", gw + "/api/dashboard/summary",
+                headers=inf)[0] in (401, 403)
+
+    # substring tokens must not grant anything (ADVICE r1)
+    bad = mint("badminton")
+    assert _req("POST", gw + "/v1/chat/completions", chat,
+                headers=bad)[0] in (401, 403)
+    assert _req("GET", gw + "/api/endpoints", headers=bad)[0] in (401,
+                                                                  403)
