@@ -63,28 +63,65 @@ def main():
             return o
 
         def v2(a, b):
-            return ops.decode_gemm_v2(a, b)
+            return ops.decode_gemm_v2(a, b, ver="2")
 
-        t_v1 = bench_one(v1, x, w)
+        def v3(a, b):
+            return ops.decode_gemm_v2(a, b, ver="3")
+
+        w8, w8s = ops.quantize_fp8_weight(w)
+
+        def v3_fp8(a, b):
+            return ops.decode_gemm_v2(a, w8, w_scale=w8s, ver="3")
+
+        ref = torch.nn.functional.linear(x, w).float()
         t_blas = bench_one(torch.nn.functional.linear, x, w)
-        try:
-            t_v2 = bench_one(v2, x, w)
-            err = (v2(x, w).float()
-                   - torch.nn.functional.linear(x, w).float()) \
-                .abs().max().item()
-        except ValueError:
-            t_v2, err = float("inf"), -1
+        res = {"shape": f"N{N} M{M} K{K}",
+               "hipblaslt_us": round(t_blas * 1e6, 1)}
         wbytes = M * K * 2
-        out[label] = {
-            "shape": f"N{N} M{M} K{K}",
-            "v1_us": round(t_v1 * 1e6, 1),
-            "v2_us": round(t_v2 * 1e6, 1),
-            "hipblaslt_us": round(t_blas * 1e6, 1),
-            "v2_tbps": round(wbytes / t_v2 / 1e12, 2),
-            "hipblaslt_tbps": round(wbytes / t_blas / 1e12, 2),
-            "v2_speedup_vs_blas": round(t_blas / t_v2, 2),
-            "v2_max_err": round(err, 4),
-        }
+        res["hipblaslt_tbps"] = round(wbytes / t_blas / 1e12, 2)
+        for name, fn, wb in (("v2", v2, wbytes), ("v3", v3, wbytes),
+                             ("v3_fp8", v3_fp8, wbytes // 2)):
+            try:
+                t = bench_one(fn, x, w)
+                d = (fn(x, w).float() - ref).abs()
+                rel = (d / ref.abs().clamp(min=1e-3)).max().item()
+            except ValueError:
+                t, rel = float("inf"), -1
+            res[f"{name}_us"] = round(t * 1e6, 1)
+            res[f"{name}_tbps"] = round(wb / t / 1e12, 2)
+            res[f"{name}_vs_blas"] = round(t_blas / t, 2)
+            res[f"{name}_rel_err"] = round(rel, 4)
+        out[label] = res
+        print(json.dumps({label: res}), flush=True)
+    # fused-swiglu arm on the down shape (gateup [N, 2K] -> [N, M])
+    for N, M, K, label in [(64, 4096, 14336, "down_fused_swiglu")]:
+        gu = torch.randn(N, 2 * K, dtype=torch.bfloat16, device=dev) * 0.3
+        w = torch.randn(M, K, dtype=torch.bfloat16, device=dev) * 0.05
+
+        def fused(a, b):
+            o = ops.swiglu_linear(a, b)
+            assert o is not None
+            return o
+
+        def unfused(a, b):
+            g, u = a[:, :K], a[:, K:]
+            act = torch.nn.functional.silu(g) * u
+            return torch.nn.functional.linear(act, b)
+
+        os.environ["LLMLB_FUSED_SWIGLU"] = "1"
+        ops._FUSED_SWIGLU = True
+        ref = unfused(gu, w).float()
+        try:
+            t_f = bench_one(fused, gu, w)
+            rel = ((fused(gu, w).float() - ref).abs()
+                   / ref.abs().clamp(min=1e-3)).max().item()
+        except (ValueError, AssertionError):
+            t_f, rel = float("inf"), -1
+        t_u = bench_one(unfused, gu, w)
+        out[label] = {"fused_us": round(t_f * 1e6, 1),
+                      "unfused_us": round(t_u * 1e6, 1),
+                      "fused_vs_unfused": round(t_u / t_f, 2),
+                      "rel_err": round(rel, 4)}
         print(json.dumps({label: out[label]}), flush=True)
     print(json.dumps(out))
 

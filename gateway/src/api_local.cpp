@@ -137,12 +137,14 @@ std::string Gateway::render_metrics_text() {
       "# TYPE llmlb_cloud_requests_total counter\n"
       "llmlb_cloud_requests_total %lld\n"
       "# TYPE llmlb_active_requests gauge\n"
-      "llmlb_active_requests %lld\n",
+      "llmlb_active_requests %lld\n"
+      "# TYPE llmlb_admission_delays_total counter\n"
+      "llmlb_admission_delays_total %lld\n",
       metrics.requests_total.load(), metrics.responses_2xx.load(),
       metrics.responses_4xx.load(), metrics.responses_5xx.load(),
       metrics.upstream_errors.load(), metrics.streamed_responses.load(),
       metrics.tokens_streamed.load(), metrics.cloud_requests.load(),
-      metrics.active_requests.load());
+      metrics.active_requests.load(), metrics.admission_delays.load());
   std::string out(buf, n);
   // per-endpoint gauges (ops dashboards scrape these to see the
   // balancer's view of each replica)

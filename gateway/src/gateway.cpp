@@ -275,6 +275,7 @@ void ClientConn::close_now() {
   if (closed) return;
   closed = true;
   w->timed_ops.erase(this);
+  w->parked.erase(this);
   if (op) {
     // client went away mid-op: tear down upstream, record failure
     if (op->timer_id) w->loop.cancel_timer(op->timer_id);
@@ -371,7 +372,9 @@ void ClientConn::dispatch() {
   // request tracing: honour a client-supplied X-Request-Id, else mint
   // one; echoed on the response and forwarded upstream
   std::string_view rid = req.header("x-request-id");
-  if (!rid.empty() && rid.size() <= 64) {
+  if (adm_delayed && !request_id.empty()) {
+    // re-dispatch after a delayed-admission hold: keep the minted id
+  } else if (!rid.empty() && rid.size() <= 64) {
     request_id.assign(rid);
   } else {
     char buf[32];
@@ -380,7 +383,12 @@ void ClientConn::dispatch() {
     request_id = buf;
   }
   w->gw->handle_request(this);
-  if (!op && !response_open) finish_request_cycle();
+  // a conn parked by the delayed-admission arm keeps its request
+  // buffered; the timer re-enters process_requests() to re-parse and
+  // re-dispatch it (head views cannot be kept across the hold: any
+  // socket read may reallocate `in`)
+  if (!op && !response_open && !w->parked.count(this))
+    finish_request_cycle();
 }
 
 void ClientConn::finish_request_cycle() {
@@ -625,6 +633,32 @@ bool Gateway::init(std::string* err) {
   registry.load_from_db();
   auth.load();
   history.start();
+  // boot reseed (bootstrap.rs:127-159): 60-min live ring from
+  // request_history, TPS EMAs from today's daily stats
+  if (db.ok()) {
+    std::vector<DbRow> rows;
+    db.exec("SELECT ts_ms, endpoint_id, model, api_kind, status, "
+            "duration_ms, prompt_tokens, completion_tokens, client_ip "
+            "FROM request_history WHERE ts_ms > ? ORDER BY ts_ms",
+            {DbParam((long long)(wall_ms() - 60 * 60 * 1000))}, &rows);
+    for (auto& r : rows) {
+      RequestRecord rec;
+      rec.ts_ms = r[0].i;
+      rec.endpoint_id = r[1].str();
+      rec.model = r[2].str();
+      rec.api_kind = r[3].str();
+      rec.status = (int)r[4].i;
+      rec.duration_ms = r[5].kind == DbValue::kReal ? r[5].d
+                                              : (double)r[5].i;
+      rec.prompt_tokens = r[6].i;
+      rec.completion_tokens = r[7].i;
+      rec.client_ip = r[8].str();
+      history.ring().push(rec);
+    }
+    if (!rows.empty())
+      LOG_INFO("reseeded %zu request-history ring entries", rows.size());
+    registry.seed_tps_from_db();
+  }
   start_ms = wall_ms();
   int nworkers = cfg.workers > 0
                      ? cfg.workers

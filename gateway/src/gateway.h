@@ -100,6 +100,7 @@ struct Metrics {
   std::atomic<long long> tokens_streamed{0};
   std::atomic<long long> cloud_requests{0};
   std::atomic<long long> active_requests{0};
+  std::atomic<long long> admission_delays{0};
 };
 
 // auth store: api-key sha256 lookup (auth/middleware.rs analog)
@@ -223,6 +224,7 @@ class ClientConn : public FdHandler {
   std::unique_ptr<ProxyOp> op;
   uint32_t events = 0;
   bool response_open = false;  // streaming response in progress
+  bool adm_delayed = false;    // delayed-admission hold already served
 
   void flush_out();            // try write; arm EPOLLOUT if partial
   void close_now();
@@ -271,6 +273,10 @@ class Worker {
   unsigned long long request_seq = 0;
   // ops with an armed inference timeout, swept every 250 ms
   std::unordered_set<ClientConn*> timed_ops;
+  // conns parked by the delayed-admission arm (50-80% load -> 10-100ms
+  // hold, balancer/mod.rs:2255-2270); membership validates the timer
+  // callback (the conn may close while parked)
+  std::unordered_set<ClientConn*> parked;
   void arm_op_sweep();
   void release_upstream(UpstreamConn* uc);   // pool for reuse
   void discard_upstream(UpstreamConn* uc);   // close

@@ -96,7 +96,8 @@ static bool build_upstream_body(std::string_view body,
 // Request dispatch
 // -------------------------------------------------------------------
 void Gateway::handle_request(ClientConn* c) {
-  metrics.requests_total.fetch_add(1, std::memory_order_relaxed);
+  if (!c->adm_delayed)   // parked re-dispatch counts once
+    metrics.requests_total.fetch_add(1, std::memory_order_relaxed);
   const HttpRequest& r = c->req;
   std::string_view body =
       c->in.view().substr(r.header_len, (size_t)r.content_length);
@@ -228,6 +229,22 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
         : sel.err_status == 429 ? "all endpoints at capacity"
                                 : "no endpoint online for this model";
     c->send_json_error(sel.err_status, sel.err_code, msg);
+    return;
+  }
+
+  // delayed-admission arm (reference admission_control 50%->delay,
+  // 80%->reject): park the request on the event loop for 10-100 ms,
+  // then re-run selection (conditions may have improved).  One hold
+  // max per request; the conn may close while parked, so membership
+  // in w->parked validates the timer.
+  if (sel.delay_ms > 0 && !c->adm_delayed) {
+    c->adm_delayed = true;
+    Worker* w = c->w;
+    w->parked.insert(c);
+    metrics.admission_delays.fetch_add(1, std::memory_order_relaxed);
+    w->loop.add_timer(sel.delay_ms, [w, c] {
+      if (w->parked.erase(c)) c->process_requests();
+    });
     return;
   }
 

@@ -341,6 +341,40 @@ void Registry::clear_tps_for_endpoint(const std::string& endpoint_id) {
   }
 }
 
+// Boot-time TPS reseed from today's persisted daily stats
+// (bootstrap.rs:143-159, balancer/mod.rs:2612-2634): one synthetic
+// EMA sample per (endpoint, model, api_kind) so a restarted gateway
+// is not score-blind until every backend re-reports.
+void Registry::seed_tps_from_db() {
+  if (!db_ || !db_->ok()) return;
+  std::vector<DbRow> rows;
+  db_->exec(
+      "SELECT endpoint_id, model, api_kind, completion_tokens, "
+      "total_duration_ms FROM endpoint_daily_stats WHERE "
+      "date = date('now') AND requests > 0 AND total_duration_ms > 0",
+      {}, &rows);
+  int64_t now = wall_ms();
+  int seeded = 0;
+  for (auto& r : rows) {
+    double tokens = (double)r[3].i;
+    double dur_s = (r[4].kind == DbValue::kReal ? r[4].d
+                                            : (double)r[4].i) /
+               1000.0;
+    if (tokens <= 0 || dur_s <= 0) continue;
+    ApiKind kind = ApiKind::kOther;
+    std::string kname = r[2].str();
+    for (ApiKind k : {ApiKind::kChat, ApiKind::kCompletions,
+                      ApiKind::kEmbeddings, ApiKind::kResponses,
+                      ApiKind::kMessages, ApiKind::kAudio,
+                      ApiKind::kImages})
+      if (kname == api_kind_name(k)) kind = k;
+    tps_state(r[0].str(), r[1].str(), kind)->seed(tokens / dur_s, now);
+    ++seeded;
+  }
+  if (seeded)
+    LOG_INFO("seeded %d TPS cells from today's daily stats", seeded);
+}
+
 SelectResult select_endpoint(const RouteTable& table,
                              const std::string& model, ApiKind kind,
                              const AdmissionConfig& adm, int64_t now_ms,
@@ -405,6 +439,18 @@ SelectResult select_endpoint(const RouteTable& table,
   uint64_t rr = route.rr->fetch_add(1, std::memory_order_relaxed);
   const Cand& pick = cands[rr % top_end];
   res.ep = route.endpoints[pick.idx];
+  {
+    // delayed-admission arm: 50%..80% load -> 10..100 ms hold
+    // (reference admission_control, balancer/mod.rs:2255-2270)
+    double frac =
+        (double)res.ep->active.load(std::memory_order_relaxed) /
+        adm.max_concurrent_per_endpoint;
+    if (frac >= adm.delay_fraction && adm.reject_fraction >
+                                          adm.delay_fraction)
+      res.delay_ms =
+          10 + (int)((frac - adm.delay_fraction) /
+                     (adm.reject_fraction - adm.delay_fraction) * 90.0);
+  }
   res.upstream_model = route.upstream_ids[pick.idx];
   res.canonical = route.canonical;
   res.tps = route.tps[pick.idx];

@@ -59,6 +59,18 @@ struct TpsState {
   double last_sample = 0.0;
   long long samples = 0;
 
+  // boot-time seed from persisted daily stats (bootstrap.rs:143-159):
+  // one synthetic sample so first selections after a restart are not
+  // score-blind; the first live sample then EMA-blends over it
+  void seed(double tps, int64_t now) {
+    std::lock_guard<std::mutex> lk(mu);
+    if (samples > 0) return;
+    ema = tps;
+    last_sample = tps;
+    samples = 1;
+    updated_ms = now;
+  }
+
   void update(double tokens, double duration_s, double alpha, int64_t now) {
     if (duration_s <= 0.0) return;
     double tps = tokens / duration_s;
@@ -185,6 +197,8 @@ class Registry {
   void record_tps(const std::string& endpoint_id, const std::string& model,
                   ApiKind kind, double tokens, double duration_s);
   void clear_tps_for_endpoint(const std::string& endpoint_id);
+  // boot reseed from endpoint_daily_stats (today's rows)
+  void seed_tps_from_db();
   double tps_score(const std::string& endpoint_id, const std::string& model,
                    ApiKind kind);
 
@@ -292,6 +306,10 @@ struct SelectResult {
   std::shared_ptr<TpsState> tps;  // the (endpoint, model, chat) EMA cell
   int err_status = 0;           // 0 = ok; else 404/429/503
   std::string err_code;
+  // delayed admission (reference admission_control,
+  // balancer/mod.rs:2255-2270): >=50% load on the picked endpoint ->
+  // hold the request 10-100 ms before forwarding
+  int delay_ms = 0;
 };
 
 // TPS-weighted selection with RR tie-break + admission control.

@@ -190,6 +190,24 @@ static void test_tps_and_selection() {
   route.endpoints[1]->active = 90;  // of max 100
   CHECK(select_endpoint(table, "m", ApiKind::kChat, adm, 5000).ep->id !=
         "e2");
+  // delayed-admission arm (balancer/mod.rs:2255-2270): picked endpoint
+  // between 50% and 80% load -> 10..100 ms hold; below 50% -> none
+  route.endpoints[1]->active = 0;
+  route.endpoints[0]->active = 65;   // 65% -> delay = 10 + 0.5*90 = 55
+  route.tps[0]->update(900, 1.0, 0.2, 5000);  // make e1 win selection
+  route.tps[1]->reset();
+  table.by_model["m"] = route;
+  {
+    auto r2 = select_endpoint(table, "m", ApiKind::kChat, adm, 5000);
+    CHECK(r2.ep->id == "e1");
+    CHECK(r2.delay_ms >= 50 && r2.delay_ms <= 60);
+    route.endpoints[0]->active = 10;  // 10% -> no delay
+    auto r3 = select_endpoint(table, "m", ApiKind::kChat, adm, 5000);
+    CHECK(r3.delay_ms == 0);
+    route.endpoints[0]->active = 0;
+    route.tps[0]->reset();
+  }
+  route.endpoints[1]->active = 90;   // restore state for HBM-gate test
   // HBM gate: e1+e3 over budget and e2 busy -> 429
   for (auto& ep : route.endpoints) {
     std::lock_guard<std::mutex> lk(ep->gpu_mu);

@@ -47,6 +47,7 @@ def _try_load() -> None:
     _lib.ll_kv_append_fp8.restype = ctypes.c_int
     _lib.ll_decode_gemm.restype = ctypes.c_int
     _lib.ll_decode_gemm_v2.restype = ctypes.c_int
+    _lib.ll_decode_gemm_v3.restype = ctypes.c_int
 
 
 _try_load()
@@ -212,10 +213,18 @@ def _split_k_ws(N: int, M: int, device) -> torch.Tensor:
     return t
 
 
+# staged decode-GEMM generation: "3" (default) = software-pipelined v3
+# (double-buffered W register sets + LDS x buffers, nt weight stream);
+# "2" = the r1 single-buffered kernel, kept for A/B.
+_DECODE_GEMM_VER = os.environ.get("LLMLB_DECODE_GEMM_VER", "3")
+
+
 def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
                    ws: torch.Tensor | None = None,
-                   w_scale: torch.Tensor | None = None) -> torch.Tensor:
-    """Staged-x decode GEMM (v2).  Raises on unsupported shapes.
+                   w_scale: torch.Tensor | None = None,
+                   ver: str | None = None) -> torch.Tensor:
+    """Staged-x decode GEMM (v3 pipelined by default; see
+    _DECODE_GEMM_VER).  Raises on unsupported shapes.
     ``w_scale`` non-None = fp8-weight arm: ``weight`` holds e4m3 bytes
     and per-row scales fold into the store (half the weight traffic on
     the bandwidth-bound decode shapes)."""
@@ -226,7 +235,9 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     need_ws = (M // 128) < 224
     if need_ws and ws is None:
         ws = _split_k_ws(N, M, x.device)
-    rc = lib.ll_decode_gemm_v2(
+    entry = (lib.ll_decode_gemm_v3
+             if (ver or _DECODE_GEMM_VER) == "3" else lib.ll_decode_gemm_v2)
+    rc = entry(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
         _ptr(weight),
         _ptr(w_scale) if w_scale is not None else None,
@@ -280,10 +291,17 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor,
             return decode_gemm_v2(x, w8[0], w_scale=w8[1])
         except ValueError:
             pass
-    legal = shape_ok and (mode != "auto" or N <= 16)
-    if legal:
+    # measured per-shape policy (profiles/r02_decode_gemm_v3_ab.md,
+    # three interleaved probe runs): the pipelined v3 wins o_proj
+    # (+15-20%) and down (+16-20%) and ties qkv; hipBLASLt keeps the
+    # fat gate_up (M=28672) at N>16; the v1-style fragment kernel (v2
+    # entry, ver="2") stays best at tiny N.
+    if shape_ok:
         try:
-            return decode_gemm_v2(x, weight)
+            if N <= 16:
+                return decode_gemm_v2(x, weight, ver="2")
+            if mode != "auto" or M <= 8192:
+                return decode_gemm_v2(x, weight, ver="3")
         except ValueError:
             pass
     return torch.nn.functional.linear(x, weight)
@@ -319,7 +337,9 @@ def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
     lib = _require_native()
     out = torch.empty(N, M, dtype=gateup.dtype, device=gateup.device)
     ws = _split_k_ws(N, M, gateup.device) if (M // 128) < 224 else None
-    rc = lib.ll_decode_gemm_v2(
+    entry = (lib.ll_decode_gemm_v3 if _DECODE_GEMM_VER == "3"
+             else lib.ll_decode_gemm_v2)
+    rc = entry(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(gateup),
         _ptr(weight), None,
         ctypes.c_int(N), ctypes.c_int(M), ctypes.c_int(K),

@@ -31,6 +31,7 @@
 
 typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ float dg_bf2f(unsigned short u) {
@@ -458,6 +459,358 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
     default: return -2;
   }
 #undef LAUNCH_V2
+  if (SK > 1) {
+    long long n = (long long)N * M;
+    int thr = 256;
+    long long blocks2 = (n + thr - 1) / thr;
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((unsigned)blocks2),
+                       dim3(thr), 0, s, (bf16*)out, (float*)ws, n);
+  }
+  hipError_t e = hipGetLastError();
+  return e == hipSuccess ? 0 : (int)e;
+}
+
+}  // extern "C"
+
+// ===========================================================================
+// v3: software-pipelined v2.  The v2 consume loop compiled to ONE
+// global_load_dwordx4 + s_waitcnt vmcnt(0) per MFMA step (hipcc reused
+// the bfrag registers), i.e. one serialized HBM round trip per 32-k
+// step — the measured 11k-cycle chunks behind the 3.0 TB/s ceiling
+// (profiles/r01_decode_gemm_v2_ab.md).  v3 keeps TWO explicit register
+// sets for the W fragments and TWO LDS x buffers: chunk c+1's W and x
+// loads issue before chunk c's MFMAs, so 8 B-fragment loads per wave
+// are in flight across the whole compute phase and the waits become
+// counted (guide §5 "glds vs register staging": T14 write-after-barrier
+// register pipeline; one barrier per chunk).  W loads are nontemporal
+// (each byte is read by exactly one workgroup — MI355X_MICROARCH
+// nt-weights row).
+// ===========================================================================
+
+template <int N_TILES, bool ATOMIC, bool SWIGLU = false, bool W8 = false,
+          int SPLIT_W = 1, int CHUNK_K = 256>
+// SPLIT_W=2 (BN=64) targets TWO 512-thread workgroups per CU (double
+// the W bytes in flight); that needs <=128 VGPRs, which N_TILES<=4
+// fits without spilling.  SPLIT_W=1 keeps the 2-wave/SIMD budget —
+// forcing 128 there made hipcc spill 16-56 B/lane to scratch.
+__global__ __launch_bounds__(512, (N_TILES <= 4 && SPLIT_W == 2 ? 4 : 2))
+void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
+                           float* __restrict__ out_ws,  // [N, M] (SK>1)
+                           const bf16* __restrict__ x,  // [N, K]
+                           const bf16* __restrict__ w,  // [M, K] (e4m3
+                                                        // bytes when W8)
+                           const float* __restrict__ w_scale,  // [M]; W8
+                           int N, int M, int K, int SK) {
+  constexpr int STEPS = CHUNK_K / 32;       // MFMA k-steps per chunk
+  constexpr int ROWS = N_TILES * 16;
+  // x pieces staged per thread per chunk (16 B each)
+  constexpr int PIECES = ROWS * CHUNK_K * 2 / (512 * 16);
+  constexpr int COLS_W = 8 / SPLIT_W;       // 16-col groups per block
+  constexpr int WSTEPS = STEPS / SPLIT_W;   // MFMA steps per wave/chunk
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int mblk = blockIdx.x;
+  const int col_group = wid % COLS_W;
+  const int kslice_w = wid / COLS_W;        // in-block k slice
+  const int col_base = mblk * (COLS_W * 16) + col_group * 16;
+  const int Ks = K / SK;                    // this BLOCK's K range
+  const int k_begin = blockIdx.y * Ks;
+  const int l_lo = lane & 15;
+  const int l_hi = lane >> 4;
+  // lane's k offset within a chunk: slice picks its half of the steps
+  const int koff = kslice_w * (CHUNK_K / SPLIT_W) + l_hi * 8;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // two x buffers, ONE __shared__ object (guide §5 trap 4a)
+  char* xs[2] = {smem, smem + (size_t)ROWS * CHUNK_K * 2};
+
+  const bf16* wrow = W8 ? nullptr : w + (size_t)(col_base + l_lo) * K;
+  const unsigned char* wrow8 =
+      W8 ? reinterpret_cast<const unsigned char*>(w) +
+               (size_t)(col_base + l_lo) * K
+         : nullptr;
+  const float wsc = W8 ? w_scale[col_base + l_lo] : 1.f;
+
+  f32x4 acc[N_TILES];
+#pragma unroll
+  for (int t = 0; t < N_TILES; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int piece0 = threadIdx.x * PIECES;
+
+  // ---- helpers as macros so the two register sets stay distinct ----
+  // issue the 8 (W8: 8 half-width) W fragment loads for chunk at k0
+#define V3_LOAD_W(dst, k0)                                                  \
+  do {                                                                      \
+    if (!W8) {                                                              \
+      _Pragma("unroll") for (int s2 = 0; s2 < WSTEPS; ++s2)                 \
+          dst[s2] = __builtin_nontemporal_load(                             \
+              reinterpret_cast<const bf16x8*>(wrow + (k0) + s2 * 32 +       \
+                                              koff));                       \
+    }                                                                       \
+  } while (0)
+#define V3_LOAD_W8(dst8, k0)                                                \
+  do {                                                                      \
+    if (W8) {                                                               \
+      _Pragma("unroll") for (int s2 = 0; s2 < WSTEPS; ++s2)                 \
+          dst8[s2] = __builtin_nontemporal_load(                            \
+              reinterpret_cast<const u32x2*>(wrow8 + (k0) + s2 * 32 +       \
+                                             koff));                        \
+    }                                                                       \
+  } while (0)
+  // issue the x piece loads for chunk at k0 into xv[] (SWIGLU: gate+up)
+#define V3_LOAD_X(xv, xv2, k0)                                              \
+  do {                                                                      \
+    _Pragma("unroll") for (int p = 0; p < PIECES; ++p) {                    \
+      const int piece = piece0 + p;                                         \
+      const int tok = piece / (CHUNK_K * 2 / 16);                           \
+      const int kb = piece % (CHUNK_K * 2 / 16);                            \
+      const int src_tok = tok < N ? tok : 0;                                \
+      if (SWIGLU) {                                                         \
+        const bf16* row = x + (size_t)src_tok * 2 * K;                      \
+        xv[p] = *reinterpret_cast<const uint4*>(row + (k0) + kb * 8);       \
+        xv2[p] = *reinterpret_cast<const uint4*>(row + K + (k0) + kb * 8);  \
+      } else {                                                              \
+        xv[p] = *reinterpret_cast<const uint4*>(                            \
+            x + (size_t)src_tok * K + (k0) + kb * 8);                       \
+      }                                                                     \
+    }                                                                       \
+  } while (0)
+  // write staged x pieces into LDS buffer b (swizzled), after loads land
+#define V3_WRITE_X(b, xv, xv2)                                              \
+  do {                                                                      \
+    _Pragma("unroll") for (int p = 0; p < PIECES; ++p) {                    \
+      const int piece = piece0 + p;                                         \
+      const int tok = piece / (CHUNK_K * 2 / 16);                           \
+      const int kb = piece % (CHUNK_K * 2 / 16);                            \
+      uint4 v = xv[p];                                                      \
+      if (SWIGLU) {                                                         \
+        float gf[8], uf[8];                                                 \
+        dg_unpack8(xv[p], gf);                                              \
+        dg_unpack8(xv2[p], uf);                                             \
+        _Pragma("unroll") for (int j = 0; j < 8; ++j)                       \
+            gf[j] = gf[j] / (1.f + __expf(-gf[j])) * uf[j];                 \
+        v = dg_pack8(gf);                                                   \
+      }                                                                     \
+      const int swz_kb = kb ^ (tok & 15);                                   \
+      *reinterpret_cast<uint4*>(xs[b] + (size_t)tok * CHUNK_K * 2 +         \
+                                swz_kb * 16) = v;                           \
+    }                                                                       \
+  } while (0)
+  // consume one chunk from LDS buffer b with W fragments bw/bw8
+#define V3_CONSUME(b, bw, bw8)                                              \
+  do {                                                                      \
+    _Pragma("unroll") for (int s2 = 0; s2 < WSTEPS; ++s2) {                 \
+      bf16x8 bfrag;                                                         \
+      if (W8) {                                                             \
+        float f[8];                                                         \
+        dg_fp8x4_to_f32(bw8[s2].x, f);                                      \
+        dg_fp8x4_to_f32(bw8[s2].y, f + 4);                                  \
+        const uint4 packed = dg_pack8(f);                                   \
+        bfrag = *reinterpret_cast<const bf16x8*>(&packed);                  \
+      } else {                                                              \
+        bfrag = bw[s2];                                                     \
+      }                                                                     \
+      const int kb = (s2 * 32 + koff) * 2 / 16;                             \
+      /* koff already contains this slice's half-chunk offset */            \
+      _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
+        const int tok = t * 16 + l_lo;                                      \
+        const int swz_kb = kb ^ (tok & 15);                                 \
+        bf16x8 afrag = *reinterpret_cast<const bf16x8*>(                    \
+            xs[b] + (size_t)tok * CHUNK_K * 2 + swz_kb * 16);               \
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,      \
+                                                         acc[t], 0, 0, 0);  \
+      }                                                                     \
+    }                                                                       \
+  } while (0)
+
+  bf16x8 bA[WSTEPS], bB[WSTEPS];
+  u32x2 bA8[WSTEPS], bB8[WSTEPS];
+  uint4 xvA[PIECES], xvB[PIECES];
+  uint4 xv2A[SWIGLU ? PIECES : 1], xv2B[SWIGLU ? PIECES : 1];
+
+  const int NC = Ks / CHUNK_K;              // chunks in this K slice
+
+  // prologue: chunk 0 x -> LDS buf0; chunk 0 W -> set A
+  V3_LOAD_X(xvA, xv2A, k_begin);
+  V3_LOAD_W(bA, k_begin);
+  V3_LOAD_W8(bA8, k_begin);
+  V3_WRITE_X(0, xvA, xv2A);
+  __syncthreads();
+
+  int c = 0;
+  // unroll-by-2 steady state: consume A while loading B, then swap
+  for (; c + 2 <= NC; c += 2) {
+    const int k1 = k_begin + (c + 1) * CHUNK_K;
+    V3_LOAD_X(xvB, xv2B, k1);
+    V3_LOAD_W(bB, k1);
+    V3_LOAD_W8(bB8, k1);
+    V3_CONSUME(0, bA, bA8);
+    V3_WRITE_X(1, xvB, xv2B);
+    __syncthreads();
+    if (c + 2 < NC) {
+      const int k2 = k_begin + (c + 2) * CHUNK_K;
+      V3_LOAD_X(xvA, xv2A, k2);
+      V3_LOAD_W(bA, k2);
+      V3_LOAD_W8(bA8, k2);
+      V3_CONSUME(1, bB, bB8);
+      V3_WRITE_X(0, xvA, xv2A);
+      __syncthreads();
+    } else {
+      V3_CONSUME(1, bB, bB8);
+    }
+  }
+  if (c < NC) {  // odd tail: its loads were issued in the last pair (or
+                 // the prologue when NC == 1)
+    V3_CONSUME(0, bA, bA8);
+  }
+#undef V3_LOAD_W
+#undef V3_LOAD_W8
+#undef V3_LOAD_X
+#undef V3_WRITE_X
+#undef V3_CONSUME
+
+  // ---- in-block split-K reduce (SPLIT_W == 2): the upper k-slice
+  // parks its accumulators in the dead x buffer, the lower adds ----
+  if (SPLIT_W == 2) {
+    float* red = reinterpret_cast<float*>(smem);  // reuse xs[0]
+    __syncthreads();                              // consumes done
+    if (kslice_w == 1) {
+#pragma unroll
+      for (int t = 0; t < N_TILES; ++t)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          red[((col_group * N_TILES + t) * 16 + l_hi * 4 + r) * 16 +
+              l_lo] = acc[t][r];
+    }
+    __syncthreads();
+    if (kslice_w == 1) return;
+#pragma unroll
+    for (int t = 0; t < N_TILES; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        acc[t][r] += red[((col_group * N_TILES + t) * 16 + l_hi * 4 +
+                          r) * 16 + l_lo];
+  }
+
+  // ---- store (same as v2) ----
+#pragma unroll
+  for (int t = 0; t < N_TILES; ++t) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = t * 16 + l_hi * 4 + r;
+      if (row < N) {
+        if (ATOMIC) {
+          atomicAdd(out_ws + (size_t)row * M + col_base + l_lo,
+                    W8 ? acc[t][r] * wsc : acc[t][r]);
+        } else {
+          union { float f; unsigned int i; } c2;
+          c2.f = W8 ? acc[t][r] * wsc : acc[t][r];
+          unsigned int bias = 0x7FFF + ((c2.i >> 16) & 1);
+          union { unsigned short u; __hip_bfloat16 b; } cv;
+          cv.u = (unsigned short)((c2.i + bias) >> 16);
+          out[(size_t)row * M + col_base + l_lo] = cv.b;
+        }
+      }
+    }
+  }
+}
+
+extern "C" {
+
+// v3 entry: same contract as ll_decode_gemm_v2.
+//
+// Grid policy (guide §5 "decomposition first"): the chip has 256 CUs
+// and N_TILES<=4 instantiations fit TWO 512-thread workgroups per CU,
+// so target >=256 blocks.  Preference order:
+//   1. BN=128, SK=1 when M/128 >= 256 (plain, no reduction)
+//   2. BN=64 in-block k-split (SPLIT_W=2), SK=1 when M/64 >= 256 —
+//      full fill with LDS-only reduction (gate_up: 224 -> 448 blocks)
+//   3. BN=64 SPLIT_W=2 + global split-K atomics (half the atomic
+//      traffic of the BN=128 equivalent at the same block count)
+//   4. BN=128 + global split-K (shape constraints force it)
+// LLMLB_GEMM_V3_SPLITW=1 forces BN=128 for A/B.
+int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
+                      const void* w_scale, int N, int M, int K,
+                      int swiglu_x, void* stream) {
+  if (K % 256 != 0 || M % 128 != 0 || N > 128) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  const int n_tiles = (N + 15) / 16;
+  static const char* fsw = getenv("LLMLB_GEMM_V3_SPLITW");
+  const bool allow_sw2 = !(fsw && fsw[0] == '1');
+
+  // A/B of the BN=64 in-block split (LLMLB_GEMM_V3_SPLITW=2) measured
+  // it a clean loss (gate_up 86 vs 72 us: doubled x traffic, halved
+  // per-wave load depth, doubled barriers) — BN=128 stays the default.
+  int split_w = (fsw && fsw[0] == '2' && allow_sw2) ? 2 : 1;
+  int SK = 1;
+  if (split_w == 2) {
+    while ((M / 64) * SK < 224 && SK < 8 && (K / (SK * 2)) % 256 == 0)
+      SK *= 2;
+  } else {
+    while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
+      SK *= 2;
+  }
+  const int mblocks = M / (split_w == 2 ? 64 : 128);
+  // chunk depth: 512-k chunks (2x W in flight, half the barriers)
+  // MEASURED SLOWER on every probe shape (gate_up 86.9 vs 71.8 us —
+  // the 236-248-VGPR register ring schedules worse than two 256-k
+  // pipelined chunks); keep 256 unless explicitly forced for A/B.
+  static const char* fck = getenv("LLMLB_GEMM_V3_CHUNK512");
+  const int chunk_k = (fck && fck[0] == '1' && !swiglu_x &&
+                       n_tiles <= 4 && split_w == 1 &&
+                       (K / SK) % 512 == 0)
+                          ? 512
+                          : 256;
+  const size_t lds = (size_t)n_tiles * 16 * chunk_k * 2 * 2;
+  if (lds > 160 * 1024) return -2;
+  if (SK > 1 && ws == nullptr) return -2;
+  dim3 grid(mblocks, SK), block(512);
+
+  const bool w8 = w_scale != nullptr;
+#define LAUNCH_V3_(NT, AT, SW, W8F, SWI)                                    \
+  do {                                                                      \
+    if (chunk_k == 512)                                                     \
+      hipLaunchKernelGGL((decode_gemm_v3_kernel<NT, AT, SW, W8F, SWI,       \
+                                                512>),                      \
+                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
+                         (const bf16*)x, (const bf16*)w,                    \
+                         (const float*)w_scale, N, M, K, SK);               \
+    else                                                                    \
+      hipLaunchKernelGGL((decode_gemm_v3_kernel<NT, AT, SW, W8F, SWI,       \
+                                                256>),                      \
+                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
+                         (const bf16*)x, (const bf16*)w,                    \
+                         (const float*)w_scale, N, M, K, SK);               \
+  } while (0)
+#define LAUNCH_V3S(NT, SWI)                                                 \
+  do {                                                                      \
+    if (SK > 1) {                                                           \
+      if (swiglu_x) LAUNCH_V3_(NT, true, true, false, SWI);                 \
+      else if (w8) LAUNCH_V3_(NT, true, false, true, SWI);                  \
+      else LAUNCH_V3_(NT, true, false, false, SWI);                         \
+    } else {                                                                \
+      if (swiglu_x) LAUNCH_V3_(NT, false, true, false, SWI);                \
+      else if (w8) LAUNCH_V3_(NT, false, false, true, SWI);                 \
+      else LAUNCH_V3_(NT, false, false, false, SWI);                        \
+    }                                                                       \
+  } while (0)
+#define LAUNCH_V3(NT)                                                       \
+  do {                                                                      \
+    if (split_w == 2) LAUNCH_V3S(NT, 2);                                    \
+    else LAUNCH_V3S(NT, 1);                                                 \
+  } while (0)
+  switch (n_tiles) {
+    case 1: LAUNCH_V3(1); break;
+    case 2: LAUNCH_V3(2); break;
+    case 3: LAUNCH_V3(3); break;
+    case 4: LAUNCH_V3(4); break;
+    case 5: case 6: LAUNCH_V3(6); break;
+    case 7: case 8: LAUNCH_V3(8); break;
+    default: return -2;
+  }
+#undef LAUNCH_V3
+#undef LAUNCH_V3S
+#undef LAUNCH_V3_
   if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;

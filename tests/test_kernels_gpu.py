@@ -292,3 +292,41 @@ def test_paged_attn_v2_odd_groups():
                                   sl.cpu(), D ** -0.5)
         rel = (out.float().cpu() - ref).abs().max() / ref.abs().max()
         assert rel < 5e-2, (nq, nkv, rel.item())
+
+
+@pytest.mark.parametrize("N,M,K", [(64, 4096, 4096),     # o_proj (SK)
+                                   (64, 28672, 4096),    # gate_up (SK=1)
+                                   (64, 4096, 14336),    # down (odd NC)
+                                   (16, 6144, 4096),     # small N
+                                   (33, 4096, 4096)])    # padded N
+def test_decode_gemm_v3_matches_v2_and_ref(N, M, K):
+    """v3 (software-pipelined) must agree with v2 and the fp32 ref on
+    every dispatch regime: split-K atomics, SK=1 direct store, odd
+    chunk counts (down: 7 chunks/slice), padded N tails."""
+    torch.manual_seed(5)
+    x = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    ref32 = torch.nn.functional.linear(x.float(), w.float())
+    o2 = ops.decode_gemm_v2(x, w, ver="2")
+    o3 = ops.decode_gemm_v2(x, w, ver="3")
+    for name, o in (("v2", o2), ("v3", o3)):
+        rel = ((o.float() - ref32).abs().max()
+               / ref32.abs().max().clamp(min=1e-6)).item()
+        assert rel < 2e-2, (name, rel)
+
+
+@pytest.mark.parametrize("N,M,K", [(64, 28672, 4096), (64, 4096, 14336)])
+def test_decode_gemm_v3_fp8_weights(N, M, K):
+    """W8A16 arm of v3: e4m3 rows + per-row scale vs the dequantized
+    fp32 reference (tolerance covers e4m3 quantization error)."""
+    torch.manual_seed(6)
+    x = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    w8, scale = ops.quantize_fp8_weight(w)
+    wd = (w8.view(torch.float8_e4m3fn).float()
+          * scale[:, None]).to(torch.bfloat16)
+    ref32 = torch.nn.functional.linear(x.float(), wd.float())
+    o = ops.decode_gemm_v2(x, w8, w_scale=scale, ver="3")
+    rel = ((o.float() - ref32).abs().max()
+           / ref32.abs().max().clamp(min=1e-6)).item()
+    assert rel < 2e-2, rel
