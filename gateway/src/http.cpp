@@ -213,10 +213,15 @@ void SseTap::line(std::string_view l) {
   }
   ++events_;
   // token estimate: one completion token per delta event that carries
-  // content; corrected by the final usage object when present
-  if (body.find("\"content\"") != std::string_view::npos ||
-      body.find("\"text\"") != std::string_view::npos)
+  // content; corrected by the final usage object when present, or by
+  // a BPE count over the accumulated text (tap keeps <= 256 KiB)
+  if (body.find("\"content\"") != std::string_view::npos) {
     ++event_tokens_;
+    append_json_string_field(body, "content", &text_, 256 * 1024);
+  } else if (body.find("\"text\"") != std::string_view::npos) {
+    ++event_tokens_;
+    append_json_string_field(body, "text", &text_, 256 * 1024);
+  }
   size_t upos = body.find("\"usage\"");
   if (upos != std::string_view::npos) {
     // cheap scan for prompt_tokens / completion_tokens numbers
@@ -236,6 +241,65 @@ void SseTap::line(std::string_view l) {
     if (pt >= 0) usage_prompt_ = pt;
     if (ct >= 0) usage_completion_ = ct;
   }
+}
+
+bool append_json_string_field(std::string_view body, std::string_view key,
+                              std::string* out, size_t cap) {
+  std::string pat = "\"";
+  pat.append(key);
+  pat += "\"";
+  size_t kp = body.find(pat);
+  if (kp == std::string_view::npos) return false;
+  size_t p = body.find_first_not_of(" \t", kp + pat.size());
+  if (p == std::string_view::npos || body[p] != ':') return false;
+  p = body.find_first_not_of(" \t", p + 1);
+  if (p == std::string_view::npos || body[p] != '"') return false;
+  ++p;
+  while (p < body.size() && out->size() < cap) {
+    char c = body[p];
+    if (c == '"') return true;
+    if (c == '\\' && p + 1 < body.size()) {
+      char e = body[p + 1];
+      p += 2;
+      switch (e) {
+        case 'n': out->push_back('\n'); break;
+        case 't': out->push_back('\t'); break;
+        case 'r': out->push_back('\r'); break;
+        case 'b': out->push_back('\b'); break;
+        case 'f': out->push_back('\f'); break;
+        case 'u': {
+          if (p + 4 <= body.size()) {
+            unsigned v = 0;
+            for (int i = 0; i < 4; ++i) {
+              char h = body[p + i];
+              v <<= 4;
+              if (h >= '0' && h <= '9') v |= (unsigned)(h - '0');
+              else if (h >= 'a' && h <= 'f') v |= (unsigned)(h - 'a' + 10);
+              else if (h >= 'A' && h <= 'F') v |= (unsigned)(h - 'A' + 10);
+            }
+            p += 4;
+            // UTF-8 encode (surrogates approximated as-is)
+            if (v < 0x80) {
+              out->push_back((char)v);
+            } else if (v < 0x800) {
+              out->push_back((char)(0xC0 | (v >> 6)));
+              out->push_back((char)(0x80 | (v & 0x3F)));
+            } else {
+              out->push_back((char)(0xE0 | (v >> 12)));
+              out->push_back((char)(0x80 | ((v >> 6) & 0x3F)));
+              out->push_back((char)(0x80 | (v & 0x3F)));
+            }
+          }
+          break;
+        }
+        default: out->push_back(e);
+      }
+      continue;
+    }
+    out->push_back(c);
+    ++p;
+  }
+  return true;
 }
 
 // ------------------------------------------------------------------ misc

@@ -157,6 +157,11 @@ class SseTap {
   long long completion_tokens() const {
     return usage_completion_ >= 0 ? usage_completion_ : event_tokens_;
   }
+  // true when the count came from the upstream usage object (exact);
+  // false = callers should BPE-estimate from text() (token/mod.rs:217)
+  bool usage_seen() const { return usage_completion_ >= 0; }
+  // accumulated delta content (capped) for the BPE estimate
+  const std::string& text() const { return text_; }
   long long prompt_tokens() const { return usage_prompt_; }
   bool saw_done() const { return saw_done_; }
   long long events() const { return events_; }
@@ -164,12 +169,19 @@ class SseTap {
  private:
   void line(std::string_view l);
   std::string carry_;
+  std::string text_;            // unescaped delta content, <= 256 KiB
   long long events_ = 0;
   long long event_tokens_ = 0;  // fallback estimate: one token per delta
   long long usage_prompt_ = -1;
   long long usage_completion_ = -1;
   bool saw_done_ = false;
 };
+
+// appends the FIRST top-level-ish "key":"..." string value of `body`,
+// JSON-unescaped, to *out (capped at `cap` bytes total); returns true
+// when a value was found
+bool append_json_string_field(std::string_view body, std::string_view key,
+                              std::string* out, size_t cap);
 
 // helpers
 std::string http_date_now();

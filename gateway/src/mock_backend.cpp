@@ -28,6 +28,8 @@ struct MockConfig {
   double health_tps = 0;     // advertise tokens_per_sec in /api/health
   int stream_chunks = 8;     // SSE deltas per streaming response
   double fail_rate = 0.0;    // inject failures (config 5 testing)
+  bool include_usage = true; // --no-usage: exercise the gateway's BPE
+                             // token-estimation fallback
   int completion_tokens = 8;
 };
 static MockConfig g_cfg;
@@ -180,16 +182,26 @@ static void canned_chat(MockConn* c, std::string_view body, bool echo_ids) {
   }
   if (!stream) {
     char bodybuf[1024];
-    int bl = snprintf(
-        bodybuf, sizeof bodybuf,
-        "{\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion\","
-        "\"created\":%lld,\"model\":\"%s\",\"choices\":[{\"index\":0,"
-        "\"message\":{\"role\":\"assistant\",\"content\":\"Hello from the "
-        "mock backend.\"},\"finish_reason\":\"stop\"}],\"usage\":{"
-        "\"prompt_tokens\":12,\"completion_tokens\":%d,\"total_tokens\":"
-        "%d}}",
-        id, (long long)(wall_ms() / 1000), model.c_str(),
-        g_cfg.completion_tokens, 12 + g_cfg.completion_tokens);
+    int bl;
+    if (g_cfg.include_usage)
+      bl = snprintf(
+          bodybuf, sizeof bodybuf,
+          "{\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion\","
+          "\"created\":%lld,\"model\":\"%s\",\"choices\":[{\"index\":0,"
+          "\"message\":{\"role\":\"assistant\",\"content\":\"Hello from the "
+          "mock backend.\"},\"finish_reason\":\"stop\"}],\"usage\":{"
+          "\"prompt_tokens\":12,\"completion_tokens\":%d,\"total_tokens\":"
+          "%d}}",
+          id, (long long)(wall_ms() / 1000), model.c_str(),
+          g_cfg.completion_tokens, 12 + g_cfg.completion_tokens);
+    else
+      bl = snprintf(
+          bodybuf, sizeof bodybuf,
+          "{\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion\","
+          "\"created\":%lld,\"model\":\"%s\",\"choices\":[{\"index\":0,"
+          "\"message\":{\"role\":\"assistant\",\"content\":\"Hello from the "
+          "mock backend.\"},\"finish_reason\":\"stop\"}]}",
+          id, (long long)(wall_ms() / 1000), model.c_str());
     append_status_line(c->out, 200, "OK");
     char h[128];
     int n = snprintf(h, sizeof h,
@@ -221,13 +233,23 @@ static void canned_chat(MockConn* c, std::string_view body, bool echo_ids) {
         id, model.c_str(), i);
     chunk(std::string_view(ev, n));
   }
-  int n = snprintf(
-      ev, sizeof ev,
-      "data: {\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion."
-      "chunk\",\"model\":\"%s\",\"choices\":[{\"index\":0,\"delta\":{},"
-      "\"finish_reason\":\"stop\"}],\"usage\":{\"prompt_tokens\":12,"
-      "\"completion_tokens\":%d,\"total_tokens\":%d}}\n\n",
-      id, model.c_str(), g_cfg.stream_chunks, 12 + g_cfg.stream_chunks);
+  int n;
+  if (g_cfg.include_usage) {
+    n = snprintf(
+        ev, sizeof ev,
+        "data: {\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion."
+        "chunk\",\"model\":\"%s\",\"choices\":[{\"index\":0,\"delta\":{},"
+        "\"finish_reason\":\"stop\"}],\"usage\":{\"prompt_tokens\":12,"
+        "\"completion_tokens\":%d,\"total_tokens\":%d}}\n\n",
+        id, model.c_str(), g_cfg.stream_chunks, 12 + g_cfg.stream_chunks);
+  } else {
+    n = snprintf(
+        ev, sizeof ev,
+        "data: {\"id\":\"chatcmpl-%lld\",\"object\":\"chat.completion."
+        "chunk\",\"model\":\"%s\",\"choices\":[{\"index\":0,\"delta\":{},"
+        "\"finish_reason\":\"stop\"}]}\n\n",
+        id, model.c_str());
+  }
   chunk(std::string_view(ev, n));
   chunk("data: [DONE]\n\n");
   c->out.append("0\r\n\r\n");
@@ -340,6 +362,7 @@ int main(int argc, char** argv) {
     else if (a == "--health-tps") g_cfg.health_tps = atof(next());
     else if (a == "--stream-chunks") g_cfg.stream_chunks = atoi(next());
     else if (a == "--fail-rate") g_cfg.fail_rate = atof(next());
+    else if (a == "--no-usage") g_cfg.include_usage = false;
   }
   signal(SIGPIPE, SIG_IGN);
   std::vector<std::unique_ptr<MockWorker>> ws;

@@ -10,6 +10,7 @@
 
 #include "anthropic.h"
 #include "gateway.h"
+#include "bpe.h"
 
 namespace gw {
 
@@ -683,6 +684,15 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
   long long completion_tokens = 0, prompt_tokens = 0;
   if (op->streaming) {
     completion_tokens = op->tap.completion_tokens();
+    // no usage from the backend -> BPE-estimate over the accumulated
+    // delta text (reference tiktoken fallback, token/mod.rs:217-223);
+    // event count remains the last resort
+    if (!op->tap.usage_seen() && !op->tap.text().empty()) {
+      const BpeTokenizer* bpe = BpeTokenizer::default_instance();
+      completion_tokens =
+          bpe ? bpe->count(op->tap.text())
+              : BpeTokenizer::heuristic_count(op->tap.text());
+    }
     prompt_tokens = std::max(0LL, op->tap.prompt_tokens());
     metrics.tokens_streamed.fetch_add(completion_tokens);
     if (success && op->lease_released_at_head && !op->endpoint_id.empty()) {
@@ -707,6 +717,18 @@ void Gateway::finalize_op(ClientConn* c, bool success, int status,
     completion_tokens = raw_int_field(op->buffered, "completion_tokens");
     if (completion_tokens < 0)
       completion_tokens = raw_int_field(op->buffered, "output_tokens");
+    if (completion_tokens < 0 && status < 400) {
+      // usage omitted by the backend: BPE-estimate from the response
+      // content field
+      std::string text;
+      if (append_json_string_field(op->buffered, "content", &text,
+                                   256 * 1024) &&
+          !text.empty()) {
+        const BpeTokenizer* bpe = BpeTokenizer::default_instance();
+        completion_tokens = bpe ? bpe->count(text)
+                                : BpeTokenizer::heuristic_count(text);
+      }
+    }
     completion_tokens = std::max(0LL, completion_tokens);
     prompt_tokens = raw_int_field(op->buffered, "prompt_tokens");
     if (prompt_tokens < 0)

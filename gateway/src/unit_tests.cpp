@@ -13,6 +13,7 @@
 #include "db.h"
 #include "history.h"
 #include "sha256.h"
+#include "bpe.h"
 
 using namespace gw;
 
@@ -393,6 +394,25 @@ static void test_secure_random() {
   CHECK(memcmp(a, b, sizeof a) != 0);  // astronomically unlikely equal
 }
 
+
+static void test_bpe_counts() {
+  BpeTokenizer bpe;
+  CHECK(bpe.load("data/llmlb.tiktoken") ||
+        bpe.load("gateway/data/llmlb.tiktoken"));
+  // golden counts computed by the HF tokenizer that trained the table
+  // (gateway/data/bpe_golden.json) — verifies pre-split + merge parity
+  CHECK(bpe.count("Hello, world!") == 4);
+  CHECK(bpe.count("The quick brown fox jumps over the lazy dog.") == 13);
+  CHECK(bpe.count("def fibonacci(n):\n    return n if n < 2 else fibonacci(n-1) + fibonacci(n-2)\n") == 40);
+  CHECK(bpe.count("Streaming tokens per second is the headline metric for LLM serving gateways.") == 20);
+  CHECK(bpe.count("   indented\ttext with   runs of whitespace\n\n") == 11);
+  CHECK(bpe.count("numbers 12345 and 2026-09-14 09:30:00") == 17);
+  CHECK(bpe.count("don't can't we'll they're I'm you've he'd") == 14);
+  CHECK(bpe.count("") == 0);
+  CHECK(bpe.count("a") == 1);
+  CHECK(BpeTokenizer::heuristic_count("abcdefgh") == 2);
+}
+
 static void test_caps_csv() {
   CHECK(caps_from_csv("chat,embeddings") == (kCapChat | kCapEmbeddings));
   CHECK(caps_to_csv(kCapAudioSpeech) == "audio_speech");
@@ -412,6 +432,7 @@ int main() {
   test_audit_archive();
   test_anthropic_stream_encoder_fragmented();
   test_permission_matrix();
+  test_bpe_counts();
   test_secure_random();
   test_caps_csv();
   if (g_failures == 0) {
