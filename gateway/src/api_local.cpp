@@ -86,6 +86,22 @@ std::string Gateway::render_models_json() {
                                         Json(route.max_tokens));
     data.emplace_back(Json(std::move(m)));
   }
+  {
+    // cloud-provider models (cached; reference openai.rs:451-467) —
+    // routable via their provider prefix, always "ready"
+    std::lock_guard<std::mutex> lk(cloud_models_mu);
+    for (auto& cm : cloud_models_cache) {
+      JsonObject m;
+      m.emplace_back("id", Json(cm.id));
+      m.emplace_back("object", Json("model"));
+      m.emplace_back("created", Json((long long)(wall_ms() / 1000)));
+      m.emplace_back("owned_by", Json(cm.provider));
+      m.emplace_back("endpoint_ids", Json(JsonArray{}));
+      m.emplace_back("capabilities", Json("chat"));
+      m.emplace_back("ready", Json(true));
+      data.emplace_back(Json(std::move(m)));
+    }
+  }
   JsonObject o;
   o.emplace_back("object", Json("list"));
   o.emplace_back("data", Json(std::move(data)));
@@ -516,6 +532,92 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       std::string out;
       telemetry_snapshot_json(&out);
       c->send_simple(200, "application/json", out);
+      return;
+    }
+    // -- audit query + full-text search (reference audit_log API +
+    //    migrations/026 FTS5) -----------------------------------------
+    if (path == "/api/audit" && method == "GET") {
+      std::string q(r.query);
+      auto param = [&](const char* key) -> std::string {
+        std::string pat = std::string(key) + "=";
+        size_t p = q.find(pat);
+        if (p != 0 && (p == std::string::npos || q[p - 1] != '&'))
+          return "";
+        size_t v = p + pat.size();
+        size_t e = q.find('&', v);
+        return url_decode(q.substr(v, e == std::string::npos
+                                          ? std::string::npos
+                                          : e - v));
+      };
+      std::string sql =
+          "SELECT id, ts_ms, method, path, status, client_ip, user, "
+          "duration_ms, hash FROM audit_log WHERE 1=1";
+      std::vector<DbParam> binds;
+      if (!param("from").empty()) {
+        sql += " AND ts_ms >= ?";
+        binds.emplace_back((long long)atoll(param("from").c_str()));
+      }
+      if (!param("to").empty()) {
+        sql += " AND ts_ms <= ?";
+        binds.emplace_back((long long)atoll(param("to").c_str()));
+      }
+      if (!param("status").empty()) {
+        sql += " AND status = ?";
+        binds.emplace_back(atoi(param("status").c_str()));
+      }
+      if (!param("path").empty()) {
+        sql += " AND path LIKE ?";
+        binds.emplace_back(param("path") + "%");
+      }
+      if (!param("user").empty()) {
+        sql += " AND user = ?";
+        binds.emplace_back(param("user"));
+      }
+      std::string search = param("q");
+      if (!search.empty()) {
+        if (db.fts5()) {
+          sql += " AND id IN (SELECT rowid FROM audit_fts WHERE "
+                 "audit_fts MATCH ?)";
+          binds.emplace_back(search);
+        } else {  // libsqlite3 without FTS5: LIKE fallback
+          sql += " AND (path LIKE ? OR method LIKE ? OR client_ip "
+                 "LIKE ? OR user LIKE ?)";
+          std::string like = "%" + search + "%";
+          for (int i = 0; i < 4; ++i) binds.emplace_back(like);
+        }
+      }
+      int limit = param("limit").empty() ? 100
+                                         : atoi(param("limit").c_str());
+      if (limit <= 0 || limit > 1000) limit = 100;
+      int offset = param("offset").empty()
+                       ? 0
+                       : atoi(param("offset").c_str());
+      sql += " ORDER BY id DESC LIMIT ? OFFSET ?";
+      binds.emplace_back(limit);
+      binds.emplace_back(offset);
+      std::vector<DbRow> rows;
+      db.exec(sql, binds, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("id", Json(row[0].i));
+        o.emplace_back("ts_ms", Json(row[1].i));
+        o.emplace_back("method", Json(row[2].str()));
+        o.emplace_back("path", Json(row[3].str()));
+        o.emplace_back("status", Json(row[4].i));
+        o.emplace_back("client_ip", Json(row[5].str()));
+        o.emplace_back("user", Json(row[6].str()));
+        o.emplace_back("duration_ms",
+                       Json(row[7].kind == DbValue::kReal
+                                ? row[7].d
+                                : (double)row[7].i));
+        o.emplace_back("hash", Json(row[8].str()));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      JsonObject out;
+      out.emplace_back("entries", Json(std::move(arr)));
+      out.emplace_back("fts", Json(db.fts5()));
+      c->send_simple(200, "application/json", Json(std::move(out)).dump());
       return;
     }
     if (path == "/api/audit/verify" && method == "GET") {

@@ -228,6 +228,21 @@ bool Db::migrate(std::string* err) {
   sqlite3_exec(db_,
                "ALTER TABLE request_history ADD COLUMN request_id TEXT",
                nullptr, nullptr, nullptr);
+  // audit full-text search (reference migrations/026): contentless-
+  // delete FTS5 over audit_log kept in sync by triggers.  Some
+  // libsqlite3 builds lack FTS5 — record availability and fall back
+  // to LIKE queries in the search endpoint.
+  int frc = sqlite3_exec(
+      db_,
+      "CREATE VIRTUAL TABLE IF NOT EXISTS audit_fts USING fts5("
+      "method, path, client_ip, user, content='audit_log', "
+      "content_rowid='id');"
+      "CREATE TRIGGER IF NOT EXISTS audit_fts_ai AFTER INSERT ON "
+      "audit_log BEGIN INSERT INTO audit_fts(rowid, method, path, "
+      "client_ip, user) VALUES (new.id, new.method, new.path, "
+      "new.client_ip, new.user); END;",
+      nullptr, nullptr, nullptr);
+  fts5_ = (frc == SQLITE_OK);
   return true;
 }
 

@@ -98,6 +98,7 @@ class Db {
             std::vector<DbRow>* out = nullptr, std::string* err = nullptr);
   long long last_insert_rowid();
   int changes();
+  bool fts5() const { return fts5_; }
   // run the schema migrations (idempotent)
   bool migrate(std::string* err = nullptr);
 
@@ -105,6 +106,7 @@ class Db {
 
  private:
   sqlite3* db_ = nullptr;
+  bool fts5_ = false;
   std::mutex mu_;  // serialize writers; SQLite handles its own locking too
 };
 

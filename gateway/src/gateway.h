@@ -305,6 +305,13 @@ class Gateway {
   Metrics metrics;
   EventRing events;
   std::atomic<bool> draining{false};   // inference gate (self-update)
+  // cached cloud-provider model listings, merged into /v1/models
+  // (openai.rs:451-467); refreshed by the health loop every 10 min
+  std::mutex cloud_models_mu;
+  struct CloudModel { std::string id, provider; };
+  std::vector<CloudModel> cloud_models_cache;
+  int64_t cloud_models_fetched_ms = 0;
+  void refresh_cloud_models();
   std::atomic<long long> start_ms{0};
 
   std::vector<std::unique_ptr<Worker>> workers;

@@ -277,12 +277,63 @@ bool Gateway::sync_endpoint_models(const EndpointPtr& ep, std::string* err) {
 }
 
 // ---------------------------------------------------------------- loop
+// fetch {base}/models from each configured cloud provider and cache
+// the ids prefixed for routing ("openai:gpt-4o" etc.); 10-min TTL like
+// the reference's cloud model cache (api/cloud_models.rs)
+void Gateway::refresh_cloud_models() {
+  struct Prov { const char* name; const std::string* base;
+                const std::string* key; };
+  const Prov provs[] = {{"openai", &cfg.openai_base, &cfg.openai_key},
+                        {"google", &cfg.google_base, &cfg.google_key},
+                        {"anthropic", &cfg.anthropic_base,
+                         &cfg.anthropic_key}};
+  std::vector<CloudModel> fresh;
+  for (auto& pv : provs) {
+    if (pv.base->empty()) continue;
+    // parse http://host:port/prefix
+    std::string b = *pv.base;
+    if (starts_with(b, "http://")) b = b.substr(7);
+    size_t slash = b.find('/');
+    std::string hostport = b.substr(0, slash);
+    std::string prefix = slash == std::string::npos ? "" : b.substr(slash);
+    size_t colon = hostport.find(':');
+    std::string host = hostport.substr(0, colon);
+    int port = colon == std::string::npos
+                   ? 80
+                   : atoi(hostport.c_str() + colon + 1);
+    std::vector<std::string> hdrs;
+    if (!pv.key->empty())
+      hdrs.push_back("Authorization: Bearer " + *pv.key);
+    MiniHttpResult r = mini_http_request(host, port, "GET",
+                                         prefix + "/models", "",
+                                         cfg.health_timeout_ms, hdrs);
+    if (r.status != 200) continue;
+    Json v;
+    if (!Json::parse(r.body, v) || !v["data"].is_arr()) continue;
+    for (auto& m : v["data"].as_arr()) {
+      std::string id = m["id"].as_str();
+      if (!id.empty())
+        fresh.push_back({std::string(pv.name) + ":" + id, pv.name});
+    }
+  }
+  std::lock_guard<std::mutex> lk(cloud_models_mu);
+  cloud_models_cache = std::move(fresh);
+  cloud_models_fetched_ms = wall_ms();
+}
+
 void Gateway::health_loop() {
   pthread_setname_np(pthread_self(), "gw-health");
   std::map<std::string, int64_t> last_sync_ms;
   int64_t last_telemetry_us = 0;
+  int64_t last_cloud_ms = 0;
   while (!stopping.load()) {
     int64_t loop_start = now_us();
+    if ((!cfg.openai_base.empty() || !cfg.google_base.empty() ||
+         !cfg.anthropic_base.empty()) &&
+        wall_ms() - last_cloud_ms > 10 * 60 * 1000) {
+      refresh_cloud_models();
+      last_cloud_ms = wall_ms();
+    }
     // amdsmi poll (MI355X-native GPU telemetry -> admission control)
     if (now_us() - last_telemetry_us >
         (int64_t)(cfg.amdsmi_poll_interval_s * 1e6)) {

@@ -48,6 +48,7 @@ def _try_load() -> None:
     _lib.ll_decode_gemm.restype = ctypes.c_int
     _lib.ll_decode_gemm_v2.restype = ctypes.c_int
     _lib.ll_decode_gemm_v3.restype = ctypes.c_int
+    _lib.ll_decode_gemm_v3_ws_floats.restype = ctypes.c_longlong
 
 
 _try_load()
@@ -213,6 +214,21 @@ def _split_k_ws(N: int, M: int, device) -> torch.Tensor:
     return t
 
 
+def _v3_ws(N: int, M: int, K: int, device) -> torch.Tensor | None:
+    """Slab+ticket workspace for the v3 split-K reduce (sized by the
+    kernel library; tickets ride in the same zero-initialized buffer
+    and are reset by the reducing block, so reuse needs no memset)."""
+    n = int(_lib.ll_decode_gemm_v3_ws_floats(N, M, K))
+    if n == 0:
+        return None
+    key = ("v3", N, M, K, str(device))
+    t = _WS_CACHE.get(key)
+    if t is None:
+        t = torch.zeros(n, dtype=torch.float32, device=device)
+        _WS_CACHE[key] = t
+    return t
+
+
 # staged decode-GEMM generation: "3" (default) = software-pipelined v3
 # (double-buffered W register sets + LDS x buffers, nt weight stream);
 # "2" = the r1 single-buffered kernel, kept for A/B.
@@ -232,11 +248,12 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     N, K = x.shape
     M = weight.shape[0]
     out = torch.empty(N, M, dtype=x.dtype, device=x.device)
-    need_ws = (M // 128) < 224
-    if need_ws and ws is None:
-        ws = _split_k_ws(N, M, x.device)
-    entry = (lib.ll_decode_gemm_v3
-             if (ver or _DECODE_GEMM_VER) == "3" else lib.ll_decode_gemm_v2)
+    use_v3 = (ver or _DECODE_GEMM_VER) == "3"
+    if ws is None:
+        ws = (_v3_ws(N, M, K, x.device) if use_v3
+              else (_split_k_ws(N, M, x.device)
+                    if (M // 128) < 224 else None))
+    entry = lib.ll_decode_gemm_v3 if use_v3 else lib.ll_decode_gemm_v2
     rc = entry(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(x),
         _ptr(weight),
@@ -336,8 +353,11 @@ def swiglu_linear(gateup: torch.Tensor, weight: torch.Tensor):
         return None
     lib = _require_native()
     out = torch.empty(N, M, dtype=gateup.dtype, device=gateup.device)
-    ws = _split_k_ws(N, M, gateup.device) if (M // 128) < 224 else None
-    entry = (lib.ll_decode_gemm_v3 if _DECODE_GEMM_VER == "3"
+    use_v3 = _DECODE_GEMM_VER == "3"
+    ws = (_v3_ws(N, M, K, gateup.device) if use_v3
+          else (_split_k_ws(N, M, gateup.device)
+                if (M // 128) < 224 else None))
+    entry = (lib.ll_decode_gemm_v3 if use_v3
              else lib.ll_decode_gemm_v2)
     rc = entry(
         _ptr(out), _ptr(ws) if ws is not None else None, _ptr(gateup),

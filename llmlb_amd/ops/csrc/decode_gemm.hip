@@ -692,7 +692,78 @@ void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
                           r) * 16 + l_lo];
   }
 
-  // ---- store (same as v2) ----
+  // ---- store ----
+  if (ATOMIC && SPLIT_W == 1) {
+    // Slab + ticket split-K reduce (guide §5 sampling-GEMM recipe):
+    // every K-slice writes its fp32 slab with plain stores, publishes
+    // with an agent release + relaxed ticket, and the LAST arriver
+    // acquires, sums the SK slabs and writes the bf16 tile directly.
+    // Replaces the r1 fp32-atomicAdd accumulator + separate
+    // f32_to_bf16 pass, whose 4704 extra ~5 us launches per probe run
+    // erased the in-engine win (profiles/r02_decode_gemm_v3_ab.md).
+    // ws layout: [mblk][kslice][N_TILES*16][128] f32, then one int
+    // ticket per mblk (zero-initialized by the allocator; the reducer
+    // resets its ticket for the next call).
+    constexpr int TILE_E = N_TILES * 16 * 128;
+    float* slab =
+        out_ws + ((size_t)mblk * SK + blockIdx.y) * TILE_E;
+#pragma unroll
+    for (int t = 0; t < N_TILES; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        slab[(t * 16 + l_hi * 4 + r) * 128 + wid * 16 + l_lo] =
+            W8 ? acc[t][r] * wsc : acc[t][r];
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    const int mblocks = gridDim.x;
+    int* tickets = reinterpret_cast<int*>(
+        out_ws + (size_t)mblocks * SK * TILE_E);
+    int* flag = reinterpret_cast<int*>(smem);  // xs is dead; ONE
+                                               // __shared__ object
+    if (threadIdx.x == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      // ROCm 7.2 drops the post-wbl2 vmcnt wait when the scoreboard
+      // looks clear — restate it where the compiler cannot remove it
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      int t = __hip_atomic_fetch_add(tickets + mblk, 1,
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+      flag[0] = (t == SK - 1) ? 1 : 0;
+    }
+    __syncthreads();
+    if (flag[0] == 0) return;
+    if (threadIdx.x == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      // write-through reset so the next call's first arriver sees 0
+      __hip_atomic_store(tickets + mblk, 0, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+    }
+    __syncthreads();
+    const float* base = out_ws + (size_t)mblk * SK * TILE_E;
+    for (int e4 = threadIdx.x; e4 < TILE_E / 4; e4 += 512) {
+      f32x4 sum = {0.f, 0.f, 0.f, 0.f};
+      for (int s2 = 0; s2 < SK; ++s2)
+        sum += *reinterpret_cast<const f32x4*>(base +
+                                               (size_t)s2 * TILE_E +
+                                               (size_t)e4 * 4);
+      const int row = (e4 * 4) / 128;
+      const int col = (e4 * 4) % 128;
+      if (row < N) {
+        unsigned short h[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          union { float f; unsigned int i; } c2;
+          c2.f = sum[j];
+          unsigned int bias = 0x7FFF + ((c2.i >> 16) & 1);
+          h[j] = (unsigned short)((c2.i + bias) >> 16);
+        }
+        *reinterpret_cast<uint2*>(out + (size_t)row * M + mblk * 128 +
+                                  col) =
+            *reinterpret_cast<const uint2*>(h);
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) {
 #pragma unroll
@@ -811,7 +882,9 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
 #undef LAUNCH_V3
 #undef LAUNCH_V3S
 #undef LAUNCH_V3_
-  if (SK > 1) {
+  if (SK > 1 && split_w == 2) {
+    // env-forced BN=64 experiment still accumulates via atomics and
+    // needs the convert pass; the default slab path does not
     long long n = (long long)N * M;
     int thr = 256;
     long long blocks2 = (n + thr - 1) / thr;
@@ -820,6 +893,20 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
   }
   hipError_t e = hipGetLastError();
   return e == hipSuccess ? 0 : (int)e;
+}
+
+// workspace floats needed by ll_decode_gemm_v3 for (N, M, K): slab
+// area + one ticket int per mblk (tickets live in the same zeroed
+// buffer).  Returns 0 when no workspace is needed.
+long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
+  if (K % 256 != 0 || M % 128 != 0 || N > 128) return 0;
+  int SK = 1;
+  while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
+    SK *= 2;
+  if (SK == 1) return 0;
+  const int n_tiles = (N + 15) / 16;
+  const long long mblocks = M / 128;
+  return mblocks * SK * n_tiles * 16 * 128 + (mblocks + 3) / 4 * 4;
 }
 
 }  // extern "C"
