@@ -534,6 +534,145 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", out);
       return;
     }
+    // -- client analytics (reference api/dashboard.rs: rankings,
+    //    heatmap, per-IP drilldown, monthly rollups) ------------------
+    if (path == "/api/dashboard/clients" && method == "GET") {
+      std::vector<DbRow> rows;
+      db.exec("SELECT client_ip, COUNT(*) requests, "
+              "SUM(completion_tokens) tokens, AVG(duration_ms) avg_ms, "
+              "SUM(CASE WHEN status >= 400 THEN 1 ELSE 0 END) failures "
+              "FROM request_history WHERE ts_ms > ? GROUP BY client_ip "
+              "ORDER BY requests DESC LIMIT 50",
+              {DbParam((long long)(wall_ms() - 7LL * 86400 * 1000))},
+              &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("client_ip", Json(row[0].str()));
+        o.emplace_back("requests", Json(row[1].i));
+        o.emplace_back("tokens", Json(row[2].i));
+        o.emplace_back("avg_ms", Json(row[3].kind == DbValue::kReal
+                                          ? row[3].d
+                                          : (double)row[3].i));
+        o.emplace_back("failures", Json(row[4].i));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (path == "/api/dashboard/clients/heatmap" && method == "GET") {
+      // weekday x hour 7x24 request counts over the last 7 days
+      std::vector<DbRow> rows;
+      db.exec("SELECT CAST(strftime('%w', ts_ms/1000, 'unixepoch') AS "
+              "INTEGER) dow, CAST(strftime('%H', ts_ms/1000, "
+              "'unixepoch') AS INTEGER) hour, COUNT(*) FROM "
+              "request_history WHERE ts_ms > ? GROUP BY dow, hour",
+              {DbParam((long long)(wall_ms() - 7LL * 86400 * 1000))},
+              &rows);
+      long long grid[7][24] = {};
+      for (auto& row : rows) {
+        int d = (int)row[0].i, h = (int)row[1].i;
+        if (d >= 0 && d < 7 && h >= 0 && h < 24) grid[d][h] = row[2].i;
+      }
+      JsonArray outer;
+      for (int d = 0; d < 7; ++d) {
+        JsonArray inner;
+        for (int h = 0; h < 24; ++h) inner.emplace_back(Json(grid[d][h]));
+        outer.emplace_back(Json(std::move(inner)));
+      }
+      JsonObject o;
+      o.emplace_back("days", Json(7));
+      o.emplace_back("requests", Json(std::move(outer)));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    if (starts_with(path, "/api/dashboard/clients/") && method == "GET") {
+      std::string ip = url_decode(
+          path.substr(strlen("/api/dashboard/clients/")));
+      long long cutoff = wall_ms() - 7LL * 86400 * 1000;
+      std::vector<DbRow> tot, models, recent;
+      db.exec("SELECT COUNT(*), SUM(prompt_tokens), "
+              "SUM(completion_tokens), AVG(duration_ms), "
+              "SUM(CASE WHEN status >= 400 THEN 1 ELSE 0 END) FROM "
+              "request_history WHERE ts_ms > ? AND client_ip = ?",
+              {DbParam(cutoff), DbParam(ip)}, &tot);
+      if (tot.empty() || tot[0][0].i == 0) {
+        c->send_json_error(404, "not_found",
+                           "no requests from this client");
+        return;
+      }
+      db.exec("SELECT model, COUNT(*), SUM(completion_tokens) FROM "
+              "request_history WHERE ts_ms > ? AND client_ip = ? "
+              "GROUP BY model ORDER BY 2 DESC",
+              {DbParam(cutoff), DbParam(ip)}, &models);
+      db.exec("SELECT ts_ms, model, api_kind, status, duration_ms, "
+              "completion_tokens FROM request_history WHERE "
+              "client_ip = ? ORDER BY ts_ms DESC LIMIT 50",
+              {DbParam(ip)}, &recent);
+      JsonObject o;
+      o.emplace_back("client_ip", Json(ip));
+      {
+        JsonObject t;
+        t.emplace_back("requests", Json(tot[0][0].i));
+        t.emplace_back("prompt_tokens", Json(tot[0][1].i));
+        t.emplace_back("completion_tokens", Json(tot[0][2].i));
+        t.emplace_back("avg_ms", Json(tot[0][3].kind == DbValue::kReal
+                                          ? tot[0][3].d
+                                          : (double)tot[0][3].i));
+        t.emplace_back("failures", Json(tot[0][4].i));
+        o.emplace_back("totals", Json(std::move(t)));
+      }
+      {
+        JsonArray arr;
+        for (auto& m : models) {
+          JsonObject mo;
+          mo.emplace_back("model", Json(m[0].str()));
+          mo.emplace_back("requests", Json(m[1].i));
+          mo.emplace_back("tokens", Json(m[2].i));
+          arr.emplace_back(Json(std::move(mo)));
+        }
+        o.emplace_back("models", Json(std::move(arr)));
+      }
+      {
+        JsonArray arr;
+        for (auto& r2 : recent) {
+          JsonObject ro;
+          ro.emplace_back("ts_ms", Json(r2[0].i));
+          ro.emplace_back("model", Json(r2[1].str()));
+          ro.emplace_back("api_kind", Json(r2[2].str()));
+          ro.emplace_back("status", Json(r2[3].i));
+          ro.emplace_back("duration_ms",
+                          Json(r2[4].kind == DbValue::kReal
+                                   ? r2[4].d
+                                   : (double)r2[4].i));
+          ro.emplace_back("completion_tokens", Json(r2[5].i));
+          arr.emplace_back(Json(std::move(ro)));
+        }
+        o.emplace_back("recent", Json(std::move(arr)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    if (path == "/api/dashboard/token-stats/monthly" && method == "GET") {
+      std::vector<DbRow> rows;
+      db.exec("SELECT substr(date, 1, 7) month, SUM(prompt_tokens), "
+              "SUM(completion_tokens), SUM(requests), SUM(failures) "
+              "FROM endpoint_daily_stats GROUP BY month ORDER BY "
+              "month DESC LIMIT 24", {}, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("month", Json(row[0].str()));
+        o.emplace_back("prompt_tokens", Json(row[1].i));
+        o.emplace_back("completion_tokens", Json(row[2].i));
+        o.emplace_back("requests", Json(row[3].i));
+        o.emplace_back("failures", Json(row[4].i));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+
     // -- audit query + full-text search (reference audit_log API +
     //    migrations/026 FTS5) -----------------------------------------
     if (path == "/api/audit" && method == "GET") {

@@ -26,8 +26,8 @@ import time
 import urllib.request
 from contextlib import contextmanager
 
-from fastapi import (Depends, FastAPI, HTTPException, Query, WebSocket,
-                     WebSocketDisconnect)
+from fastapi import (Depends, FastAPI, Header, HTTPException, Query,
+                     Request, WebSocket, WebSocketDisconnect)
 from fastapi.responses import JSONResponse
 from pydantic import BaseModel
 
@@ -142,6 +142,7 @@ def create_app(db_path: str | None = None,
 
     app = FastAPI(title="llmlb control plane", docs_url=None,
                   redoc_url=None)
+    app.state.db_path = db_path
 
     @contextmanager
     def db():
@@ -179,17 +180,33 @@ def create_app(db_path: str | None = None,
     # ---- auth dependency ----
     from fastapi import Header
 
-    def auth_user(authorization: str | None = Header(None)):
-        if not authorization:
+    def auth_user(request: Request,
+                  authorization: str | None = Header(None),
+                  x_csrf_token: str | None = Header(None)):
+        token = None
+        from_cookie = False
+        if authorization:
+            token = authorization
+            if token.lower().startswith("bearer "):
+                token = token[7:]
+        else:
+            # browser cookie session (reference JWT-via-cookie path,
+            # auth/middleware.rs).  Cookie auth on state-changing
+            # methods requires the CSRF double-submit header to match
+            # the csrf claim minted at login.
+            token = request.cookies.get("llmlb_session")
+            from_cookie = token is not None
+        if not token:
             raise HTTPException(401, "missing Authorization")
-        token = authorization
-        if token.lower().startswith("bearer "):
-            token = token[7:]
         if admin_token and token == admin_token:
             return {"username": "admin-token", "role": "admin"}
         payload = jwt_decode(token, jwt_secret)
         if not payload:
             raise HTTPException(401, "invalid token")
+        if from_cookie and request.method not in ("GET", "HEAD",
+                                                  "OPTIONS"):
+            if not x_csrf_token or x_csrf_token != payload.get("csrf"):
+                raise HTTPException(403, "CSRF token missing or invalid")
         return payload
 
     def require_admin(user=Depends(auth_user)):
@@ -208,12 +225,20 @@ def create_app(db_path: str | None = None,
         if not row or not verify_password(body.password,
                                           row["password_hash"]):
             raise HTTPException(401, "invalid credentials")
+        csrf = secrets.token_hex(16)
         token = jwt_encode(
             {"sub": row["id"], "username": row["username"],
-             "role": row["role"], "exp": time.time() + 24 * 3600},
+             "role": row["role"], "csrf": csrf,
+             "exp": time.time() + 24 * 3600},
             jwt_secret)
-        return {"token": token, "role": row["role"],
-                "must_change_password": bool(row["must_change_password"])}
+        resp = JSONResponse(
+            {"token": token, "role": row["role"], "csrf_token": csrf,
+             "must_change_password": bool(row["must_change_password"])})
+        # httpOnly session cookie for browser flows; SPA keeps using
+        # the bearer token, curl/browser forms get cookie + CSRF
+        resp.set_cookie("llmlb_session", token, httponly=True,
+                        samesite="lax", max_age=24 * 3600)
+        return resp
 
     @app.post("/api/auth/change-password")
     def change_password(body: ChangePasswordBody, user=Depends(auth_user)):
@@ -368,6 +393,107 @@ def create_app(db_path: str | None = None,
                 "WHERE ts_ms > ? GROUP BY bucket ORDER BY bucket",
                 (bucket_ms, bucket_ms, cutoff)).fetchall()
         return [dict(r) for r in rows]
+
+    @app.get("/api/dashboard/clients/heatmap")
+    def client_heatmap(days: int = Query(7, le=90),
+                       client_ip: str | None = None,
+                       user=Depends(auth_user)):
+        """Weekday x hour activity heatmap (reference client-analytics
+        heatmap, api/dashboard.rs): 7x24 request counts over the
+        window, optionally for one client."""
+        cutoff = int(time.time() * 1000) - days * 86400 * 1000
+        q = ("SELECT CAST(strftime('%w', ts_ms/1000, 'unixepoch') AS "
+             "INTEGER) dow, CAST(strftime('%H', ts_ms/1000, "
+             "'unixepoch') AS INTEGER) hour, COUNT(*) requests, "
+             "SUM(completion_tokens) tokens FROM request_history "
+             "WHERE ts_ms > ?")
+        args: list = [cutoff]
+        if client_ip:
+            q += " AND client_ip = ?"
+            args.append(client_ip)
+        q += " GROUP BY dow, hour"
+        grid = [[0] * 24 for _ in range(7)]
+        tokens = [[0] * 24 for _ in range(7)]
+        with db() as conn:
+            for r in conn.execute(q, args):
+                grid[r["dow"]][r["hour"]] = r["requests"]
+                tokens[r["dow"]][r["hour"]] = r["tokens"] or 0
+        return {"days": days, "requests": grid, "tokens": tokens}
+
+    @app.get("/api/dashboard/clients/{client_ip}")
+    def client_drilldown(client_ip: str, days: int = Query(7, le=90),
+                         user=Depends(auth_user)):
+        """Per-IP drilldown (reference api/dashboard.rs per-client
+        view): totals, per-model split, hourly timeline, recent
+        requests."""
+        cutoff = int(time.time() * 1000) - days * 86400 * 1000
+        with db() as conn:
+            tot = conn.execute(
+                "SELECT COUNT(*) requests, SUM(prompt_tokens) pt, "
+                "SUM(completion_tokens) ct, AVG(duration_ms) avg_ms, "
+                "SUM(CASE WHEN status >= 400 THEN 1 ELSE 0 END) "
+                "failures FROM request_history WHERE ts_ms > ? AND "
+                "client_ip = ?", (cutoff, client_ip)).fetchone()
+            models = conn.execute(
+                "SELECT model, COUNT(*) requests, "
+                "SUM(completion_tokens) tokens FROM request_history "
+                "WHERE ts_ms > ? AND client_ip = ? GROUP BY model "
+                "ORDER BY requests DESC", (cutoff, client_ip)).fetchall()
+            hourly = conn.execute(
+                "SELECT (ts_ms / 3600000) * 3600000 bucket, COUNT(*) "
+                "requests FROM request_history WHERE ts_ms > ? AND "
+                "client_ip = ? GROUP BY bucket ORDER BY bucket",
+                (cutoff, client_ip)).fetchall()
+            recent = conn.execute(
+                "SELECT ts_ms, model, api_kind, status, duration_ms, "
+                "completion_tokens FROM request_history WHERE "
+                "client_ip = ? ORDER BY ts_ms DESC LIMIT 50",
+                (client_ip,)).fetchall()
+        if not tot or tot["requests"] == 0:
+            raise HTTPException(404, "no requests from this client")
+        return {"client_ip": client_ip, "totals": dict(tot),
+                "models": [dict(m) for m in models],
+                "hourly": [dict(h) for h in hourly],
+                "recent": [dict(r) for r in recent]}
+
+    @app.get("/api/dashboard/token-stats/monthly")
+    def token_stats_monthly(months: int = Query(12, le=36),
+                            user=Depends(auth_user)):
+        """Monthly token rollups (reference api/dashboard.rs monthly
+        token stats) aggregated from endpoint_daily_stats."""
+        with db() as conn:
+            rows = conn.execute(
+                "SELECT substr(date, 1, 7) month, "
+                "SUM(prompt_tokens) prompt_tokens, "
+                "SUM(completion_tokens) completion_tokens, "
+                "SUM(requests) requests, SUM(failures) failures "
+                "FROM endpoint_daily_stats GROUP BY month "
+                "ORDER BY month DESC LIMIT ?", (months,)).fetchall()
+        return [dict(r) for r in rows]
+
+    # ---------------- settings KV (reference db/settings.rs +
+    # dashboard settings pages) ----------------
+    @app.get("/api/settings")
+    def get_settings(user=Depends(auth_user)):
+        with db() as conn:
+            conn.execute("CREATE TABLE IF NOT EXISTS settings ("
+                         "key TEXT PRIMARY KEY, value TEXT NOT NULL)")
+            rows = conn.execute("SELECT key, value FROM settings")
+            return {r["key"]: r["value"] for r in rows}
+
+    @app.put("/api/settings")
+    def put_settings(body: dict, user=Depends(require_admin)):
+        if not isinstance(body, dict) or len(body) > 100:
+            raise HTTPException(400, "expected a flat key->value object")
+        with db() as conn:
+            conn.execute("CREATE TABLE IF NOT EXISTS settings ("
+                         "key TEXT PRIMARY KEY, value TEXT NOT NULL)")
+            for k, v in body.items():
+                conn.execute(
+                    "INSERT INTO settings (key, value) VALUES (?, ?) "
+                    "ON CONFLICT(key) DO UPDATE SET value=excluded.value",
+                    (str(k)[:128], str(v)[:4096]))
+        return {"updated": len(body)}
 
     # ---------------- TPS benchmarks (api/benchmarks.rs) ----------------
     bench_results: dict[str, dict] = {}

@@ -122,3 +122,44 @@ def test_audit_query_and_search(gw):
     assert any("dashboard" in x["path"] for x in out["entries"])
     status, body = _req("GET", gw + "/api/audit/verify", headers=admin)
     assert json.loads(body)["valid"] is True
+
+
+def test_client_analytics_dataplane(gw):
+    """C++ data-plane analytics: rankings, heatmap grid shape, per-IP
+    drilldown, monthly rollups (reference api/dashboard.rs)."""
+    admin = {"Authorization": "Bearer t"}
+    chat = {"model": "localm",
+            "messages": [{"role": "user", "content": "x"}]}
+    for _ in range(4):
+        assert _req("POST", gw + "/v1/chat/completions", chat)[0] == 200
+    time.sleep(0.6)
+    status, body = _req("GET", gw + "/api/dashboard/clients",
+                        headers=admin)
+    assert status == 200
+    clients = json.loads(body)
+    assert clients and clients[0]["client_ip"] == "127.0.0.1"
+    assert clients[0]["requests"] >= 4
+
+    status, body = _req("GET", gw + "/api/dashboard/clients/heatmap",
+                        headers=admin)
+    hm = json.loads(body)
+    assert len(hm["requests"]) == 7
+    assert all(len(r) == 24 for r in hm["requests"])
+    assert sum(sum(r) for r in hm["requests"]) >= 4
+
+    status, body = _req("GET", gw + "/api/dashboard/clients/127.0.0.1",
+                        headers=admin)
+    assert status == 200
+    d = json.loads(body)
+    assert d["totals"]["requests"] >= 4
+    assert d["models"][0]["model"] == "localm"
+    assert d["recent"]
+    assert _req("GET", gw + "/api/dashboard/clients/9.9.9.9",
+                headers=admin)[0] == 404
+
+    status, body = _req("GET",
+                        gw + "/api/dashboard/token-stats/monthly",
+                        headers=admin)
+    assert status == 200
+    months = json.loads(body)
+    assert months and months[0]["requests"] >= 4

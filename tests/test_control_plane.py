@@ -462,3 +462,91 @@ def test_tps_benchmark_runner(tmp_path):
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def _seed_history(client_db_path=None, client=None):
+    pass
+
+
+def test_analytics_heatmap_drilldown_monthly(app_client):
+    import sqlite3 as _sq
+    hdrs = _login(app_client)
+    # seed request history + daily stats through the CP's own db path
+    db_path = app_client.app.state.db_path
+    conn = _sq.connect(db_path)
+    now = int(time.time() * 1000)
+    for i in range(30):
+        conn.execute(
+            "INSERT INTO request_history (ts_ms, endpoint_id, model, "
+            "api_kind, status, duration_ms, prompt_tokens, "
+            "completion_tokens, client_ip) VALUES (?,?,?,?,?,?,?,?,?)",
+            (now - i * 3600 * 1000, "e1", "m1", "chat_completions",
+             200 if i % 5 else 500, 12.5, 10, 20,
+             "10.0.0.1" if i % 2 else "10.0.0.2"))
+    conn.execute(
+        "INSERT INTO endpoint_daily_stats VALUES "
+        "('2026-08-01','e1','m1','chat_completions',10,1,100,200,50)")
+    conn.execute(
+        "INSERT INTO endpoint_daily_stats VALUES "
+        "('2026-09-01','e1','m1','chat_completions',20,0,300,400,60)")
+    conn.commit()
+    conn.close()
+
+    r = app_client.get("/api/dashboard/clients/heatmap?days=7",
+                       headers=hdrs)
+    assert r.status_code == 200
+    hm = r.json()
+    assert len(hm["requests"]) == 7
+    assert all(len(row) == 24 for row in hm["requests"])
+    assert sum(sum(row) for row in hm["requests"]) >= 30
+
+    r = app_client.get("/api/dashboard/clients/10.0.0.1", headers=hdrs)
+    assert r.status_code == 200
+    d = r.json()
+    assert d["totals"]["requests"] >= 10
+    assert d["models"][0]["model"] == "m1"
+    assert d["hourly"] and d["recent"]
+    assert app_client.get("/api/dashboard/clients/10.9.9.9",
+                          headers=hdrs).status_code == 404
+
+    r = app_client.get("/api/dashboard/token-stats/monthly",
+                       headers=hdrs)
+    assert r.status_code == 200
+    months = {m["month"]: m for m in r.json()}
+    assert months["2026-08"]["completion_tokens"] == 200
+    assert months["2026-09"]["requests"] == 20
+
+
+def test_settings_kv(app_client):
+    hdrs = _login(app_client)
+    assert app_client.get("/api/settings", headers=hdrs).json() == {}
+    r = app_client.put("/api/settings",
+                       json={"theme": "dark", "rows": "50"},
+                       headers=hdrs)
+    assert r.status_code == 200 and r.json()["updated"] == 2
+    got = app_client.get("/api/settings", headers=hdrs).json()
+    assert got == {"theme": "dark", "rows": "50"}
+
+
+def test_cookie_session_requires_csrf(app_client):
+    """Browser cookie sessions enforce CSRF double-submit on mutating
+    requests (reference auth CSRF middleware)."""
+    r = app_client.post("/api/auth/login",
+                        json={"username": "root",
+                              "password": "rootpw123"})
+    assert r.status_code == 200
+    csrf = r.json()["csrf_token"]
+    assert "llmlb_session" in r.cookies
+    # cookie-only GET works (TestClient persists cookies)
+    assert app_client.get("/api/auth/me").status_code == 200
+    # mutating request with cookie but no CSRF header -> 403
+    r2 = app_client.put("/api/settings", json={"a": "1"})
+    assert r2.status_code == 403
+    # with the CSRF header -> accepted
+    r3 = app_client.put("/api/settings", json={"a": "1"},
+                        headers={"X-CSRF-Token": csrf})
+    assert r3.status_code == 200
+    # bearer-token requests never need CSRF
+    hdrs = _login(app_client)
+    assert app_client.put("/api/settings", json={"b": "2"},
+                          headers=hdrs).status_code == 200
