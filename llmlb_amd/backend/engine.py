@@ -100,7 +100,8 @@ class GenerationResult:
 class _Sequence:
     __slots__ = ("rid", "prompt_ids", "out_ids", "params", "block_table",
                  "length", "out_queue", "generator", "t_submit", "t_first",
-                 "row", "cached_len", "hashes", "bigram_pos", "max_new")
+                 "row", "cached_len", "hashes", "bigram_pos",
+                 "trigram_pos", "spec_k", "max_new")
 
     def __init__(self, rid, prompt_ids, params, out_queue, generator):
         self.rid = rid
@@ -117,6 +118,8 @@ class _Sequence:
         self.cached_len = 0                  # tokens reused from PrefixCache
         self.hashes: list[bytes] | None = None
         self.bigram_pos: dict | None = None  # spec: bigram -> last pos
+        self.trigram_pos: dict | None = None  # spec: trigram -> last pos
+        self.spec_k = 0                       # adaptive draft window
         self.max_new = params.max_new_tokens # clamped to max_pos at submit
 
 
@@ -730,30 +733,57 @@ class Engine:
 
     # -- prompt-lookup speculative decoding ------------------------------
     def _draft_for(self, seq: _Sequence) -> list[int]:
-        """Up to spec_ngram draft tokens: continuation after the last
-        earlier occurrence of the current trailing bigram (prompt-lookup
-        decoding).  The bigram index is maintained incrementally."""
+        """Prompt-lookup draft v2: match the trailing TRIGRAM first
+        (higher precision), fall back to the bigram, and cap the draft
+        by the sequence's ADAPTIVE window — grown on full acceptance,
+        shrunk to the measured match length on a miss — so sequences
+        in non-repetitive phases stop paying for long rejected drafts
+        (profiles/r01_spec_decode.md follow-up)."""
         ctx = seq.prompt_ids + seq.out_ids
         if len(ctx) < 3:
             return []
         if seq.bigram_pos is None:
             seq.bigram_pos = {}
+            seq.trigram_pos = {}
             for i in range(len(ctx) - 2):      # exclude trailing bigram
                 seq.bigram_pos[(ctx[i], ctx[i + 1])] = i
-        pos = seq.bigram_pos.get((ctx[-2], ctx[-1]))
+            for i in range(len(ctx) - 3):
+                seq.trigram_pos[(ctx[i], ctx[i + 1], ctx[i + 2])] = i
+            seq.spec_k = min(4, self.spec_ngram) or self.spec_ngram
+        pos = None
+        if len(ctx) >= 4:
+            tpos = seq.trigram_pos.get((ctx[-3], ctx[-2], ctx[-1]))
+            if tpos is not None:
+                pos = tpos + 3
+        if pos is None:
+            bpos = seq.bigram_pos.get((ctx[-2], ctx[-1]))
+            if bpos is not None:
+                pos = bpos + 2
         if pos is None:
             return []
-        start = pos + 2
-        return ctx[start:start + self.spec_ngram]
+        k = max(1, min(seq.spec_k or self.spec_ngram, self.spec_ngram))
+        return ctx[pos:pos + k]
+
+    def _adapt_spec_k(self, seq: _Sequence, proposed: int,
+                      accepted: int) -> None:
+        if proposed == 0:
+            return
+        if accepted == proposed:
+            seq.spec_k = min(max(seq.spec_k * 2, seq.spec_k + 1),
+                             self.spec_ngram)
+        else:
+            seq.spec_k = max(2, accepted + 1)
 
     def _index_token(self, seq: _Sequence) -> None:
-        """Advance the bigram index by the token just emitted."""
+        """Advance the n-gram indexes by the token just emitted."""
         if seq.bigram_pos is None:
             return
         ctx = seq.prompt_ids + seq.out_ids
         if len(ctx) >= 3:
             # the bigram ENDING at the previous token becomes eligible
             seq.bigram_pos[(ctx[-3], ctx[-2])] = len(ctx) - 3
+        if len(ctx) >= 4 and seq.trigram_pos is not None:
+            seq.trigram_pos[(ctx[-4], ctx[-3], ctx[-2])] = len(ctx) - 4
 
     @torch.inference_mode()
     def _spec_step(self) -> None:
@@ -861,6 +891,7 @@ class Engine:
             emit = preds[i][:m + 1]
             self.spec_proposed += len(d)
             self.spec_accepted += m
+            self._adapt_spec_k(s, len(d), m)
             s.length += 1 + m
             rows_idx.append(s.row)
             new_last.append(emit[-1])

@@ -716,31 +716,31 @@ void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     const int mblocks = gridDim.x;
+    // two counters per tile: arrivals, then observers (for the reset)
     int* tickets = reinterpret_cast<int*>(
         out_ws + (size_t)mblocks * SK * TILE_E);
-    int* flag = reinterpret_cast<int*>(smem);  // xs is dead; ONE
-                                               // __shared__ object
+    int* observed = tickets + mblocks;
     if (threadIdx.x == 0) {
       __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
       // ROCm 7.2 drops the post-wbl2 vmcnt wait when the scoreboard
       // looks clear — restate it where the compiler cannot remove it
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      int t = __hip_atomic_fetch_add(tickets + mblk, 1,
-                                     __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-      flag[0] = (t == SK - 1) ? 1 : 0;
-    }
-    __syncthreads();
-    if (flag[0] == 0) return;
-    if (threadIdx.x == 0) {
+      __hip_atomic_fetch_add(tickets + mblk, 1, __ATOMIC_RELAXED,
+                             __HIP_MEMORY_SCOPE_AGENT);
+      // EVERY slice reduces 1/SK of the tile (the last-arriver form
+      // serialized a 256 KB slab read per tile and cost ~1.2 ms/step
+      // in-engine): spin until all siblings have published
+      while (__hip_atomic_load(tickets + mblk, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT) < SK)
+        __builtin_amdgcn_s_sleep(2);
       __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-      // write-through reset so the next call's first arriver sees 0
-      __hip_atomic_store(tickets + mblk, 0, __ATOMIC_RELAXED,
-                         __HIP_MEMORY_SCOPE_AGENT);
     }
     __syncthreads();
+    // this slice reduces rows of its share across all SK slabs
     const float* base = out_ws + (size_t)mblk * SK * TILE_E;
-    for (int e4 = threadIdx.x; e4 < TILE_E / 4; e4 += 512) {
+    const int share = TILE_E / 4 / SK;  // f32x4 groups per slice
+    const int e0 = (int)blockIdx.y * share;
+    for (int e4 = e0 + threadIdx.x; e4 < e0 + share; e4 += 512) {
       f32x4 sum = {0.f, 0.f, 0.f, 0.f};
       for (int s2 = 0; s2 < SK; ++s2)
         sum += *reinterpret_cast<const f32x4*>(base +
@@ -760,6 +760,20 @@ void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
         *reinterpret_cast<uint2*>(out + (size_t)row * M + mblk * 128 +
                                   col) =
             *reinterpret_cast<const uint2*>(h);
+      }
+    }
+    // reset protocol: last block to finish observing resets both
+    // counters for the next call
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      int o = __hip_atomic_fetch_add(observed + mblk, 1,
+                                     __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+      if (o == SK - 1) {
+        __hip_atomic_store(tickets + mblk, 0, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        __hip_atomic_store(observed + mblk, 0, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
       }
     }
     return;
@@ -818,7 +832,12 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
     while ((M / 64) * SK < 224 && SK < 8 && (K / (SK * 2)) % 256 == 0)
       SK *= 2;
   } else {
-    while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
+    // grid cap 256 = the chip's one-WG-per-CU residency: the split-K
+    // reduce spins on sibling slices, and a block that cannot be
+    // resident would deadlock the spin (every block must be dispatched
+    // while its siblings spin)
+    while ((M / 128) * SK < 224 && SK < 16 &&
+           (M / 128) * SK * 2 <= 256 && (K / (SK * 2)) % 256 == 0)
       SK *= 2;
   }
   const int mblocks = M / (split_w == 2 ? 64 : 128);
@@ -901,12 +920,14 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
 long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
   if (K % 256 != 0 || M % 128 != 0 || N > 128) return 0;
   int SK = 1;
-  while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
+  while ((M / 128) * SK < 224 && SK < 16 &&
+         (M / 128) * SK * 2 <= 256 && (K / (SK * 2)) % 256 == 0)
     SK *= 2;
   if (SK == 1) return 0;
   const int n_tiles = (N + 15) / 16;
   const long long mblocks = M / 128;
-  return mblocks * SK * n_tiles * 16 * 128 + (mblocks + 3) / 4 * 4;
+  return mblocks * SK * n_tiles * 16 * 128 +
+         2 * ((mblocks + 3) / 4 * 4);
 }
 
 }  // extern "C"

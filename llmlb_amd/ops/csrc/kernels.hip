@@ -507,6 +507,10 @@ void paged_attn_decode_kernel(
 // within the workgroup), then across grid partitions by the shared
 // paged_attn_reduce_kernel.
 // ---------------------------------------------------------------------------
+// GROUP <= 4 uses 8 lanes/token x 16-dim slices; GROUP 5..8 switches
+// to 16 lanes/token x 8-dim slices so qf+acc stay ~128 VGPRs and the
+// whole group runs in ONE pass over the pages (the r1 fallback ran
+// group 8 as two GROUP=4 launches and paid the page reads twice).
 template <int GROUP, bool KV8 = false>
 __global__ __launch_bounds__(PA_WAVES * WAVE)
 void paged_attn_decode_v2_kernel(
@@ -523,7 +527,9 @@ void paged_attn_decode_v2_kernel(
     int n_q, int n_kv, int max_blocks, int n_parts, float scale,
     int q_rstride, int group_stride, int g_base) {
   constexpr int D = 128;
-  constexpr int DS = 16;                 // dim-slice (8 lanes per token)
+  constexpr int LPT = (GROUP > 4) ? 16 : 8;  // lanes per token
+  constexpr int DS = D / LPT;            // dims per lane (16 or 8)
+  constexpr int TPP = WAVE / LPT;        // tokens per pass (8 or 4)
   const int seq = blockIdx.x;
   const int kvh = blockIdx.y;
   const int part = blockIdx.z;
@@ -537,8 +543,8 @@ void paged_attn_decode_v2_kernel(
   const int per_slice = (n_blocks + slices - 1) / slices;
   const int blk_begin = slice * per_slice;
   const int blk_end = min(blk_begin + per_slice, n_blocks);
-  const int t_local = lane >> 3;         // token within half-page (0..7)
-  const int p = lane & 7;                // dim slice
+  const int t_local = lane / LPT;        // token within the pass
+  const int p = lane % LPT;              // dim slice
   const int* btab = block_tables + (size_t)seq * max_blocks;
 
   // q fragments for every head of the group, pre-scaled
@@ -549,7 +555,7 @@ void paged_attn_decode_v2_kernel(
         q + (size_t)seq * q_rstride
         + (size_t)(kvh * group_stride + g_base + g) * D + p * DS);
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
+    for (int i = 0; i < DS / 8; ++i) {
       float f[8];
       unpack8(qv[i], f);
 #pragma unroll
@@ -568,33 +574,40 @@ void paged_attn_decode_v2_kernel(
   for (int b = blk_begin; b < blk_end; ++b) {
     const int blk_id = btab[b];
 #pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const int trow = half * 8 + t_local;
+    for (int half = 0; half < PA_BLOCK / TPP; ++half) {
+      const int trow = half * TPP + t_local;
       const int tok = b * PA_BLOCK + trow;
       const bool valid = tok < seq_len;
       const size_t row = ((size_t)blk_id * n_kv + kvh) * PA_BLOCK + trow;
       float kf[DS];
       if (KV8) {
-        const uint4 raw = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const unsigned char*>(kcache) + row * D +
-            p * DS);
         const float sc = k_scale[row];
-        fp8x4_to_f32(raw.x, kf + 0);
-        fp8x4_to_f32(raw.y, kf + 4);
-        fp8x4_to_f32(raw.z, kf + 8);
-        fp8x4_to_f32(raw.w, kf + 12);
+        const unsigned char* kb =
+            reinterpret_cast<const unsigned char*>(kcache) + row * D +
+            p * DS;
+        if (DS == 16) {
+          const uint4 raw = *reinterpret_cast<const uint4*>(kb);
+          fp8x4_to_f32(raw.x, kf + 0);
+          fp8x4_to_f32(raw.y, kf + 4);
+          fp8x4_to_f32(raw.z, kf + 8);
+          fp8x4_to_f32(raw.w, kf + 12);
+        } else {
+          const uint2 raw = *reinterpret_cast<const uint2*>(kb);
+          fp8x4_to_f32(raw.x, kf + 0);
+          fp8x4_to_f32(raw.y, kf + 4);
+        }
 #pragma unroll
         for (int j = 0; j < DS; ++j) kf[j] *= sc;
       } else {
         const uint4* kv4 = reinterpret_cast<const uint4*>(
             kcache + row * D + p * DS);
-        float f[8];
-        unpack8(kv4[0], f);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kf[j] = f[j];
-        unpack8(kv4[1], f);
+        for (int i = 0; i < DS / 8; ++i) {
+          float f[8];
+          unpack8(kv4[i], f);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kf[8 + j] = f[j];
+          for (int j = 0; j < 8; ++j) kf[i * 8 + j] = f[j];
+        }
       }
       float s[GROUP];
 #pragma unroll
@@ -602,40 +615,48 @@ void paged_attn_decode_v2_kernel(
         float acc_s = 0.f;
 #pragma unroll
         for (int j = 0; j < DS; ++j) acc_s += qf[g][j] * kf[j];
-        // reduce over the 8-lane dim group (xor 1,2,4)
+        // reduce over the LPT-lane dim group
         acc_s += __shfl_xor(acc_s, 1, WAVE);
         acc_s += __shfl_xor(acc_s, 2, WAVE);
         acc_s += __shfl_xor(acc_s, 4, WAVE);
+        if (LPT == 16) acc_s += __shfl_xor(acc_s, 8, WAVE);
         s[g] = valid ? acc_s : -INFINITY;
       }
       // block max over the 8 tokens of this pass (lane bits 3..5)
       float vf[DS];
       if (KV8) {
-        const uint4 raw = *reinterpret_cast<const uint4*>(
-            reinterpret_cast<const unsigned char*>(vcache) + row * D +
-            p * DS);
         const float sc = v_scale[row];
-        fp8x4_to_f32(raw.x, vf + 0);
-        fp8x4_to_f32(raw.y, vf + 4);
-        fp8x4_to_f32(raw.z, vf + 8);
-        fp8x4_to_f32(raw.w, vf + 12);
+        const unsigned char* vb =
+            reinterpret_cast<const unsigned char*>(vcache) + row * D +
+            p * DS;
+        if (DS == 16) {
+          const uint4 raw = *reinterpret_cast<const uint4*>(vb);
+          fp8x4_to_f32(raw.x, vf + 0);
+          fp8x4_to_f32(raw.y, vf + 4);
+          fp8x4_to_f32(raw.z, vf + 8);
+          fp8x4_to_f32(raw.w, vf + 12);
+        } else {
+          const uint2 raw = *reinterpret_cast<const uint2*>(vb);
+          fp8x4_to_f32(raw.x, vf + 0);
+          fp8x4_to_f32(raw.y, vf + 4);
+        }
 #pragma unroll
         for (int j = 0; j < DS; ++j) vf[j] *= sc;
       } else {
         const uint4* vv4 = reinterpret_cast<const uint4*>(
             vcache + row * D + p * DS);
-        float f[8];
-        unpack8(vv4[0], f);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vf[j] = f[j];
-        unpack8(vv4[1], f);
+        for (int i = 0; i < DS / 8; ++i) {
+          float f[8];
+          unpack8(vv4[i], f);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vf[8 + j] = f[j];
+          for (int j = 0; j < 8; ++j) vf[i * 8 + j] = f[j];
+        }
       }
 #pragma unroll
       for (int g = 0; g < GROUP; ++g) {
         float bm = s[g];
-        bm = fmaxf(bm, __shfl_xor(bm, 8, WAVE));
+        if (LPT == 8) bm = fmaxf(bm, __shfl_xor(bm, 8, WAVE));
         bm = fmaxf(bm, __shfl_xor(bm, 16, WAVE));
         bm = fmaxf(bm, __shfl_xor(bm, 32, WAVE));
         const float m_new = fmaxf(m[g], bm);
@@ -652,9 +673,9 @@ void paged_attn_decode_v2_kernel(
       }
     }
   }
-  // reduce l and acc across the 8 token groups (butterfly, bits 3..5)
+  // reduce l and acc across the token groups (butterfly)
 #pragma unroll
-  for (int off = 8; off <= 32; off <<= 1) {
+  for (int off = LPT; off <= 32; off <<= 1) {
 #pragma unroll
     for (int g = 0; g < GROUP; ++g) {
       l[g] += __shfl_xor(l[g], off, WAVE);
@@ -899,13 +920,18 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
                      (const bf16*)vcache, nullptr, nullptr,                 \
                      (const int*)block_tables, (const int*)seq_lens, n_q,   \
                      n_kv, max_blocks, n_parts, scale, q_rstride, GS, GB)
-    // any GQA group: greedy 4/2/1 passes at the same group_stride —
-    // ceil(group/4) page reads per workgroup instead of `group` under
-    // the v1 head-per-wave layout (e.g. 70B group 8 -> 2 reads,
-    // Qwen2-7B group 7 -> 3)
+    // any GQA group: greedy 8/4/2/1 passes at the same group_stride.
+    // GROUP=8 runs SINGLE-PASS (16-lane/token layout) so the 70B
+    // shape reads each page once instead of twice (r2; the r1 +16%
+    // A/B left that amortization on the table)
+    // LLMLB_PA_MAX_G=4 forces the r1 two-pass fallback for A/B
+    static const char* mgenv = getenv("LLMLB_PA_MAX_G");
+    const int max_g = mgenv ? atoi(mgenv) : 8;
     for (int gb = 0, rem = group; rem > 0;) {
-      const int g = rem >= 4 ? 4 : (rem >= 2 ? 2 : 1);
+      int g = rem >= 8 ? 8 : (rem >= 4 ? 4 : (rem >= 2 ? 2 : 1));
+      if (g > max_g) g = max_g;
       switch (g) {
+        case 8: LAUNCH_PA_V2(8, group, gb); break;
         case 4: LAUNCH_PA_V2(4, group, gb); break;
         case 2: LAUNCH_PA_V2(2, group, gb); break;
         default: LAUNCH_PA_V2(1, group, gb); break;
@@ -971,8 +997,9 @@ int ll_paged_attn_decode_fp8(void* out, void* ws_acc, void* ws_ml,
                      (const int*)seq_lens, n_q, n_kv, max_blocks,           \
                      n_parts, scale, q_rstride, GS, GB)
   for (int gb = 0, rem = group; rem > 0;) {
-    const int g = rem >= 4 ? 4 : (rem >= 2 ? 2 : 1);
+    const int g = rem >= 8 ? 8 : (rem >= 4 ? 4 : (rem >= 2 ? 2 : 1));
     switch (g) {
+      case 8: LAUNCH_V2_FP8(8, group, gb); break;
       case 4: LAUNCH_V2_FP8(4, group, gb); break;
       case 2: LAUNCH_V2_FP8(2, group, gb); break;
       default: LAUNCH_V2_FP8(1, group, gb); break;

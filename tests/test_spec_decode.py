@@ -116,3 +116,27 @@ def test_spec_near_context_window_end():
     assert res.finish_reason in ("length", "stop")
     assert len(res.token_ids) <= 64 - 40
     assert eng.kv.free_blocks() >= 0
+
+
+def test_adaptive_draft_window_and_trigram():
+    """v2: the draft window grows on full acceptance and shrinks to the
+    observed match length on a miss; trigram matches take priority."""
+    eng = _engine(spec=8)
+    from llmlb_amd.backend.engine import _Sequence
+    from llmlb_amd.backend.sampler import SamplingParams
+    import queue as _q
+    params = SamplingParams(max_new_tokens=4, temperature=0.0)
+    seq = _Sequence(1, [5, 6, 7, 5, 6, 7, 5, 6], params, _q.Queue(),
+                    None)
+    d = eng._draft_for(seq)
+    assert d, "repetitive context must produce a draft"
+    # trigram (7,5,6) last seen at i=2 -> continuation starts at ctx[5:]
+    assert d[0] == 7
+    k0 = seq.spec_k
+    eng._adapt_spec_k(seq, proposed=k0, accepted=k0)   # full accept
+    assert seq.spec_k >= min(k0 * 2, eng.spec_ngram) or \
+        seq.spec_k == eng.spec_ngram
+    eng._adapt_spec_k(seq, proposed=seq.spec_k, accepted=0)  # miss
+    assert seq.spec_k == 2
+    d2 = eng._draft_for(seq)
+    assert len(d2) <= 2
