@@ -397,6 +397,117 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
                                : r.header("content-type"));
         return;
       }
+      // -- per-engine model adapters (reference metadata/*.rs,
+      //    download/mod.rs, delete: types/endpoint.rs:164-166) -------
+      if (starts_with(action, "models/") &&
+          ends_with(action, "/metadata") && method == "GET") {
+        std::string model = url_decode(action.substr(
+            strlen("models/"),
+            action.size() - strlen("models/") - strlen("/metadata")));
+        MiniHttpResult res;
+        if (ep->type == "ollama") {
+          res = mini_http_request(ep->host, ep->port, "POST",
+                                  ep->path_prefix + "/api/show",
+                                  "{\"name\":\"" + model + "\"}",
+                                  cfg.health_timeout_ms);
+        } else if (ep->type == "lm_studio") {
+          res = mini_http_request(ep->host, ep->port, "GET",
+                                  ep->path_prefix + "/api/v1/models",
+                                  "", cfg.health_timeout_ms);
+        } else if (ep->type == "xllm") {
+          res = mini_http_request(
+              ep->host, ep->port, "GET",
+              ep->path_prefix + "/api/models/" + model + "/metadata",
+              "", cfg.health_timeout_ms);
+        } else {
+          c->send_json_error(501, "unsupported_engine",
+                             "metadata adapter: ollama, lm_studio and "
+                             "xllm endpoints only");
+          return;
+        }
+        if (res.status != 200) {
+          c->send_json_error(res.status == 404 ? 404 : 502,
+                             "metadata_unavailable",
+                             res.error.empty() ? res.body.substr(0, 200)
+                                               : res.error);
+          return;
+        }
+        c->send_simple(200, "application/json", res.body);
+        return;
+      }
+      if (action == "models/download" && method == "POST") {
+        Json in;
+        Json::parse(body, in);
+        std::string model = in["model"].as_str();
+        if (model.empty()) model = in["name"].as_str();
+        if (model.empty()) {
+          c->send_json_error(400, "missing_model",
+                             "body must name a model");
+          return;
+        }
+        MiniHttpResult res;
+        if (ep->type == "ollama") {
+          res = mini_http_request(
+              ep->host, ep->port, "POST", ep->path_prefix + "/api/pull",
+              "{\"name\":\"" + model + "\",\"stream\":false}",
+              ep->inference_timeout_s * 1000);
+        } else if (ep->type == "xllm") {
+          res = mini_http_request(
+              ep->host, ep->port, "POST",
+              ep->path_prefix + "/api/models/download",
+              "{\"model\":\"" + model + "\"}",
+              ep->inference_timeout_s * 1000);
+        } else if (ep->type == "lm_studio") {
+          res = mini_http_request(
+              ep->host, ep->port, "POST",
+              ep->path_prefix + "/api/v1/models/download",
+              "{\"model\":\"" + model + "\"}",
+              ep->inference_timeout_s * 1000);
+        } else {
+          c->send_json_error(501, "unsupported_engine",
+                             "download adapter: ollama, lm_studio and "
+                             "xllm endpoints only");
+          return;
+        }
+        if (res.status != 200) {
+          c->send_json_error(502, "download_failed",
+                             res.error.empty() ? res.body.substr(0, 200)
+                                               : res.error);
+          return;
+        }
+        std::string serr;
+        sync_endpoint_models(ep, &serr);   // pick up the new model
+        c->send_simple(200, "application/json",
+                       "{\"downloaded\":\"" + model + "\"}");
+        return;
+      }
+      if (starts_with(action, "models/") && method == "DELETE") {
+        std::string model =
+            url_decode(action.substr(strlen("models/")));
+        if (ep->type != "ollama") {
+          // reference parity: model delete is an ollama-only
+          // capability (types/endpoint.rs:164-166)
+          c->send_json_error(501, "unsupported_engine",
+                             "model delete: ollama endpoints only");
+          return;
+        }
+        MiniHttpResult res = mini_http_request(
+            ep->host, ep->port, "DELETE",
+            ep->path_prefix + "/api/delete",
+            "{\"name\":\"" + model + "\"}", cfg.health_timeout_ms);
+        if (res.status != 200) {
+          c->send_json_error(res.status == 404 ? 404 : 502,
+                             "delete_failed",
+                             res.error.empty() ? res.body.substr(0, 200)
+                                               : res.error);
+          return;
+        }
+        std::string serr;
+        sync_endpoint_models(ep, &serr);
+        c->send_simple(200, "application/json",
+                       "{\"deleted\":\"" + model + "\"}");
+        return;
+      }
       if (action == "test" && method == "POST") {
         MiniHttpResult res = mini_http_request(
             ep->host, ep->port, "GET", ep->path_prefix + "/v1/models", "",

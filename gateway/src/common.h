@@ -143,6 +143,10 @@ std::string uuid4();
 bool secure_random_bytes(void* buf, size_t n);
 // true for 127.0.0.0/8, ::1 and their IPv4-mapped forms
 bool is_loopback_ip(std::string_view ip);
+inline bool ends_with(std::string_view s, std::string_view sfx) {
+  return s.size() >= sfx.size() &&
+         s.compare(s.size() - sfx.size(), sfx.size(), sfx) == 0;
+}
 
 // base64 / base64url
 std::string b64_encode(std::string_view in, bool url_safe = false,

@@ -30,6 +30,9 @@ struct MockConfig {
   double fail_rate = 0.0;    // inject failures (config 5 testing)
   bool include_usage = true; // --no-usage: exercise the gateway's BPE
                              // token-estimation fallback
+  std::string engine;        // --engine ollama: emulate the ollama
+                             // control surface (tags/show/pull/delete)
+                             // for the per-engine adapter tests
   int completion_tokens = 8;
 };
 static MockConfig g_cfg;
@@ -261,6 +264,63 @@ void MockConn::respond() {
   std::string path(req.path);
   auto do_respond = [this, path, bodystr = std::string(body)]() {
     std::string_view body2 = bodystr;
+    auto send_json = [this](int status, const std::string& b) {
+      append_status_line(out, status,
+                         status == 200 ? "OK" : "Not Found");
+      char h[128];
+      int n = snprintf(h, sizeof h,
+                       "Content-Type: application/json\r\n"
+                       "Content-Length: %zu\r\n\r\n", b.size());
+      out.append(h, n);
+      out.append(b);
+    };
+    if (g_cfg.engine == "ollama" && starts_with(path, "/api/")) {
+      // minimal ollama control surface (adapter tests; the reference
+      // talks to these endpoints for sync/metadata/pull/delete)
+      if (path == "/api/tags") {
+        std::string b = "{\"models\":[";
+        for (size_t i = 0; i < g_cfg.models.size(); ++i) {
+          if (i) b += ',';
+          b += "{\"name\":\"" + g_cfg.models[i] +
+               "\",\"size\":123456789}";
+        }
+        b += "]}";
+        send_json(200, b);
+      } else if (path == "/api/show") {
+        std::string name = raw_string_field(body2, "name");
+        bool have = false;
+        for (auto& m : g_cfg.models) have |= (m == name);
+        if (!have) send_json(404, "{\"error\":\"model not found\"}");
+        else
+          send_json(200,
+                    "{\"modelfile\":\"FROM " + name +
+                        "\",\"details\":{\"family\":\"llama\","
+                        "\"parameter_size\":\"8B\","
+                        "\"quantization_level\":\"Q4_K_M\"},"
+                        "\"model_info\":{\"general.parameter_count\":"
+                        "8030000000}}");
+      } else if (path == "/api/pull") {
+        std::string name = raw_string_field(body2, "name");
+        if (!name.empty()) {
+          bool have = false;
+          for (auto& m : g_cfg.models) have |= (m == name);
+          if (!have) g_cfg.models.push_back(name);
+        }
+        send_json(200, "{\"status\":\"success\"}");
+      } else if (path == "/api/delete") {
+        std::string name = raw_string_field(body2, "name");
+        auto& ms = g_cfg.models;
+        size_t before = ms.size();
+        ms.erase(std::remove(ms.begin(), ms.end(), name), ms.end());
+        send_json(before != ms.size() ? 200 : 404,
+                  before != ms.size() ? "{}"
+                                      : "{\"error\":\"not found\"}");
+      } else {
+        send_json(404, "{}");
+      }
+      flush();
+      return;
+    }
     if (path == "/v1/models" || path == "/api/v1/models") {
       std::string b = "{\"object\":\"list\",\"data\":[";
       for (size_t i = 0; i < g_cfg.models.size(); ++i) {
@@ -363,6 +423,7 @@ int main(int argc, char** argv) {
     else if (a == "--stream-chunks") g_cfg.stream_chunks = atoi(next());
     else if (a == "--fail-rate") g_cfg.fail_rate = atof(next());
     else if (a == "--no-usage") g_cfg.include_usage = false;
+    else if (a == "--engine") g_cfg.engine = next();
   }
   signal(SIGPIPE, SIG_IGN);
   std::vector<std::unique_ptr<MockWorker>> ws;
