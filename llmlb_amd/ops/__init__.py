@@ -308,17 +308,22 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor,
             return decode_gemm_v2(x, w8[0], w_scale=w8[1])
         except ValueError:
             pass
-    # measured per-shape policy (profiles/r02_decode_gemm_v3_ab.md,
-    # three interleaved probe runs): the pipelined v3 wins o_proj
-    # (+15-20%) and down (+16-20%) and ties qkv; hipBLASLt keeps the
-    # fat gate_up (M=28672) at N>16; the v1-style fragment kernel (v2
-    # entry, ver="2") stays best at tiny N.
+    # measured policy (profiles/r02_decode_gemm_v3_ab.md): the
+    # pipelined v3 beats hipBLASLt 15-20% in ISOLATED probes on
+    # o_proj/down, and the rocprof trace shows its kernels ~4% faster
+    # in-engine too — but every split-K reduction scheme tried (fp32
+    # atomics + convert pass, last-arriver slab, every-slice spin)
+    # costs more than that margin inside the captured decode step
+    # (best custom arm 5.32 vs blas 5.25 ms/step at bs64).  auto
+    # therefore keeps hipBLASLt at N>16 and the v1-style fragment
+    # kernel at tiny N where it wins outright; explicit modes force
+    # the staged kernels for experiments.
     if shape_ok:
         try:
             if N <= 16:
                 return decode_gemm_v2(x, weight, ver="2")
-            if mode != "auto" or M <= 8192:
-                return decode_gemm_v2(x, weight, ver="3")
+            if mode != "auto":
+                return decode_gemm_v2(x, weight)
         except ValueError:
             pass
     return torch.nn.functional.linear(x, weight)

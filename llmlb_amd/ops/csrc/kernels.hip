@@ -924,9 +924,13 @@ int ll_paged_attn_decode(void* out, void* ws_acc, void* ws_ml, const void* q,
     // GROUP=8 runs SINGLE-PASS (16-lane/token layout) so the 70B
     // shape reads each page once instead of twice (r2; the r1 +16%
     // A/B left that amortization on the table)
-    // LLMLB_PA_MAX_G=4 forces the r1 two-pass fallback for A/B
+    // Default two-pass (max_g 4): the single-pass GROUP=8 kernel
+    // measured 40.08 vs 39.63 ms/step on llama-3-70b bs32 ctx1024 —
+    // the second pass's K/V re-reads are L2-hot while the 16-lane
+    // layout pays ~3x the cross-lane shuffles (profiles/
+    // r02_group8_single_pass_ab.md).  LLMLB_PA_MAX_G=8 enables it.
     static const char* mgenv = getenv("LLMLB_PA_MAX_G");
-    const int max_g = mgenv ? atoi(mgenv) : 8;
+    const int max_g = mgenv ? atoi(mgenv) : 4;
     for (int gb = 0, rem = group; rem > 0;) {
       int g = rem >= 8 ? 8 : (rem >= 4 ? 4 : (rem >= 2 ? 2 : 1));
       if (g > max_g) g = max_g;
