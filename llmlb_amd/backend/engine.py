@@ -526,6 +526,14 @@ class Engine:
                 if c1 == L:
                     break
                 c0 = c1
+                # decode piggyback: one decode step for
+                # the running streams between prefill chunks, so a long
+                # prompt no longer stalls everyone's time-between-tokens
+                # for its whole multi-chunk prefill.  Intermediate chunk
+                # activations are discarded anyway (only the KV writes
+                # matter), so interleaving here is state-safe.
+                if self.running:
+                    self._decode_step()
             Ls = L - c0                      # tokens in the final chunk
             last_idx = torch.arange(1, B + 1, device=dev) * Ls - 1
             logits = self.model.final(h[last_idx], residual[last_idx])
