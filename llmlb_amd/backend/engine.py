@@ -238,6 +238,10 @@ class Engine:
         import os as _os
         self.spec_ngram = int(_os.environ.get("LLMLB_SPEC_NGRAM",
                                               str(cfg.spec_ngram)) or 0)
+        # decode piggyback at prefill-chunk boundaries (A/B knob)
+        self._piggyback = _os.environ.get(
+            "LLMLB_PREFILL_PIGGYBACK", "1").lower() not in (
+                "0", "false", "off")
         self.spec_proposed = 0
         self.spec_accepted = 0
         self.prefix = (PrefixCache(self.kv)
@@ -532,7 +536,7 @@ class Engine:
                 # for its whole multi-chunk prefill.  Intermediate chunk
                 # activations are discarded anyway (only the KV writes
                 # matter), so interleaving here is state-safe.
-                if self.running:
+                if self.running and self._piggyback:
                     self._decode_step()
             Ls = L - c0                      # tokens in the final chunk
             last_idx = torch.arange(1, B + 1, device=dev) * Ls - 1
