@@ -688,6 +688,15 @@ void Gateway::shutdown_threads() {
   for (auto& w : workers) w->loop.stop();
   if (health_thread.joinable()) health_thread.join();
   for (auto& w : workers) w->join();
+  // drain the per-worker record buffers before stopping the writer —
+  // the 200 ms flush timer may not have fired for the last requests,
+  // and losing them breaks the boot-time history/TPS reseed
+  for (auto& w : workers) {
+    if (!w->pending_records.empty())
+      history.record_batch(w->pending_records);
+    if (!w->pending_audits.empty())
+      history.audit_batch(w->pending_audits);
+  }
   history.stop();
 }
 

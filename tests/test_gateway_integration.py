@@ -606,6 +606,23 @@ def test_endpoint_persistence_across_restart():
                 break
             time.sleep(0.5)
         assert ok, f"model not routable after restart (last {status})"
+        # boot reseed (bootstrap.rs:127-159 analog): the restarted
+        # gateway rebuilds the 60-min history ring from SQLite and
+        # seeds TPS EMAs from today's daily stats — the summary shows
+        # pre-restart traffic immediately and selections are not
+        # score-blind
+        status, body = _req("GET", base + "/api/dashboard/summary",
+                            headers={"Authorization": "Bearer t"})
+        assert status == 200
+        summary = json.loads(body)
+        assert summary["last_hour"]["requests"] >= 1, \
+            "history ring not reseeded after restart"
+        found_tps = False
+        for ep in summary["endpoints"]:
+            for m in ep.get("models", []):
+                if m.get("tokens_per_sec"):
+                    found_tps = True
+        assert found_tps, "TPS EMA not reseeded from daily stats"
     finally:
         gw.terminate()
         mock.terminate()
