@@ -170,8 +170,11 @@ class _DecodeGraph:
         residual = None
         for li in range(eng.cfg.model.layers):
             q, k, v, residual = eng.model.layer_attn_prelude(
-                li, h, residual, self.lens)
-            eng._kv_append(li, k, v, self.slots)
+                li, h, residual, self.lens,
+                kv_sink=(None if eng.kv.fp8 else
+                         (eng.kv.k[li], eng.kv.v[li], self.slots)))
+            if eng.kv.fp8:
+                eng._kv_append(li, k, v, self.slots)
             attn = ops.paged_attn_decode(
                 q, eng.kv.k[li], eng.kv.v[li], self.bt, self.seq_lens,
                 eng.scale, n_parts=self.n_parts, workspace=self.ws,
@@ -698,8 +701,11 @@ class Engine:
             residual = None
             for li in range(self.cfg.model.layers):
                 q, k, v, residual = self.model.layer_attn_prelude(
-                    li, h, residual, lens)
-                self._kv_append(li, k, v, slots)
+                    li, h, residual, lens,
+                    kv_sink=(None if self.kv.fp8 else
+                             (self.kv.k[li], self.kv.v[li], slots)))
+                if self.kv.fp8:
+                    self._kv_append(li, k, v, slots)
                 attn = ops.paged_attn_decode(
                     q, self.kv.k[li], self.kv.v[li], bt, seq_lens,
                     self.scale,

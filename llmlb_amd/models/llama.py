@@ -337,8 +337,12 @@ class LlamaForCausalLM(nn.Module):
 
     def layer_attn_prelude(self, layer_idx: int, h: torch.Tensor,
                            residual: torch.Tensor | None,
-                           positions: torch.Tensor):
-        """norm -> qkv -> rope.  Returns (q, k, v, residual)."""
+                           positions: torch.Tensor, kv_sink=None):
+        """norm -> qkv -> rope.  Returns (q, k, v, residual).
+
+        ``kv_sink=(kcache, vcache, slots)`` fuses the paged-cache
+        append into the RoPE launch (decode hot path: one kernel
+        instead of two); the caller must then skip its own append."""
         layer = self.layers_[layer_idx]
         normed, residual = ops.rmsnorm(h, layer.input_norm_w,
                                        self.cfg.norm_eps, residual)
@@ -346,7 +350,12 @@ class LlamaForCausalLM(nn.Module):
             residual = h
         qkv = layer.qkv(normed)
         q, k, v = layer.split_qkv(qkv)
-        q, k = ops.rope_apply(q, k, self.cos_sin, positions)
+        if kv_sink is not None:
+            kcache, vcache, slots = kv_sink
+            ops.rope_kv_append(q, k, v, self.cos_sin, positions,
+                               kcache, vcache, slots)
+        else:
+            q, k = ops.rope_apply(q, k, self.cos_sin, positions)
         return q, k, v, residual
 
     def layer_attn_epilogue(self, layer_idx: int, attn_out: torch.Tensor,

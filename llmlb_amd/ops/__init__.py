@@ -42,6 +42,7 @@ def _try_load() -> None:
     _lib.ll_rope.restype = ctypes.c_int
     _lib.ll_swiglu.restype = ctypes.c_int
     _lib.ll_kv_append.restype = ctypes.c_int
+    _lib.ll_rope_kv_append.restype = ctypes.c_int
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
     _lib.ll_paged_attn_decode_fp8.restype = ctypes.c_int
     _lib.ll_kv_append_fp8.restype = ctypes.c_int
@@ -136,6 +137,42 @@ def rope_apply(q: torch.Tensor, k: torch.Tensor, cos_sin: torch.Tensor,
         ctypes.c_int(q.stride(0)), ctypes.c_int(k.stride(0)), _stream())
     _check(rc, "ll_rope")
     return q, k
+
+
+def rope_kv_append(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                   cos_sin: torch.Tensor, positions: torch.Tensor,
+                   kcache: torch.Tensor, vcache: torch.Tensor,
+                   slots: torch.Tensor) -> None:
+    """Fused RoPE + paged-cache append: rotates q in place, rotates k
+    in place AND into the K cache, copies v into the V cache — one
+    launch instead of the rope + kv_append pair on the decode path.
+    bf16 caches only (the fp8 path keeps the separate quantizing
+    append)."""
+    if (not q.is_cuda or q.dtype != torch.bfloat16
+            or kcache.dtype != torch.bfloat16):
+        # the reference rope returns NEW tensors; emulate the fused
+        # kernel's in-place contract by copying back into the views
+        q2, k2 = reference.rope_apply(q, k, cos_sin, positions)
+        q.copy_(q2)
+        k.copy_(k2)
+        reference.kv_append(kcache, vcache, k, v, slots,
+                            kcache.shape[2])
+        return
+    lib = _require_native()
+    assert q.stride(-1) == 1 and q.stride(1) == q.shape[2]
+    assert k.stride(-1) == 1 and k.stride(1) == k.shape[2]
+    assert v.stride(-1) == 1 and v.stride(1) == v.shape[2]
+    assert cos_sin.dtype == torch.float32
+    pos = positions.to(torch.int32)
+    slots_i = slots.to(torch.int32)
+    rc = lib.ll_rope_kv_append(
+        _ptr(q), _ptr(k), _ptr(v), _ptr(cos_sin), _ptr(pos),
+        _ptr(kcache), _ptr(vcache), _ptr(slots_i),
+        ctypes.c_int(q.shape[0]), ctypes.c_int(q.shape[1]),
+        ctypes.c_int(k.shape[1]), ctypes.c_int(q.shape[2]),
+        ctypes.c_int(kcache.shape[2]), ctypes.c_int(q.stride(0)),
+        ctypes.c_int(k.stride(0)), ctypes.c_int(v.stride(0)), _stream())
+    _check(rc, "ll_rope_kv_append")
 
 
 def swiglu(gateup: torch.Tensor) -> torch.Tensor:

@@ -216,6 +216,89 @@ __global__ void swiglu_kernel(bf16* __restrict__ out,
 // slots[t] = block_id * block_size + offset   (-1 = skip)
 // One thread per 8 contiguous elements.
 // ---------------------------------------------------------------------------
+
+// ---------------------------------------------------------------------------
+// Fused RoPE + KV append: one launch rotates q (in place), rotates k
+// (in place AND straight into the paged K cache) and copies v into the
+// paged V cache — replaces the rope_kernel + kv_append_kernel pair on
+// the decode hot path (two ~5 us launches per layer -> one).
+// Thread ranges: [0, R) = rope work (q then k heads), [R, R+V) = v copy.
+// ---------------------------------------------------------------------------
+__global__ void rope_kv_append_kernel(
+    bf16* __restrict__ q, bf16* __restrict__ k,
+    const bf16* __restrict__ v, const float* __restrict__ cos_sin,
+    const int* __restrict__ positions, bf16* __restrict__ kcache,
+    bf16* __restrict__ vcache, const int* __restrict__ slots,
+    int n_tokens, int n_q, int n_kv, int D, int block_size,
+    int q_rstride, int k_rstride, int v_rstride) {
+  const int half = D / 2;
+  const int heads = n_q + n_kv;
+  const int pairs_per_thread = 4;
+  const int tph = half / pairs_per_thread;       // threads per head
+  const long long rope_total = (long long)n_tokens * heads * tph;
+  const int nvec = D >> 3;
+  const long long v_total = (long long)n_tokens * n_kv * nvec;
+  const long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= rope_total + v_total) return;
+
+  if (idx >= rope_total) {                       // ---- v copy ----
+    const long long vi = idx - rope_total;
+    const int t = vi / (n_kv * nvec);
+    const int rem = vi % (n_kv * nvec);
+    const int h = rem / nvec;
+    const int i = rem % nvec;
+    const int slot = slots[t];
+    if (slot < 0) return;
+    const size_t dst = (((size_t)(slot / block_size) * n_kv + h) *
+                            block_size +
+                        slot % block_size) * D;
+    reinterpret_cast<uint4*>(vcache + dst)[i] =
+        reinterpret_cast<const uint4*>(
+            v + (size_t)t * v_rstride + (size_t)h * D)[i];
+    return;
+  }
+  // ---- rope (q heads, then k heads which also land in the cache) ----
+  const int t = idx / (heads * tph);
+  const int rem = idx % (heads * tph);
+  const int h = rem / tph;
+  const int j0 = (rem % tph) * pairs_per_thread;
+  const int pos = positions[t];
+  const bool is_k = h >= n_q;
+  bf16* base = is_k ? (k + (size_t)t * k_rstride +
+                       (size_t)(h - n_q) * D)
+                    : (q + (size_t)t * q_rstride + (size_t)h * D);
+  float x1[4], x2[4];
+  uint2 lo = *reinterpret_cast<uint2*>(base + j0);
+  uint2 hi = *reinterpret_cast<uint2*>(base + half + j0);
+  x1[0] = bf2f(lo.x & 0xffff); x1[1] = bf2f(lo.x >> 16);
+  x1[2] = bf2f(lo.y & 0xffff); x1[3] = bf2f(lo.y >> 16);
+  x2[0] = bf2f(hi.x & 0xffff); x2[1] = bf2f(hi.x >> 16);
+  x2[2] = bf2f(hi.y & 0xffff); x2[3] = bf2f(hi.y >> 16);
+  const float* cs = cos_sin + (size_t)pos * D;
+  float4 c = *reinterpret_cast<const float4*>(cs + j0);
+  float4 sn = *reinterpret_cast<const float4*>(cs + half + j0);
+  float o1[4], o2[4];
+  o1[0] = x1[0] * c.x - x2[0] * sn.x;  o2[0] = x2[0] * c.x + x1[0] * sn.x;
+  o1[1] = x1[1] * c.y - x2[1] * sn.y;  o2[1] = x2[1] * c.y + x1[1] * sn.y;
+  o1[2] = x1[2] * c.z - x2[2] * sn.z;  o2[2] = x2[2] * c.z + x1[2] * sn.z;
+  o1[3] = x1[3] * c.w - x2[3] * sn.w;  o2[3] = x2[3] * c.w + x1[3] * sn.w;
+  lo.x = (unsigned)f2bf(o1[0]) | ((unsigned)f2bf(o1[1]) << 16);
+  lo.y = (unsigned)f2bf(o1[2]) | ((unsigned)f2bf(o1[3]) << 16);
+  hi.x = (unsigned)f2bf(o2[0]) | ((unsigned)f2bf(o2[1]) << 16);
+  hi.y = (unsigned)f2bf(o2[2]) | ((unsigned)f2bf(o2[3]) << 16);
+  *reinterpret_cast<uint2*>(base + j0) = lo;
+  *reinterpret_cast<uint2*>(base + half + j0) = hi;
+  if (is_k) {
+    const int slot = slots[t];
+    if (slot < 0) return;
+    const size_t dst = (((size_t)(slot / block_size) * n_kv +
+                         (h - n_q)) * block_size +
+                        slot % block_size) * D;
+    *reinterpret_cast<uint2*>(kcache + dst + j0) = lo;
+    *reinterpret_cast<uint2*>(kcache + dst + half + j0) = hi;
+  }
+}
+
 __global__ void kv_append_kernel(bf16* __restrict__ kcache,
                                  bf16* __restrict__ vcache,
                                  const bf16* __restrict__ k,
@@ -857,6 +940,31 @@ int ll_swiglu(void* out, const void* gateup, long long n_tokens, int inter,
   hipLaunchKernelGGL(swiglu_kernel, dim3((unsigned)grid), dim3(block), 0,
                      (hipStream_t)stream, (bf16*)out, (const bf16*)gateup,
                      n_tokens, inter);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+
+int ll_rope_kv_append(void* q, void* k, const void* v,
+                      const void* cos_sin, const void* positions,
+                      void* kcache, void* vcache, const void* slots,
+                      int n_tokens, int n_q, int n_kv, int D,
+                      int block_size, int q_rstride, int k_rstride,
+                      int v_rstride, void* stream) {
+  if (D % 16 != 0) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  const long long rope_total =
+      (long long)n_tokens * (n_q + n_kv) * (D / 8);
+  const long long v_total = (long long)n_tokens * n_kv * (D / 8);
+  const int block = 256;
+  const long long grid = (rope_total + v_total + block - 1) / block;
+  hipLaunchKernelGGL(rope_kv_append_kernel, dim3((unsigned)grid),
+                     dim3(block), 0, s, (bf16*)q, (bf16*)k,
+                     (const bf16*)v, (const float*)cos_sin,
+                     (const int*)positions, (bf16*)kcache,
+                     (bf16*)vcache, (const int*)slots, n_tokens, n_q,
+                     n_kv, D, block_size, q_rstride, k_rstride,
+                     v_rstride);
   LL_CHECK(hipGetLastError());
   return 0;
 }

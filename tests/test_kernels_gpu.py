@@ -330,3 +330,40 @@ def test_decode_gemm_v3_fp8_weights(N, M, K):
     rel = ((o.float() - ref32).abs().max()
            / ref32.abs().max().clamp(min=1e-6)).item()
     assert rel < 2e-2, rel
+
+
+def test_rope_kv_append_fused_matches_separate():
+    """Fused RoPE+append must be bit-identical to rope_apply followed
+    by kv_append on strided qkv views (the decode hot path shape)."""
+    torch.manual_seed(9)
+    dev = "cuda:0"
+    T, nq, nkv, D, BS, nb = 24, 8, 2, 128, 16, 8
+    width = (nq + 2 * nkv) * D
+    qkv = torch.randn(T, width, dtype=torch.bfloat16, device=dev)
+    qkv2 = qkv.clone()
+
+    def views(t):
+        qs, ks, vs = t.split([nq * D, nkv * D, nkv * D], dim=-1)
+        return (qs.view(T, nq, D), ks.view(T, nkv, D),
+                vs.view(T, nkv, D))
+
+    from llmlb_amd.ops.reference import build_cos_sin_table
+    cs = build_cos_sin_table(256, D, 1e4, device=dev)
+    pos = torch.arange(T, dtype=torch.int32, device=dev)
+    slots = torch.arange(16, 16 + T, dtype=torch.int32, device=dev)
+
+    q1, k1, v1 = views(qkv)
+    kc1 = torch.zeros(nb, nkv, BS, D, dtype=torch.bfloat16, device=dev)
+    vc1 = torch.zeros_like(kc1)
+    ops.rope_apply(q1, k1, cs, pos)
+    ops.kv_append(kc1, vc1, k1, v1, slots)
+
+    q2, k2, v2 = views(qkv2)
+    kc2 = torch.zeros_like(kc1)
+    vc2 = torch.zeros_like(kc1)
+    ops.rope_kv_append(q2, k2, v2, cs, pos, kc2, vc2, slots)
+
+    assert torch.equal(q1.contiguous(), q2.contiguous())
+    assert torch.equal(k1.contiguous(), k2.contiguous())
+    assert torch.equal(kc1, kc2)
+    assert torch.equal(vc1, vc2)
