@@ -890,6 +890,36 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", "{\"draining\":false}");
       return;
     }
+    // -- settings KV (reference db/settings.rs + dashboard settings
+    //    pages) --------------------------------------------------------
+    if (path == "/api/settings" && method == "GET") {
+      std::vector<DbRow> rows;
+      db.exec("SELECT key, value FROM settings", {}, &rows);
+      JsonObject o;
+      for (auto& row : rows)
+        o.emplace_back(row[0].str(), Json(row[1].str()));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    if (path == "/api/settings" && (method == "PUT" || method == "POST")) {
+      Json in;
+      if (!Json::parse(body, in) || !in.is_obj()) {
+        c->send_json_error(400, "bad_json", "expected a key->value object");
+        return;
+      }
+      int nset = 0;
+      for (auto& [key, val] : in.as_obj()) {
+        if (key.size() > 128 || nset > 100) continue;
+        db.exec("INSERT INTO settings (key, value) VALUES (?, ?) ON "
+                "CONFLICT(key) DO UPDATE SET value=excluded.value",
+                {DbParam(key), DbParam(val.is_str() ? val.as_str()
+                                                    : val.dump())});
+        ++nset;
+      }
+      c->send_simple(200, "application/json",
+                     "{\"updated\":" + std::to_string(nset) + "}");
+      return;
+    }
     if (path == "/api/system/info" && method == "GET") {
       JsonObject o;
       o.emplace_back("version", Json("0.1.0"));

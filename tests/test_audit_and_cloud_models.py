@@ -163,3 +163,15 @@ def test_client_analytics_dataplane(gw):
     assert status == 200
     months = json.loads(body)
     assert months and months[0]["requests"] >= 4
+
+
+def test_settings_kv_dataplane(gw):
+    admin = {"Authorization": "Bearer t"}
+    status, body = _req("GET", gw + "/api/settings", headers=admin)
+    assert status == 200
+    status, body = _req("PUT", gw + "/api/settings",
+                        {"theme": "dark", "rows": "25"}, headers=admin)
+    assert status == 200 and json.loads(body)["updated"] == 2
+    status, body = _req("GET", gw + "/api/settings", headers=admin)
+    got = json.loads(body)
+    assert got["theme"] == "dark" and got["rows"] == "25"
