@@ -183,11 +183,12 @@ long long BpeTokenizer::count_piece(std::string_view piece) const {
 }
 
 const BpeTokenizer* BpeTokenizer::default_instance() {
-  static BpeTokenizer tok;
-  static bool tried = false;
-  static bool ok = false;
-  if (!tried) {
-    tried = true;
+  // magic-static init: several worker threads can hit the first
+  // estimation concurrently, so the load must be once-only and
+  // thread-safe
+  static const BpeTokenizer* inst = []() -> const BpeTokenizer* {
+    static BpeTokenizer tok;
+    bool ok = false;
     const char* env = getenv("LLMLB_TIKTOKEN_FILE");
     if (env && *env && tok.load(env)) {
       ok = true;
@@ -210,8 +211,9 @@ const BpeTokenizer* BpeTokenizer::default_instance() {
     else
       LOG_WARN("no BPE table found: token estimation falls back to "
                "chars/4 (set LLMLB_TIKTOKEN_FILE)");
-  }
-  return ok ? &tok : nullptr;
+    return ok ? &tok : nullptr;
+  }();
+  return inst;
 }
 
 }  // namespace gw
