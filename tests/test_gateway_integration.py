@@ -778,3 +778,63 @@ def test_anthropic_unicode_roundtrip(stack):
     assert out["type"] == "message"
     assert out["content"][0]["type"] == "text"
     assert out["usage"]["input_tokens"] > 0
+
+
+def test_delayed_admission_between_50_and_80pct():
+    """LLMLB_QUEUE_MAX=4 with 2 slots held (50%): the next request is
+    HELD 10-100 ms on the event loop and then served (reference
+    admission_control AcceptWithDelay), visible both in its latency
+    and in the llmlb_admission_delays_total counter."""
+    mock_port, gw_port = _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    slow = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(mock_port), "--workers",
+         "1", "--models", "dm", "--delay-ms", "4000",
+         "--delay-inference-only"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t", "LLMLB_QUEUE_MAX": "4",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        for _ in range(50):
+            try:
+                _req("GET", base + "/health", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        status, _ = _req("POST", base + "/api/endpoints",
+                         {"name": "d",
+                          "base_url": f"http://127.0.0.1:{mock_port}"},
+                         headers={"Authorization": "Bearer t"})
+        assert status == 201
+        import threading
+        body = {"model": "dm",
+                "messages": [{"role": "user", "content": "hi"}]}
+        holders = [threading.Thread(
+            target=lambda: _req("POST", base + "/v1/chat/completions",
+                                body, timeout=30)) for _ in range(2)]
+        for h in holders:
+            h.start()
+        time.sleep(1.0)   # 2/4 active = 50% -> delay band
+        status, _ = _req("POST", base + "/v1/chat/completions", body,
+                         timeout=30)
+        assert status == 200
+        status, text = _req("GET", base + "/api/metrics")
+        delays = [ln for ln in text.splitlines()
+                  if ln.startswith("llmlb_admission_delays_total")]
+        assert delays and int(delays[0].split()[-1]) >= 1, delays
+        for h in holders:
+            h.join(timeout=30)
+    finally:
+        gw.terminate()
+        slow.terminate()
+        for p in (gw, slow):
+            try:
+                p.wait(timeout=5)
+            except subprocess.TimeoutExpired:
+                p.kill()
