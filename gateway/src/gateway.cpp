@@ -147,10 +147,28 @@ void UpstreamConn::close_now() {
     if (o->client) {
       ClientConn* c = o->client;
       if (!o->resp_head_seen) {
-        c->w->gw->metrics.upstream_errors.fetch_add(1);
-        c->send_json_error(502, "endpoint_upstream_error",
-                           "upstream connection failed");
-        c->w->gw->finalize_op(c, false, 502, "upstream_conn_failed");
+        // TCP connect never completed on a TPS-selected endpoint:
+        // retry ONCE against the next-best endpoint (the dead one
+        // keeps the top TPS score until the health loop ejects it;
+        // the request bytes are still buffered).  The reference
+        // surfaces the 502 instead (openai.rs:1006-1079) — this is a
+        // deliberate availability improvement: a killed backend costs
+        // clients one extra connect, not an error, during the
+        // eject window.
+        if (!connected && o->selected && c->retry_exclude_ep.empty() &&
+            !o->endpoint_id.empty()) {
+          c->retry_exclude_ep = o->endpoint_id;
+          LOG_WARN("connect to endpoint %s failed; retrying selection "
+                   "without it", o->endpoint_id.c_str());
+          c->w->gw->metrics.upstream_errors.fetch_add(1);
+          c->w->gw->abandon_op_for_retry(c);
+          c->process_requests();
+        } else {
+          c->w->gw->metrics.upstream_errors.fetch_add(1);
+          c->send_json_error(502, "endpoint_upstream_error",
+                             "upstream connection failed");
+          c->w->gw->finalize_op(c, false, 502, "upstream_conn_failed");
+        }
       } else if (o->streaming) {
         if (o->body_reader.eof_terminated()) {
           // EOF-terminated stream: this IS normal completion
@@ -372,8 +390,8 @@ void ClientConn::dispatch() {
   // request tracing: honour a client-supplied X-Request-Id, else mint
   // one; echoed on the response and forwarded upstream
   std::string_view rid = req.header("x-request-id");
-  if (adm_delayed && !request_id.empty()) {
-    // re-dispatch after a delayed-admission hold: keep the minted id
+  if ((adm_delayed || !retry_exclude_ep.empty()) && !request_id.empty()) {
+    // re-dispatch (delayed-admission hold / connect retry): keep the id
   } else if (!rid.empty() && rid.size() <= 64) {
     request_id.assign(rid);
   } else {
@@ -395,6 +413,8 @@ void ClientConn::finish_request_cycle() {
   size_t need = req.header_len + (size_t)req.content_length;
   in.consume(need);
   expect_ack_sent = false;
+  adm_delayed = false;
+  retry_exclude_ep.clear();
   req.headers.clear();  // keep vector capacity across requests
   req.header_len = 0;
   req.content_length = 0;

@@ -146,6 +146,7 @@ struct ProxyOp {
   UpstreamConn* upstream = nullptr;
   RequestLease lease;
   BodyReader body_reader;
+  bool selected = false;      // came through TPS selection (retryable)
   SseTap tap;
   HttpResponse resp;          // views valid only at head-parse time
   std::string resp_content_type;  // copied at head time (views dangle)
@@ -225,6 +226,9 @@ class ClientConn : public FdHandler {
   uint32_t events = 0;
   bool response_open = false;  // streaming response in progress
   bool adm_delayed = false;    // delayed-admission hold already served
+  // one-shot connect-failure retry: the endpoint id to exclude on the
+  // re-selection (non-empty also means "already retried once")
+  std::string retry_exclude_ep;
 
   void flush_out();            // try write; arm EPOLLOUT if partial
   void close_now();
@@ -337,6 +341,10 @@ class Gateway {
   void start_control_plane_proxy(ClientConn* c);
   void finalize_op(ClientConn* c, bool success, int status,
                    const std::string& error);
+  // teardown for the connect-failure retry: releases the lease and
+  // counters but KEEPS the request buffered (finalize_op would consume
+  // it and finish the cycle)
+  void abandon_op_for_retry(ClientConn* c);
 
   // health checking — health.cpp
   void health_loop();

@@ -97,7 +97,7 @@ static bool build_upstream_body(std::string_view body,
 // Request dispatch
 // -------------------------------------------------------------------
 void Gateway::handle_request(ClientConn* c) {
-  if (!c->adm_delayed)   // parked re-dispatch counts once
+  if (!c->adm_delayed && c->retry_exclude_ep.empty())
     metrics.requests_total.fetch_add(1, std::memory_order_relaxed);
   const HttpRequest& r = c->req;
   std::string_view body =
@@ -223,7 +223,8 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
 
   auto table = registry.table();
   SelectResult sel = select_endpoint(*table, model, kind, cfg.admission,
-                                     wall_ms(), required_caps);
+                                     wall_ms(), required_caps,
+                                     &c->retry_exclude_ep);
   if (sel.err_status != 0) {
     const char* msg =
         sel.err_status == 404 ? "model not found on any online endpoint"
@@ -279,6 +280,7 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
     op->captured_request = std::string(body.substr(0, 16384));
   op->req_prompt_chars = (long long)body.size();
   op->tps = sel.tps;
+  op->selected = true;
   op->lease = RequestLease(sel.ep, &registry, sel.canonical, kind);
   op->lease.set_tps(sel.tps);
   c->op = std::move(op);
@@ -667,6 +669,16 @@ static void respond_buffered(Gateway* gw, ClientConn* c) {
 // -------------------------------------------------------------------
 // Op finalization: history + TPS + cleanup, then resume the client
 // -------------------------------------------------------------------
+void Gateway::abandon_op_for_retry(ClientConn* c) {
+  std::unique_ptr<ProxyOp> op = std::move(c->op);
+  if (!op) return;
+  c->w->timed_ops.erase(c);
+  if (op->timer_id) c->w->loop.cancel_timer(op->timer_id);
+  op->upstream = nullptr;   // the upstream conn is tearing itself down
+  metrics.active_requests.fetch_sub(1);
+  if (op->lease.active()) op->lease.complete_failure();
+}
+
 void Gateway::finalize_op(ClientConn* c, bool success, int status,
                           const std::string& error) {
   std::unique_ptr<ProxyOp> op = std::move(c->op);

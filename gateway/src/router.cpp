@@ -378,7 +378,8 @@ void Registry::seed_tps_from_db() {
 SelectResult select_endpoint(const RouteTable& table,
                              const std::string& model, ApiKind kind,
                              const AdmissionConfig& adm, int64_t now_ms,
-                             uint32_t required_caps) {
+                             uint32_t required_caps,
+                             const std::string* exclude) {
   SelectResult res;
   auto it = table.by_model.find(model);
   if (it == table.by_model.end()) {
@@ -405,6 +406,7 @@ SelectResult select_endpoint(const RouteTable& table,
     if (ep->status.load(std::memory_order_relaxed) !=
         EndpointStatus::kOnline)
       continue;
+    if (exclude && !exclude->empty() && ep->id == *exclude) continue;
     any_online = true;
     int active = ep->active.load(std::memory_order_relaxed);
     double frac = (double)active / adm.max_concurrent_per_endpoint;

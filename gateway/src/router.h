@@ -313,9 +313,13 @@ struct SelectResult {
 };
 
 // TPS-weighted selection with RR tie-break + admission control.
+// exclude: endpoint id to skip (single retry after a connect failure
+// — the dead endpoint keeps the best TPS score until the health loop
+// ejects it, so without exclusion a retry would re-pick it)
 SelectResult select_endpoint(const RouteTable& table,
                              const std::string& model, ApiKind kind,
                              const AdmissionConfig& adm, int64_t now_ms,
-                             uint32_t required_caps = 0);
+                             uint32_t required_caps = 0,
+                             const std::string* exclude = nullptr);
 
 }  // namespace gw

@@ -838,3 +838,67 @@ def test_delayed_admission_between_50_and_80pct():
                 p.wait(timeout=5)
             except subprocess.TimeoutExpired:
                 p.kill()
+
+
+def test_connect_failure_retries_next_endpoint():
+    """A killed backend whose TPS score still tops selection must not
+    surface errors: the gateway retries the request once against the
+    next-best endpoint (deliberate improvement over the reference,
+    which 502s until the health eject)."""
+    m1, m2, gw_port = _free_port(), _free_port(), _free_port()
+    db = tempfile.mktemp(suffix=".sqlite3")
+    fast = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(m1), "--workers", "1",
+         "--models", "ha-model"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    backup = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(m2), "--workers", "1",
+         "--models", "ha-model"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    env = os.environ.copy()
+    env.update({"LLMLB_PORT": str(gw_port), "LLMLB_DATABASE_URL": db,
+                "LLMLB_ADMIN_TOKEN": "t",
+                "LLMLB_HEALTH_CHECK_INTERVAL": "60"})
+    gw = subprocess.Popen([str(BIN / "llmlb-gw"), "--workers", "1"],
+                          env=env, stdout=subprocess.DEVNULL,
+                          stderr=subprocess.DEVNULL)
+    base = f"http://127.0.0.1:{gw_port}"
+    try:
+        for _ in range(50):
+            try:
+                _req("GET", base + "/health", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        for name, port in (("fast", m1), ("backup", m2)):
+            status, _ = _req("POST", base + "/api/endpoints",
+                             {"name": name,
+                              "base_url": f"http://127.0.0.1:{port}"},
+                             headers={"Authorization": "Bearer t"})
+            assert status == 201
+        body = {"model": "ha-model",
+                "messages": [{"role": "user", "content": "x"}]}
+        # earn the first endpoint a TPS score so it keeps winning
+        for _ in range(6):
+            assert _req("POST", base + "/v1/chat/completions",
+                        body)[0] == 200
+        fast.terminate()
+        fast.wait(timeout=5)
+        # health interval is 60 s: no eject will save us — every
+        # request must succeed via the transparent connect retry
+        for i in range(10):
+            status, resp = _req("POST", base + "/v1/chat/completions",
+                                body, timeout=10)
+            assert status == 200, (i, status, resp)
+        # streaming too
+        status, resp = _req("POST", base + "/v1/chat/completions",
+                            dict(body, stream=True), timeout=10)
+        assert status == 200
+    finally:
+        gw.terminate()
+        backup.terminate()
+        for p in (gw, fast, backup):
+            try:
+                p.wait(timeout=5)
+            except Exception:
+                pass
