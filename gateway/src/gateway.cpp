@@ -139,6 +139,17 @@ void UpstreamConn::want(uint32_t ev) {
 void UpstreamConn::close_now() {
   if (closed) return;
   closed = true;
+  // a conn closing while IDLE in the keep-alive pool (peer closed /
+  // RST) must leave the pool NOW: deferred deletion runs on the next
+  // loop turn, and acquire_upstream popping a freed pointer was a
+  // use-after-free (caught by the r2 TSAN pass)
+  if (!op && !pool_key.empty()) {
+    auto it = w->pool.find(pool_key);
+    if (it != w->pool.end()) {
+      auto& v = it->second;
+      v.erase(std::remove(v.begin(), v.end(), this), v.end());
+    }
+  }
   if (op) {
     // upstream died with an active op: surface to the client
     ProxyOp* o = op;

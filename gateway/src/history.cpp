@@ -31,9 +31,18 @@ void HistoryWriter::loop() {
     std::vector<AuditRecord> audits;
     {
       std::unique_lock<std::mutex> lk(mu_);
-      cv_.wait_for(lk, std::chrono::milliseconds(500), [this] {
-        return stop_ || queue_.size() > 512 || !audit_queue_.empty();
-      });
+      // system_clock wait_until (NOT steady-clock wait_for): libstdc++
+      // lowers wait_for to pthread_cond_clockwait, which this image's
+      // libtsan does not intercept — TSAN then believes the writer
+      // holds mu_ forever and cascades hundreds of false lock reports
+      // over every worker (r2 sanitizer pass).  Semantics unchanged.
+      cv_.wait_until(lk,
+                     std::chrono::system_clock::now() +
+                         std::chrono::milliseconds(500),
+                     [this] {
+                       return stop_ || queue_.size() > 512 ||
+                              !audit_queue_.empty();
+                     });
       recs.assign(queue_.begin(), queue_.end());
       queue_.clear();
       audits.assign(audit_queue_.begin(), audit_queue_.end());

@@ -77,7 +77,12 @@ int main(int argc, char** argv) {
     }
   }
 
-  gw::Gateway gateway(cfg);
+  // heap-allocate (never freed: process-lifetime object).  On the
+  // stack, TSAN conflated the HistoryWriter mutex address with prior
+  // stack frames and cascaded false lock reports; heap placement also
+  // matches the object's actual lifetime.
+  auto* gateway_ptr = new gw::Gateway(cfg);
+  gw::Gateway& gateway = *gateway_ptr;
   g_gw = &gateway;
   std::string err;
   if (!gateway.init(&err)) {
