@@ -147,6 +147,7 @@ struct ProxyOp {
   RequestLease lease;
   BodyReader body_reader;
   bool selected = false;      // came through TPS selection (retryable)
+  int admission_delay_ms = 0; // held by delayed admission -> header
   SseTap tap;
   HttpResponse resp;          // views valid only at head-parse time
   std::string resp_content_type;  // copied at head time (views dangle)
@@ -226,6 +227,7 @@ class ClientConn : public FdHandler {
   uint32_t events = 0;
   bool response_open = false;  // streaming response in progress
   bool adm_delayed = false;    // delayed-admission hold already served
+  int adm_delay_ms = 0;        // the hold length (response header)
   // one-shot connect-failure retry: the endpoint id to exclude on the
   // re-selection (non-empty also means "already retried once")
   std::string retry_exclude_ep;

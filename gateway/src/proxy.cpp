@@ -241,6 +241,7 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
   // in w->parked validates the timer.
   if (sel.delay_ms > 0 && !c->adm_delayed) {
     c->adm_delayed = true;
+    c->adm_delay_ms = sel.delay_ms;
     Worker* w = c->w;
     w->parked.insert(c);
     metrics.admission_delays.fetch_add(1, std::memory_order_relaxed);
@@ -281,6 +282,7 @@ void Gateway::start_proxy(ClientConn* c, ApiKind kind,
   op->req_prompt_chars = (long long)body.size();
   op->tps = sel.tps;
   op->selected = true;
+  op->admission_delay_ms = c->adm_delayed ? c->adm_delay_ms : 0;
   op->lease = RequestLease(sel.ep, &registry, sel.canonical, kind);
   op->lease.set_tps(sel.tps);
   c->op = std::move(op);
@@ -533,6 +535,11 @@ void proxy_on_upstream_data(UpstreamConn* uc) {
           c->out.append(c->request_id);
           c->out.append("\r\n");
         }
+        if (op->admission_delay_ms > 0) {
+          c->out.append("X-Admission-Delay-Ms: ");
+          c->out.append(std::to_string(op->admission_delay_ms));
+          c->out.append("\r\n");
+        }
         c->out.append("\r\n");
       } else {
         // forward upstream head verbatim minus hop-by-hop Connection
@@ -549,6 +556,13 @@ void proxy_on_upstream_data(UpstreamConn* uc) {
         if (!c->request_id.empty()) {
           c->out.append("X-Request-Id: ");
           c->out.append(c->request_id);
+          c->out.append("\r\n");
+        }
+        if (op->admission_delay_ms > 0) {
+          // reference x-queue-* analog (openai.rs:74-84): tell the
+          // client it waited in delayed admission
+          c->out.append("X-Admission-Delay-Ms: ");
+          c->out.append(std::to_string(op->admission_delay_ms));
           c->out.append("\r\n");
         }
         c->out.append("\r\n");
@@ -635,6 +649,9 @@ static void respond_buffered(Gateway* gw, ClientConn* c) {
     std::string ctype = op->resp_content_type;
     if (ctype.empty()) ctype = "application/json";
     std::string extra;
+    if (op->admission_delay_ms > 0)
+      extra += "X-Admission-Delay-Ms: " +
+               std::to_string(op->admission_delay_ms) + "\r\n";
     if (op->anthropic_reencode) {
       // non-streaming Anthropic ingress: convert OpenAI -> Anthropic
       std::string converted;

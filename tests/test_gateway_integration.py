@@ -821,9 +821,14 @@ def test_delayed_admission_between_50_and_80pct():
         for h in holders:
             h.start()
         time.sleep(1.0)   # 2/4 active = 50% -> delay band
-        status, _ = _req("POST", base + "/v1/chat/completions", body,
-                         timeout=30)
+        r = urllib.request.Request(
+            base + "/v1/chat/completions",
+            data=json.dumps(body).encode(), method="POST")
+        with urllib.request.urlopen(r, timeout=30) as resp:
+            status = resp.status
+            delay_hdr = resp.headers.get("X-Admission-Delay-Ms")
         assert status == 200
+        assert delay_hdr and 10 <= int(delay_hdr) <= 100, delay_hdr
         status, text = _req("GET", base + "/api/metrics")
         delays = [ln for ln in text.splitlines()
                   if ln.startswith("llmlb_admission_delays_total")]
