@@ -672,9 +672,16 @@ def create_app(db_path: str | None = None,
         return {"lines": out[-lines:]}
 
     # ---------------- model registry metadata (registry::models) -----
+    def _ensure_models_table(conn):
+        conn.execute(
+            "CREATE TABLE IF NOT EXISTS models (id TEXT PRIMARY KEY, "
+            "name TEXT, source TEXT, capabilities TEXT, "
+            "size_bytes INTEGER, created_at TEXT)")
+
     @app.get("/api/models")
     def list_registered_models(user=Depends(auth_user)):
         with db() as conn:
+            _ensure_models_table(conn)
             rows = conn.execute(
                 "SELECT id,name,source,capabilities,size_bytes,created_at"
                 " FROM models").fetchall()
@@ -683,6 +690,7 @@ def create_app(db_path: str | None = None,
     @app.post("/api/models", status_code=201)
     def register_model(body: ModelRegBody, user=Depends(require_admin)):
         with db() as conn:
+            _ensure_models_table(conn)
             conn.execute(
                 "INSERT OR REPLACE INTO models (id,name,source,"
                 "capabilities,size_bytes,created_at) VALUES (?,?,?,?,?,?)",

@@ -550,3 +550,39 @@ def test_cookie_session_requires_csrf(app_client):
     hdrs = _login(app_client)
     assert app_client.put("/api/settings", json={"b": "2"},
                           headers=hdrs).status_code == 200
+
+
+CP_CASES = [
+    ("GET", "/api/auth/me", None, True, 200, {"username", "role"}),
+    ("GET", "/api/users", None, True, 200, None),
+    ("GET", "/api/invitations", None, True, 200, None),
+    ("GET", "/api/audit", None, True, 200, None),
+    ("GET", "/api/dashboard/token-stats", None, True, 200, None),
+    ("GET", "/api/dashboard/token-stats/monthly", None, True, 200,
+     None),
+    ("GET", "/api/dashboard/clients", None, True, 200, None),
+    ("GET", "/api/dashboard/clients/heatmap", None, True, 200,
+     {"requests"}),
+    ("GET", "/api/dashboard/timeline", None, True, 200, None),
+    ("GET", "/api/settings", None, True, 200, None),
+    ("GET", "/api/models", None, True, 200, None),
+    ("GET", "/api/catalog/search?q=llama", None, True, 200, None),
+    ("GET", "/api/logs", None, True, 200, {"lines"}),
+    ("GET", "/api/update/status", None, True, 200, None),
+    ("GET", "/api/users", None, False, 401, None),
+    ("GET", "/api/auth/me", None, False, 401, None),
+]
+
+
+@pytest.mark.parametrize("method,path,body,authed,want,keys", CP_CASES,
+                         ids=[f"{m} {p} -> {w}"
+                              for m, p, b, a, w, keys in CP_CASES])
+def test_cp_route_contract(app_client, method, path, body, authed,
+                           want, keys):
+    hdrs = _login(app_client) if authed else {}
+    r = app_client.request(method, path,
+                           json=body if body is not None else None,
+                           headers=hdrs)
+    assert r.status_code == want, (r.status_code, r.text[:200])
+    if want == 200 and keys:
+        assert keys <= set(r.json()), r.text[:200]
