@@ -73,6 +73,13 @@ def main():
         def v3_fp8(a, b):
             return ops.decode_gemm_v2(a, w8, w_scale=w8s, ver="3")
 
+        from llmlb_amd.ops import reference as _R
+        xq, xs = _R.quantize_rows_fp8(x)
+        xq, xs = xq.to(x.device), xs.to(x.device)
+
+        def w8a8(a, b):
+            return ops.decode_gemm_w8a8(xq, xs, w8, w8s)
+
         ref = torch.nn.functional.linear(x, w).float()
         t_blas = bench_one(torch.nn.functional.linear, x, w)
         res = {"shape": f"N{N} M{M} K{K}",
@@ -80,7 +87,8 @@ def main():
         wbytes = M * K * 2
         res["hipblaslt_tbps"] = round(wbytes / t_blas / 1e12, 2)
         for name, fn, wb in (("v2", v2, wbytes), ("v3", v3, wbytes),
-                             ("v3_fp8", v3_fp8, wbytes // 2)):
+                             ("v3_fp8", v3_fp8, wbytes // 2),
+                             ("w8a8", w8a8, wbytes // 2)):
             try:
                 t = bench_one(fn, x, w)
                 d = (fn(x, w).float() - ref).abs()

@@ -344,11 +344,25 @@ class LlamaForCausalLM(nn.Module):
         append into the RoPE launch (decode hot path: one kernel
         instead of two); the caller must then skip its own append."""
         layer = self.layers_[layer_idx]
-        normed, residual = ops.rmsnorm(h, layer.input_norm_w,
-                                       self.cfg.norm_eps, residual)
-        if residual is None:
-            residual = h
-        qkv = layer.qkv(normed)
+        qkv = None
+        if (ops.w8a8_active() and h.is_cuda and h.dim() == 2
+                and h.shape[0] <= 128 and h.dtype == torch.bfloat16):
+            # W8A8 decode arm: the rmsnorm emits the quantized row +
+            # per-token scale directly; the fp8 MFMA GEMM consumes it
+            nq, xs_, residual = ops.rmsnorm_q(h, layer.input_norm_w,
+                                              self.cfg.norm_eps,
+                                              residual)
+            if residual is None:
+                residual = h
+            qkv = layer.qkv.forward_w8a8(nq, xs_)
+            if qkv is not None and layer.qkv.bias is not None:
+                qkv = qkv + layer.qkv.bias
+        if qkv is None:
+            normed, residual = ops.rmsnorm(h, layer.input_norm_w,
+                                           self.cfg.norm_eps, residual)
+            if residual is None:
+                residual = h
+            qkv = layer.qkv(normed)
         q, k, v = layer.split_qkv(qkv)
         if kv_sink is not None:
             kcache, vcache, slots = kv_sink
@@ -364,6 +378,18 @@ class LlamaForCausalLM(nn.Module):
         layer = self.layers_[layer_idx]
         T = attn_out.shape[0]
         o = layer.o_proj(attn_out.reshape(T, -1))
+        if (ops.w8a8_active() and o.is_cuda and T <= 128
+                and o.dtype == torch.bfloat16):
+            nq, xs_, residual = ops.rmsnorm_q(o, layer.post_norm_w,
+                                              self.cfg.norm_eps,
+                                              residual)
+            gateup = layer.gate_up.forward_w8a8(nq, xs_)
+            if gateup is not None:
+                h = layer.down.forward_swiglu_w8a8(gateup)
+                if h is not None:
+                    return h, residual
+                # fall back mid-path: finish with bf16 swiglu+down
+                return layer.down.forward(ops.swiglu(gateup)), residual
         normed, residual = ops.rmsnorm(o, layer.post_norm_w,
                                        self.cfg.norm_eps, residual)
         gateup = layer.gate_up(normed)

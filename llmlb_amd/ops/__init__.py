@@ -43,6 +43,10 @@ def _try_load() -> None:
     _lib.ll_swiglu.restype = ctypes.c_int
     _lib.ll_kv_append.restype = ctypes.c_int
     _lib.ll_rope_kv_append.restype = ctypes.c_int
+    _lib.ll_rmsnorm_q.restype = ctypes.c_int
+    _lib.ll_swiglu_q.restype = ctypes.c_int
+    _lib.ll_decode_gemm_w8a8.restype = ctypes.c_int
+    _lib.ll_decode_gemm_w8a8_ws_floats.restype = ctypes.c_longlong
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
     _lib.ll_paged_attn_decode_fp8.restype = ctypes.c_int
     _lib.ll_kv_append_fp8.restype = ctypes.c_int
@@ -303,13 +307,89 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
     return out
 
 
-# EXPERIMENTAL: measured SLOWER than bf16+hipBLASLt on the current v2
-# kernel structure (the e4m3->bf16 converts sit on the load-latency
-# critical path and outweigh the halved weight traffic —
-# profiles/r01_fp8_weights_ab.md).  Kept for the round-2 pipelined
-# staging kernel; default OFF.
+# EXPERIMENTAL: W8A16 measured SLOWER than bf16+hipBLASLt (the
+# e4m3->bf16 converts sit on the MFMA operand path —
+# profiles/r01_fp8_weights_ab.md).  Kept for A/B; default OFF.
 _FP8_WEIGHTS = os.environ.get("LLMLB_FP8_WEIGHTS",
                               "0").lower() in ("1", "true", "on")
+
+# W8A8: BOTH operands e4m3, consumed directly by the gfx950 fp8 MFMA
+# (no convert chain); activations are quantized per-token INSIDE the
+# producing kernels (rmsnorm_q / swiglu_q).  Implies fp8 weights.
+# Opt-in: reduced model precision is a capacity/throughput trade the
+# operator must choose (the bf16 default serves the BASELINE bench).
+_FP8_A8 = os.environ.get("LLMLB_FP8_A8",
+                         "0").lower() in ("1", "true", "on")
+
+
+def w8a8_active() -> bool:
+    return _FP8_A8 and _lib is not None
+
+
+def rmsnorm_q(x: torch.Tensor, weight: torch.Tensor, eps: float,
+              residual: torch.Tensor | None = None):
+    """Fused rmsnorm + per-token e4m3 quantization: returns
+    (q u8 [T, H], scale f32 [T], residual').  GPU bf16 only; the CPU
+    reference path normalizes then simulates e4m3 via torch float8."""
+    if not x.is_cuda or x.dtype != torch.bfloat16:
+        out, residual = reference.rmsnorm(x, weight, eps, residual)
+        q, sc = reference.quantize_rows_fp8(out)
+        return q, sc, residual
+    lib = _require_native()
+    T, H = x.shape[0], x.shape[-1]
+    q = torch.empty(T, H, dtype=torch.uint8, device=x.device)
+    sc = torch.empty(T, dtype=torch.float32, device=x.device)
+    rc = lib.ll_rmsnorm_q(
+        _ptr(q), _ptr(sc), _ptr(x), _ptr(weight),
+        _ptr(residual) if residual is not None else None,
+        ctypes.c_int(T), ctypes.c_int(H), ctypes.c_float(eps), _stream())
+    _check(rc, "ll_rmsnorm_q")
+    return q, sc, residual
+
+
+def swiglu_q(gateup: torch.Tensor):
+    """Fused silu(gate)*up + per-token e4m3 quantization: returns
+    (q u8 [T, inter], scale f32 [T])."""
+    if not gateup.is_cuda or gateup.dtype != torch.bfloat16:
+        act = reference.swiglu(gateup)
+        return reference.quantize_rows_fp8(act)
+    lib = _require_native()
+    T = gateup.shape[0]
+    inter = gateup.shape[-1] // 2
+    q = torch.empty(T, inter, dtype=torch.uint8, device=gateup.device)
+    sc = torch.empty(T, dtype=torch.float32, device=gateup.device)
+    rc = lib.ll_swiglu_q(_ptr(q), _ptr(sc), _ptr(gateup),
+                         ctypes.c_int(T), ctypes.c_int(inter), _stream())
+    _check(rc, "ll_swiglu_q")
+    return q, sc
+
+
+def decode_gemm_w8a8(xq: torch.Tensor, x_scale: torch.Tensor,
+                     wq: torch.Tensor,
+                     w_scale: torch.Tensor) -> torch.Tensor:
+    """out[r][c] = x_scale[r] * w_scale[c] * (q_x . q_w) via the fp8
+    MFMA.  Raises ValueError on unsupported shapes (caller falls back
+    to the bf16 path)."""
+    lib = _require_native()
+    N, K = xq.shape
+    M = wq.shape[0]
+    out = torch.empty(N, M, dtype=torch.bfloat16, device=xq.device)
+    nws = int(lib.ll_decode_gemm_w8a8_ws_floats(N, M, K))
+    ws = None
+    if nws:
+        key = ("a8", N, M, K, str(xq.device))
+        ws = _WS_CACHE.get(key)
+        if ws is None:
+            ws = torch.zeros(nws, dtype=torch.float32, device=xq.device)
+            _WS_CACHE[key] = ws
+    rc = lib.ll_decode_gemm_w8a8(
+        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(xq),
+        _ptr(x_scale), _ptr(wq), _ptr(w_scale), ctypes.c_int(N),
+        ctypes.c_int(M), ctypes.c_int(K), _stream())
+    if rc == -2:
+        raise ValueError("unsupported shape for decode_gemm_w8a8")
+    _check(rc, "ll_decode_gemm_w8a8")
+    return out
 
 
 def quantize_fp8_weight(w: torch.Tensor):

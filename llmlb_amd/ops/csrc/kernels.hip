@@ -1138,3 +1138,193 @@ int ll_device_sync() {
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// Quantizing producers for the W8A8 decode arm: the activation feeding
+// a decode GEMM is emitted as OCP e4m3 bytes + one f32 scale per row
+// BY THE KERNEL THAT PRODUCES IT (a standalone quant launch would cost
+// ~5 us x 3 per layer and eat the fp8 win, the same failure mode as
+// the W8A16 convert pass).  Rounding via the native v_cvt_pk_fp8_f32.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ unsigned int pack4_fp8(const float* f,
+                                                  float inv_scale) {
+  unsigned int lo = __builtin_amdgcn_cvt_pk_fp8_f32(
+      f[0] * inv_scale, f[1] * inv_scale, 0, false);
+  return __builtin_amdgcn_cvt_pk_fp8_f32(
+      f[2] * inv_scale, f[3] * inv_scale, lo, true);
+}
+
+// rmsnorm + quant: same row-per-block shape as rmsnorm_kernel; emits
+// ONLY the quantized row (the bf16 normed output has no other consumer
+// on the decode path).  residual semantics identical to rmsnorm<true>.
+template <bool FUSE_RESIDUAL>
+__global__ void rmsnorm_q_kernel(unsigned char* __restrict__ out_q,
+                                 float* __restrict__ out_scale,
+                                 const bf16* __restrict__ in,
+                                 const bf16* __restrict__ weight,
+                                 bf16* __restrict__ residual,
+                                 int hidden, float eps) {
+  const int row = blockIdx.x;
+  const uint4* inv = reinterpret_cast<const uint4*>(in + (size_t)row * hidden);
+  uint4* resv = FUSE_RESIDUAL
+      ? reinterpret_cast<uint4*>(residual + (size_t)row * hidden) : nullptr;
+  const uint4* wv = reinterpret_cast<const uint4*>(weight);
+  const int nvec = hidden >> 3;
+
+  float ssq = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    uint4 v = inv[i];
+    float f[8];
+    unpack8(v, f);
+    if (FUSE_RESIDUAL) {
+      float r[8];
+      unpack8(resv[i], r);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) f[j] += r[j];
+      uint4 packed = pack8(f);
+      resv[i] = packed;
+      unpack8(packed, f);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ssq += f[j] * f[j];
+  }
+  for (int off = 32; off > 0; off >>= 1) ssq += __shfl_xor(ssq, off, WAVE);
+  __shared__ float warp_red[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_red[wid] = ssq;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float total = 0.f;
+#pragma unroll
+  for (int w = 0; w < 4; ++w) total += (w < nw) ? warp_red[w] : 0.f;
+  const float rrms = rsqrtf(total / hidden + eps);
+
+  // pass 2a: row amax of the normed values (recomputed pass 2b)
+  float amax = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    uint4 v = FUSE_RESIDUAL ? resv[i] : inv[i];
+    float f[8], w[8];
+    unpack8(v, f);
+    unpack8(wv[i], w);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      amax = fmaxf(amax, fabsf(f[j] * rrms * w[j]));
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, WAVE));
+  __syncthreads();   // warp_red reuse
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_red[wid] = amax;
+  __syncthreads();
+  float amax_tot = 0.f;
+#pragma unroll
+  for (int w = 0; w < 4; ++w)
+    amax_tot = fmaxf(amax_tot, (w < nw) ? warp_red[w] : 0.f);
+  const float scale = amax_tot > 0.f ? amax_tot / 448.f : 1.f;
+  const float inv_scale = 1.f / scale;
+  if (threadIdx.x == 0) out_scale[row] = scale;
+
+  // pass 2b: quantize + store bytes (8 per thread-iter = one dword x2)
+  uint2* outv = reinterpret_cast<uint2*>(out_q + (size_t)row * hidden);
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    uint4 v = FUSE_RESIDUAL ? resv[i] : inv[i];
+    float f[8], w[8];
+    unpack8(v, f);
+    unpack8(wv[i], w);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) f[j] = f[j] * rrms * w[j];
+    uint2 q;
+    q.x = pack4_fp8(f, inv_scale);
+    q.y = pack4_fp8(f + 4, inv_scale);
+    outv[i] = q;
+  }
+}
+
+// swiglu + quant: row per block over gateup [T, 2*inter]
+__global__ void swiglu_q_kernel(unsigned char* __restrict__ out_q,
+                                float* __restrict__ out_scale,
+                                const bf16* __restrict__ gateup,
+                                int inter) {
+  const int row = blockIdx.x;
+  const uint4* gv = reinterpret_cast<const uint4*>(
+      gateup + (size_t)row * 2 * inter);
+  const uint4* uv = reinterpret_cast<const uint4*>(
+      gateup + (size_t)row * 2 * inter + inter);
+  const int nvec = inter >> 3;
+
+  float amax = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float g[8], u[8];
+    unpack8(gv[i], g);
+    unpack8(uv[i], u);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float a = g[j] / (1.f + __expf(-g[j])) * u[j];
+      amax = fmaxf(amax, fabsf(a));
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, WAVE));
+  __shared__ float warp_red[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_red[wid] = amax;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float amax_tot = 0.f;
+#pragma unroll
+  for (int w = 0; w < 4; ++w)
+    amax_tot = fmaxf(amax_tot, (w < nw) ? warp_red[w] : 0.f);
+  const float scale = amax_tot > 0.f ? amax_tot / 448.f : 1.f;
+  const float inv_scale = 1.f / scale;
+  if (threadIdx.x == 0) out_scale[row] = scale;
+
+  uint2* outv = reinterpret_cast<uint2*>(out_q + (size_t)row * inter);
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float g[8], u[8], f[8];
+    unpack8(gv[i], g);
+    unpack8(uv[i], u);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      f[j] = g[j] / (1.f + __expf(-g[j])) * u[j];
+    uint2 q;
+    q.x = pack4_fp8(f, inv_scale);
+    q.y = pack4_fp8(f + 4, inv_scale);
+    outv[i] = q;
+  }
+}
+
+extern "C" {
+
+int ll_rmsnorm_q(void* out_q, void* out_scale, const void* in,
+                 const void* weight, void* residual, int n_tokens,
+                 int hidden, float eps, void* stream) {
+  if (hidden % 8 != 0) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(n_tokens), block(256);
+  if (residual)
+    hipLaunchKernelGGL((rmsnorm_q_kernel<true>), grid, block, 0, s,
+                       (unsigned char*)out_q, (float*)out_scale,
+                       (const bf16*)in, (const bf16*)weight,
+                       (bf16*)residual, hidden, eps);
+  else
+    hipLaunchKernelGGL((rmsnorm_q_kernel<false>), grid, block, 0, s,
+                       (unsigned char*)out_q, (float*)out_scale,
+                       (const bf16*)in, (const bf16*)weight, nullptr,
+                       hidden, eps);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+int ll_swiglu_q(void* out_q, void* out_scale, const void* gateup,
+                int n_tokens, int inter, void* stream) {
+  if (inter % 8 != 0) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(n_tokens), block(256);
+  hipLaunchKernelGGL(swiglu_q_kernel, grid, block, 0, s,
+                     (unsigned char*)out_q, (float*)out_scale,
+                     (const bf16*)gateup, inter);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
+
+}  // extern "C"

@@ -144,3 +144,22 @@ def paged_attn_decode(q: torch.Tensor, kcache: torch.Tensor,
             w = torch.softmax(scores, dim=0)
             out[s, h] = (w @ vf[:, kvh]).to(q.dtype)
     return out
+
+
+def quantize_rows_fp8(x: "torch.Tensor"):
+    """Per-row e4m3 quantization reference: returns (bytes u8, scale
+    f32 [rows]); mirrors the GPU kernels' amax/448 scaling."""
+    import torch
+    xf = x.float()
+    amax = xf.abs().amax(dim=-1)
+    scale = torch.where(amax > 0, amax / 448.0, torch.ones_like(amax))
+    q = (xf / scale[:, None]).clamp(-448, 448).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8).contiguous(), scale.contiguous()
+
+
+def w8a8_matmul(xq, x_scale, wq, w_scale):
+    """Dequantized fp32 reference of the W8A8 GEMM."""
+    import torch
+    xd = xq.view(torch.float8_e4m3fn).float() * x_scale[:, None]
+    wd = wq.view(torch.float8_e4m3fn).float() * w_scale[:, None]
+    return xd @ wd.T

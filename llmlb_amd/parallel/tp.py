@@ -76,14 +76,27 @@ class _Fp8WeightMixin:
 
     _w8 = None
 
-    def _w8_pair(self):
+    def _w8_pair(self, force: bool = False):
         from .. import ops as _ops
-        if not _ops._FP8_WEIGHTS or not self.weight.is_cuda \
+        if not (force or _ops._FP8_WEIGHTS) or not self.weight.is_cuda \
                 or self.weight.dtype != torch.bfloat16:
             return None
         if self._w8 is None:
             self._w8 = _ops.quantize_fp8_weight(self.weight.data)
         return self._w8
+
+    def forward_w8a8(self, xq: torch.Tensor, x_scale: torch.Tensor):
+        """W8A8 decode GEMM on pre-quantized activations (rmsnorm_q /
+        swiglu_q outputs).  Returns None when the path does not apply
+        (caller recomputes via the bf16 path)."""
+        from .. import ops as _ops
+        pair = self._w8_pair(force=True)
+        if pair is None or xq.shape[0] > 128:
+            return None
+        try:
+            return _ops.decode_gemm_w8a8(xq, x_scale, pair[0], pair[1])
+        except ValueError:
+            return None
 
 
 class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
@@ -131,6 +144,16 @@ class RowParallelLinear(nn.Module, _Fp8WeightMixin):
         else:
             y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)
+
+    def forward_swiglu_w8a8(self, gateup: torch.Tensor):
+        """swiglu_q (quantizing activation kernel) + W8A8 down GEMM;
+        None when inapplicable."""
+        from .. import ops as _ops
+        if gateup.shape[0] > 128 or not gateup.is_cuda:
+            return None
+        q, sc = _ops.swiglu_q(gateup)
+        y = self.forward_w8a8(q, sc)
+        return self.tp.all_reduce(y) if y is not None else None
 
     def forward_swiglu(self, gateup: torch.Tensor) -> torch.Tensor:
         """down(silu(gate)*up) with the SwiGLU fused into the decode

@@ -367,3 +367,73 @@ def test_rope_kv_append_fused_matches_separate():
     assert torch.equal(k1.contiguous(), k2.contiguous())
     assert torch.equal(kc1, kc2)
     assert torch.equal(vc1, vc2)
+
+
+@pytest.mark.parametrize("N,M,K", [(64, 4096, 4096),    # split-K
+                                   (64, 28672, 4096),   # SK=1
+                                   (64, 4096, 14336),   # odd NC
+                                   (16, 6144, 4096),
+                                   (33, 4096, 4096)])   # padded N
+def test_decode_gemm_w8a8(N, M, K):
+    """W8A8 (fp8 MFMA, no convert chain) vs the dequantized fp32
+    reference — summation order is the only difference."""
+    torch.manual_seed(12)
+    x = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    xq, xs = ref_mod.quantize_rows_fp8(x)
+    xq, xs = xq.to(DEV), xs.to(DEV)
+    wq, ws = ops.quantize_fp8_weight(w)
+    out = ops.decode_gemm_w8a8(xq, xs, wq, ws)
+    ref32 = ref_mod.w8a8_matmul(xq.cpu(), xs.cpu(), wq.cpu(), ws.cpu())
+    rel = ((out.float().cpu() - ref32).abs().max()
+           / ref32.abs().max().clamp(min=1e-6)).item()
+    assert rel < 5e-3, rel
+
+
+def test_rmsnorm_q_matches_reference():
+    torch.manual_seed(13)
+    T, H = 64, 4096
+    x = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    res = torch.randn(T, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    q, sc, res_out = ops.rmsnorm_q(x, w, 1e-5, res.clone())
+    # dequantized GPU output vs the fp32 rmsnorm of the same inputs
+    deq = q.view(torch.float8_e4m3fn).float() * sc[:, None]
+    ref_out, _ = ref_mod.rmsnorm(x.float().cpu(), w.float().cpu(),
+                                 1e-5, res.float().cpu())
+    rel = ((deq.cpu() - ref_out).abs().max()
+           / ref_out.abs().max().clamp(min=1e-6)).item()
+    assert rel < 2e-2, rel
+
+
+def test_swiglu_q_matches_reference():
+    torch.manual_seed(14)
+    T, inter = 64, 14336
+    gu = torch.randn(T, 2 * inter, dtype=torch.bfloat16, device=DEV)
+    q, sc = ops.swiglu_q(gu)
+    deq = q.view(torch.float8_e4m3fn).float() * sc[:, None]
+    ref_out = ref_mod.swiglu(gu.float().cpu())
+    rel = ((deq.cpu() - ref_out).abs().max()
+           / ref_out.abs().max().clamp(min=1e-6)).item()
+    assert rel < 2e-2, rel
+
+
+def test_w8a8_engine_decode_smoke(monkeypatch):
+    """LLMLB_FP8_A8 end to end: the engine decodes through the fp8
+    MFMA GEMMs + quantizing producers without falling back (random
+    weights flip greedy argmax vs bf16, so this checks execution and
+    sanity, not token identity)."""
+    import llmlb_amd.ops as ops_mod
+    monkeypatch.setattr(ops_mod, "_FP8_A8", True)
+    monkeypatch.setattr(ops_mod, "_FP8_WEIGHTS", True)
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.launcher import PRESETS
+    from llmlb_amd.backend.sampler import SamplingParams
+    eng = Engine(EngineConfig(model=PRESETS["llama-small"](),
+                              device=DEV, kv_cache_blocks=512,
+                              seed=5))
+    res = eng.generate_sync([3, 5, 7, 11], max_new_tokens=16,
+                            params=SamplingParams(temperature=0.0,
+                                                  max_new_tokens=16))
+    assert len(res.token_ids) == 16
+    assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
