@@ -256,9 +256,11 @@ def _split_k_ws(N: int, M: int, device) -> torch.Tensor:
 
 
 def _v3_ws(N: int, M: int, K: int, device) -> torch.Tensor | None:
-    """Slab+ticket workspace for the v3 split-K reduce (sized by the
-    kernel library; tickets ride in the same zero-initialized buffer
-    and are reset by the reducing block, so reuse needs no memset)."""
+    """fp32 accumulator workspace for the v3 split-K atomic reduce
+    (sized by the kernel library; the f32->bf16 convert pass re-zeroes
+    it as it drains, so reuse needs no memset).  Both slab-reduce
+    variants (last-arriver and every-slice spin) measured WORSE than
+    atomics+convert on the GPU — see profiles/r02_decode_gemm_v3_ab.md."""
     n = int(_lib.ll_decode_gemm_v3_ws_floats(N, M, K))
     if n == 0:
         return None

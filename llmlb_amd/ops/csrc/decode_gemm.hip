@@ -32,6 +32,7 @@
 typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ float dg_bf2f(unsigned short u) {
@@ -693,91 +694,11 @@ void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
   }
 
   // ---- store ----
-  if (ATOMIC && SPLIT_W == 1) {
-    // Slab + ticket split-K reduce (guide §5 sampling-GEMM recipe):
-    // every K-slice writes its fp32 slab with plain stores, publishes
-    // with an agent release + relaxed ticket, and the LAST arriver
-    // acquires, sums the SK slabs and writes the bf16 tile directly.
-    // Replaces the r1 fp32-atomicAdd accumulator + separate
-    // f32_to_bf16 pass, whose 4704 extra ~5 us launches per probe run
-    // erased the in-engine win (profiles/r02_decode_gemm_v3_ab.md).
-    // ws layout: [mblk][kslice][N_TILES*16][128] f32, then one int
-    // ticket per mblk (zero-initialized by the allocator; the reducer
-    // resets its ticket for the next call).
-    constexpr int TILE_E = N_TILES * 16 * 128;
-    float* slab =
-        out_ws + ((size_t)mblk * SK + blockIdx.y) * TILE_E;
-#pragma unroll
-    for (int t = 0; t < N_TILES; ++t)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        slab[(t * 16 + l_hi * 4 + r) * 128 + wid * 16 + l_lo] =
-            W8 ? acc[t][r] * wsc : acc[t][r];
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    const int mblocks = gridDim.x;
-    // two counters per tile: arrivals, then observers (for the reset)
-    int* tickets = reinterpret_cast<int*>(
-        out_ws + (size_t)mblocks * SK * TILE_E);
-    int* observed = tickets + mblocks;
-    if (threadIdx.x == 0) {
-      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-      // ROCm 7.2 drops the post-wbl2 vmcnt wait when the scoreboard
-      // looks clear — restate it where the compiler cannot remove it
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __hip_atomic_fetch_add(tickets + mblk, 1, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-      // EVERY slice reduces 1/SK of the tile (the last-arriver form
-      // serialized a 256 KB slab read per tile and cost ~1.2 ms/step
-      // in-engine): spin until all siblings have published
-      while (__hip_atomic_load(tickets + mblk, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT) < SK)
-        __builtin_amdgcn_s_sleep(2);
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    }
-    __syncthreads();
-    // this slice reduces rows of its share across all SK slabs
-    const float* base = out_ws + (size_t)mblk * SK * TILE_E;
-    const int share = TILE_E / 4 / SK;  // f32x4 groups per slice
-    const int e0 = (int)blockIdx.y * share;
-    for (int e4 = e0 + threadIdx.x; e4 < e0 + share; e4 += 512) {
-      f32x4 sum = {0.f, 0.f, 0.f, 0.f};
-      for (int s2 = 0; s2 < SK; ++s2)
-        sum += *reinterpret_cast<const f32x4*>(base +
-                                               (size_t)s2 * TILE_E +
-                                               (size_t)e4 * 4);
-      const int row = (e4 * 4) / 128;
-      const int col = (e4 * 4) % 128;
-      if (row < N) {
-        unsigned short h[4];
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          union { float f; unsigned int i; } c2;
-          c2.f = sum[j];
-          unsigned int bias = 0x7FFF + ((c2.i >> 16) & 1);
-          h[j] = (unsigned short)((c2.i + bias) >> 16);
-        }
-        *reinterpret_cast<uint2*>(out + (size_t)row * M + mblk * 128 +
-                                  col) =
-            *reinterpret_cast<const uint2*>(h);
-      }
-    }
-    // reset protocol: last block to finish observing resets both
-    // counters for the next call
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      int o = __hip_atomic_fetch_add(observed + mblk, 1,
-                                     __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-      if (o == SK - 1) {
-        __hip_atomic_store(tickets + mblk, 0, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-        __hip_atomic_store(observed + mblk, 0, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-      }
-    }
-    return;
-  }
+  // split-K returns to fp32 atomics + the f32->bf16 convert pass:
+  // the last-arriver slab reduce serialized 256 KB reads per tile
+  // (-1.2 ms/step) and the every-slice spin both cost more than the
+  // convert pass AND forced a <=256-block residency cap that broke
+  // the qkv fill (r2 A/B history in profiles/r02_decode_gemm_v3_ab.md)
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) {
 #pragma unroll
@@ -832,12 +753,7 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
     while ((M / 64) * SK < 224 && SK < 8 && (K / (SK * 2)) % 256 == 0)
       SK *= 2;
   } else {
-    // grid cap 256 = the chip's one-WG-per-CU residency: the split-K
-    // reduce spins on sibling slices, and a block that cannot be
-    // resident would deadlock the spin (every block must be dispatched
-    // while its siblings spin)
-    while ((M / 128) * SK < 224 && SK < 16 &&
-           (M / 128) * SK * 2 <= 256 && (K / (SK * 2)) % 256 == 0)
+    while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
       SK *= 2;
   }
   const int mblocks = M / (split_w == 2 ? 64 : 128);
@@ -901,9 +817,7 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
 #undef LAUNCH_V3
 #undef LAUNCH_V3S
 #undef LAUNCH_V3_
-  if (SK > 1 && split_w == 2) {
-    // env-forced BN=64 experiment still accumulates via atomics and
-    // needs the convert pass; the default slab path does not
+  if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;
     long long blocks2 = (n + thr - 1) / thr;
@@ -935,14 +849,14 @@ long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
 // ===========================================================================
 // W8A8: both operands OCP e4m3, consumed DIRECTLY by the gfx950 fp8
 // MFMA (v_mfma_f32_16x16x32_fp8_fp8) — no convert chain at all, which
-// was the measured killer of the W8A16 arm (profiles/
-// r01_fp8_weights_ab.md, r02_decode_gemm_v3_ab.md).  out[r][c] =
-// x_scale[r] * w_scale[c] * sum(q_x * q_w): the per-row/per-col scales
-// factor EXACTLY out of the MFMA accumulation.  Same software pipeline
-// as v3 (two W register sets, two LDS x buffers, nt weight loads,
-// counted waits); x is staged as BYTES (half the v3 LDS/stage traffic).
-// Weight bytes halve vs bf16 — the bandwidth ceiling the r1 VERDICT
-// called the biggest unclaimed win.
+// was the measured killer of the W8A16 arm.  out[r][c] = x_scale[r] *
+// w_scale[c] * sum(q_x*q_w): per-row/col scales factor EXACTLY out of
+// the accumulation.  512-k chunks with 16-byte W fragment loads keep
+// 128 B/lane of the weight stream in flight (the first 8-byte-load
+// cut measured HALF the bytes in the SAME wall time — the stream is
+// latency-bound, so in-flight BYTES are the currency, not requests).
+// Split-K uses fp32 atomics + the shared f32->bf16 convert pass (both
+// slab-reduce forms measured worse; see r02_decode_gemm_v3_ab.md).
 // ===========================================================================
 
 typedef long long i64_t;
@@ -950,17 +864,15 @@ typedef long long i64_t;
 template <int N_TILES, bool ATOMIC>
 __global__ __launch_bounds__(512, 2)
 void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
-                           float* __restrict__ out_ws,       // slab ws
+                           float* __restrict__ out_ws,       // [N, M]
                            const unsigned char* __restrict__ xq,  // [N,K]
                            const float* __restrict__ x_scale,     // [N]
                            const unsigned char* __restrict__ wq,  // [M,K]
                            const float* __restrict__ w_scale,     // [M]
                            int N, int M, int K, int SK) {
-  constexpr int CHUNK_K = 256;              // k per staged chunk
-  constexpr int STEPS = CHUNK_K / 32;
+  constexpr int CHUNK_K = 512;              // k (= bytes) per chunk
+  constexpr int STEPS = CHUNK_K / 32;       // 16 MFMA k-steps
   constexpr int ROWS = N_TILES * 16;
-  // bytes: 16-byte pieces; small N_TILES has fewer pieces than
-  // threads, so guard with TOTAL_P
   constexpr int TOTAL_P = ROWS * CHUNK_K / 16;
   constexpr int PIECES = (TOTAL_P + 511) / 512;
   const int wid = threadIdx.x / WAVE;
@@ -971,7 +883,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   const int k_begin = (int)blockIdx.y * Ks;
   const int l_lo = lane & 15;
   const int l_hi = lane >> 4;
-  const int koff = l_hi * 8;                // byte == k offset
+  const int koff = l_hi * 8;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xs[2] = {smem, smem + (size_t)ROWS * CHUNK_K};
@@ -983,17 +895,15 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  // stride-512 piece mapping with CLAMP (not predication: divergent
-  // guards around the staging array made hipcc spill it to scratch
-  // and wait vmcnt(0) per load).  Overflow threads re-stage the last
-  // piece — duplicate identical LDS writes are benign.
   const int piece_base = threadIdx.x;
 
+  // W fragment u: covers steps 2u and 2u+1 (k = u*64 + [0,32) halves);
+  // lane k offset koff applies within each 32-k step
 #define A8_LOAD_W(dst, k0)                                                  \
   do {                                                                      \
-    _Pragma("unroll") for (int s2 = 0; s2 < STEPS; ++s2)                    \
-        dst[s2] = __builtin_nontemporal_load(                               \
-            reinterpret_cast<const u32x2*>(wrow + (k0) + s2 * 32 + koff));  \
+    _Pragma("unroll") for (int u = 0; u < STEPS; ++u)                       \
+        dst[u] = __builtin_nontemporal_load(                                \
+            reinterpret_cast<const u32x2*>(wrow + (k0) + u * 32 + koff));   \
   } while (0)
 #define A8_LOAD_X(xv, k0)                                                   \
   do {                                                                      \
@@ -1002,7 +912,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
       const int tok = piece / (CHUNK_K / 16);                               \
       const int kb = piece % (CHUNK_K / 16);                                \
       const int src_tok = tok < N ? tok : 0;                                \
-      xv[p] = *reinterpret_cast<const uint4*>(                              \
+      xv[p] = *reinterpret_cast<const u32x4*>(                              \
           xq + (size_t)src_tok * K + (k0) + kb * 16);                       \
     }                                                                       \
   } while (0)
@@ -1013,7 +923,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
       const int tok = piece / (CHUNK_K / 16);                               \
       const int kb = piece % (CHUNK_K / 16);                                \
       const int swz_kb = kb ^ (tok & 15);                                   \
-      *reinterpret_cast<uint4*>(xs[b] + (size_t)tok * CHUNK_K +             \
+      *reinterpret_cast<u32x4*>(xs[b] + (size_t)tok * CHUNK_K +             \
                                 swz_kb * 16) = xv[p];                       \
     }                                                                       \
   } while (0)
@@ -1021,127 +931,68 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   do {                                                                      \
     _Pragma("unroll") for (int s2 = 0; s2 < STEPS; ++s2) {                  \
       const i64_t bfrag = *reinterpret_cast<const i64_t*>(&bw[s2]);         \
-      const int kbyte = s2 * 32 + koff;                                     \
-      const int kb = kbyte >> 4;                                            \
-      const int off_in = kbyte & 15;                                        \
+      const int kk = s2 * 32 + koff;                                        \
+      const int kb = kk >> 4;                                               \
       _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
         const int tok = t * 16 + l_lo;                                      \
         const int swz_kb = kb ^ (tok & 15);                                 \
         const i64_t afrag = *reinterpret_cast<const i64_t*>(                \
-            xs[b] + (size_t)tok * CHUNK_K + swz_kb * 16 + off_in);          \
+            xs[b] + (size_t)tok * CHUNK_K + swz_kb * 16 + (kk & 15));       \
         acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(                \
             afrag, bfrag, acc[t], 0, 0, 0);                                 \
       }                                                                     \
     }                                                                       \
   } while (0)
 
-  u32x2 bA[STEPS], bB[STEPS];
-  uint4 xvA[PIECES], xvB[PIECES];
+  u32x2 wA[STEPS], wB[STEPS];
+  u32x4 xv[PIECES];
   const int NC = Ks / CHUNK_K;
 
-  A8_LOAD_X(xvA, k_begin);
-  A8_LOAD_W(bA, k_begin);
-  A8_WRITE_X(0, xvA);
+  A8_LOAD_X(xv, k_begin);
+  A8_LOAD_W(wA, k_begin);
+  A8_WRITE_X(0, xv);
   __syncthreads();
 
   int c = 0;
   for (; c + 2 <= NC; c += 2) {
     const int k1 = k_begin + (c + 1) * CHUNK_K;
-    A8_LOAD_X(xvB, k1);
-    A8_LOAD_W(bB, k1);
-    A8_CONSUME(0, bA);
-    A8_WRITE_X(1, xvB);
+    A8_LOAD_X(xv, k1);
+    A8_LOAD_W(wB, k1);
+    A8_CONSUME(0, wA);
+    A8_WRITE_X(1, xv);
     __syncthreads();
     if (c + 2 < NC) {
       const int k2 = k_begin + (c + 2) * CHUNK_K;
-      A8_LOAD_X(xvA, k2);
-      A8_LOAD_W(bA, k2);
-      A8_CONSUME(1, bB);
-      A8_WRITE_X(0, xvA);
+      A8_LOAD_X(xv, k2);
+      A8_LOAD_W(wA, k2);
+      A8_CONSUME(1, wB);
+      A8_WRITE_X(0, xv);
       __syncthreads();
     } else {
-      A8_CONSUME(1, bB);
+      A8_CONSUME(1, wB);
     }
   }
-  if (c < NC) A8_CONSUME(0, bA);
+  if (c < NC) A8_CONSUME(0, wA);
 #undef A8_LOAD_W
 #undef A8_LOAD_X
 #undef A8_WRITE_X
 #undef A8_CONSUME
 
-  // ---- store: scales fold here (exact factorization) ----
-  if (ATOMIC) {
-    // split-K: scaled partials into the slab; every-slice reduce as v3
-    constexpr int TILE_E = N_TILES * 16 * 128;
-    float* slab = out_ws + ((size_t)mblk * SK + blockIdx.y) * TILE_E;
-#pragma unroll
-    for (int t = 0; t < N_TILES; ++t)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        slab[(t * 16 + l_hi * 4 + r) * 128 + wid * 16 + l_lo] =
-            acc[t][r] * wsc;
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    const int mblocks = gridDim.x;
-    int* tickets = reinterpret_cast<int*>(
-        out_ws + (size_t)mblocks * SK * TILE_E);
-    int* observed = tickets + mblocks;
-    if (threadIdx.x == 0) {
-      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __hip_atomic_fetch_add(tickets + mblk, 1, __ATOMIC_RELAXED,
-                             __HIP_MEMORY_SCOPE_AGENT);
-      while (__hip_atomic_load(tickets + mblk, __ATOMIC_RELAXED,
-                               __HIP_MEMORY_SCOPE_AGENT) < SK)
-        __builtin_amdgcn_s_sleep(2);
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    }
-    __syncthreads();
-    const float* base = out_ws + (size_t)mblk * SK * TILE_E;
-    const int share = TILE_E / 4 / SK;
-    const int e0 = (int)blockIdx.y * share;
-    for (int e4 = e0 + threadIdx.x; e4 < e0 + share; e4 += 512) {
-      f32x4 sum = {0.f, 0.f, 0.f, 0.f};
-      for (int s2 = 0; s2 < SK; ++s2)
-        sum += *reinterpret_cast<const f32x4*>(base +
-                                               (size_t)s2 * TILE_E +
-                                               (size_t)e4 * 4);
-      const int row = (e4 * 4) / 128;
-      const int col = (e4 * 4) % 128;
-      if (row < N) {
-        const float sx = x_scale[row];
-        unsigned short h2[4];
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          h2[j] = dg_f2bf(sum[j] * sx);
-        *reinterpret_cast<uint2*>(out + (size_t)row * M + mblk * 128 +
-                                  col) =
-            *reinterpret_cast<const uint2*>(h2);
-      }
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      int o = __hip_atomic_fetch_add(observed + mblk, 1,
-                                     __ATOMIC_RELAXED,
-                                     __HIP_MEMORY_SCOPE_AGENT);
-      if (o == SK - 1) {
-        __hip_atomic_store(tickets + mblk, 0, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-        __hip_atomic_store(observed + mblk, 0, __ATOMIC_RELAXED,
-                           __HIP_MEMORY_SCOPE_AGENT);
-      }
-    }
-    return;
-  }
+  // ---- store: scales fold exactly here ----
 #pragma unroll
   for (int t = 0; t < N_TILES; ++t) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = t * 16 + l_hi * 4 + r;
       if (row < N) {
-        union { unsigned short u; __hip_bfloat16 b; } cv;
-        cv.u = dg_f2bf(acc[t][r] * wsc * x_scale[row]);
-        out[(size_t)row * M + col_base + l_lo] = cv.b;
+        const float v = acc[t][r] * wsc * x_scale[row];
+        if (ATOMIC) {
+          atomicAdd(out_ws + (size_t)row * M + col_base + l_lo, v);
+        } else {
+          union { unsigned short u; __hip_bfloat16 b; } cv;
+          cv.u = dg_f2bf(v);
+          out[(size_t)row * M + col_base + l_lo] = cv.b;
+        }
       }
     }
   }
@@ -1153,16 +1004,28 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
                         const void* x_scale, const void* wq,
                         const void* w_scale, int N, int M, int K,
                         void* stream) {
-  if (K % 256 != 0 || M % 128 != 0 || N > 128) return -2;
+  if (K % 512 != 0 || M % 128 != 0 || N > 128) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
+  // SK = smallest divisor of K/512 reaching >=224 blocks (chunk
+  // divisibility first; non-power-of-two allowed, e.g. down K=14336
+  // -> SK=7 for 224 blocks of NC=4)
+  const int mblocks = M / 128;
+  const int kc = K / 512;
   int SK = 1;
-  while ((M / 128) * SK < 224 && SK < 16 &&
-         (M / 128) * SK * 2 <= 256 && (K / (SK * 2)) % 256 == 0)
-    SK *= 2;
-  const size_t lds = (size_t)n_tiles * 16 * 256 * 2;  // byte x, 2 bufs
+  if (mblocks < 224) {
+    int best = 1;
+    for (int d = 1; d <= kc && d <= 16; ++d) {
+      if (kc % d) continue;
+      best = d;
+      if (mblocks * d >= 224) break;
+    }
+    SK = best;
+  }
+  const size_t lds = (size_t)n_tiles * 16 * 512 * 2;  // byte x, 2 bufs
+  if (lds > 160 * 1024) return -2;
   if (SK > 1 && ws == nullptr) return -2;
-  dim3 grid(M / 128, SK), block(512);
+  dim3 grid(mblocks, SK), block(512);
 #define LAUNCH_A8(NT)                                                       \
   do {                                                                      \
     if (SK > 1)                                                             \
@@ -1188,13 +1051,21 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
     default: return -2;
   }
 #undef LAUNCH_A8
+  if (SK > 1) {
+    long long n = (long long)N * M;
+    int thr = 256;
+    long long blocks2 = (n + thr - 1) / thr;
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((unsigned)blocks2),
+                       dim3(thr), 0, s, (bf16*)out, (float*)ws, n);
+  }
   hipError_t e = hipGetLastError();
   return e == hipSuccess ? 0 : (int)e;
 }
 
-// same ws contract as v3 (slabs + two counters)
 long long ll_decode_gemm_w8a8_ws_floats(int N, int M, int K) {
-  return ll_decode_gemm_v3_ws_floats(N, M, K);
+  if (K % 512 != 0 || M % 128 != 0 || N > 128) return 0;
+  if (M / 128 >= 224) return 0;
+  return (long long)N * M;   // fp32 atomic accumulator
 }
 
 }  // extern "C"

