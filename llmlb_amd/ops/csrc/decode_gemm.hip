@@ -847,9 +847,18 @@ long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
 }  // extern "C"
 
 // ===========================================================================
-// W8A8: both operands OCP e4m3, consumed DIRECTLY by the gfx950 fp8
-// MFMA (v_mfma_f32_16x16x32_fp8_fp8) — no convert chain at all, which
-// was the measured killer of the W8A16 arm.  out[r][c] = x_scale[r] *
+// W8A8: both operands OCP e4m3, consumed DIRECTLY by the gfx950
+// block-scaled MFMA v_mfma_scale_f32_16x16x128_f8f6f4 with unit E8M0
+// scales (byte 127) — 128 k per instruction, 4x the k-rate of the
+// plain fp8 16x16x32 form.  The first cut used the x32 fp8 MFMA and
+// measured HALF the bytes of the bf16 v3 kernel in the SAME wall time
+// (same instruction count per k => the decode GEMM is ISSUE-bound at
+// this occupancy, not bandwidth-bound); only a larger-K MFMA converts
+// the halved weight stream into wall-clock.  Operand layout verified
+// empirically by benchmarks/mfma128_probe.hip: any consistent A/B
+// byte-position bijection contracts pairwise (lane&15 = row/col,
+// lane>>4 = k-group), so we pack k(g, j) = g*32 + j — 32 CONTIGUOUS
+// bytes per lane, exactly what wide loads want.  out[r][c] = x_scale[r] *
 // w_scale[c] * sum(q_x*q_w): per-row/col scales factor EXACTLY out of
 // the accumulation.  512-k chunks with 16-byte W fragment loads keep
 // 128 B/lane of the weight stream in flight (the first 8-byte-load
@@ -860,6 +869,7 @@ long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
 // ===========================================================================
 
 typedef long long i64_t;
+typedef __attribute__((ext_vector_type(8))) int dg_i32x8;
 
 template <int N_TILES, bool ATOMIC>
 __global__ __launch_bounds__(512, 2)
@@ -871,7 +881,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            const float* __restrict__ w_scale,     // [M]
                            int N, int M, int K, int SK) {
   constexpr int CHUNK_K = 512;              // k (= bytes) per chunk
-  constexpr int STEPS = CHUNK_K / 32;       // 16 MFMA k-steps
+  constexpr int S128 = CHUNK_K / 128;       // 4 scaled-MFMA k-steps
   constexpr int ROWS = N_TILES * 16;
   constexpr int TOTAL_P = ROWS * CHUNK_K / 16;
   constexpr int PIECES = (TOTAL_P + 511) / 512;
@@ -883,7 +893,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   const int k_begin = (int)blockIdx.y * Ks;
   const int l_lo = lane & 15;
   const int l_hi = lane >> 4;
-  const int koff = l_hi * 8;
+  const int koff = l_hi * 32;               // 32 contiguous k per lane
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xs[2] = {smem, smem + (size_t)ROWS * CHUNK_K};
@@ -897,13 +907,14 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 
   const int piece_base = threadIdx.x;
 
-  // W fragment u: covers steps 2u and 2u+1 (k = u*64 + [0,32) halves);
-  // lane k offset koff applies within each 32-k step
 #define A8_LOAD_W(dst, k0)                                                  \
   do {                                                                      \
-    _Pragma("unroll") for (int u = 0; u < STEPS; ++u)                       \
-        dst[u] = __builtin_nontemporal_load(                                \
-            reinterpret_cast<const u32x2*>(wrow + (k0) + u * 32 + koff));   \
+    _Pragma("unroll") for (int u = 0; u < S128; ++u) {                      \
+      dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const       \
+          u32x4*>(wrow + (k0) + u * 128 + koff));                           \
+      dst[u].h[1] = __builtin_nontemporal_load(reinterpret_cast<const       \
+          u32x4*>(wrow + (k0) + u * 128 + koff + 16));                      \
+    }                                                                       \
   } while (0)
 #define A8_LOAD_X(xv, k0)                                                   \
   do {                                                                      \
@@ -929,22 +940,24 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   } while (0)
 #define A8_CONSUME(b, bw)                                                   \
   do {                                                                      \
-    _Pragma("unroll") for (int s2 = 0; s2 < STEPS; ++s2) {                  \
-      const i64_t bfrag = *reinterpret_cast<const i64_t*>(&bw[s2]);         \
-      const int kk = s2 * 32 + koff;                                        \
+    _Pragma("unroll") for (int u = 0; u < S128; ++u) {                      \
+      const int kk = u * 128 + koff;                                        \
       const int kb = kk >> 4;                                               \
       _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
         const int tok = t * 16 + l_lo;                                      \
-        const int swz_kb = kb ^ (tok & 15);                                 \
-        const i64_t afrag = *reinterpret_cast<const i64_t*>(                \
-            xs[b] + (size_t)tok * CHUNK_K + swz_kb * 16 + (kk & 15));       \
-        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(                \
-            afrag, bfrag, acc[t], 0, 0, 0);                                 \
+        w32 af;                                                             \
+        af.h[0] = *reinterpret_cast<const u32x4*>(                          \
+            xs[b] + (size_t)tok * CHUNK_K + (kb ^ (tok & 15)) * 16);        \
+        af.h[1] = *reinterpret_cast<const u32x4*>(                          \
+            xs[b] + (size_t)tok * CHUNK_K + ((kb + 1) ^ (tok & 15)) * 16);  \
+        acc[t] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(          \
+            af.v, bw[u].v, acc[t], 0, 0, 0, 127, 0, 127);                   \
       }                                                                     \
     }                                                                       \
   } while (0)
 
-  u32x2 wA[STEPS], wB[STEPS];
+  union w32 { dg_i32x8 v; u32x4 h[2]; };
+  w32 wA[S128], wB[S128];
   u32x4 xv[PIECES];
   const int NC = Ks / CHUNK_K;
 
