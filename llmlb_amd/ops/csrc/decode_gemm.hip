@@ -871,8 +871,13 @@ long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
 typedef long long i64_t;
 typedef __attribute__((ext_vector_type(8))) int dg_i32x8;
 
-template <int N_TILES, bool ATOMIC>
-__global__ __launch_bounds__(512, 2)
+// WAVES: waves per workgroup.  8 => M-tile 128, 1 WG/CU at N_TILES=4
+// (194 VGPRs); 4 => M-tile 64, TWO 4-wave WGs per CU — while one WG
+// sits in its barrier/LDS-write phase the other keeps the HBM pipe
+// full (the asm shows healthy counted waits, so the residual gap to
+// the bandwidth floor is barrier drain, not serialization).
+template <int N_TILES, bool ATOMIC, int WAVES = 8>
+__global__ __launch_bounds__(WAVES * 64, 2)
 void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            float* __restrict__ out_ws,       // [N, M]
                            const unsigned char* __restrict__ xq,  // [N,K]
@@ -884,11 +889,11 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   constexpr int S128 = CHUNK_K / 128;       // 4 scaled-MFMA k-steps
   constexpr int ROWS = N_TILES * 16;
   constexpr int TOTAL_P = ROWS * CHUNK_K / 16;
-  constexpr int PIECES = (TOTAL_P + 511) / 512;
+  constexpr int PIECES = (TOTAL_P + WAVES * 64 - 1) / (WAVES * 64);
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int mblk = blockIdx.x;
-  const int col_base = mblk * 128 + wid * 16;
+  const int col_base = mblk * (WAVES * 16) + wid * 16;
   const int Ks = K / SK;
   const int k_begin = (int)blockIdx.y * Ks;
   const int l_lo = lane & 15;
@@ -919,7 +924,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 #define A8_LOAD_X(xv, k0)                                                   \
   do {                                                                      \
     _Pragma("unroll") for (int p = 0; p < PIECES; ++p) {                    \
-      const int piece = min(piece_base + p * 512, TOTAL_P - 1);             \
+      const int piece = min(piece_base + p * (WAVES * 64), TOTAL_P - 1);             \
       const int tok = piece / (CHUNK_K / 16);                               \
       const int kb = piece % (CHUNK_K / 16);                                \
       const int src_tok = tok < N ? tok : 0;                                \
@@ -930,7 +935,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 #define A8_WRITE_X(b, xv)                                                   \
   do {                                                                      \
     _Pragma("unroll") for (int p = 0; p < PIECES; ++p) {                    \
-      const int piece = min(piece_base + p * 512, TOTAL_P - 1);             \
+      const int piece = min(piece_base + p * (WAVES * 64), TOTAL_P - 1);             \
       const int tok = piece / (CHUNK_K / 16);                               \
       const int kb = piece % (CHUNK_K / 16);                                \
       const int swz_kb = kb ^ (tok & 15);                                   \
@@ -1020,50 +1025,57 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
   if (K % 512 != 0 || M % 128 != 0 || N > 128) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
-  // SK = smallest divisor of K/512 reaching >=224 blocks (chunk
-  // divisibility first; non-power-of-two allowed, e.g. down K=14336
-  // -> SK=7 for 224 blocks of NC=4)
-  const int mblocks = M / 128;
-  const int kc = K / 512;
-  int SK = 1;
-  if (mblocks < 224) {
-    int best = 1;
-    for (int d = 1; d <= kc && d <= 16; ++d) {
-      if (kc % d) continue;
-      best = d;
-      if (mblocks * d >= 224) break;
-    }
-    SK = best;
-  }
+  static const int waves = [] {
+    const char* e = getenv("LLMLB_A8_WAVES");
+    return (e && e[0] == '8') ? 8 : 4;   // 4-wave default (2 WG/CU)
+  }();
   const size_t lds = (size_t)n_tiles * 16 * 512 * 2;  // byte x, 2 bufs
   if (lds > 160 * 1024) return -2;
-  if (SK > 1 && ws == nullptr) return -2;
-  dim3 grid(mblocks, SK), block(512);
-#define LAUNCH_A8(NT)                                                       \
+  const int kc = K / 512;
+  int SK = 1;          // set per-launch below (depends on the M tile)
+#define LAUNCH_A8W(NT, W)                                                   \
   do {                                                                      \
+    const int mblocks = M / (16 * (W));                                     \
+    SK = 1;                                                                 \
+    if (mblocks < 224) {                                                    \
+      for (int d = 1; d <= kc && d <= 16; ++d) {                            \
+        if (kc % d) continue;                                               \
+        SK = d;                                                             \
+        if (mblocks * d >= 224) break;                                      \
+      }                                                                     \
+    }                                                                       \
+    if (SK > 1 && ws == nullptr) return -2;                                 \
+    dim3 grid(mblocks, SK), block((W) * 64);                                \
     if (SK > 1)                                                             \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true>), grid, block,    \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true, W>), grid, block, \
                          lds, s, (bf16*)out, (float*)ws,                    \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
                          N, M, K, SK);                                      \
     else                                                                    \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false>), grid, block,   \
-                         lds, s, (bf16*)out, (float*)ws,                    \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W>), grid,       \
+                         block, lds, s, (bf16*)out, (float*)ws,             \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
                          N, M, K, SK);                                      \
+  } while (0)
+#define LAUNCH_A8(NT)                                                       \
+  do {                                                                      \
+    if (waves == 8) LAUNCH_A8W(NT, 8); else LAUNCH_A8W(NT, 4);              \
   } while (0)
   switch (n_tiles) {
     case 1: LAUNCH_A8(1); break;
     case 2: LAUNCH_A8(2); break;
     case 3: LAUNCH_A8(3); break;
     case 4: LAUNCH_A8(4); break;
-    case 5: case 6: LAUNCH_A8(6); break;
-    case 7: case 8: LAUNCH_A8(8); break;
+    // NT >= 6 at 4 waves spills (x staging registers double); the
+    // 8-wave variant is the only clean fit for the big row tiles
+    case 5: case 6: LAUNCH_A8W(6, 8); break;
+    case 7: case 8: LAUNCH_A8W(8, 8); break;
     default: return -2;
   }
 #undef LAUNCH_A8
+#undef LAUNCH_A8W
   if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;
