@@ -1025,10 +1025,14 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
   if (K % 512 != 0 || M % 128 != 0 || N > 128) return -2;
   hipStream_t s = (hipStream_t)stream;
   const int n_tiles = (N + 15) / 16;
-  static const int waves = [] {
+  static const int wave_env = [] {
     const char* e = getenv("LLMLB_A8_WAVES");
-    return (e && e[0] == '8') ? 8 : 4;   // 4-wave default (2 WG/CU)
+    return e ? (e[0] == '8' ? 8 : 4) : 0;
   }();
+  // measured (profiles/r02_w8a8.md): 4-wave (two WGs/CU) wins every
+  // shape except gate_up, where the 8-wave M-tile-128 grid already
+  // fills 224 CUs without split-K and wins by ~3%
+  const int waves = wave_env ? wave_env : (M / 128 >= 224 ? 8 : 4);
   const size_t lds = (size_t)n_tiles * 16 * 512 * 2;  // byte x, 2 bufs
   if (lds > 160 * 1024) return -2;
   const int kc = K / 512;
