@@ -489,12 +489,15 @@ int ll_decode_gemm_v2(void* out, void* ws, const void* x, const void* w,
 // ===========================================================================
 
 template <int N_TILES, bool ATOMIC, bool SWIGLU = false, bool W8 = false,
-          int SPLIT_W = 1, int CHUNK_K = 256>
+          int SPLIT_W = 1, int CHUNK_K = 256, int WAVES = 8>
 // SPLIT_W=2 (BN=64) targets TWO 512-thread workgroups per CU (double
 // the W bytes in flight); that needs <=128 VGPRs, which N_TILES<=4
 // fits without spilling.  SPLIT_W=1 keeps the 2-wave/SIMD budget —
 // forcing 128 there made hipcc spill 16-56 B/lane to scratch.
-__global__ __launch_bounds__(512, (N_TILES <= 4 && SPLIT_W == 2 ? 4 : 2))
+// WAVES=4 (M-tile 64, two WGs/CU) transfers the A8 kernel's
+// barrier-drain fix to the bf16 path.
+__global__ __launch_bounds__(WAVES * 64,
+                             (N_TILES <= 4 && SPLIT_W == 2 ? 4 : 2))
 void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
                            float* __restrict__ out_ws,  // [N, M] (SK>1)
                            const bf16* __restrict__ x,  // [N, K]
@@ -505,8 +508,8 @@ void decode_gemm_v3_kernel(bf16* __restrict__ out,      // [N, M]
   constexpr int STEPS = CHUNK_K / 32;       // MFMA k-steps per chunk
   constexpr int ROWS = N_TILES * 16;
   // x pieces staged per thread per chunk (16 B each)
-  constexpr int PIECES = ROWS * CHUNK_K * 2 / (512 * 16);
-  constexpr int COLS_W = 8 / SPLIT_W;       // 16-col groups per block
+  constexpr int PIECES = ROWS * CHUNK_K * 2 / (WAVES * 64 * 16);
+  constexpr int COLS_W = WAVES / SPLIT_W;   // 16-col groups per block
   constexpr int WSTEPS = STEPS / SPLIT_W;   // MFMA steps per wave/chunk
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
@@ -767,15 +770,35 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
                        (K / SK) % 512 == 0)
                           ? 512
                           : 256;
+  // 4-wave variant (A8's two-WGs-per-CU barrier-drain fix, bf16 form):
+  // plain path only (no swiglu/W8/SPLIT_W2/chunk512 instantiations)
+  static const char* fwv = getenv("LLMLB_GEMM_V3_WAVES");
+  int waves = 8;
+  if (fwv && fwv[0] == '4' && !swiglu_x && w_scale == nullptr &&
+      split_w == 1 && chunk_k == 256 && M % 64 == 0 &&
+      (n_tiles <= 4 || n_tiles >= 7)) {  // NT=6 at 4 waves spills
+    waves = 4;
+    SK = 1;
+    const int mb4 = M / 64;
+    while (mb4 * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
+      SK *= 2;
+  }
+  const int mblocks_w = waves == 4 ? M / 64 : mblocks;
   const size_t lds = (size_t)n_tiles * 16 * chunk_k * 2 * 2;
   if (lds > 160 * 1024) return -2;
   if (SK > 1 && ws == nullptr) return -2;
-  dim3 grid(mblocks, SK), block(512);
+  dim3 grid(mblocks_w, SK), block(waves * 64);
 
   const bool w8 = w_scale != nullptr;
 #define LAUNCH_V3_(NT, AT, SW, W8F, SWI)                                    \
   do {                                                                      \
-    if (chunk_k == 512)                                                     \
+    if (waves == 4 && !(SW) && !(W8F) && (SWI) == 1)                        \
+      hipLaunchKernelGGL((decode_gemm_v3_kernel<NT, AT, false, false, 1,    \
+                                                256, 4>),                   \
+                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
+                         (const bf16*)x, (const bf16*)w,                    \
+                         (const float*)w_scale, N, M, K, SK);               \
+    else if (chunk_k == 512)                                                \
       hipLaunchKernelGGL((decode_gemm_v3_kernel<NT, AT, SW, W8F, SWI,       \
                                                 512>),                      \
                          grid, block, lds, s, (bf16*)out, (float*)ws,       \
@@ -834,8 +857,7 @@ int ll_decode_gemm_v3(void* out, void* ws, const void* x, const void* w,
 long long ll_decode_gemm_v3_ws_floats(int N, int M, int K) {
   if (K % 256 != 0 || M % 128 != 0 || N > 128) return 0;
   int SK = 1;
-  while ((M / 128) * SK < 224 && SK < 16 &&
-         (M / 128) * SK * 2 <= 256 && (K / (SK * 2)) % 256 == 0)
+  while ((M / 128) * SK < 224 && SK < 16 && (K / (SK * 2)) % 256 == 0)
     SK *= 2;
   if (SK == 1) return 0;
   const int n_tiles = (N + 15) / 16;
