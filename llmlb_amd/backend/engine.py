@@ -172,7 +172,8 @@ class _DecodeGraph:
             q, k, v, residual = eng.model.layer_attn_prelude(
                 li, h, residual, self.lens,
                 kv_sink=(None if eng.kv.fp8 else
-                         (eng.kv.k[li], eng.kv.v[li], self.slots)))
+                         (eng.kv.k[li], eng.kv.v[li], self.slots)),
+                quant_ok=True)
             if eng.kv.fp8:
                 eng._kv_append(li, k, v, self.slots)
             attn = ops.paged_attn_decode(
@@ -180,7 +181,8 @@ class _DecodeGraph:
                 eng.scale, n_parts=self.n_parts, workspace=self.ws,
                 k_scale=(eng.kv.k_scale[li] if eng.kv.fp8 else None),
                 v_scale=(eng.kv.v_scale[li] if eng.kv.fp8 else None))
-            h, residual = eng.model.layer_attn_epilogue(li, attn, residual)
+            h, residual = eng.model.layer_attn_epilogue(li, attn, residual,
+                                                        quant_ok=True)
         return eng.model.final(h, residual)
 
     def capture(self) -> None:
@@ -703,7 +705,8 @@ class Engine:
                 q, k, v, residual = self.model.layer_attn_prelude(
                     li, h, residual, lens,
                     kv_sink=(None if self.kv.fp8 else
-                             (self.kv.k[li], self.kv.v[li], slots)))
+                             (self.kv.k[li], self.kv.v[li], slots)),
+                    quant_ok=True)
                 if self.kv.fp8:
                     self._kv_append(li, k, v, slots)
                 attn = ops.paged_attn_decode(
@@ -713,8 +716,8 @@ class Engine:
                              else None),
                     v_scale=(self.kv.v_scale[li] if self.kv.fp8
                              else None))
-                h, residual = self.model.layer_attn_epilogue(li, attn,
-                                                             residual)
+                h, residual = self.model.layer_attn_epilogue(
+                    li, attn, residual, quant_ok=True)
             logits = self.model.final(h, residual)
         if self.timing is not None:
             torch.cuda.synchronize() if self.device.type == "cuda" else None

@@ -337,7 +337,8 @@ class LlamaForCausalLM(nn.Module):
 
     def layer_attn_prelude(self, layer_idx: int, h: torch.Tensor,
                            residual: torch.Tensor | None,
-                           positions: torch.Tensor, kv_sink=None):
+                           positions: torch.Tensor, kv_sink=None,
+                           quant_ok: bool = False):
         """norm -> qkv -> rope.  Returns (q, k, v, residual).
 
         ``kv_sink=(kcache, vcache, slots)`` fuses the paged-cache
@@ -345,7 +346,11 @@ class LlamaForCausalLM(nn.Module):
         instead of two); the caller must then skip its own append."""
         layer = self.layers_[layer_idx]
         qkv = None
-        if (ops.w8a8_active() and h.is_cuda and h.dim() == 2
+        # quant_ok: only the DECODE phase opts in — prefill chunks and
+        # spec-verify stacks that happen to have T <= 128 measured a
+        # 4.3x prefill slowdown on llama-3-70b (per-sequence quantize +
+        # split-K overhead on tiny batches; profiles/r02_w8a8.md)
+        if (quant_ok and ops.w8a8_active() and h.is_cuda and h.dim() == 2
                 and h.shape[0] <= 128 and h.dtype == torch.bfloat16):
             # W8A8 decode arm: the rmsnorm emits the quantized row +
             # per-token scale directly; the fp8 MFMA GEMM consumes it
@@ -373,12 +378,13 @@ class LlamaForCausalLM(nn.Module):
         return q, k, v, residual
 
     def layer_attn_epilogue(self, layer_idx: int, attn_out: torch.Tensor,
-                            residual: torch.Tensor):
+                            residual: torch.Tensor,
+                            quant_ok: bool = False):
         """o_proj -> +residual -> norm -> mlp.  Returns (h, residual)."""
         layer = self.layers_[layer_idx]
         T = attn_out.shape[0]
         o = layer.o_proj(attn_out.reshape(T, -1))
-        if (ops.w8a8_active() and o.is_cuda and T <= 128
+        if (quant_ok and ops.w8a8_active() and o.is_cuda and T <= 128
                 and o.dtype == torch.bfloat16):
             nq, xs_, residual = ops.rmsnorm_q(o, layer.post_norm_w,
                                               self.cfg.norm_eps,

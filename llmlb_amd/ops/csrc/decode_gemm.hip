@@ -898,7 +898,12 @@ typedef __attribute__((ext_vector_type(8))) int dg_i32x8;
 // sits in its barrier/LDS-write phase the other keeps the HBM pipe
 // full (the asm shows healthy counted waits, so the residual gap to
 // the bandwidth floor is barrier drain, not serialization).
-template <int N_TILES, bool ATOMIC, int WAVES = 8>
+// NOLDS: skip the LDS x staging entirely — each lane reads its own
+// 32-byte A fragments straight from global x (L2-resident: the whole
+// activation block is ~256 KB shared by every M block).  Removes the
+// per-chunk __syncthreads(), which PMC showed as the bound (waves
+// 53.5% parked in WAIT_ANY, HBM at only ~2.5 of ~6.3 TB/s).
+template <int N_TILES, bool ATOMIC, int WAVES = 8, bool NOLDS = false>
 __global__ __launch_bounds__(WAVES * 64, 2)
 void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            float* __restrict__ out_ws,       // [N, M]
@@ -965,7 +970,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                                 swz_kb * 16) = xv[p];                       \
     }                                                                       \
   } while (0)
-#define A8_CONSUME(b, bw)                                                   \
+#define A8_CONSUME(b, bw, kx)                                                   \
   do {                                                                      \
     _Pragma("unroll") for (int u = 0; u < S128; ++u) {                      \
       const int kk = u * 128 + koff;                                        \
@@ -973,10 +978,17 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
       _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
         const int tok = t * 16 + l_lo;                                      \
         w32 af;                                                             \
-        af.h[0] = *reinterpret_cast<const u32x4*>(                          \
-            xs[b] + (size_t)tok * CHUNK_K + (kb ^ (tok & 15)) * 16);        \
-        af.h[1] = *reinterpret_cast<const u32x4*>(                          \
-            xs[b] + (size_t)tok * CHUNK_K + ((kb + 1) ^ (tok & 15)) * 16);  \
+        if (NOLDS) {                                                        \
+          const unsigned char* xr =                                         \
+              xq + (size_t)(tok < N ? tok : 0) * K + kx + kk;               \
+          af.h[0] = *reinterpret_cast<const u32x4*>(xr);                    \
+          af.h[1] = *reinterpret_cast<const u32x4*>(xr + 16);               \
+        } else {                                                            \
+          af.h[0] = *reinterpret_cast<const u32x4*>(                        \
+              xs[b] + (size_t)tok * CHUNK_K + (kb ^ (tok & 15)) * 16);      \
+          af.h[1] = *reinterpret_cast<const u32x4*>(                        \
+              xs[b] + (size_t)tok * CHUNK_K + ((kb + 1) ^ (tok & 15)) * 16); \
+        }                                                                   \
         acc[t] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(          \
             af.v, bw[u].v, acc[t], 0, 0, 0, 127, 0, 127);                   \
       }                                                                     \
@@ -988,31 +1000,37 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   u32x4 xv[PIECES];
   const int NC = Ks / CHUNK_K;
 
-  A8_LOAD_X(xv, k_begin);
+  if (!NOLDS) A8_LOAD_X(xv, k_begin);
   A8_LOAD_W(wA, k_begin);
-  A8_WRITE_X(0, xv);
-  __syncthreads();
+  if (!NOLDS) {
+    A8_WRITE_X(0, xv);
+    __syncthreads();
+  }
 
   int c = 0;
   for (; c + 2 <= NC; c += 2) {
     const int k1 = k_begin + (c + 1) * CHUNK_K;
-    A8_LOAD_X(xv, k1);
+    if (!NOLDS) A8_LOAD_X(xv, k1);
     A8_LOAD_W(wB, k1);
-    A8_CONSUME(0, wA);
-    A8_WRITE_X(1, xv);
-    __syncthreads();
+    A8_CONSUME(0, wA, k_begin + c * CHUNK_K);
+    if (!NOLDS) {
+      A8_WRITE_X(1, xv);
+      __syncthreads();
+    }
     if (c + 2 < NC) {
       const int k2 = k_begin + (c + 2) * CHUNK_K;
-      A8_LOAD_X(xv, k2);
+      if (!NOLDS) A8_LOAD_X(xv, k2);
       A8_LOAD_W(wA, k2);
-      A8_CONSUME(1, wB);
-      A8_WRITE_X(0, xv);
-      __syncthreads();
+      A8_CONSUME(1, wB, k_begin + (c + 1) * CHUNK_K);
+      if (!NOLDS) {
+        A8_WRITE_X(0, xv);
+        __syncthreads();
+      }
     } else {
-      A8_CONSUME(1, wB);
+      A8_CONSUME(1, wB, k_begin + (c + 1) * CHUNK_K);
     }
   }
-  if (c < NC) A8_CONSUME(0, wA);
+  if (c < NC) A8_CONSUME(0, wA, k_begin + c * CHUNK_K);
 #undef A8_LOAD_W
 #undef A8_LOAD_X
 #undef A8_WRITE_X
@@ -1055,11 +1073,15 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
   // shape except gate_up, where the 8-wave M-tile-128 grid already
   // fills 224 CUs without split-K and wins by ~3%
   const int waves = wave_env ? wave_env : (M / 128 >= 224 ? 8 : 4);
+  static const bool nolds = [] {
+    const char* e = getenv("LLMLB_A8_NOLDS");
+    return e && e[0] == '1';
+  }();
   const size_t lds = (size_t)n_tiles * 16 * 512 * 2;  // byte x, 2 bufs
   if (lds > 160 * 1024) return -2;
   const int kc = K / 512;
   int SK = 1;          // set per-launch below (depends on the M tile)
-#define LAUNCH_A8W(NT, W)                                                   \
+#define LAUNCH_A8W_(NT, W, NL, LDSB)                                        \
   do {                                                                      \
     const int mblocks = M / (16 * (W));                                     \
     SK = 1;                                                                 \
@@ -1073,17 +1095,22 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
     if (SK > 1 && ws == nullptr) return -2;                                 \
     dim3 grid(mblocks, SK), block((W) * 64);                                \
     if (SK > 1)                                                             \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true, W>), grid, block, \
-                         lds, s, (bf16*)out, (float*)ws,                    \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true, W, NL>), grid,    \
+                         block, LDSB, s, (bf16*)out, (float*)ws,            \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
                          N, M, K, SK);                                      \
     else                                                                    \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W>), grid,       \
-                         block, lds, s, (bf16*)out, (float*)ws,             \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W, NL>), grid,   \
+                         block, LDSB, s, (bf16*)out, (float*)ws,            \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
                          N, M, K, SK);                                      \
+  } while (0)
+#define LAUNCH_A8W(NT, W)                                                   \
+  do {                                                                      \
+    if (nolds) LAUNCH_A8W_(NT, W, true, 0);                                 \
+    else LAUNCH_A8W_(NT, W, false, lds);                                    \
   } while (0)
 #define LAUNCH_A8(NT)                                                       \
   do {                                                                      \
@@ -1102,6 +1129,7 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
   }
 #undef LAUNCH_A8
 #undef LAUNCH_A8W
+#undef LAUNCH_A8W_
   if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;
