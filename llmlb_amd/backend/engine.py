@@ -166,6 +166,7 @@ class _DecodeGraph:
 
     def _forward(self) -> torch.Tensor:
         eng = self._engine
+        ops.set_decode_phase(True)   # captured: bakes the decode arms in
         h = eng.model.embed_tokens(self.ids)
         residual = None
         for li in range(eng.cfg.model.layers):
@@ -183,7 +184,9 @@ class _DecodeGraph:
                 v_scale=(eng.kv.v_scale[li] if eng.kv.fp8 else None))
             h, residual = eng.model.layer_attn_epilogue(li, attn, residual,
                                                         quant_ok=True)
-        return eng.model.final(h, residual)
+        out = eng.model.final(h, residual)
+        ops.set_decode_phase(False)
+        return out
 
     def capture(self) -> None:
         torch.cuda.synchronize()
@@ -699,6 +702,7 @@ class Engine:
         if graph is not None:
             logits = graph.run(ids, lens, bt, slots, seq_lens)
         else:
+            ops.set_decode_phase(True)
             h = self.model.embed_tokens(ids)
             residual = None
             for li in range(self.cfg.model.layers):
@@ -719,6 +723,7 @@ class Engine:
                 h, residual = self.model.layer_attn_epilogue(
                     li, attn, residual, quant_ok=True)
             logits = self.model.final(h, residual)
+            ops.set_decode_phase(False)
         if self.timing is not None:
             torch.cuda.synchronize() if self.device.type == "cuda" else None
             self.timing["forward"] += time.monotonic() - _t1

@@ -312,6 +312,18 @@ def decode_gemm_v2(x: torch.Tensor, weight: torch.Tensor,
 # EXPERIMENTAL: W8A16 measured SLOWER than bf16+hipBLASLt (the
 # e4m3->bf16 converts sit on the MFMA operand path —
 # profiles/r01_fp8_weights_ab.md).  Kept for A/B; default OFF.
+# engine-maintained phase flag: the fp8-weight (W8A16) decode arm and
+# similar decode-only dispatches must not trigger on small PREFILL
+# fragments (a 16-token prompt chunk has T <= 128 too) — measured 4.3x
+# prefill slowdown on llama-3-70b before the gate (profiles/r02_w8a8.md)
+_DECODE_PHASE = False
+
+
+def set_decode_phase(v: bool) -> None:
+    global _DECODE_PHASE
+    _DECODE_PHASE = bool(v)
+
+
 _FP8_WEIGHTS = os.environ.get("LLMLB_FP8_WEIGHTS",
                               "0").lower() in ("1", "true", "on")
 
@@ -421,7 +433,7 @@ def linear_decode(x: torch.Tensor, weight: torch.Tensor,
     N, K = x.shape
     M = weight.shape[0]
     shape_ok = N <= 128 and K % 256 == 0 and M % 128 == 0 and K >= 4096
-    if w8 is not None and shape_ok:
+    if w8 is not None and shape_ok and _DECODE_PHASE:
         # fp8 weights halve the dominant traffic: always take them
         try:
             return decode_gemm_v2(x, w8[0], w_scale=w8[1])
