@@ -921,8 +921,15 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   const int lane = threadIdx.x % WAVE;
   const int mblk = blockIdx.x;
   const int col_base = mblk * (WAVES * 16) + wid * 16;
-  const int Ks = K / SK;
-  const int k_begin = (int)blockIdx.y * Ks;
+  // balanced split-K over the B = K/512 chunks: slice y takes
+  // q + (y < r) chunks (SK need not divide B — e.g. qwen2-7b down has
+  // B = 37 prime, which a divisor-only scheme left at 56 blocks)
+  const int BCH = K / CHUNK_K;
+  const int qch = BCH / SK, rch = BCH % SK;
+  const int y = (int)blockIdx.y;
+  const int c_begin = y * qch + min(y, rch);
+  const int NC = qch + (y < rch ? 1 : 0);
+  const int k_begin = c_begin * CHUNK_K;
   const int l_lo = lane & 15;
   const int l_hi = lane >> 4;
   const int koff = l_hi * 32;               // 32 contiguous k per lane
@@ -998,7 +1005,6 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   union w32 { dg_i32x8 v; u32x4 h[2]; };
   w32 wA[S128], wB[S128];
   u32x4 xv[PIECES];
-  const int NC = Ks / CHUNK_K;
 
   if (!NOLDS) A8_LOAD_X(xv, k_begin);
   A8_LOAD_W(wA, k_begin);
@@ -1086,11 +1092,11 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
     const int mblocks = M / (16 * (W));                                     \
     SK = 1;                                                                 \
     if (mblocks < 224) {                                                    \
-      for (int d = 1; d <= kc && d <= 16; ++d) {                            \
-        if (kc % d) continue;                                               \
-        SK = d;                                                             \
-        if (mblocks * d >= 224) break;                                      \
-      }                                                                     \
+      SK = (224 + mblocks - 1) / mblocks;  /* minimal fill */               \
+      if (SK > 16) SK = 16;                                                 \
+      if (SK > kc) SK = kc;                                                 \
+      for (int d = SK; d <= 16 && d <= kc; ++d)                             \
+        if (kc % d == 0) { SK = d; break; }  /* prefer even slices */       \
     }                                                                       \
     if (SK > 1 && ws == nullptr) return -2;                                 \
     dim3 grid(mblocks, SK), block((W) * 64);                                \
