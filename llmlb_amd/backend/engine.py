@@ -184,7 +184,7 @@ class _DecodeGraph:
                 v_scale=(eng.kv.v_scale[li] if eng.kv.fp8 else None))
             h, residual = eng.model.layer_attn_epilogue(li, attn, residual,
                                                         quant_ok=True)
-        out = eng.model.final(h, residual)
+        out = eng.model.final(h, residual, quant_ok=True)
         ops.set_decode_phase(False)
         return out
 
@@ -237,6 +237,12 @@ class Engine:
             for mod in self.model.modules():
                 if hasattr(mod, "_w8_pair"):
                     mod._w8_pair(force=True)
+            if ops.w8a8_active() and self.model._lm8 is None:
+                try:
+                    self.model._lm8 = ops.quantize_fp8_weight(
+                        self.model.lm_head.data)
+                except Exception:
+                    pass
         m = cfg.model
         n_kv_local = m.kv_heads // self.model.tp.world_size
         import os as _os0
@@ -731,7 +737,7 @@ class Engine:
                              else None))
                 h, residual = self.model.layer_attn_epilogue(
                     li, attn, residual, quant_ok=True)
-            logits = self.model.final(h, residual)
+            logits = self.model.final(h, residual, quant_ok=True)
             ops.set_decode_phase(False)
         if self.timing is not None:
             torch.cuda.synchronize() if self.device.type == "cuda" else None

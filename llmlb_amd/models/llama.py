@@ -402,7 +402,27 @@ class LlamaForCausalLM(nn.Module):
         h = layer.down.forward_swiglu(gateup)
         return h, residual
 
-    def final(self, h: torch.Tensor, residual: torch.Tensor) -> torch.Tensor:
+    _lm8 = None
+
+    def final(self, h: torch.Tensor, residual: torch.Tensor,
+              quant_ok: bool = False) -> torch.Tensor:
+        # W8A8 lm_head on the decode path: the vocab projection is the
+        # single largest per-step weight stream (1.05 GB bf16 on the
+        # 128k llama-3 vocab) and the final rmsnorm is its natural
+        # quantizing producer
+        if (quant_ok and ops.w8a8_active() and h.is_cuda
+                and h.dim() == 2 and h.shape[0] <= 128
+                and h.dtype == torch.bfloat16):
+            if self._lm8 is None and self.lm_head.is_cuda:
+                self._lm8 = ops.quantize_fp8_weight(self.lm_head.data)
+            if self._lm8 is not None:
+                nq, xs_, _ = ops.rmsnorm_q(h, self.final_norm_w,
+                                           self.cfg.norm_eps, residual)
+                try:
+                    return ops.decode_gemm_w8a8(nq, xs_, self._lm8[0],
+                                                self._lm8[1])
+                except ValueError:
+                    pass
         normed, _ = ops.rmsnorm(h, self.final_norm_w, self.cfg.norm_eps,
                                 residual)
         return torch.nn.functional.linear(normed, self.lm_head)
