@@ -237,6 +237,8 @@ class Engine:
             for mod in self.model.modules():
                 if hasattr(mod, "_w8_pair"):
                     mod._w8_pair(force=True)
+                if hasattr(mod, "_w4_pair"):
+                    mod._w4_pair()
             if ops.w8a8_active() and self.model._lm8 is None:
                 try:
                     self.model._lm8 = ops.quantize_fp8_weight(

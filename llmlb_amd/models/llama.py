@@ -403,6 +403,7 @@ class LlamaForCausalLM(nn.Module):
         return h, residual
 
     _lm8 = None
+    _lm4 = None
 
     def final(self, h: torch.Tensor, residual: torch.Tensor,
               quant_ok: bool = False) -> torch.Tensor:
@@ -415,9 +416,20 @@ class LlamaForCausalLM(nn.Module):
                 and h.dtype == torch.bfloat16):
             if self._lm8 is None and self.lm_head.is_cuda:
                 self._lm8 = ops.quantize_fp8_weight(self.lm_head.data)
-            if self._lm8 is not None:
+            if (ops.w4_active() and self._lm4 is None
+                    and self.lm_head.is_cuda
+                    and self.lm_head.shape[1] % 512 == 0):
+                self._lm4 = ops.quantize_fp4_weight(self.lm_head.data)
+            if self._lm8 is not None or self._lm4 is not None:
                 nq, xs_, _ = ops.rmsnorm_q(h, self.final_norm_w,
                                            self.cfg.norm_eps, residual)
+                if self._lm4 is not None:
+                    try:
+                        return ops.decode_gemm_w4a8(
+                            nq, xs_, self._lm4[0], self._lm4[1],
+                            self.lm_head.shape[0])
+                    except ValueError:
+                        pass
                 try:
                     return ops.decode_gemm_w8a8(nq, xs_, self._lm8[0],
                                                 self._lm8[1])

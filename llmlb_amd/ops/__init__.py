@@ -46,6 +46,7 @@ def _try_load() -> None:
     _lib.ll_rmsnorm_q.restype = ctypes.c_int
     _lib.ll_swiglu_q.restype = ctypes.c_int
     _lib.ll_decode_gemm_w8a8.restype = ctypes.c_int
+    _lib.ll_decode_gemm_w4a8.restype = ctypes.c_int
     _lib.ll_decode_gemm_w8a8_ws_floats.restype = ctypes.c_longlong
     _lib.ll_paged_attn_decode.restype = ctypes.c_int
     _lib.ll_paged_attn_decode_fp8.restype = ctypes.c_int
@@ -332,6 +333,16 @@ _FP8_WEIGHTS = os.environ.get("LLMLB_FP8_WEIGHTS",
 # producing kernels (rmsnorm_q / swiglu_q).  Implies fp8 weights.
 # Opt-in: reduced model precision is a capacity/throughput trade the
 # operator must choose (the bf16 default serves the BASELINE bench).
+# MXFP4 weights under the A8 activation path (W4A8): block-scaled fp4
+# via the f8f6f4 MFMA's scale operands.  Requires LLMLB_FP8_A8=1 too.
+_FP4_WEIGHTS = os.environ.get("LLMLB_FP4_WEIGHTS",
+                              "0").lower() in ("1", "true", "on")
+
+
+def w4_active() -> bool:
+    return _FP4_WEIGHTS and _lib is not None
+
+
 _FP8_A8 = os.environ.get("LLMLB_FP8_A8",
                          "0").lower() in ("1", "true", "on")
 
@@ -403,6 +414,39 @@ def decode_gemm_w8a8(xq: torch.Tensor, x_scale: torch.Tensor,
     if rc == -2:
         raise ValueError("unsupported shape for decode_gemm_w8a8")
     _check(rc, "ll_decode_gemm_w8a8")
+    return out
+
+
+def quantize_fp4_weight(w: torch.Tensor):
+    """MXFP4 weight quantization (kernel register layout); see
+    reference.quantize_fp4_weight."""
+    return reference.quantize_fp4_weight(w)
+
+
+def decode_gemm_w4a8(xq: torch.Tensor, x_scale: torch.Tensor,
+                     w4: torch.Tensor, w4_scales: torch.Tensor,
+                     M: int) -> torch.Tensor:
+    """out[r][c] = x_scale[r] * (q_x . deq_mxfp4(w4)) via the
+    block-scaled fp8 x fp4 MFMA (per-32-value E8M0 scales applied by
+    the matrix core).  Raises ValueError on unsupported shapes."""
+    lib = _require_native()
+    N, K = xq.shape
+    out = torch.empty(N, M, dtype=torch.bfloat16, device=xq.device)
+    nws = int(lib.ll_decode_gemm_w8a8_ws_floats(N, M, K))
+    ws = None
+    if nws:
+        key = ("a8", N, M, K, str(xq.device))
+        ws = _WS_CACHE.get(key)
+        if ws is None:
+            ws = torch.zeros(nws, dtype=torch.float32, device=xq.device)
+            _WS_CACHE[key] = ws
+    rc = lib.ll_decode_gemm_w4a8(
+        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(xq),
+        _ptr(x_scale), _ptr(w4), _ptr(w4_scales), ctypes.c_int(N),
+        ctypes.c_int(M), ctypes.c_int(K), _stream())
+    if rc == -2:
+        raise ValueError("unsupported shape for decode_gemm_w4a8")
+    _check(rc, "ll_decode_gemm_w4a8")
     return out
 
 

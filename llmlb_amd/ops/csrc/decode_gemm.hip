@@ -903,7 +903,14 @@ typedef __attribute__((ext_vector_type(8))) int dg_i32x8;
 // activation block is ~256 KB shared by every M block).  Removes the
 // per-chunk __syncthreads(), which PMC showed as the bound (waves
 // 53.5% parked in WAIT_ANY, HBM at only ~2.5 of ~6.3 TB/s).
-template <int N_TILES, bool ATOMIC, int WAVES = 8, bool NOLDS = false>
+// WBITS=4: MXFP4 weights — e2m1 nibbles + per-32-value E8M0 block
+// scales fed through the instruction's scale_b operand (one MX block
+// per lane per MFMA; layout established by benchmarks/mfma128_probe).
+// Weight stream drops to K/2 bytes per row; dequantization is FREE
+// (done by the matrix core).  wq then points at the packed nibbles in
+// quantize_fp4_weight's register layout and w4sc at the E8M0 bytes.
+template <int N_TILES, bool ATOMIC, int WAVES = 8, bool NOLDS = false,
+          int WBITS = 8>
 __global__ __launch_bounds__(WAVES * 64, 2)
 void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            float* __restrict__ out_ws,       // [N, M]
@@ -911,6 +918,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            const float* __restrict__ x_scale,     // [N]
                            const unsigned char* __restrict__ wq,  // [M,K]
                            const float* __restrict__ w_scale,     // [M]
+                           const unsigned char* __restrict__ w4sc,
                            int N, int M, int K, int SK) {
   constexpr int CHUNK_K = 512;              // k (= bytes) per chunk
   constexpr int S128 = CHUNK_K / 128;       // 4 scaled-MFMA k-steps
@@ -937,8 +945,13 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xs[2] = {smem, smem + (size_t)ROWS * CHUNK_K};
 
-  const unsigned char* wrow = wq + (size_t)(col_base + l_lo) * K;
-  const float wsc = w_scale[col_base + l_lo];
+  const int SBLK = K / 128;                 // superblocks per row
+  const unsigned char* wrow =
+      WBITS == 4 ? wq + (size_t)(col_base + l_lo) * (K / 2)
+                 : wq + (size_t)(col_base + l_lo) * K;
+  const unsigned char* scrow =
+      WBITS == 4 ? w4sc + (size_t)(col_base + l_lo) * (K / 32) : nullptr;
+  const float wsc = WBITS == 4 ? 1.f : w_scale[col_base + l_lo];
 
   f32x4 acc[N_TILES];
 #pragma unroll
@@ -946,13 +959,22 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 
   const int piece_base = threadIdx.x;
 
-#define A8_LOAD_W(dst, k0)                                                  \
+#define A8_LOAD_W(dst, dsc, k0)                                             \
   do {                                                                      \
-    _Pragma("unroll") for (int u = 0; u < S128; ++u) {                      \
-      dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const       \
-          u32x4*>(wrow + (k0) + u * 128 + koff));                           \
-      dst[u].h[1] = __builtin_nontemporal_load(reinterpret_cast<const       \
-          u32x4*>(wrow + (k0) + u * 128 + koff + 16));                      \
+    if (WBITS == 4) {                                                       \
+      const int c0 = (k0) / CHUNK_K;                                        \
+      _Pragma("unroll") for (int u = 0; u < S128; ++u)                      \
+          dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const   \
+              u32x4*>(wrow + ((size_t)(c0 * 4 + u) * 64) + l_hi * 16));     \
+      dsc = __builtin_nontemporal_load(                                     \
+          reinterpret_cast<const u32x4*>(scrow + (size_t)c0 * 16));         \
+    } else {                                                                \
+      _Pragma("unroll") for (int u = 0; u < S128; ++u) {                    \
+        dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const     \
+            u32x4*>(wrow + (k0) + u * 128 + koff));                         \
+        dst[u].h[1] = __builtin_nontemporal_load(reinterpret_cast<const     \
+            u32x4*>(wrow + (k0) + u * 128 + koff + 16));                    \
+      }                                                                     \
     }                                                                       \
   } while (0)
 #define A8_LOAD_X(xv, k0)                                                   \
@@ -977,11 +999,13 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                                 swz_kb * 16) = xv[p];                       \
     }                                                                       \
   } while (0)
-#define A8_CONSUME(b, bw, kx)                                                   \
+#define A8_CONSUME(b, bw, bsc, kx)                                          \
   do {                                                                      \
     _Pragma("unroll") for (int u = 0; u < S128; ++u) {                      \
       const int kk = u * 128 + koff;                                        \
       const int kb = kk >> 4;                                               \
+      const int sbyte =                                                     \
+          WBITS == 4 ? (int)((bsc[u] >> (l_hi * 8)) & 0xFF) : 127;          \
       _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
         const int tok = t * 16 + l_lo;                                      \
         w32 af;                                                             \
@@ -996,18 +1020,30 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
           af.h[1] = *reinterpret_cast<const u32x4*>(                        \
               xs[b] + (size_t)tok * CHUNK_K + ((kb + 1) ^ (tok & 15)) * 16); \
         }                                                                   \
-        acc[t] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(          \
-            af.v, bw[u].v, acc[t], 0, 0, 0, 127, 0, 127);                   \
+        if (WBITS == 4)                                                     \
+          acc[t] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(        \
+              af.v, bw[u].v, acc[t], 0, 4, 0, 127, 0, sbyte);               \
+        else                                                                \
+          acc[t] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(        \
+              af.v, bw[u].v, acc[t], 0, 0, 0, 127, 0, 127);                 \
       }                                                                     \
     }                                                                       \
   } while (0)
 
   union w32 { dg_i32x8 v; u32x4 h[2]; };
   w32 wA[S128], wB[S128];
+  if (WBITS == 4) {   // hardware reads only the low 16 B per lane
+#pragma unroll
+    for (int u = 0; u < S128; ++u) {
+      wA[u].h[1] = u32x4{0, 0, 0, 0};
+      wB[u].h[1] = u32x4{0, 0, 0, 0};
+    }
+  }
+  u32x4 scA{0, 0, 0, 0}, scB{0, 0, 0, 0};
   u32x4 xv[PIECES];
 
   if (!NOLDS) A8_LOAD_X(xv, k_begin);
-  A8_LOAD_W(wA, k_begin);
+  A8_LOAD_W(wA, scA, k_begin);
   if (!NOLDS) {
     A8_WRITE_X(0, xv);
     __syncthreads();
@@ -1017,8 +1053,8 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
   for (; c + 2 <= NC; c += 2) {
     const int k1 = k_begin + (c + 1) * CHUNK_K;
     if (!NOLDS) A8_LOAD_X(xv, k1);
-    A8_LOAD_W(wB, k1);
-    A8_CONSUME(0, wA, k_begin + c * CHUNK_K);
+    A8_LOAD_W(wB, scB, k1);
+    A8_CONSUME(0, wA, scA, k_begin + c * CHUNK_K);
     if (!NOLDS) {
       A8_WRITE_X(1, xv);
       __syncthreads();
@@ -1026,17 +1062,17 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
     if (c + 2 < NC) {
       const int k2 = k_begin + (c + 2) * CHUNK_K;
       if (!NOLDS) A8_LOAD_X(xv, k2);
-      A8_LOAD_W(wA, k2);
-      A8_CONSUME(1, wB, k_begin + (c + 1) * CHUNK_K);
+      A8_LOAD_W(wA, scA, k2);
+      A8_CONSUME(1, wB, scB, k_begin + (c + 1) * CHUNK_K);
       if (!NOLDS) {
         A8_WRITE_X(0, xv);
         __syncthreads();
       }
     } else {
-      A8_CONSUME(1, wB, k_begin + (c + 1) * CHUNK_K);
+      A8_CONSUME(1, wB, scB, k_begin + (c + 1) * CHUNK_K);
     }
   }
-  if (c < NC) A8_CONSUME(0, wA, k_begin + c * CHUNK_K);
+  if (c < NC) A8_CONSUME(0, wA, scA, k_begin + c * CHUNK_K);
 #undef A8_LOAD_W
 #undef A8_LOAD_X
 #undef A8_WRITE_X
@@ -1105,13 +1141,13 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
                          block, LDSB, s, (bf16*)out, (float*)ws,            \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
-                         N, M, K, SK);                                      \
+                         (const unsigned char*)nullptr, N, M, K, SK);       \
     else                                                                    \
       hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W, NL>), grid,   \
                          block, LDSB, s, (bf16*)out, (float*)ws,            \
                          (const unsigned char*)xq, (const float*)x_scale,   \
                          (const unsigned char*)wq, (const float*)w_scale,   \
-                         N, M, K, SK);                                      \
+                         (const unsigned char*)nullptr, N, M, K, SK);       \
   } while (0)
 #define LAUNCH_A8W(NT, W)                                                   \
   do {                                                                      \
@@ -1136,6 +1172,72 @@ int ll_decode_gemm_w8a8(void* out, void* ws, const void* xq,
 #undef LAUNCH_A8
 #undef LAUNCH_A8W
 #undef LAUNCH_A8W_
+  if (SK > 1) {
+    long long n = (long long)N * M;
+    int thr = 256;
+    long long blocks2 = (n + thr - 1) / thr;
+    hipLaunchKernelGGL(f32_to_bf16_kernel, dim3((unsigned)blocks2),
+                       dim3(thr), 0, s, (bf16*)out, (float*)ws, n);
+  }
+  hipError_t e = hipGetLastError();
+  return e == hipSuccess ? 0 : (int)e;
+}
+
+// W4A8: MXFP4 weights (quantize_fp4_weight layout) x e4m3
+// activations.  Same grid/SK/wave policies as W8A8; the weight stream
+// is K/2 bytes per row and block scales ride the MFMA scale operand.
+int ll_decode_gemm_w4a8(void* out, void* ws, const void* xq,
+                        const void* x_scale, const void* w4,
+                        const void* w4_scales, int N, int M, int K,
+                        void* stream) {
+  if (K % 512 != 0 || M % 128 != 0 || N > 128) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  const int n_tiles = (N + 15) / 16;
+  static const int wave_env4 = [] {
+    const char* e = getenv("LLMLB_A8_WAVES");
+    return e ? (e[0] == '8' ? 8 : 4) : 0;
+  }();
+  const int waves = wave_env4 ? wave_env4 : (M / 128 >= 224 ? 8 : 4);
+  const size_t lds = (size_t)n_tiles * 16 * 512 * 2;
+  if (lds > 160 * 1024) return -2;
+  const int kc = K / 512;
+  int SK = 1;
+#define LAUNCH_W4(NT, W)                                                    \
+  do {                                                                      \
+    const int mblocks = M / (16 * (W));                                     \
+    SK = 1;                                                                 \
+    if (mblocks < 224) {                                                    \
+      SK = (224 + mblocks - 1) / mblocks;                                   \
+      if (SK > 16) SK = 16;                                                 \
+      if (SK > kc) SK = kc;                                                 \
+      for (int d = SK; d <= 16 && d <= kc; ++d)                             \
+        if (kc % d == 0) { SK = d; break; }                                 \
+    }                                                                       \
+    if (SK > 1 && ws == nullptr) return -2;                                 \
+    dim3 grid(mblocks, SK), block((W) * 64);                                \
+    if (SK > 1)                                                             \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true, W, false, 4>),    \
+                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
+                         (const unsigned char*)xq, (const float*)x_scale,   \
+                         (const unsigned char*)w4, (const float*)nullptr,   \
+                         (const unsigned char*)w4_scales, N, M, K, SK);     \
+    else                                                                    \
+      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W, false, 4>),   \
+                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
+                         (const unsigned char*)xq, (const float*)x_scale,   \
+                         (const unsigned char*)w4, (const float*)nullptr,   \
+                         (const unsigned char*)w4_scales, N, M, K, SK);     \
+  } while (0)
+  switch (n_tiles) {
+    case 1: if (waves == 8) LAUNCH_W4(1, 8); else LAUNCH_W4(1, 4); break;
+    case 2: if (waves == 8) LAUNCH_W4(2, 8); else LAUNCH_W4(2, 4); break;
+    case 3: if (waves == 8) LAUNCH_W4(3, 8); else LAUNCH_W4(3, 4); break;
+    case 4: if (waves == 8) LAUNCH_W4(4, 8); else LAUNCH_W4(4, 4); break;
+    case 5: case 6: LAUNCH_W4(6, 8); break;
+    case 7: case 8: LAUNCH_W4(8, 8); break;
+    default: return -2;
+  }
+#undef LAUNCH_W4
   if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;

@@ -163,3 +163,81 @@ def w8a8_matmul(xq, x_scale, wq, w_scale):
     xd = xq.view(torch.float8_e4m3fn).float() * x_scale[:, None]
     wd = wq.view(torch.float8_e4m3fn).float() * w_scale[:, None]
     return xd @ wd.T
+
+
+# ---------------------------------------------------------------------------
+# MXFP4 weights (W4A8): e2m1 nibbles + per-32-value E8M0 block scales in
+# the EXACT register layout of v_mfma_scale_f32_16x16x128_f8f6f4 with
+# blgp=4, established empirically by benchmarks/mfma128_probe.hip:
+#   lane g (= lane>>4) supplies nibble (byte j, half h) holding
+#   k_local = (g&1)*64 + (g>>1)*16 + (j>>3)*32 + (j&7)*2 + h
+# within each 128-k superblock; its E8M0 scale operand covers exactly
+# those 32 k (one MX block per lane per instruction).
+# ---------------------------------------------------------------------------
+
+_E2M1_VALS = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0]
+
+
+def _fp4_k_local():
+    """[4, 16, 2] int64: k_local for (g, byte j, nibble h)."""
+    g = torch.arange(4).view(4, 1, 1)
+    j = torch.arange(16).view(1, 16, 1)
+    h = torch.arange(2).view(1, 1, 2)
+    return ((g & 1) * 64 + (g >> 1) * 16 + (j >> 3) * 32 +
+            (j & 7) * 2 + h)
+
+
+def quantize_fp4_weight(w):
+    """[M, K] bf16/fp32 -> (w4 u8 [M, K//2], scales u8 [M, K//32]),
+    both in the kernel/register layout (NOT plain row-major nibbles):
+      w4[(m*S + s)*64 + g*16 + j] = nibbles of k_local(g, j, 0/1)
+      scales[(m*(S//4) + c)*16 + u*4 + g] = E8M0 of block (s=c*4+u, g)
+    K must be a multiple of 512 (4 superblocks per 512-k chunk)."""
+    M, K = w.shape
+    assert K % 512 == 0, K
+    S = K // 128
+    wf = w.float().view(M, S, 128)
+    kl = _fp4_k_local().to(w.device)                    # [4,16,2]
+    vals = wf[:, :, kl.reshape(-1)].view(M, S, 4, 32)   # [M,S,g,32]
+    amax = vals.abs().amax(dim=-1)                      # [M,S,4]
+    e = torch.ceil(torch.log2((amax / 6.0).clamp(min=2.0 ** -64)))
+    e = e.clamp(-64, 64)
+    scale = torch.pow(2.0, e)                           # [M,S,4]
+    q = vals / scale.unsqueeze(-1)
+    mags = torch.tensor(_E2M1_VALS, device=w.device)
+    mids = (mags[1:] + mags[:-1]) / 2                   # 7 midpoints
+    code = torch.bucketize(q.abs(), mids)               # [M,S,4,32] 0..7
+    nib = (code | ((q < 0).long() << 3)).to(torch.uint8)
+    nib = nib.view(M, S, 4, 16, 2)
+    packed = nib[..., 0] | (nib[..., 1] << 4)           # [M,S,4,16]
+    w4 = packed.reshape(M, K // 2).contiguous()
+    ebyte = (e + 127).to(torch.uint8)                   # [M,S,4]
+    scales = ebyte.view(M, S // 4, 4, 4).reshape(M, K // 32).contiguous()
+    return w4, scales
+
+
+def dequantize_fp4_weight(w4, scales, K):
+    """Inverse of quantize_fp4_weight -> float32 [M, K]."""
+    M = w4.shape[0]
+    S = K // 128
+    packed = w4.view(M, S, 4, 16)
+    nib0 = (packed & 15).long()
+    nib1 = (packed >> 4).long()
+    nib = torch.stack([nib0, nib1], dim=-1)             # [M,S,4,16,2]
+    mags = torch.tensor(_E2M1_VALS, device=w4.device)
+    v = mags[nib & 7] * torch.where(nib >= 8,
+                                    torch.tensor(-1.0), torch.tensor(1.0))
+    e = scales.view(M, S // 4, 4, 4).reshape(M, S, 4).float() - 127.0
+    v = v.view(M, S, 4, 32) * torch.pow(2.0, e).unsqueeze(-1)
+    kl = _fp4_k_local().to(w4.device).reshape(-1)       # [128]
+    out = torch.empty(M, S, 128)
+    out[:, :, kl] = v.view(M, S, 128)
+    return out.view(M, K)
+
+
+def w4a8_matmul(xq, x_scale, w4, w4_scales):
+    """Reference: dequantized fp32 matmul for the W4A8 kernel."""
+    K = xq.shape[1]
+    xf = xq.view(torch.float8_e4m3fn).float() * x_scale[:, None].float()
+    wf = dequantize_fp4_weight(w4.cpu(), w4_scales.cpu(), K)
+    return xf.cpu() @ wf.T

@@ -75,6 +75,21 @@ class _Fp8WeightMixin:
     and large batches keep the bf16 weights + hipBLASLt."""
 
     _w8 = None
+    _w4 = None
+
+    def _w4_pair(self):
+        """MXFP4 weight copy (LLMLB_FP4_WEIGHTS=1): packed e2m1
+        nibbles + E8M0 block scales for the W4A8 decode GEMM."""
+        from .. import ops as _ops
+        if not _ops.w4_active() or not self.weight.is_cuda \
+                or self.weight.dtype != torch.bfloat16:
+            return None
+        if self._w4 is None:
+            K = self.weight.shape[1]
+            if K % 512 != 0:
+                return None
+            self._w4 = _ops.quantize_fp4_weight(self.weight.data)
+        return self._w4
 
     def _w8_pair(self, force: bool = False):
         from .. import ops as _ops
@@ -86,12 +101,22 @@ class _Fp8WeightMixin:
         return self._w8
 
     def forward_w8a8(self, xq: torch.Tensor, x_scale: torch.Tensor):
-        """W8A8 decode GEMM on pre-quantized activations (rmsnorm_q /
-        swiglu_q outputs).  Returns None when the path does not apply
-        (caller recomputes via the bf16 path)."""
+        """W8A8 (or W4A8 under LLMLB_FP4_WEIGHTS=1) decode GEMM on
+        pre-quantized activations (rmsnorm_q / swiglu_q outputs).
+        Returns None when the path does not apply (caller recomputes
+        via the bf16 path)."""
         from .. import ops as _ops
+        if xq.shape[0] > 128:
+            return None
+        w4 = self._w4_pair()
+        if w4 is not None:
+            try:
+                return _ops.decode_gemm_w4a8(xq, x_scale, w4[0], w4[1],
+                                             self.weight.shape[0])
+            except ValueError:
+                pass
         pair = self._w8_pair(force=True)
-        if pair is None or xq.shape[0] > 128:
+        if pair is None:
             return None
         try:
             return _ops.decode_gemm_w8a8(xq, x_scale, pair[0], pair[1])

@@ -390,6 +390,27 @@ def test_decode_gemm_w8a8(N, M, K):
     assert rel < 5e-3, rel
 
 
+@pytest.mark.parametrize("N,M,K", [(64, 4096, 4096),    # split-K
+                                   (64, 28672, 4096),   # SK=1, 8-wave
+                                   (64, 4096, 14336),
+                                   (16, 6144, 4096),
+                                   (33, 4096, 4096)])
+def test_decode_gemm_w4a8(N, M, K):
+    """W4A8 (MXFP4 weights through the f8f6f4 MFMA's block-scale
+    operands) vs the dequantized fp32 reference."""
+    torch.manual_seed(21)
+    x = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.3
+    w = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) * 0.05
+    xq, xs = ref_mod.quantize_rows_fp8(x)
+    xq, xs = xq.to(DEV), xs.to(DEV)
+    w4, w4s = ref_mod.quantize_fp4_weight(w)
+    out = ops.decode_gemm_w4a8(xq, xs, w4.to(DEV), w4s.to(DEV), M)
+    ref32 = ref_mod.w4a8_matmul(xq.cpu(), xs.cpu(), w4.cpu(), w4s.cpu())
+    rel = ((out.float().cpu() - ref32).abs().max()
+           / ref32.abs().max().clamp(min=1e-6)).item()
+    assert rel < 5e-3, rel
+
+
 def test_rmsnorm_q_matches_reference():
     torch.manual_seed(13)
     T, H = 64, 4096

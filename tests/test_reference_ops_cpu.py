@@ -71,3 +71,28 @@ def test_paged_attn_reference_matches_sdpa():
             vs.transpose(0, 1).repeat_interleave(nq // nkv, 0),
         ).transpose(0, 1).squeeze(0)
         assert torch.allclose(out[s], ref, atol=1e-4), s
+
+
+def test_fp4_quantizer_round_trip():
+    """MXFP4 quantize -> dequantize: block-bounded error and exact
+    representability of in-grid values."""
+    import torch
+    from llmlb_amd.ops import reference as R
+    torch.manual_seed(5)
+    w = torch.randn(128, 1024) * 0.05
+    w4, sc = R.quantize_fp4_weight(w)
+    assert w4.shape == (128, 512) and sc.shape == (128, 32)
+    wd = R.dequantize_fp4_weight(w4, sc, 1024)
+    # per-element error bounded by half an e2m1 step at the block scale
+    S = 1024 // 128
+    e = sc.view(128, S // 4, 4, 4).reshape(128, S, 4).float() - 127.0
+    scale = torch.pow(2.0, e)                    # [M, S, 4]
+    kl = R._fp4_k_local().reshape(-1)
+    verr = (wd - w).view(128, S, 128)[:, :, kl].view(128, S, 4, 32).abs()
+    # largest e2m1 gap is 2.0 (4 -> 6); half-gap = 1.0 x scale
+    assert (verr <= scale.unsqueeze(-1) * 1.0 + 1e-6).all()
+    # exact grid values survive the round trip
+    g = torch.tensor([0.0, 0.5, 1.0, -1.5, 2.0, -3.0, 4.0, 6.0])
+    wg = g.repeat(2, 64)                         # [2, 512]
+    q4, qs = R.quantize_fp4_weight(wg)
+    assert torch.allclose(R.dequantize_fp4_weight(q4, qs, 512), wg)
