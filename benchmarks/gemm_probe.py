@@ -80,6 +80,12 @@ def main():
         def w8a8(a, b):
             return ops.decode_gemm_w8a8(xq, xs, w8, w8s)
 
+        w4, w4s = ops.quantize_fp4_weight(w)
+        w4, w4s = w4.to(x.device), w4s.to(x.device)
+
+        def w4a8(a, b):
+            return ops.decode_gemm_w4a8(xq, xs, w4, w4s, M)
+
         ref = torch.nn.functional.linear(x, w).float()
         t_blas = bench_one(torch.nn.functional.linear, x, w)
         res = {"shape": f"N{N} M{M} K{K}",
@@ -88,7 +94,8 @@ def main():
         res["hipblaslt_tbps"] = round(wbytes / t_blas / 1e12, 2)
         for name, fn, wb in (("v2", v2, wbytes), ("v3", v3, wbytes),
                              ("v3_fp8", v3_fp8, wbytes // 2),
-                             ("w8a8", w8a8, wbytes // 2)):
+                             ("w8a8", w8a8, wbytes // 2),
+                             ("w4a8", w4a8, wbytes // 4)):
             try:
                 t = bench_one(fn, x, w)
                 d = (fn(x, w).float() - ref).abs()
