@@ -234,15 +234,28 @@ class Engine:
             # built inside the first decode's graph capture, which the
             # first step (and any prefill-latency measurement around
             # it) absorbs — 0.3 s on llama-3-70b
+            # under W4 the fp8 weight copies are dead weight (the
+            # decode GEMMs take the fp4 path; only o_proj's W8A16 arm
+            # still wants w8 and builds it lazily) — on llama-3-70b
+            # pre-building both (140 GB bf16 + 70 GB fp8 + 17.5 GB
+            # fp4) left no room for the KV cache
+            w4_mode = ops.w4_active()
             for mod in self.model.modules():
-                if hasattr(mod, "_w8_pair"):
-                    mod._w8_pair(force=True)
-                if hasattr(mod, "_w4_pair"):
+                if w4_mode and hasattr(mod, "_w4_pair"):
                     mod._w4_pair()
-            if ops.w8a8_active() and self.model._lm8 is None:
+                elif not w4_mode and hasattr(mod, "_w8_pair"):
+                    mod._w8_pair(force=True)
+            if ops.w8a8_active():
                 try:
-                    self.model._lm8 = ops.quantize_fp8_weight(
-                        self.model.lm_head.data)
+                    if w4_mode:
+                        if (self.model._lm4 is None
+                                and self.model.lm_head.shape[1] % 512
+                                == 0):
+                            self.model._lm4 = ops.quantize_fp4_weight(
+                                self.model.lm_head.data)
+                    elif self.model._lm8 is None:
+                        self.model._lm8 = ops.quantize_fp8_weight(
+                            self.model.lm_head.data)
                 except Exception:
                     pass
         m = cfg.model

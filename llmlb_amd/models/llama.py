@@ -414,7 +414,8 @@ class LlamaForCausalLM(nn.Module):
         if (quant_ok and ops.w8a8_active() and h.is_cuda
                 and h.dim() == 2 and h.shape[0] <= 128
                 and h.dtype == torch.bfloat16):
-            if self._lm8 is None and self.lm_head.is_cuda:
+            if (self._lm8 is None and self.lm_head.is_cuda
+                    and not ops.w4_active()):
                 self._lm8 = ops.quantize_fp8_weight(self.lm_head.data)
             if (ops.w4_active() and self._lm4 is None
                     and self.lm_head.is_cuda
@@ -430,11 +431,13 @@ class LlamaForCausalLM(nn.Module):
                             self.lm_head.shape[0])
                     except ValueError:
                         pass
-                try:
-                    return ops.decode_gemm_w8a8(nq, xs_, self._lm8[0],
-                                                self._lm8[1])
-                except ValueError:
-                    pass
+                if self._lm8 is not None:
+                    try:
+                        return ops.decode_gemm_w8a8(nq, xs_,
+                                                    self._lm8[0],
+                                                    self._lm8[1])
+                    except ValueError:
+                        pass
         normed, _ = ops.rmsnorm(h, self.final_norm_w, self.cfg.norm_eps,
                                 residual)
         return torch.nn.functional.linear(normed, self.lm_head)
