@@ -463,3 +463,22 @@ def test_w8a8_engine_decode_smoke(monkeypatch):
                                                   max_new_tokens=16))
     assert len(res.token_ids) == 16
     assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
+
+
+def test_w4a8_engine_decode_smoke(monkeypatch):
+    """LLMLB_FP4_WEIGHTS on top of A8: the engine decodes through the
+    MXFP4 GEMMs (HW block scales) end to end."""
+    import llmlb_amd.ops as ops_mod
+    monkeypatch.setattr(ops_mod, "_FP8_A8", True)
+    monkeypatch.setattr(ops_mod, "_FP4_WEIGHTS", True)
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.launcher import PRESETS
+    from llmlb_amd.backend.sampler import SamplingParams
+    eng = Engine(EngineConfig(model=PRESETS["llama-small"](),
+                              device=DEV, kv_cache_blocks=512,
+                              seed=6))
+    res = eng.generate_sync([2, 4, 6, 8], max_new_tokens=16,
+                            params=SamplingParams(temperature=0.0,
+                                                  max_new_tokens=16))
+    assert len(res.token_ids) == 16
+    assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
