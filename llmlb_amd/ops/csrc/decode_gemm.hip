@@ -910,7 +910,7 @@ typedef __attribute__((ext_vector_type(8))) int dg_i32x8;
 // (done by the matrix core).  wq then points at the packed nibbles in
 // quantize_fp4_weight's register layout and w4sc at the E8M0 bytes.
 template <int N_TILES, bool ATOMIC, int WAVES = 8, bool NOLDS = false,
-          int WBITS = 8>
+          int WBITS = 8, int CHUNK_T = 512>
 __global__ __launch_bounds__(WAVES * 64, 2)
 void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            float* __restrict__ out_ws,       // [N, M]
@@ -920,7 +920,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
                            const float* __restrict__ w_scale,     // [M]
                            const unsigned char* __restrict__ w4sc,
                            int N, int M, int K, int SK) {
-  constexpr int CHUNK_K = 512;              // k (= bytes) per chunk
+  constexpr int CHUNK_K = CHUNK_T;          // k (= bytes of x) per chunk
   constexpr int S128 = CHUNK_K / 128;       // 4 scaled-MFMA k-steps
   constexpr int ROWS = N_TILES * 16;
   constexpr int TOTAL_P = ROWS * CHUNK_K / 16;
@@ -962,12 +962,13 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
 #define A8_LOAD_W(dst, dsc, k0)                                             \
   do {                                                                      \
     if (WBITS == 4) {                                                       \
-      const int c0 = (k0) / CHUNK_K;                                        \
+      const int c0 = (k0) / 128;       /* absolute superblock index */      \
       _Pragma("unroll") for (int u = 0; u < S128; ++u)                      \
           dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const   \
-              u32x4*>(wrow + ((size_t)(c0 * 4 + u) * 64) + l_hi * 16));     \
-      dsc = __builtin_nontemporal_load(                                     \
-          reinterpret_cast<const u32x4*>(scrow + (size_t)c0 * 16));         \
+              u32x4*>(wrow + ((size_t)(c0 + u) * 64) + l_hi * 16));         \
+      _Pragma("unroll") for (int q2 = 0; q2 < S128 / 4; ++q2)               \
+          dsc[q2] = __builtin_nontemporal_load(reinterpret_cast<const       \
+              u32x4*>(scrow + (size_t)c0 * 4 + q2 * 16));                   \
     } else {                                                                \
       _Pragma("unroll") for (int u = 0; u < S128; ++u) {                    \
         dst[u].h[0] = __builtin_nontemporal_load(reinterpret_cast<const     \
@@ -1005,7 +1006,8 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
       const int kk = u * 128 + koff;                                        \
       const int kb = kk >> 4;                                               \
       const int sbyte =                                                     \
-          WBITS == 4 ? (int)((bsc[u] >> (l_hi * 8)) & 0xFF) : 127;          \
+          WBITS == 4 ? (int)((bsc[u >> 2][u & 3] >> (l_hi * 8)) & 0xFF)     \
+                     : 127;                                                 \
       _Pragma("unroll") for (int t = 0; t < N_TILES; ++t) {                 \
         const int tok = t * 16 + l_lo;                                      \
         w32 af;                                                             \
@@ -1039,7 +1041,7 @@ void decode_gemm_a8_kernel(bf16* __restrict__ out,           // [N, M]
       wB[u].h[1] = u32x4{0, 0, 0, 0};
     }
   }
-  u32x4 scA{0, 0, 0, 0}, scB{0, 0, 0, 0};
+  u32x4 scA[S128 / 4] = {}, scB[S128 / 4] = {};
   u32x4 xv[PIECES];
 
   if (!NOLDS) A8_LOAD_X(xv, k_begin);
@@ -1198,11 +1200,19 @@ int ll_decode_gemm_w4a8(void* out, void* ws, const void* xq,
     return e ? (e[0] == '8' ? 8 : 4) : 0;
   }();
   const int waves = wave_env4 ? wave_env4 : (M / 128 >= 224 ? 8 : 4);
-  const size_t lds = (size_t)n_tiles * 16 * 512 * 2;
+  // 1024-k chunks double the in-flight W bytes and halve the barrier
+  // count (W4 streams only 64 B/lane per 512-chunk); A/B via env
+  static const bool big_chunk = [] {
+    const char* e = getenv("LLMLB_A8_CHUNK");
+    return e && e[0] == '1' && e[1] == '0';   // "1024"
+  }();
+  const int chunk = (big_chunk && K % 1024 == 0 && n_tiles <= 4)
+                        ? 1024 : 512;
+  const size_t lds = (size_t)n_tiles * 16 * chunk * 2;
   if (lds > 160 * 1024) return -2;
-  const int kc = K / 512;
+  const int kc = K / chunk;
   int SK = 1;
-#define LAUNCH_W4(NT, W)                                                    \
+#define LAUNCH_W4_(NT, W, CH)                                               \
   do {                                                                      \
     const int mblocks = M / (16 * (W));                                     \
     SK = 1;                                                                 \
@@ -1216,17 +1226,24 @@ int ll_decode_gemm_w4a8(void* out, void* ws, const void* xq,
     if (SK > 1 && ws == nullptr) return -2;                                 \
     dim3 grid(mblocks, SK), block((W) * 64);                                \
     if (SK > 1)                                                             \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, true, W, false, 4>),    \
-                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
-                         (const unsigned char*)xq, (const float*)x_scale,   \
-                         (const unsigned char*)w4, (const float*)nullptr,   \
-                         (const unsigned char*)w4_scales, N, M, K, SK);     \
+      hipLaunchKernelGGL(                                                   \
+          (decode_gemm_a8_kernel<NT, true, W, false, 4, CH>), grid,         \
+          block, lds, s, (bf16*)out, (float*)ws,                            \
+          (const unsigned char*)xq, (const float*)x_scale,                  \
+          (const unsigned char*)w4, (const float*)nullptr,                  \
+          (const unsigned char*)w4_scales, N, M, K, SK);                    \
     else                                                                    \
-      hipLaunchKernelGGL((decode_gemm_a8_kernel<NT, false, W, false, 4>),   \
-                         grid, block, lds, s, (bf16*)out, (float*)ws,       \
-                         (const unsigned char*)xq, (const float*)x_scale,   \
-                         (const unsigned char*)w4, (const float*)nullptr,   \
-                         (const unsigned char*)w4_scales, N, M, K, SK);     \
+      hipLaunchKernelGGL(                                                   \
+          (decode_gemm_a8_kernel<NT, false, W, false, 4, CH>), grid,        \
+          block, lds, s, (bf16*)out, (float*)ws,                            \
+          (const unsigned char*)xq, (const float*)x_scale,                  \
+          (const unsigned char*)w4, (const float*)nullptr,                  \
+          (const unsigned char*)w4_scales, N, M, K, SK);                    \
+  } while (0)
+#define LAUNCH_W4(NT, W)                                                    \
+  do {                                                                      \
+    if (chunk == 1024) LAUNCH_W4_(NT, W, 1024);                             \
+    else LAUNCH_W4_(NT, W, 512);                                            \
   } while (0)
   switch (n_tiles) {
     case 1: if (waves == 8) LAUNCH_W4(1, 8); else LAUNCH_W4(1, 4); break;
@@ -1238,6 +1255,7 @@ int ll_decode_gemm_w4a8(void* out, void* ws, const void* xq,
     default: return -2;
   }
 #undef LAUNCH_W4
+#undef LAUNCH_W4_
   if (SK > 1) {
     long long n = (long long)N * M;
     int thr = 256;
