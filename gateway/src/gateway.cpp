@@ -364,11 +364,15 @@ void ClientConn::process_requests() {
     int rc = parse_request_head(in.view(), req);
     if (rc == 0) return;
     if (rc < 0) {
+      cur_method.clear();
+      cur_path.clear();
       send_simple(400, "text/plain", "bad request");
       want_close = true;
       flush_out();
       return;
     }
+    cur_method.assign(req.method);
+    cur_path.assign(req.path);
     if (req.chunked) {
       send_json_error(411, "length_required",
                       "chunked request bodies are not supported");
@@ -464,13 +468,14 @@ void ClientConn::send_simple(int status, std::string_view content_type,
                              std::string_view extra_headers) {
   // every-request audit trail (excl. health/metrics/static —
   // audit/middleware.rs:23-45), buffered per worker
-  if (getenv("LLMLB_NO_AUDIT") == nullptr && !req.path.empty() && req.path != "/health" &&
-      req.path != "/api/health" && req.path != "/metrics" &&
-      req.path != "/api/metrics" && !starts_with(req.path, "/dashboard")) {
+  if (getenv("LLMLB_NO_AUDIT") == nullptr && !cur_path.empty() &&
+      cur_path != "/health" && cur_path != "/api/health" &&
+      cur_path != "/metrics" && cur_path != "/api/metrics" &&
+      !starts_with(cur_path, "/dashboard")) {
     AuditRecord a;
     a.ts_ms = wall_ms();
-    a.method = std::string(req.method);
-    a.path = std::string(req.path);
+    a.method = cur_method;
+    a.path = cur_path;
     a.status = status;
     a.client_ip = peer_ip;
     w->pending_audits.push_back(std::move(a));

@@ -217,6 +217,13 @@ class ClientConn : public FdHandler {
   int fd;
   std::string peer_ip;
   std::string request_id;   // X-Request-Id: client-supplied or generated
+  // OWNED copies of the parsed head's method/path: the parsed
+  // HttpRequest holds string_views INTO `in`, and any later socket
+  // read may reallocate it while an async op is in flight — the
+  // audit block in send_simple() read those views after the upstream
+  // round trip (heap-use-after-free under pipelined load; found by a
+  // 280 s ASAN soak at 100 conns)
+  std::string cur_method, cur_path;
   Buffer in, out;
   HttpRequest req;
   bool have_head = false;
