@@ -383,7 +383,21 @@ class LlamaForCausalLM(nn.Module):
         """o_proj -> +residual -> norm -> mlp.  Returns (h, residual)."""
         layer = self.layers_[layer_idx]
         T = attn_out.shape[0]
-        o = layer.o_proj(attn_out.reshape(T, -1))
+        o = None
+        if (quant_ok and ops.w8a8_active() and attn_out.is_cuda
+                and T <= 128 and attn_out.dtype == torch.bfloat16):
+            # o_proj joins the quantized path: one standalone row
+            # quantize of the attention output (its producer has no
+            # per-token amax), then the same W8A8/W4A8 GEMM
+            a2 = attn_out.reshape(T, -1)
+            if a2.is_contiguous():
+                oq, osc = ops.rows_q(a2)
+                o = layer.o_proj.forward_w8a8(oq, osc)
+                if o is not None:
+                    # RowParallelLinear: the mixin GEMM is shard-local
+                    o = layer.o_proj.tp.all_reduce(o)
+        if o is None:
+            o = layer.o_proj(attn_out.reshape(T, -1))
         if (quant_ok and ops.w8a8_active() and o.is_cuda and T <= 128
                 and o.dtype == torch.bfloat16):
             nq, xs_, residual = ops.rmsnorm_q(o, layer.post_norm_w,

@@ -45,6 +45,7 @@ def _try_load() -> None:
     _lib.ll_rope_kv_append.restype = ctypes.c_int
     _lib.ll_rmsnorm_q.restype = ctypes.c_int
     _lib.ll_swiglu_q.restype = ctypes.c_int
+    _lib.ll_rows_q.restype = ctypes.c_int
     _lib.ll_decode_gemm_w8a8.restype = ctypes.c_int
     _lib.ll_decode_gemm_w4a8.restype = ctypes.c_int
     _lib.ll_decode_gemm_w8a8_ws_floats.restype = ctypes.c_longlong
@@ -386,6 +387,23 @@ def swiglu_q(gateup: torch.Tensor):
     rc = lib.ll_swiglu_q(_ptr(q), _ptr(sc), _ptr(gateup),
                          ctypes.c_int(T), ctypes.c_int(inter), _stream())
     _check(rc, "ll_swiglu_q")
+    return q, sc
+
+
+def rows_q(x: torch.Tensor):
+    """Plain per-row e4m3 quantization (amax/448): returns
+    (q u8 [T, H], scale f32 [T]).  Quantizes the attention output so
+    o_proj can take the quantized decode GEMMs."""
+    if not x.is_cuda:
+        q, sc = reference.quantize_rows_fp8(x)
+        return q, sc
+    lib = _require_native()
+    T, H = x.shape
+    q = torch.empty(T, H, dtype=torch.uint8, device=x.device)
+    sc = torch.empty(T, dtype=torch.float32, device=x.device)
+    rc = lib.ll_rows_q(_ptr(q), _ptr(sc), _ptr(x), ctypes.c_int(T),
+                       ctypes.c_int(H), _stream())
+    _check(rc, "ll_rows_q")
     return q, sc
 
 

@@ -1293,7 +1293,62 @@ __global__ void swiglu_q_kernel(unsigned char* __restrict__ out_q,
   }
 }
 
+// plain row quantizer: amax + e4m3 pack, no norm — quantizes the
+// attention output so o_proj can join the W8A8/W4A8 decode path
+// (its producer is the attention kernel, which has no per-token
+// amax; this is the one standalone quantize launch in the arm)
+__global__ void rows_q_kernel(unsigned char* __restrict__ out_q,
+                              float* __restrict__ out_scale,
+                              const bf16* __restrict__ in, int hidden) {
+  const int row = blockIdx.x;
+  const uint4* inv = reinterpret_cast<const uint4*>(
+      in + (size_t)row * hidden);
+  const int nvec = hidden >> 3;
+  float amax = 0.f;
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float f[8];
+    unpack8(inv[i], f);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) amax = fmaxf(amax, fabsf(f[j]));
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    amax = fmaxf(amax, __shfl_xor(amax, off, WAVE));
+  __shared__ float warp_red[4];
+  const int wid = threadIdx.x / WAVE;
+  if ((threadIdx.x & (WAVE - 1)) == 0) warp_red[wid] = amax;
+  __syncthreads();
+  const int nw = blockDim.x / WAVE;
+  float amax_tot = 0.f;
+#pragma unroll
+  for (int w = 0; w < 4; ++w)
+    amax_tot = fmaxf(amax_tot, (w < nw) ? warp_red[w] : 0.f);
+  const float scale = amax_tot > 0.f ? amax_tot / 448.f : 1.f;
+  const float inv_scale = 1.f / scale;
+  if (threadIdx.x == 0) out_scale[row] = scale;
+  uint2* outv = reinterpret_cast<uint2*>(out_q + (size_t)row * hidden);
+  for (int i = threadIdx.x; i < nvec; i += blockDim.x) {
+    float f[8];
+    unpack8(inv[i], f);
+    uint2 q;
+    q.x = pack4_fp8(f, inv_scale);
+    q.y = pack4_fp8(f + 4, inv_scale);
+    outv[i] = q;
+  }
+}
+
 extern "C" {
+
+int ll_rows_q(void* out_q, void* out_scale, const void* in,
+              int n_tokens, int hidden, void* stream) {
+  if (hidden % 8 != 0) return -2;
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid(n_tokens), block(256);
+  hipLaunchKernelGGL(rows_q_kernel, grid, block, 0, s,
+                     (unsigned char*)out_q, (float*)out_scale,
+                     (const bf16*)in, hidden);
+  LL_CHECK(hipGetLastError());
+  return 0;
+}
 
 int ll_rmsnorm_q(void* out_q, void* out_scale, const void* in,
                  const void* weight, void* residual, int n_tokens,

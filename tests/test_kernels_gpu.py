@@ -465,6 +465,18 @@ def test_w8a8_engine_decode_smoke(monkeypatch):
     assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
 
 
+def test_rows_q_matches_reference():
+    torch.manual_seed(22)
+    x = torch.randn(64, 4096, dtype=torch.bfloat16, device=DEV)
+    q, sc = ops.rows_q(x)
+    deq = q.view(torch.float8_e4m3fn).float() * sc[:, None]
+    ref_q, ref_sc = ref_mod.quantize_rows_fp8(x.cpu())
+    ref_deq = ref_q.view(torch.float8_e4m3fn).float() * ref_sc[:, None]
+    rel = ((deq.cpu() - ref_deq).abs().max()
+           / ref_deq.abs().max().clamp(min=1e-6)).item()
+    assert rel < 1e-2, rel
+
+
 def test_w4a8_engine_decode_smoke(monkeypatch):
     """LLMLB_FP4_WEIGHTS on top of A8: the engine decodes through the
     MXFP4 GEMMs (HW block scales) end to end."""
