@@ -474,7 +474,11 @@ def test_rows_q_matches_reference():
     ref_deq = ref_q.view(torch.float8_e4m3fn).float() * ref_sc[:, None]
     rel = ((deq.cpu() - ref_deq).abs().max()
            / ref_deq.abs().max().clamp(min=1e-6)).item()
-    assert rel < 1e-2, rel
+    # the HW packer (v_cvt_pk_fp8_f32) and torch's float8 cast can
+    # round a near-amax value to adjacent e4m3 codes: one ulp at the
+    # top of the range is 32*scale = 7% of amax, so half-ulp
+    # disagreements reach ~3.6%
+    assert rel < 8e-2, rel
 
 
 def test_w4a8_engine_decode_smoke(monkeypatch):
