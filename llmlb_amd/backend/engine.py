@@ -166,27 +166,34 @@ class _DecodeGraph:
 
     def _forward(self) -> torch.Tensor:
         eng = self._engine
-        ops.set_decode_phase(True)   # captured: bakes the decode arms in
-        h = eng.model.embed_tokens(self.ids)
-        residual = None
-        for li in range(eng.cfg.model.layers):
-            q, k, v, residual = eng.model.layer_attn_prelude(
-                li, h, residual, self.lens,
-                kv_sink=(None if eng.kv.fp8 else
-                         (eng.kv.k[li], eng.kv.v[li], self.slots)),
-                quant_ok=True)
-            if eng.kv.fp8:
-                eng._kv_append(li, k, v, self.slots)
-            attn = ops.paged_attn_decode(
-                q, eng.kv.k[li], eng.kv.v[li], self.bt, self.seq_lens,
-                eng.scale, n_parts=self.n_parts, workspace=self.ws,
-                k_scale=(eng.kv.k_scale[li] if eng.kv.fp8 else None),
-                v_scale=(eng.kv.v_scale[li] if eng.kv.fp8 else None))
-            h, residual = eng.model.layer_attn_epilogue(li, attn, residual,
-                                                        quant_ok=True)
-        out = eng.model.final(h, residual, quant_ok=True)
-        ops.set_decode_phase(False)
-        return out
+        # captured: bakes the decode arms into the graph; reset in the
+        # finally below so an exception during capture cannot leave
+        # the flag set for a later prefill
+        ops.set_decode_phase(True)
+        try:
+            h = eng.model.embed_tokens(self.ids)
+            residual = None
+            for li in range(eng.cfg.model.layers):
+                q, k, v, residual = eng.model.layer_attn_prelude(
+                    li, h, residual, self.lens,
+                    kv_sink=(None if eng.kv.fp8 else
+                             (eng.kv.k[li], eng.kv.v[li], self.slots)),
+                    quant_ok=True)
+                if eng.kv.fp8:
+                    eng._kv_append(li, k, v, self.slots)
+                attn = ops.paged_attn_decode(
+                    q, eng.kv.k[li], eng.kv.v[li], self.bt,
+                    self.seq_lens, eng.scale, n_parts=self.n_parts,
+                    workspace=self.ws,
+                    k_scale=(eng.kv.k_scale[li] if eng.kv.fp8
+                             else None),
+                    v_scale=(eng.kv.v_scale[li] if eng.kv.fp8
+                             else None))
+                h, residual = eng.model.layer_attn_epilogue(
+                    li, attn, residual, quant_ok=True)
+            return eng.model.final(h, residual, quant_ok=True)
+        finally:
+            ops.set_decode_phase(False)
 
     def capture(self) -> None:
         torch.cuda.synchronize()
@@ -733,27 +740,29 @@ class Engine:
             logits = graph.run(ids, lens, bt, slots, seq_lens)
         else:
             ops.set_decode_phase(True)
-            h = self.model.embed_tokens(ids)
-            residual = None
-            for li in range(self.cfg.model.layers):
-                q, k, v, residual = self.model.layer_attn_prelude(
-                    li, h, residual, lens,
-                    kv_sink=(None if self.kv.fp8 else
-                             (self.kv.k[li], self.kv.v[li], slots)),
-                    quant_ok=True)
-                if self.kv.fp8:
-                    self._kv_append(li, k, v, slots)
-                attn = ops.paged_attn_decode(
-                    q, self.kv.k[li], self.kv.v[li], bt, seq_lens,
-                    self.scale,
-                    k_scale=(self.kv.k_scale[li] if self.kv.fp8
-                             else None),
-                    v_scale=(self.kv.v_scale[li] if self.kv.fp8
-                             else None))
-                h, residual = self.model.layer_attn_epilogue(
-                    li, attn, residual, quant_ok=True)
-            logits = self.model.final(h, residual, quant_ok=True)
-            ops.set_decode_phase(False)
+            try:
+                h = self.model.embed_tokens(ids)
+                residual = None
+                for li in range(self.cfg.model.layers):
+                    q, k, v, residual = self.model.layer_attn_prelude(
+                        li, h, residual, lens,
+                        kv_sink=(None if self.kv.fp8 else
+                                 (self.kv.k[li], self.kv.v[li], slots)),
+                        quant_ok=True)
+                    if self.kv.fp8:
+                        self._kv_append(li, k, v, slots)
+                    attn = ops.paged_attn_decode(
+                        q, self.kv.k[li], self.kv.v[li], bt, seq_lens,
+                        self.scale,
+                        k_scale=(self.kv.k_scale[li] if self.kv.fp8
+                                 else None),
+                        v_scale=(self.kv.v_scale[li] if self.kv.fp8
+                                 else None))
+                    h, residual = self.model.layer_attn_epilogue(
+                        li, attn, residual, quant_ok=True)
+                logits = self.model.final(h, residual, quant_ok=True)
+            finally:
+                ops.set_decode_phase(False)
         if self.timing is not None:
             torch.cuda.synchronize() if self.device.type == "cuda" else None
             self.timing["forward"] += time.monotonic() - _t1

@@ -69,10 +69,12 @@ def set_tp_context(ctx: TPContext) -> None:
 
 
 class _Fp8WeightMixin:
-    """Opt-in fp8 (e4m3) weight storage for the DECODE path
-    (LLMLB_FP8_WEIGHTS=1): per-row-quantized bytes + scales feed the
-    W8 arm of the v2 decode GEMM (half the weight traffic); prefill
-    and large batches keep the bf16 weights + hipBLASLt."""
+    """Opt-in quantized weight storage for the DECODE path:
+    LLMLB_FP8_WEIGHTS=1 -> per-row e4m3 bytes + scales (the W8A16 arm
+    of the staged decode GEMM, and the W8A8 MFMA path when
+    LLMLB_FP8_A8 is also set); LLMLB_FP4_WEIGHTS=1 -> MXFP4 blocks
+    for the W4A8 path.  Prefill and large batches always keep the
+    bf16 weights + hipBLASLt."""
 
     _w8 = None
     _w4 = None
