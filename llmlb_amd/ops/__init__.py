@@ -417,21 +417,30 @@ def decode_gemm_w8a8(xq: torch.Tensor, x_scale: torch.Tensor,
     N, K = xq.shape
     M = wq.shape[0]
     out = torch.empty(N, M, dtype=torch.bfloat16, device=xq.device)
-    nws = int(lib.ll_decode_gemm_w8a8_ws_floats(N, M, K))
-    ws = None
-    if nws:
-        key = ("a8", N, M, K, str(xq.device))
-        ws = _WS_CACHE.get(key)
-        if ws is None:
-            ws = torch.zeros(nws, dtype=torch.float32, device=xq.device)
-            _WS_CACHE[key] = ws
-    rc = lib.ll_decode_gemm_w8a8(
-        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(xq),
-        _ptr(x_scale), _ptr(wq), _ptr(w_scale), ctypes.c_int(N),
-        ctypes.c_int(M), ctypes.c_int(K), _stream())
-    if rc == -2:
-        raise ValueError("unsupported shape for decode_gemm_w8a8")
-    _check(rc, "ll_decode_gemm_w8a8")
+    # kernel tiles cap at 128 rows; larger decode buckets (bs 256)
+    # run as two row-halves into the same output (weights stream
+    # twice, but fp8/fp4 still totals less than one bf16 pass)
+    for lo in range(0, N, 128):
+        hi = min(lo + 128, N)
+        n = hi - lo
+        nws = int(lib.ll_decode_gemm_w8a8_ws_floats(n, M, K))
+        ws = None
+        if nws:
+            key = ("a8", n, M, K, str(xq.device))
+            ws = _WS_CACHE.get(key)
+            if ws is None:
+                ws = torch.zeros(nws, dtype=torch.float32,
+                                 device=xq.device)
+                _WS_CACHE[key] = ws
+        rc = lib.ll_decode_gemm_w8a8(
+            _ptr(out[lo:hi]), _ptr(ws) if ws is not None else None,
+            _ptr(xq[lo:hi]), ctypes.c_void_p(x_scale.data_ptr()
+                                             + lo * 4),
+            _ptr(wq), _ptr(w_scale), ctypes.c_int(n),
+            ctypes.c_int(M), ctypes.c_int(K), _stream())
+        if rc == -2:
+            raise ValueError("unsupported shape for decode_gemm_w8a8")
+        _check(rc, "ll_decode_gemm_w8a8")
     return out
 
 
@@ -450,21 +459,27 @@ def decode_gemm_w4a8(xq: torch.Tensor, x_scale: torch.Tensor,
     lib = _require_native()
     N, K = xq.shape
     out = torch.empty(N, M, dtype=torch.bfloat16, device=xq.device)
-    nws = int(lib.ll_decode_gemm_w8a8_ws_floats(N, M, K))
-    ws = None
-    if nws:
-        key = ("a8", N, M, K, str(xq.device))
-        ws = _WS_CACHE.get(key)
-        if ws is None:
-            ws = torch.zeros(nws, dtype=torch.float32, device=xq.device)
-            _WS_CACHE[key] = ws
-    rc = lib.ll_decode_gemm_w4a8(
-        _ptr(out), _ptr(ws) if ws is not None else None, _ptr(xq),
-        _ptr(x_scale), _ptr(w4), _ptr(w4_scales), ctypes.c_int(N),
-        ctypes.c_int(M), ctypes.c_int(K), _stream())
-    if rc == -2:
-        raise ValueError("unsupported shape for decode_gemm_w4a8")
-    _check(rc, "ll_decode_gemm_w4a8")
+    for lo in range(0, N, 128):     # see decode_gemm_w8a8
+        hi = min(lo + 128, N)
+        n = hi - lo
+        nws = int(lib.ll_decode_gemm_w8a8_ws_floats(n, M, K))
+        ws = None
+        if nws:
+            key = ("a8", n, M, K, str(xq.device))
+            ws = _WS_CACHE.get(key)
+            if ws is None:
+                ws = torch.zeros(nws, dtype=torch.float32,
+                                 device=xq.device)
+                _WS_CACHE[key] = ws
+        rc = lib.ll_decode_gemm_w4a8(
+            _ptr(out[lo:hi]), _ptr(ws) if ws is not None else None,
+            _ptr(xq[lo:hi]), ctypes.c_void_p(x_scale.data_ptr()
+                                             + lo * 4),
+            _ptr(w4), _ptr(w4_scales), ctypes.c_int(n),
+            ctypes.c_int(M), ctypes.c_int(K), _stream())
+        if rc == -2:
+            raise ValueError("unsupported shape for decode_gemm_w4a8")
+        _check(rc, "ll_decode_gemm_w4a8")
     return out
 
 

@@ -108,7 +108,7 @@ class _Fp8WeightMixin:
         Returns None when the path does not apply (caller recomputes
         via the bf16 path)."""
         from .. import ops as _ops
-        if xq.shape[0] > 128:
+        if xq.shape[0] > 256:
             return None
         w4 = self._w4_pair()
         if w4 is not None:
@@ -176,7 +176,7 @@ class RowParallelLinear(nn.Module, _Fp8WeightMixin):
         """swiglu_q (quantizing activation kernel) + W8A8 down GEMM;
         None when inapplicable."""
         from .. import ops as _ops
-        if gateup.shape[0] > 128 or not gateup.is_cuda:
+        if gateup.shape[0] > 256 or not gateup.is_cuda:
             return None
         q, sc = _ops.swiglu_q(gateup)
         y = self.forward_w8a8(q, sc)
