@@ -351,7 +351,7 @@ class LlamaForCausalLM(nn.Module):
         # 4.3x prefill slowdown on llama-3-70b (per-sequence quantize +
         # split-K overhead on tiny batches; profiles/r02_w8a8.md)
         if (quant_ok and ops.w8a8_active() and h.is_cuda and h.dim() == 2
-                and h.shape[0] <= 256 and h.dtype == torch.bfloat16):
+                and h.shape[0] <= 128 and h.dtype == torch.bfloat16):
             # W8A8 decode arm: the rmsnorm emits the quantized row +
             # per-token scale directly; the fp8 MFMA GEMM consumes it
             nq, xs_, residual = ops.rmsnorm_q(h, layer.input_norm_w,
@@ -385,7 +385,7 @@ class LlamaForCausalLM(nn.Module):
         T = attn_out.shape[0]
         o = None
         if (quant_ok and ops.w8a8_active() and attn_out.is_cuda
-                and T <= 256 and attn_out.dtype == torch.bfloat16):
+                and T <= 128 and attn_out.dtype == torch.bfloat16):
             # o_proj joins the quantized path: one standalone row
             # quantize of the attention output (its producer has no
             # per-token amax), then the same W8A8/W4A8 GEMM
@@ -398,7 +398,7 @@ class LlamaForCausalLM(nn.Module):
                     o = layer.o_proj.tp.all_reduce(o)
         if o is None:
             o = layer.o_proj(attn_out.reshape(T, -1))
-        if (quant_ok and ops.w8a8_active() and o.is_cuda and T <= 256
+        if (quant_ok and ops.w8a8_active() and o.is_cuda and T <= 128
                 and o.dtype == torch.bfloat16):
             nq, xs_, residual = ops.rmsnorm_q(o, layer.post_norm_w,
                                               self.cfg.norm_eps,
@@ -426,7 +426,7 @@ class LlamaForCausalLM(nn.Module):
         # 128k llama-3 vocab) and the final rmsnorm is its natural
         # quantizing producer
         if (quant_ok and ops.w8a8_active() and h.is_cuda
-                and h.dim() == 2 and h.shape[0] <= 256
+                and h.dim() == 2 and h.shape[0] <= 128
                 and h.dtype == torch.bfloat16):
             if (self._lm8 is None and self.lm_head.is_cuda
                     and not ops.w4_active()):
