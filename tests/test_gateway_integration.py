@@ -907,3 +907,75 @@ def test_connect_failure_retries_next_endpoint():
                 p.wait(timeout=5)
             except Exception:
                 pass
+
+
+def test_pipelined_request_does_not_dangle_head_views(stack):
+    """Regression for the r2 use-after-free: a pipelined second request
+    reallocates the connection's input buffer while an upstream op is
+    in flight; the audit path then read the FIRST request's parsed
+    method/path string_views (which pointed into the old buffer).
+    Fixed with owned per-request copies (gateway.cpp cur_method/path).
+    This drives the exact sequence: slow upstream -> op in flight ->
+    large pipelined request forces the realloc -> first response's
+    audit runs after it.  Under an ASAN build this test crashes the
+    old gateway; on a normal build it asserts both responses and a
+    live gateway (profiles/r02_gateway_uaf_fix.md)."""
+    import re as _re
+    base = stack["base"]
+    host, port = base.replace("http://", "").split(":")
+    # slow mock: 400 ms inference latency keeps the op window open
+    slow_port = _free_port()
+    slow = subprocess.Popen(
+        [str(BIN / "llmlb-mock"), "--port", str(slow_port), "--workers",
+         "1", "--models", "uaf-model", "--delay-ms", "400",
+         "--delay-inference-only"],
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+    try:
+        _req("POST", base + "/api/endpoints",
+             {"name": "uafslow", "base_url": f"http://127.0.0.1:{slow_port}"},
+             headers=stack["admin"])
+        deadline = time.time() + 15
+        while time.time() < deadline:
+            st, body = _req("GET", base + "/v1/models",
+                            headers=stack["auth"])
+            if st == 200 and "uaf-model" in body:
+                break
+            time.sleep(0.3)
+        assert "uaf-model" in body
+
+        key = stack["auth"]["Authorization"]
+        def chat_req(model, pad):
+            payload = json.dumps({
+                "model": model,
+                "messages": [{"role": "user", "content": "x" * pad}]})
+            return (f"POST /v1/chat/completions HTTP/1.1\r\n"
+                    f"Host: {host}\r\nAuthorization: {key}\r\n"
+                    f"Content-Type: application/json\r\n"
+                    f"Content-Length: {len(payload)}\r\n\r\n"
+                    ).encode() + payload.encode()
+
+        c = socket.create_connection((host, int(port)), timeout=10)
+        try:
+            # request A: slow upstream -> async op stays in flight
+            c.sendall(chat_req("uaf-model", 8))
+            time.sleep(0.1)          # let the gateway dispatch A
+            # request B: ~200 KB body pipelined on the same conn ->
+            # guaranteed input-buffer reallocation
+            c.sendall(chat_req("uaf-model", 200_000))
+            c.settimeout(15)
+            data = b""
+            while data.count(b"HTTP/1.1") < 2 or not data.endswith(b"}"):
+                chunk = c.recv(65536)
+                if not chunk:
+                    break
+                data = data + chunk
+            statuses = _re.findall(rb"HTTP/1\.1 (\d+)", data)
+            assert len(statuses) == 2, data[:200]
+            assert statuses[0] == b"200", data[:200]
+        finally:
+            c.close()
+        # the gateway survived and still serves
+        st, _ = _req("GET", base + "/health")
+        assert st == 200
+    finally:
+        slow.terminate()
