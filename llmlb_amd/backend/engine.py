@@ -247,10 +247,24 @@ class Engine:
             # pre-building both (140 GB bf16 + 70 GB fp8 + 17.5 GB
             # fp4) left no room for the KV cache
             w4_mode = ops.w4_active()
+            # fp8 prefill reuses the per-row w8 copies; under W4 they
+            # are otherwise unused, so only build them when the extra
+            # half-weight footprint clearly fits (llama-3-70b + W4 +
+            # w8 copies left no room for the KV cache)
+            want_w8 = not w4_mode
+            if w4_mode and ops.fp8_prefill_active():
+                try:
+                    free_b = torch.cuda.mem_get_info(self.device)[0]
+                    w8_bytes = sum(
+                        m.weight.numel() for m in self.model.modules()
+                        if hasattr(m, "_w8_pair"))
+                    want_w8 = w8_bytes < 0.3 * free_b
+                except Exception:
+                    want_w8 = False
             for mod in self.model.modules():
                 if w4_mode and hasattr(mod, "_w4_pair"):
                     mod._w4_pair()
-                elif not w4_mode and hasattr(mod, "_w8_pair"):
+                if want_w8 and hasattr(mod, "_w8_pair"):
                     mod._w8_pair(force=True)
             if ops.w8a8_active():
                 try:

@@ -344,6 +344,42 @@ def w4_active() -> bool:
     return _FP4_WEIGHTS and _lib is not None
 
 
+# fp8 PREFILL GEMMs (torch._scaled_mm -> hipBLASLt fp8, rowwise
+# scales): measured 1.45-2.0x the bf16 prefill GEMMs at T>=1024
+# (benchmarks/scaled_mm_probe.py).  Auto-enabled with the A8 arm
+# (same e4m3-activation numerics class); LLMLB_FP8_PREFILL=0 opts out.
+_FP8_PREFILL = os.environ.get("LLMLB_FP8_PREFILL",
+                              "1").lower() not in ("0", "false", "off")
+
+
+def fp8_prefill_active() -> bool:
+    return _FP8_A8 and _FP8_PREFILL and _lib is not None
+
+
+def scaled_mm_prefill(x: torch.Tensor, w8) -> torch.Tensor | None:
+    """Rowwise-scaled fp8 prefill GEMM: quantize x per token (rows_q),
+    reuse the per-row fp8 weight copies, dispatch hipBLASLt fp8 via
+    torch._scaled_mm.  Returns None when inapplicable."""
+    global _FP8_PREFILL
+    if w8 is None:
+        return None
+    wq, wsc = w8
+    T, K = x.shape
+    M = wq.shape[0]
+    if K % 16 != 0 or M % 16 != 0:
+        return None
+    xq, xs = rows_q(x)
+    try:
+        return torch._scaled_mm(
+            xq.view(torch.float8_e4m3fn),
+            wq.view(torch.float8_e4m3fn).t(),
+            scale_a=xs.unsqueeze(1), scale_b=wsc.unsqueeze(0),
+            out_dtype=torch.bfloat16)
+    except Exception:  # unsupported torch build: disable for the run
+        _FP8_PREFILL = False
+        return None
+
+
 _FP8_A8 = os.environ.get("LLMLB_FP8_A8",
                          "0").lower() in ("1", "true", "on")
 

@@ -149,6 +149,12 @@ class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
         if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
             y = linear_decode(x, self.weight, w8=self._w8_pair())
             return y + self.bias if self.bias is not None else y
+        from .. import ops as _ops
+        if (x.dim() == 2 and x.is_cuda and x.dtype == torch.bfloat16
+                and _ops.fp8_prefill_active()):
+            y = _ops.scaled_mm_prefill(x, self._w8_pair(force=True))
+            if y is not None:
+                return y + self.bias if self.bias is not None else y
         return torch.nn.functional.linear(x, self.weight, self.bias)
 
 
@@ -169,7 +175,14 @@ class RowParallelLinear(nn.Module, _Fp8WeightMixin):
         if x.dim() == 2 and x.shape[0] <= 128 and x.is_cuda:
             y = linear_decode(x, self.weight, w8=self._w8_pair())
         else:
-            y = torch.nn.functional.linear(x, self.weight)
+            from .. import ops as _ops
+            y = None
+            if (x.dim() == 2 and x.is_cuda
+                    and x.dtype == torch.bfloat16
+                    and _ops.fp8_prefill_active()):
+                y = _ops.scaled_mm_prefill(x, self._w8_pair(force=True))
+            if y is None:
+                y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)
 
     def forward_swiglu_w8a8(self, gateup: torch.Tensor):
