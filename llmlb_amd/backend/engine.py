@@ -261,6 +261,13 @@ class Engine:
                     want_w8 = w8_bytes < 0.3 * free_b
                 except Exception:
                     want_w8 = False
+            if w4_mode and not want_w8:
+                # make the guard authoritative for the whole run: with
+                # LLMLB_FP8_WEIGHTS also set, the decode branch's
+                # non-force _w8_pair() would lazily rebuild the full
+                # w8 set (~70 GB on llama-3-70b) during short prefill
+                # fragments, bypassing this decision
+                ops._FP8_WEIGHTS = False
             for mod in self.model.modules():
                 if w4_mode and hasattr(mod, "_w4_pair"):
                     mod._w4_pair()

@@ -560,3 +560,33 @@ def test_fp8_prefill_respects_w4_hbm_guard(monkeypatch):
     assert len(res.token_ids) == 4
     # the guard decision survived the prefill: no lazy w8 on W4 weights
     assert all(m._w8 is None for m in w4_served())
+
+
+def test_w4_guard_disables_w8_weight_arm(monkeypatch):
+    """All three quant envs set while the w8-copy guard trips: the
+    engine must switch the W8 weight arm off for the run, or the
+    decode branch's non-force _w8_pair() would lazily rebuild the
+    full w8 set during short prefill fragments."""
+    import llmlb_amd.ops as ops_mod
+    from llmlb_amd.backend.engine import Engine, EngineConfig
+    from llmlb_amd.backend.launcher import PRESETS
+    from llmlb_amd.backend.sampler import SamplingParams
+
+    monkeypatch.setattr(ops_mod, "_FP8_A8", True)
+    monkeypatch.setattr(ops_mod, "_FP8_WEIGHTS", True)
+    monkeypatch.setattr(ops_mod, "_FP4_WEIGHTS", True)
+    monkeypatch.setattr(torch.cuda, "mem_get_info",
+                        lambda *a, **k: (0, 0))
+    eng = Engine(EngineConfig(model=PRESETS["llama-small"](),
+                              device=DEV, kv_cache_blocks=512,
+                              seed=10))
+    assert not ops_mod._FP8_WEIGHTS
+    # short prompt: prefill fragment takes the T<=128 decode-shaped
+    # branch, which must no longer build w8 pairs
+    res = eng.generate_sync([3, 5, 7], max_new_tokens=4,
+                            params=SamplingParams(temperature=0.0,
+                                                  max_new_tokens=4))
+    assert len(res.token_ids) == 4
+    w4_served = [m for m in eng.model.modules()
+                 if getattr(m, "_w4", None) is not None]
+    assert w4_served and all(m._w8 is None for m in w4_served)
