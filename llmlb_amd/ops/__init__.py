@@ -345,21 +345,61 @@ def w4_active() -> bool:
 
 
 # fp8 PREFILL GEMMs (torch._scaled_mm -> hipBLASLt fp8, rowwise
-# scales): measured 1.45-2.0x the bf16 prefill GEMMs at T>=1024
-# (benchmarks/scaled_mm_probe.py).  Auto-enabled with the A8 arm
-# (same e4m3-activation numerics class); LLMLB_FP8_PREFILL=0 opts out.
+# scales): measured 1.45-2.0x the bf16 prefill GEMMs at T>=1024 on
+# llama shapes (benchmarks/scaled_mm_probe.py, +16.7% in-engine 8B)
+# but a 25% LOSS on qwen2-7b shapes (hipBLASLt's rowwise-fp8 kernel
+# is shape-sensitive) — so the choice is AUTO-TUNED per weight shape
+# at first use: both paths are timed once per (M, K) and the winner
+# is cached for the run.  Auto-enabled with the A8 arm (same
+# e4m3-activation numerics class); LLMLB_FP8_PREFILL=0 opts out.
 _FP8_PREFILL = os.environ.get("LLMLB_FP8_PREFILL",
                               "1").lower() not in ("0", "false", "off")
+_PREFILL_CHOICE: dict[tuple[int, int], bool] = {}
 
 
 def fp8_prefill_active() -> bool:
     return _FP8_A8 and _FP8_PREFILL and _lib is not None
 
 
-def scaled_mm_prefill(x: torch.Tensor, w8) -> torch.Tensor | None:
+def _scaled_mm_raw(x: torch.Tensor, wq, wsc) -> torch.Tensor:
+    xq, xs = rows_q(x)
+    return torch._scaled_mm(
+        xq.view(torch.float8_e4m3fn),
+        wq.view(torch.float8_e4m3fn).t(),
+        scale_a=xs.unsqueeze(1), scale_b=wsc.unsqueeze(0),
+        out_dtype=torch.bfloat16)
+
+
+def _tune_prefill(x: torch.Tensor, weight, wq, wsc) -> bool:
+    """Time the fp8 path against the bf16 GEMM on this exact call's
+    shapes (3 iters each after one warmup); runs ONCE per (M, K).
+    ~10 ms, paid inside the first long prefill of that shape."""
+    import torch.nn.functional as F
+    ev = [torch.cuda.Event(enable_timing=True) for _ in range(4)]
+    _scaled_mm_raw(x, wq, wsc)
+    F.linear(x, weight)
+    torch.cuda.synchronize(x.device)
+    ev[0].record()
+    for _ in range(3):
+        _scaled_mm_raw(x, wq, wsc)
+    ev[1].record()
+    ev[2].record()
+    for _ in range(3):
+        F.linear(x, weight)
+    ev[3].record()
+    torch.cuda.synchronize(x.device)
+    t_fp8 = ev[0].elapsed_time(ev[1])
+    t_bf16 = ev[2].elapsed_time(ev[3])
+    return t_fp8 < t_bf16
+
+
+def scaled_mm_prefill(x: torch.Tensor, w8,
+                      weight: torch.Tensor | None = None
+                      ) -> torch.Tensor | None:
     """Rowwise-scaled fp8 prefill GEMM: quantize x per token (rows_q),
     reuse the per-row fp8 weight copies, dispatch hipBLASLt fp8 via
-    torch._scaled_mm.  Returns None when inapplicable."""
+    torch._scaled_mm.  Returns None when inapplicable or when the
+    auto-tuner picked the bf16 GEMM for this weight shape."""
     global _FP8_PREFILL
     if w8 is None:
         return None
@@ -368,13 +408,17 @@ def scaled_mm_prefill(x: torch.Tensor, w8) -> torch.Tensor | None:
     M = wq.shape[0]
     if K % 16 != 0 or M % 16 != 0:
         return None
-    xq, xs = rows_q(x)
     try:
-        return torch._scaled_mm(
-            xq.view(torch.float8_e4m3fn),
-            wq.view(torch.float8_e4m3fn).t(),
-            scale_a=xs.unsqueeze(1), scale_b=wsc.unsqueeze(0),
-            out_dtype=torch.bfloat16)
+        use = _PREFILL_CHOICE.get((M, K))
+        if use is None:
+            if weight is None or T < 512:
+                # too small to be representative: don't lock a choice
+                return None
+            use = _tune_prefill(x, weight, wq, wsc)
+            _PREFILL_CHOICE[(M, K)] = use
+        if not use:
+            return None
+        return _scaled_mm_raw(x, wq, wsc)
     except Exception:  # unsupported torch build: disable for the run
         _FP8_PREFILL = False
         return None

@@ -155,7 +155,7 @@ class ColumnParallelLinear(nn.Module, _Fp8WeightMixin):
         # building them lazily here would bypass that HBM guard
         if (x.dim() == 2 and x.is_cuda and x.dtype == torch.bfloat16
                 and _ops.fp8_prefill_active() and self._w8 is not None):
-            y = _ops.scaled_mm_prefill(x, self._w8)
+            y = _ops.scaled_mm_prefill(x, self._w8, self.weight)
             if y is not None:
                 return y + self.bias if self.bias is not None else y
         return torch.nn.functional.linear(x, self.weight, self.bias)
@@ -185,7 +185,8 @@ class RowParallelLinear(nn.Module, _Fp8WeightMixin):
                     and x.dtype == torch.bfloat16
                     and _ops.fp8_prefill_active()
                     and self._w8 is not None):
-                y = _ops.scaled_mm_prefill(x, self._w8)
+                y = _ops.scaled_mm_prefill(x, self._w8,
+                                           self.weight)
             if y is None:
                 y = torch.nn.functional.linear(x, self.weight)
         return self.tp.all_reduce(y)

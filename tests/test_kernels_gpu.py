@@ -512,7 +512,9 @@ def test_fp8_prefill_engine_smoke(monkeypatch):
     from llmlb_amd.backend.sampler import SamplingParams
 
     assert ops_mod._FP8_PREFILL, "fp8 prefill should default on"
-    prompt = [(7 * i + 3) % 31999 for i in range(200)]
+    # >= 512 tokens so the per-shape auto-tuner runs (T<512 prefills
+    # stay bf16 — too small to lock a choice)
+    prompt = [(7 * i + 3) % 31999 for i in range(600)]
 
     monkeypatch.setattr(ops_mod, "_FP8_A8", True)
     monkeypatch.setattr(ops_mod, "_FP8_WEIGHTS", True)
@@ -523,11 +525,15 @@ def test_fp8_prefill_engine_smoke(monkeypatch):
     built = [m for m in eng.model.modules()
              if getattr(m, "_w8", None) is not None]
     assert built, "A8 engine init must prebuild w8 pairs"
+    monkeypatch.setattr(ops_mod, "_PREFILL_CHOICE", {})
     res = eng.generate_sync(prompt, max_new_tokens=8,
                             params=SamplingParams(temperature=0.0,
                                                   max_new_tokens=8))
     assert len(res.token_ids) == 8
     assert all(0 <= t < eng.cfg.model.vocab for t in res.token_ids)
+    # the per-shape auto-tuner ran and recorded choices for the
+    # prefill GEMM shapes it saw
+    assert ops_mod._PREFILL_CHOICE, "prefill auto-tuner did not run"
 
 
 def test_fp8_prefill_respects_w4_hbm_guard(monkeypatch):
