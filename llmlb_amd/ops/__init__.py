@@ -354,7 +354,11 @@ def w4_active() -> bool:
 # e4m3-activation numerics class); LLMLB_FP8_PREFILL=0 opts out.
 _FP8_PREFILL = os.environ.get("LLMLB_FP8_PREFILL",
                               "1").lower() not in ("0", "false", "off")
-_PREFILL_CHOICE: dict[tuple[int, int], bool] = {}
+# keyed by (M, K, pow2-bucket of T): hipBLASLt's rowwise-fp8 kernels
+# are non-monotonic in the row count (measured: +16.7% whole-prefill
+# at 8x2048 on llama-3-8b, 2.2x SLOWER at 8x6144), so the winner is
+# re-measured per row-count bucket, ~15 ms once per bucket
+_PREFILL_CHOICE: dict[tuple[int, int, int], bool] = {}
 
 
 def fp8_prefill_active() -> bool:
@@ -426,13 +430,14 @@ def scaled_mm_prefill(x: torch.Tensor, w8,
     if K % 16 != 0 or M % 16 != 0:
         return None
     try:
-        use = _PREFILL_CHOICE.get((M, K))
+        key = (M, K, max(T, 1).bit_length())
+        use = _PREFILL_CHOICE.get(key)
         if use is None:
             if weight is None or T < 512:
                 # too small to be representative: don't lock a choice
                 return None
             use = _tune_prefill(x, weight, wq, wsc)
-            _PREFILL_CHOICE[(M, K)] = use
+            _PREFILL_CHOICE[key] = use
         if not use:
             return None
         return _scaled_mm_raw(x, wq, wsc)
