@@ -414,32 +414,69 @@ def _tune_prefill(x: torch.Tensor, weight, wq, wsc) -> bool:
     return t_fp8 < t_bf16
 
 
+TUNE_BUCKETS = (1024, 4096, 16384, 32768)
+
+
+def tune_prefill_shapes(pairs, device) -> int:
+    """Tune the fp8-vs-bf16 prefill choice for every (weight, w8) pair
+    at the representative row counts in ``TUNE_BUCKETS``, filling
+    ``_PREFILL_CHOICE``.  Called ONCE at engine init: tuning inline
+    during a live prefill was measured to cost ~2 s — each of its
+    ``torch.cuda.synchronize`` calls drains the host's 32-layer
+    launch-ahead queue mid-prefill (profiles/r02_fp8_prefill_ab.md).
+    Returns the number of (shape, bucket) decisions recorded."""
+    global _FP8_PREFILL
+    done = set()
+    n = 0
+    for weight, (wq, wsc) in pairs:
+        M, K = wq.shape[0], weight.shape[1]
+        if (M, K) in done or K % 16 != 0 or M % 16 != 0:
+            continue
+        done.add((M, K))
+        for T in TUNE_BUCKETS:
+            key = (M, K, T.bit_length())
+            if key in _PREFILL_CHOICE:
+                n += 1
+                continue
+            x = torch.randn(T, K, dtype=torch.bfloat16, device=device)
+            try:
+                _PREFILL_CHOICE[key] = _tune_prefill(x, weight, wq, wsc)
+                n += 1
+            except Exception:
+                _FP8_PREFILL = False
+                return n
+            finally:
+                del x
+    return n
+
+
 def scaled_mm_prefill(x: torch.Tensor, w8,
                       weight: torch.Tensor | None = None
                       ) -> torch.Tensor | None:
     """Rowwise-scaled fp8 prefill GEMM: quantize x per token (rows_q),
     reuse the per-row fp8 weight copies, dispatch hipBLASLt fp8 via
-    torch._scaled_mm.  Returns None when inapplicable or when the
-    auto-tuner picked the bf16 GEMM for this weight shape."""
+    torch._scaled_mm.  Pure LOOKUP against the init-time tuning table
+    (nearest row-count bucket); returns None when inapplicable, when
+    the tuner picked bf16 for this shape, or when the shape was never
+    tuned — never measures (and never syncs) on the serving path."""
     global _FP8_PREFILL
     if w8 is None:
         return None
     wq, wsc = w8
     T, K = x.shape
     M = wq.shape[0]
-    if K % 16 != 0 or M % 16 != 0:
+    if K % 16 != 0 or M % 16 != 0 or T < 512:
+        return None
+    tb = max(T, 1).bit_length()
+    use = _PREFILL_CHOICE.get((M, K, tb))
+    if use is None:
+        cands = [k for k in _PREFILL_CHOICE if k[0] == M and k[1] == K]
+        if not cands:
+            return None
+        use = _PREFILL_CHOICE[min(cands, key=lambda k: abs(k[2] - tb))]
+    if not use:
         return None
     try:
-        key = (M, K, max(T, 1).bit_length())
-        use = _PREFILL_CHOICE.get(key)
-        if use is None:
-            if weight is None or T < 512:
-                # too small to be representative: don't lock a choice
-                return None
-            use = _tune_prefill(x, weight, wq, wsc)
-            _PREFILL_CHOICE[key] = use
-        if not use:
-            return None
         return _scaled_mm_raw(x, wq, wsc)
     except Exception:  # unsupported torch build: disable for the run
         _FP8_PREFILL = False
