@@ -414,7 +414,16 @@ def _tune_prefill(x: torch.Tensor, weight, wq, wsc) -> bool:
     return t_fp8 < t_bf16
 
 
-TUNE_BUCKETS = (1024, 4096, 16384, 32768)
+TUNE_BUCKETS = (1024, 4096, 16384)
+
+# fp8 prefill row cap: at 32768-row chunks the prefill loop is HOST
+# launch-bound (rocprof: 0.8-1.0 s of GPU idle in the chunk window for
+# BOTH arms) and the fp8 path's extra launches (rows_q per GEMM) tip
+# it further behind — measured 3.83 s vs 2.18 s bf16 on an 8x6144
+# prefill even though the fp8 GEMMs are faster in isolation at that
+# size.  <= 16384-row calls measured +84% (8x2048) and neutral on the
+# 16k-row second chunk (profiles/r02_fp8_prefill_ab.md).
+MAX_FP8_PREFILL_ROWS = 16384
 
 
 def tune_prefill_shapes(pairs, device) -> int:
@@ -465,7 +474,8 @@ def scaled_mm_prefill(x: torch.Tensor, w8,
     wq, wsc = w8
     T, K = x.shape
     M = wq.shape[0]
-    if K % 16 != 0 or M % 16 != 0 or T < 512:
+    if K % 16 != 0 or M % 16 != 0 or T < 512 \
+            or T > MAX_FP8_PREFILL_ROWS:
         return None
     tb = max(T, 1).bit_length()
     use = _PREFILL_CHOICE.get((M, K, tb))
