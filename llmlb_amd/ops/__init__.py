@@ -365,8 +365,33 @@ def fp8_prefill_active() -> bool:
     return _FP8_A8 and _FP8_PREFILL and _lib is not None
 
 
+# persistent per-K quantization workspaces for the prefill path: a
+# fresh 100-700 MB xq allocation per GEMM call, interleaved with the
+# chunk-2 math-path attention's ~13 GB/layer buffers, splits the
+# caching allocator's large blocks and forces hipMalloc/free-sync
+# churn (measured 2.3x whole-prefill regression at 8x6144).  One
+# buffer per K, sliced to the call's row count, removes the churn.
+_PREFILL_WS: dict[int, tuple[torch.Tensor, torch.Tensor]] = {}
+
+
+def _rows_q_ws(x: torch.Tensor):
+    T, K = x.shape
+    ws = _PREFILL_WS.get(K)
+    if ws is None or ws[0].shape[0] < T or ws[0].device != x.device:
+        rows = max(T, MAX_FP8_PREFILL_ROWS)
+        ws = (torch.empty(rows, K, dtype=torch.uint8, device=x.device),
+              torch.empty(rows, dtype=torch.float32, device=x.device))
+        _PREFILL_WS[K] = ws
+    q, sc = ws[0][:T], ws[1][:T]
+    lib = _require_native()
+    rc = lib.ll_rows_q(_ptr(q), _ptr(sc), _ptr(x), ctypes.c_int(T),
+                       ctypes.c_int(K), _stream())
+    _check(rc, "ll_rows_q")
+    return q, sc
+
+
 def _scaled_mm_raw(x: torch.Tensor, wq, wsc) -> torch.Tensor:
-    xq, xs = rows_q(x)
+    xq, xs = _rows_q_ws(x.contiguous())
     return torch._scaled_mm(
         xq.view(torch.float8_e4m3fn),
         wq.view(torch.float8_e4m3fn).t(),
