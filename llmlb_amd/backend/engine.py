@@ -661,6 +661,11 @@ class Engine:
         Returns the final (h, residual) of the chunk's tokens."""
         dev = self.device
         Ls = L - P
+        # fp8 prefill GEMMs only on P==0 chunks: continuation chunks
+        # run the mask-path SDPA (fp32 math fallback) whose ~13 GB
+        # buffers lose 30%+ wall time with the fp8 path interleaved
+        # (profiles/r02_fp8_prefill_ab.md, same-box 8x6144 A/B)
+        ops.set_prefill_fp8_ok(P == 0)
         ids = torch.tensor([s.prompt_ids[P:L] for s in group],
                            dtype=torch.long, device=dev).reshape(-1)
         positions = torch.arange(P, L, dtype=torch.int32, device=dev) \

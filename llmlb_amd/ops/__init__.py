@@ -450,6 +450,20 @@ TUNE_BUCKETS = (1024, 4096, 16384)
 # 16k-row second chunk (profiles/r02_fp8_prefill_ab.md).
 MAX_FP8_PREFILL_ROWS = 16384
 
+# fp8 prefill applies only to P==0 chunks (first/only chunk of a
+# prefill group).  Continuation chunks (P>0: cached prefix or chunk 2+
+# of a long prompt) run the mask-path SDPA whose fp32 math fallback
+# dominates the step AND loses 30%+ wall time when the fp8 GEMMs are
+# interleaved with its ~13 GB/layer buffers (same-box A/B at 8x6144);
+# single-chunk prefills measured +74-84% (profiles/r02_fp8_prefill_ab
+# .md).  The engine toggles this around each chunk.
+_PREFILL_FP8_OK = True
+
+
+def set_prefill_fp8_ok(v: bool) -> None:
+    global _PREFILL_FP8_OK
+    _PREFILL_FP8_OK = bool(v)
+
 
 def tune_prefill_shapes(pairs, device) -> int:
     """Tune the fp8-vs-bf16 prefill choice for every (weight, w8) pair
@@ -500,7 +514,7 @@ def scaled_mm_prefill(x: torch.Tensor, w8,
     T, K = x.shape
     M = wq.shape[0]
     if K % 16 != 0 or M % 16 != 0 or T < 512 \
-            or T > MAX_FP8_PREFILL_ROWS:
+            or T > MAX_FP8_PREFILL_ROWS or not _PREFILL_FP8_OK:
         return None
     tb = max(T, 1).bit_length()
     use = _PREFILL_CHOICE.get((M, K, tb))
