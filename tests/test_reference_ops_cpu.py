@@ -96,3 +96,54 @@ def test_fp4_quantizer_round_trip():
     wg = g.repeat(2, 64)                         # [2, 512]
     q4, qs = R.quantize_fp4_weight(wg)
     assert torch.allclose(R.dequantize_fp4_weight(q4, qs, 512), wg)
+
+
+def test_fp8_prefill_gating_logic(monkeypatch):
+    """The fp8-prefill dispatch is a pure lookup with hard gates; all
+    early-return paths are CUDA-free and must reject without touching
+    the fp8 kernels: row caps, the P==0 chunk flag, untuned shapes,
+    and nearest-bucket fallback (profiles/r02_fp8_prefill_ab.md)."""
+    import llmlb_amd.ops as ops
+
+    calls = []
+    monkeypatch.setattr(ops, "_scaled_mm_raw",
+                        lambda x, wq, wsc: calls.append(x.shape) or "HIT")
+    wq = torch.zeros(64, 32, dtype=torch.uint8)
+    wsc = torch.ones(64)
+    w8 = (wq, wsc)
+
+    def x(rows, k=32):
+        return torch.zeros(rows, k, dtype=torch.bfloat16)
+
+    # no w8 copies (HBM guard skipped them) -> bf16
+    assert ops.scaled_mm_prefill(x(1024), None) is None
+    # too few rows to have been representative at tune time
+    assert ops.scaled_mm_prefill(x(256), w8) is None
+    # above the host-launch-bound row cap
+    assert ops.scaled_mm_prefill(x(ops.MAX_FP8_PREFILL_ROWS + 1),
+                                 w8) is None
+    # K not 16-aligned
+    assert ops.scaled_mm_prefill(x(1024, 24), (torch.zeros(
+        64, 24, dtype=torch.uint8), wsc)) is None
+    # untuned shape -> bf16 (never measures on the serving path)
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE", {})
+    assert ops.scaled_mm_prefill(x(1024), w8) is None
+    # tuned: exact bucket hit
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE",
+                        {(64, 32, (1024).bit_length()): True})
+    assert ops.scaled_mm_prefill(x(1024), w8) == "HIT"
+    # nearest-bucket fallback (2048 rows, only the 1024 bucket tuned)
+    assert ops.scaled_mm_prefill(x(2048), w8) == "HIT"
+    # tuner picked bf16 for the shape
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE",
+                        {(64, 32, (1024).bit_length()): False})
+    assert ops.scaled_mm_prefill(x(1024), w8) is None
+    # continuation chunk (P>0): engine clears the flag
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE",
+                        {(64, 32, (1024).bit_length()): True})
+    ops.set_prefill_fp8_ok(False)
+    try:
+        assert ops.scaled_mm_prefill(x(1024), w8) is None
+    finally:
+        ops.set_prefill_fp8_ok(True)
+    assert calls == [(1024, 32), (2048, 32)]
