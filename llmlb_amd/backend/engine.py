@@ -700,11 +700,23 @@ class Engine:
                 vh = torch.cat([vpre.view(B, P, nkv, D).to(vh.dtype), vh],
                                dim=1)
             kh, vh = kh.transpose(1, 2), vh.transpose(1, 2)
-            o = torch.nn.functional.scaled_dot_product_attention(
-                qh, kh, vh,
-                attn_mask=mask if P else None,
-                is_causal=Ls > 1 if not P else False,
-                enable_gqa=True)
+            if P:
+                # masked SDPA + enable_gqa has no fused kernel on ROCm
+                # (the head-count mismatch forces torch's fp32 math
+                # path: 35.4 ms/layer and ~13 GB of transient fp32 at
+                # the 8x6144 shape).  Repeating KV to the full head
+                # count keeps the efficient kernel: 6.1 ms/layer
+                # measured (benchmarks/gqa_mask_probe.py); the ~50 us
+                # copy is noise against the 29 ms saved.
+                G = nq // nkv
+                if G > 1:
+                    kh = kh.repeat_interleave(G, dim=1)
+                    vh = vh.repeat_interleave(G, dim=1)
+                o = torch.nn.functional.scaled_dot_product_attention(
+                    qh, kh, vh, attn_mask=mask)
+            else:
+                o = torch.nn.functional.scaled_dot_product_attention(
+                    qh, kh, vh, is_causal=Ls > 1, enable_gqa=True)
             attn = o.transpose(1, 2).reshape(B * Ls, nq, D).contiguous()
             h, residual = self.model.layer_attn_epilogue(li, attn,
                                                          residual)
