@@ -108,6 +108,9 @@ def test_fp8_prefill_gating_logic(monkeypatch):
     calls = []
     monkeypatch.setattr(ops, "_scaled_mm_raw",
                         lambda x, wq, wsc: calls.append(x.shape) or "HIT")
+    # the engine sets the chunk flag per prefill chunk and may leave
+    # it False after a P>0 chunk (harmless outside prefill); pin it
+    monkeypatch.setattr(ops, "_PREFILL_FP8_OK", True)
     wq = torch.zeros(64, 32, dtype=torch.uint8)
     wsc = torch.ones(64)
     w8 = (wq, wsc)
@@ -141,9 +144,6 @@ def test_fp8_prefill_gating_logic(monkeypatch):
     # continuation chunk (P>0): engine clears the flag
     monkeypatch.setattr(ops, "_PREFILL_CHOICE",
                         {(64, 32, (1024).bit_length()): True})
-    ops.set_prefill_fp8_ok(False)
-    try:
-        assert ops.scaled_mm_prefill(x(1024), w8) is None
-    finally:
-        ops.set_prefill_fp8_ok(True)
+    monkeypatch.setattr(ops, "_PREFILL_FP8_OK", False)
+    assert ops.scaled_mm_prefill(x(1024), w8) is None
     assert calls == [(1024, 32), (2048, 32)]
