@@ -147,3 +147,38 @@ def test_fp8_prefill_gating_logic(monkeypatch):
     monkeypatch.setattr(ops, "_PREFILL_FP8_OK", False)
     assert ops.scaled_mm_prefill(x(1024), w8) is None
     assert calls == [(1024, 32), (2048, 32)]
+
+
+def test_tune_prefill_shapes_contract(monkeypatch):
+    """Engine-init tuning fills one decision per (M, K, bucket) for
+    every distinct weight shape, dedupes repeated shapes, and skips
+    non-16-aligned shapes; a tuner failure disables the arm instead
+    of raising into engine init."""
+    import llmlb_amd.ops as ops
+
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE", {})
+    monkeypatch.setattr(ops, "_tune_prefill",
+                        lambda x, w, wq, wsc: x.shape[0] <= 4096)
+    def pair(m, k):
+        return (torch.zeros(m, k, dtype=torch.bfloat16),
+                (torch.zeros(m, k, dtype=torch.uint8), torch.ones(m)))
+
+    n = ops.tune_prefill_shapes(
+        [pair(64, 32), pair(64, 32), pair(48, 64), pair(50, 24)], "cpu")
+    # 2 distinct aligned shapes x all buckets; (50, 24) skipped (K%16)
+    assert n == 2 * len(ops.TUNE_BUCKETS)
+    for (m, k) in [(64, 32), (48, 64)]:
+        for t in ops.TUNE_BUCKETS:
+            assert ops._PREFILL_CHOICE[(m, k, t.bit_length())] == \
+                (t <= 4096)
+    assert (50, 24, ops.TUNE_BUCKETS[0].bit_length()) \
+        not in ops._PREFILL_CHOICE
+
+    # tuner failure -> arm disabled for the run, no raise
+    monkeypatch.setattr(ops, "_PREFILL_CHOICE", {})
+    monkeypatch.setattr(ops, "_FP8_PREFILL", True)
+    def boom(x, w, wq, wsc):
+        raise RuntimeError("no fp8 support")
+    monkeypatch.setattr(ops, "_tune_prefill", boom)
+    ops.tune_prefill_shapes([pair(64, 32)], "cpu")
+    assert not ops._FP8_PREFILL
