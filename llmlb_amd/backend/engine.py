@@ -987,8 +987,14 @@ class Engine:
             qh = q.view(B, F, nq, D).transpose(1, 2)
             kh = kg.view(B, Lmax, nkv, D).to(q.dtype).transpose(1, 2)
             vh = vg.view(B, Lmax, nkv, D).to(q.dtype).transpose(1, 2)
+            # masked SDPA + enable_gqa has no fused ROCm kernel (falls
+            # to the fp32 math path); repeat KV to full heads instead
+            # — same fix as the chunked-prefill attention above
+            if nq != nkv:
+                kh = kh.repeat_interleave(nq // nkv, dim=1)
+                vh = vh.repeat_interleave(nq // nkv, dim=1)
             o = torch.nn.functional.scaled_dot_product_attention(
-                qh, kh, vh, attn_mask=mask, enable_gqa=True)
+                qh, kh, vh, attn_mask=mask)
             attn = o.transpose(1, 2).reshape(B * F, nq, D).contiguous()
             h, residual = self.model.layer_attn_epilogue(li, attn,
                                                          residual)
