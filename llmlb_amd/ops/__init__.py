@@ -345,19 +345,18 @@ def w4_active() -> bool:
 
 
 # fp8 PREFILL GEMMs (torch._scaled_mm -> hipBLASLt fp8, rowwise
-# scales): measured 1.45-2.0x the bf16 prefill GEMMs at T>=1024 on
-# llama shapes (benchmarks/scaled_mm_probe.py, +16.7% in-engine 8B)
-# but a 25% LOSS on qwen2-7b shapes (hipBLASLt's rowwise-fp8 kernel
-# is shape-sensitive) — so the choice is AUTO-TUNED per weight shape
-# at first use: both paths are timed once per (M, K) and the winner
-# is cached for the run.  Auto-enabled with the A8 arm (same
-# e4m3-activation numerics class); LLMLB_FP8_PREFILL=0 opts out.
+# scales): +86% whole-prefill on llama-3-8b and +76% on qwen2-7b at
+# 8x2048 (profiles/r02_fp8_prefill_ab.md — read the measured arc
+# before changing ANY of the policy pieces below).  The fp8-vs-bf16
+# choice is tuned per (M, K, row-bucket) at ENGINE INIT
+# (tune_prefill_shapes); the serving path is a pure table lookup.
+# Auto-enabled with the A8 arm (same e4m3-activation numerics
+# class); LLMLB_FP8_PREFILL=0 opts out.
 _FP8_PREFILL = os.environ.get("LLMLB_FP8_PREFILL",
                               "1").lower() not in ("0", "false", "off")
 # keyed by (M, K, pow2-bucket of T): hipBLASLt's rowwise-fp8 kernels
-# are non-monotonic in the row count (measured: +16.7% whole-prefill
-# at 8x2048 on llama-3-8b, 2.2x SLOWER at 8x6144), so the winner is
-# re-measured per row-count bucket, ~15 ms once per bucket
+# are shape- and row-count-sensitive, so the winner is measured per
+# row-count bucket (~15 ms per bucket, once, at engine init)
 _PREFILL_CHOICE: dict[tuple[int, int, int], bool] = {}
 
 
@@ -441,13 +440,12 @@ def _tune_prefill(x: torch.Tensor, weight, wq, wsc) -> bool:
 
 TUNE_BUCKETS = (1024, 4096, 16384)
 
-# fp8 prefill row cap: at 32768-row chunks the prefill loop is HOST
-# launch-bound (rocprof: 0.8-1.0 s of GPU idle in the chunk window for
-# BOTH arms) and the fp8 path's extra launches (rows_q per GEMM) tip
-# it further behind — measured 3.83 s vs 2.18 s bf16 on an 8x6144
-# prefill even though the fp8 GEMMs are faster in isolation at that
-# size.  <= 16384-row calls measured +84% (8x2048) and neutral on the
-# 16k-row second chunk (profiles/r02_fp8_prefill_ab.md).
+# fp8 prefill row cap: 32k-row chunks regressed in-engine while the
+# fp8 GEMMs win in isolation at that size; the regression was later
+# traced to the chunk-2 math-attention interleaving (now gated by
+# _PREFILL_FP8_OK), so this cap is conservative and unproven — first
+# thing to re-measure next round (profiles/r02_fp8_prefill_ab.md).
+# <= 16384-row calls measured +74-84% at 8x2048.
 MAX_FP8_PREFILL_ROWS = 16384
 
 # fp8 prefill applies only to P==0 chunks (first/only chunk of a
