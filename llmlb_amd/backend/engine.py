@@ -275,9 +275,10 @@ class Engine:
                     mod._w8_pair(force=True)
             if ops.fp8_prefill_active():
                 # fill the fp8-vs-bf16 prefill table now (one-time,
-                # ~0.3 s); measuring inline during a live prefill
-                # drains the launch-ahead queue and was measured to
-                # cost ~2 s on a 49k-token prefill
+                # ~1.5 s incl. hipBLASLt algo search); measuring
+                # inline during a live prefill drains the launch-ahead
+                # queue and was measured to cost ~2 s per 49k-token
+                # prefill
                 pairs = [(m.weight, m._w8)
                          for m in self.model.modules()
                          if getattr(m, "_w8", None) is not None]
