@@ -189,6 +189,26 @@ std::string Gateway::render_metrics_text() {
 // capability-routed backend pick: FIRST online endpoint with the
 // capability (reference api/audio.rs:160-190 — deliberately no TPS
 // weighting on the audio/image paths)
+// ModelStatEntry rows -> JSON (reference db/endpoint_daily_stats.rs:25-37):
+// rows are (model, requests, failures, completion_tokens, duration_ms)
+static Json model_stats_json(const std::vector<DbRow>& rows) {
+  JsonArray arr;
+  for (auto& row : rows) {
+    long long req = row[1].i, fail = row[2].i;
+    JsonObject o;
+    o.emplace_back("model_id", Json(row[0].str()));
+    o.emplace_back("total_requests", Json(req));
+    o.emplace_back("successful_requests", Json(req - fail));
+    o.emplace_back("failed_requests", Json(fail));
+    o.emplace_back("total_output_tokens", Json((long long)row[3].i));
+    o.emplace_back(
+        "total_duration_ms",
+        Json(row[4].kind == DbValue::kReal ? row[4].d : (double)row[4].i));
+    arr.emplace_back(Json(std::move(o)));
+  }
+  return Json(std::move(arr));
+}
+
 static EndpointPtr pick_by_capability(Registry& reg, uint32_t cap) {
   for (auto& ep : reg.list()) {
     if (ep->status.load() != EndpointStatus::kOnline) continue;
@@ -385,6 +405,67 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
                                : "{\"synced\":false,\"error\":\"" + err +
                                      "\"}";
         c->send_simple(ok ? 200 : 502, "application/json", bodyj);
+        return;
+      }
+      // -- per-endpoint request statistics (reference
+      //    api/dashboard.rs:919-976, db/endpoint_daily_stats.rs) ----
+      if (action == "today" && method == "GET") {
+        char date[16];
+        time_t t = time(nullptr);
+        struct tm tmv;
+        gmtime_r(&t, &tmv);
+        strftime(date, sizeof date, "%Y-%m-%d", &tmv);
+        std::vector<DbRow> rows;
+        db.exec("SELECT COALESCE(SUM(requests),0),"
+                "COALESCE(SUM(failures),0) FROM endpoint_daily_stats "
+                "WHERE endpoint_id=? AND date=?",
+                {id, date}, &rows);
+        long long req = rows.empty() ? 0 : rows[0][0].i;
+        long long fail = rows.empty() ? 0 : rows[0][1].i;
+        JsonObject o;
+        o.emplace_back("date", Json(std::string(date)));
+        o.emplace_back("total_requests", Json(req));
+        o.emplace_back("successful_requests", Json(req - fail));
+        o.emplace_back("failed_requests", Json(fail));
+        c->send_simple(200, "application/json",
+                       Json(std::move(o)).dump());
+        return;
+      }
+      if (action == "daily-stats" && method == "GET") {
+        long long days = 7;
+        std::string q(r.query);
+        size_t p = q.find("days=");
+        if (p != std::string::npos) days = atoll(q.c_str() + p + 5);
+        if (days < 1) days = 1;
+        if (days > 365) days = 365;
+        std::vector<DbRow> rows;
+        db.exec("SELECT date, SUM(requests), SUM(failures) FROM "
+                "endpoint_daily_stats WHERE endpoint_id=? GROUP BY "
+                "date ORDER BY date DESC LIMIT ?",
+                {id, days}, &rows);
+        JsonArray arr;
+        for (auto& row : rows) {
+          long long req = row[1].i, fail = row[2].i;
+          JsonObject o;
+          o.emplace_back("date", Json(row[0].str()));
+          o.emplace_back("total_requests", Json(req));
+          o.emplace_back("successful_requests", Json(req - fail));
+          o.emplace_back("failed_requests", Json(fail));
+          arr.emplace_back(Json(std::move(o)));
+        }
+        c->send_simple(200, "application/json",
+                       Json(std::move(arr)).dump());
+        return;
+      }
+      if (action == "model-stats" && method == "GET") {
+        std::vector<DbRow> rows;
+        db.exec("SELECT model, SUM(requests), SUM(failures), "
+                "SUM(completion_tokens), SUM(total_duration_ms) FROM "
+                "endpoint_daily_stats WHERE endpoint_id=? GROUP BY "
+                "model ORDER BY SUM(requests) DESC",
+                {id}, &rows);
+        c->send_simple(200, "application/json",
+                       model_stats_json(rows).dump());
         return;
       }
       if (action == "chat" && method == "POST") {
@@ -593,6 +674,19 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
         arr.emplace_back(Json(std::move(o)));
       }
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
+    if (path == "/api/dashboard/all-model-stats" && method == "GET") {
+      // cross-endpoint per-model rollup (reference
+      // api/dashboard.rs:969-976 get_all_model_stats)
+      std::vector<DbRow> rows;
+      db.exec("SELECT model, SUM(requests), SUM(failures), "
+              "SUM(completion_tokens), SUM(total_duration_ms) FROM "
+              "endpoint_daily_stats GROUP BY model ORDER BY "
+              "SUM(requests) DESC",
+              {}, &rows);
+      c->send_simple(200, "application/json",
+                     model_stats_json(rows).dump());
       return;
     }
     if (path == "/api/dashboard/daily-stats" && method == "GET") {

@@ -120,6 +120,7 @@ CASES = [
     ("GET", "/api/dashboard/summary", None, ADMIN, 200,
      {"endpoints", "active_requests", "last_hour"}),
     ("GET", "/api/dashboard/daily-stats", None, ADMIN, 200, None),
+    ("GET", "/api/dashboard/all-model-stats", None, ADMIN, 200, None),
     ("GET", "/api/dashboard/clients", None, ADMIN, 200, None),
     ("GET", "/api/dashboard/clients/heatmap", None, ADMIN, 200,
      {"requests"}),

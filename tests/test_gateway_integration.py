@@ -979,3 +979,47 @@ def test_pipelined_request_does_not_dangle_head_views(stack):
         assert st == 200
     finally:
         slow.terminate()
+
+
+def test_per_endpoint_stats_routes(stack):
+    """Per-endpoint statistics family (reference api/dashboard.rs:919-976):
+    today / daily-stats / model-stats on one endpoint, plus the
+    cross-endpoint /api/dashboard/all-model-stats rollup."""
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "mock-model", "messages": []}, headers=stack["auth"])
+    time.sleep(0.6)  # batched daily-stats flush
+    status, body = _req("GET", stack["base"] + "/api/endpoints",
+                        headers=stack["admin"])
+    eid = json.loads(body)[0]["id"]
+    base = stack["base"] + "/api/endpoints/" + eid
+
+    status, body = _req("GET", base + "/today", headers=stack["admin"])
+    assert status == 200
+    today = json.loads(body)
+    assert today["total_requests"] >= 1
+    assert today["total_requests"] == (today["successful_requests"]
+                                       + today["failed_requests"])
+
+    status, body = _req("GET", base + "/daily-stats?days=3",
+                        headers=stack["admin"])
+    assert status == 200
+    days = json.loads(body)
+    assert days and days[0]["date"] == today["date"]
+    assert days[0]["total_requests"] >= 1
+
+    status, body = _req("GET", base + "/model-stats",
+                        headers=stack["admin"])
+    assert status == 200
+    ms = json.loads(body)
+    row = next(m for m in ms if m["model_id"] == "mock-model")
+    assert row["total_requests"] >= 1
+    assert "total_output_tokens" in row and "total_duration_ms" in row
+
+    status, body = _req("GET",
+                        stack["base"] + "/api/dashboard/all-model-stats",
+                        headers=stack["admin"])
+    assert status == 200
+    allms = json.loads(body)
+    assert any(m["model_id"] == "mock-model"
+               and m["total_requests"] >= row["total_requests"]
+               for m in allms)
