@@ -468,6 +468,42 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
                        model_stats_json(rows).dump());
         return;
       }
+      if (action == "model-tps" && method == "GET") {
+        // live per-model TPS EMAs for this endpoint enriched with
+        // persisted totals (reference api/dashboard.rs:1206-1225;
+        // source is Production — benchmark-sourced TPS lives under
+        // /api/benchmarks/tps runs)
+        auto list = registry.model_tps_list(id);
+        JsonArray arr;
+        for (auto& e : list) {
+          std::vector<DbRow> rows;
+          db.exec("SELECT COALESCE(SUM(completion_tokens),0),"
+                  "COALESCE(SUM(total_duration_ms),0),"
+                  "COALESCE(SUM(requests),0) FROM endpoint_daily_stats "
+                  "WHERE endpoint_id=? AND model=?",
+                  {id, e.model}, &rows);
+          long long toks = rows.empty() ? 0 : rows[0][0].i;
+          double dur = rows.empty()
+                           ? 0.0
+                           : (rows[0][1].kind == DbValue::kReal
+                                  ? rows[0][1].d
+                                  : (double)rows[0][1].i);
+          long long reqs = rows.empty() ? 0 : rows[0][2].i;
+          JsonObject o;
+          o.emplace_back("model_id", Json(e.model));
+          o.emplace_back("api_kind", Json(e.kind));
+          o.emplace_back("source", Json(std::string("Production")));
+          o.emplace_back("tps", Json(e.tps));
+          o.emplace_back("request_count", Json(e.samples));
+          o.emplace_back("total_output_tokens", Json(toks));
+          o.emplace_back("average_duration_ms",
+                         Json(reqs > 0 ? dur / reqs : 0.0));
+          arr.emplace_back(Json(std::move(o)));
+        }
+        c->send_simple(200, "application/json",
+                       Json(std::move(arr)).dump());
+        return;
+      }
       if (action == "chat" && method == "POST") {
         // playground: chat through THIS endpoint, bypassing selection
         // (reference api/endpoints.rs:1079 proxy_chat_completions)

@@ -334,6 +334,24 @@ double Registry::tps_score(const std::string& endpoint_id,
   return tps_state(endpoint_id, model, kind)->score(wall_ms());
 }
 
+std::vector<Registry::ModelTps> Registry::model_tps_list(
+    const std::string& endpoint_id) {
+  std::vector<ModelTps> out;
+  std::lock_guard<std::mutex> lk(tps_mu_);
+  const std::string prefix = endpoint_id + "|";
+  for (auto& [key, st] : tps_) {
+    if (!starts_with(key, prefix)) continue;
+    std::string rest = key.substr(prefix.size());   // "model|kind"
+    size_t bar = rest.rfind('|');
+    if (bar == std::string::npos) continue;
+    std::lock_guard<std::mutex> slk(st->mu);
+    if (st->samples == 0) continue;
+    out.push_back({rest.substr(0, bar), rest.substr(bar + 1), st->ema,
+                   st->samples});
+  }
+  return out;
+}
+
 void Registry::clear_tps_for_endpoint(const std::string& endpoint_id) {
   std::lock_guard<std::mutex> lk(tps_mu_);
   for (auto& [key, st] : tps_) {

@@ -197,6 +197,14 @@ class Registry {
   void record_tps(const std::string& endpoint_id, const std::string& model,
                   ApiKind kind, double tokens, double duration_s);
   void clear_tps_for_endpoint(const std::string& endpoint_id);
+  // per-model TPS snapshot for one endpoint (dashboard.rs:1206-1225):
+  // (model, kind, ema tokens/s, samples), live entries only
+  struct ModelTps {
+    std::string model, kind;
+    double tps;
+    long long samples;
+  };
+  std::vector<ModelTps> model_tps_list(const std::string& endpoint_id);
   // boot reseed from endpoint_daily_stats (today's rows)
   void seed_tps_from_db();
   double tps_score(const std::string& endpoint_id, const std::string& model,

@@ -1023,3 +1023,24 @@ def test_per_endpoint_stats_routes(stack):
     assert any(m["model_id"] == "mock-model"
                and m["total_requests"] >= row["total_requests"]
                for m in allms)
+
+
+def test_endpoint_model_tps_route(stack):
+    """Live per-model TPS EMAs for one endpoint
+    (reference api/dashboard.rs:1206-1225)."""
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "mock-model", "messages": []}, headers=stack["auth"])
+    time.sleep(0.6)
+    status, body = _req("GET", stack["base"] + "/api/endpoints",
+                        headers=stack["admin"])
+    eid = json.loads(body)[0]["id"]
+    status, body = _req("GET",
+                        stack["base"] + "/api/endpoints/" + eid
+                        + "/model-tps", headers=stack["admin"])
+    assert status == 200
+    entries = json.loads(body)
+    row = next(e for e in entries if e["model_id"] == "mock-model")
+    assert row["tps"] > 0 and row["request_count"] >= 1
+    assert row["source"] == "Production"
+    assert row["api_kind"] == "chat_completions"
+    assert "average_duration_ms" in row
