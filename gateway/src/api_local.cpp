@@ -712,6 +712,36 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }
+    if (starts_with(path, "/api/dashboard/metrics/") && method == "GET") {
+      // per-endpoint health-metrics history (reference
+      // api/dashboard.rs:205-211 get_node_metrics): the health loop's
+      // persisted probe results, newest first
+      std::string eid =
+          url_decode(path.substr(strlen("/api/dashboard/metrics/")));
+      std::vector<DbRow> rows;
+      db.exec("SELECT ts_ms, ok, latency_ms, detail FROM health_checks "
+              "WHERE endpoint_id=? ORDER BY ts_ms DESC LIMIT 500",
+              {eid}, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        JsonObject o;
+        o.emplace_back("ts_ms", Json((long long)row[0].i));
+        o.emplace_back("ok", Json(row[1].i != 0));
+        o.emplace_back("latency_ms",
+                       Json(row[2].kind == DbValue::kReal
+                                ? row[2].d
+                                : (double)row[2].i));
+        std::string detail = row[3].str();
+        Json dj;
+        if (Json::parse(detail, dj))
+          o.emplace_back("detail", std::move(dj));
+        else if (!detail.empty())
+          o.emplace_back("detail", Json(detail));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
     if (path == "/api/dashboard/all-model-stats" && method == "GET") {
       // cross-endpoint per-model rollup (reference
       // api/dashboard.rs:969-976 get_all_model_stats)

@@ -1044,3 +1044,23 @@ def test_endpoint_model_tps_route(stack):
     assert row["source"] == "Production"
     assert row["api_kind"] == "chat_completions"
     assert "average_duration_ms" in row
+
+
+def test_node_metrics_history_route(stack):
+    """Per-endpoint health-metrics history (reference
+    api/dashboard.rs:205-211): the health loop's persisted probes."""
+    status, body = _req("GET", stack["base"] + "/api/endpoints",
+                        headers=stack["admin"])
+    eid = json.loads(body)[0]["id"]
+    url = stack["base"] + "/api/dashboard/metrics/" + eid
+    deadline = time.time() + 8
+    while time.time() < deadline:
+        status, body = _req("GET", url, headers=stack["admin"])
+        assert status == 200
+        rows = json.loads(body)
+        if rows:
+            assert rows[0]["ok"] is True
+            assert rows[0]["latency_ms"] >= 0
+            return
+        time.sleep(0.5)
+    raise AssertionError("no health_checks rows recorded")
