@@ -712,6 +712,51 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }
+    if (path == "/api/dashboard/stats/tokens" && method == "GET") {
+      // all-time token totals (reference api/dashboard.rs:216-225,
+      // db/request_history.rs:946-953 TokenStatistics)
+      std::vector<DbRow> rows;
+      db.exec("SELECT COALESCE(SUM(prompt_tokens),0),"
+              "COALESCE(SUM(completion_tokens),0) FROM "
+              "endpoint_daily_stats",
+              {}, &rows);
+      long long in_t = rows.empty() ? 0 : rows[0][0].i;
+      long long out_t = rows.empty() ? 0 : rows[0][1].i;
+      JsonObject o;
+      o.emplace_back("total_input_tokens", Json(in_t));
+      o.emplace_back("total_output_tokens", Json(out_t));
+      o.emplace_back("total_tokens", Json(in_t + out_t));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
+    if (path == "/api/dashboard/stats/tokens/daily" && method == "GET") {
+      // daily token totals (reference api/dashboard.rs:257-309
+      // DailyTokenStats, default 30 days)
+      long long days = 30;
+      std::string q(r.query);
+      size_t p = q.find("days=");
+      if (p != std::string::npos) days = atoll(q.c_str() + p + 5);
+      if (days < 1) days = 1;
+      if (days > 365) days = 365;
+      std::vector<DbRow> rows;
+      db.exec("SELECT date, SUM(prompt_tokens), SUM(completion_tokens),"
+              "SUM(requests) FROM endpoint_daily_stats GROUP BY date "
+              "ORDER BY date DESC LIMIT ?",
+              {days}, &rows);
+      JsonArray arr;
+      for (auto& row : rows) {
+        long long in_t = row[1].i, out_t = row[2].i;
+        JsonObject o;
+        o.emplace_back("date", Json(row[0].str()));
+        o.emplace_back("total_input_tokens", Json(in_t));
+        o.emplace_back("total_output_tokens", Json(out_t));
+        o.emplace_back("total_tokens", Json(in_t + out_t));
+        o.emplace_back("request_count", Json((long long)row[3].i));
+        arr.emplace_back(Json(std::move(o)));
+      }
+      c->send_simple(200, "application/json", Json(std::move(arr)).dump());
+      return;
+    }
     if (starts_with(path, "/api/dashboard/metrics/") && method == "GET") {
       // per-endpoint health-metrics history (reference
       // api/dashboard.rs:205-211 get_node_metrics): the health loop's

@@ -121,6 +121,10 @@ CASES = [
      {"endpoints", "active_requests", "last_hour"}),
     ("GET", "/api/dashboard/daily-stats", None, ADMIN, 200, None),
     ("GET", "/api/dashboard/all-model-stats", None, ADMIN, 200, None),
+    ("GET", "/api/dashboard/stats/tokens", None, ADMIN, 200,
+     {"total_input_tokens", "total_output_tokens", "total_tokens"}),
+    ("GET", "/api/dashboard/stats/tokens/daily?days=7", None, ADMIN,
+     200, None),
     ("GET", "/api/dashboard/clients", None, ADMIN, 200, None),
     ("GET", "/api/dashboard/clients/heatmap", None, ADMIN, 200,
      {"requests"}),
