@@ -690,6 +690,45 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", render_summary_json());
       return;
     }
+    if (path == "/api/dashboard/overview" && method == "GET") {
+      // one-call dashboard bootstrap (reference api/dashboard.rs:186-202
+      // get_overview): summary + recent history + per-endpoint model
+      // TPS, stamped with generation time
+      int64_t t0 = now_us();
+      Json summary;
+      Json::parse(render_summary_json(), summary);
+      JsonArray hist;
+      for (auto& rec : history.ring().snapshot(100)) {
+        JsonObject o;
+        o.emplace_back("ts_ms", Json((long long)rec.ts_ms));
+        o.emplace_back("endpoint_id", Json(rec.endpoint_id));
+        o.emplace_back("model", Json(rec.model));
+        o.emplace_back("status", Json((long long)rec.status));
+        o.emplace_back("duration_ms", Json(rec.duration_ms));
+        hist.emplace_back(Json(std::move(o)));
+      }
+      JsonArray tps;
+      for (auto& ep : registry.list()) {
+        for (auto& e : registry.model_tps_list(ep->id)) {
+          JsonObject o;
+          o.emplace_back("endpoint_id", Json(ep->id));
+          o.emplace_back("model_id", Json(e.model));
+          o.emplace_back("api_kind", Json(e.kind));
+          o.emplace_back("tps", Json(e.tps));
+          o.emplace_back("request_count", Json(e.samples));
+          tps.emplace_back(Json(std::move(o)));
+        }
+      }
+      JsonObject out;
+      out.emplace_back("summary", std::move(summary));
+      out.emplace_back("history", Json(std::move(hist)));
+      out.emplace_back("endpoint_tps", Json(std::move(tps)));
+      out.emplace_back("generated_at", Json(iso8601_now()));
+      out.emplace_back("generation_time_ms",
+                       Json((double)(now_us() - t0) / 1000.0));
+      c->send_simple(200, "application/json", Json(std::move(out)).dump());
+      return;
+    }
     if (path == "/api/requests" && method == "GET") {
       auto recs = history.ring().snapshot(500);
       JsonArray arr;

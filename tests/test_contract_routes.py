@@ -120,6 +120,9 @@ CASES = [
     ("GET", "/api/dashboard/summary", None, ADMIN, 200,
      {"endpoints", "active_requests", "last_hour"}),
     ("GET", "/api/dashboard/daily-stats", None, ADMIN, 200, None),
+    ("GET", "/api/dashboard/overview", None, ADMIN, 200,
+     {"summary", "history", "endpoint_tps", "generated_at",
+      "generation_time_ms"}),
     ("GET", "/api/dashboard/all-model-stats", None, ADMIN, 200, None),
     ("GET", "/api/dashboard/stats/tokens", None, ADMIN, 200,
      {"total_input_tokens", "total_output_tokens", "total_tokens"}),
