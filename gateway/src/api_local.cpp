@@ -751,6 +751,44 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }
+    if (starts_with(path, "/api/requests/") && method == "GET") {
+      // request-history detail incl. captured bodies (reference
+      // api/dashboard.rs:692-705 get_request_response_detail);
+      // accepts the numeric row id or the X-Request-Id value
+      std::string rid = url_decode(path.substr(strlen("/api/requests/")));
+      std::vector<DbRow> rows;
+      db.exec("SELECT id,ts_ms,endpoint_id,model,api_kind,status,"
+              "duration_ms,prompt_tokens,completion_tokens,client_ip,"
+              "error,request_body,response_body,request_id FROM "
+              "request_history WHERE id=? OR request_id=? LIMIT 1",
+              {rid, rid}, &rows);
+      if (rows.empty()) {
+        c->send_json_error(404, "not_found", "no such request record");
+        return;
+      }
+      auto& row = rows[0];
+      JsonObject o;
+      o.emplace_back("id", Json((long long)row[0].i));
+      o.emplace_back("ts_ms", Json((long long)row[1].i));
+      o.emplace_back("endpoint_id", Json(row[2].str()));
+      o.emplace_back("model", Json(row[3].str()));
+      o.emplace_back("api_kind", Json(row[4].str()));
+      o.emplace_back("status", Json((long long)row[5].i));
+      o.emplace_back("duration_ms",
+                     Json(row[6].kind == DbValue::kReal ? row[6].d
+                                                        : (double)row[6].i));
+      o.emplace_back("prompt_tokens", Json((long long)row[7].i));
+      o.emplace_back("completion_tokens", Json((long long)row[8].i));
+      o.emplace_back("client_ip", Json(row[9].str()));
+      if (!row[10].str().empty())
+        o.emplace_back("error", Json(row[10].str()));
+      o.emplace_back("request_body", Json(row[11].str()));
+      o.emplace_back("response_body", Json(row[12].str()));
+      if (!row[13].str().empty())
+        o.emplace_back("request_id", Json(row[13].str()));
+      c->send_simple(200, "application/json", Json(std::move(o)).dump());
+      return;
+    }
     if (path == "/api/dashboard/stats/tokens" && method == "GET") {
       // all-time token totals (reference api/dashboard.rs:216-225,
       // db/request_history.rs:946-953 TokenStatistics)

@@ -1064,3 +1064,29 @@ def test_node_metrics_history_route(stack):
             return
         time.sleep(0.5)
     raise AssertionError("no health_checks rows recorded")
+
+
+def test_request_detail_route(stack):
+    """Request-history detail by row id or X-Request-Id (reference
+    api/dashboard.rs:692-705)."""
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "mock-model", "messages": []},
+         headers={**stack["auth"], "X-Request-Id": "detail-route-1"})
+    time.sleep(0.6)
+    status, body = _req("GET",
+                        stack["base"] + "/api/requests/detail-route-1",
+                        headers=stack["admin"])
+    assert status == 200
+    rec = json.loads(body)
+    assert rec["request_id"] == "detail-route-1"
+    assert rec["model"] == "mock-model" and rec["status"] == 200
+    # numeric row id resolves to the same record
+    status, body = _req("GET",
+                        stack["base"] + "/api/requests/" + str(rec["id"]),
+                        headers=stack["admin"])
+    assert status == 200
+    assert json.loads(body)["request_id"] == "detail-route-1"
+    # unknown id -> 404
+    status, _ = _req("GET", stack["base"] + "/api/requests/nope-404",
+                     headers=stack["admin"])
+    assert status == 404
