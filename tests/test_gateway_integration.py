@@ -981,6 +981,15 @@ def test_pipelined_request_does_not_dangle_head_views(stack):
         slow.terminate()
 
 
+def _mock1_id(stack):
+    _, body = _req("GET", stack["base"] + "/api/endpoints",
+                   headers=stack["admin"])
+    eps = json.loads(body)
+    # the module-scoped stack accumulates endpoints from earlier
+    # tests; the traffic-bearing one is the original "mock1"
+    return next(e["id"] for e in eps if e["name"] == "mock1")
+
+
 def test_per_endpoint_stats_routes(stack):
     """Per-endpoint statistics family (reference api/dashboard.rs:919-976):
     today / daily-stats / model-stats on one endpoint, plus the
@@ -988,10 +997,7 @@ def test_per_endpoint_stats_routes(stack):
     _req("POST", stack["base"] + "/v1/chat/completions",
          {"model": "mock-model", "messages": []}, headers=stack["auth"])
     time.sleep(0.6)  # batched daily-stats flush
-    status, body = _req("GET", stack["base"] + "/api/endpoints",
-                        headers=stack["admin"])
-    eid = json.loads(body)[0]["id"]
-    base = stack["base"] + "/api/endpoints/" + eid
+    base = stack["base"] + "/api/endpoints/" + _mock1_id(stack)
 
     status, body = _req("GET", base + "/today", headers=stack["admin"])
     assert status == 200
@@ -1031,12 +1037,10 @@ def test_endpoint_model_tps_route(stack):
     _req("POST", stack["base"] + "/v1/chat/completions",
          {"model": "mock-model", "messages": []}, headers=stack["auth"])
     time.sleep(0.6)
-    status, body = _req("GET", stack["base"] + "/api/endpoints",
-                        headers=stack["admin"])
-    eid = json.loads(body)[0]["id"]
     status, body = _req("GET",
-                        stack["base"] + "/api/endpoints/" + eid
-                        + "/model-tps", headers=stack["admin"])
+                        stack["base"] + "/api/endpoints/"
+                        + _mock1_id(stack) + "/model-tps",
+                        headers=stack["admin"])
     assert status == 200
     entries = json.loads(body)
     row = next(e for e in entries if e["model_id"] == "mock-model")
@@ -1049,10 +1053,8 @@ def test_endpoint_model_tps_route(stack):
 def test_node_metrics_history_route(stack):
     """Per-endpoint health-metrics history (reference
     api/dashboard.rs:205-211): the health loop's persisted probes."""
-    status, body = _req("GET", stack["base"] + "/api/endpoints",
-                        headers=stack["admin"])
-    eid = json.loads(body)[0]["id"]
-    url = stack["base"] + "/api/dashboard/metrics/" + eid
+    url = (stack["base"] + "/api/dashboard/metrics/"
+           + _mock1_id(stack))
     deadline = time.time() + 8
     while time.time() < deadline:
         status, body = _req("GET", url, headers=stack["admin"])
