@@ -730,9 +730,36 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
       return;
     }
     if (path == "/api/requests" && method == "GET") {
+      // optional filters (reference api/dashboard.rs:661-689
+      // list_request_responses / RequestHistoryQuery): model=, status=
+      // (exact or "4xx"/"5xx" class), limit=
+      std::string q(r.query);
+      auto qget = [&](const char* k) -> std::string {
+        std::string needle = std::string(k) + "=";
+        size_t p = q.find(needle);
+        if (p != 0 && (p == std::string::npos || q[p - 1] != '&'))
+          return "";
+        size_t v = p + needle.size();
+        return url_decode(q.substr(v, q.find('&', v) - v));
+      };
+      std::string f_model = qget("model"), f_status = qget("status");
+      long long limit = 500;
+      if (!qget("limit").empty()) limit = atoll(qget("limit").c_str());
+      if (limit < 1) limit = 1;
+      if (limit > 500) limit = 500;
       auto recs = history.ring().snapshot(500);
       JsonArray arr;
       for (auto& rec : recs) {
+        if (!f_model.empty() && rec.model != f_model) continue;
+        if (!f_status.empty()) {
+          if (f_status == "4xx") {
+            if (rec.status < 400 || rec.status > 499) continue;
+          } else if (f_status == "5xx") {
+            if (rec.status < 500 || rec.status > 599) continue;
+          } else if (rec.status != atoi(f_status.c_str())) {
+            continue;
+          }
+        }
         JsonObject o;
         o.emplace_back("ts_ms", Json((long long)rec.ts_ms));
         o.emplace_back("endpoint_id", Json(rec.endpoint_id));
@@ -748,6 +775,9 @@ void Gateway::handle_local(ClientConn* c, const std::string& path,
         if (!rec.error.empty()) o.emplace_back("error", Json(rec.error));
         arr.emplace_back(Json(std::move(o)));
       }
+      // snapshot is oldest-first: a limit keeps the NEWEST records
+      if ((long long)arr.size() > limit)
+        arr.erase(arr.begin(), arr.end() - limit);
       c->send_simple(200, "application/json", Json(std::move(arr)).dump());
       return;
     }

@@ -1092,3 +1092,40 @@ def test_request_detail_route(stack):
     status, _ = _req("GET", stack["base"] + "/api/requests/nope-404",
                      headers=stack["admin"])
     assert status == 404
+
+
+def test_requests_list_filters(stack):
+    """Request-history list filters model=/status=/limit= (reference
+    api/dashboard.rs:661-689 RequestHistoryQuery)."""
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "mock-model", "messages": []}, headers=stack["auth"])
+    _req("POST", stack["base"] + "/v1/chat/completions",
+         {"model": "ghost-model", "messages": []}, headers=stack["auth"])
+    time.sleep(0.3)
+    status, body = _req("GET",
+                        stack["base"] + "/api/requests?model=mock-model",
+                        headers=stack["admin"])
+    assert status == 200
+    recs = json.loads(body)
+    assert recs and all(r["model"] == "mock-model" for r in recs)
+
+    status, body = _req("GET", stack["base"] + "/api/requests?status=200",
+                        headers=stack["admin"])
+    recs = json.loads(body)
+    assert recs and all(r["status"] == 200 for r in recs)
+    status, body = _req("GET",
+                        stack["base"] + "/api/requests?model=no-such",
+                        headers=stack["admin"])
+    assert json.loads(body) == []
+
+    status, body = _req("GET",
+                        stack["base"]
+                        + "/api/requests?model=mock-model&limit=1",
+                        headers=stack["admin"])
+    recs = json.loads(body)
+    assert len(recs) == 1 and recs[0]["model"] == "mock-model"
+    # the newest matching record is kept, not the oldest
+    status, body = _req("GET",
+                        stack["base"] + "/api/requests?model=mock-model",
+                        headers=stack["admin"])
+    assert recs[0]["ts_ms"] == json.loads(body)[-1]["ts_ms"]
