@@ -119,6 +119,13 @@ class UpdateBody(BaseModel):
     target_path: str = ""
 
 
+class ScheduleBody(BaseModel):
+    artifact_path: str
+    target_path: str = ""
+    at_epoch: float | None = None
+    delay_s: float | None = None
+
+
 class TpsBenchmarkBody(BaseModel):
     # reference api/benchmarks.rs:25-34 defaults and caps
     model: str
@@ -719,14 +726,16 @@ def create_app(db_path: str | None = None,
     def update_status(user=Depends(require_admin)):
         return update_state
 
-    @app.post("/api/update/apply", status_code=202)
-    async def update_apply(body: UpdateBody, user=Depends(require_admin)):
-        import shutil
-        target = body.target_path or os.path.join(
+    def _resolve_target(target_path: str) -> str:
+        return target_path or os.path.join(
             os.path.dirname(os.path.dirname(os.path.dirname(
                 os.path.abspath(__file__)))), "gateway", "bin", "llmlb-gw")
-        if not os.path.exists(body.artifact_path):
-            raise HTTPException(400, "artifact not found")
+
+    def _start_apply(artifact_path: str, target: str) -> None:
+        """Drain -> swap binary (+.bak) -> resume; shared by the
+        immediate apply and the scheduler (reference
+        api/system.rs:129-150, 218-330)."""
+        import shutil
         update_state.update({"state": "draining"})
 
         def drain_and_apply():
@@ -754,7 +763,7 @@ def create_app(db_path: str | None = None,
                 bak = target + ".bak"
                 if os.path.exists(target):
                     shutil.copy2(target, bak)
-                shutil.copy2(body.artifact_path, target)
+                shutil.copy2(artifact_path, target)
                 os.chmod(target, 0o755)
                 update_state.update({"state": "applied",
                                      "rollback": bak})
@@ -775,7 +784,62 @@ def create_app(db_path: str | None = None,
 
         import threading as _threading
         _threading.Thread(target=drain_and_apply, daemon=True).start()
+
+    @app.post("/api/update/apply", status_code=202)
+    async def update_apply(body: UpdateBody, user=Depends(require_admin)):
+        if not os.path.exists(body.artifact_path):
+            raise HTTPException(400, "artifact not found")
+        _start_apply(body.artifact_path, _resolve_target(body.target_path))
         return {"accepted": True}
+
+    # scheduled update (reference api/system.rs:218-330): one pending
+    # schedule at a time; a timer fires the same drain->apply path
+    update_schedule: dict = {}
+
+    @app.post("/api/update/schedule", status_code=202)
+    def update_schedule_create(body: ScheduleBody,
+                               user=Depends(require_admin)):
+        import threading as _threading
+        if update_schedule.get("timer") is not None:
+            raise HTTPException(409, "a schedule already exists")
+        if not os.path.exists(body.artifact_path):
+            raise HTTPException(400, "artifact not found")
+        delay = (body.at_epoch - time.time()) if body.at_epoch             else body.delay_s
+        if delay is None or delay < 0:
+            raise HTTPException(400, "schedule time is in the past")
+        target = _resolve_target(body.target_path)
+
+        def fire():
+            update_schedule.clear()
+            _start_apply(body.artifact_path, target)
+
+        t = _threading.Timer(delay, fire)
+        t.daemon = True
+        update_schedule.update({"timer": t,
+                                "artifact_path": body.artifact_path,
+                                "target_path": target,
+                                "scheduled_at": time.time() + delay})
+        t.start()
+        return {"scheduled": True,
+                "scheduled_at": update_schedule["scheduled_at"]}
+
+    @app.get("/api/update/schedule")
+    def update_schedule_get(user=Depends(require_admin)):
+        if update_schedule.get("timer") is None:
+            return {"scheduled": False}
+        return {"scheduled": True,
+                "artifact_path": update_schedule["artifact_path"],
+                "target_path": update_schedule["target_path"],
+                "scheduled_at": update_schedule["scheduled_at"]}
+
+    @app.delete("/api/update/schedule")
+    def update_schedule_cancel(user=Depends(require_admin)):
+        t = update_schedule.get("timer")
+        if t is None:
+            raise HTTPException(404, "no schedule")
+        t.cancel()
+        update_schedule.clear()
+        return {"cancelled": True}
 
     @app.post("/api/update/rollback")
     def update_rollback(user=Depends(require_admin)):

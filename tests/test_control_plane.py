@@ -586,3 +586,92 @@ def test_cp_route_contract(app_client, method, path, body, authed,
     assert r.status_code == want, (r.status_code, r.text[:200])
     if want == 200 and keys:
         assert keys <= set(r.json()), r.text[:200]
+
+
+def test_scheduled_update(tmp_path):
+    """Scheduled self-update (reference api/system.rs:218-330): create
+    fires the drain->apply path at the scheduled time; get reflects
+    the pending schedule; cancel stops it; one schedule at a time."""
+    import http.server
+    import socket as sk
+    import threading as th
+
+    class StubGW(http.server.BaseHTTPRequestHandler):
+        def log_message(self, *a):
+            pass
+
+        def do_POST(self):
+            self.send_response(200)
+            self.send_header("Content-Length", "2")
+            self.end_headers()
+            self.wfile.write(b"{}")
+
+        def do_GET(self):
+            body = b"llmlb_active_requests 0\n"
+            self.send_response(200)
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+    s = sk.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    httpd = http.server.ThreadingHTTPServer(("127.0.0.1", port), StubGW)
+    th.Thread(target=httpd.serve_forever, daemon=True).start()
+
+    from llmlb_amd.api import control_plane as cp
+    from fastapi.testclient import TestClient
+    app = cp.create_app(db_path=":memory:",
+                        gateway_url=f"http://127.0.0.1:{port}",
+                        jwt_secret="s", admin_token="admtok")
+    client = TestClient(app)
+    hdrs = {"Authorization": "Bearer admtok"}
+
+    old = tmp_path / "llmlb-gw"
+    old.write_bytes(b"OLD BINARY")
+    new = tmp_path / "artifact"
+    new.write_bytes(b"NEW BINARY")
+
+    # cancel without a schedule -> 404; past time -> 400
+    assert client.delete("/api/update/schedule",
+                         headers=hdrs).status_code == 404
+    r = client.post("/api/update/schedule", headers=hdrs,
+                    json={"artifact_path": str(new),
+                          "target_path": str(old),
+                          "at_epoch": time.time() - 10})
+    assert r.status_code == 400
+
+    # schedule, observe pending, duplicate -> 409, cancel works
+    r = client.post("/api/update/schedule", headers=hdrs,
+                    json={"artifact_path": str(new),
+                          "target_path": str(old), "delay_s": 30})
+    assert r.status_code == 202
+    st = client.get("/api/update/schedule", headers=hdrs).json()
+    assert st["scheduled"] and st["artifact_path"] == str(new)
+    assert client.post("/api/update/schedule", headers=hdrs,
+                       json={"artifact_path": str(new),
+                             "delay_s": 30}).status_code == 409
+    assert client.delete("/api/update/schedule",
+                         headers=hdrs).status_code == 200
+    assert client.get("/api/update/schedule",
+                      headers=hdrs).json() == {"scheduled": False}
+    assert old.read_bytes() == b"OLD BINARY"  # never fired
+
+    # short schedule actually applies
+    r = client.post("/api/update/schedule", headers=hdrs,
+                    json={"artifact_path": str(new),
+                          "target_path": str(old), "delay_s": 0.2})
+    assert r.status_code == 202
+    deadline = time.time() + 15
+    st = {}
+    while time.time() < deadline:
+        st = client.get("/api/update/status", headers=hdrs).json()
+        if st.get("state") in ("applied", "failed"):
+            break
+        time.sleep(0.2)
+    assert st.get("state") == "applied", st
+    assert old.read_bytes() == b"NEW BINARY"
+    assert client.get("/api/update/schedule",
+                      headers=hdrs).json() == {"scheduled": False}
+    httpd.shutdown()
